@@ -1,0 +1,92 @@
+"""Checkpoint layout — compatible with the reference's structure
+(SURVEY.md §3.4): ``path/metadata`` (params JSON), ``path/words`` (one word
+per line, line number == row index), plus binary shard files of both
+matrices (one per shard, concatenable by row interleave).
+
+Reference: save at mllib:493-498 / ml:499-560, load at mllib:683-725.
+"""
+from __future__ import annotations
+
+import json
+import os
+import time
+from typing import List, Tuple
+
+import numpy as np
+
+from .config import Word2VecConfig
+from .vocab import Vocabulary
+
+_DTYPES = {"float32": np.float32, "bfloat16": np.uint16}  # bf16 stored as raw u16
+
+
+def _shard_rows(vocab_size: int, shard: int, num_shards: int) -> np.ndarray:
+    """Row r lives in shard r % num_shards (round-robin row interleave —
+    balances Zipf-hot rows across shards/GPUs)."""
+    return np.arange(shard, vocab_size, num_shards, dtype=np.int64)
+
+
+def save_model(path: str, config: Word2VecConfig, vocab: Vocabulary,
+               syn0: np.ndarray, syn1: np.ndarray | None = None,
+               num_shards: int = 1) -> None:
+    os.makedirs(path, exist_ok=True)
+    os.makedirs(os.path.join(path, "shards"), exist_ok=True)
+    meta = {
+        "class": "glint_word2vec_amd.GlintWord2VecModel",
+        "timestamp": int(time.time() * 1000),
+        "numWords": vocab.num_words,
+        "vectorSize": int(syn0.shape[1]),
+        "paramMap": config.to_dict(),
+    }
+    with open(os.path.join(path, "metadata"), "w") as f:
+        json.dump(meta, f, indent=2, sort_keys=True)
+    vocab.save_words(os.path.join(path, "words"))
+    np.save(os.path.join(path, "counts.npy"), vocab.counts)
+    dt = syn0.dtype
+    index = {
+        "num_shards": num_shards,
+        "vocab": int(syn0.shape[0]),
+        "dim": int(syn0.shape[1]),
+        "dtype": str(dt),
+        "layout": "row_mod",
+        "has_syn1": syn1 is not None,
+    }
+    with open(os.path.join(path, "shards", "index.json"), "w") as f:
+        json.dump(index, f, indent=2)
+    for s in range(num_shards):
+        rows = _shard_rows(syn0.shape[0], s, num_shards)
+        syn0[rows].tofile(os.path.join(path, "shards", f"syn0-{s:05d}.bin"))
+        if syn1 is not None:
+            syn1[rows].tofile(os.path.join(path, "shards", f"syn1-{s:05d}.bin"))
+
+
+def load_model(path: str) -> Tuple[Word2VecConfig, Vocabulary, np.ndarray, np.ndarray | None]:
+    with open(os.path.join(path, "metadata")) as f:
+        meta = json.load(f)
+    config = Word2VecConfig.from_dict(meta["paramMap"])
+    counts_p = os.path.join(path, "counts.npy")
+    counts = np.load(counts_p) if os.path.exists(counts_p) else None
+    vocab = Vocabulary.load_words(os.path.join(path, "words"), counts)
+    with open(os.path.join(path, "shards", "index.json")) as f:
+        index = json.load(f)
+    V, D, k = index["vocab"], index["dim"], index["num_shards"]
+    dt = np.dtype(index["dtype"])
+    syn0 = np.empty((V, D), dtype=dt)
+    syn1 = np.empty((V, D), dtype=dt) if index.get("has_syn1") else None
+    for s in range(k):
+        rows = _shard_rows(V, s, k)
+        buf = np.fromfile(os.path.join(path, "shards", f"syn0-{s:05d}.bin"), dtype=dt)
+        syn0[rows] = buf.reshape(len(rows), D)
+        if syn1 is not None:
+            buf = np.fromfile(os.path.join(path, "shards", f"syn1-{s:05d}.bin"), dtype=dt)
+            syn1[rows] = buf.reshape(len(rows), D)
+    return config, vocab, syn0, syn1
+
+
+def save_word2vec_text(path: str, words: List[str], vectors: np.ndarray) -> None:
+    """Classic word2vec/gensim text format (the `toLocal` export analog,
+    mllib:651-659)."""
+    with open(path, "w", encoding="utf-8") as f:
+        f.write(f"{len(words)} {vectors.shape[1]}\n")
+        for w, v in zip(words, vectors):
+            f.write(w + " " + " ".join(f"{x:.6g}" for x in v) + "\n")

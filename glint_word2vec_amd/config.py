@@ -1,0 +1,104 @@
+"""Configuration surface for the MI355X-native Word2Vec trainer.
+
+Mirrors the full user-visible knob set of the reference
+(/root/reference/src/main/scala/org/apache/spark/mllib/feature/ServerSideGlintWord2Vec.scala:67-85,246-251
+and the ml.feature Params at ml:40-222), with the parameter-server/HOCON tier
+replaced by a device/communication section appropriate for a single 8xMI355X
+node.  Defaults are identical to the reference where the knob carries over.
+"""
+from __future__ import annotations
+
+import dataclasses
+import json
+from dataclasses import dataclass, field
+from typing import Optional
+
+# Constants carried over from the reference (mllib:246-251).
+MAX_EXP = 6.0          # sigmoid clip range [-6, 6] (mllib:248, 281-302)
+EXP_TABLE_SIZE = 1000  # only used by the optional LUT-parity sigmoid mode
+
+
+@dataclass
+class Word2VecConfig:
+    """All user knobs.  Names are snake_case; the estimator exposes the
+    reference's fluent ``setX`` aliases on top of this.
+
+    Reference defaults: vectorSize=100, stepSize=0.01875, numPartitions=1,
+    numIterations=1, minCount=5, maxSentenceLength=1000, window=5,
+    batchSize=50, n=5, subsampleRatio=1e-6, unigramTableSize=1e8
+    (mllib:67-85).
+    """
+
+    # --- model shape -------------------------------------------------------
+    vector_size: int = 100           # embedding dimension (mllib:67)
+    # --- optimisation ------------------------------------------------------
+    learning_rate: float = 0.01875   # stepSize (mllib:69)
+    num_iterations: int = 1          # epochs (mllib:73)
+    batch_size: int = 50             # positions per mini-batch (mllib:79)
+    window: int = 5                  # max context window (mllib:77)
+    n: int = 5                       # negatives per positive pair (mllib:81)
+    subsample_ratio: float = 1e-6    # frequency subsampling ratio (mllib:83)
+    seed: Optional[int] = None       # RNG seed (mllib:71); None = random
+    # --- vocabulary --------------------------------------------------------
+    min_count: int = 5               # (mllib:75)
+    max_sentence_length: int = 1000  # sentence chunk length (mllib:88-97)
+    unigram_table_size: int = 100_000_000  # negative-sampling table (mllib:85)
+    unigram_power: float = 0.75      # classic word2vec table exponent
+    # --- parallelism (replaces numPartitions / numParameterServers) --------
+    num_partitions: int = 1          # concurrent async workers per GPU
+    num_shards: Optional[int] = None  # row shards; None = world size
+    # --- device ------------------------------------------------------------
+    dtype: str = "float32"           # "float32" | "bfloat16" table storage
+    device: str = "auto"             # "auto" | "cpu" | "cuda"
+    words_per_step: int = 1 << 20    # tokens fed to the GPU per training step
+    atomic_updates: bool = False     # atomics vs hogwild plain RMW in kernel
+    # --- semantics switches ---------------------
+    # The reference's subsampling is a de-facto no-op (integer-division bug,
+    # mllib:375-377).  We implement the intended math; set
+    # legacy_subsample=True to reproduce the reference behaviour (keep all).
+    legacy_subsample: bool = False
+    # Reference windows are asymmetric b-left/(b-1)-right with possible empty
+    # context (mllib:385-387).  Default is the canonical symmetric shrunk
+    # window; "reference" reproduces B2.
+    window_mode: str = "canonical"   # "canonical" | "reference"
+
+    def __post_init__(self) -> None:
+        self.validate()
+
+    def validate(self) -> None:
+        if self.vector_size <= 0:
+            raise ValueError("vector_size must be > 0")
+        if self.window <= 0:
+            raise ValueError("window must be > 0")
+        if self.n < 0:
+            raise ValueError("n (negatives) must be >= 0")
+        if self.batch_size <= 0:
+            raise ValueError("batch_size must be > 0")
+        if self.num_iterations <= 0:
+            raise ValueError("num_iterations must be > 0")
+        if self.min_count < 0:
+            raise ValueError("min_count must be >= 0")
+        if self.max_sentence_length <= 0:
+            raise ValueError("max_sentence_length must be > 0")
+        if self.unigram_table_size <= 0:
+            raise ValueError("unigram_table_size must be > 0")
+        if self.dtype not in ("float32", "bfloat16"):
+            raise ValueError(f"unsupported dtype {self.dtype!r}")
+        if self.window_mode not in ("canonical", "reference"):
+            raise ValueError(f"unsupported window_mode {self.window_mode!r}")
+
+    # -- (de)serialisation used by the checkpoint metadata ------------------
+    def to_dict(self) -> dict:
+        return dataclasses.asdict(self)
+
+    @classmethod
+    def from_dict(cls, d: dict) -> "Word2VecConfig":
+        known = {f.name for f in dataclasses.fields(cls)}
+        return cls(**{k: v for k, v in d.items() if k in known})
+
+    def to_json(self) -> str:
+        return json.dumps(self.to_dict(), indent=2, sort_keys=True)
+
+    @classmethod
+    def from_json(cls, s: str) -> "Word2VecConfig":
+        return cls.from_dict(json.loads(s))

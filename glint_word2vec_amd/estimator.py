@@ -1,0 +1,305 @@
+"""User-facing API: ``GlintWord2Vec`` (estimator) and ``GlintWord2VecModel``.
+
+Mirrors the reference's Spark-ML surface (ml.feature.ServerSideGlintWord2Vec
+fit/transform/findSynonyms/getVectors/save/load/toLocal/stop — ml:284-497,
+mllib:460-726) without Spark: corpora are paths, iterables of token lists,
+or pandas DataFrames with an ``input_col`` of token lists.  The reference's
+fluent ``setX`` setters are provided as aliases.
+"""
+from __future__ import annotations
+
+import logging
+import math
+import time
+from typing import Iterable, List, Optional, Sequence, Union
+
+import numpy as np
+
+from .config import Word2VecConfig
+from .checkpoint import load_model, save_model, save_word2vec_text
+from .data import batch_sentences, read_text_corpus
+from .models import sgns
+from .vocab import (Vocabulary, build_unigram_table, build_vocab,
+                    encode_sentences, keep_probabilities)
+
+log = logging.getLogger("glint_word2vec_amd")
+
+CorpusLike = Union[str, Iterable[Sequence[str]]]
+
+
+def _corpus_reader(corpus: CorpusLike, input_col: Optional[str]):
+    """Return a zero-arg callable yielding token sequences (re-iterable)."""
+    if isinstance(corpus, str):
+        return lambda: read_text_corpus(corpus)
+    try:
+        import pandas as pd  # noqa
+        if hasattr(corpus, "columns") and input_col is not None:
+            col = corpus[input_col]
+            return lambda: iter(col.tolist())
+    except ImportError:
+        pass
+    if hasattr(corpus, "__iter__") and not hasattr(corpus, "__next__"):
+        return lambda: iter(corpus)
+    # one-shot iterator: materialise
+    sents = [list(s) for s in corpus]
+    return lambda: iter(sents)
+
+
+class GlintWord2Vec:
+    """Estimator.  Keyword args are Word2VecConfig fields; reference-style
+    fluent setters are also available (setVectorSize, setStepSize, ...)."""
+
+    def __init__(self, input_col: str = "sentence", output_col: str = "vector",
+                 config: Optional[Word2VecConfig] = None, **kwargs):
+        self.input_col = input_col
+        self.output_col = output_col
+        self.config = config if config is not None else Word2VecConfig(**kwargs)
+
+    # --- reference-parity fluent setters (ml:40-222) ----------------------
+    def _set(self, **kw) -> "GlintWord2Vec":
+        for k, v in kw.items():
+            setattr(self.config, k, v)
+        self.config.validate()
+        return self
+
+    def setVectorSize(self, v):        return self._set(vector_size=v)
+    def setStepSize(self, v):          return self._set(learning_rate=v)
+    def setLearningRate(self, v):      return self._set(learning_rate=v)
+    def setNumPartitions(self, v):     return self._set(num_partitions=v)
+    def setNumIterations(self, v):     return self._set(num_iterations=v)
+    def setMaxIter(self, v):           return self._set(num_iterations=v)
+    def setSeed(self, v):              return self._set(seed=v)
+    def setMinCount(self, v):          return self._set(min_count=v)
+    def setMaxSentenceLength(self, v): return self._set(max_sentence_length=v)
+    def setWindowSize(self, v):        return self._set(window=v)
+    def setBatchSize(self, v):         return self._set(batch_size=v)
+    def setN(self, v):                 return self._set(n=v)
+    def setSubsampleRatio(self, v):    return self._set(subsample_ratio=v)
+    def setUnigramTableSize(self, v):  return self._set(unigram_table_size=v)
+    def setNumParameterServers(self, v):
+        # PS count maps to GPU shard count in this framework.
+        return self._set(num_shards=v)
+
+    def setInputCol(self, v):
+        self.input_col = v
+        return self
+
+    def setOutputCol(self, v):
+        self.output_col = v
+        return self
+
+    # ----------------------------------------------------------------------
+    def fit(self, corpus: CorpusLike) -> "GlintWord2VecModel":
+        cfg = self.config
+        seed = cfg.seed if cfg.seed is not None else np.random.SeedSequence().entropy % (2 ** 63)
+        seed = int(seed)
+        reader = _corpus_reader(corpus, self.input_col)
+        vocab = build_vocab(reader(), min_count=cfg.min_count)
+        if vocab.num_words == 0:
+            raise ValueError("empty vocabulary — corpus has no word above min_count")
+        log.info("vocab: %d words, %d train words", vocab.num_words, vocab.train_words_count)
+
+        device = cfg.device
+        if device == "auto":
+            import torch
+            device = "cuda" if torch.cuda.is_available() else "cpu"
+        if device == "cuda":
+            from .parallel.engine import train_gpu
+            syn0, syn1 = train_gpu(cfg, vocab, reader, seed)
+        else:
+            syn0, syn1 = self._fit_cpu(cfg, vocab, reader, seed)
+        return GlintWord2VecModel(config=cfg, vocab=vocab, syn0=syn0, syn1=syn1,
+                                  output_col=self.output_col, input_col=self.input_col)
+
+    # --- single-process CPU trainer (BASELINE config 1) --------------------
+    def _fit_cpu(self, cfg: Word2VecConfig, vocab: Vocabulary, reader, seed: int):
+        syn0, syn1 = sgns.init_tables(vocab.num_words, cfg.vector_size, seed)
+        kp = None
+        if not cfg.legacy_subsample and cfg.subsample_ratio > 0:
+            kp = keep_probabilities(vocab.counts, vocab.train_words_count,
+                                    cfg.subsample_ratio)
+            if np.all(kp >= 1.0):
+                kp = None
+        table = build_unigram_table(vocab.counts, cfg.unigram_table_size,
+                                    cfg.unigram_power)
+        try:
+            from . import _cpu_native
+        except ImportError:
+            _cpu_native = None
+            log.warning("_cpu_native extension not built; falling back to the "
+                        "slow vectorized path (run `python setup.py build_ext "
+                        "--inplace`)")
+        rng = np.random.default_rng(seed)
+        total_words = vocab.train_words_count * cfg.num_iterations
+        processed = 0
+        sent_base = 0
+        t0 = time.time()
+        for it in range(cfg.num_iterations):
+            for batch in batch_sentences(
+                    encode_sentences(reader(), vocab, cfg.max_sentence_length),
+                    cfg.words_per_step):
+                alpha = cfg.learning_rate * max(
+                    1e-4, 1.0 - processed / (total_words + 1))
+                if _cpu_native is not None:
+                    st = _cpu_native.train_batch(
+                        syn0, syn1, batch.tokens, batch.offsets, kp, table,
+                        alpha, cfg.window, cfg.n, seed, sent_base,
+                        cfg.window_mode, cfg.num_partitions)
+                    npos, sum_fp = st["positives"], st["sum_fplus"]
+                else:
+                    plan = sgns.make_plan(batch.tokens, batch.offsets, kp,
+                                          table, cfg.window, cfg.n, rng,
+                                          cfg.window_mode)
+                    npos, sum_fp = sgns.train_plan_minibatched(
+                        syn0, syn1, plan, alpha, minibatch=256)
+                sent_base += batch.num_sentences
+                processed += batch.num_tokens
+                wps = processed / max(time.time() - t0, 1e-9)
+                log.info("iter %d: %d/%d words, alpha=%.5f, %.0f words/s, "
+                         "mean_fplus=%.4f", it, processed, total_words, alpha,
+                         wps, sum_fp / max(npos, 1))
+        return syn0, syn1
+
+
+class LocalWord2VecModel:
+    """`toLocal` result (mllib:651-659): plain in-memory word->vector map."""
+
+    def __init__(self, words: List[str], vectors: np.ndarray):
+        self.words = words
+        self.vectors = vectors
+        self.index = {w: i for i, w in enumerate(words)}
+
+    def __getitem__(self, word: str) -> np.ndarray:
+        return self.vectors[self.index[word]]
+
+    def save(self, path: str) -> None:
+        save_word2vec_text(path, self.words, self.vectors)
+
+
+class GlintWord2VecModel:
+    """Fitted model.  Holds the input matrix syn0 (and syn1 when kept for
+    resume); model ops follow the reference op semantics.
+    """
+
+    def __init__(self, config: Word2VecConfig, vocab: Vocabulary,
+                 syn0: np.ndarray, syn1: Optional[np.ndarray] = None,
+                 input_col: str = "sentence", output_col: str = "vector"):
+        self.config = config
+        self.vocab = vocab
+        self.syn0 = syn0
+        self.syn1 = syn1
+        self.input_col = input_col
+        self.output_col = output_col
+        self._norms: Optional[np.ndarray] = None
+
+    # --- helpers ----------------------------------------------------------
+    @property
+    def num_words(self) -> int:
+        return self.vocab.num_words
+
+    @property
+    def vector_size(self) -> int:
+        return int(self.syn0.shape[1])
+
+    def _f32(self) -> np.ndarray:
+        return self.syn0.astype(np.float32, copy=False)
+
+    def norms(self) -> np.ndarray:
+        """Lazily cached row norms (mllib:486)."""
+        if self._norms is None:
+            self._norms = np.linalg.norm(self._f32(), axis=1)
+        return self._norms
+
+    # --- transform --------------------------------------------------------
+    def transform(self, x):
+        """word -> vector; token sequence -> average vector (ml:432-460);
+        pandas DataFrame -> copy with output_col of averaged vectors."""
+        if isinstance(x, str):
+            return self._f32()[self.vocab[x]].copy()
+        if hasattr(x, "columns"):  # DataFrame
+            out = x.copy()
+            out[self.output_col] = [self.transform_sentence(s)
+                                    for s in x[self.input_col]]
+            return out
+        return self.transform_sentence(x)
+
+    def transform_sentence(self, tokens: Sequence[str]) -> np.ndarray:
+        idx = [self.vocab[w] for w in tokens if w in self.vocab]
+        if not idx:
+            return np.zeros(self.vector_size, dtype=np.float32)
+        return self._f32()[idx].mean(axis=0)
+
+    def transform_words(self, words: Iterable[str]) -> np.ndarray:
+        """Batched per-word lookup (mllib:529-543)."""
+        idx = [self.vocab[w] for w in words]
+        return self._f32()[idx]
+
+    # --- similarity -------------------------------------------------------
+    def find_synonyms(self, word_or_vec, num: int) -> List[tuple]:
+        """Top-``num`` cosine-similar words (mllib:554-630).  When queried by
+        word, the word itself is excluded."""
+        if isinstance(word_or_vec, str):
+            query_word = word_or_vec
+            vec = self._f32()[self.vocab[word_or_vec]]
+        else:
+            query_word = None
+            vec = np.asarray(word_or_vec, dtype=np.float32)
+        qn = np.linalg.norm(vec)
+        if qn > 0:
+            vec = vec / qn
+        scores = self._f32() @ vec              # `multiply` (mllib:598)
+        norms = self.norms()
+        with np.errstate(divide="ignore", invalid="ignore"):
+            cos = np.where(norms > 0, scores / norms, 0.0)
+        k = min(num + 1, len(cos))
+        top = np.argpartition(-cos, k - 1)[:k]
+        top = top[np.argsort(-cos[top])]
+        out = []
+        for i in top:
+            w = self.vocab.words[i]
+            if w == query_word:
+                continue
+            out.append((w, float(cos[i])))
+            if len(out) == num:
+                break
+        return out
+
+    def analogy(self, pos: List[str], neg: List[str], num: int = 10) -> List[tuple]:
+        """wien - oesterreich + deutschland -> berlin style queries
+        (IT spec :327-382)."""
+        v = np.zeros(self.vector_size, dtype=np.float32)
+        for w in pos:
+            v += self._f32()[self.vocab[w]]
+        for w in neg:
+            v -= self._f32()[self.vocab[w]]
+        skip = set(pos) | set(neg)
+        res = self.find_synonyms(v, num + len(skip))
+        return [(w, c) for w, c in res if w not in skip][:num]
+
+    # --- export / persistence --------------------------------------------
+    def get_vectors(self) -> dict:
+        f = self._f32()
+        return {w: f[i].copy() for i, w in enumerate(self.vocab.words)}
+
+    def to_local(self) -> LocalWord2VecModel:
+        return LocalWord2VecModel(list(self.vocab.words), self._f32().copy())
+
+    def save(self, path: str, num_shards: int = 1) -> None:
+        save_model(path, self.config, self.vocab, self.syn0, self.syn1,
+                   num_shards=num_shards)
+
+    @classmethod
+    def load(cls, path: str) -> "GlintWord2VecModel":
+        config, vocab, syn0, syn1 = load_model(path)
+        return cls(config=config, vocab=vocab, syn0=syn0, syn1=syn1)
+
+    def stop(self, terminate_other_clients: bool = False) -> None:
+        """Release training resources (reference: stops PS cluster,
+        mllib:664-667).  Tears down the torch.distributed group if this
+        process created one."""
+        try:
+            import torch.distributed as dist
+            if dist.is_initialized():
+                dist.destroy_process_group()
+        except Exception:
+            pass

@@ -1,0 +1,154 @@
+"""SGNS model: initialisation, vectorized pair-plan generation, and a fast
+vectorized CPU trainer.
+
+Two training paths exist:
+  * the fused HIP kernel (ops/gpu.py) — sampling + training in one kernel,
+    exact-oracle-matched (ops/cpu_ref.py);
+  * this module's vectorized path — generates an explicit pair plan
+    (subsample -> window -> negatives) then applies mini-batched updates.
+    It is the CPU production path (quality gates, plumbing config 1 of
+    BASELINE.json) and the plan generator also feeds the sharded multi-GPU
+    path, where remote rows must be known before the kernel runs.
+
+The plan's RNG is numpy's (performance); exact-RNG parity with the kernel is
+the oracle's job, not this path's.
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass
+from typing import Iterator, Tuple
+
+import numpy as np
+
+from ..config import MAX_EXP, Word2VecConfig
+
+
+def init_tables(vocab_size: int, dim: int, seed: int) -> Tuple[np.ndarray, np.ndarray]:
+    """Canonical word2vec init: syn0 ~ U(-0.5/dim, 0.5/dim), syn1 = 0."""
+    rng = np.random.default_rng(seed)
+    syn0 = ((rng.random((vocab_size, dim), dtype=np.float32) - 0.5) / dim).astype(np.float32)
+    syn1 = np.zeros((vocab_size, dim), dtype=np.float32)
+    return syn0, syn1
+
+
+@dataclass
+class PairPlan:
+    """Flat training work for one step: positives and negatives interleaved
+    per positive (targets of one center are contiguous)."""
+    center: np.ndarray   # int32 [P]
+    target: np.ndarray   # int32 [P]
+    label: np.ndarray    # float32 [P] (1.0 positive, 0.0 negative)
+
+    @property
+    def num_pairs(self) -> int:
+        return len(self.center)
+
+
+def subsample_batch(tokens: np.ndarray, offsets: np.ndarray,
+                    keep_prob: np.ndarray | None,
+                    rng: np.random.Generator) -> Tuple[np.ndarray, np.ndarray]:
+    """Apply frequency subsampling, preserving sentence boundaries.
+    Returns (kept_tokens, new_offsets)."""
+    if keep_prob is None:
+        return tokens, offsets
+    u = rng.random(len(tokens))
+    keep = u < keep_prob[tokens]
+    kept = tokens[keep]
+    # new offsets: count kept per sentence
+    counts = np.add.reduceat(keep.astype(np.int64), offsets[:-1]) if len(tokens) else \
+        np.zeros(len(offsets) - 1, dtype=np.int64)
+    # reduceat quirk: empty sentences at the end still index; guard zero-len
+    sent_len = np.diff(offsets)
+    counts = np.where(sent_len > 0, counts, 0)
+    new_offsets = np.zeros(len(offsets), dtype=np.int32)
+    np.cumsum(counts, out=new_offsets[1:])
+    return kept.astype(np.int32), new_offsets
+
+
+def make_plan(tokens: np.ndarray, offsets: np.ndarray,
+              keep_prob: np.ndarray | None, table: np.ndarray,
+              window: int, n_neg: int, rng: np.random.Generator,
+              window_mode: str = "canonical") -> PairPlan:
+    """Vectorized subsample -> shrunk-window pair generation -> negative
+    draws.  Negatives colliding with their positive target are dropped
+    (matching the oracle's skip rule)."""
+    toks, offs = subsample_batch(tokens, offsets, keep_prob, rng)
+    T = len(toks)
+    if T == 0:
+        z = np.zeros(0, dtype=np.int32)
+        return PairPlan(z, z, np.zeros(0, dtype=np.float32))
+    # sentence id and in-sentence bounds per position
+    sent_id = np.repeat(np.arange(len(offs) - 1, dtype=np.int64), np.diff(offs))
+    sent_lo = offs[:-1].astype(np.int64)[sent_id]
+    sent_hi = offs[1:].astype(np.int64)[sent_id]          # exclusive
+    pos = np.arange(T, dtype=np.int64)
+    if window_mode == "canonical":
+        b = rng.integers(1, window + 1, size=T)
+        left, right = b, b
+    else:  # reference-mode: left b, right b-1, b in [0, window-1]
+        b = rng.integers(0, window, size=T)
+        left, right = b, np.maximum(b - 1, 0) * (b > 0)
+    centers_l, targets_l = [], []
+    for o in range(1, window + 1):
+        m = (o <= left) & (pos - o >= sent_lo)
+        centers_l.append(pos[m]); targets_l.append(pos[m] - o)
+        m = (o <= right) & (pos + o < sent_hi)
+        centers_l.append(pos[m]); targets_l.append(pos[m] + o)
+    cpos = np.concatenate(centers_l) if centers_l else np.zeros(0, dtype=np.int64)
+    tpos = np.concatenate(targets_l) if targets_l else np.zeros(0, dtype=np.int64)
+    # keep targets of one center contiguous & deterministic: sort by (center, target)
+    order = np.lexsort((tpos, cpos))
+    cpos, tpos = cpos[order], tpos[order]
+    pc = toks[cpos].astype(np.int32)     # positive centers
+    pt = toks[tpos].astype(np.int32)     # positive targets
+    P = len(pc)
+    if n_neg > 0 and P > 0:
+        nidx = rng.integers(0, len(table), size=(P, n_neg))
+        negs = table[nidx].astype(np.int32)               # [P, n]
+        valid = negs != pt[:, None]
+        # interleave: for each positive, its negatives follow it
+        ctr = np.repeat(pc, n_neg).reshape(P, n_neg)
+        all_center = np.concatenate([pc[:, None], ctr], axis=1)      # [P, 1+n]
+        all_target = np.concatenate([pt[:, None], negs], axis=1)
+        all_label = np.concatenate(
+            [np.ones((P, 1), np.float32), np.zeros((P, n_neg), np.float32)], axis=1)
+        all_valid = np.concatenate([np.ones((P, 1), bool), valid], axis=1)
+        flat = all_valid.ravel()
+        return PairPlan(all_center.ravel()[flat].astype(np.int32),
+                        all_target.ravel()[flat].astype(np.int32),
+                        all_label.ravel()[flat])
+    return PairPlan(pc, pt, np.ones(P, dtype=np.float32))
+
+
+def train_plan_minibatched(syn0: np.ndarray, syn1: np.ndarray, plan: PairPlan,
+                           alpha: float, minibatch: int = 16384) -> Tuple[int, float]:
+    """Fast vectorized CPU SGD over a pair plan (torch index_add under the
+    hood).  Updates within a mini-batch see pre-batch rows (mini-batched
+    hogwild — same asynchrony class as the reference's concurrent workers,
+    mllib:392-433).  Returns (num_positive_pairs, sum_f_plus) for the
+    divergence canary (mllib:411-412)."""
+    import torch
+    s0 = torch.from_numpy(syn0)
+    s1 = torch.from_numpy(syn1)
+    c_all = torch.from_numpy(plan.center.astype(np.int64))
+    t_all = torch.from_numpy(plan.target.astype(np.int64))
+    l_all = torch.from_numpy(plan.label)
+    n_pos = 0
+    sum_fplus = 0.0
+    for s in range(0, plan.num_pairs, minibatch):
+        c = c_all[s:s + minibatch]
+        t = t_all[s:s + minibatch]
+        lab = l_all[s:s + minibatch]
+        c_rows = s0.index_select(0, c)
+        t_rows = s1.index_select(0, t)
+        f = (c_rows * t_rows).sum(dim=1)
+        sig = torch.sigmoid(f.clamp(-MAX_EXP, MAX_EXP))
+        sig = torch.where(f > MAX_EXP, torch.ones_like(sig), sig)
+        sig = torch.where(f < -MAX_EXP, torch.zeros_like(sig), sig)
+        g = (lab - sig) * alpha
+        s1.index_add_(0, t, g[:, None] * c_rows)
+        s0.index_add_(0, c, g[:, None] * t_rows)
+        pos_mask = lab > 0.5
+        n_pos += int(pos_mask.sum())
+        sum_fplus += float(f[pos_mask].sum())
+    return n_pos, sum_fplus
