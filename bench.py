@@ -1,0 +1,173 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: SGNS training words/sec (whole node).
+
+BASELINE.json metric: "training words/sec (whole node), vocab=1M dim=300
+SGNS, at 1/2/4/8 MI355X".  Synthetic Zipf token stream, random-init weights,
+bf16 tables (f32 math in-kernel).
+
+Usage (driver contract):
+  python bench.py --gpus N --steps K --warmup W
+For N>1 the driver launches via torch.distributed.run with one rank per GPU
+(RCCL); rank r reads RANK/LOCAL_RANK/WORLD_SIZE from the env.  Weak scaling:
+each GPU trains words_per_step tokens per step on its own corpus partition.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import numpy as np
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--vocab", type=int, default=1_000_000)
+    p.add_argument("--dim", type=int, default=300)
+    p.add_argument("--neg", type=int, default=5)
+    p.add_argument("--window", type=int, default=5)
+    p.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "fp32"])
+    p.add_argument("--words-per-step", type=int, default=1_000_000)
+    p.add_argument("--sentence-len", type=int, default=250)
+    p.add_argument("--table-size", type=int, default=100_000_000)
+    p.add_argument("--subsample", type=float, default=1e-4,
+                   help="subsample ratio (drawn in-kernel; 0 disables)")
+    p.add_argument("--blocks", type=int, default=0, help="grid blocks override")
+    p.add_argument("--atomic", action="store_true")
+    p.add_argument("--profile-steps", type=int, default=0,
+                   help="run only this many steps, no warmup JSON (rocprof)")
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    import torch
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    distributed = world > 1
+    if distributed:
+        import torch.distributed as dist
+        torch.cuda.set_device(local_rank)
+        dist.init_process_group(backend="nccl")
+    device = torch.device("cuda", local_rank)
+    torch.cuda.set_device(device)
+
+    from glint_word2vec_amd.data import synthetic_corpus
+    from glint_word2vec_amd.vocab import build_unigram_table
+
+    dtype = "bfloat16" if args.dtype == "bf16" else "float32"
+    n_steps = args.profile_steps if args.profile_steps else args.steps
+    total_launches = n_steps + (0 if args.profile_steps else args.warmup)
+
+    # --- synthetic data: one step's batch, reused with varying sent_id_base
+    # (RNG/negative draws differ per step; token stream is Zipf-realistic).
+    batch = synthetic_corpus(args.vocab, args.words_per_step,
+                             sentence_len=args.sentence_len,
+                             seed=1234 + rank)
+    counts = np.bincount(batch.tokens, minlength=args.vocab).astype(np.int64) + 1
+
+    if distributed:
+        from glint_word2vec_amd.parallel.sharded import ShardedSgns
+        eng = ShardedSgns(args.vocab, args.dim, dtype=dtype, device=str(device),
+                          seed=1, counts=counts, table_size=args.table_size,
+                          subsample=args.subsample)
+        trainer = eng
+    else:
+        from glint_word2vec_amd.ops.gpu import GpuSgns
+        gs = GpuSgns(args.vocab, args.dim, dtype=dtype, device=str(device), seed=1)
+        table = build_unigram_table(counts, args.table_size)
+        gs.set_table(table)
+        if args.subsample > 0:
+            gs.set_subsample(counts, int(counts.sum()), args.subsample)
+        trainer = gs
+
+    tok = torch.from_numpy(batch.tokens).to(device)
+    off = torch.from_numpy(batch.offsets).to(device)
+    alpha = 0.01875
+    nsent = batch.num_sentences
+
+    def step(i):
+        if distributed:
+            trainer.train_step(tok, off, alpha, args.window, args.neg,
+                               seed=99, sent_id_base=i * nsent)
+        else:
+            trainer.train_batch(tok, off, alpha, args.window, args.neg, 99,
+                                sent_id_base=i * nsent,
+                                atomic=args.atomic,
+                                blocks=args.blocks or None)
+
+    def barrier_sync():
+        torch.cuda.synchronize(device)
+        if distributed:
+            import torch.distributed as dist
+            dist.barrier()
+        torch.cuda.synchronize(device)
+
+    if args.profile_steps:
+        for i in range(n_steps):
+            step(i)
+        torch.cuda.synchronize(device)
+        if rank == 0:
+            st = trainer.read_stats()
+            print(f"profiled {n_steps} steps: {st}")
+        return
+
+    for i in range(args.warmup):
+        step(i)
+    barrier_sync()
+    trainer.read_stats()  # reset
+    t0 = time.time()
+    for i in range(args.steps):
+        step(args.warmup + i)
+    barrier_sync()
+    elapsed = time.time() - t0
+
+    if distributed:
+        import torch.distributed as dist
+        t = torch.tensor([elapsed], device=device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    st = trainer.read_stats()
+    words = args.words_per_step * args.steps * world
+    wps = words / elapsed
+    if rank == 0:
+        out = {
+            "metric": "training words/sec (whole node), vocab=1M dim=300 SGNS",
+            "value": wps,
+            "unit": "words/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000.0,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if args.dtype == "bf16" else "fp32",
+            "data": "synthetic (Zipf 1.05 token stream, random-init weights)",
+            "config": {
+                "model": f"sgns vocab={args.vocab} dim={args.dim} "
+                         f"neg={args.neg} window={args.window}",
+                "global_batch": args.words_per_step * world,
+                "seq_len": args.sentence_len,
+                "parallelism": (f"rowshard-alltoallv-dp{world}" if distributed
+                                else "hogwild-1gpu"),
+            },
+            "pairs_per_step": st.pairs / max(args.steps, 1),
+            "mean_fplus": st.sum_fplus / max(st.positives, 1),
+        }
+        print(json.dumps(out))
+    if distributed:
+        import torch.distributed as dist
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

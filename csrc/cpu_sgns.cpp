@@ -27,29 +27,30 @@ namespace py = pybind11;
 
 static constexpr float kMaxExp = 6.0f;
 
-// --- RNG (must match glint_word2vec_amd/rng.py bit-for-bit) ---------------
+// --- counter-based RNG (must match glint_word2vec_amd/rng.py bit-for-bit) -
+static constexpr uint64_t kGolden = 0x9E3779B97F4A7C15ULL;
+static constexpr uint64_t kWinBase = 1ULL << 20;
+static constexpr uint64_t kNegBase = 1ULL << 21;
+
 static inline uint64_t splitmix64(uint64_t x) {
-  uint64_t z = x + 0x9E3779B97F4A7C15ULL;
+  uint64_t z = x + kGolden;
   z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ULL;
   z = (z ^ (z >> 27)) * 0x94D049BB133111EBULL;
   return z ^ (z >> 31);
 }
 
-struct XorShift64Star {
-  uint64_t state;
-  XorShift64Star(uint64_t seed, uint64_t sentence_id) {
-    uint64_t s = splitmix64(seed ^ (sentence_id * 0x9E3779B97F4A7C15ULL));
-    state = s ? s : 1;
-  }
-  inline uint32_t next_u32() {
-    uint64_t x = state;
-    x ^= x >> 12;
-    x ^= x << 25;
-    x ^= x >> 27;
-    state = x;
-    return (uint32_t)((x * 0x2545F4914F6CDD1DULL) >> 32);
-  }
-};
+static inline uint64_t sentence_base(uint64_t seed, uint64_t sentence_id) {
+  return splitmix64(seed ^ (sentence_id * kGolden));
+}
+
+static inline uint32_t draw_u32(uint64_t base, uint64_t k) {
+  return (uint32_t)(splitmix64(base + k * kGolden) >> 32);
+}
+
+static inline uint32_t keep_thr(float kp) {
+  double t = (double)kp * 4294967296.0;
+  return t >= 4294967295.0 ? 0xFFFFFFFFu : (uint32_t)t;
+}
 
 static inline float sigmoid_clipped(float f) {
   if (f > kMaxExp) return 1.0f;
@@ -79,13 +80,13 @@ static void train_sentences(float* syn0, float* syn1, int64_t dim,
   for (int64_t s = s_begin; s < s_end; ++s) {
     const int32_t* sent = tokens + offsets[s];
     int64_t len = offsets[s + 1] - offsets[s];
-    XorShift64Star rng(seed, (uint64_t)(sent_id_base + s));
+    uint64_t base = sentence_base(seed, (uint64_t)(sent_id_base + s));
     kept.clear();
     if (keep_prob) {
       for (int64_t p = 0; p < len; ++p) {
-        uint32_t u = rng.next_u32();
+        uint32_t u = draw_u32(base, (uint64_t)p);
         int32_t w = sent[p];
-        if ((double)u < (double)keep_prob[w] * 4294967296.0) kept.push_back(w);
+        if (u <= keep_thr(keep_prob[w])) kept.push_back(w);
       }
     } else {
       kept.assign(sent, sent + len);
@@ -93,7 +94,7 @@ static void train_sentences(float* syn0, float* syn1, int64_t dim,
     int64_t L = (int64_t)kept.size();
     for (int64_t i = 0; i < L; ++i) {
       int32_t c = kept[i];
-      uint32_t u = rng.next_u32();
+      uint32_t u = draw_u32(base, kWinBase + (uint64_t)i);
       int64_t lo, hi;
       if (!reference_window) {
         int64_t b = 1 + (int64_t)(u % (uint32_t)window);
@@ -128,8 +129,10 @@ static void train_sentences(float* syn0, float* syn1, int64_t dim,
           }
           stats->pairs++; stats->positives++; stats->sum_fplus += f;
         }
+        uint64_t kbase = kNegBase +
+            (uint64_t)(i * (2 * window + 1) + (j - i + window)) * (uint64_t)n_neg;
         for (int k = 0; k < n_neg; ++k) {
-          uint32_t un = rng.next_u32();
+          uint32_t un = draw_u32(base, kbase + (uint64_t)k);
           int32_t neg = table[un % (uint64_t)table_size];
           if (neg == t) continue;
           float* t1 = syn1 + (int64_t)neg * dim;

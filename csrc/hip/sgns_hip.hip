@@ -1,0 +1,538 @@
+// Fused SGNS training kernel for MI355X (gfx950, CDNA4).
+//
+// This is the MI355X-native replacement for the reference's entire hot path:
+// the per-minibatch client loop (mllib ServerSideGlintWord2Vec.scala:419-429)
+// plus the Glint parameter-server dotprod/adjust ops.  One
+// kernel launch consumes a step's worth of encoded sentences and performs,
+// fully fused on-device:
+//   frequency subsampling -> shrunk-window pair generation -> unigram-table
+//   negative draws -> embedding row gather -> center*target dots (wave
+//   reduction) -> clipped sigmoid + gradient -> SGD scatter-add on both
+//   tables.
+//
+// Execution model: one 64-lane wave per sentence (4 waves per 256-thread
+// workgroup), sentences assigned round-robin over the launched waves.  The
+// per-GPU asynchrony of the reference's numPartitions workers (mllib:120-127)
+// becomes hogwild waves: row updates are plain read-modify-write by default
+// (races embraced exactly as the reference embraces async PS updates), with
+// an atomicAdd variant for fp32 when determinism-vs-throughput is preferred.
+//
+// Rows are laid out [vocab][stride] with stride a multiple of 64 elements;
+// each lane owns elements in chunk-pairs of 128 (2 adjacent elements per
+// lane -> dwordx2/dword coalesced 512B/256B per instruction, guide G13),
+// plus one 64-element tail chunk when NC is odd.  Padding elements stay 0
+// forever (0-init, updates scale by row values), so no masking is needed.
+//
+// RNG: counter-based splitmix64 streams, bit-identical to
+// glint_word2vec_amd/rng.py and csrc/cpu_sgns.cpp — see rng.py for the
+// normative draw-index layout.  This makes every draw computable at any
+// lane with no sequential state.
+
+#include <hip/hip_runtime.h>
+#include <pybind11/pybind11.h>
+
+#include <cstdint>
+#include <cstring>
+#include <stdexcept>
+#include <string>
+
+namespace py = pybind11;
+
+#define HIP_CHECK(expr)                                                     \
+  do {                                                                      \
+    hipError_t _e = (expr);                                                 \
+    if (_e != hipSuccess)                                                   \
+      throw std::runtime_error(std::string("HIP error: ") +                 \
+                               hipGetErrorString(_e) + " at " __FILE__ ":" + \
+                               std::to_string(__LINE__));                   \
+  } while (0)
+
+// ---------------------------------------------------------------------------
+// RNG (normative contract in glint_word2vec_amd/rng.py)
+// ---------------------------------------------------------------------------
+__host__ __device__ __forceinline__ uint64_t splitmix64(uint64_t x) {
+  uint64_t z = x + 0x9E3779B97F4A7C15ULL;
+  z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ULL;
+  z = (z ^ (z >> 27)) * 0x94D049BB133111EBULL;
+  return z ^ (z >> 31);
+}
+__host__ __device__ __forceinline__ uint64_t sent_base(uint64_t seed, uint64_t sid) {
+  return splitmix64(seed ^ (sid * 0x9E3779B97F4A7C15ULL));
+}
+__device__ __forceinline__ uint32_t draw_u32(uint64_t base, uint64_t k) {
+  return (uint32_t)(splitmix64(base + k * 0x9E3779B97F4A7C15ULL) >> 32);
+}
+constexpr uint64_t kWinBase = 1ULL << 20;
+constexpr uint64_t kNegBase = 1ULL << 21;
+
+// ---------------------------------------------------------------------------
+// Row I/O: float32 and bf16 (raw u16) storage, f32 math.
+// Element of register slot k (k < NC, NC = stride/64):
+//   paired slots 2m / 2m+1 : elements 128*m + 2*lane + {0,1}
+//   odd-NC tail slot NC-1  : element  64*(NC-1) + lane
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ float bf16_to_f32(uint16_t h) {
+  uint32_t u = ((uint32_t)h) << 16;
+  return __uint_as_float(u);
+}
+__device__ __forceinline__ uint16_t f32_to_bf16_rne(float f) {
+  uint32_t x = __float_as_uint(f);
+  uint32_t r = (x + 0x7FFFu + ((x >> 16) & 1u)) >> 16;
+  return (uint16_t)r;
+}
+
+template <typename T, int NC>
+struct RowIO;
+
+template <int NC>
+struct RowIO<float, NC> {
+  static __device__ __forceinline__ void load(const float* row, float v[NC], int lane) {
+#pragma unroll
+    for (int m = 0; m < NC / 2; ++m) {
+      const float2 p = *reinterpret_cast<const float2*>(row + 128 * m + 2 * lane);
+      v[2 * m] = p.x;
+      v[2 * m + 1] = p.y;
+    }
+    if (NC & 1) v[NC - 1] = row[64 * (NC - 1) + lane];
+  }
+  static __device__ __forceinline__ void store(float* row, const float v[NC], int lane) {
+#pragma unroll
+    for (int m = 0; m < NC / 2; ++m) {
+      *reinterpret_cast<float2*>(row + 128 * m + 2 * lane) =
+          make_float2(v[2 * m], v[2 * m + 1]);
+    }
+    if (NC & 1) row[64 * (NC - 1) + lane] = v[NC - 1];
+  }
+  static __device__ __forceinline__ void atomic_add(float* row, const float v[NC], int lane) {
+#pragma unroll
+    for (int m = 0; m < NC / 2; ++m) {
+      atomicAdd(row + 128 * m + 2 * lane, v[2 * m]);
+      atomicAdd(row + 128 * m + 2 * lane + 1, v[2 * m + 1]);
+    }
+    if (NC & 1) atomicAdd(row + 64 * (NC - 1) + lane, v[NC - 1]);
+  }
+};
+
+template <int NC>
+struct RowIO<uint16_t, NC> {
+  static __device__ __forceinline__ void load(const uint16_t* row, float v[NC], int lane) {
+#pragma unroll
+    for (int m = 0; m < NC / 2; ++m) {
+      const uint32_t p = *reinterpret_cast<const uint32_t*>(row + 128 * m + 2 * lane);
+      v[2 * m] = bf16_to_f32((uint16_t)(p & 0xFFFF));
+      v[2 * m + 1] = bf16_to_f32((uint16_t)(p >> 16));
+    }
+    if (NC & 1) v[NC - 1] = bf16_to_f32(row[64 * (NC - 1) + lane]);
+  }
+  static __device__ __forceinline__ void store(uint16_t* row, const float v[NC], int lane) {
+#pragma unroll
+    for (int m = 0; m < NC / 2; ++m) {
+      uint32_t p = (uint32_t)f32_to_bf16_rne(v[2 * m]) |
+                   ((uint32_t)f32_to_bf16_rne(v[2 * m + 1]) << 16);
+      *reinterpret_cast<uint32_t*>(row + 128 * m + 2 * lane) = p;
+    }
+    if (NC & 1) row[64 * (NC - 1) + lane] = f32_to_bf16_rne(v[NC - 1]);
+  }
+  static __device__ __forceinline__ void atomic_add(uint16_t*, const float[NC], int) {
+    // bf16 atomic variant not supported (wrapper rejects it)
+  }
+};
+
+// ---------------------------------------------------------------------------
+// sigmoid with the reference's +-MAX_EXP clip (mllib:281-302 semantics,
+// exact sigmoid instead of the 1000-entry LUT approximation)
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ float sigmoid_clipped(float f) {
+  if (f > 6.0f) return 1.0f;
+  if (f < -6.0f) return 0.0f;
+  return 1.0f / (1.0f + __expf(-f));
+}
+
+struct KernelArgs {
+  void* syn0;
+  void* syn1;
+  const int32_t* tokens;
+  const int32_t* offsets;
+  int64_t num_sentences;
+  const uint32_t* keep_thr;   // nullptr = subsampling off
+  const int32_t* table;
+  uint32_t table_size;
+  float alpha;
+  int window;
+  int n_neg;
+  uint64_t seed;
+  int64_t sent_id_base;
+  int64_t stride;             // row stride in elements (= 64*NC)
+  int ref_window;             // 0 canonical, 1 reference (B2) semantics
+  // stats
+  unsigned long long* d_pairs;
+  unsigned long long* d_positives;
+  unsigned long long* d_words;
+  double* d_sum_fplus;
+};
+
+constexpr int kMaxSent = 1024;
+constexpr int kWavesPerBlock = 4;
+
+template <typename T, int NC, bool ATOMIC>
+__global__ __launch_bounds__(64 * kWavesPerBlock) void sgns_train_kernel(KernelArgs a) {
+  __shared__ int32_t sbuf[kWavesPerBlock][kMaxSent];
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int64_t wave_gid = (int64_t)blockIdx.x * kWavesPerBlock + wave;
+  const int64_t total_waves = (int64_t)gridDim.x * kWavesPerBlock;
+  int32_t* sent_lds = sbuf[wave];
+
+  T* syn0 = (T*)a.syn0;
+  T* syn1 = (T*)a.syn1;
+
+  unsigned long long w_pairs = 0, w_pos = 0, w_words = 0;
+  float w_fplus = 0.0f;
+
+  for (int64_t s = wave_gid; s < a.num_sentences; s += total_waves) {
+    const int64_t off = a.offsets[s];
+    const int len = (int)(a.offsets[s + 1] - off);
+    const uint64_t base = sent_base(a.seed, (uint64_t)(a.sent_id_base + s));
+
+    // ---- 1. subsample + compact into LDS --------------------------------
+    int L = 0;
+    for (int p0 = 0; p0 < len; p0 += 64) {
+      const int p = p0 + lane;
+      bool keep = false;
+      int32_t w = 0;
+      if (p < len) {
+        w = a.tokens[off + p];
+        if (a.keep_thr) {
+          const uint32_t u = draw_u32(base, (uint64_t)p);
+          keep = u <= a.keep_thr[w];
+        } else {
+          keep = true;
+        }
+      }
+      const uint64_t m = __ballot(keep);
+      const int pos = __popcll(m & ((1ULL << lane) - 1ULL));
+      if (keep) sent_lds[L + pos] = w;
+      L += __popcll(m);
+    }
+    // (no __syncthreads needed: LDS buffer is private to this wave)
+
+    // ---- 2. per-position window loop ------------------------------------
+    for (int i = 0; i < L; ++i) {
+      const int32_t c = sent_lds[i];
+      const uint32_t u = draw_u32(base, kWinBase + (uint64_t)i);
+      int lo, hi;
+      if (!a.ref_window) {
+        const int b = 1 + (int)(u % (uint32_t)a.window);
+        lo = i - b < 0 ? 0 : i - b;
+        hi = i + b >= L ? L - 1 : i + b;
+      } else {
+        const int b = (int)(u % (uint32_t)a.window);
+        if (b == 0) { lo = i; hi = i; }
+        else {
+          lo = i - b < 0 ? 0 : i - b;
+          hi = i + b - 1 >= L ? L - 1 : i + b - 1;
+        }
+      }
+      if (lo == i && hi == i) continue;
+      bool any = (lo < i) || (hi > i);
+      if (!any) continue;
+
+      T* c_ptr = syn0 + (int64_t)c * a.stride;
+      float c_row[NC], grad[NC];
+      RowIO<T, NC>::load(c_ptr, c_row, lane);
+#pragma unroll
+      for (int k = 0; k < NC; ++k) grad[k] = 0.0f;
+
+      for (int j = lo; j <= hi; ++j) {
+        if (j == i) continue;
+        const int32_t t = sent_lds[j];
+        // positive pair, then its negatives — same inner body
+        const uint64_t kbase =
+            kNegBase + (uint64_t)(i * (2 * a.window + 1) + (j - i + a.window)) *
+                           (uint64_t)a.n_neg;
+        for (int slot = -1; slot < a.n_neg; ++slot) {
+          int32_t tgt;
+          float label;
+          if (slot < 0) {
+            tgt = t;
+            label = 1.0f;
+          } else {
+            const uint32_t un = draw_u32(base, kbase + (uint64_t)slot);
+            tgt = a.table[un % a.table_size];
+            if (tgt == t) continue;   // discard colliding negative
+            label = 0.0f;
+          }
+          T* t_ptr = syn1 + (int64_t)tgt * a.stride;
+          float t_row[NC];
+          RowIO<T, NC>::load(t_ptr, t_row, lane);
+          float f = 0.0f;
+#pragma unroll
+          for (int k = 0; k < NC; ++k) f += c_row[k] * t_row[k];
+#pragma unroll
+          for (int sh = 32; sh > 0; sh >>= 1) f += __shfl_xor(f, sh, 64);
+          const float g = (label - sigmoid_clipped(f)) * a.alpha;
+#pragma unroll
+          for (int k = 0; k < NC; ++k) grad[k] += g * t_row[k];
+          if (ATOMIC) {
+            float delta[NC];
+#pragma unroll
+            for (int k = 0; k < NC; ++k) delta[k] = g * c_row[k];
+            RowIO<T, NC>::atomic_add(t_ptr, delta, lane);
+          } else {
+#pragma unroll
+            for (int k = 0; k < NC; ++k) t_row[k] += g * c_row[k];
+            RowIO<T, NC>::store(t_ptr, t_row, lane);
+          }
+          ++w_pairs;
+          if (slot < 0) {
+            ++w_pos;
+            w_fplus += f;
+          }
+        }
+      }
+      // center row update (hogwild: re-read current value, add, store)
+      if (ATOMIC) {
+        RowIO<T, NC>::atomic_add(c_ptr, grad, lane);
+      } else {
+        float cur[NC];
+        RowIO<T, NC>::load(c_ptr, cur, lane);
+#pragma unroll
+        for (int k = 0; k < NC; ++k) cur[k] += grad[k];
+        RowIO<T, NC>::store(c_ptr, cur, lane);
+      }
+      ++w_words;
+    }
+  }
+
+  // ---- stats (device-scope atomics, once per wave) -----------------------
+  if (lane == 0 && a.d_pairs) {
+    atomicAdd(a.d_pairs, w_pairs);
+    atomicAdd(a.d_positives, w_pos);
+    atomicAdd(a.d_words, w_words);
+    atomicAdd(a.d_sum_fplus, (double)w_fplus);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// pullAverage: per-sentence mean of syn0 rows (Glint pullAverage, ml:453).
+// One wave per sentence; output f32 [num_sentences][dim_out<=stride].
+// ---------------------------------------------------------------------------
+template <typename T, int NC>
+__global__ __launch_bounds__(256) void pull_average_kernel(
+    const T* __restrict__ syn0, const int32_t* __restrict__ tokens,
+    const int32_t* __restrict__ offsets, int64_t num_sentences,
+    int64_t stride, float* __restrict__ out) {
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int64_t wave_gid = (int64_t)blockIdx.x * kWavesPerBlock + wave;
+  const int64_t total_waves = (int64_t)gridDim.x * kWavesPerBlock;
+  for (int64_t s = wave_gid; s < num_sentences; s += total_waves) {
+    const int64_t off = offsets[s];
+    const int len = (int)(offsets[s + 1] - off);
+    float acc[NC];
+#pragma unroll
+    for (int k = 0; k < NC; ++k) acc[k] = 0.0f;
+    for (int p = 0; p < len; ++p) {
+      const T* row = syn0 + (int64_t)tokens[off + p] * stride;
+      float v[NC];
+      RowIO<T, NC>::load(row, v, lane);
+#pragma unroll
+      for (int k = 0; k < NC; ++k) acc[k] += v[k];
+    }
+    const float inv = len > 0 ? 1.0f / (float)len : 0.0f;
+#pragma unroll
+    for (int k = 0; k < NC; ++k) acc[k] *= inv;
+    RowIO<float, NC>::store(out + s * stride, acc, lane);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// norms: Euclidean norm of every row (Glint norms, mllib:486).
+// ---------------------------------------------------------------------------
+template <typename T, int NC>
+__global__ __launch_bounds__(256) void norms_kernel(
+    const T* __restrict__ syn0, int64_t vocab, int64_t stride,
+    float* __restrict__ out) {
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int64_t wave_gid = (int64_t)blockIdx.x * kWavesPerBlock + wave;
+  const int64_t total_waves = (int64_t)gridDim.x * kWavesPerBlock;
+  for (int64_t r = wave_gid; r < vocab; r += total_waves) {
+    float v[NC];
+    RowIO<T, NC>::load(syn0 + r * stride, v, lane);
+    float ss = 0.0f;
+#pragma unroll
+    for (int k = 0; k < NC; ++k) ss += v[k] * v[k];
+#pragma unroll
+    for (int sh = 32; sh > 0; sh >>= 1) ss += __shfl_xor(ss, sh, 64);
+    if (lane == 0) out[r] = sqrtf(ss);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Host-side dispatch
+// ---------------------------------------------------------------------------
+template <typename T, int NC>
+static void launch_train_nc(const KernelArgs& a, bool atomic, int blocks,
+                            hipStream_t stream) {
+  if (atomic)
+    hipLaunchKernelGGL((sgns_train_kernel<T, NC, true>), dim3(blocks),
+                       dim3(64 * kWavesPerBlock), 0, stream, a);
+  else
+    hipLaunchKernelGGL((sgns_train_kernel<T, NC, false>), dim3(blocks),
+                       dim3(64 * kWavesPerBlock), 0, stream, a);
+}
+
+// Supported NC values (stride = 64*NC).  The wrapper rounds the row stride
+// up to the nearest supported NC.
+#define FOR_EACH_NC(X) \
+  X(1) X(2) X(3) X(4) X(5) X(6) X(8) X(10) X(12) X(16) X(20) X(24) X(32)
+
+template <typename T>
+static void launch_train(const KernelArgs& a, int nc, bool atomic, int blocks,
+                         hipStream_t stream) {
+  switch (nc) {
+#define CASE_NC(N)                                        \
+  case N:                                                 \
+    launch_train_nc<T, N>(a, atomic, blocks, stream);     \
+    return;
+    FOR_EACH_NC(CASE_NC)
+#undef CASE_NC
+    default:
+      throw std::runtime_error("unsupported NC=" + std::to_string(nc));
+  }
+}
+
+static int supported_nc(int nc_min) {
+  static const int ncs[] = {1, 2, 3, 4, 5, 6, 8, 10, 12, 16, 20, 24, 32};
+  for (int nc : ncs)
+    if (nc >= nc_min) return nc;
+  return -1;
+}
+
+static void sgns_train(uintptr_t syn0, uintptr_t syn1, int is_bf16,
+                       int64_t stride, uintptr_t tokens, uintptr_t offsets,
+                       int64_t num_sentences, uintptr_t keep_thr,
+                       uintptr_t table, int64_t table_size, double alpha,
+                       int window, int n_neg, uint64_t seed,
+                       int64_t sent_id_base, int ref_window, int atomic,
+                       uintptr_t stats, int blocks, uintptr_t stream_ptr) {
+  if (stride % 64 != 0) throw std::runtime_error("stride must be a multiple of 64");
+  const int nc = (int)(stride / 64);
+  if (supported_nc(nc) != nc) throw std::runtime_error("stride/64 not a supported NC");
+  if (table_size <= 0 || table_size > 0xFFFFFFFFLL)
+    throw std::runtime_error("table_size out of range");
+  if (atomic && is_bf16) throw std::runtime_error("atomic updates unsupported for bf16");
+  KernelArgs a{};
+  a.syn0 = (void*)syn0;
+  a.syn1 = (void*)syn1;
+  a.tokens = (const int32_t*)tokens;
+  a.offsets = (const int32_t*)offsets;
+  a.num_sentences = num_sentences;
+  a.keep_thr = (const uint32_t*)keep_thr;
+  a.table = (const int32_t*)table;
+  a.table_size = (uint32_t)table_size;
+  a.alpha = (float)alpha;
+  a.window = window;
+  a.n_neg = n_neg;
+  a.seed = seed;
+  a.sent_id_base = sent_id_base;
+  a.stride = stride;
+  a.ref_window = ref_window;
+  unsigned long long* st = (unsigned long long*)stats;
+  if (st) {
+    a.d_pairs = st + 0;
+    a.d_positives = st + 1;
+    a.d_words = st + 2;
+    a.d_sum_fplus = (double*)(st + 3);
+  }
+  hipStream_t stream = (hipStream_t)stream_ptr;
+  if (is_bf16)
+    launch_train<uint16_t>(a, nc, atomic != 0, blocks, stream);
+  else
+    launch_train<float>(a, nc, atomic != 0, blocks, stream);
+  HIP_CHECK(hipGetLastError());
+}
+
+template <typename T>
+static void launch_pull_average_t(uintptr_t syn0, uintptr_t tokens,
+                                  uintptr_t offsets, int64_t num_sentences,
+                                  int64_t stride, uintptr_t out, int blocks,
+                                  hipStream_t stream) {
+  const int nc = (int)(stride / 64);
+  switch (nc) {
+#define CASE_NC(N)                                                          \
+  case N:                                                                   \
+    hipLaunchKernelGGL((pull_average_kernel<T, N>), dim3(blocks), dim3(256),\
+                       0, stream, (const T*)syn0, (const int32_t*)tokens,   \
+                       (const int32_t*)offsets, num_sentences, stride,      \
+                       (float*)out);                                        \
+    return;
+    FOR_EACH_NC(CASE_NC)
+#undef CASE_NC
+    default:
+      throw std::runtime_error("unsupported NC");
+  }
+}
+
+static void pull_average(uintptr_t syn0, int is_bf16, int64_t stride,
+                         uintptr_t tokens, uintptr_t offsets,
+                         int64_t num_sentences, uintptr_t out, int blocks,
+                         uintptr_t stream_ptr) {
+  hipStream_t stream = (hipStream_t)stream_ptr;
+  if (is_bf16)
+    launch_pull_average_t<uint16_t>(syn0, tokens, offsets, num_sentences,
+                                    stride, out, blocks, stream);
+  else
+    launch_pull_average_t<float>(syn0, tokens, offsets, num_sentences, stride,
+                                 out, blocks, stream);
+  HIP_CHECK(hipGetLastError());
+}
+
+template <typename T>
+static void launch_norms_t(uintptr_t syn0, int64_t vocab, int64_t stride,
+                           uintptr_t out, int blocks, hipStream_t stream) {
+  const int nc = (int)(stride / 64);
+  switch (nc) {
+#define CASE_NC(N)                                                        \
+  case N:                                                                 \
+    hipLaunchKernelGGL((norms_kernel<T, N>), dim3(blocks), dim3(256), 0,  \
+                       stream, (const T*)syn0, vocab, stride, (float*)out);\
+    return;
+    FOR_EACH_NC(CASE_NC)
+#undef CASE_NC
+    default:
+      throw std::runtime_error("unsupported NC");
+  }
+}
+
+static void norms(uintptr_t syn0, int is_bf16, int64_t vocab, int64_t stride,
+                  uintptr_t out, int blocks, uintptr_t stream_ptr) {
+  hipStream_t stream = (hipStream_t)stream_ptr;
+  if (is_bf16)
+    launch_norms_t<uint16_t>(syn0, vocab, stride, out, blocks, stream);
+  else
+    launch_norms_t<float>(syn0, vocab, stride, out, blocks, stream);
+  HIP_CHECK(hipGetLastError());
+}
+
+static int round_stride(int dim) {
+  int nc = supported_nc((dim + 63) / 64);
+  if (nc < 0) throw std::runtime_error("dim too large (max 2048)");
+  return 64 * nc;
+}
+
+PYBIND11_MODULE(_hip_native, m) {
+  m.doc() = "MI355X (gfx950) fused SGNS kernels";
+  m.def("sgns_train", &sgns_train, py::arg("syn0"), py::arg("syn1"),
+        py::arg("is_bf16"), py::arg("stride"), py::arg("tokens"),
+        py::arg("offsets"), py::arg("num_sentences"), py::arg("keep_thr"),
+        py::arg("table"), py::arg("table_size"), py::arg("alpha"),
+        py::arg("window"), py::arg("n_neg"), py::arg("seed"),
+        py::arg("sent_id_base"), py::arg("ref_window"), py::arg("atomic"),
+        py::arg("stats"), py::arg("blocks"), py::arg("stream"));
+  m.def("pull_average", &pull_average);
+  m.def("norms", &norms);
+  m.def("round_stride", &round_stride);
+  m.def("max_sentence_length", []() { return (int)kMaxSent; });
+}

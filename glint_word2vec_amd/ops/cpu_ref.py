@@ -25,7 +25,7 @@ from dataclasses import dataclass
 import numpy as np
 
 from ..config import MAX_EXP
-from ..rng import XorShift64Star
+from ..rng import NEG_BASE, WIN_BASE, draw_u32, keep_threshold, sentence_base
 
 
 def sigmoid_clipped(f: float) -> float:
@@ -62,13 +62,13 @@ def train_batch_oracle(syn0: np.ndarray, syn1: np.ndarray,
     do_subsample = keep_prob is not None
     for s in range(num_sentences):
         sent = tokens[offsets[s]:offsets[s + 1]]
-        rng = XorShift64Star(seed, sent_id_base + s)
+        base = sentence_base(seed, sent_id_base + s)
         # 1. subsample
         if do_subsample:
             kept = []
-            for w in sent:
-                u = rng.next_u32()
-                if u < int(keep_prob[w] * 4294967296.0):
+            for p, w in enumerate(sent):
+                u = draw_u32(base, p)
+                if u <= keep_threshold(float(keep_prob[w])):
                     kept.append(int(w))
         else:
             kept = [int(w) for w in sent]
@@ -76,7 +76,7 @@ def train_batch_oracle(syn0: np.ndarray, syn1: np.ndarray,
         # 2./3. windows and training
         for i in range(L):
             c = kept[i]
-            u = rng.next_u32()
+            u = draw_u32(base, WIN_BASE + i)
             if window_mode == "canonical":
                 b = 1 + (u % window)              # symmetric +-b
                 lo, hi = max(0, i - b), min(L - 1, i + b)
@@ -102,8 +102,9 @@ def train_batch_oracle(syn0: np.ndarray, syn1: np.ndarray,
                 stats.positives += 1
                 stats.sum_fplus += f
                 # negatives
+                kbase = NEG_BASE + (i * (2 * window + 1) + (j - i + window)) * n_neg
                 for k in range(n_neg):
-                    u = rng.next_u32()
+                    u = draw_u32(base, kbase + k)
                     neg = int(table[u % table_size])
                     if neg == t:
                         continue

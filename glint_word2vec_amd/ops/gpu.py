@@ -1,0 +1,171 @@
+"""GPU-resident SGNS state + launch wrappers around the fused HIP kernels.
+
+The embedding tables live in HBM as [vocab, stride] tensors (stride = dim
+rounded up to the kernel's supported chunk counts); padding columns are
+zero-initialised and provably stay zero (every update scales existing row
+values), so dots over the padded width are exact.
+
+On a machine with a GPU the HIP extension is REQUIRED: there is no silent
+eager/torch fallback — a missing or broken _hip_native raises immediately.
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass
+from typing import Optional
+
+import numpy as np
+import torch
+
+from ..config import Word2VecConfig
+from ..vocab import keep_probabilities
+
+
+def _load_native():
+    try:
+        from .. import _hip_native
+        return _hip_native
+    except ImportError as e:
+        raise RuntimeError(
+            "glint_word2vec_amd._hip_native is not built but a GPU path was "
+            "requested. Build it in-tree with `python setup.py build_ext "
+            "--inplace` (hipcc, gfx950). Refusing to fall back to eager."
+        ) from e
+
+
+def round_stride(dim: int) -> int:
+    return _load_native().round_stride(dim)
+
+
+@dataclass
+class GpuStats:
+    pairs: int
+    positives: int
+    words_trained: int
+    sum_fplus: float
+
+
+class GpuSgns:
+    """Owns the device tables and launches the fused kernels.
+
+    The full-table (non-sharded) case: both matrices resident on one GPU —
+    BASELINE.json config 2.  The sharded engine (parallel/engine.py) holds
+    one GpuSgns per rank over its row shard plus staging buffers.
+    """
+
+    def __init__(self, vocab_size: int, dim: int, dtype: str = "float32",
+                 device: str = "cuda", seed: int = 1,
+                 init: str = "word2vec",
+                 syn0_host: Optional[np.ndarray] = None,
+                 syn1_host: Optional[np.ndarray] = None):
+        self.native = _load_native()
+        self.device = torch.device(device)
+        self.vocab_size = vocab_size
+        self.dim = dim
+        self.stride = self.native.round_stride(dim)
+        self.is_bf16 = dtype == "bfloat16"
+        tdtype = torch.bfloat16 if self.is_bf16 else torch.float32
+        self.syn0 = torch.zeros((vocab_size, self.stride), dtype=tdtype,
+                                device=self.device)
+        self.syn1 = torch.zeros((vocab_size, self.stride), dtype=tdtype,
+                                device=self.device)
+        if syn0_host is not None:
+            self.load_host(syn0_host, syn1_host)
+        elif init == "word2vec":
+            g = torch.Generator(device="cpu").manual_seed(seed)
+            w = (torch.rand((vocab_size, dim), generator=g) - 0.5) / dim
+            self.syn0[:, :dim] = w.to(tdtype).to(self.device)
+        # stats buffer: [pairs u64, positives u64, words u64, sum_fplus f64]
+        self._stats = torch.zeros(4, dtype=torch.int64, device=self.device)
+        self.keep_thr: Optional[torch.Tensor] = None
+        self.table: Optional[torch.Tensor] = None
+
+    # --- state management --------------------------------------------------
+    def load_host(self, syn0: np.ndarray, syn1: Optional[np.ndarray]) -> None:
+        tdtype = self.syn0.dtype
+        s0 = torch.from_numpy(np.ascontiguousarray(syn0, dtype=np.float32))
+        self.syn0[:, :self.dim] = s0.to(tdtype).to(self.device)
+        if syn1 is not None:
+            s1 = torch.from_numpy(np.ascontiguousarray(syn1, dtype=np.float32))
+            self.syn1[:, :self.dim] = s1.to(tdtype).to(self.device)
+
+    def to_host(self) -> tuple[np.ndarray, np.ndarray]:
+        s0 = self.syn0[:, :self.dim].float().cpu().numpy()
+        s1 = self.syn1[:, :self.dim].float().cpu().numpy()
+        return s0, s1
+
+    def set_subsample(self, counts: np.ndarray, train_words: int,
+                      ratio: float) -> None:
+        if ratio <= 0:
+            self.keep_thr = None
+            return
+        kp = keep_probabilities(counts, train_words, ratio).astype(np.float64)
+        thr = np.minimum(kp * 4294967296.0, 4294967295.0).astype(np.uint32)
+        self.keep_thr = torch.from_numpy(thr.view(np.int32)).to(self.device)
+
+    def set_table(self, table: np.ndarray) -> None:
+        self.table = torch.from_numpy(np.ascontiguousarray(table, dtype=np.int32)) \
+            .to(self.device)
+
+    # --- ops ---------------------------------------------------------------
+    def train_batch(self, tokens: torch.Tensor, offsets: torch.Tensor,
+                    alpha: float, window: int, n_neg: int, seed: int,
+                    sent_id_base: int = 0, window_mode: str = "canonical",
+                    atomic: bool = False, blocks: Optional[int] = None,
+                    serial: bool = False,
+                    stream: Optional[torch.cuda.Stream] = None) -> None:
+        """Launch the fused train kernel (async on the given/current stream).
+        Stats accumulate on-device; read with read_stats()."""
+        assert self.table is not None, "call set_table first"
+        num_sent = int(offsets.numel() - 1)
+        if num_sent <= 0:
+            return
+        if serial:
+            nblocks = 1
+        elif blocks is None:
+            # 4 waves per block; oversubscribe the 256 CUs
+            nblocks = max(1, min((num_sent + 3) // 4, 8192))
+        else:
+            nblocks = blocks
+        s = stream if stream is not None else torch.cuda.current_stream(self.device)
+        self.native.sgns_train(
+            self.syn0.data_ptr(), self.syn1.data_ptr(), int(self.is_bf16),
+            self.stride, tokens.data_ptr(), offsets.data_ptr(), num_sent,
+            0 if self.keep_thr is None else self.keep_thr.data_ptr(),
+            self.table.data_ptr(), int(self.table.numel()), float(alpha),
+            int(window), int(n_neg), seed & 0xFFFFFFFFFFFFFFFF,
+            int(sent_id_base), int(window_mode == "reference"), int(atomic),
+            self._stats.data_ptr(), nblocks, s.cuda_stream)
+
+    def read_stats(self, reset: bool = True) -> GpuStats:
+        h = self._stats.cpu()
+        pairs, pos, words = int(h[0]), int(h[1]), int(h[2])
+        sum_fplus = float(h[3:4].view(torch.float64)[0])
+        if reset:
+            self._stats.zero_()
+        return GpuStats(pairs, pos, words, sum_fplus)
+
+    def pull_average(self, tokens: torch.Tensor, offsets: torch.Tensor) -> torch.Tensor:
+        num_sent = int(offsets.numel() - 1)
+        out = torch.empty((num_sent, self.stride), dtype=torch.float32,
+                          device=self.device)
+        s = torch.cuda.current_stream(self.device)
+        self.native.pull_average(
+            self.syn0.data_ptr(), int(self.is_bf16), self.stride,
+            tokens.data_ptr(), offsets.data_ptr(), num_sent, out.data_ptr(),
+            max(1, min((num_sent + 3) // 4, 8192)), s.cuda_stream)
+        return out[:, :self.dim]
+
+    def norms(self) -> torch.Tensor:
+        out = torch.empty(self.vocab_size, dtype=torch.float32, device=self.device)
+        s = torch.cuda.current_stream(self.device)
+        self.native.norms(self.syn0.data_ptr(), int(self.is_bf16),
+                          self.vocab_size, self.stride, out.data_ptr(),
+                          2048, s.cuda_stream)
+        return out
+
+    def multiply(self, vec: torch.Tensor) -> torch.Tensor:
+        """Whole-table GEMV — a plain library GEMM-shaped op: rocBLAS via
+        torch.matmul is the right tool (guide: hand-write only fused ops)."""
+        v = torch.zeros(self.stride, dtype=self.syn0.dtype, device=self.device)
+        v[:self.dim] = vec.to(self.syn0.dtype)
+        return (self.syn0 @ v).float()

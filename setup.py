@@ -32,10 +32,40 @@ cpu_ext = Extension(
 )
 
 
+def build_hip_extension(force: bool = False) -> str | None:
+    """Compile the HIP extension in-tree with hipcc for gfx950.
+
+    Returns the path of the built .so, or None when hipcc is unavailable.
+    Deliberately bypasses torch's JIT extension cache: the .so must live
+    in-tree so it travels with the repo snapshot to the GPU box.
+    """
+    hipcc = shutil.which("hipcc") or "/opt/rocm/bin/hipcc"
+    if not os.path.exists(hipcc):
+        return None
+    import sysconfig
+    ext_suffix = sysconfig.get_config_var("EXT_SUFFIX") or ".so"
+    src = os.path.join(ROOT, "csrc", "hip", "sgns_hip.hip")
+    out = os.path.join(PKG, "_hip_native" + ext_suffix)
+    if not force and os.path.exists(out) and \
+            os.path.getmtime(out) > os.path.getmtime(src):
+        return out
+    cmd = [
+        hipcc, "--offload-arch=gfx950", "-O3", "-std=c++17", "-fPIC",
+        "-shared", "-fvisibility=hidden",
+        "-I", pybind11.get_include(),
+        "-I", sysconfig.get_paths()["include"],
+        "-x", "hip", src,
+        "-o", out,
+    ]
+    print("building HIP extension:", " ".join(cmd))
+    subprocess.check_call(cmd, cwd=ROOT)
+    return out
+
+
 class BuildExt(build_ext):
     def run(self):
         super().run()
-        # copy built .so in-tree (--inplace does this already; keep both paths safe)
+        build_hip_extension()
 
 
 setup(

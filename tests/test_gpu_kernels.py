@@ -1,0 +1,179 @@
+"""Numerics tests for the fused HIP kernels vs the plain-fp32 CPU oracle.
+All tests here require an MI355X (marked gpu)."""
+import numpy as np
+import pytest
+
+import torch
+
+from glint_word2vec_amd.models import sgns
+from glint_word2vec_amd.ops import cpu_ref
+from glint_word2vec_amd.vocab import build_unigram_table, keep_probabilities
+
+pytestmark = pytest.mark.gpu
+
+
+def _problem(vocab=50, dim=20, n_tokens=150, seed=11, sentences=6):
+    rng = np.random.default_rng(seed)
+    tokens = rng.integers(0, vocab, n_tokens).astype(np.int32)
+    offsets = np.linspace(0, n_tokens, sentences + 1).astype(np.int32)
+    counts = np.bincount(tokens, minlength=vocab).astype(np.int64) + 1
+    table = build_unigram_table(counts, 1009)
+    syn0, syn1 = sgns.init_tables(vocab, dim, seed)
+    return tokens, offsets, counts, table, syn0, syn1
+
+
+def _gpu_setup(syn0, syn1, table, dtype="float32"):
+    from glint_word2vec_amd.ops.gpu import GpuSgns
+    gs = GpuSgns(syn0.shape[0], syn0.shape[1], dtype=dtype, device="cuda")
+    gs.load_host(syn0, syn1)
+    gs.set_table(table)
+    return gs
+
+
+def _to_dev(a):
+    return torch.from_numpy(a).cuda()
+
+
+@pytest.mark.parametrize("window_mode", ["canonical", "reference"])
+def test_serial_kernel_matches_oracle(window_mode):
+    tokens, offsets, counts, table, syn0, syn1 = _problem()
+    a0, a1 = syn0.copy(), syn1.copy()
+    st_py = cpu_ref.train_batch_oracle(a0, a1, tokens, offsets, None, table,
+                                       0.03, 3, 4, seed=77, sent_id_base=5,
+                                       window_mode=window_mode)
+    gs = _gpu_setup(syn0, syn1, table)
+    gs.train_batch(_to_dev(tokens), _to_dev(offsets), 0.03, 3, 4, 77,
+                   sent_id_base=5, window_mode=window_mode, serial=True)
+    torch.cuda.synchronize()
+    st = gs.read_stats()
+    assert st.pairs == st_py.pairs
+    assert st.positives == st_py.positives
+    assert st.words_trained == st_py.words_trained
+    assert st.sum_fplus == pytest.approx(st_py.sum_fplus, rel=1e-3, abs=1e-3)
+    g0, g1 = gs.to_host()
+    np.testing.assert_allclose(g0, a0, rtol=2e-4, atol=2e-6)
+    np.testing.assert_allclose(g1, a1, rtol=2e-4, atol=2e-6)
+
+
+def test_serial_kernel_with_subsampling():
+    tokens, offsets, counts, table, syn0, syn1 = _problem()
+    total = int(counts.sum())
+    kp = keep_probabilities(counts, total, 0.3)
+    assert (kp < 1).any()
+    a0, a1 = syn0.copy(), syn1.copy()
+    st_py = cpu_ref.train_batch_oracle(a0, a1, tokens, offsets, kp, table,
+                                       0.03, 3, 4, seed=13)
+    gs = _gpu_setup(syn0, syn1, table)
+    gs.set_subsample(counts, total, 0.3)
+    gs.train_batch(_to_dev(tokens), _to_dev(offsets), 0.03, 3, 4, 13,
+                   serial=True)
+    torch.cuda.synchronize()
+    st = gs.read_stats()
+    assert st.pairs == st_py.pairs
+    assert st.words_trained == st_py.words_trained
+    g0, g1 = gs.to_host()
+    np.testing.assert_allclose(g0, a0, rtol=2e-4, atol=2e-6)
+
+
+def test_parallel_kernel_exact_counts_stable_values():
+    """Hogwild launch: RNG-driven pair counts are schedule-independent and
+    must match the oracle exactly; values race but must stay finite/close
+    in aggregate."""
+    tokens, offsets, counts, table, syn0, syn1 = _problem(
+        vocab=200, dim=32, n_tokens=4000, sentences=50)
+    a0, a1 = syn0.copy(), syn1.copy()
+    st_py = cpu_ref.train_batch_oracle(a0, a1, tokens, offsets, None, table,
+                                       0.03, 4, 5, seed=3)
+    gs = _gpu_setup(syn0, syn1, table)
+    gs.train_batch(_to_dev(tokens), _to_dev(offsets), 0.03, 4, 5, 3)
+    torch.cuda.synchronize()
+    st = gs.read_stats()
+    assert st.pairs == st_py.pairs
+    assert st.positives == st_py.positives
+    g0, g1 = gs.to_host()
+    assert np.isfinite(g0).all() and np.isfinite(g1).all()
+    # aggregate movement should be similar magnitude to the oracle's
+    assert np.linalg.norm(g0 - syn0) == pytest.approx(
+        np.linalg.norm(a0 - syn0), rel=0.5)
+
+
+def test_atomic_variant_fp32():
+    tokens, offsets, counts, table, syn0, syn1 = _problem()
+    gs = _gpu_setup(syn0, syn1, table)
+    gs.train_batch(_to_dev(tokens), _to_dev(offsets), 0.03, 3, 4, 77,
+                   atomic=True)
+    torch.cuda.synchronize()
+    st = gs.read_stats()
+    assert st.pairs > 0
+    g0, _ = gs.to_host()
+    assert np.isfinite(g0).all()
+    assert not np.allclose(g0, syn0)
+
+
+def test_bf16_kernel_counts_and_direction():
+    tokens, offsets, counts, table, syn0, syn1 = _problem(dim=64)
+    a0, a1 = syn0.copy(), syn1.copy()
+    st_py = cpu_ref.train_batch_oracle(a0, a1, tokens, offsets, None, table,
+                                       0.03, 3, 4, seed=77)
+    gs = _gpu_setup(syn0, syn1, table, dtype="bfloat16")
+    gs.train_batch(_to_dev(tokens), _to_dev(offsets), 0.03, 3, 4, 77,
+                   serial=True)
+    torch.cuda.synchronize()
+    st = gs.read_stats()
+    assert st.pairs == st_py.pairs      # RNG identical regardless of dtype
+    g0, g1 = gs.to_host()
+    # bf16 storage: loose agreement with the fp32 oracle
+    assert np.isfinite(g0).all()
+    diff_gpu = g0 - syn0.astype(np.float32)
+    diff_ref = a0 - syn0
+    # update directions should correlate strongly
+    num = (diff_gpu * diff_ref).sum()
+    den = np.linalg.norm(diff_gpu) * np.linalg.norm(diff_ref)
+    assert den > 0 and num / den > 0.98
+
+
+def test_pull_average_matches_numpy():
+    tokens, offsets, counts, table, syn0, syn1 = _problem()
+    gs = _gpu_setup(syn0, syn1, table)
+    out = gs.pull_average(_to_dev(tokens), _to_dev(offsets))
+    torch.cuda.synchronize()
+    sentences = [tokens[offsets[i]:offsets[i + 1]] for i in range(len(offsets) - 1)]
+    ref = cpu_ref.pull_average(syn0, sentences)
+    np.testing.assert_allclose(out.cpu().numpy(), ref, rtol=1e-5, atol=1e-7)
+
+
+def test_norms_matches_numpy():
+    tokens, offsets, counts, table, syn0, syn1 = _problem(vocab=300, dim=45)
+    gs = _gpu_setup(syn0, syn1, table)
+    out = gs.norms().cpu().numpy()
+    ref = cpu_ref.norms(syn0)
+    np.testing.assert_allclose(out, ref, rtol=1e-5, atol=1e-7)
+
+
+def test_multiply_matches_numpy():
+    tokens, offsets, counts, table, syn0, syn1 = _problem(vocab=128, dim=64)
+    gs = _gpu_setup(syn0, syn1, table)
+    rng = np.random.default_rng(0)
+    v = rng.standard_normal(64).astype(np.float32)
+    out = gs.multiply(torch.from_numpy(v).cuda()).cpu().numpy()
+    ref = cpu_ref.multiply(syn0, v)
+    np.testing.assert_allclose(out, ref, rtol=1e-4, atol=1e-5)
+
+
+def test_estimator_end_to_end_gpu():
+    rng = np.random.default_rng(5)
+    sents = []
+    for _ in range(600):
+        head = "aa" if rng.random() < 0.5 else "a2"
+        filler = [f"x{rng.integers(0, 20)}" for _ in range(3)]
+        sents.append([head, "bb"] + filler + [head, "bb"])
+    from glint_word2vec_amd import GlintWord2Vec
+    est = (GlintWord2Vec().setVectorSize(32).setMinCount(1).setSeed(4)
+           .setNumIterations(6).setWindowSize(2).setN(5)
+           .setUnigramTableSize(100000).setStepSize(0.05)
+           .setSubsampleRatio(0.0))
+    est.config.device = "cuda"
+    m = est.fit(sents)
+    assert np.isfinite(m.syn0).all()
+    syns = [w for w, _ in m.find_synonyms("aa", 3)]
+    assert "a2" in syns

@@ -1,0 +1,52 @@
+"""C++ native trainer vs Python oracle: identical RNG draws (exact pair
+counts) and matching table updates (fp tolerance for reassociation)."""
+import numpy as np
+import pytest
+
+from glint_word2vec_amd.models import sgns
+from glint_word2vec_amd.ops import cpu_ref
+from glint_word2vec_amd.vocab import build_unigram_table, keep_probabilities
+
+native = pytest.importorskip("glint_word2vec_amd._cpu_native")
+
+
+def _problem(vocab=30, dim=12, n_tokens=80, seed=3):
+    rng = np.random.default_rng(seed)
+    tokens = rng.integers(0, vocab, n_tokens).astype(np.int32)
+    offsets = np.array([0, 25, 60, n_tokens], dtype=np.int32)
+    counts = np.bincount(tokens, minlength=vocab).astype(np.int64) + 1
+    table = build_unigram_table(counts, 997)
+    syn0, syn1 = sgns.init_tables(vocab, dim, seed)
+    return tokens, offsets, counts, table, syn0, syn1
+
+
+@pytest.mark.parametrize("window_mode", ["canonical", "reference"])
+@pytest.mark.parametrize("subsample", [False, True])
+def test_cpp_matches_python_oracle(window_mode, subsample):
+    tokens, offsets, counts, table, syn0, syn1 = _problem()
+    kp = keep_probabilities(counts, int(counts.sum()), 0.2) if subsample else None
+    a0, a1 = syn0.copy(), syn1.copy()
+    b0, b1 = syn0.copy(), syn1.copy()
+    st_py = cpu_ref.train_batch_oracle(a0, a1, tokens, offsets, kp, table,
+                                       0.03, 3, 4, seed=99, sent_id_base=7,
+                                       window_mode=window_mode)
+    st_c = native.train_batch(b0, b1, tokens, offsets, kp, table,
+                              0.03, 3, 4, 99, 7, window_mode, 1)
+    assert st_c["pairs"] == st_py.pairs
+    assert st_c["positives"] == st_py.positives
+    assert st_c["words_trained"] == st_py.words_trained
+    assert st_c["sum_fplus"] == pytest.approx(st_py.sum_fplus, rel=1e-4, abs=1e-4)
+    np.testing.assert_allclose(a0, b0, rtol=1e-5, atol=1e-7)
+    np.testing.assert_allclose(a1, b1, rtol=1e-5, atol=1e-7)
+
+
+def test_cpp_multithread_stats_close():
+    """Hogwild threads race on rows but must process the same pair count."""
+    tokens, offsets, counts, table, syn0, syn1 = _problem(n_tokens=200)
+    offsets = np.arange(0, 201, 20, dtype=np.int32)
+    st1 = native.train_batch(syn0.copy(), syn1.copy(), tokens, offsets, None,
+                             table, 0.03, 3, 4, 5, 0, "canonical", 1)
+    st4 = native.train_batch(syn0.copy(), syn1.copy(), tokens, offsets, None,
+                             table, 0.03, 3, 4, 5, 0, "canonical", 4)
+    assert st1["pairs"] == st4["pairs"]
+    assert st1["positives"] == st4["positives"]

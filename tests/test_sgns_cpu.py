@@ -5,22 +5,29 @@ import pytest
 
 from glint_word2vec_amd.models import sgns
 from glint_word2vec_amd.ops import cpu_ref
-from glint_word2vec_amd.rng import XorShift64Star, splitmix64
+from glint_word2vec_amd.rng import (draw_u32, keep_threshold, sentence_base,
+                                    splitmix64)
 from glint_word2vec_amd.vocab import build_unigram_table
 
 
-def test_rng_deterministic_and_nonzero():
-    a = XorShift64Star(123, 0)
-    b = XorShift64Star(123, 0)
-    c = XorShift64Star(123, 1)
-    seq_a = [a.next_u32() for _ in range(100)]
-    seq_b = [b.next_u32() for _ in range(100)]
-    seq_c = [c.next_u32() for _ in range(100)]
-    assert seq_a == seq_b
-    assert seq_a != seq_c
-    assert all(0 <= x < 2 ** 32 for x in seq_a)
+def test_rng_deterministic_and_uniform():
+    b0 = sentence_base(123, 0)
+    b1 = sentence_base(123, 1)
+    assert b0 == sentence_base(123, 0)
+    assert b0 != b1
+    seq = [draw_u32(b0, k) for k in range(200)]
+    assert seq == [draw_u32(b0, k) for k in range(200)]
+    assert all(0 <= x < 2 ** 32 for x in seq)
+    assert len(set(seq)) == len(seq)
     # crude uniformity
-    assert 0.4 < np.mean(np.asarray(seq_a) / 2 ** 32) < 0.6
+    assert 0.4 < np.mean(np.asarray(seq, dtype=np.float64) / 2 ** 32) < 0.6
+
+
+def test_keep_threshold_clamp():
+    assert keep_threshold(1.0) == 2 ** 32 - 1
+    assert keep_threshold(2.0) == 2 ** 32 - 1
+    assert keep_threshold(0.0) == 0
+    assert keep_threshold(0.5) == 2 ** 31
 
 
 def test_sigmoid_clip():
