@@ -145,7 +145,9 @@ struct RowIO<uint16_t, NC> {
 __device__ __forceinline__ float sigmoid_clipped(float f) {
   if (f > 6.0f) return 1.0f;
   if (f < -6.0f) return 0.0f;
-  return 1.0f / (1.0f + __expf(-f));
+  // precise expf (not __expf): one call per pair per wave — cost is
+  // negligible and it keeps parity with the CPU oracle tight
+  return 1.0f / (1.0f + expf(-f));
 }
 
 struct KernelArgs {
@@ -179,8 +181,9 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void sgns_train_kernel(KernelA
   __shared__ int32_t sbuf[kWavesPerBlock][kMaxSent];
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
-  const int64_t wave_gid = (int64_t)blockIdx.x * kWavesPerBlock + wave;
-  const int64_t total_waves = (int64_t)gridDim.x * kWavesPerBlock;
+  const int waves_in_block = blockDim.x >> 6;   // 1 in serial mode, else 4
+  const int64_t wave_gid = (int64_t)blockIdx.x * waves_in_block + wave;
+  const int64_t total_waves = (int64_t)gridDim.x * waves_in_block;
   int32_t* sent_lds = sbuf[wave];
 
   T* syn0 = (T*)a.syn0;
@@ -374,13 +377,13 @@ __global__ __launch_bounds__(256) void norms_kernel(
 // ---------------------------------------------------------------------------
 template <typename T, int NC>
 static void launch_train_nc(const KernelArgs& a, bool atomic, int blocks,
-                            hipStream_t stream) {
+                            int threads, hipStream_t stream) {
   if (atomic)
     hipLaunchKernelGGL((sgns_train_kernel<T, NC, true>), dim3(blocks),
-                       dim3(64 * kWavesPerBlock), 0, stream, a);
+                       dim3(threads), 0, stream, a);
   else
     hipLaunchKernelGGL((sgns_train_kernel<T, NC, false>), dim3(blocks),
-                       dim3(64 * kWavesPerBlock), 0, stream, a);
+                       dim3(threads), 0, stream, a);
 }
 
 // Supported NC values (stride = 64*NC).  The wrapper rounds the row stride
@@ -390,11 +393,11 @@ static void launch_train_nc(const KernelArgs& a, bool atomic, int blocks,
 
 template <typename T>
 static void launch_train(const KernelArgs& a, int nc, bool atomic, int blocks,
-                         hipStream_t stream) {
+                         int threads, hipStream_t stream) {
   switch (nc) {
-#define CASE_NC(N)                                        \
-  case N:                                                 \
-    launch_train_nc<T, N>(a, atomic, blocks, stream);     \
+#define CASE_NC(N)                                             \
+  case N:                                                      \
+    launch_train_nc<T, N>(a, atomic, blocks, threads, stream); \
     return;
     FOR_EACH_NC(CASE_NC)
 #undef CASE_NC
@@ -416,7 +419,10 @@ static void sgns_train(uintptr_t syn0, uintptr_t syn1, int is_bf16,
                        uintptr_t table, int64_t table_size, double alpha,
                        int window, int n_neg, uint64_t seed,
                        int64_t sent_id_base, int ref_window, int atomic,
-                       uintptr_t stats, int blocks, uintptr_t stream_ptr) {
+                       uintptr_t stats, int blocks, int threads,
+                       uintptr_t stream_ptr) {
+  if (threads != 64 && threads != 256)
+    throw std::runtime_error("threads must be 64 (serial) or 256");
   if (stride % 64 != 0) throw std::runtime_error("stride must be a multiple of 64");
   const int nc = (int)(stride / 64);
   if (supported_nc(nc) != nc) throw std::runtime_error("stride/64 not a supported NC");
@@ -448,9 +454,9 @@ static void sgns_train(uintptr_t syn0, uintptr_t syn1, int is_bf16,
   }
   hipStream_t stream = (hipStream_t)stream_ptr;
   if (is_bf16)
-    launch_train<uint16_t>(a, nc, atomic != 0, blocks, stream);
+    launch_train<uint16_t>(a, nc, atomic != 0, blocks, threads, stream);
   else
-    launch_train<float>(a, nc, atomic != 0, blocks, stream);
+    launch_train<float>(a, nc, atomic != 0, blocks, threads, stream);
   HIP_CHECK(hipGetLastError());
 }
 
@@ -530,7 +536,8 @@ PYBIND11_MODULE(_hip_native, m) {
         py::arg("table"), py::arg("table_size"), py::arg("alpha"),
         py::arg("window"), py::arg("n_neg"), py::arg("seed"),
         py::arg("sent_id_base"), py::arg("ref_window"), py::arg("atomic"),
-        py::arg("stats"), py::arg("blocks"), py::arg("stream"));
+        py::arg("stats"), py::arg("blocks"), py::arg("threads"),
+        py::arg("stream"));
   m.def("pull_average", &pull_average);
   m.def("norms", &norms);
   m.def("round_stride", &round_stride);

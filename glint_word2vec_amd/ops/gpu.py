@@ -120,12 +120,14 @@ class GpuSgns:
         if num_sent <= 0:
             return
         if serial:
-            nblocks = 1
-        elif blocks is None:
-            # 4 waves per block; oversubscribe the 256 CUs
-            nblocks = max(1, min((num_sent + 3) // 4, 8192))
+            nblocks, nthreads = 1, 64        # exactly one wave: oracle order
         else:
-            nblocks = blocks
+            nthreads = 256
+            if blocks is None:
+                # 4 waves per block; oversubscribe the 256 CUs
+                nblocks = max(1, min((num_sent + 3) // 4, 8192))
+            else:
+                nblocks = blocks
         s = stream if stream is not None else torch.cuda.current_stream(self.device)
         self.native.sgns_train(
             self.syn0.data_ptr(), self.syn1.data_ptr(), int(self.is_bf16),
@@ -134,7 +136,7 @@ class GpuSgns:
             self.table.data_ptr(), int(self.table.numel()), float(alpha),
             int(window), int(n_neg), seed & 0xFFFFFFFFFFFFFFFF,
             int(sent_id_base), int(window_mode == "reference"), int(atomic),
-            self._stats.data_ptr(), nblocks, s.cuda_stream)
+            self._stats.data_ptr(), nblocks, nthreads, s.cuda_stream)
 
     def read_stats(self, reset: bool = True) -> GpuStats:
         h = self._stats.cpu()
