@@ -68,17 +68,20 @@ def main():
 
     # --- synthetic data: one step's batch, reused with varying sent_id_base
     # (RNG/negative draws differ per step; token stream is Zipf-realistic).
-    batch = synthetic_corpus(args.vocab, args.words_per_step,
-                             sentence_len=args.sentence_len,
-                             seed=1234 + rank)
+    # Dim-sharded (distributed): every rank walks the SAME global batch of
+    # world*words_per_step tokens, each computing its dim-slice -> per-GPU
+    # work is fixed as N grows (weak scaling).
+    batch = synthetic_corpus(args.vocab,
+                             args.words_per_step * (world if distributed else 1),
+                             sentence_len=args.sentence_len, seed=1234)
     counts = np.bincount(batch.tokens, minlength=args.vocab).astype(np.int64) + 1
 
     if distributed:
-        from glint_word2vec_amd.parallel.sharded import ShardedSgns
-        eng = ShardedSgns(args.vocab, args.dim, dtype=dtype, device=str(device),
-                          seed=1, counts=counts, table_size=args.table_size,
-                          subsample=args.subsample)
-        trainer = eng
+        from glint_word2vec_amd.parallel.dim_sharded import DimShardedSgns
+        trainer = DimShardedSgns(args.vocab, args.dim, dtype=dtype,
+                                 device=str(device), seed=1, counts=counts,
+                                 table_size=args.table_size,
+                                 subsample=args.subsample)
     else:
         from glint_word2vec_amd.ops.gpu import GpuSgns
         gs = GpuSgns(args.vocab, args.dim, dtype=dtype, device=str(device), seed=1)
@@ -96,7 +99,8 @@ def main():
     def step(i):
         if distributed:
             trainer.train_step(tok, off, alpha, args.window, args.neg,
-                               seed=99, sent_id_base=i * nsent)
+                               seed=99, sent_id_base=i * nsent,
+                               offsets_host=batch.offsets)
         else:
             trainer.train_batch(tok, off, alpha, args.window, args.neg, 99,
                                 sent_id_base=i * nsent,
@@ -157,8 +161,8 @@ def main():
                          f"neg={args.neg} window={args.window}",
                 "global_batch": args.words_per_step * world,
                 "seq_len": args.sentence_len,
-                "parallelism": (f"rowshard-alltoallv-dp{world}" if distributed
-                                else "hogwild-1gpu"),
+                "parallelism": (f"dimshard-rccl-allreduce-x{world}"
+                                if distributed else "hogwild-1gpu"),
             },
             "pairs_per_step": st.pairs / max(args.steps, 1),
             "mean_fplus": st.sum_fplus / max(st.positives, 1),

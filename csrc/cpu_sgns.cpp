@@ -152,6 +152,309 @@ static void train_sentences(float* syn0, float* syn1, int64_t dim,
   }
 }
 
+// ---------------------------------------------------------------------------
+// Shared sentence walker for the dimension-sharded engine:
+// every phase (count / partial-dot / update) re-derives the identical pair
+// enumeration from the counter-based RNG, so ranks need no coordination.
+// Phase must provide:
+//   void begin_position(int32_t c);
+//   void pair(int32_t c, int32_t tgt, float label, int64_t pair_idx);
+//   void end_position(int32_t c, bool trained);
+// pair_idx counts emitted pairs within the sentence, in walk order.
+// ---------------------------------------------------------------------------
+template <typename Phase>
+static void walk_sentence(const int32_t* sent, int64_t len, uint64_t base,
+                          const float* keep_prob, const int32_t* table,
+                          int64_t table_size, int window, int n_neg,
+                          bool reference_window, std::vector<int32_t>& kept,
+                          Phase& ph) {
+  kept.clear();
+  if (keep_prob) {
+    for (int64_t p = 0; p < len; ++p) {
+      uint32_t u = draw_u32(base, (uint64_t)p);
+      int32_t w = sent[p];
+      if (u <= keep_thr(keep_prob[w])) kept.push_back(w);
+    }
+  } else {
+    kept.assign(sent, sent + len);
+  }
+  int64_t L = (int64_t)kept.size();
+  int64_t pair_idx = 0;
+  for (int64_t i = 0; i < L; ++i) {
+    int32_t c = kept[i];
+    uint32_t u = draw_u32(base, kWinBase + (uint64_t)i);
+    int64_t lo, hi;
+    if (!reference_window) {
+      int64_t b = 1 + (int64_t)(u % (uint32_t)window);
+      lo = i - b < 0 ? 0 : i - b;
+      hi = i + b >= L ? L - 1 : i + b;
+    } else {
+      int64_t b = (int64_t)(u % (uint32_t)window);
+      if (b == 0) { lo = i; hi = i; }
+      else {
+        lo = i - b < 0 ? 0 : i - b;
+        hi = i + b - 1 >= L ? L - 1 : i + b - 1;
+      }
+    }
+    bool any = false;
+    for (int64_t j = lo; j <= hi; ++j) if (j != i) { any = true; break; }
+    if (!any) continue;
+    ph.begin_position(c);
+    for (int64_t j = lo; j <= hi; ++j) {
+      if (j == i) continue;
+      int32_t t = kept[j];
+      ph.pair(c, t, 1.0f, pair_idx++);
+      uint64_t kbase = kNegBase +
+          (uint64_t)(i * (2 * window + 1) + (j - i + window)) * (uint64_t)n_neg;
+      for (int k = 0; k < n_neg; ++k) {
+        uint32_t un = draw_u32(base, kbase + (uint64_t)k);
+        int32_t neg = table[un % (uint64_t)table_size];
+        if (neg == t) continue;
+        ph.pair(c, neg, 0.0f, pair_idx++);
+      }
+    }
+    ph.end_position(c, true);
+  }
+}
+
+// Phase: count pairs only.
+struct CountPhase {
+  int64_t pairs = 0;
+  void begin_position(int32_t) {}
+  void pair(int32_t, int32_t, float, int64_t) { ++pairs; }
+  void end_position(int32_t, bool) {}
+};
+
+// Phase: partial dot over a dim-slice.
+struct DotPhase {
+  const float* s0;
+  const float* s1;
+  int64_t width;
+  float* f_out;   // sentence-local base
+  void begin_position(int32_t) {}
+  void pair(int32_t c, int32_t tgt, float, int64_t idx) {
+    const float* a = s0 + (int64_t)c * width;
+    const float* b = s1 + (int64_t)tgt * width;
+    float f = 0.0f;
+    for (int64_t d = 0; d < width; ++d) f += a[d] * b[d];
+    f_out[idx] = f;
+  }
+  void end_position(int32_t, bool) {}
+};
+
+// Phase: apply updates to a dim-slice using precomputed full dots.
+// With f_correction (world_scale > 0), the stale allreduced dot is
+// freshened by extrapolating this rank's local drift:
+//   f_used = f_total + world * (local_partial_now - local_partial_at_pass1)
+// At world=1 this is exactly the sequential (fused-kernel) semantics; at
+// world>1 it restores bounded sigmoid feedback within a chunk.
+struct UpdatePhase {
+  float* s0;
+  float* s1;
+  int64_t width;
+  const float* f_in;      // sentence-local base (full, allreduced dots)
+  const float* f_loc;     // sentence-local base (pass-1 local partials); may be null
+  float world_scale;      // world size as float; 0 = correction off
+  float alpha;
+  Stats* stats;
+  std::vector<float>* c_row;
+  std::vector<float>* grad;
+  void begin_position(int32_t c) {
+    std::memcpy(c_row->data(), s0 + (int64_t)c * width, width * sizeof(float));
+    std::memset(grad->data(), 0, width * sizeof(float));
+  }
+  void pair(int32_t, int32_t tgt, float label, int64_t idx) {
+    float* t1 = s1 + (int64_t)tgt * width;
+    float* cr = c_row->data();
+    float f = f_in[idx];
+    if (f_loc) {
+      float fresh = 0.0f;
+      for (int64_t d = 0; d < width; ++d) fresh += cr[d] * t1[d];
+      f += world_scale * (fresh - f_loc[idx]);
+    }
+    float g = (label - sigmoid_clipped(f)) * alpha;
+    float* gr = grad->data();
+    for (int64_t d = 0; d < width; ++d) {
+      gr[d] += g * t1[d];
+      t1[d] += g * cr[d];
+    }
+    stats->pairs++;
+    if (label > 0.5f) { stats->positives++; stats->sum_fplus += f; }
+  }
+  void end_position(int32_t c, bool) {
+    float* c0 = s0 + (int64_t)c * width;
+    const float* gr = grad->data();
+    for (int64_t d = 0; d < width; ++d) c0[d] += gr[d];
+    stats->words_trained++;
+  }
+};
+
+static py::array_t<int64_t> count_pairs(
+    py::array_t<int32_t, py::array::c_style> tokens,
+    py::array_t<int32_t, py::array::c_style> offsets,
+    py::object keep_prob_obj,
+    py::array_t<int32_t, py::array::c_style> table,
+    int window, int n_neg, uint64_t seed, int64_t sent_id_base,
+    std::string window_mode) {
+  int64_t num_sent = offsets.shape(0) - 1;
+  const float* keep_prob = nullptr;
+  py::array_t<float, py::array::c_style> kp_arr;
+  if (!keep_prob_obj.is_none()) {
+    kp_arr = keep_prob_obj.cast<py::array_t<float, py::array::c_style>>();
+    keep_prob = kp_arr.data();
+  }
+  bool ref_window = (window_mode == "reference");
+  auto out = py::array_t<int64_t>(num_sent);
+  int64_t* o = out.mutable_data();
+  const int32_t* tok = tokens.data();
+  const int32_t* off = offsets.data();
+  std::vector<int32_t> kept;
+  for (int64_t s = 0; s < num_sent; ++s) {
+    CountPhase ph;
+    walk_sentence(tok + off[s], off[s + 1] - off[s],
+                  sentence_base(seed, (uint64_t)(sent_id_base + s)), keep_prob,
+                  table.data(), table.shape(0), window, n_neg, ref_window,
+                  kept, ph);
+    o[s] = ph.pairs;
+  }
+  return out;
+}
+
+static void dots_slice(
+    py::array_t<float, py::array::c_style> syn0,
+    py::array_t<float, py::array::c_style> syn1,
+    py::array_t<int32_t, py::array::c_style> tokens,
+    py::array_t<int32_t, py::array::c_style> offsets,
+    py::object keep_prob_obj,
+    py::array_t<int32_t, py::array::c_style> table,
+    int window, int n_neg, uint64_t seed, int64_t sent_id_base,
+    std::string window_mode,
+    py::array_t<int64_t, py::array::c_style> pair_offsets,
+    py::array_t<float, py::array::c_style> f_out) {
+  int64_t num_sent = offsets.shape(0) - 1;
+  const float* keep_prob = nullptr;
+  py::array_t<float, py::array::c_style> kp_arr;
+  if (!keep_prob_obj.is_none()) {
+    kp_arr = keep_prob_obj.cast<py::array_t<float, py::array::c_style>>();
+    keep_prob = kp_arr.data();
+  }
+  bool ref_window = (window_mode == "reference");
+  const int32_t* tok = tokens.data();
+  const int32_t* off = offsets.data();
+  const int64_t* poff = pair_offsets.data();
+  float* f = f_out.mutable_data();
+  std::vector<int32_t> kept;
+  for (int64_t s = 0; s < num_sent; ++s) {
+    DotPhase ph{syn0.data(), syn1.data(), syn0.shape(1), f + poff[s]};
+    walk_sentence(tok + off[s], off[s + 1] - off[s],
+                  sentence_base(seed, (uint64_t)(sent_id_base + s)), keep_prob,
+                  table.data(), table.shape(0), window, n_neg, ref_window,
+                  kept, ph);
+  }
+}
+
+static py::dict update_slice(
+    py::array_t<float, py::array::c_style> syn0,
+    py::array_t<float, py::array::c_style> syn1,
+    py::array_t<int32_t, py::array::c_style> tokens,
+    py::array_t<int32_t, py::array::c_style> offsets,
+    py::object keep_prob_obj,
+    py::array_t<int32_t, py::array::c_style> table,
+    float alpha, int window, int n_neg, uint64_t seed, int64_t sent_id_base,
+    std::string window_mode,
+    py::array_t<int64_t, py::array::c_style> pair_offsets,
+    py::array_t<float, py::array::c_style> f_in,
+    py::object f_loc_obj, float world_scale) {
+  int64_t num_sent = offsets.shape(0) - 1;
+  const float* keep_prob = nullptr;
+  py::array_t<float, py::array::c_style> kp_arr;
+  if (!keep_prob_obj.is_none()) {
+    kp_arr = keep_prob_obj.cast<py::array_t<float, py::array::c_style>>();
+    keep_prob = kp_arr.data();
+  }
+  const float* f_loc = nullptr;
+  py::array_t<float, py::array::c_style> floc_arr;
+  if (!f_loc_obj.is_none()) {
+    floc_arr = f_loc_obj.cast<py::array_t<float, py::array::c_style>>();
+    f_loc = floc_arr.data();
+  }
+  bool ref_window = (window_mode == "reference");
+  int64_t width = syn0.shape(1);
+  const int32_t* tok = tokens.data();
+  const int32_t* off = offsets.data();
+  const int64_t* poff = pair_offsets.data();
+  const float* f = f_in.data();
+  Stats st;
+  std::vector<int32_t> kept;
+  std::vector<float> c_row(width), grad(width);
+  for (int64_t s = 0; s < num_sent; ++s) {
+    UpdatePhase ph{syn0.mutable_data(), syn1.mutable_data(), width,
+                   f + poff[s], f_loc ? f_loc + poff[s] : nullptr, world_scale,
+                   alpha, &st, &c_row, &grad};
+    walk_sentence(tok + off[s], off[s + 1] - off[s],
+                  sentence_base(seed, (uint64_t)(sent_id_base + s)), keep_prob,
+                  table.data(), table.shape(0), window, n_neg, ref_window,
+                  kept, ph);
+  }
+  py::dict d;
+  d["pairs"] = st.pairs;
+  d["positives"] = st.positives;
+  d["sum_fplus"] = st.sum_fplus;
+  d["words_trained"] = st.words_trained;
+  return d;
+}
+
+// ---------------------------------------------------------------------------
+// Pairs trainer for the row-sharded engine: sequential SGD over an explicit
+// grouped pair plan against local f32 caches (pulled rows).  Group = one
+// center position; its targets are contiguous.
+// ---------------------------------------------------------------------------
+static py::dict train_pairs(
+    py::array_t<float, py::array::c_style> cache0,
+    py::array_t<float, py::array::c_style> cache1,
+    py::array_t<int32_t, py::array::c_style> group_center,
+    py::array_t<int64_t, py::array::c_style> group_offsets,
+    py::array_t<int32_t, py::array::c_style> pair_target,
+    py::array_t<float, py::array::c_style> pair_label,
+    float alpha) {
+  int64_t width = cache0.shape(1);
+  if (cache1.shape(1) != width) throw std::runtime_error("width mismatch");
+  int64_t G = group_center.shape(0);
+  float* c0 = cache0.mutable_data();
+  float* c1 = cache1.mutable_data();
+  const int32_t* gc = group_center.data();
+  const int64_t* go = group_offsets.data();
+  const int32_t* pt = pair_target.data();
+  const float* pl = pair_label.data();
+  Stats st;
+  std::vector<float> c_row(width), grad(width);
+  for (int64_t g = 0; g < G; ++g) {
+    float* crow = c0 + (int64_t)gc[g] * width;
+    std::memcpy(c_row.data(), crow, width * sizeof(float));
+    std::memset(grad.data(), 0, width * sizeof(float));
+    for (int64_t p = go[g]; p < go[g + 1]; ++p) {
+      float* trow = c1 + (int64_t)pt[p] * width;
+      float f = 0.0f;
+      for (int64_t d = 0; d < width; ++d) f += c_row[d] * trow[d];
+      float gg = (pl[p] - sigmoid_clipped(f)) * alpha;
+      for (int64_t d = 0; d < width; ++d) {
+        grad[d] += gg * trow[d];
+        trow[d] += gg * c_row[d];
+      }
+      st.pairs++;
+      if (pl[p] > 0.5f) { st.positives++; st.sum_fplus += f; }
+    }
+    for (int64_t d = 0; d < width; ++d) crow[d] += grad[d];
+    st.words_trained++;
+  }
+  py::dict d;
+  d["pairs"] = st.pairs;
+  d["positives"] = st.positives;
+  d["sum_fplus"] = st.sum_fplus;
+  d["words_trained"] = st.words_trained;
+  return d;
+}
+
 static py::dict train_batch(
     py::array_t<float, py::array::c_style> syn0,
     py::array_t<float, py::array::c_style> syn1,
@@ -229,4 +532,22 @@ PYBIND11_MODULE(_cpu_native, m) {
         py::arg("alpha"), py::arg("window"), py::arg("n_neg"),
         py::arg("seed"), py::arg("sent_id_base") = 0,
         py::arg("window_mode") = "canonical", py::arg("num_threads") = 1);
+  m.def("count_pairs", &count_pairs, py::arg("tokens"), py::arg("offsets"),
+        py::arg("keep_prob"), py::arg("table"), py::arg("window"),
+        py::arg("n_neg"), py::arg("seed"), py::arg("sent_id_base") = 0,
+        py::arg("window_mode") = "canonical");
+  m.def("dots_slice", &dots_slice, py::arg("syn0"), py::arg("syn1"),
+        py::arg("tokens"), py::arg("offsets"), py::arg("keep_prob"),
+        py::arg("table"), py::arg("window"), py::arg("n_neg"), py::arg("seed"),
+        py::arg("sent_id_base"), py::arg("window_mode"),
+        py::arg("pair_offsets"), py::arg("f_out"));
+  m.def("update_slice", &update_slice, py::arg("syn0"), py::arg("syn1"),
+        py::arg("tokens"), py::arg("offsets"), py::arg("keep_prob"),
+        py::arg("table"), py::arg("alpha"), py::arg("window"), py::arg("n_neg"),
+        py::arg("seed"), py::arg("sent_id_base"), py::arg("window_mode"),
+        py::arg("pair_offsets"), py::arg("f_in"),
+        py::arg("f_loc") = py::none(), py::arg("world_scale") = 0.0f);
+  m.def("train_pairs", &train_pairs, py::arg("cache0"), py::arg("cache1"),
+        py::arg("group_center"), py::arg("group_offsets"),
+        py::arg("pair_target"), py::arg("pair_label"), py::arg("alpha"));
 }

@@ -176,6 +176,145 @@ struct KernelArgs {
 constexpr int kMaxSent = 1024;
 constexpr int kWavesPerBlock = 4;
 
+// ---------------------------------------------------------------------------
+// Shared sentence walker.  Subsample+compact into the wave's LDS buffer,
+// then iterate positions/pairs in the normative order (rng.py), invoking
+// Phase:  begin_position(c) -> bool want_pairs;
+//         pair(tgt, label, pair_idx);
+//         end_position(c).
+// pair_idx counts emitted pairs within the sentence, identical on every
+// rank/implementation — the spine of the dim-sharded engine.
+// ---------------------------------------------------------------------------
+template <typename Phase>
+__device__ __forceinline__ void walk_sentence_dev(
+    const int32_t* __restrict__ tokens, int64_t off, int len, uint64_t base,
+    const uint32_t* __restrict__ keep_thr, const int32_t* __restrict__ table,
+    uint32_t table_size, int window, int n_neg, int ref_window, int lane,
+    int32_t* sent_lds, Phase& ph) {
+  // ---- subsample + wave compaction into LDS -----------------------------
+  int L = 0;
+  for (int p0 = 0; p0 < len; p0 += 64) {
+    const int p = p0 + lane;
+    bool keep = false;
+    int32_t w = 0;
+    if (p < len) {
+      w = tokens[off + p];
+      if (keep_thr) {
+        const uint32_t u = draw_u32(base, (uint64_t)p);
+        keep = u <= keep_thr[w];
+      } else {
+        keep = true;
+      }
+    }
+    const uint64_t m = __ballot(keep);
+    const int pos = __popcll(m & ((1ULL << lane) - 1ULL));
+    if (keep) sent_lds[L + pos] = w;
+    L += __popcll(m);
+  }
+  // (no __syncthreads needed: LDS buffer is private to this wave)
+
+  int64_t pair_idx = 0;
+  for (int i = 0; i < L; ++i) {
+    const int32_t c = sent_lds[i];
+    const uint32_t u = draw_u32(base, kWinBase + (uint64_t)i);
+    int lo, hi;
+    if (!ref_window) {
+      const int b = 1 + (int)(u % (uint32_t)window);
+      lo = i - b < 0 ? 0 : i - b;
+      hi = i + b >= L ? L - 1 : i + b;
+    } else {
+      const int b = (int)(u % (uint32_t)window);
+      if (b == 0) { lo = i; hi = i; }
+      else {
+        lo = i - b < 0 ? 0 : i - b;
+        hi = i + b - 1 >= L ? L - 1 : i + b - 1;
+      }
+    }
+    if (!((lo < i) || (hi > i))) continue;
+    ph.begin_position(c);
+    for (int j = lo; j <= hi; ++j) {
+      if (j == i) continue;
+      const int32_t t = sent_lds[j];
+      ph.pair(t, 1.0f, pair_idx++);
+      const uint64_t kbase =
+          kNegBase + (uint64_t)(i * (2 * window + 1) + (j - i + window)) *
+                         (uint64_t)n_neg;
+      for (int slot = 0; slot < n_neg; ++slot) {
+        const uint32_t un = draw_u32(base, kbase + (uint64_t)slot);
+        const int32_t neg = table[un % table_size];
+        if (neg == t) continue;   // discard colliding negative
+        ph.pair(neg, 0.0f, pair_idx++);
+      }
+    }
+    ph.end_position(c);
+  }
+}
+
+// ---- Phase: fully fused train (dot + sigmoid + update in one pass) -------
+template <typename T, int NC, bool ATOMIC>
+struct TrainPhase {
+  T* syn0;
+  T* syn1;
+  int64_t stride;
+  float alpha;
+  int lane;
+  // per-position state
+  T* c_ptr;
+  float c_row[NC];
+  float grad[NC];
+  // stats
+  unsigned long long w_pairs = 0, w_pos = 0, w_words = 0;
+  float w_fplus = 0.0f;
+
+  __device__ __forceinline__ void begin_position(int32_t c) {
+    c_ptr = syn0 + (int64_t)c * stride;
+    RowIO<T, NC>::load(c_ptr, c_row, lane);
+#pragma unroll
+    for (int k = 0; k < NC; ++k) grad[k] = 0.0f;
+  }
+  __device__ __forceinline__ void pair(int32_t tgt, float label, int64_t) {
+    T* t_ptr = syn1 + (int64_t)tgt * stride;
+    float t_row[NC];
+    RowIO<T, NC>::load(t_ptr, t_row, lane);
+    float f = 0.0f;
+#pragma unroll
+    for (int k = 0; k < NC; ++k) f += c_row[k] * t_row[k];
+#pragma unroll
+    for (int sh = 32; sh > 0; sh >>= 1) f += __shfl_xor(f, sh, 64);
+    const float g = (label - sigmoid_clipped(f)) * alpha;
+#pragma unroll
+    for (int k = 0; k < NC; ++k) grad[k] += g * t_row[k];
+    if (ATOMIC) {
+      float delta[NC];
+#pragma unroll
+      for (int k = 0; k < NC; ++k) delta[k] = g * c_row[k];
+      RowIO<T, NC>::atomic_add(t_ptr, delta, lane);
+    } else {
+#pragma unroll
+      for (int k = 0; k < NC; ++k) t_row[k] += g * c_row[k];
+      RowIO<T, NC>::store(t_ptr, t_row, lane);
+    }
+    ++w_pairs;
+    if (label > 0.5f) {
+      ++w_pos;
+      w_fplus += f;
+    }
+  }
+  __device__ __forceinline__ void end_position(int32_t) {
+    // center row update (hogwild: re-read current value, add, store)
+    if (ATOMIC) {
+      RowIO<T, NC>::atomic_add(c_ptr, grad, lane);
+    } else {
+      float cur[NC];
+      RowIO<T, NC>::load(c_ptr, cur, lane);
+#pragma unroll
+      for (int k = 0; k < NC; ++k) cur[k] += grad[k];
+      RowIO<T, NC>::store(c_ptr, cur, lane);
+    }
+    ++w_words;
+  }
+};
+
 template <typename T, int NC, bool ATOMIC>
 __global__ __launch_bounds__(64 * kWavesPerBlock) void sgns_train_kernel(KernelArgs a) {
   __shared__ int32_t sbuf[kWavesPerBlock][kMaxSent];
@@ -184,135 +323,261 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void sgns_train_kernel(KernelA
   const int waves_in_block = blockDim.x >> 6;   // 1 in serial mode, else 4
   const int64_t wave_gid = (int64_t)blockIdx.x * waves_in_block + wave;
   const int64_t total_waves = (int64_t)gridDim.x * waves_in_block;
-  int32_t* sent_lds = sbuf[wave];
 
-  T* syn0 = (T*)a.syn0;
-  T* syn1 = (T*)a.syn1;
-
-  unsigned long long w_pairs = 0, w_pos = 0, w_words = 0;
-  float w_fplus = 0.0f;
+  TrainPhase<T, NC, ATOMIC> ph{};
+  ph.syn0 = (T*)a.syn0;
+  ph.syn1 = (T*)a.syn1;
+  ph.stride = a.stride;
+  ph.alpha = a.alpha;
+  ph.lane = lane;
 
   for (int64_t s = wave_gid; s < a.num_sentences; s += total_waves) {
-    const int64_t off = a.offsets[s];
-    const int len = (int)(a.offsets[s + 1] - off);
     const uint64_t base = sent_base(a.seed, (uint64_t)(a.sent_id_base + s));
-
-    // ---- 1. subsample + compact into LDS --------------------------------
-    int L = 0;
-    for (int p0 = 0; p0 < len; p0 += 64) {
-      const int p = p0 + lane;
-      bool keep = false;
-      int32_t w = 0;
-      if (p < len) {
-        w = a.tokens[off + p];
-        if (a.keep_thr) {
-          const uint32_t u = draw_u32(base, (uint64_t)p);
-          keep = u <= a.keep_thr[w];
-        } else {
-          keep = true;
-        }
-      }
-      const uint64_t m = __ballot(keep);
-      const int pos = __popcll(m & ((1ULL << lane) - 1ULL));
-      if (keep) sent_lds[L + pos] = w;
-      L += __popcll(m);
-    }
-    // (no __syncthreads needed: LDS buffer is private to this wave)
-
-    // ---- 2. per-position window loop ------------------------------------
-    for (int i = 0; i < L; ++i) {
-      const int32_t c = sent_lds[i];
-      const uint32_t u = draw_u32(base, kWinBase + (uint64_t)i);
-      int lo, hi;
-      if (!a.ref_window) {
-        const int b = 1 + (int)(u % (uint32_t)a.window);
-        lo = i - b < 0 ? 0 : i - b;
-        hi = i + b >= L ? L - 1 : i + b;
-      } else {
-        const int b = (int)(u % (uint32_t)a.window);
-        if (b == 0) { lo = i; hi = i; }
-        else {
-          lo = i - b < 0 ? 0 : i - b;
-          hi = i + b - 1 >= L ? L - 1 : i + b - 1;
-        }
-      }
-      if (lo == i && hi == i) continue;
-      bool any = (lo < i) || (hi > i);
-      if (!any) continue;
-
-      T* c_ptr = syn0 + (int64_t)c * a.stride;
-      float c_row[NC], grad[NC];
-      RowIO<T, NC>::load(c_ptr, c_row, lane);
-#pragma unroll
-      for (int k = 0; k < NC; ++k) grad[k] = 0.0f;
-
-      for (int j = lo; j <= hi; ++j) {
-        if (j == i) continue;
-        const int32_t t = sent_lds[j];
-        // positive pair, then its negatives — same inner body
-        const uint64_t kbase =
-            kNegBase + (uint64_t)(i * (2 * a.window + 1) + (j - i + a.window)) *
-                           (uint64_t)a.n_neg;
-        for (int slot = -1; slot < a.n_neg; ++slot) {
-          int32_t tgt;
-          float label;
-          if (slot < 0) {
-            tgt = t;
-            label = 1.0f;
-          } else {
-            const uint32_t un = draw_u32(base, kbase + (uint64_t)slot);
-            tgt = a.table[un % a.table_size];
-            if (tgt == t) continue;   // discard colliding negative
-            label = 0.0f;
-          }
-          T* t_ptr = syn1 + (int64_t)tgt * a.stride;
-          float t_row[NC];
-          RowIO<T, NC>::load(t_ptr, t_row, lane);
-          float f = 0.0f;
-#pragma unroll
-          for (int k = 0; k < NC; ++k) f += c_row[k] * t_row[k];
-#pragma unroll
-          for (int sh = 32; sh > 0; sh >>= 1) f += __shfl_xor(f, sh, 64);
-          const float g = (label - sigmoid_clipped(f)) * a.alpha;
-#pragma unroll
-          for (int k = 0; k < NC; ++k) grad[k] += g * t_row[k];
-          if (ATOMIC) {
-            float delta[NC];
-#pragma unroll
-            for (int k = 0; k < NC; ++k) delta[k] = g * c_row[k];
-            RowIO<T, NC>::atomic_add(t_ptr, delta, lane);
-          } else {
-#pragma unroll
-            for (int k = 0; k < NC; ++k) t_row[k] += g * c_row[k];
-            RowIO<T, NC>::store(t_ptr, t_row, lane);
-          }
-          ++w_pairs;
-          if (slot < 0) {
-            ++w_pos;
-            w_fplus += f;
-          }
-        }
-      }
-      // center row update (hogwild: re-read current value, add, store)
-      if (ATOMIC) {
-        RowIO<T, NC>::atomic_add(c_ptr, grad, lane);
-      } else {
-        float cur[NC];
-        RowIO<T, NC>::load(c_ptr, cur, lane);
-#pragma unroll
-        for (int k = 0; k < NC; ++k) cur[k] += grad[k];
-        RowIO<T, NC>::store(c_ptr, cur, lane);
-      }
-      ++w_words;
-    }
+    walk_sentence_dev(a.tokens, a.offsets[s],
+                      (int)(a.offsets[s + 1] - a.offsets[s]), base, a.keep_thr,
+                      a.table, a.table_size, a.window, a.n_neg, a.ref_window,
+                      lane, sbuf[wave], ph);
   }
 
   // ---- stats (device-scope atomics, once per wave) -----------------------
   if (lane == 0 && a.d_pairs) {
-    atomicAdd(a.d_pairs, w_pairs);
-    atomicAdd(a.d_positives, w_pos);
-    atomicAdd(a.d_words, w_words);
-    atomicAdd(a.d_sum_fplus, (double)w_fplus);
+    atomicAdd(a.d_pairs, ph.w_pairs);
+    atomicAdd(a.d_positives, ph.w_pos);
+    atomicAdd(a.d_words, ph.w_words);
+    atomicAdd(a.d_sum_fplus, (double)ph.w_fplus);
+  }
+}
+
+// ---- Phase: count pairs (dim-sharded phase 0) ----------------------------
+struct CountPhase {
+  int64_t pairs = 0;
+  __device__ __forceinline__ void begin_position(int32_t) {}
+  __device__ __forceinline__ void pair(int32_t, float, int64_t) { ++pairs; }
+  __device__ __forceinline__ void end_position(int32_t) {}
+};
+
+__global__ __launch_bounds__(64 * kWavesPerBlock) void count_pairs_kernel(
+    KernelArgs a, int64_t* __restrict__ counts) {
+  __shared__ int32_t sbuf[kWavesPerBlock][kMaxSent];
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int waves_in_block = blockDim.x >> 6;
+  const int64_t wave_gid = (int64_t)blockIdx.x * waves_in_block + wave;
+  const int64_t total_waves = (int64_t)gridDim.x * waves_in_block;
+  for (int64_t s = wave_gid; s < a.num_sentences; s += total_waves) {
+    CountPhase ph{};
+    const uint64_t base = sent_base(a.seed, (uint64_t)(a.sent_id_base + s));
+    walk_sentence_dev(a.tokens, a.offsets[s],
+                      (int)(a.offsets[s + 1] - a.offsets[s]), base, a.keep_thr,
+                      a.table, a.table_size, a.window, a.n_neg, a.ref_window,
+                      lane, sbuf[wave], ph);
+    if (lane == 0) counts[s] = ph.pairs;
+  }
+}
+
+// ---- Phase: partial dots over a dim-slice (dim-sharded phase 1) ----------
+template <typename T, int NC>
+struct DotPhase {
+  const T* syn0;
+  const T* syn1;
+  int64_t stride;
+  float* f_base;   // f output, sentence-local
+  int lane;
+  float c_row[NC];
+  __device__ __forceinline__ void begin_position(int32_t c) {
+    RowIO<T, NC>::load(syn0 + (int64_t)c * stride, c_row, lane);
+  }
+  __device__ __forceinline__ void pair(int32_t tgt, float, int64_t idx) {
+    float t_row[NC];
+    RowIO<T, NC>::load(syn1 + (int64_t)tgt * stride, t_row, lane);
+    float f = 0.0f;
+#pragma unroll
+    for (int k = 0; k < NC; ++k) f += c_row[k] * t_row[k];
+#pragma unroll
+    for (int sh = 32; sh > 0; sh >>= 1) f += __shfl_xor(f, sh, 64);
+    if (lane == 0) f_base[idx] = f;
+  }
+  __device__ __forceinline__ void end_position(int32_t) {}
+};
+
+template <typename T, int NC>
+__global__ __launch_bounds__(64 * kWavesPerBlock) void dots_slice_kernel(
+    KernelArgs a, const int64_t* __restrict__ pair_offsets,
+    float* __restrict__ f_out) {
+  __shared__ int32_t sbuf[kWavesPerBlock][kMaxSent];
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int waves_in_block = blockDim.x >> 6;
+  const int64_t wave_gid = (int64_t)blockIdx.x * waves_in_block + wave;
+  const int64_t total_waves = (int64_t)gridDim.x * waves_in_block;
+  for (int64_t s = wave_gid; s < a.num_sentences; s += total_waves) {
+    DotPhase<T, NC> ph{};
+    ph.syn0 = (const T*)a.syn0;
+    ph.syn1 = (const T*)a.syn1;
+    ph.stride = a.stride;
+    ph.f_base = f_out + pair_offsets[s];
+    ph.lane = lane;
+    const uint64_t base = sent_base(a.seed, (uint64_t)(a.sent_id_base + s));
+    walk_sentence_dev(a.tokens, a.offsets[s],
+                      (int)(a.offsets[s + 1] - a.offsets[s]), base, a.keep_thr,
+                      a.table, a.table_size, a.window, a.n_neg, a.ref_window,
+                      lane, sbuf[wave], ph);
+  }
+}
+
+// ---- Phase: slice update from allreduced dots (dim-sharded phase 2) ------
+// With f_loc != null, the stale allreduced dot is freshened by extrapolating
+// this rank's local drift:
+//   f_used = f_total + world * (local_partial_now - local_partial_at_pass1)
+template <typename T, int NC>
+struct UpdateSlicePhase {
+  T* syn0;
+  T* syn1;
+  int64_t stride;
+  const float* f_base;   // full (summed) dots, sentence-local
+  const float* f_loc;    // pass-1 local partials (null = correction off)
+  float world_scale;
+  float alpha;
+  int lane;
+  T* c_ptr;
+  float c_row[NC];
+  float grad[NC];
+  unsigned long long w_pairs = 0, w_pos = 0, w_words = 0;
+  float w_fplus = 0.0f;
+  __device__ __forceinline__ void begin_position(int32_t c) {
+    c_ptr = syn0 + (int64_t)c * stride;
+    RowIO<T, NC>::load(c_ptr, c_row, lane);
+#pragma unroll
+    for (int k = 0; k < NC; ++k) grad[k] = 0.0f;
+  }
+  __device__ __forceinline__ void pair(int32_t tgt, float label, int64_t idx) {
+    T* t_ptr = syn1 + (int64_t)tgt * stride;
+    float t_row[NC];
+    RowIO<T, NC>::load(t_ptr, t_row, lane);
+    float f = f_base[idx];
+    if (f_loc) {
+      float fresh = 0.0f;
+#pragma unroll
+      for (int k = 0; k < NC; ++k) fresh += c_row[k] * t_row[k];
+#pragma unroll
+      for (int sh = 32; sh > 0; sh >>= 1) fresh += __shfl_xor(fresh, sh, 64);
+      f += world_scale * (fresh - f_loc[idx]);
+    }
+    const float g = (label - sigmoid_clipped(f)) * alpha;
+#pragma unroll
+    for (int k = 0; k < NC; ++k) {
+      grad[k] += g * t_row[k];
+      t_row[k] += g * c_row[k];
+    }
+    RowIO<T, NC>::store(t_ptr, t_row, lane);
+    ++w_pairs;
+    if (label > 0.5f) {
+      ++w_pos;
+      w_fplus += f;
+    }
+  }
+  __device__ __forceinline__ void end_position(int32_t) {
+    float cur[NC];
+    RowIO<T, NC>::load(c_ptr, cur, lane);
+#pragma unroll
+    for (int k = 0; k < NC; ++k) cur[k] += grad[k];
+    RowIO<T, NC>::store(c_ptr, cur, lane);
+    ++w_words;
+  }
+};
+
+template <typename T, int NC>
+__global__ __launch_bounds__(64 * kWavesPerBlock) void update_slice_kernel(
+    KernelArgs a, const int64_t* __restrict__ pair_offsets,
+    const float* __restrict__ f_in, const float* __restrict__ f_loc,
+    float world_scale) {
+  __shared__ int32_t sbuf[kWavesPerBlock][kMaxSent];
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int waves_in_block = blockDim.x >> 6;
+  const int64_t wave_gid = (int64_t)blockIdx.x * waves_in_block + wave;
+  const int64_t total_waves = (int64_t)gridDim.x * waves_in_block;
+  UpdateSlicePhase<T, NC> ph{};
+  ph.syn0 = (T*)a.syn0;
+  ph.syn1 = (T*)a.syn1;
+  ph.stride = a.stride;
+  ph.alpha = a.alpha;
+  ph.lane = lane;
+  ph.world_scale = world_scale;
+  for (int64_t s = wave_gid; s < a.num_sentences; s += total_waves) {
+    ph.f_base = f_in + pair_offsets[s];
+    ph.f_loc = f_loc ? f_loc + pair_offsets[s] : nullptr;
+    const uint64_t base = sent_base(a.seed, (uint64_t)(a.sent_id_base + s));
+    walk_sentence_dev(a.tokens, a.offsets[s],
+                      (int)(a.offsets[s + 1] - a.offsets[s]), base, a.keep_thr,
+                      a.table, a.table_size, a.window, a.n_neg, a.ref_window,
+                      lane, sbuf[wave], ph);
+  }
+  if (lane == 0 && a.d_pairs) {
+    atomicAdd(a.d_pairs, ph.w_pairs);
+    atomicAdd(a.d_positives, ph.w_pos);
+    atomicAdd(a.d_words, ph.w_words);
+    atomicAdd(a.d_sum_fplus, (double)ph.w_fplus);
+  }
+}
+
+// ---- pairs trainer (row-sharded engine): grouped explicit plan against
+// local f32 caches; hogwild across groups with fp32 atomics on the shared
+// cache rows.  Group = one center position; targets contiguous. ------------
+template <int NC>
+__global__ __launch_bounds__(64 * kWavesPerBlock) void train_pairs_kernel(
+    float* __restrict__ cache0, float* __restrict__ cache1, int64_t stride,
+    const int32_t* __restrict__ group_center,
+    const int64_t* __restrict__ group_offsets, int64_t num_groups,
+    const int32_t* __restrict__ pair_target,
+    const float* __restrict__ pair_label, float alpha,
+    unsigned long long* d_pairs, unsigned long long* d_positives,
+    unsigned long long* d_words, double* d_sum_fplus) {
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int waves_in_block = blockDim.x >> 6;
+  const int64_t wave_gid = (int64_t)blockIdx.x * waves_in_block + wave;
+  const int64_t total_waves = (int64_t)gridDim.x * waves_in_block;
+  unsigned long long w_pairs = 0, w_pos = 0, w_words = 0;
+  float w_fplus = 0.0f;
+  for (int64_t g = wave_gid; g < num_groups; g += total_waves) {
+    float* c_ptr = cache0 + (int64_t)group_center[g] * stride;
+    float c_row[NC], grad[NC];
+    RowIO<float, NC>::load(c_ptr, c_row, lane);
+#pragma unroll
+    for (int k = 0; k < NC; ++k) grad[k] = 0.0f;
+    for (int64_t p = group_offsets[g]; p < group_offsets[g + 1]; ++p) {
+      float* t_ptr = cache1 + (int64_t)pair_target[p] * stride;
+      float t_row[NC];
+      RowIO<float, NC>::load(t_ptr, t_row, lane);
+      float f = 0.0f;
+#pragma unroll
+      for (int k = 0; k < NC; ++k) f += c_row[k] * t_row[k];
+#pragma unroll
+      for (int sh = 32; sh > 0; sh >>= 1) f += __shfl_xor(f, sh, 64);
+      const float label = pair_label[p];
+      const float gg = (label - sigmoid_clipped(f)) * alpha;
+      float delta[NC];
+#pragma unroll
+      for (int k = 0; k < NC; ++k) {
+        grad[k] += gg * t_row[k];
+        delta[k] = gg * c_row[k];
+      }
+      RowIO<float, NC>::atomic_add(t_ptr, delta, lane);
+      ++w_pairs;
+      if (label > 0.5f) {
+        ++w_pos;
+        w_fplus += f;
+      }
+    }
+    RowIO<float, NC>::atomic_add(c_ptr, grad, lane);
+    ++w_words;
+  }
+  if (lane == 0 && d_pairs) {
+    atomicAdd(d_pairs, w_pairs);
+    atomicAdd(d_positives, w_pos);
+    atomicAdd(d_words, w_words);
+    atomicAdd(d_sum_fplus, (double)w_fplus);
   }
 }
 
@@ -460,6 +725,153 @@ static void sgns_train(uintptr_t syn0, uintptr_t syn1, int is_bf16,
   HIP_CHECK(hipGetLastError());
 }
 
+static KernelArgs make_walk_args(uintptr_t syn0, uintptr_t syn1,
+                                 int64_t stride, uintptr_t tokens,
+                                 uintptr_t offsets, int64_t num_sentences,
+                                 uintptr_t keep_thr, uintptr_t table,
+                                 int64_t table_size, double alpha, int window,
+                                 int n_neg, uint64_t seed,
+                                 int64_t sent_id_base, int ref_window,
+                                 uintptr_t stats) {
+  if (stride % 64 != 0) throw std::runtime_error("stride must be a multiple of 64");
+  if (table_size <= 0 || table_size > 0xFFFFFFFFLL)
+    throw std::runtime_error("table_size out of range");
+  KernelArgs a{};
+  a.syn0 = (void*)syn0;
+  a.syn1 = (void*)syn1;
+  a.tokens = (const int32_t*)tokens;
+  a.offsets = (const int32_t*)offsets;
+  a.num_sentences = num_sentences;
+  a.keep_thr = (const uint32_t*)keep_thr;
+  a.table = (const int32_t*)table;
+  a.table_size = (uint32_t)table_size;
+  a.alpha = (float)alpha;
+  a.window = window;
+  a.n_neg = n_neg;
+  a.seed = seed;
+  a.sent_id_base = sent_id_base;
+  a.stride = stride;
+  a.ref_window = ref_window;
+  unsigned long long* st = (unsigned long long*)stats;
+  if (st) {
+    a.d_pairs = st + 0;
+    a.d_positives = st + 1;
+    a.d_words = st + 2;
+    a.d_sum_fplus = (double*)(st + 3);
+  }
+  return a;
+}
+
+static void count_pairs(uintptr_t tokens, uintptr_t offsets,
+                        int64_t num_sentences, uintptr_t keep_thr,
+                        uintptr_t table, int64_t table_size, int window,
+                        int n_neg, uint64_t seed, int64_t sent_id_base,
+                        int ref_window, uintptr_t counts_out, int blocks,
+                        int threads, uintptr_t stream_ptr) {
+  KernelArgs a = make_walk_args(0, 0, 64, tokens, offsets, num_sentences,
+                                keep_thr, table, table_size, 0.0, window,
+                                n_neg, seed, sent_id_base, ref_window, 0);
+  hipLaunchKernelGGL(count_pairs_kernel, dim3(blocks), dim3(threads), 0,
+                     (hipStream_t)stream_ptr, a, (int64_t*)counts_out);
+  HIP_CHECK(hipGetLastError());
+}
+
+static void dots_slice(uintptr_t syn0, uintptr_t syn1, int is_bf16,
+                       int64_t stride, uintptr_t tokens, uintptr_t offsets,
+                       int64_t num_sentences, uintptr_t keep_thr,
+                       uintptr_t table, int64_t table_size, int window,
+                       int n_neg, uint64_t seed, int64_t sent_id_base,
+                       int ref_window, uintptr_t pair_offsets, uintptr_t f_out,
+                       int blocks, int threads, uintptr_t stream_ptr) {
+  KernelArgs a = make_walk_args(syn0, syn1, stride, tokens, offsets,
+                                num_sentences, keep_thr, table, table_size,
+                                0.0, window, n_neg, seed, sent_id_base,
+                                ref_window, 0);
+  const int nc = (int)(stride / 64);
+  hipStream_t stream = (hipStream_t)stream_ptr;
+#define DOTS_CASE(T, N)                                                       \
+  hipLaunchKernelGGL((dots_slice_kernel<T, N>), dim3(blocks), dim3(threads),  \
+                     0, stream, a, (const int64_t*)pair_offsets, (float*)f_out)
+  switch (nc) {
+#define CASE_NC(N)                                   \
+  case N:                                            \
+    if (is_bf16) { DOTS_CASE(uint16_t, N); }         \
+    else { DOTS_CASE(float, N); }                    \
+    break;
+    FOR_EACH_NC(CASE_NC)
+#undef CASE_NC
+    default:
+      throw std::runtime_error("unsupported NC");
+  }
+#undef DOTS_CASE
+  HIP_CHECK(hipGetLastError());
+}
+
+static void update_slice(uintptr_t syn0, uintptr_t syn1, int is_bf16,
+                         int64_t stride, uintptr_t tokens, uintptr_t offsets,
+                         int64_t num_sentences, uintptr_t keep_thr,
+                         uintptr_t table, int64_t table_size, double alpha,
+                         int window, int n_neg, uint64_t seed,
+                         int64_t sent_id_base, int ref_window,
+                         uintptr_t pair_offsets, uintptr_t f_in,
+                         uintptr_t f_loc, double world_scale,
+                         uintptr_t stats, int blocks, int threads,
+                         uintptr_t stream_ptr) {
+  KernelArgs a = make_walk_args(syn0, syn1, stride, tokens, offsets,
+                                num_sentences, keep_thr, table, table_size,
+                                alpha, window, n_neg, seed, sent_id_base,
+                                ref_window, stats);
+  const int nc = (int)(stride / 64);
+  hipStream_t stream = (hipStream_t)stream_ptr;
+#define UPD_CASE(T, N)                                                        \
+  hipLaunchKernelGGL((update_slice_kernel<T, N>), dim3(blocks), dim3(threads),\
+                     0, stream, a, (const int64_t*)pair_offsets,              \
+                     (const float*)f_in, (const float*)f_loc,                 \
+                     (float)world_scale)
+  switch (nc) {
+#define CASE_NC(N)                                  \
+  case N:                                           \
+    if (is_bf16) { UPD_CASE(uint16_t, N); }         \
+    else { UPD_CASE(float, N); }                    \
+    break;
+    FOR_EACH_NC(CASE_NC)
+#undef CASE_NC
+    default:
+      throw std::runtime_error("unsupported NC");
+  }
+#undef UPD_CASE
+  HIP_CHECK(hipGetLastError());
+}
+
+static void train_pairs(uintptr_t cache0, uintptr_t cache1, int64_t stride,
+                        uintptr_t group_center, uintptr_t group_offsets,
+                        int64_t num_groups, uintptr_t pair_target,
+                        uintptr_t pair_label, double alpha, uintptr_t stats,
+                        int blocks, int threads, uintptr_t stream_ptr) {
+  if (stride % 64 != 0) throw std::runtime_error("stride must be a multiple of 64");
+  const int nc = (int)(stride / 64);
+  hipStream_t stream = (hipStream_t)stream_ptr;
+  unsigned long long* st = (unsigned long long*)stats;
+  switch (nc) {
+#define CASE_NC(N)                                                            \
+  case N:                                                                     \
+    hipLaunchKernelGGL((train_pairs_kernel<N>), dim3(blocks), dim3(threads),  \
+                       0, stream, (float*)cache0, (float*)cache1, stride,     \
+                       (const int32_t*)group_center,                          \
+                       (const int64_t*)group_offsets, num_groups,             \
+                       (const int32_t*)pair_target, (const float*)pair_label, \
+                       (float)alpha, st ? st + 0 : nullptr,                   \
+                       st ? st + 1 : nullptr, st ? st + 2 : nullptr,          \
+                       st ? (double*)(st + 3) : nullptr);                     \
+    break;
+    FOR_EACH_NC(CASE_NC)
+#undef CASE_NC
+    default:
+      throw std::runtime_error("unsupported NC");
+  }
+  HIP_CHECK(hipGetLastError());
+}
+
 template <typename T>
 static void launch_pull_average_t(uintptr_t syn0, uintptr_t tokens,
                                   uintptr_t offsets, int64_t num_sentences,
@@ -538,6 +950,10 @@ PYBIND11_MODULE(_hip_native, m) {
         py::arg("sent_id_base"), py::arg("ref_window"), py::arg("atomic"),
         py::arg("stats"), py::arg("blocks"), py::arg("threads"),
         py::arg("stream"));
+  m.def("count_pairs", &count_pairs);
+  m.def("dots_slice", &dots_slice);
+  m.def("update_slice", &update_slice);
+  m.def("train_pairs", &train_pairs);
   m.def("pull_average", &pull_average);
   m.def("norms", &norms);
   m.def("round_stride", &round_stride);

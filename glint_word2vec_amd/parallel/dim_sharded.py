@@ -1,0 +1,248 @@
+"""Dimension-sharded multi-GPU SGNS engine (CIKM'16 scheme on RCCL/xGMI —
+DESIGN.md).
+
+Each rank owns a contiguous dim-slice of BOTH tables.  Every rank walks the
+same token batch with the same counter-based RNG, so pair enumeration is
+identical everywhere.  Per chunk:
+
+    count pairs -> partial dots over my slice -> allreduce(f) ->
+    sigmoid/gradient + slice updates (fully local)
+
+Network traffic is dimension-independent: ~4 bytes per pair (the f scalar),
+never an index, never a row — the reference's network-efficiency claim
+(README.md:7-9) realized with one RCCL allreduce.
+
+Works on CUDA (HIP kernels) and CPU (C++ twins, used by the gloo
+multi-process tests).
+"""
+from __future__ import annotations
+
+import logging
+from dataclasses import dataclass
+from typing import Optional
+
+import numpy as np
+import torch
+
+from ..vocab import build_unigram_table, keep_probabilities
+from . import comm
+from ..ops.gpu import GpuStats
+
+log = logging.getLogger("glint_word2vec_amd")
+
+
+def slice_bounds(dim: int, rank: int, world: int) -> tuple[int, int]:
+    lo = rank * dim // world
+    hi = (rank + 1) * dim // world
+    return lo, hi
+
+
+class DimShardedSgns:
+    def __init__(self, vocab_size: int, dim: int, dtype: str = "float32",
+                 device: str = "cuda", seed: int = 1,
+                 counts: Optional[np.ndarray] = None,
+                 table_size: int = 1_000_000, subsample: float = 0.0,
+                 window_mode: str = "canonical", chunk_words: int = 1 << 16,
+                 f_correction: bool = True, init_full_limit: int = 1 << 28):
+        self.rank, self.world = (comm.init_from_env() if torch.distributed.is_available()
+                                 else (0, 1))
+        self.vocab_size = vocab_size
+        self.dim = dim
+        self.device = torch.device(device)
+        self.is_cuda = self.device.type == "cuda"
+        self.is_bf16 = dtype == "bfloat16"
+        self.window_mode = window_mode
+        self.chunk_words = chunk_words
+        self.f_correction = f_correction
+        self.lo, self.hi = slice_bounds(dim, self.rank, self.world)
+        self.width = self.hi - self.lo
+
+        if self.is_cuda:
+            from .. import _hip_native
+            self.native = _hip_native
+            self.stride = self.native.round_stride(max(self.width, 1))
+        else:
+            from .. import _cpu_native
+            self.native = _cpu_native
+            self.stride = self.width
+        tdtype = torch.bfloat16 if self.is_bf16 else torch.float32
+        if self.is_bf16 and not self.is_cuda:
+            raise ValueError("bf16 dim-sharded path is GPU-only")
+        self.syn0 = torch.zeros((vocab_size, self.stride), dtype=tdtype,
+                                device=self.device)
+        self.syn1 = torch.zeros((vocab_size, self.stride), dtype=tdtype,
+                                device=self.device)
+        self._init_slices(seed, init_full_limit)
+
+        counts = (np.ones(vocab_size, dtype=np.int64) if counts is None
+                  else counts)
+        table = build_unigram_table(counts, table_size)
+        self.table = torch.from_numpy(table).to(self.device)
+        self.keep_prob: Optional[np.ndarray] = None
+        self.keep_thr: Optional[torch.Tensor] = None
+        if subsample > 0:
+            kp = keep_probabilities(counts, int(counts.sum()), subsample)
+            self.keep_prob = kp
+            thr = np.minimum(kp.astype(np.float64) * 4294967296.0,
+                             4294967295.0).astype(np.uint32)
+            self.keep_thr = torch.from_numpy(thr.view(np.int32)).to(self.device)
+        self._stats = torch.zeros(4, dtype=torch.int64, device=self.device)
+        self._cpu_stats = dict(pairs=0, positives=0, words_trained=0,
+                               sum_fplus=0.0)
+        self.serial = False   # tests: single-wave launches (oracle order)
+
+    def _init_slices(self, seed: int, full_limit: int) -> None:
+        """Same init as the single-GPU path: full-matrix U(-.5/dim,.5/dim)
+        sliced by column, so results are world-size invariant (tested).  For
+        huge vocab*dim the init is generated per row-block."""
+        tdtype = self.syn0.dtype
+        block = max(1, full_limit // max(self.dim, 1))
+        rng = np.random.default_rng(seed)
+        for r0 in range(0, self.vocab_size, block):
+            r1 = min(self.vocab_size, r0 + block)
+            full = (rng.random((r1 - r0, self.dim), dtype=np.float32) - 0.5) / self.dim
+            sl = torch.from_numpy(np.ascontiguousarray(full[:, self.lo:self.hi]))
+            self.syn0[r0:r1, :self.width] = sl.to(tdtype).to(self.device)
+
+    # ------------------------------------------------------------------
+    def train_step(self, tokens: torch.Tensor, offsets: torch.Tensor,
+                   alpha: float, window: int, n_neg: int, seed: int,
+                   sent_id_base: int = 0,
+                   offsets_host: Optional[np.ndarray] = None) -> None:
+        """One step over a batch.  ``tokens``/``offsets`` must be identical
+        on every rank (dim-sharding splits compute by dimension, not data)."""
+        if offsets_host is None:
+            offsets_host = offsets.cpu().numpy()
+        num_sent = len(offsets_host) - 1
+        if num_sent <= 0:
+            return
+        # chunk boundaries: greedily pack sentences up to chunk_words tokens
+        sent_lens = np.diff(offsets_host)
+        chunks = []
+        s0 = 0
+        acc = 0
+        for s in range(num_sent):
+            acc += sent_lens[s]
+            if acc >= self.chunk_words or s == num_sent - 1:
+                chunks.append((s0, s + 1))
+                s0, acc = s + 1, 0
+        for (a, b) in chunks:
+            self._train_chunk(tokens, offsets, a, b, alpha, window, n_neg,
+                              seed, sent_id_base)
+
+    def _train_chunk(self, tokens, offsets, s0, s1, alpha, window, n_neg,
+                     seed, sent_id_base):
+        n = s1 - s0
+        off_view = offsets[s0:s1 + 1]
+        base = sent_id_base + s0
+        if self.is_cuda:
+            stream = torch.cuda.current_stream(self.device)
+            nb = 1 if self.serial else max(1, min((n + 3) // 4, 8192))
+            nt = 64 if self.serial else 256
+            counts = torch.empty(n, dtype=torch.int64, device=self.device)
+            self.native.count_pairs(
+                tokens.data_ptr(), off_view.data_ptr(), n,
+                0 if self.keep_thr is None else self.keep_thr.data_ptr(),
+                self.table.data_ptr(), int(self.table.numel()), window, n_neg,
+                seed & 0xFFFFFFFFFFFFFFFF, base,
+                int(self.window_mode == "reference"), counts.data_ptr(),
+                nb, nt, stream.cuda_stream)
+            poff = torch.zeros(n + 1, dtype=torch.int64, device=self.device)
+            torch.cumsum(counts, 0, out=poff[1:])
+            total = int(poff[-1].item())
+            if total == 0:
+                return
+            f = torch.zeros(total, dtype=torch.float32, device=self.device)
+            self.native.dots_slice(
+                self.syn0.data_ptr(), self.syn1.data_ptr(), int(self.is_bf16),
+                self.stride, tokens.data_ptr(), off_view.data_ptr(), n,
+                0 if self.keep_thr is None else self.keep_thr.data_ptr(),
+                self.table.data_ptr(), int(self.table.numel()), window, n_neg,
+                seed & 0xFFFFFFFFFFFFFFFF, base,
+                int(self.window_mode == "reference"), poff.data_ptr(),
+                f.data_ptr(), nb, nt, stream.cuda_stream)
+            f_loc = f.clone() if self.f_correction else None
+            comm.all_reduce_sum(f)
+            self.native.update_slice(
+                self.syn0.data_ptr(), self.syn1.data_ptr(), int(self.is_bf16),
+                self.stride, tokens.data_ptr(), off_view.data_ptr(), n,
+                0 if self.keep_thr is None else self.keep_thr.data_ptr(),
+                self.table.data_ptr(), int(self.table.numel()), float(alpha),
+                window, n_neg, seed & 0xFFFFFFFFFFFFFFFF, base,
+                int(self.window_mode == "reference"), poff.data_ptr(),
+                f.data_ptr(),
+                0 if f_loc is None else f_loc.data_ptr(),
+                float(self.world), self._stats.data_ptr(),
+                nb, nt, stream.cuda_stream)
+        else:
+            tok_np = tokens.numpy()
+            off_np = off_view.numpy()
+            tab_np = self.table.numpy()
+            wm = self.window_mode
+            cnts = self.native.count_pairs(tok_np, off_np, self.keep_prob,
+                                           tab_np, window, n_neg,
+                                           seed & 0xFFFFFFFFFFFFFFFF, base, wm)
+            poff = np.zeros(n + 1, dtype=np.int64)
+            np.cumsum(cnts, out=poff[1:])
+            total = int(poff[-1])
+            if total == 0:
+                return
+            f = torch.zeros(total, dtype=torch.float32)
+            s0_np = self.syn0.numpy()
+            s1_np = self.syn1.numpy()
+            self.native.dots_slice(s0_np, s1_np, tok_np, off_np,
+                                   self.keep_prob, tab_np, window, n_neg,
+                                   seed & 0xFFFFFFFFFFFFFFFF, base, wm, poff,
+                                   f.numpy())
+            f_loc = f.numpy().copy() if self.f_correction else None
+            comm.all_reduce_sum(f)
+            st = self.native.update_slice(s0_np, s1_np, tok_np, off_np,
+                                          self.keep_prob, tab_np, float(alpha),
+                                          window, n_neg,
+                                          seed & 0xFFFFFFFFFFFFFFFF, base, wm,
+                                          poff, f.numpy(), f_loc,
+                                          float(self.world))
+            for k in ("pairs", "positives", "words_trained"):
+                self._cpu_stats[k] += st[k]
+            self._cpu_stats["sum_fplus"] += st["sum_fplus"]
+
+    # ------------------------------------------------------------------
+    def read_stats(self, reset: bool = True) -> GpuStats:
+        if self.is_cuda:
+            h = self._stats.cpu()
+            out = GpuStats(int(h[0]), int(h[1]), int(h[2]),
+                           float(h[3:4].view(torch.float64)[0]))
+            if reset:
+                self._stats.zero_()
+            return out
+        st = self._cpu_stats
+        out = GpuStats(st["pairs"], st["positives"], st["words_trained"],
+                       st["sum_fplus"])
+        if reset:
+            self._cpu_stats = dict(pairs=0, positives=0, words_trained=0,
+                                   sum_fplus=0.0)
+        return out
+
+    def to_host(self) -> tuple[np.ndarray, np.ndarray]:
+        """Assemble the full [vocab, dim] matrices on every rank (allgather
+        of slices; small-model path for save/model ops)."""
+        out = []
+        for t in (self.syn0, self.syn1):
+            my = t[:, :self.width].float()
+            if self.world == 1:
+                out.append(my.cpu().numpy())
+                continue
+            maxw = max(slice_bounds(self.dim, r, self.world)[1] -
+                       slice_bounds(self.dim, r, self.world)[0]
+                       for r in range(self.world))
+            padded = torch.zeros((self.vocab_size, maxw), dtype=torch.float32,
+                                 device=self.device)
+            padded[:, :self.width] = my
+            gathered = [torch.empty_like(padded) for _ in range(self.world)]
+            torch.distributed.all_gather(gathered, padded)
+            full = np.empty((self.vocab_size, self.dim), dtype=np.float32)
+            for r in range(self.world):
+                lo, hi = slice_bounds(self.dim, r, self.world)
+                full[:, lo:hi] = gathered[r][:, :hi - lo].cpu().numpy()
+            out.append(full)
+        return out[0], out[1]

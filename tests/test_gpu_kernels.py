@@ -160,6 +160,53 @@ def test_multiply_matches_numpy():
     np.testing.assert_allclose(out, ref, rtol=1e-4, atol=1e-5)
 
 
+def test_dim_sharded_gpu_world1_matches_sequential():
+    """GPU dim-sharded phases (count/dots/update+f-correction), serial
+    launch, world=1: must reproduce the sequential C++ trainer."""
+    native = pytest.importorskip("glint_word2vec_amd._cpu_native")
+    from glint_word2vec_amd.parallel.dim_sharded import DimShardedSgns
+    from glint_word2vec_amd.data import synthetic_corpus
+    batch = synthetic_corpus(vocab_size=60, num_tokens=400, sentence_len=40,
+                             seed=5, zipf_a=1.01)
+    counts = np.bincount(batch.tokens, minlength=60).astype(np.int64) + 1
+    table = build_unigram_table(counts, 1009)
+    syn0, syn1 = sgns.init_tables(60, 24, 3)
+    st_ref = native.train_batch(syn0, syn1, batch.tokens, batch.offsets, None,
+                                table, 0.03, 3, 4, 42, 0, "canonical", 1)
+    eng = DimShardedSgns(60, 24, device="cuda", seed=3, counts=counts,
+                         table_size=1009, chunk_words=10 ** 9,
+                         f_correction=True)
+    eng.serial = True
+    tok = torch.from_numpy(batch.tokens).cuda()
+    off = torch.from_numpy(batch.offsets).cuda()
+    eng.train_step(tok, off, 0.03, 3, 4, seed=42, offsets_host=batch.offsets)
+    torch.cuda.synchronize()
+    st = eng.read_stats()
+    assert st.pairs == st_ref["pairs"]
+    assert st.positives == st_ref["positives"]
+    s0, s1 = eng.to_host()
+    np.testing.assert_allclose(s0, syn0, rtol=2e-4, atol=2e-6)
+    np.testing.assert_allclose(s1, syn1, rtol=2e-4, atol=2e-6)
+
+
+def test_dim_sharded_gpu_parallel_stable():
+    from glint_word2vec_amd.parallel.dim_sharded import DimShardedSgns
+    from glint_word2vec_amd.data import synthetic_corpus
+    batch = synthetic_corpus(vocab_size=5000, num_tokens=50_000,
+                             sentence_len=100, seed=9)
+    counts = np.bincount(batch.tokens, minlength=5000).astype(np.int64) + 1
+    eng = DimShardedSgns(5000, 64, device="cuda", seed=3, counts=counts,
+                         table_size=100_003, chunk_words=16384)
+    tok = torch.from_numpy(batch.tokens).cuda()
+    off = torch.from_numpy(batch.offsets).cuda()
+    eng.train_step(tok, off, 0.025, 5, 5, seed=1, offsets_host=batch.offsets)
+    torch.cuda.synchronize()
+    st = eng.read_stats()
+    assert st.pairs > 100_000
+    s0, s1 = eng.to_host()
+    assert np.isfinite(s0).all() and np.isfinite(s1).all()
+
+
 def test_estimator_end_to_end_gpu():
     rng = np.random.default_rng(5)
     sents = []
