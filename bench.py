@@ -38,7 +38,8 @@ def parse_args():
     p.add_argument("--subsample", type=float, default=1e-4,
                    help="subsample ratio (drawn in-kernel; 0 disables)")
     p.add_argument("--blocks", type=int, default=0, help="grid blocks override")
-    p.add_argument("--atomic", action="store_true")
+    p.add_argument("--hogwild", action="store_true",
+                   help="plain read-modify-write row updates (no atomics)")
     p.add_argument("--profile-steps", type=int, default=0,
                    help="run only this many steps, no warmup JSON (rocprof)")
     return p.parse_args()
@@ -104,7 +105,7 @@ def main():
         else:
             trainer.train_batch(tok, off, alpha, args.window, args.neg, 99,
                                 sent_id_base=i * nsent,
-                                atomic=args.atomic,
+                                atomic=not args.hogwild,
                                 blocks=args.blocks or None)
 
     def barrier_sync():

@@ -71,6 +71,8 @@ constexpr uint64_t kNegBase = 1ULL << 21;
 //   paired slots 2m / 2m+1 : elements 128*m + 2*lane + {0,1}
 //   odd-NC tail slot NC-1  : element  64*(NC-1) + lane
 // ---------------------------------------------------------------------------
+typedef __attribute__((ext_vector_type(2))) __bf16 v2bf16;
+
 __device__ __forceinline__ float bf16_to_f32(uint16_t h) {
   uint32_t u = ((uint32_t)h) << 16;
   return __uint_as_float(u);
@@ -133,8 +135,28 @@ struct RowIO<uint16_t, NC> {
     }
     if (NC & 1) row[64 * (NC - 1) + lane] = f32_to_bf16_rne(v[NC - 1]);
   }
-  static __device__ __forceinline__ void atomic_add(uint16_t*, const float[NC], int) {
-    // bf16 atomic variant not supported (wrapper rejects it)
+  // gfx950 packed-bf16 atomic add (global_atomic_pk_add_bf16): adds the f32
+  // delta rounded to bf16 — no lost updates on contended (hot Zipf) rows.
+  static __device__ __forceinline__ void atomic_add(uint16_t* row, const float v[NC], int lane) {
+#pragma unroll
+    for (int m = 0; m < NC / 2; ++m) {
+      v2bf16 d;
+      d[0] = (__bf16)v[2 * m];
+      d[1] = (__bf16)v[2 * m + 1];
+      __builtin_amdgcn_global_atomic_fadd_v2bf16(
+          (v2bf16*)(row + 128 * m + 2 * lane), d);
+    }
+    if (NC & 1) {
+      // tail: one element per lane; hit the containing aligned pair with the
+      // other half = +0.0 (x + 0.0 == x in bf16)
+      uint16_t* p = row + 64 * (NC - 1) + lane;
+      const bool odd = ((uintptr_t)p >> 1) & 1;
+      v2bf16 d;
+      d[0] = odd ? (__bf16)0.0f : (__bf16)v[NC - 1];
+      d[1] = odd ? (__bf16)v[NC - 1] : (__bf16)0.0f;
+      __builtin_amdgcn_global_atomic_fadd_v2bf16(
+          (v2bf16*)((uintptr_t)p & ~(uintptr_t)3), d);
+    }
   }
 };
 
@@ -429,7 +451,7 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void dots_slice_kernel(
 // With f_loc != null, the stale allreduced dot is freshened by extrapolating
 // this rank's local drift:
 //   f_used = f_total + world * (local_partial_now - local_partial_at_pass1)
-template <typename T, int NC>
+template <typename T, int NC, bool ATOMIC>
 struct UpdateSlicePhase {
   T* syn0;
   T* syn1;
@@ -464,12 +486,22 @@ struct UpdateSlicePhase {
       f += world_scale * (fresh - f_loc[idx]);
     }
     const float g = (label - sigmoid_clipped(f)) * alpha;
+    if (ATOMIC) {
+      float delta[NC];
 #pragma unroll
-    for (int k = 0; k < NC; ++k) {
-      grad[k] += g * t_row[k];
-      t_row[k] += g * c_row[k];
+      for (int k = 0; k < NC; ++k) {
+        grad[k] += g * t_row[k];
+        delta[k] = g * c_row[k];
+      }
+      RowIO<T, NC>::atomic_add(t_ptr, delta, lane);
+    } else {
+#pragma unroll
+      for (int k = 0; k < NC; ++k) {
+        grad[k] += g * t_row[k];
+        t_row[k] += g * c_row[k];
+      }
+      RowIO<T, NC>::store(t_ptr, t_row, lane);
     }
-    RowIO<T, NC>::store(t_ptr, t_row, lane);
     ++w_pairs;
     if (label > 0.5f) {
       ++w_pos;
@@ -477,16 +509,20 @@ struct UpdateSlicePhase {
     }
   }
   __device__ __forceinline__ void end_position(int32_t) {
-    float cur[NC];
-    RowIO<T, NC>::load(c_ptr, cur, lane);
+    if (ATOMIC) {
+      RowIO<T, NC>::atomic_add(c_ptr, grad, lane);
+    } else {
+      float cur[NC];
+      RowIO<T, NC>::load(c_ptr, cur, lane);
 #pragma unroll
-    for (int k = 0; k < NC; ++k) cur[k] += grad[k];
-    RowIO<T, NC>::store(c_ptr, cur, lane);
+      for (int k = 0; k < NC; ++k) cur[k] += grad[k];
+      RowIO<T, NC>::store(c_ptr, cur, lane);
+    }
     ++w_words;
   }
 };
 
-template <typename T, int NC>
+template <typename T, int NC, bool ATOMIC>
 __global__ __launch_bounds__(64 * kWavesPerBlock) void update_slice_kernel(
     KernelArgs a, const int64_t* __restrict__ pair_offsets,
     const float* __restrict__ f_in, const float* __restrict__ f_loc,
@@ -497,7 +533,7 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void update_slice_kernel(
   const int waves_in_block = blockDim.x >> 6;
   const int64_t wave_gid = (int64_t)blockIdx.x * waves_in_block + wave;
   const int64_t total_waves = (int64_t)gridDim.x * waves_in_block;
-  UpdateSlicePhase<T, NC> ph{};
+  UpdateSlicePhase<T, NC, ATOMIC> ph{};
   ph.syn0 = (T*)a.syn0;
   ph.syn1 = (T*)a.syn1;
   ph.stride = a.stride;
@@ -693,7 +729,6 @@ static void sgns_train(uintptr_t syn0, uintptr_t syn1, int is_bf16,
   if (supported_nc(nc) != nc) throw std::runtime_error("stride/64 not a supported NC");
   if (table_size <= 0 || table_size > 0xFFFFFFFFLL)
     throw std::runtime_error("table_size out of range");
-  if (atomic && is_bf16) throw std::runtime_error("atomic updates unsupported for bf16");
   KernelArgs a{};
   a.syn0 = (void*)syn0;
   a.syn1 = (void*)syn1;
@@ -814,7 +849,7 @@ static void update_slice(uintptr_t syn0, uintptr_t syn1, int is_bf16,
                          int window, int n_neg, uint64_t seed,
                          int64_t sent_id_base, int ref_window,
                          uintptr_t pair_offsets, uintptr_t f_in,
-                         uintptr_t f_loc, double world_scale,
+                         uintptr_t f_loc, double world_scale, int atomic,
                          uintptr_t stats, int blocks, int threads,
                          uintptr_t stream_ptr) {
   KernelArgs a = make_walk_args(syn0, syn1, stride, tokens, offsets,
@@ -824,10 +859,18 @@ static void update_slice(uintptr_t syn0, uintptr_t syn1, int is_bf16,
   const int nc = (int)(stride / 64);
   hipStream_t stream = (hipStream_t)stream_ptr;
 #define UPD_CASE(T, N)                                                        \
-  hipLaunchKernelGGL((update_slice_kernel<T, N>), dim3(blocks), dim3(threads),\
-                     0, stream, a, (const int64_t*)pair_offsets,              \
-                     (const float*)f_in, (const float*)f_loc,                 \
-                     (float)world_scale)
+  do {                                                                        \
+    if (atomic)                                                               \
+      hipLaunchKernelGGL((update_slice_kernel<T, N, true>), dim3(blocks),     \
+                         dim3(threads), 0, stream, a,                         \
+                         (const int64_t*)pair_offsets, (const float*)f_in,    \
+                         (const float*)f_loc, (float)world_scale);            \
+    else                                                                      \
+      hipLaunchKernelGGL((update_slice_kernel<T, N, false>), dim3(blocks),    \
+                         dim3(threads), 0, stream, a,                         \
+                         (const int64_t*)pair_offsets, (const float*)f_in,    \
+                         (const float*)f_loc, (float)world_scale);            \
+  } while (0)
   switch (nc) {
 #define CASE_NC(N)                                  \
   case N:                                           \

@@ -51,7 +51,11 @@ class Word2VecConfig:
     dtype: str = "float32"           # "float32" | "bfloat16" table storage
     device: str = "auto"             # "auto" | "cpu" | "cuda"
     words_per_step: int = 1 << 20    # tokens fed to the GPU per training step
-    atomic_updates: bool = False     # atomics vs hogwild plain RMW in kernel
+    # Row updates: atomics (no lost updates on hot rows — fp32 atomicAdd /
+    # gfx950 packed-bf16 atomic) vs plain hogwild read-modify-write (the
+    # reference's races-embraced semantics, slightly faster, loses updates
+    # under heavy contention).
+    atomic_updates: bool = True
     # --- semantics switches ---------------------
     # The reference's subsampling is a de-facto no-op (integer-division bug,
     # mllib:375-377).  We implement the intended math; set

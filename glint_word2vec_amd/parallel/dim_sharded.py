@@ -43,7 +43,8 @@ class DimShardedSgns:
                  counts: Optional[np.ndarray] = None,
                  table_size: int = 1_000_000, subsample: float = 0.0,
                  window_mode: str = "canonical", chunk_words: int = 1 << 16,
-                 f_correction: bool = True, init_full_limit: int = 1 << 28):
+                 f_correction: bool = True, atomic: bool = True,
+                 init_full_limit: int = 1 << 28):
         self.rank, self.world = (comm.init_from_env() if torch.distributed.is_available()
                                  else (0, 1))
         self.vocab_size = vocab_size
@@ -54,6 +55,7 @@ class DimShardedSgns:
         self.window_mode = window_mode
         self.chunk_words = chunk_words
         self.f_correction = f_correction
+        self.atomic = atomic
         self.lo, self.hi = slice_bounds(dim, self.rank, self.world)
         self.width = self.hi - self.lo
 
@@ -172,7 +174,7 @@ class DimShardedSgns:
                 int(self.window_mode == "reference"), poff.data_ptr(),
                 f.data_ptr(),
                 0 if f_loc is None else f_loc.data_ptr(),
-                float(self.world), self._stats.data_ptr(),
+                float(self.world), int(self.atomic), self._stats.data_ptr(),
                 nb, nt, stream.cuda_stream)
         else:
             tok_np = tokens.numpy()

@@ -58,13 +58,13 @@ def test_serial_kernel_matches_oracle(window_mode):
 def test_serial_kernel_with_subsampling():
     tokens, offsets, counts, table, syn0, syn1 = _problem()
     total = int(counts.sum())
-    kp = keep_probabilities(counts, total, 0.3)
+    kp = keep_probabilities(counts, total, 0.005)
     assert (kp < 1).any()
     a0, a1 = syn0.copy(), syn1.copy()
     st_py = cpu_ref.train_batch_oracle(a0, a1, tokens, offsets, kp, table,
                                        0.03, 3, 4, seed=13)
     gs = _gpu_setup(syn0, syn1, table)
-    gs.set_subsample(counts, total, 0.3)
+    gs.set_subsample(counts, total, 0.005)
     gs.train_batch(_to_dev(tokens), _to_dev(offsets), 0.03, 3, 4, 13,
                    serial=True)
     torch.cuda.synchronize()
@@ -80,21 +80,22 @@ def test_parallel_kernel_exact_counts_stable_values():
     must match the oracle exactly; values race but must stay finite/close
     in aggregate."""
     tokens, offsets, counts, table, syn0, syn1 = _problem(
-        vocab=200, dim=32, n_tokens=4000, sentences=50)
+        vocab=20000, dim=32, n_tokens=40000, sentences=400)
     a0, a1 = syn0.copy(), syn1.copy()
     st_py = cpu_ref.train_batch_oracle(a0, a1, tokens, offsets, None, table,
                                        0.03, 4, 5, seed=3)
     gs = _gpu_setup(syn0, syn1, table)
-    gs.train_batch(_to_dev(tokens), _to_dev(offsets), 0.03, 4, 5, 3)
+    gs.train_batch(_to_dev(tokens), _to_dev(offsets), 0.03, 4, 5, 3,
+                   atomic=True)
     torch.cuda.synchronize()
     st = gs.read_stats()
     assert st.pairs == st_py.pairs
     assert st.positives == st_py.positives
     g0, g1 = gs.to_host()
     assert np.isfinite(g0).all() and np.isfinite(g1).all()
-    # aggregate movement should be similar magnitude to the oracle's
+    # atomics: no lost updates -> aggregate movement matches the oracle's
     assert np.linalg.norm(g0 - syn0) == pytest.approx(
-        np.linalg.norm(a0 - syn0), rel=0.5)
+        np.linalg.norm(a0 - syn0), rel=0.3)
 
 
 def test_atomic_variant_fp32():
