@@ -40,6 +40,8 @@ def parse_args():
     p.add_argument("--blocks", type=int, default=0, help="grid blocks override")
     p.add_argument("--hogwild", action="store_true",
                    help="plain read-modify-write row updates (no atomics)")
+    p.add_argument("--atomic-below", type=int, default=None,
+                   help="atomics only for rows < K (hot rows); default all")
     p.add_argument("--profile-steps", type=int, default=0,
                    help="run only this many steps, no warmup JSON (rocprof)")
     return p.parse_args()
@@ -106,6 +108,7 @@ def main():
             trainer.train_batch(tok, off, alpha, args.window, args.neg, 99,
                                 sent_id_base=i * nsent,
                                 atomic=not args.hogwild,
+                                atomic_below=args.atomic_below,
                                 blocks=args.blocks or None)
 
     def barrier_sync():

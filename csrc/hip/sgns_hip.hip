@@ -31,6 +31,7 @@
 #include <hip/hip_runtime.h>
 #include <pybind11/pybind11.h>
 
+#include <algorithm>
 #include <cstdint>
 #include <cstring>
 #include <stdexcept>
@@ -188,6 +189,7 @@ struct KernelArgs {
   int64_t sent_id_base;
   int64_t stride;             // row stride in elements (= 64*NC)
   int ref_window;             // 0 canonical, 1 reference (B2) semantics
+  int32_t atomic_below;       // rows < this use atomics (hot rows); rest plain
   // stats
   unsigned long long* d_pairs;
   unsigned long long* d_positives;
@@ -212,7 +214,7 @@ __device__ __forceinline__ void walk_sentence_dev(
     const int32_t* __restrict__ tokens, int64_t off, int len, uint64_t base,
     const uint32_t* __restrict__ keep_thr, const int32_t* __restrict__ table,
     uint32_t table_size, int window, int n_neg, int ref_window, int lane,
-    int32_t* sent_lds, Phase& ph) {
+    int32_t* sent_lds, uint32_t* tgt_lds, Phase& ph) {
   // ---- subsample + wave compaction into LDS -----------------------------
   int L = 0;
   for (int p0 = 0; p0 < len; p0 += 64) {
@@ -233,9 +235,10 @@ __device__ __forceinline__ void walk_sentence_dev(
     if (keep) sent_lds[L + pos] = w;
     L += __popcll(m);
   }
-  // (no __syncthreads needed: LDS buffer is private to this wave)
+  // (no __syncthreads needed: LDS buffers are private to this wave)
 
   int64_t pair_idx = 0;
+  const int per_ctx = 1 + n_neg;
   for (int i = 0; i < L; ++i) {
     const int32_t c = sent_lds[i];
     const uint32_t u = draw_u32(base, kWinBase + (uint64_t)i);
@@ -253,19 +256,42 @@ __device__ __forceinline__ void walk_sentence_dev(
       }
     }
     if (!((lo < i) || (hi > i))) continue;
+    const int n_ctx = (hi - lo + 1) - ((lo <= i && i <= hi) ? 1 : 0);
+    const int total_slots = n_ctx * per_ctx;
     ph.begin_position(c);
-    for (int j = lo; j <= hi; ++j) {
-      if (j == i) continue;
-      const int32_t t = sent_lds[j];
-      ph.pair(t, 1.0f, pair_idx++);
-      const uint64_t kbase =
-          kNegBase + (uint64_t)(i * (2 * window + 1) + (j - i + window)) *
-                         (uint64_t)n_neg;
-      for (int slot = 0; slot < n_neg; ++slot) {
-        const uint32_t un = draw_u32(base, kbase + (uint64_t)slot);
-        const int32_t neg = table[un % table_size];
-        if (neg == t) continue;   // discard colliding negative
-        ph.pair(neg, 0.0f, pair_idx++);
+    // Materialize the target list 64 slots at a time (all lanes draw in
+    // parallel — negative RNG + table gathers vectorize across the wave),
+    // then the phase processes the compacted chunk serially with prefetch.
+    for (int chunk = 0; chunk < total_slots; chunk += 64) {
+      const int slot = chunk + lane;
+      bool valid = false;
+      uint32_t enc = 0;
+      if (slot < total_slots) {
+        const int ctx_i = slot / per_ctx;
+        const int s_in = slot - ctx_i * per_ctx;
+        int j = lo + ctx_i;
+        if (j >= i) ++j;                       // skip the center position
+        const int32_t t = sent_lds[j];
+        if (s_in == 0) {
+          enc = (uint32_t)t | 0x80000000u;     // positive: bit 31 set
+          valid = true;
+        } else {
+          const uint64_t kbase = kNegBase +
+              (uint64_t)(i * (2 * window + 1) + (j - i + window)) *
+                  (uint64_t)n_neg;
+          const uint32_t un = draw_u32(base, kbase + (uint64_t)(s_in - 1));
+          const int32_t neg = table[un % table_size];
+          valid = (neg != t);                  // discard colliding negative
+          enc = (uint32_t)neg;
+        }
+      }
+      const uint64_t m = __ballot(valid);
+      const int pos = __popcll(m & ((1ULL << lane) - 1ULL));
+      if (valid) tgt_lds[pos] = enc;
+      const int count = __popcll(m);
+      if (count > 0) {
+        ph.process_pairs(tgt_lds, count, pair_idx);
+        pair_idx += count;
       }
     }
     ph.end_position(c);
@@ -273,6 +299,10 @@ __device__ __forceinline__ void walk_sentence_dev(
 }
 
 // ---- Phase: fully fused train (dot + sigmoid + update in one pass) -------
+// process_pairs: 2-deep software pipeline — the next target row's loads are
+// issued before the current pair's dot/update consumes its row, hiding the
+// HBM/LLC gather latency under compute (guide G7/G15; explicit ping-pong
+// buffers keep indices compile-time, rule 20).
 template <typename T, int NC, bool ATOMIC>
 struct TrainPhase {
   T* syn0;
@@ -280,8 +310,10 @@ struct TrainPhase {
   int64_t stride;
   float alpha;
   int lane;
+  int32_t atomic_below;
   // per-position state
   T* c_ptr;
+  int32_t c_idx;
   float c_row[NC];
   float grad[NC];
   // stats
@@ -289,42 +321,80 @@ struct TrainPhase {
   float w_fplus = 0.0f;
 
   __device__ __forceinline__ void begin_position(int32_t c) {
+    c_idx = c;
     c_ptr = syn0 + (int64_t)c * stride;
     RowIO<T, NC>::load(c_ptr, c_row, lane);
 #pragma unroll
     for (int k = 0; k < NC; ++k) grad[k] = 0.0f;
   }
-  __device__ __forceinline__ void pair(int32_t tgt, float label, int64_t) {
-    T* t_ptr = syn1 + (int64_t)tgt * stride;
-    float t_row[NC];
-    RowIO<T, NC>::load(t_ptr, t_row, lane);
+
+  __device__ __forceinline__ T* rowptr(uint32_t enc) const {
+    return syn1 + (int64_t)(enc & 0x7FFFFFFFu) * stride;
+  }
+
+  __device__ __forceinline__ void do_pair(uint32_t enc, float (&t_row)[NC],
+                                          T* t_ptr) {
+    const float label = (enc & 0x80000000u) ? 1.0f : 0.0f;
     float f = 0.0f;
 #pragma unroll
     for (int k = 0; k < NC; ++k) f += c_row[k] * t_row[k];
 #pragma unroll
     for (int sh = 32; sh > 0; sh >>= 1) f += __shfl_xor(f, sh, 64);
     const float g = (label - sigmoid_clipped(f)) * alpha;
-#pragma unroll
-    for (int k = 0; k < NC; ++k) grad[k] += g * t_row[k];
-    if (ATOMIC) {
+    const bool use_atomic =
+        ATOMIC && ((int32_t)(enc & 0x7FFFFFFFu) < atomic_below);
+    if (use_atomic) {
       float delta[NC];
 #pragma unroll
-      for (int k = 0; k < NC; ++k) delta[k] = g * c_row[k];
+      for (int k = 0; k < NC; ++k) {
+        grad[k] += g * t_row[k];
+        delta[k] = g * c_row[k];
+      }
       RowIO<T, NC>::atomic_add(t_ptr, delta, lane);
     } else {
 #pragma unroll
-      for (int k = 0; k < NC; ++k) t_row[k] += g * c_row[k];
+      for (int k = 0; k < NC; ++k) {
+        grad[k] += g * t_row[k];
+        t_row[k] += g * c_row[k];
+      }
       RowIO<T, NC>::store(t_ptr, t_row, lane);
     }
     ++w_pairs;
-    if (label > 0.5f) {
+    if (enc & 0x80000000u) {
       ++w_pos;
       w_fplus += f;
     }
   }
+
+  __device__ __forceinline__ void process_pairs(const uint32_t* tl, int count,
+                                                int64_t) {
+    float b0[NC], b1[NC];
+    uint32_t e0 = tl[0], e1 = 0;
+    T* p0 = rowptr(e0);
+    T* p1 = nullptr;
+    RowIO<T, NC>::load(p0, b0, lane);
+    int k = 0;
+    for (;;) {
+      if (k + 1 < count) {
+        e1 = tl[k + 1];
+        p1 = rowptr(e1);
+        RowIO<T, NC>::load(p1, b1, lane);
+      }
+      do_pair(e0, b0, p0);
+      if (++k >= count) break;
+      if (k + 1 < count) {
+        e0 = tl[k + 1];
+        p0 = rowptr(e0);
+        RowIO<T, NC>::load(p0, b0, lane);
+      }
+      do_pair(e1, b1, p1);
+      if (++k >= count) break;
+    }
+  }
+
   __device__ __forceinline__ void end_position(int32_t) {
     // center row update (hogwild: re-read current value, add, store)
-    if (ATOMIC) {
+    if (ATOMIC && (c_idx < atomic_below)) {
       RowIO<T, NC>::atomic_add(c_ptr, grad, lane);
     } else {
       float cur[NC];
@@ -340,6 +410,7 @@ struct TrainPhase {
 template <typename T, int NC, bool ATOMIC>
 __global__ __launch_bounds__(64 * kWavesPerBlock) void sgns_train_kernel(KernelArgs a) {
   __shared__ int32_t sbuf[kWavesPerBlock][kMaxSent];
+  __shared__ uint32_t tbuf[kWavesPerBlock][64];
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int waves_in_block = blockDim.x >> 6;   // 1 in serial mode, else 4
@@ -352,13 +423,14 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void sgns_train_kernel(KernelA
   ph.stride = a.stride;
   ph.alpha = a.alpha;
   ph.lane = lane;
+  ph.atomic_below = a.atomic_below;
 
   for (int64_t s = wave_gid; s < a.num_sentences; s += total_waves) {
     const uint64_t base = sent_base(a.seed, (uint64_t)(a.sent_id_base + s));
     walk_sentence_dev(a.tokens, a.offsets[s],
                       (int)(a.offsets[s + 1] - a.offsets[s]), base, a.keep_thr,
                       a.table, a.table_size, a.window, a.n_neg, a.ref_window,
-                      lane, sbuf[wave], ph);
+                      lane, sbuf[wave], tbuf[wave], ph);
   }
 
   // ---- stats (device-scope atomics, once per wave) -----------------------
@@ -374,13 +446,17 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void sgns_train_kernel(KernelA
 struct CountPhase {
   int64_t pairs = 0;
   __device__ __forceinline__ void begin_position(int32_t) {}
-  __device__ __forceinline__ void pair(int32_t, float, int64_t) { ++pairs; }
+  __device__ __forceinline__ void process_pairs(const uint32_t*, int count,
+                                                int64_t) {
+    pairs += count;
+  }
   __device__ __forceinline__ void end_position(int32_t) {}
 };
 
 __global__ __launch_bounds__(64 * kWavesPerBlock) void count_pairs_kernel(
     KernelArgs a, int64_t* __restrict__ counts) {
   __shared__ int32_t sbuf[kWavesPerBlock][kMaxSent];
+  __shared__ uint32_t tbuf[kWavesPerBlock][64];
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int waves_in_block = blockDim.x >> 6;
@@ -392,7 +468,7 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void count_pairs_kernel(
     walk_sentence_dev(a.tokens, a.offsets[s],
                       (int)(a.offsets[s + 1] - a.offsets[s]), base, a.keep_thr,
                       a.table, a.table_size, a.window, a.n_neg, a.ref_window,
-                      lane, sbuf[wave], ph);
+                      lane, sbuf[wave], tbuf[wave], ph);
     if (lane == 0) counts[s] = ph.pairs;
   }
 }
@@ -409,15 +485,31 @@ struct DotPhase {
   __device__ __forceinline__ void begin_position(int32_t c) {
     RowIO<T, NC>::load(syn0 + (int64_t)c * stride, c_row, lane);
   }
-  __device__ __forceinline__ void pair(int32_t tgt, float, int64_t idx) {
-    float t_row[NC];
-    RowIO<T, NC>::load(syn1 + (int64_t)tgt * stride, t_row, lane);
+  __device__ __forceinline__ const T* rowptr(uint32_t enc) const {
+    return syn1 + (int64_t)(enc & 0x7FFFFFFFu) * stride;
+  }
+  __device__ __forceinline__ void dot_one(const float (&t_row)[NC], int64_t idx) {
     float f = 0.0f;
 #pragma unroll
     for (int k = 0; k < NC; ++k) f += c_row[k] * t_row[k];
 #pragma unroll
     for (int sh = 32; sh > 0; sh >>= 1) f += __shfl_xor(f, sh, 64);
     if (lane == 0) f_base[idx] = f;
+  }
+  __device__ __forceinline__ void process_pairs(const uint32_t* tl, int count,
+                                                int64_t idx_base) {
+    float b0[NC], b1[NC];
+    const T* p0 = rowptr(tl[0]);
+    RowIO<T, NC>::load(p0, b0, lane);
+    int k = 0;
+    for (;;) {
+      if (k + 1 < count) RowIO<T, NC>::load(rowptr(tl[k + 1]), b1, lane);
+      dot_one(b0, idx_base + k);
+      if (++k >= count) break;
+      if (k + 1 < count) RowIO<T, NC>::load(rowptr(tl[k + 1]), b0, lane);
+      dot_one(b1, idx_base + k);
+      if (++k >= count) break;
+    }
   }
   __device__ __forceinline__ void end_position(int32_t) {}
 };
@@ -427,6 +519,7 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void dots_slice_kernel(
     KernelArgs a, const int64_t* __restrict__ pair_offsets,
     float* __restrict__ f_out) {
   __shared__ int32_t sbuf[kWavesPerBlock][kMaxSent];
+  __shared__ uint32_t tbuf[kWavesPerBlock][64];
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int waves_in_block = blockDim.x >> 6;
@@ -443,7 +536,7 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void dots_slice_kernel(
     walk_sentence_dev(a.tokens, a.offsets[s],
                       (int)(a.offsets[s + 1] - a.offsets[s]), base, a.keep_thr,
                       a.table, a.table_size, a.window, a.n_neg, a.ref_window,
-                      lane, sbuf[wave], ph);
+                      lane, sbuf[wave], tbuf[wave], ph);
   }
 }
 
@@ -461,21 +554,26 @@ struct UpdateSlicePhase {
   float world_scale;
   float alpha;
   int lane;
+  int32_t atomic_below;
   T* c_ptr;
+  int32_t c_idx;
   float c_row[NC];
   float grad[NC];
   unsigned long long w_pairs = 0, w_pos = 0, w_words = 0;
   float w_fplus = 0.0f;
   __device__ __forceinline__ void begin_position(int32_t c) {
+    c_idx = c;
     c_ptr = syn0 + (int64_t)c * stride;
     RowIO<T, NC>::load(c_ptr, c_row, lane);
 #pragma unroll
     for (int k = 0; k < NC; ++k) grad[k] = 0.0f;
   }
-  __device__ __forceinline__ void pair(int32_t tgt, float label, int64_t idx) {
-    T* t_ptr = syn1 + (int64_t)tgt * stride;
-    float t_row[NC];
-    RowIO<T, NC>::load(t_ptr, t_row, lane);
+  __device__ __forceinline__ T* rowptr(uint32_t enc) const {
+    return syn1 + (int64_t)(enc & 0x7FFFFFFFu) * stride;
+  }
+  __device__ __forceinline__ void do_pair(uint32_t enc, float (&t_row)[NC],
+                                          T* t_ptr, int64_t idx) {
+    const float label = (enc & 0x80000000u) ? 1.0f : 0.0f;
     float f = f_base[idx];
     if (f_loc) {
       float fresh = 0.0f;
@@ -486,7 +584,9 @@ struct UpdateSlicePhase {
       f += world_scale * (fresh - f_loc[idx]);
     }
     const float g = (label - sigmoid_clipped(f)) * alpha;
-    if (ATOMIC) {
+    const bool use_atomic =
+        ATOMIC && ((int32_t)(enc & 0x7FFFFFFFu) < atomic_below);
+    if (use_atomic) {
       float delta[NC];
 #pragma unroll
       for (int k = 0; k < NC; ++k) {
@@ -503,13 +603,38 @@ struct UpdateSlicePhase {
       RowIO<T, NC>::store(t_ptr, t_row, lane);
     }
     ++w_pairs;
-    if (label > 0.5f) {
+    if (enc & 0x80000000u) {
       ++w_pos;
       w_fplus += f;
     }
   }
+  __device__ __forceinline__ void process_pairs(const uint32_t* tl, int count,
+                                                int64_t idx_base) {
+    float b0[NC], b1[NC];
+    uint32_t e0 = tl[0], e1 = 0;
+    T* p0 = rowptr(e0);
+    T* p1 = nullptr;
+    RowIO<T, NC>::load(p0, b0, lane);
+    int k = 0;
+    for (;;) {
+      if (k + 1 < count) {
+        e1 = tl[k + 1];
+        p1 = rowptr(e1);
+        RowIO<T, NC>::load(p1, b1, lane);
+      }
+      do_pair(e0, b0, p0, idx_base + k);
+      if (++k >= count) break;
+      if (k + 1 < count) {
+        e0 = tl[k + 1];
+        p0 = rowptr(e0);
+        RowIO<T, NC>::load(p0, b0, lane);
+      }
+      do_pair(e1, b1, p1, idx_base + k);
+      if (++k >= count) break;
+    }
+  }
   __device__ __forceinline__ void end_position(int32_t) {
-    if (ATOMIC) {
+    if (ATOMIC && (c_idx < atomic_below)) {
       RowIO<T, NC>::atomic_add(c_ptr, grad, lane);
     } else {
       float cur[NC];
@@ -528,6 +653,7 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void update_slice_kernel(
     const float* __restrict__ f_in, const float* __restrict__ f_loc,
     float world_scale) {
   __shared__ int32_t sbuf[kWavesPerBlock][kMaxSent];
+  __shared__ uint32_t tbuf[kWavesPerBlock][64];
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int waves_in_block = blockDim.x >> 6;
@@ -540,6 +666,7 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void update_slice_kernel(
   ph.alpha = a.alpha;
   ph.lane = lane;
   ph.world_scale = world_scale;
+  ph.atomic_below = a.atomic_below;
   for (int64_t s = wave_gid; s < a.num_sentences; s += total_waves) {
     ph.f_base = f_in + pair_offsets[s];
     ph.f_loc = f_loc ? f_loc + pair_offsets[s] : nullptr;
@@ -547,7 +674,7 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void update_slice_kernel(
     walk_sentence_dev(a.tokens, a.offsets[s],
                       (int)(a.offsets[s + 1] - a.offsets[s]), base, a.keep_thr,
                       a.table, a.table_size, a.window, a.n_neg, a.ref_window,
-                      lane, sbuf[wave], ph);
+                      lane, sbuf[wave], tbuf[wave], ph);
   }
   if (lane == 0 && a.d_pairs) {
     atomicAdd(a.d_pairs, ph.w_pairs);
@@ -719,7 +846,8 @@ static void sgns_train(uintptr_t syn0, uintptr_t syn1, int is_bf16,
                        int64_t num_sentences, uintptr_t keep_thr,
                        uintptr_t table, int64_t table_size, double alpha,
                        int window, int n_neg, uint64_t seed,
-                       int64_t sent_id_base, int ref_window, int atomic,
+                       int64_t sent_id_base, int ref_window,
+                       int64_t atomic_below,
                        uintptr_t stats, int blocks, int threads,
                        uintptr_t stream_ptr) {
   if (threads != 64 && threads != 256)
@@ -745,6 +873,7 @@ static void sgns_train(uintptr_t syn0, uintptr_t syn1, int is_bf16,
   a.sent_id_base = sent_id_base;
   a.stride = stride;
   a.ref_window = ref_window;
+  a.atomic_below = (int32_t)std::min<int64_t>(atomic_below, 0x7FFFFFFFLL);
   unsigned long long* st = (unsigned long long*)stats;
   if (st) {
     a.d_pairs = st + 0;
@@ -753,10 +882,11 @@ static void sgns_train(uintptr_t syn0, uintptr_t syn1, int is_bf16,
     a.d_sum_fplus = (double*)(st + 3);
   }
   hipStream_t stream = (hipStream_t)stream_ptr;
+  const bool use_atomic = atomic_below > 0;
   if (is_bf16)
-    launch_train<uint16_t>(a, nc, atomic != 0, blocks, threads, stream);
+    launch_train<uint16_t>(a, nc, use_atomic, blocks, threads, stream);
   else
-    launch_train<float>(a, nc, atomic != 0, blocks, threads, stream);
+    launch_train<float>(a, nc, use_atomic, blocks, threads, stream);
   HIP_CHECK(hipGetLastError());
 }
 
@@ -849,13 +979,16 @@ static void update_slice(uintptr_t syn0, uintptr_t syn1, int is_bf16,
                          int window, int n_neg, uint64_t seed,
                          int64_t sent_id_base, int ref_window,
                          uintptr_t pair_offsets, uintptr_t f_in,
-                         uintptr_t f_loc, double world_scale, int atomic,
+                         uintptr_t f_loc, double world_scale,
+                         int64_t atomic_below,
                          uintptr_t stats, int blocks, int threads,
                          uintptr_t stream_ptr) {
   KernelArgs a = make_walk_args(syn0, syn1, stride, tokens, offsets,
                                 num_sentences, keep_thr, table, table_size,
                                 alpha, window, n_neg, seed, sent_id_base,
                                 ref_window, stats);
+  a.atomic_below = (int32_t)std::min<int64_t>(atomic_below, 0x7FFFFFFFLL);
+  const int atomic = atomic_below > 0;
   const int nc = (int)(stride / 64);
   hipStream_t stream = (hipStream_t)stream_ptr;
 #define UPD_CASE(T, N)                                                        \

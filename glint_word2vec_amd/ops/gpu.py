@@ -110,7 +110,8 @@ class GpuSgns:
     def train_batch(self, tokens: torch.Tensor, offsets: torch.Tensor,
                     alpha: float, window: int, n_neg: int, seed: int,
                     sent_id_base: int = 0, window_mode: str = "canonical",
-                    atomic: bool = False, blocks: Optional[int] = None,
+                    atomic: bool = True, atomic_below: Optional[int] = None,
+                    blocks: Optional[int] = None,
                     serial: bool = False,
                     stream: Optional[torch.cuda.Stream] = None) -> None:
         """Launch the fused train kernel (async on the given/current stream).
@@ -135,7 +136,9 @@ class GpuSgns:
             0 if self.keep_thr is None else self.keep_thr.data_ptr(),
             self.table.data_ptr(), int(self.table.numel()), float(alpha),
             int(window), int(n_neg), seed & 0xFFFFFFFFFFFFFFFF,
-            int(sent_id_base), int(window_mode == "reference"), int(atomic),
+            int(sent_id_base), int(window_mode == "reference"),
+            0 if not atomic else (2 ** 31 - 1 if atomic_below is None
+                                  else int(atomic_below)),
             self._stats.data_ptr(), nblocks, nthreads, s.cuda_stream)
 
     def read_stats(self, reset: bool = True) -> GpuStats:

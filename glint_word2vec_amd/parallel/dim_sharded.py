@@ -174,7 +174,8 @@ class DimShardedSgns:
                 int(self.window_mode == "reference"), poff.data_ptr(),
                 f.data_ptr(),
                 0 if f_loc is None else f_loc.data_ptr(),
-                float(self.world), int(self.atomic), self._stats.data_ptr(),
+                float(self.world),
+                (2 ** 31 - 1) if self.atomic else 0, self._stats.data_ptr(),
                 nb, nt, stream.cuda_stream)
         else:
             tok_np = tokens.numpy()
