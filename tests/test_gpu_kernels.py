@@ -42,8 +42,10 @@ def test_serial_kernel_matches_oracle(window_mode):
                                        0.03, 3, 4, seed=77, sent_id_base=5,
                                        window_mode=window_mode)
     gs = _gpu_setup(syn0, syn1, table)
+    # serial: no races, plain RMW matches the oracle's rounding exactly
     gs.train_batch(_to_dev(tokens), _to_dev(offsets), 0.03, 3, 4, 77,
-                   sent_id_base=5, window_mode=window_mode, serial=True)
+                   sent_id_base=5, window_mode=window_mode, serial=True,
+                   atomic=False)
     torch.cuda.synchronize()
     st = gs.read_stats()
     assert st.pairs == st_py.pairs
@@ -66,7 +68,7 @@ def test_serial_kernel_with_subsampling():
     gs = _gpu_setup(syn0, syn1, table)
     gs.set_subsample(counts, total, 0.005)
     gs.train_batch(_to_dev(tokens), _to_dev(offsets), 0.03, 3, 4, 13,
-                   serial=True)
+                   serial=True, atomic=False)
     torch.cuda.synchronize()
     st = gs.read_stats()
     assert st.pairs == st_py.pairs
@@ -118,7 +120,7 @@ def test_bf16_kernel_counts_and_direction():
                                        0.03, 3, 4, seed=77)
     gs = _gpu_setup(syn0, syn1, table, dtype="bfloat16")
     gs.train_batch(_to_dev(tokens), _to_dev(offsets), 0.03, 3, 4, 77,
-                   serial=True)
+                   serial=True, atomic=False)
     torch.cuda.synchronize()
     st = gs.read_stats()
     assert st.pairs == st_py.pairs      # RNG identical regardless of dtype
@@ -176,7 +178,7 @@ def test_dim_sharded_gpu_world1_matches_sequential():
                                 table, 0.03, 3, 4, 42, 0, "canonical", 1)
     eng = DimShardedSgns(60, 24, device="cuda", seed=3, counts=counts,
                          table_size=1009, chunk_words=10 ** 9,
-                         f_correction=True)
+                         f_correction=True, atomic=False)
     eng.serial = True
     tok = torch.from_numpy(batch.tokens).cuda()
     off = torch.from_numpy(batch.offsets).cuda()
