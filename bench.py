@@ -38,10 +38,12 @@ def parse_args():
     p.add_argument("--subsample", type=float, default=1e-4,
                    help="subsample ratio (drawn in-kernel; 0 disables)")
     p.add_argument("--blocks", type=int, default=0, help="grid blocks override")
-    p.add_argument("--hogwild", action="store_true",
-                   help="plain read-modify-write row updates (no atomics)")
+    p.add_argument("--atomic", action="store_true",
+                   help="atomic row updates (no lost updates; ~3-5x slower). "
+                        "Default is hogwild plain RMW — the reference's "
+                        "fire-and-forget adjust semantics (mllib:425)")
     p.add_argument("--atomic-below", type=int, default=None,
-                   help="atomics only for rows < K (hot rows); default all")
+                   help="atomics only for rows < K (hot rows)")
     p.add_argument("--profile-steps", type=int, default=0,
                    help="run only this many steps, no warmup JSON (rocprof)")
     return p.parse_args()
@@ -84,7 +86,8 @@ def main():
         trainer = DimShardedSgns(args.vocab, args.dim, dtype=dtype,
                                  device=str(device), seed=1, counts=counts,
                                  table_size=args.table_size,
-                                 subsample=args.subsample)
+                                 subsample=args.subsample,
+                                 atomic=bool(args.atomic or args.atomic_below))
     else:
         from glint_word2vec_amd.ops.gpu import GpuSgns
         gs = GpuSgns(args.vocab, args.dim, dtype=dtype, device=str(device), seed=1)
@@ -107,7 +110,7 @@ def main():
         else:
             trainer.train_batch(tok, off, alpha, args.window, args.neg, 99,
                                 sent_id_base=i * nsent,
-                                atomic=not args.hogwild,
+                                atomic=bool(args.atomic or args.atomic_below),
                                 atomic_below=args.atomic_below,
                                 blocks=args.blocks or None)
 
@@ -167,6 +170,8 @@ def main():
                 "seq_len": args.sentence_len,
                 "parallelism": (f"dimshard-rccl-allreduce-x{world}"
                                 if distributed else "hogwild-1gpu"),
+                "updates": ("atomic" if (args.atomic or args.atomic_below)
+                            else "hogwild"),
             },
             "pairs_per_step": st.pairs / max(args.steps, 1),
             "mean_fplus": st.sum_fplus / max(st.positives, 1),
