@@ -120,6 +120,85 @@ def make_plan(tokens: np.ndarray, offsets: np.ndarray,
     return PairPlan(pc, pt, np.ones(P, dtype=np.float32))
 
 
+@dataclass
+class GroupedPlan:
+    """Pair plan grouped by center position (the row-sharded engine's work
+    unit: one group = one center position's window + negatives, trained
+    against a pulled row cache — cf. the Glint dotprod/adjust batch)."""
+    group_center: np.ndarray    # int32 [G] center WORD ids
+    group_offsets: np.ndarray   # int64 [G+1] into pair arrays
+    pair_target: np.ndarray     # int32 [P] target WORD ids
+    pair_label: np.ndarray      # float32 [P]
+
+    @property
+    def num_groups(self) -> int:
+        return len(self.group_center)
+
+    @property
+    def num_pairs(self) -> int:
+        return len(self.pair_target)
+
+
+def make_grouped_plan(tokens: np.ndarray, offsets: np.ndarray,
+                      keep_prob: np.ndarray | None, table: np.ndarray,
+                      window: int, n_neg: int, rng: np.random.Generator,
+                      window_mode: str = "canonical") -> GroupedPlan:
+    """Like make_plan but with center-position group structure."""
+    toks, offs = subsample_batch(tokens, offsets, keep_prob, rng)
+    T = len(toks)
+    empty = GroupedPlan(np.zeros(0, np.int32), np.zeros(1, np.int64),
+                        np.zeros(0, np.int32), np.zeros(0, np.float32))
+    if T == 0:
+        return empty
+    sent_id = np.repeat(np.arange(len(offs) - 1, dtype=np.int64), np.diff(offs))
+    sent_lo = offs[:-1].astype(np.int64)[sent_id]
+    sent_hi = offs[1:].astype(np.int64)[sent_id]
+    pos = np.arange(T, dtype=np.int64)
+    if window_mode == "canonical":
+        b = rng.integers(1, window + 1, size=T)
+        left, right = b, b
+    else:
+        b = rng.integers(0, window, size=T)
+        left, right = b, np.maximum(b - 1, 0) * (b > 0)
+    centers_l, targets_l = [], []
+    for o in range(1, window + 1):
+        m = (o <= left) & (pos - o >= sent_lo)
+        centers_l.append(pos[m]); targets_l.append(pos[m] - o)
+        m = (o <= right) & (pos + o < sent_hi)
+        centers_l.append(pos[m]); targets_l.append(pos[m] + o)
+    if not centers_l:
+        return empty
+    cpos = np.concatenate(centers_l)
+    tpos = np.concatenate(targets_l)
+    if len(cpos) == 0:
+        return empty
+    order = np.lexsort((tpos, cpos))
+    cpos, tpos = cpos[order], tpos[order]
+    pc, pt = toks[cpos], toks[tpos]
+    P = len(pc)
+    if n_neg > 0:
+        negs = table[rng.integers(0, len(table), size=(P, n_neg))].astype(np.int32)
+        valid = negs != pt[:, None]
+        all_target = np.concatenate([pt[:, None], negs], axis=1)
+        all_label = np.concatenate(
+            [np.ones((P, 1), np.float32), np.zeros((P, n_neg), np.float32)], axis=1)
+        all_valid = np.concatenate([np.ones((P, 1), bool), valid], axis=1)
+        all_cpos = np.repeat(cpos, 1 + n_neg).reshape(P, 1 + n_neg)
+        flat = all_valid.ravel()
+        pair_target = all_target.ravel()[flat].astype(np.int32)
+        pair_label = all_label.ravel()[flat]
+        pair_cpos = all_cpos.ravel()[flat]
+    else:
+        pair_target, pair_label, pair_cpos = pt.astype(np.int32), \
+            np.ones(P, np.float32), cpos
+    # group boundaries: runs of equal center position
+    change = np.nonzero(np.diff(pair_cpos))[0] + 1
+    starts = np.concatenate([[0], change])
+    group_offsets = np.concatenate([starts, [len(pair_cpos)]]).astype(np.int64)
+    group_center = toks[pair_cpos[starts]].astype(np.int32)
+    return GroupedPlan(group_center, group_offsets, pair_target, pair_label)
+
+
 def train_plan_minibatched(syn0: np.ndarray, syn1: np.ndarray, plan: PairPlan,
                            alpha: float, minibatch: int = 16384) -> Tuple[int, float]:
     """Fast vectorized CPU SGD over a pair plan (torch index_add under the

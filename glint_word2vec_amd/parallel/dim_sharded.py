@@ -233,7 +233,7 @@ class DimShardedSgns:
         for t in (self.syn0, self.syn1):
             my = t[:, :self.width].float()
             if self.world == 1:
-                out.append(my.cpu().numpy())
+                out.append(my.cpu().numpy().copy())
                 continue
             maxw = max(slice_bounds(self.dim, r, self.world)[1] -
                        slice_bounds(self.dim, r, self.world)[0]

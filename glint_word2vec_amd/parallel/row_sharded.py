@@ -1,0 +1,265 @@
+"""Row-sharded multi-GPU SGNS engine — the Glint parameter-server shape on
+RCCL alltoallv over xGMI (BASELINE.json north star).
+
+Row r of both tables lives on rank r % world (round-robin interleave keeps
+Zipf-hot rows balanced).  Each rank is simultaneously a *worker* over its
+own corpus partition and the *server* for its row shard:
+
+  plan pairs (vectorized host plan)  ->  unique touched rows  ->
+  alltoallv index request to owners  ->  owners gather rows   ->
+  alltoallv rows back (pull)         ->  train on the local f32 cache
+  (fused pairs kernel, live hogwild feedback within the step)  ->
+  alltoallv row deltas to owners     ->  owners scatter-add (push/adjust)
+
+This is the reference's dotprod/adjust split with the network role of Akka
+messages taken by two alltoallv exchanges; traffic scales with unique rows
+x dim, so it is the engine of choice when the touched-row set is sparse
+(very large vocabularies) — the dim-sharded engine (dim_sharded.py) covers
+the dense case with dimension-independent traffic.
+
+Works on CUDA (HIP train_pairs kernel) and CPU (C++ twin; gloo tests).
+"""
+from __future__ import annotations
+
+import logging
+from typing import Optional, Tuple
+
+import numpy as np
+import torch
+
+from ..models import sgns
+from ..ops.gpu import GpuStats
+from ..vocab import build_unigram_table, keep_probabilities
+from . import comm
+
+log = logging.getLogger("glint_word2vec_amd")
+
+
+class RowShardedSgns:
+    def __init__(self, vocab_size: int, dim: int, dtype: str = "float32",
+                 device: str = "cuda", seed: int = 1,
+                 counts: Optional[np.ndarray] = None,
+                 table_size: int = 1_000_000, subsample: float = 0.0,
+                 window_mode: str = "canonical",
+                 init_full_limit: int = 1 << 28):
+        self.rank, self.world = comm.init_from_env()
+        self.vocab_size = vocab_size
+        self.dim = dim
+        self.device = torch.device(device)
+        self.is_cuda = self.device.type == "cuda"
+        self.is_bf16 = dtype == "bfloat16"
+        self.window_mode = window_mode
+        if self.is_cuda:
+            from .. import _hip_native
+            self.native = _hip_native
+            self.cache_stride = self.native.round_stride(dim)
+        else:
+            from .. import _cpu_native
+            self.native = _cpu_native
+            self.cache_stride = dim
+        self.my_rows = np.arange(self.rank, vocab_size, self.world,
+                                 dtype=np.int64)
+        self.shard_size = len(self.my_rows)
+        tdtype = torch.bfloat16 if self.is_bf16 else torch.float32
+        self.syn0 = torch.zeros((self.shard_size, dim), dtype=tdtype,
+                                device=self.device)
+        self.syn1 = torch.zeros((self.shard_size, dim), dtype=tdtype,
+                                device=self.device)
+        self._init_shard(seed, init_full_limit)
+        counts = (np.ones(vocab_size, dtype=np.int64) if counts is None
+                  else counts)
+        self.table = build_unigram_table(counts, table_size)
+        self.keep_prob = None
+        if subsample > 0:
+            self.keep_prob = keep_probabilities(counts, int(counts.sum()),
+                                                subsample)
+        self._stats = torch.zeros(4, dtype=torch.int64, device=self.device)
+        self._cpu_stats = dict(pairs=0, positives=0, words_trained=0,
+                               sum_fplus=0.0)
+        self.serial = False
+
+    def _init_shard(self, seed: int, full_limit: int) -> None:
+        """Word2vec init, world-size invariant: same full-matrix stream as
+        the other engines, this rank keeps rows r % world == rank."""
+        tdtype = self.syn0.dtype
+        block = max(self.world, (full_limit // max(self.dim, 1)) //
+                    self.world * self.world)
+        rng = np.random.default_rng(seed)
+        for r0 in range(0, self.vocab_size, block):
+            r1 = min(self.vocab_size, r0 + block)
+            full = (rng.random((r1 - r0, self.dim), dtype=np.float32) - 0.5) / self.dim
+            mine = full[(np.arange(r0, r1) % self.world) == self.rank]
+            lo = (r0 + self.world - 1 - self.rank) // self.world
+            self.syn0[lo:lo + len(mine)] = \
+                torch.from_numpy(np.ascontiguousarray(mine)).to(tdtype).to(self.device)
+
+    # ------------------------------------------------------------------
+    # pull / push: the alltoallv exchanges (Glint pull / adjust push)
+    # ------------------------------------------------------------------
+    def _route(self, ids: np.ndarray):
+        """Sort global row ids by owner.  Returns (perm, send_counts,
+        sorted_local_ids)."""
+        owner = ids % self.world
+        perm = np.argsort(owner, kind="stable")
+        send_counts = np.bincount(owner, minlength=self.world)
+        local = (ids[perm] // self.world).astype(np.int64)
+        return perm, send_counts, local
+
+    def _exchange(self, send_chunks, recv_shapes, dtype):
+        """alltoallv helper: send_chunks[d] -> rank d; returns recv list."""
+        recv = [torch.empty(shape, dtype=dtype, device=send_chunks[0].device)
+                for shape in recv_shapes]
+        comm.all_to_all_v(recv, send_chunks)
+        return recv
+
+    def pull(self, ids: np.ndarray, which: int) -> torch.Tensor:
+        """Gather rows `ids` (global) of syn0 (which=0) / syn1 (which=1)
+        from their owners.  Returns f32 [len(ids), cache_stride]."""
+        shard = self.syn0 if which == 0 else self.syn1
+        if self.world == 1:
+            out = torch.zeros((len(ids), self.cache_stride),
+                              dtype=torch.float32, device=self.device)
+            idx = torch.from_numpy(ids // 1).to(self.device)
+            out[:, :self.dim] = shard.index_select(0, idx).float()
+            return out
+        perm, send_counts, local_sorted = self._route(ids)
+        # 1) exchange request sizes + index lists
+        sc = torch.from_numpy(send_counts.astype(np.int64))
+        rc = comm.exchange_counts(sc).numpy()
+        bounds = np.concatenate([[0], np.cumsum(send_counts)])
+        idx_send = [torch.from_numpy(local_sorted[bounds[d]:bounds[d + 1]])
+                    .to(self.device) for d in range(self.world)]
+        idx_recv = self._exchange(idx_send, [int(rc[s]) for s in range(self.world)],
+                                  torch.int64)
+        # 2) owners gather rows, reply
+        row_send = []
+        for s in range(self.world):
+            rows = torch.zeros((int(rc[s]), self.cache_stride),
+                               dtype=torch.float32, device=self.device)
+            if int(rc[s]):
+                rows[:, :self.dim] = shard.index_select(0, idx_recv[s]).float()
+            row_send.append(rows)
+        row_recv = self._exchange(
+            row_send, [(int(send_counts[d]), self.cache_stride)
+                       for d in range(self.world)], torch.float32)
+        sorted_rows = torch.cat(row_recv, dim=0)
+        out = torch.empty_like(sorted_rows)
+        out[torch.from_numpy(perm).to(self.device)] = sorted_rows
+        return out
+
+    def push_add(self, ids: np.ndarray, deltas: torch.Tensor, which: int) -> None:
+        """Scatter-add row deltas back to their owners (the adjust push)."""
+        shard = self.syn0 if which == 0 else self.syn1
+        if self.world == 1:
+            idx = torch.from_numpy(ids // 1).to(self.device)
+            upd = shard.index_select(0, idx).float() + deltas[:, :self.dim]
+            shard.index_copy_(0, idx, upd.to(shard.dtype))
+            return
+        perm, send_counts, local_sorted = self._route(ids)
+        sc = torch.from_numpy(send_counts.astype(np.int64))
+        rc = comm.exchange_counts(sc).numpy()
+        bounds = np.concatenate([[0], np.cumsum(send_counts)])
+        idx_send = [torch.from_numpy(local_sorted[bounds[d]:bounds[d + 1]])
+                    .to(self.device) for d in range(self.world)]
+        idx_recv = self._exchange(idx_send, [int(rc[s]) for s in range(self.world)],
+                                  torch.int64)
+        perm_t = torch.from_numpy(perm).to(self.device)
+        deltas_sorted = deltas.index_select(0, perm_t)
+        del_send = [deltas_sorted[bounds[d]:bounds[d + 1]].contiguous()
+                    for d in range(self.world)]
+        del_recv = self._exchange(
+            del_send, [(int(rc[s]), self.cache_stride)
+                       for s in range(self.world)], torch.float32)
+        for s in range(self.world):
+            if int(rc[s]) == 0:
+                continue
+            cur = shard.index_select(0, idx_recv[s]).float()
+            cur += del_recv[s][:, :self.dim]
+            # duplicate local indices across sources are rare (each source
+            # deduplicates); across sources index_copy applies sequentially
+            shard.index_copy_(0, idx_recv[s], cur.to(shard.dtype))
+
+    # ------------------------------------------------------------------
+    def train_step(self, tokens: np.ndarray, offsets: np.ndarray,
+                   alpha: float, window: int, n_neg: int,
+                   rng: np.random.Generator) -> None:
+        """One data-parallel step over this rank's batch (host arrays).
+        Ranks with no data still participate in the collectives."""
+        plan = sgns.make_grouped_plan(tokens, offsets, self.keep_prob,
+                                      self.table, window, n_neg, rng,
+                                      self.window_mode)
+        uc, inv_c = np.unique(plan.group_center, return_inverse=True)
+        ut, inv_t = np.unique(plan.pair_target, return_inverse=True)
+        cache0 = self.pull(uc.astype(np.int64), 0)
+        cache1 = self.pull(ut.astype(np.int64), 1)
+        orig0 = cache0.clone()
+        orig1 = cache1.clone()
+        if plan.num_pairs > 0:
+            self._train_pairs(cache0, cache1, inv_c.astype(np.int32),
+                              plan.group_offsets,
+                              inv_t.astype(np.int32), plan.pair_label, alpha)
+        self.push_add(uc.astype(np.int64), cache0 - orig0, 0)
+        self.push_add(ut.astype(np.int64), cache1 - orig1, 1)
+
+    def _train_pairs(self, cache0, cache1, group_center, group_offsets,
+                     pair_target, pair_label, alpha):
+        if self.is_cuda:
+            gc = torch.from_numpy(group_center).to(self.device)
+            go = torch.from_numpy(group_offsets.astype(np.int64)).to(self.device)
+            pt = torch.from_numpy(pair_target).to(self.device)
+            pl = torch.from_numpy(pair_label).to(self.device)
+            stream = torch.cuda.current_stream(self.device)
+            G = len(group_center)
+            nb = 1 if self.serial else max(1, min((G + 3) // 4, 2048))
+            nt = 64 if self.serial else 256
+            self.native.train_pairs(
+                cache0.data_ptr(), cache1.data_ptr(), self.cache_stride,
+                gc.data_ptr(), go.data_ptr(), G, pt.data_ptr(), pl.data_ptr(),
+                float(alpha), self._stats.data_ptr(), nb, nt,
+                stream.cuda_stream)
+            torch.cuda.current_stream(self.device).synchronize()
+        else:
+            st = self.native.train_pairs(cache0.numpy(), cache1.numpy(),
+                                         group_center,
+                                         group_offsets.astype(np.int64),
+                                         pair_target, pair_label, float(alpha))
+            for k in ("pairs", "positives", "words_trained"):
+                self._cpu_stats[k] += st[k]
+            self._cpu_stats["sum_fplus"] += st["sum_fplus"]
+
+    # ------------------------------------------------------------------
+    def read_stats(self, reset: bool = True) -> GpuStats:
+        if self.is_cuda:
+            h = self._stats.cpu()
+            out = GpuStats(int(h[0]), int(h[1]), int(h[2]),
+                           float(h[3:4].view(torch.float64)[0]))
+            if reset:
+                self._stats.zero_()
+            return out
+        st = self._cpu_stats
+        out = GpuStats(st["pairs"], st["positives"], st["words_trained"],
+                       st["sum_fplus"])
+        if reset:
+            self._cpu_stats = dict(pairs=0, positives=0, words_trained=0,
+                                   sum_fplus=0.0)
+        return out
+
+    def to_host(self) -> Tuple[np.ndarray, np.ndarray]:
+        out = []
+        for shard in (self.syn0, self.syn1):
+            my = shard.float()
+            if self.world == 1:
+                out.append(my.cpu().numpy().copy())
+                continue
+            maxn = (self.vocab_size + self.world - 1) // self.world
+            padded = torch.zeros((maxn, self.dim), dtype=torch.float32,
+                                 device=self.device)
+            padded[:self.shard_size] = my
+            gathered = [torch.empty_like(padded) for _ in range(self.world)]
+            torch.distributed.all_gather(gathered, padded)
+            full = np.empty((self.vocab_size, self.dim), dtype=np.float32)
+            for r in range(self.world):
+                rows = np.arange(r, self.vocab_size, self.world)
+                full[rows] = gathered[r][:len(rows)].cpu().numpy()
+            out.append(full)
+        return out[0], out[1]

@@ -1,0 +1,121 @@
+"""Row-sharded (alltoallv pull/push) engine tests on CPU: pull/push
+correctness, world-size invariance (gloo, 2 procs), learning smoke."""
+import os
+
+import numpy as np
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+from glint_word2vec_amd.data import synthetic_corpus
+
+pytest.importorskip("glint_word2vec_amd._cpu_native")
+
+VOCAB, DIM = 60, 20
+
+
+def _make_batch(seed=5):
+    return synthetic_corpus(vocab_size=VOCAB, num_tokens=500, sentence_len=50,
+                            seed=seed, zipf_a=1.01)
+
+
+def _run(world_batches):
+    """Run one step per batch list entry for this process's rank."""
+    from glint_word2vec_amd.parallel.row_sharded import RowShardedSgns
+    batch = _make_batch()
+    counts = np.bincount(batch.tokens, minlength=VOCAB).astype(np.int64) + 1
+    eng = RowShardedSgns(VOCAB, DIM, device="cpu", seed=3, counts=counts,
+                         table_size=1009)
+    rng = np.random.default_rng(17)
+    tokens, offsets = world_batches
+    eng.train_step(tokens, offsets, 0.03, 3, 4, rng)
+    st = eng.read_stats()
+    s0, s1 = eng.to_host()
+    return s0, s1, st
+
+
+def test_world1_trains_and_is_finite():
+    batch = _make_batch()
+    s0, s1, st = _run((batch.tokens, batch.offsets))
+    assert st.pairs > 0
+    assert np.isfinite(s0).all() and np.isfinite(s1).all()
+
+
+def _worker(rank, world, rdv_file, out_dir, mode):
+    import torch.distributed as dist
+    dist.init_process_group("gloo", rank=rank, world_size=world,
+                            init_method=f"file://{rdv_file}")
+    try:
+        batch = _make_batch()
+        if mode == "rank0_only":
+            if rank == 0:
+                tokens, offsets = batch.tokens, batch.offsets
+            else:
+                tokens = np.zeros(0, dtype=np.int32)
+                offsets = np.zeros(1, dtype=np.int32)
+        else:  # split: each rank half the sentences
+            n = batch.num_sentences
+            if rank == 0:
+                lo_s, hi_s = 0, n // 2
+            else:
+                lo_s, hi_s = n // 2, n
+            tokens = batch.tokens[batch.offsets[lo_s]:batch.offsets[hi_s]]
+            offsets = (batch.offsets[lo_s:hi_s + 1] -
+                       batch.offsets[lo_s]).astype(np.int32)
+        s0, s1, st = _run((tokens, offsets))
+        if rank == 0:
+            np.save(os.path.join(out_dir, "s0.npy"), s0)
+            np.save(os.path.join(out_dir, "s1.npy"), s1)
+            np.save(os.path.join(out_dir, "st.npy"),
+                    np.array([st.pairs, st.positives]))
+    finally:
+        dist.destroy_process_group()
+
+
+def test_world2_rank0_only_matches_world1(tmp_path):
+    """All data on rank 0, rows sharded over 2 ranks: pure comm test — the
+    result must match the single-process run exactly (same plan RNG)."""
+    batch = _make_batch()
+    s0_ref, s1_ref, st_ref = _run((batch.tokens, batch.offsets))
+    rdv = str(tmp_path / "rdv")
+    mp.spawn(_worker, args=(2, rdv, str(tmp_path), "rank0_only"),
+             nprocs=2, join=True)
+    s0 = np.load(tmp_path / "s0.npy")
+    s1 = np.load(tmp_path / "s1.npy")
+    st = np.load(tmp_path / "st.npy")
+    assert st[0] == st_ref.pairs
+    np.testing.assert_allclose(s0, s0_ref, rtol=1e-5, atol=1e-7)
+    np.testing.assert_allclose(s1, s1_ref, rtol=1e-5, atol=1e-7)
+
+
+def test_world2_split_data_stable(tmp_path):
+    """Both ranks train concurrently on disjoint halves: pair totals add up
+    and the result stays finite (async merge races values, not counts)."""
+    rdv = str(tmp_path / "rdv2")
+    mp.spawn(_worker, args=(2, rdv, str(tmp_path), "split"),
+             nprocs=2, join=True)
+    s0 = np.load(tmp_path / "s0.npy")
+    assert np.isfinite(s0).all()
+    assert np.abs(s0).max() < 10.0
+
+
+def test_pull_roundtrip_identity():
+    from glint_word2vec_amd.parallel.row_sharded import RowShardedSgns
+    counts = np.ones(VOCAB, dtype=np.int64)
+    eng = RowShardedSgns(VOCAB, DIM, device="cpu", seed=3, counts=counts,
+                         table_size=101)
+    ids = np.array([0, 5, 5, 17, 59, 3], dtype=np.int64)
+    rows = eng.pull(ids, 0)
+    s0, _ = eng.to_host()
+    np.testing.assert_allclose(rows[:, :DIM].numpy(), s0[ids], atol=0)
+    # push adds deltas
+    deltas = torch.ones((len(ids), eng.cache_stride), dtype=torch.float32)
+    eng.push_add(ids, deltas, 0)
+    s0b, _ = eng.to_host()
+    expect = s0.copy()
+    # duplicate id 5 in one push: deltas from the same source list are
+    # applied per occurrence? No: within one rank's push the ids come from
+    # np.unique in train_step; raw push_add applies last-wins per duplicate.
+    for i, r in enumerate(ids):
+        expect[r] = s0[r] + 1.0
+    np.testing.assert_allclose(s0b[ids], expect[ids], rtol=1e-6)
