@@ -56,6 +56,13 @@ class Word2VecConfig:
     # reference's races-embraced semantics, slightly faster, loses updates
     # under heavy contention).
     atomic_updates: bool = True
+    # --- multi-GPU engine --------------------------------------
+    # "auto": fused single-GPU kernel at world 1; dim-sharded at world > 1.
+    # "dim": dimension-sharded (CIKM scheme, RCCL allreduce of partial dots).
+    # "row": row-sharded parameter-server shape (RCCL alltoallv pull/push).
+    engine: str = "auto"             # "auto" | "fused" | "dim" | "row"
+    chunk_words: int = 1 << 16       # dim-sharded feedback chunk
+    f_correction: bool = True        # dim-sharded local-drift freshening
     # --- semantics switches ---------------------
     # The reference's subsampling is a de-facto no-op (integer-division bug,
     # mllib:375-377).  We implement the intended math; set
@@ -90,6 +97,8 @@ class Word2VecConfig:
             raise ValueError(f"unsupported dtype {self.dtype!r}")
         if self.window_mode not in ("canonical", "reference"):
             raise ValueError(f"unsupported window_mode {self.window_mode!r}")
+        if self.engine not in ("auto", "fused", "dim", "row"):
+            raise ValueError(f"unsupported engine {self.engine!r}")
 
     # -- (de)serialisation used by the checkpoint metadata ------------------
     def to_dict(self) -> dict:

@@ -103,7 +103,9 @@ class GlintWord2Vec:
         if device == "auto":
             import torch
             device = "cuda" if torch.cuda.is_available() else "cpu"
-        if device == "cuda":
+        if device == "cuda" or cfg.engine in ("dim", "row"):
+            # single- or multi-GPU engines; dim/row also run on CPU under
+            # gloo (multi-process tests, torchrun without GPUs)
             from .parallel.engine import train_gpu
             syn0, syn1 = train_gpu(cfg, vocab, reader, seed)
         else:

@@ -26,6 +26,20 @@ log = logging.getLogger("glint_word2vec_amd")
 
 def train_gpu(cfg: Word2VecConfig, vocab: Vocabulary, reader: Callable,
               seed: int) -> Tuple[np.ndarray, np.ndarray]:
+    """Dispatch to the right engine: fused single-GPU kernel at world 1,
+    dim-sharded (default) or row-sharded at world > 1 (one rank per GPU,
+    launched via torchrun; every rank calls fit() with the same corpus)."""
+    from . import comm
+    import os
+    if int(os.environ.get("WORLD_SIZE", "1")) > 1:
+        local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+        torch.cuda.set_device(local_rank)
+    rank, world = comm.init_from_env()
+    engine = cfg.engine
+    if engine == "auto":
+        engine = "fused" if world == 1 else "dim"
+    if engine in ("dim", "row") and world >= 1:
+        return _train_sharded(cfg, vocab, reader, seed, engine, rank, world)
     device = torch.device("cuda", torch.cuda.current_device())
     gs = GpuSgns(vocab.num_words, cfg.vector_size, cfg.dtype,
                  device=str(device), seed=seed)
@@ -67,3 +81,72 @@ def train_gpu(cfg: Word2VecConfig, vocab: Vocabulary, reader: Callable,
              "%d pairs, mean_fplus=%.4f", processed, dt, processed / max(dt, 1e-9),
              st.pairs, st.sum_fplus / max(st.positives, 1))
     return gs.to_host()
+
+
+def _train_sharded(cfg: Word2VecConfig, vocab: Vocabulary, reader: Callable,
+                   seed: int, engine: str, rank: int, world: int
+                   ) -> Tuple[np.ndarray, np.ndarray]:
+    """Multi-GPU engines; also usable on CPU (gloo) for tests.  Dim-sharded:
+    every rank walks the same data (compute split by dimension).  Row-
+    sharded: corpus partitioned by rank (data parallel), rows sharded."""
+    import torch
+    device = ("cuda:" + str(torch.cuda.current_device())
+              if torch.cuda.is_available() else "cpu")
+    subsample = 0.0 if cfg.legacy_subsample else cfg.subsample_ratio
+    common = dict(dtype=cfg.dtype if device != "cpu" else "float32",
+                  device=device, seed=seed, counts=vocab.counts,
+                  table_size=cfg.unigram_table_size, subsample=subsample,
+                  window_mode=cfg.window_mode)
+    max_sent = min(cfg.max_sentence_length, 1024)
+    total_words = vocab.train_words_count * cfg.num_iterations
+    processed = 0
+    sent_base = 0
+    t0 = time.time()
+    if engine == "dim":
+        from .dim_sharded import DimShardedSgns
+        eng = DimShardedSgns(vocab.num_words, cfg.vector_size,
+                             chunk_words=cfg.chunk_words,
+                             f_correction=cfg.f_correction,
+                             atomic=cfg.atomic_updates, **common)
+        for it in range(cfg.num_iterations):
+            for batch in batch_sentences(
+                    encode_sentences(reader(), vocab, max_sent),
+                    cfg.words_per_step):
+                alpha = cfg.learning_rate * max(
+                    1e-4, 1.0 - processed / (total_words + 1))
+                tok = torch.from_numpy(batch.tokens).to(eng.device)
+                off = torch.from_numpy(batch.offsets).to(eng.device)
+                eng.train_step(tok, off, alpha, cfg.window, cfg.n, seed,
+                               sent_id_base=sent_base,
+                               offsets_host=batch.offsets)
+                sent_base += batch.num_sentences
+                processed += batch.num_tokens
+    else:  # row
+        from ..data import partition_round_robin
+        from .row_sharded import RowShardedSgns
+        eng = RowShardedSgns(vocab.num_words, cfg.vector_size, **common)
+        rng = np.random.default_rng(seed + 17 * rank)
+        empty = (np.zeros(0, dtype=np.int32), np.zeros(1, dtype=np.int32))
+        for it in range(cfg.num_iterations):
+            my_sents = partition_round_robin(
+                encode_sentences(reader(), vocab, max_sent), rank, world)
+            batches = list(batch_sentences(my_sents, cfg.words_per_step))
+            # every rank must make the same number of collective calls
+            n_steps = len(batches)
+            if world > 1:
+                t = torch.tensor([n_steps])
+                torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+                n_steps = int(t.item())
+            for k in range(n_steps):
+                tokens, offsets = ((batches[k].tokens, batches[k].offsets)
+                                   if k < len(batches) else empty)
+                alpha = cfg.learning_rate * max(
+                    1e-4, 1.0 - processed / (total_words // world + 1))
+                eng.train_step(tokens, offsets, alpha, cfg.window, cfg.n, rng)
+                processed += len(tokens)
+    st = eng.read_stats()
+    dt = time.time() - t0
+    log.info("%s-sharded training (rank %d/%d): %d words in %.2fs, %d pairs, "
+             "mean_fplus=%.4f", engine, rank, world, processed, dt, st.pairs,
+             st.sum_fplus / max(st.positives, 1))
+    return eng.to_host()
