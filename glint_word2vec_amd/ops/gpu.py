@@ -71,9 +71,13 @@ class GpuSgns:
         if syn0_host is not None:
             self.load_host(syn0_host, syn1_host)
         elif init == "word2vec":
+            # blockwise so 80M-vocab tables never materialise on host
             g = torch.Generator(device="cpu").manual_seed(seed)
-            w = (torch.rand((vocab_size, dim), generator=g) - 0.5) / dim
-            self.syn0[:, :dim] = w.to(tdtype).to(self.device)
+            block = max(1, (1 << 28) // max(dim, 1))
+            for r0 in range(0, vocab_size, block):
+                r1 = min(vocab_size, r0 + block)
+                w = (torch.rand((r1 - r0, dim), generator=g) - 0.5) / dim
+                self.syn0[r0:r1, :dim] = w.to(tdtype).to(self.device)
         # stats buffer: [pairs u64, positives u64, words u64, sum_fplus f64]
         self._stats = torch.zeros(4, dtype=torch.int64, device=self.device)
         self.keep_thr: Optional[torch.Tensor] = None
