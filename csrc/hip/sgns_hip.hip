@@ -382,6 +382,9 @@ struct TrainPhase {
       }
       do_pair(e0, b0, p0);
       if (++k >= count) break;
+      // adjacent pairs on the same row: the prefetch predates this pair's
+      // update — refresh so sequential semantics (and oracle parity) hold
+      if (((e1 ^ e0) & 0x7FFFFFFFu) == 0) RowIO<T, NC>::load(p1, b1, lane);
       if (k + 1 < count) {
         e0 = tl[k + 1];
         p0 = rowptr(e0);
@@ -389,6 +392,7 @@ struct TrainPhase {
       }
       do_pair(e1, b1, p1);
       if (++k >= count) break;
+      if (((e0 ^ e1) & 0x7FFFFFFFu) == 0) RowIO<T, NC>::load(p0, b0, lane);
     }
   }
 
@@ -624,6 +628,7 @@ struct UpdateSlicePhase {
       }
       do_pair(e0, b0, p0, idx_base + k);
       if (++k >= count) break;
+      if (((e1 ^ e0) & 0x7FFFFFFFu) == 0) RowIO<T, NC>::load(p1, b1, lane);
       if (k + 1 < count) {
         e0 = tl[k + 1];
         p0 = rowptr(e0);
@@ -631,6 +636,7 @@ struct UpdateSlicePhase {
       }
       do_pair(e1, b1, p1, idx_base + k);
       if (++k >= count) break;
+      if (((e0 ^ e1) & 0x7FFFFFFFu) == 0) RowIO<T, NC>::load(p0, b0, lane);
     }
   }
   __device__ __forceinline__ void end_position(int32_t) {
