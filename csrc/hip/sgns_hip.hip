@@ -128,13 +128,17 @@ struct RowIO<uint16_t, NC> {
     if (NC & 1) v[NC - 1] = bf16_to_f32(row[64 * (NC - 1) + lane]);
   }
   static __device__ __forceinline__ void store(uint16_t* row, const float v[NC], int lane) {
+    // plain casts: hipcc fuses pairs into v_cvt_pk_bf16_f32 (RNE), one
+    // instruction per two elements vs ~4 VALU each for bit-math RNE
 #pragma unroll
     for (int m = 0; m < NC / 2; ++m) {
-      uint32_t p = (uint32_t)f32_to_bf16_rne(v[2 * m]) |
-                   ((uint32_t)f32_to_bf16_rne(v[2 * m + 1]) << 16);
-      *reinterpret_cast<uint32_t*>(row + 128 * m + 2 * lane) = p;
+      v2bf16 d;
+      d[0] = (__bf16)v[2 * m];
+      d[1] = (__bf16)v[2 * m + 1];
+      *reinterpret_cast<v2bf16*>(row + 128 * m + 2 * lane) = d;
     }
-    if (NC & 1) row[64 * (NC - 1) + lane] = f32_to_bf16_rne(v[NC - 1]);
+    if (NC & 1)
+      *reinterpret_cast<__bf16*>(row + 64 * (NC - 1) + lane) = (__bf16)v[NC - 1];
   }
   // gfx950 packed-bf16 atomic add (global_atomic_pk_add_bf16): adds the f32
   // delta rounded to bf16 — no lost updates on contended (hot Zipf) rows.
@@ -171,6 +175,25 @@ __device__ __forceinline__ float sigmoid_clipped(float f) {
   // precise expf (not __expf): one call per pair per wave — cost is
   // negligible and it keeps parity with the CPU oracle tight
   return 1.0f / (1.0f + expf(-f));
+}
+
+// Fast full-wave (64-lane) float sum: 4x row_shr + row_bcast15/31 DPP adds
+// (bound_ctrl=0-fill), total in lane 63, broadcast via readlane.  Replaces
+// the 6-step __shfl_xor chain (ds_bpermute + mask setup, ~7x more VALU).
+__device__ __forceinline__ float wave_sum_f32(float v) {
+  typedef int i32;
+  i32 x = __float_as_int(v);
+#define DPP_ADD(ctrl)                                                        \
+  x = __float_as_int(__int_as_float(x) +                                     \
+      __int_as_float(__builtin_amdgcn_update_dpp(0, x, ctrl, 0xF, 0xF, true)))
+  DPP_ADD(0x111);   // row_shr:1
+  DPP_ADD(0x112);   // row_shr:2
+  DPP_ADD(0x114);   // row_shr:4
+  DPP_ADD(0x118);   // row_shr:8
+  DPP_ADD(0x142);   // row_bcast:15
+  DPP_ADD(0x143);   // row_bcast:31
+#undef DPP_ADD
+  return __int_as_float(__builtin_amdgcn_readlane(x, 63));
 }
 
 struct KernelArgs {
@@ -338,8 +361,7 @@ struct TrainPhase {
     float f = 0.0f;
 #pragma unroll
     for (int k = 0; k < NC; ++k) f += c_row[k] * t_row[k];
-#pragma unroll
-    for (int sh = 32; sh > 0; sh >>= 1) f += __shfl_xor(f, sh, 64);
+    f = wave_sum_f32(f);
     const float g = (label - sigmoid_clipped(f)) * alpha;
     const bool use_atomic =
         ATOMIC && ((int32_t)(enc & 0x7FFFFFFFu) < atomic_below);
@@ -496,8 +518,7 @@ struct DotPhase {
     float f = 0.0f;
 #pragma unroll
     for (int k = 0; k < NC; ++k) f += c_row[k] * t_row[k];
-#pragma unroll
-    for (int sh = 32; sh > 0; sh >>= 1) f += __shfl_xor(f, sh, 64);
+    f = wave_sum_f32(f);
     if (lane == 0) f_base[idx] = f;
   }
   __device__ __forceinline__ void process_pairs(const uint32_t* tl, int count,
@@ -583,8 +604,7 @@ struct UpdateSlicePhase {
       float fresh = 0.0f;
 #pragma unroll
       for (int k = 0; k < NC; ++k) fresh += c_row[k] * t_row[k];
-#pragma unroll
-      for (int sh = 32; sh > 0; sh >>= 1) fresh += __shfl_xor(fresh, sh, 64);
+      fresh = wave_sum_f32(fresh);
       f += world_scale * (fresh - f_loc[idx]);
     }
     const float g = (label - sigmoid_clipped(f)) * alpha;
@@ -800,8 +820,7 @@ __global__ __launch_bounds__(256) void norms_kernel(
     float ss = 0.0f;
 #pragma unroll
     for (int k = 0; k < NC; ++k) ss += v[k] * v[k];
-#pragma unroll
-    for (int sh = 32; sh > 0; sh >>= 1) ss += __shfl_xor(ss, sh, 64);
+    ss = wave_sum_f32(ss);
     if (lane == 0) out[r] = sqrtf(ss);
   }
 }
