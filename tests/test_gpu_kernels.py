@@ -220,25 +220,27 @@ def test_row_sharded_gpu_world1_trains():
                              seed=5, zipf_a=1.01)
     counts = np.bincount(batch.tokens, minlength=500).astype(np.int64) + 1
 
-    def run(device):
+    def run(device, serial=False):
         eng = RowShardedSgns(500, 48, device=device, seed=3, counts=counts,
                              table_size=1009)
+        eng.serial = serial
         rng = np.random.default_rng(17)
         eng.train_step(batch.tokens, batch.offsets, 0.03, 3, 4, rng)
         st = eng.read_stats()
         s0, s1 = eng.to_host()
         return s0, s1, st
 
-    g0, g1, gst = run("cuda")
+    # serial GPU launch processes groups in CPU order -> tight comparison
+    g0, g1, gst = run("cuda", serial=True)
     c0, c1, cst = run("cpu")
     assert gst.pairs == cst.pairs
     assert gst.positives == cst.positives
-    assert np.isfinite(g0).all()
-    # GPU pairs kernel races across groups (atomic adds, different order);
-    # aggregate movement must match the sequential CPU engine closely
-    # GPU pairs kernel hogwilds across groups; compare aggregates loosely
-    assert np.linalg.norm(g0) == pytest.approx(np.linalg.norm(c0), rel=0.25)
-    assert np.linalg.norm(g1) == pytest.approx(np.linalg.norm(c1), rel=0.5)
+    np.testing.assert_allclose(g0, c0, rtol=5e-3, atol=5e-5)
+    np.testing.assert_allclose(g1, c1, rtol=5e-3, atol=5e-5)
+    # parallel launch: hogwild across groups; counts exact, values finite
+    p0, p1, pst = run("cuda", serial=False)
+    assert pst.pairs == cst.pairs
+    assert np.isfinite(p0).all() and np.isfinite(p1).all()
 
 
 def test_estimator_end_to_end_gpu():
