@@ -193,6 +193,7 @@ class GlintWord2VecModel:
         self.input_col = input_col
         self.output_col = output_col
         self._norms: Optional[np.ndarray] = None
+        self._gpu = None          # GpuSgns when to_gpu() was called
 
     # --- helpers ----------------------------------------------------------
     @property
@@ -209,8 +210,42 @@ class GlintWord2VecModel:
     def norms(self) -> np.ndarray:
         """Lazily cached row norms (mllib:486)."""
         if self._norms is None:
-            self._norms = np.linalg.norm(self._f32(), axis=1)
+            if self._gpu is not None:
+                self._norms = self._gpu.norms().cpu().numpy()
+            else:
+                self._norms = np.linalg.norm(self._f32(), axis=1)
         return self._norms
+
+    # --- GPU-resident serving (the PS-serving analog; inverse of toLocal) --
+    def to_gpu(self, device: str = "cuda", dtype: str = "float32"
+               ) -> "GlintWord2VecModel":
+        """Upload syn0 to a GPU; findSynonyms (rocBLAS GEMV + norms kernel)
+        and sentence-average transform (pull_average kernel) then run
+        device-side — the reference's server-side model ops."""
+        from .ops.gpu import GpuSgns
+        self._gpu = GpuSgns(self.num_words, self.vector_size, dtype=dtype,
+                            device=device, syn0_host=self._f32())
+        self._norms = None
+        return self
+
+    def transform_sentences_gpu(self, sentences) -> np.ndarray:
+        """Batched sentence-average transform on the GPU (ml:443-456)."""
+        assert self._gpu is not None, "call to_gpu() first"
+        import torch
+        idx_lists = [[self.vocab[w] for w in s if w in self.vocab]
+                     for s in sentences]
+        offsets = np.zeros(len(idx_lists) + 1, dtype=np.int32)
+        np.cumsum([len(x) for x in idx_lists], out=offsets[1:])
+        tokens = np.concatenate([np.asarray(x, dtype=np.int32)
+                                 for x in idx_lists if x] or
+                                [np.zeros(0, dtype=np.int32)])
+        dev = self._gpu.device
+        out = self._gpu.pull_average(
+            torch.from_numpy(tokens).to(dev),
+            torch.from_numpy(offsets).to(dev))
+        import torch as _t
+        _t.cuda.synchronize(dev)
+        return out.cpu().numpy()
 
     # --- transform --------------------------------------------------------
     def transform(self, x):
@@ -249,7 +284,12 @@ class GlintWord2VecModel:
         qn = np.linalg.norm(vec)
         if qn > 0:
             vec = vec / qn
-        scores = self._f32() @ vec              # `multiply` (mllib:598)
+        if self._gpu is not None:
+            import torch
+            scores = self._gpu.multiply(
+                torch.from_numpy(vec).to(self._gpu.device)).cpu().numpy()
+        else:
+            scores = self._f32() @ vec          # `multiply` (mllib:598)
         norms = self.norms()
         with np.errstate(divide="ignore", invalid="ignore"):
             cos = np.where(norms > 0, scores / norms, 0.0)

@@ -1,0 +1,48 @@
+"""CLI end-to-end: train -> similar -> analogy -> export -> info."""
+import subprocess
+import sys
+
+import pytest
+
+
+def _run(args, **kw):
+    return subprocess.run([sys.executable, "-m", "glint_word2vec_amd"] + args,
+                          capture_output=True, text=True, timeout=300, **kw)
+
+
+def test_cli_roundtrip(tmp_path):
+    corpus = tmp_path / "corpus.txt"
+    lines = []
+    import numpy as np
+    rng = np.random.default_rng(3)
+    for _ in range(300):
+        head = "aa" if rng.random() < 0.5 else "a2"
+        lines.append(" ".join([head, "bb"] +
+                              [f"x{rng.integers(0, 15)}" for _ in range(3)] +
+                              [head, "bb"]))
+    corpus.write_text("\n".join(lines))
+    model_dir = str(tmp_path / "model")
+    r = _run(["train", str(corpus), model_dir, "--vector-size", "16",
+              "--min-count", "1", "--iterations", "4", "--window", "2",
+              "--learning-rate", "0.05", "--unigram-table-size", "50000",
+              "--subsample", "0", "--seed", "7", "--device", "cpu",
+              "--workers", "2", "--shards", "2"])
+    assert r.returncode == 0, r.stderr
+    assert "model saved" in r.stdout
+
+    r = _run(["similar", model_dir, "aa", "-n", "3"])
+    assert r.returncode == 0, r.stderr
+    assert "a2" in r.stdout
+
+    r = _run(["analogy", model_dir, "aa", "bb", "-m", "a2", "-n", "3"])
+    assert r.returncode == 0, r.stderr
+
+    out = tmp_path / "vecs.txt"
+    r = _run(["export", model_dir, str(out)])
+    assert r.returncode == 0, r.stderr
+    head = out.read_text().splitlines()[0].split()
+    assert int(head[1]) == 16
+
+    r = _run(["info", model_dir])
+    assert r.returncode == 0
+    assert "vectorSize" in r.stdout

@@ -243,6 +243,28 @@ def test_row_sharded_gpu_world1_trains():
     assert np.isfinite(p0).all() and np.isfinite(p1).all()
 
 
+def test_model_gpu_serving_ops():
+    """to_gpu(): device-side findSynonyms + sentence-average transform must
+    agree with the host implementations."""
+    from glint_word2vec_amd import GlintWord2VecModel
+    from glint_word2vec_amd.config import Word2VecConfig
+    from glint_word2vec_amd.vocab import build_vocab
+    rng = np.random.default_rng(0)
+    sents = [[f"w{i}", f"w{(i + 1) % 40}"] for i in range(40)] * 3
+    vocab = build_vocab(sents, min_count=1)
+    syn0 = rng.standard_normal((vocab.num_words, 48)).astype(np.float32)
+    m = GlintWord2VecModel(Word2VecConfig(vector_size=48), vocab, syn0)
+    host_syns = m.find_synonyms("w0", 5)
+    host_avg = np.stack([m.transform_sentence(s) for s in sents[:7]])
+    m.to_gpu()
+    gpu_syns = m.find_synonyms("w0", 5)
+    assert [w for w, _ in gpu_syns] == [w for w, _ in host_syns]
+    for (w1, c1), (w2, c2) in zip(host_syns, gpu_syns):
+        assert c2 == pytest.approx(c1, rel=1e-4)
+    gpu_avg = m.transform_sentences_gpu(sents[:7])
+    np.testing.assert_allclose(gpu_avg, host_avg, rtol=1e-4, atol=1e-6)
+
+
 def test_estimator_end_to_end_gpu():
     rng = np.random.default_rng(5)
     sents = []
