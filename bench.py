@@ -46,6 +46,11 @@ def parse_args():
                    help="atomics only for rows < K (hot rows)")
     p.add_argument("--profile-steps", type=int, default=0,
                    help="run only this many steps, no warmup JSON (rocprof)")
+    p.add_argument("--device", choices=["cuda", "cpu"], default="cuda",
+                   help="cpu = gloo plumbing test of the distributed path")
+    p.add_argument("--engine", choices=["auto", "fused", "dim", "row"],
+                   default="auto",
+                   help="auto: fused kernel at world 1, dim-sharded beyond")
     return p.parse_args()
 
 
@@ -57,17 +62,25 @@ def main():
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
     distributed = world > 1
+    use_cpu = args.device == "cpu"
     if distributed:
         import torch.distributed as dist
-        torch.cuda.set_device(local_rank)
-        dist.init_process_group(backend="nccl")
-    device = torch.device("cuda", local_rank)
-    torch.cuda.set_device(device)
+        if use_cpu:
+            dist.init_process_group(backend="gloo")
+        else:
+            torch.cuda.set_device(local_rank)
+            dist.init_process_group(backend="nccl")
+    if use_cpu:
+        device = torch.device("cpu")
+    else:
+        device = torch.device("cuda", local_rank)
+        torch.cuda.set_device(device)
 
     from glint_word2vec_amd.data import synthetic_corpus
     from glint_word2vec_amd.vocab import build_unigram_table
 
-    dtype = "bfloat16" if args.dtype == "bf16" else "float32"
+    dtype = ("float32" if use_cpu else
+             ("bfloat16" if args.dtype == "bf16" else "float32"))
     n_steps = args.profile_steps if args.profile_steps else args.steps
     total_launches = n_steps + (0 if args.profile_steps else args.warmup)
 
@@ -81,13 +94,24 @@ def main():
                              sentence_len=args.sentence_len, seed=1234)
     counts = np.bincount(batch.tokens, minlength=args.vocab).astype(np.int64) + 1
 
-    if distributed:
+    engine = args.engine
+    if engine == "auto":
+        engine = "dim" if (distributed or use_cpu) else "fused"
+    if engine == "dim":
         from glint_word2vec_amd.parallel.dim_sharded import DimShardedSgns
         trainer = DimShardedSgns(args.vocab, args.dim, dtype=dtype,
                                  device=str(device), seed=1, counts=counts,
                                  table_size=args.table_size,
                                  subsample=args.subsample,
                                  atomic=bool(args.atomic or args.atomic_below))
+        dist_mode = True
+    elif engine == "row":
+        from glint_word2vec_amd.parallel.row_sharded import RowShardedSgns
+        trainer = RowShardedSgns(args.vocab, args.dim, dtype=dtype,
+                                 device=str(device), seed=1, counts=counts,
+                                 table_size=args.table_size,
+                                 subsample=args.subsample)
+        dist_mode = "row"
     else:
         from glint_word2vec_amd.ops.gpu import GpuSgns
         gs = GpuSgns(args.vocab, args.dim, dtype=dtype, device=str(device), seed=1)
@@ -96,14 +120,20 @@ def main():
         if args.subsample > 0:
             gs.set_subsample(counts, int(counts.sum()), args.subsample)
         trainer = gs
+        dist_mode = False
 
     tok = torch.from_numpy(batch.tokens).to(device)
     off = torch.from_numpy(batch.offsets).to(device)
     alpha = 0.01875
     nsent = batch.num_sentences
 
+    row_rng = np.random.default_rng(99 + rank)
+
     def step(i):
-        if distributed:
+        if dist_mode == "row":
+            trainer.train_step(batch.tokens, batch.offsets, alpha,
+                               args.window, args.neg, row_rng)
+        elif dist_mode:
             trainer.train_step(tok, off, alpha, args.window, args.neg,
                                seed=99, sent_id_base=i * nsent,
                                offsets_host=batch.offsets)
@@ -115,16 +145,19 @@ def main():
                                 blocks=args.blocks or None)
 
     def barrier_sync():
-        torch.cuda.synchronize(device)
+        if not use_cpu:
+            torch.cuda.synchronize(device)
         if distributed:
             import torch.distributed as dist
             dist.barrier()
-        torch.cuda.synchronize(device)
+        if not use_cpu:
+            torch.cuda.synchronize(device)
 
     if args.profile_steps:
         for i in range(n_steps):
             step(i)
-        torch.cuda.synchronize(device)
+        if not use_cpu:
+            torch.cuda.synchronize(device)
         if rank == 0:
             st = trainer.read_stats()
             print(f"profiled {n_steps} steps: {st}")
@@ -168,8 +201,8 @@ def main():
                          f"neg={args.neg} window={args.window}",
                 "global_batch": args.words_per_step * world,
                 "seq_len": args.sentence_len,
-                "parallelism": (f"dimshard-rccl-allreduce-x{world}"
-                                if distributed else "hogwild-1gpu"),
+                "parallelism": (f"{engine}shard-rccl-x{world}" if dist_mode
+                                else "hogwild-1gpu"),
                 "updates": ("atomic" if (args.atomic or args.atomic_below)
                             else "hogwild"),
             },

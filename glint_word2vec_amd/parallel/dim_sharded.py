@@ -128,56 +128,98 @@ class DimShardedSgns:
             if acc >= self.chunk_words or s == num_sent - 1:
                 chunks.append((s0, s + 1))
                 s0, acc = s + 1, 0
-        for (a, b) in chunks:
-            self._train_chunk(tokens, offsets, a, b, alpha, window, n_neg,
-                              seed, sent_id_base)
+        if self.is_cuda:
+            self._train_step_gpu(tokens, offsets, chunks, alpha, window,
+                                 n_neg, seed, sent_id_base)
+        else:
+            for (a, b) in chunks:
+                self._train_chunk_cpu(tokens, offsets, a, b, alpha, window,
+                                      n_neg, seed, sent_id_base)
 
-    def _train_chunk(self, tokens, offsets, s0, s1, alpha, window, n_neg,
-                     seed, sent_id_base):
+    def _train_step_gpu(self, tokens, offsets, chunks, alpha, window, n_neg,
+                        seed, sent_id_base):
+        """Pipelined phases: one pair-count pass + host sync per STEP, then
+        per chunk {dots -> allreduce(f slice) on a comm stream -> update},
+        with chunk k's allreduce overlapping chunk k+1's dots.  The one-
+        chunk lookahead (dots k+1 run before update k lands) is absorbed by
+        the f-correction."""
+        device = self.device
+        comp = torch.cuda.current_stream(device)
+        if not hasattr(self, "_comm_stream"):
+            self._comm_stream = torch.cuda.Stream(device)
+        cs = self._comm_stream
+        num_sent = len(offsets) - 1
+        seed &= 0xFFFFFFFFFFFFFFFF
+        ref = int(self.window_mode == "reference")
+        kthr = 0 if self.keep_thr is None else self.keep_thr.data_ptr()
+        nb_all = 1 if self.serial else max(1, min((num_sent + 3) // 4, 8192))
+        nt = 64 if self.serial else 256
+        counts = torch.empty(num_sent, dtype=torch.int64, device=device)
+        self.native.count_pairs(
+            tokens.data_ptr(), offsets.data_ptr(), num_sent, kthr,
+            self.table.data_ptr(), int(self.table.numel()), window, n_neg,
+            seed, sent_id_base, ref, counts.data_ptr(), nb_all, nt,
+            comp.cuda_stream)
+        poff = torch.zeros(num_sent + 1, dtype=torch.int64, device=device)
+        torch.cumsum(counts, 0, out=poff[1:])
+        poff_host = poff.cpu().numpy()          # one sync per step
+        total = int(poff_host[-1])
+        if total == 0:
+            return
+        f = torch.zeros(total, dtype=torch.float32, device=device)
+        f_loc = torch.empty_like(f) if self.f_correction else None
+        prev = None   # (chunk, ar_event)
+
+        def launch_update(c):
+            n = c[1] - c[0]
+            nb = 1 if self.serial else max(1, min((n + 3) // 4, 8192))
+            self.native.update_slice(
+                self.syn0.data_ptr(), self.syn1.data_ptr(), int(self.is_bf16),
+                self.stride, tokens.data_ptr(),
+                offsets[c[0]:c[1] + 1].data_ptr(), n, kthr,
+                self.table.data_ptr(), int(self.table.numel()), float(alpha),
+                window, n_neg, seed, sent_id_base + c[0], ref,
+                poff[c[0]:c[1] + 1].data_ptr(), f.data_ptr(),
+                0 if f_loc is None else f_loc.data_ptr(), float(self.world),
+                (2 ** 31 - 1) if self.atomic else 0, self._stats.data_ptr(),
+                nb, nt, comp.cuda_stream)
+
+        for (a, b) in chunks:
+            n = b - a
+            lo, hi = int(poff_host[a]), int(poff_host[b])
+            if hi == lo:
+                continue
+            nb = 1 if self.serial else max(1, min((n + 3) // 4, 8192))
+            self.native.dots_slice(
+                self.syn0.data_ptr(), self.syn1.data_ptr(), int(self.is_bf16),
+                self.stride, tokens.data_ptr(), offsets[a:b + 1].data_ptr(),
+                n, kthr, self.table.data_ptr(), int(self.table.numel()),
+                window, n_neg, seed, sent_id_base + a, ref,
+                poff[a:b + 1].data_ptr(), f.data_ptr(), nb, nt,
+                comp.cuda_stream)
+            if f_loc is not None:
+                f_loc[lo:hi] = f[lo:hi]
+            ev = torch.cuda.Event()
+            ev.record(comp)
+            with torch.cuda.stream(cs):
+                cs.wait_event(ev)
+                comm.all_reduce_sum(f[lo:hi])
+                ar_ev = torch.cuda.Event()
+                ar_ev.record(cs)
+            if prev is not None:
+                comp.wait_event(prev[1])
+                launch_update(prev[0])
+            prev = ((a, b), ar_ev)
+        if prev is not None:
+            comp.wait_event(prev[1])
+            launch_update(prev[0])
+
+    def _train_chunk_cpu(self, tokens, offsets, s0, s1, alpha, window, n_neg,
+                         seed, sent_id_base):
         n = s1 - s0
         off_view = offsets[s0:s1 + 1]
         base = sent_id_base + s0
-        if self.is_cuda:
-            stream = torch.cuda.current_stream(self.device)
-            nb = 1 if self.serial else max(1, min((n + 3) // 4, 8192))
-            nt = 64 if self.serial else 256
-            counts = torch.empty(n, dtype=torch.int64, device=self.device)
-            self.native.count_pairs(
-                tokens.data_ptr(), off_view.data_ptr(), n,
-                0 if self.keep_thr is None else self.keep_thr.data_ptr(),
-                self.table.data_ptr(), int(self.table.numel()), window, n_neg,
-                seed & 0xFFFFFFFFFFFFFFFF, base,
-                int(self.window_mode == "reference"), counts.data_ptr(),
-                nb, nt, stream.cuda_stream)
-            poff = torch.zeros(n + 1, dtype=torch.int64, device=self.device)
-            torch.cumsum(counts, 0, out=poff[1:])
-            total = int(poff[-1].item())
-            if total == 0:
-                return
-            f = torch.zeros(total, dtype=torch.float32, device=self.device)
-            self.native.dots_slice(
-                self.syn0.data_ptr(), self.syn1.data_ptr(), int(self.is_bf16),
-                self.stride, tokens.data_ptr(), off_view.data_ptr(), n,
-                0 if self.keep_thr is None else self.keep_thr.data_ptr(),
-                self.table.data_ptr(), int(self.table.numel()), window, n_neg,
-                seed & 0xFFFFFFFFFFFFFFFF, base,
-                int(self.window_mode == "reference"), poff.data_ptr(),
-                f.data_ptr(), nb, nt, stream.cuda_stream)
-            f_loc = f.clone() if self.f_correction else None
-            comm.all_reduce_sum(f)
-            self.native.update_slice(
-                self.syn0.data_ptr(), self.syn1.data_ptr(), int(self.is_bf16),
-                self.stride, tokens.data_ptr(), off_view.data_ptr(), n,
-                0 if self.keep_thr is None else self.keep_thr.data_ptr(),
-                self.table.data_ptr(), int(self.table.numel()), float(alpha),
-                window, n_neg, seed & 0xFFFFFFFFFFFFFFFF, base,
-                int(self.window_mode == "reference"), poff.data_ptr(),
-                f.data_ptr(),
-                0 if f_loc is None else f_loc.data_ptr(),
-                float(self.world),
-                (2 ** 31 - 1) if self.atomic else 0, self._stats.data_ptr(),
-                nb, nt, stream.cuda_stream)
-        else:
+        if True:
             tok_np = tokens.numpy()
             off_np = off_view.numpy()
             tab_np = self.table.numpy()
