@@ -51,6 +51,8 @@ def parse_args():
     p.add_argument("--engine", choices=["auto", "fused", "dim", "row"],
                    default="auto",
                    help="auto: fused kernel at world 1, dim-sharded beyond")
+    p.add_argument("--chunk-words", type=int, default=1 << 19,
+                   help="dim-sharded feedback chunk size")
     return p.parse_args()
 
 
@@ -103,6 +105,7 @@ def main():
                                  device=str(device), seed=1, counts=counts,
                                  table_size=args.table_size,
                                  subsample=args.subsample,
+                                 chunk_words=args.chunk_words,
                                  atomic=bool(args.atomic or args.atomic_below))
         dist_mode = True
     elif engine == "row":

@@ -42,7 +42,7 @@ class DimShardedSgns:
                  device: str = "cuda", seed: int = 1,
                  counts: Optional[np.ndarray] = None,
                  table_size: int = 1_000_000, subsample: float = 0.0,
-                 window_mode: str = "canonical", chunk_words: int = 1 << 16,
+                 window_mode: str = "canonical", chunk_words: int = 1 << 19,
                  f_correction: bool = True, atomic: bool = True,
                  init_full_limit: int = 1 << 28):
         self.rank, self.world = (comm.init_from_env() if torch.distributed.is_available()
