@@ -48,9 +48,12 @@ def parse_args():
                    help="run only this many steps, no warmup JSON (rocprof)")
     p.add_argument("--device", choices=["cuda", "cpu"], default="cuda",
                    help="cpu = gloo plumbing test of the distributed path")
-    p.add_argument("--engine", choices=["auto", "fused", "dim", "row"],
+    p.add_argument("--engine", choices=["auto", "fused", "dim", "row", "dp"],
                    default="auto",
-                   help="auto: fused kernel at world 1, dim-sharded beyond")
+                   help="auto: fused at world 1; beyond: dp (replicated + "
+                        "delta-allreduce) when tables fit, else dim-sharded")
+    p.add_argument("--sync-every", type=int, default=4,
+                   help="dp engine: steps between delta merges")
     p.add_argument("--chunk-words", type=int, default=1 << 19,
                    help="dim-sharded feedback chunk size")
     return p.parse_args()
@@ -86,19 +89,39 @@ def main():
     n_steps = args.profile_steps if args.profile_steps else args.steps
     total_launches = n_steps + (0 if args.profile_steps else args.warmup)
 
-    # --- synthetic data: one step's batch, reused with varying sent_id_base
-    # (RNG/negative draws differ per step; token stream is Zipf-realistic).
-    # Dim-sharded (distributed): every rank walks the SAME global batch of
-    # world*words_per_step tokens, each computing its dim-slice -> per-GPU
-    # work is fixed as N grows (weak scaling).
-    batch = synthetic_corpus(args.vocab,
-                             args.words_per_step * (world if distributed else 1),
-                             sentence_len=args.sentence_len, seed=1234)
-    counts = np.bincount(batch.tokens, minlength=args.vocab).astype(np.int64) + 1
-
     engine = args.engine
     if engine == "auto":
-        engine = "dim" if (distributed or use_cpu) else "fused"
+        if distributed or use_cpu:
+            esize = 2 if args.dtype == "bf16" else 4
+            table_bytes = 2 * args.vocab * (args.dim + 64) * esize
+            engine = "dp" if table_bytes <= (8 << 30) else "dim"
+        else:
+            engine = "fused"
+
+    # --- synthetic data (Zipf-realistic token stream), reused each step
+    # with varying sent_id_base so RNG/negative draws differ per step.
+    # dim engine: every rank walks the SAME global batch of
+    # world*words_per_step tokens, each computing its dim-slice.
+    # dp/row engines: each rank has its OWN words_per_step batch (data
+    # parallel).  Either way per-GPU work is fixed as N grows (weak scaling).
+    if engine == "dim" and distributed:
+        batch = synthetic_corpus(args.vocab, args.words_per_step * world,
+                                 sentence_len=args.sentence_len, seed=1234)
+    else:
+        batch = synthetic_corpus(args.vocab, args.words_per_step,
+                                 sentence_len=args.sentence_len,
+                                 seed=1234 + (rank if engine in ("dp", "row")
+                                              else 0))
+    counts = np.bincount(batch.tokens, minlength=args.vocab).astype(np.int64) + 1
+    if distributed and engine in ("dp", "row"):
+        # table/subsample stats must be identical across ranks
+        import torch.distributed as dist
+        ct = torch.from_numpy(counts.copy())
+        if use_cpu:
+            dist.all_reduce(ct)
+        else:
+            ctd = ct.to(device); dist.all_reduce(ctd); ct = ctd.cpu()
+        counts = ct.numpy()
     if engine == "dim":
         from glint_word2vec_amd.parallel.dim_sharded import DimShardedSgns
         trainer = DimShardedSgns(args.vocab, args.dim, dtype=dtype,
@@ -108,6 +131,15 @@ def main():
                                  chunk_words=args.chunk_words,
                                  atomic=bool(args.atomic or args.atomic_below))
         dist_mode = True
+    elif engine == "dp":
+        from glint_word2vec_amd.parallel.replicated import ReplicatedSgns
+        trainer = ReplicatedSgns(args.vocab, args.dim, dtype=dtype,
+                                 device=str(device), seed=1, counts=counts,
+                                 table_size=args.table_size,
+                                 subsample=args.subsample,
+                                 sync_every=args.sync_every,
+                                 atomic=bool(args.atomic or args.atomic_below))
+        dist_mode = "dp"
     elif engine == "row":
         from glint_word2vec_amd.parallel.row_sharded import RowShardedSgns
         trainer = RowShardedSgns(args.vocab, args.dim, dtype=dtype,
@@ -133,7 +165,10 @@ def main():
     row_rng = np.random.default_rng(99 + rank)
 
     def step(i):
-        if dist_mode == "row":
+        if dist_mode == "dp":
+            trainer.train_step(tok, off, alpha, args.window, args.neg,
+                               seed=99 + rank, sent_id_base=i * nsent)
+        elif dist_mode == "row":
             trainer.train_step(batch.tokens, batch.offsets, alpha,
                                args.window, args.neg, row_rng)
         elif dist_mode:

@@ -60,9 +60,12 @@ class Word2VecConfig:
     # "auto": fused single-GPU kernel at world 1; dim-sharded at world > 1.
     # "dim": dimension-sharded (CIKM scheme, RCCL allreduce of partial dots).
     # "row": row-sharded parameter-server shape (RCCL alltoallv pull/push).
-    engine: str = "auto"             # "auto" | "fused" | "dim" | "row"
-    chunk_words: int = 1 << 16       # dim-sharded feedback chunk
+    # "dp": replicated tables + periodic delta-allreduce (fused kernel per
+    # GPU; for vocabularies whose tables fit comfortably in HBM).
+    engine: str = "auto"             # "auto" | "fused" | "dim" | "row" | "dp"
+    chunk_words: int = 1 << 19       # dim-sharded feedback chunk
     f_correction: bool = True        # dim-sharded local-drift freshening
+    sync_every: int = 4              # dp engine: steps between delta merges
     # --- semantics switches ---------------------
     # The reference's subsampling is a de-facto no-op (integer-division bug,
     # mllib:375-377).  We implement the intended math; set
@@ -97,7 +100,7 @@ class Word2VecConfig:
             raise ValueError(f"unsupported dtype {self.dtype!r}")
         if self.window_mode not in ("canonical", "reference"):
             raise ValueError(f"unsupported window_mode {self.window_mode!r}")
-        if self.engine not in ("auto", "fused", "dim", "row"):
+        if self.engine not in ("auto", "fused", "dim", "row", "dp"):
             raise ValueError(f"unsupported engine {self.engine!r}")
 
     # -- (de)serialisation used by the checkpoint metadata ------------------

@@ -103,7 +103,7 @@ class GlintWord2Vec:
         if device == "auto":
             import torch
             device = "cuda" if torch.cuda.is_available() else "cpu"
-        if device == "cuda" or cfg.engine in ("dim", "row"):
+        if device == "cuda" or cfg.engine in ("dim", "row", "dp"):
             # single- or multi-GPU engines; dim/row also run on CPU under
             # gloo (multi-process tests, torchrun without GPUs)
             from .parallel.engine import train_gpu

@@ -38,7 +38,7 @@ def train_gpu(cfg: Word2VecConfig, vocab: Vocabulary, reader: Callable,
     engine = cfg.engine
     if engine == "auto":
         engine = "fused" if world == 1 else "dim"
-    if engine in ("dim", "row") and world >= 1:
+    if engine in ("dim", "row", "dp") and world >= 1:
         return _train_sharded(cfg, vocab, reader, seed, engine, rank, world)
     device = torch.device("cuda", torch.cuda.current_device())
     gs = GpuSgns(vocab.num_words, cfg.vector_size, cfg.dtype,
@@ -121,6 +121,37 @@ def _train_sharded(cfg: Word2VecConfig, vocab: Vocabulary, reader: Callable,
                                offsets_host=batch.offsets)
                 sent_base += batch.num_sentences
                 processed += batch.num_tokens
+    elif engine == "dp":
+        from ..data import partition_round_robin
+        from .replicated import ReplicatedSgns
+        eng = ReplicatedSgns(vocab.num_words, cfg.vector_size,
+                             sync_every=cfg.sync_every,
+                             atomic=cfg.atomic_updates, **common)
+        empty = (np.zeros(0, dtype=np.int32), np.zeros(1, dtype=np.int32))
+        sent_base = 10_000_000 * rank   # disjoint RNG streams per rank
+        for it in range(cfg.num_iterations):
+            my_sents = partition_round_robin(
+                encode_sentences(reader(), vocab, max_sent), rank, world)
+            batches = list(batch_sentences(my_sents, cfg.words_per_step))
+            n_steps = len(batches)
+            if world > 1:
+                t = torch.tensor([n_steps])
+                torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+                n_steps = int(t.item())
+            for k in range(n_steps):
+                tokens, offsets = ((batches[k].tokens, batches[k].offsets)
+                                   if k < len(batches) else empty)
+                alpha = cfg.learning_rate * max(
+                    1e-4, 1.0 - processed / (total_words // world + 1))
+                if eng.is_cuda:
+                    tok = torch.from_numpy(tokens).to(eng.device)
+                    off = torch.from_numpy(offsets).to(eng.device)
+                else:
+                    tok, off = tokens, offsets
+                eng.train_step(tok, off, alpha, cfg.window, cfg.n, seed,
+                               sent_id_base=sent_base)
+                sent_base += max(len(offsets) - 1, 0)
+                processed += len(tokens)
     else:  # row
         from ..data import partition_round_robin
         from .row_sharded import RowShardedSgns

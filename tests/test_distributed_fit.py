@@ -42,7 +42,7 @@ def _worker(rank, world, rdv, out_dir, engine):
         dist.destroy_process_group()
 
 
-@pytest.mark.parametrize("engine", ["dim", "row"])
+@pytest.mark.parametrize("engine", ["dim", "row", "dp"])
 def test_distributed_fit_two_ranks(tmp_path, engine):
     rdv = str(tmp_path / f"rdv_{engine}")
     mp.spawn(_worker, args=(2, rdv, str(tmp_path), engine), nprocs=2,
@@ -52,6 +52,46 @@ def test_distributed_fit_two_ranks(tmp_path, engine):
     # both ranks hold the full assembled model
     np.testing.assert_allclose(s0, s1, rtol=1e-5, atol=1e-7)
     assert np.isfinite(s0).all()
+
+
+def _worker_de(rank, world, rdv, out_dir):
+    import torch.distributed as dist
+    dist.init_process_group("gloo", rank=rank, world_size=world,
+                            init_method=f"file://{rdv}")
+    try:
+        import os as _os
+        from glint_word2vec_amd import GlintWord2Vec
+        corpus = _os.path.join(_os.path.dirname(__file__), "fixtures",
+                               "de_wikipedia_articles_country_capitals.txt")
+        est = (GlintWord2Vec().setSeed(1).setStepSize(0.025)
+               .setUnigramTableSize(1_000_000).setSubsampleRatio(0.0)
+               .setNumIterations(5))
+        est.config.device = "cpu"
+        est.config.engine = "dp"
+        est.config.sync_every = 2
+        est.config.window_mode = "reference"
+        m = est.fit(corpus)
+        if rank == 0:
+            syns = m.find_synonyms("österreich", 10)
+            res = m.analogy(["wien", "deutschland"], ["österreich"], 10)
+            with open(os.path.join(out_dir, "gate.txt"), "w") as f:
+                f.write(repr(([w for w, _ in syns], dict(syns).get("wien", 0),
+                              [w for w, _ in res])))
+            ok = ("wien" in [w for w, _ in syns] and dict(syns)["wien"] > 0.85
+                  and "berlin" in [w for w, _ in res])
+            np.save(os.path.join(out_dir, "gate.npy"), np.array([int(ok)]))
+    finally:
+        dist.destroy_process_group()
+
+
+def test_dp_engine_world2_quality_gate(tmp_path):
+    """The dp (replicated + delta-allreduce) engine — the default multi-GPU
+    strategy for HBM-resident tables — must preserve embedding quality with
+    the corpus split across 2 workers (the reference gate, Spec:290-382)."""
+    rdv = str(tmp_path / "rdv_de")
+    mp.spawn(_worker_de, args=(2, rdv, str(tmp_path)), nprocs=2, join=True)
+    ok = np.load(tmp_path / "gate.npy")
+    assert ok[0] == 1, open(tmp_path / "gate.txt").read()
 
 
 def test_single_process_engine_dim_fit_quality():
