@@ -524,7 +524,105 @@ static py::dict train_batch(
   return d;
 }
 
+// ---------------------------------------------------------------------------
+// Native corpus pipeline (the reference's Spark-side vocabulary build and
+// sentence encoding, mllib:258-279 and 335-343, as host C++):
+// whitespace-tokenized text, one sentence per line.
+// ---------------------------------------------------------------------------
+#include <fstream>
+#include <string_view>
+#include <unordered_map>
+#include <algorithm>
+
+static py::tuple build_vocab_file(const std::string& path, int64_t min_count) {
+  std::ifstream in(path, std::ios::binary);
+  if (!in) throw std::runtime_error("cannot open corpus: " + path);
+  std::unordered_map<std::string, int64_t> counter;
+  counter.reserve(1 << 20);
+  {
+    py::gil_scoped_release nogil;
+    std::string line;
+    while (std::getline(in, line)) {
+      size_t i = 0, n = line.size();
+      while (i < n) {
+        while (i < n && std::isspace((unsigned char)line[i])) ++i;
+        size_t j = i;
+        while (j < n && !std::isspace((unsigned char)line[j])) ++j;
+        if (j > i) counter[line.substr(i, j - i)]++;
+        i = j;
+      }
+    }
+  }
+  std::vector<std::pair<std::string, int64_t>> items;
+  items.reserve(counter.size());
+  for (auto& kv : counter)
+    if (kv.second >= min_count) items.emplace_back(kv.first, kv.second);
+  // count desc, word asc (deterministic — matches vocab.build_vocab)
+  std::sort(items.begin(), items.end(), [](const auto& a, const auto& b) {
+    if (a.second != b.second) return a.second > b.second;
+    return a.first < b.first;
+  });
+  py::list words;
+  auto counts = py::array_t<int64_t>((py::ssize_t)items.size());
+  int64_t* c = counts.mutable_data();
+  int64_t total = 0;
+  for (size_t i = 0; i < items.size(); ++i) {
+    words.append(py::str(items[i].first));
+    c[i] = items[i].second;
+    total += items[i].second;
+  }
+  return py::make_tuple(words, counts, total);
+}
+
+static py::tuple encode_corpus(const std::string& path, py::list words,
+                               int max_sentence_length) {
+  std::unordered_map<std::string, int32_t> index;
+  index.reserve(words.size() * 2);
+  int32_t id = 0;
+  for (auto w : words) index.emplace(py::cast<std::string>(w), id++);
+  std::ifstream in(path, std::ios::binary);
+  if (!in) throw std::runtime_error("cannot open corpus: " + path);
+  std::vector<int32_t> tokens;
+  std::vector<int32_t> offsets{0};
+  {
+    py::gil_scoped_release nogil;
+    std::string line;
+    int in_sentence = 0;
+    while (std::getline(in, line)) {
+      size_t i = 0, n = line.size();
+      in_sentence = 0;
+      while (i < n) {
+        while (i < n && std::isspace((unsigned char)line[i])) ++i;
+        size_t j = i;
+        while (j < n && !std::isspace((unsigned char)line[j])) ++j;
+        if (j > i) {
+          auto it = index.find(line.substr(i, j - i));
+          if (it != index.end()) {
+            if (in_sentence == max_sentence_length) {   // chunk (mllib:341)
+              offsets.push_back((int32_t)tokens.size());
+              in_sentence = 0;
+            }
+            tokens.push_back(it->second);
+            ++in_sentence;
+          }
+        }
+        i = j;
+      }
+      if (in_sentence > 0) offsets.push_back((int32_t)tokens.size());
+    }
+  }
+  auto tok = py::array_t<int32_t>((py::ssize_t)tokens.size());
+  std::memcpy(tok.mutable_data(), tokens.data(), tokens.size() * 4);
+  auto off = py::array_t<int32_t>((py::ssize_t)offsets.size());
+  std::memcpy(off.mutable_data(), offsets.data(), offsets.size() * 4);
+  return py::make_tuple(tok, off);
+}
+
 PYBIND11_MODULE(_cpu_native, m) {
+  m.def("build_vocab_file", &build_vocab_file, py::arg("path"),
+        py::arg("min_count") = 5);
+  m.def("encode_corpus", &encode_corpus, py::arg("path"), py::arg("words"),
+        py::arg("max_sentence_length") = 1000);
   m.doc() = "CPU-native fused SGNS trainer (exact twin of the HIP kernel)";
   m.def("train_batch", &train_batch,
         py::arg("syn0"), py::arg("syn1"), py::arg("tokens"),

@@ -62,6 +62,24 @@ def batch_sentences(encoded: Iterable[np.ndarray], words_per_step: int) -> Itera
                             offsets=np.asarray(offsets, dtype=np.int32))
 
 
+def batches_from_arrays(tokens: np.ndarray, offsets: np.ndarray,
+                        words_per_step: int) -> Iterator[SentenceBatch]:
+    """Slice a pre-encoded CSR corpus (e.g. from the native encoder) into
+    step-sized batches along sentence boundaries."""
+    num_sent = len(offsets) - 1
+    s0 = 0
+    while s0 < num_sent:
+        s1 = s0
+        start_tok = offsets[s0]
+        while s1 < num_sent and offsets[s1 + 1] - start_tok < words_per_step:
+            s1 += 1
+        s1 = min(max(s1 + 1, s0 + 1), num_sent)
+        yield SentenceBatch(
+            tokens=np.ascontiguousarray(tokens[offsets[s0]:offsets[s1]]),
+            offsets=np.ascontiguousarray(offsets[s0:s1 + 1] - offsets[s0]))
+        s0 = s1
+
+
 def partition_round_robin(items: Iterable, rank: int, world: int) -> Iterator:
     """Deterministic corpus partitioning across ranks (the reference's
     repartition(numPartitions), mllib:345)."""

@@ -93,28 +93,60 @@ class GlintWord2Vec:
         cfg = self.config
         seed = cfg.seed if cfg.seed is not None else np.random.SeedSequence().entropy % (2 ** 63)
         seed = int(seed)
-        reader = _corpus_reader(corpus, self.input_col)
-        vocab = build_vocab(reader(), min_count=cfg.min_count)
+        native = None
+        if isinstance(corpus, str):
+            try:
+                from . import _cpu_native as native
+            except ImportError:
+                native = None
+        if native is not None:
+            # native corpus pipeline (csrc/cpu_sgns.cpp): count/sort/encode
+            # in C++ — same semantics as the Python path (tested)
+            words, counts, total = native.build_vocab_file(corpus, cfg.min_count)
+            vocab = Vocabulary(words=list(words), counts=counts,
+                               index={w: i for i, w in enumerate(words)},
+                               train_words_count=int(total))
+            reader = _corpus_reader(corpus, self.input_col)
+        else:
+            reader = _corpus_reader(corpus, self.input_col)
+            vocab = build_vocab(reader(), min_count=cfg.min_count)
         if vocab.num_words == 0:
             raise ValueError("empty vocabulary — corpus has no word above min_count")
         log.info("vocab: %d words, %d train words", vocab.num_words, vocab.train_words_count)
+
+        max_sent = min(cfg.max_sentence_length, 1024)
+        if native is not None:
+            enc_tokens, enc_offsets = native.encode_corpus(
+                corpus, list(vocab.words), max_sent)
+
+            def batches():
+                from .data import batches_from_arrays
+                return batches_from_arrays(enc_tokens, enc_offsets,
+                                           cfg.words_per_step)
+        else:
+            def batches():
+                return batch_sentences(
+                    encode_sentences(reader(), vocab, max_sent),
+                    cfg.words_per_step)
+        self._batches = batches   # engines pick this up via fit internals
 
         device = cfg.device
         if device == "auto":
             import torch
             device = "cuda" if torch.cuda.is_available() else "cpu"
         if device == "cuda" or cfg.engine in ("dim", "row", "dp"):
-            # single- or multi-GPU engines; dim/row also run on CPU under
+            # single- or multi-GPU engines; dim/row/dp also run on CPU under
             # gloo (multi-process tests, torchrun without GPUs)
             from .parallel.engine import train_gpu
-            syn0, syn1 = train_gpu(cfg, vocab, reader, seed)
+            syn0, syn1 = train_gpu(cfg, vocab, batches, seed)
         else:
-            syn0, syn1 = self._fit_cpu(cfg, vocab, reader, seed)
+            syn0, syn1 = self._fit_cpu(cfg, vocab, batches, seed)
         return GlintWord2VecModel(config=cfg, vocab=vocab, syn0=syn0, syn1=syn1,
                                   output_col=self.output_col, input_col=self.input_col)
 
     # --- single-process CPU trainer (BASELINE config 1) --------------------
-    def _fit_cpu(self, cfg: Word2VecConfig, vocab: Vocabulary, reader, seed: int):
+    def _fit_cpu(self, cfg: Word2VecConfig, vocab: Vocabulary, batches_fn,
+                 seed: int):
         syn0, syn1 = sgns.init_tables(vocab.num_words, cfg.vector_size, seed)
         kp = None
         if not cfg.legacy_subsample and cfg.subsample_ratio > 0:
@@ -137,9 +169,7 @@ class GlintWord2Vec:
         sent_base = 0
         t0 = time.time()
         for it in range(cfg.num_iterations):
-            for batch in batch_sentences(
-                    encode_sentences(reader(), vocab, cfg.max_sentence_length),
-                    cfg.words_per_step):
+            for batch in batches_fn():
                 alpha = cfg.learning_rate * max(
                     1e-4, 1.0 - processed / (total_words + 1))
                 if _cpu_native is not None:

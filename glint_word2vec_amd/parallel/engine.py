@@ -24,11 +24,12 @@ from ..vocab import Vocabulary, build_unigram_table, encode_sentences
 log = logging.getLogger("glint_word2vec_amd")
 
 
-def train_gpu(cfg: Word2VecConfig, vocab: Vocabulary, reader: Callable,
+def train_gpu(cfg: Word2VecConfig, vocab: Vocabulary, batches_fn: Callable,
               seed: int) -> Tuple[np.ndarray, np.ndarray]:
     """Dispatch to the right engine: fused single-GPU kernel at world 1,
-    dim-sharded (default) or row-sharded at world > 1 (one rank per GPU,
-    launched via torchrun; every rank calls fit() with the same corpus)."""
+    dp/dim/row-sharded at world > 1 (one rank per GPU, launched via
+    torchrun; every rank calls fit() with the same corpus).
+    ``batches_fn()`` yields SentenceBatch objects for one epoch."""
     from . import comm
     import os
     if int(os.environ.get("WORLD_SIZE", "1")) > 1:
@@ -39,7 +40,7 @@ def train_gpu(cfg: Word2VecConfig, vocab: Vocabulary, reader: Callable,
     if engine == "auto":
         engine = "fused" if world == 1 else "dim"
     if engine in ("dim", "row", "dp") and world >= 1:
-        return _train_sharded(cfg, vocab, reader, seed, engine, rank, world)
+        return _train_sharded(cfg, vocab, batches_fn, seed, engine, rank, world)
     device = torch.device("cuda", torch.cuda.current_device())
     gs = GpuSgns(vocab.num_words, cfg.vector_size, cfg.dtype,
                  device=str(device), seed=seed)
@@ -48,8 +49,6 @@ def train_gpu(cfg: Word2VecConfig, vocab: Vocabulary, reader: Callable,
                          cfg.subsample_ratio)
     gs.set_table(build_unigram_table(vocab.counts, cfg.unigram_table_size,
                                      cfg.unigram_power))
-    max_sent = min(cfg.max_sentence_length, 1024)
-
     copy_stream = torch.cuda.Stream(device)
     compute_stream = torch.cuda.current_stream(device)
 
@@ -59,9 +58,7 @@ def train_gpu(cfg: Word2VecConfig, vocab: Vocabulary, reader: Callable,
     t0 = time.time()
     prev_done = None
     for it in range(cfg.num_iterations):
-        for batch in batch_sentences(
-                encode_sentences(reader(), vocab, max_sent),
-                cfg.words_per_step):
+        for batch in batches_fn():
             alpha = cfg.learning_rate * max(1e-4, 1.0 - processed / (total_words + 1))
             with torch.cuda.stream(copy_stream):
                 tok = torch.from_numpy(batch.tokens).to(device, non_blocking=True)
@@ -83,9 +80,9 @@ def train_gpu(cfg: Word2VecConfig, vocab: Vocabulary, reader: Callable,
     return gs.to_host()
 
 
-def _train_sharded(cfg: Word2VecConfig, vocab: Vocabulary, reader: Callable,
-                   seed: int, engine: str, rank: int, world: int
-                   ) -> Tuple[np.ndarray, np.ndarray]:
+def _train_sharded(cfg: Word2VecConfig, vocab: Vocabulary,
+                   batches_fn: Callable, seed: int, engine: str, rank: int,
+                   world: int) -> Tuple[np.ndarray, np.ndarray]:
     """Multi-GPU engines; also usable on CPU (gloo) for tests.  Dim-sharded:
     every rank walks the same data (compute split by dimension).  Row-
     sharded: corpus partitioned by rank (data parallel), rows sharded."""
@@ -97,7 +94,6 @@ def _train_sharded(cfg: Word2VecConfig, vocab: Vocabulary, reader: Callable,
                   device=device, seed=seed, counts=vocab.counts,
                   table_size=cfg.unigram_table_size, subsample=subsample,
                   window_mode=cfg.window_mode)
-    max_sent = min(cfg.max_sentence_length, 1024)
     total_words = vocab.train_words_count * cfg.num_iterations
     processed = 0
     sent_base = 0
@@ -109,9 +105,7 @@ def _train_sharded(cfg: Word2VecConfig, vocab: Vocabulary, reader: Callable,
                              f_correction=cfg.f_correction,
                              atomic=cfg.atomic_updates, **common)
         for it in range(cfg.num_iterations):
-            for batch in batch_sentences(
-                    encode_sentences(reader(), vocab, max_sent),
-                    cfg.words_per_step):
+            for batch in batches_fn():
                 alpha = cfg.learning_rate * max(
                     1e-4, 1.0 - processed / (total_words + 1))
                 tok = torch.from_numpy(batch.tokens).to(eng.device)
@@ -122,7 +116,6 @@ def _train_sharded(cfg: Word2VecConfig, vocab: Vocabulary, reader: Callable,
                 sent_base += batch.num_sentences
                 processed += batch.num_tokens
     elif engine == "dp":
-        from ..data import partition_round_robin
         from .replicated import ReplicatedSgns
         eng = ReplicatedSgns(vocab.num_words, cfg.vector_size,
                              sync_every=cfg.sync_every,
@@ -130,9 +123,9 @@ def _train_sharded(cfg: Word2VecConfig, vocab: Vocabulary, reader: Callable,
         empty = (np.zeros(0, dtype=np.int32), np.zeros(1, dtype=np.int32))
         sent_base = 10_000_000 * rank   # disjoint RNG streams per rank
         for it in range(cfg.num_iterations):
-            my_sents = partition_round_robin(
-                encode_sentences(reader(), vocab, max_sent), rank, world)
-            batches = list(batch_sentences(my_sents, cfg.words_per_step))
+            # data parallel: round-robin whole batches across ranks
+            batches = [b for i, b in enumerate(batches_fn())
+                       if i % world == rank]
             n_steps = len(batches)
             if world > 1:
                 t = torch.tensor([n_steps])
@@ -153,15 +146,13 @@ def _train_sharded(cfg: Word2VecConfig, vocab: Vocabulary, reader: Callable,
                 sent_base += max(len(offsets) - 1, 0)
                 processed += len(tokens)
     else:  # row
-        from ..data import partition_round_robin
         from .row_sharded import RowShardedSgns
         eng = RowShardedSgns(vocab.num_words, cfg.vector_size, **common)
         rng = np.random.default_rng(seed + 17 * rank)
         empty = (np.zeros(0, dtype=np.int32), np.zeros(1, dtype=np.int32))
         for it in range(cfg.num_iterations):
-            my_sents = partition_round_robin(
-                encode_sentences(reader(), vocab, max_sent), rank, world)
-            batches = list(batch_sentences(my_sents, cfg.words_per_step))
+            batches = [b for i, b in enumerate(batches_fn())
+                       if i % world == rank]
             # every rank must make the same number of collective calls
             n_steps = len(batches)
             if world > 1:

@@ -7,8 +7,11 @@ Mirrors the IT spec scenarios (ServerSideGlintWord2VecSpec.scala:290-382):
   * analogy wien - österreich + deutschland -> "berlin" with cosine > 0.9
 Reference settings (Spec:83-95): seed 1, lr 0.025, table 1e6, defaults
 elsewhere; subsampling effectively off (B1) and the B2 asymmetric window —
-window_mode="reference" reproduces those semantics.  numIterations > 1
-compensates for our synchronous-epoch structure vs their streaming workers.
+window_mode="reference" reproduces those semantics.  numIterations=2 (vs
+the reference's 1): robust margin across seeds (wien rank 4-6 of 3609,
+cosine ~0.99) while meaningfully trained — at 1 iteration every word is
+still cosine ~1.0 of every other, which is the (degenerate) regime the
+reference's own gate passes in.
 """
 import numpy as np
 import pytest
@@ -20,9 +23,9 @@ from glint_word2vec_amd import GlintWord2Vec, GlintWord2VecModel
 def de_model(de_corpus_path):
     est = (GlintWord2Vec()
            .setSeed(1).setStepSize(0.025).setUnigramTableSize(1_000_000)
-           .setSubsampleRatio(0.0).setNumIterations(5))
+           .setSubsampleRatio(0.0).setNumIterations(2))
     est.config.device = "cpu"
-    est.config.num_partitions = 4
+    est.config.num_partitions = 1   # deterministic (hogwild threads race)
     est.config.window_mode = "reference"
     return est.fit(de_corpus_path)
 
