@@ -243,6 +243,24 @@ def test_row_sharded_gpu_world1_trains():
     assert np.isfinite(p0).all() and np.isfinite(p1).all()
 
 
+def test_estimator_german_corpus_gpu():
+    """fit() from a corpus path on GPU: native C++ vocab/encode feeding the
+    fused kernel; the robust analogy gate must hold (Spec:327-382)."""
+    import os
+    from glint_word2vec_amd import GlintWord2Vec
+    corpus = os.path.join(os.path.dirname(__file__), "fixtures",
+                          "de_wikipedia_articles_country_capitals.txt")
+    est = (GlintWord2Vec().setSeed(1).setStepSize(0.025)
+           .setUnigramTableSize(1_000_000).setSubsampleRatio(0.0)
+           .setNumIterations(2))
+    est.config.device = "cuda"
+    est.config.window_mode = "reference"
+    m = est.fit(corpus)
+    assert 3500 < m.num_words < 3700
+    res = m.analogy(["wien", "deutschland"], ["österreich"], 10)
+    assert "berlin" in [w for w, _ in res]
+
+
 def test_model_gpu_serving_ops():
     """to_gpu(): device-side findSynonyms + sentence-average transform must
     agree with the host implementations."""
