@@ -39,6 +39,10 @@
 
 namespace py = pybind11;
 
+// hipGetLastError is sticky per-thread: clear stale errors (e.g. from
+// torch's own device probing) at wrapper entry so HIP_CHECK after our
+// launches reports only our result.
+#define HIP_CLEAR_ERROR() (void)hipGetLastError()
 #define HIP_CHECK(expr)                                                     \
   do {                                                                      \
     hipError_t _e = (expr);                                                 \
@@ -875,6 +879,7 @@ static void sgns_train(uintptr_t syn0, uintptr_t syn1, int is_bf16,
                        int64_t atomic_below,
                        uintptr_t stats, int blocks, int threads,
                        uintptr_t stream_ptr) {
+  HIP_CLEAR_ERROR();
   if (threads != 64 && threads != 256)
     throw std::runtime_error("threads must be 64 (serial) or 256");
   if (stride % 64 != 0) throw std::runtime_error("stride must be a multiple of 64");
@@ -958,6 +963,7 @@ static void count_pairs(uintptr_t tokens, uintptr_t offsets,
                         int n_neg, uint64_t seed, int64_t sent_id_base,
                         int ref_window, uintptr_t counts_out, int blocks,
                         int threads, uintptr_t stream_ptr) {
+  HIP_CLEAR_ERROR();
   KernelArgs a = make_walk_args(0, 0, 64, tokens, offsets, num_sentences,
                                 keep_thr, table, table_size, 0.0, window,
                                 n_neg, seed, sent_id_base, ref_window, 0);
@@ -973,6 +979,7 @@ static void dots_slice(uintptr_t syn0, uintptr_t syn1, int is_bf16,
                        int n_neg, uint64_t seed, int64_t sent_id_base,
                        int ref_window, uintptr_t pair_offsets, uintptr_t f_out,
                        int blocks, int threads, uintptr_t stream_ptr) {
+  HIP_CLEAR_ERROR();
   KernelArgs a = make_walk_args(syn0, syn1, stride, tokens, offsets,
                                 num_sentences, keep_thr, table, table_size,
                                 0.0, window, n_neg, seed, sent_id_base,
@@ -1008,6 +1015,7 @@ static void update_slice(uintptr_t syn0, uintptr_t syn1, int is_bf16,
                          int64_t atomic_below,
                          uintptr_t stats, int blocks, int threads,
                          uintptr_t stream_ptr) {
+  HIP_CLEAR_ERROR();
   KernelArgs a = make_walk_args(syn0, syn1, stride, tokens, offsets,
                                 num_sentences, keep_thr, table, table_size,
                                 alpha, window, n_neg, seed, sent_id_base,
@@ -1049,6 +1057,7 @@ static void train_pairs(uintptr_t cache0, uintptr_t cache1, int64_t stride,
                         int64_t num_groups, uintptr_t pair_target,
                         uintptr_t pair_label, double alpha, uintptr_t stats,
                         int blocks, int threads, uintptr_t stream_ptr) {
+  HIP_CLEAR_ERROR();
   if (stride % 64 != 0) throw std::runtime_error("stride must be a multiple of 64");
   const int nc = (int)(stride / 64);
   hipStream_t stream = (hipStream_t)stream_ptr;
@@ -1098,6 +1107,7 @@ static void pull_average(uintptr_t syn0, int is_bf16, int64_t stride,
                          uintptr_t tokens, uintptr_t offsets,
                          int64_t num_sentences, uintptr_t out, int blocks,
                          uintptr_t stream_ptr) {
+  HIP_CLEAR_ERROR();
   hipStream_t stream = (hipStream_t)stream_ptr;
   if (is_bf16)
     launch_pull_average_t<uint16_t>(syn0, tokens, offsets, num_sentences,
@@ -1127,6 +1137,7 @@ static void launch_norms_t(uintptr_t syn0, int64_t vocab, int64_t stride,
 
 static void norms(uintptr_t syn0, int is_bf16, int64_t vocab, int64_t stride,
                   uintptr_t out, int blocks, uintptr_t stream_ptr) {
+  HIP_CLEAR_ERROR();
   hipStream_t stream = (hipStream_t)stream_ptr;
   if (is_bf16)
     launch_norms_t<uint16_t>(syn0, vocab, stride, out, blocks, stream);
