@@ -226,6 +226,7 @@ struct KernelArgs {
 
 constexpr int kMaxSent = 1024;
 constexpr int kWavesPerBlock = 4;
+constexpr int kPosBlock = 96;   // kept positions per wave when grid is 2-D
 
 // ---------------------------------------------------------------------------
 // Shared sentence walker.  Subsample+compact into the wave's LDS buffer,
@@ -236,12 +237,18 @@ constexpr int kWavesPerBlock = 4;
 // pair_idx counts emitted pairs within the sentence, identical on every
 // rank/implementation — the spine of the dim-sharded engine.
 // ---------------------------------------------------------------------------
+// pos_lo/pos_hi bound the kept-position range this wave processes (position
+// blocks let long sentences spread over many waves; RNG is counter-based so
+// any wave can process any position).  pair_idx stays GLOBAL to the
+// sentence: the walker enumerates all positions, skipping work outside the
+// range but keeping the pair numbering identical (dim-sharded f indexing).
 template <typename Phase>
 __device__ __forceinline__ void walk_sentence_dev(
     const int32_t* __restrict__ tokens, int64_t off, int len, uint64_t base,
     const uint32_t* __restrict__ keep_thr, const int32_t* __restrict__ table,
     uint32_t table_size, int window, int n_neg, int ref_window, int lane,
-    int32_t* sent_lds, uint32_t* tgt_lds, Phase& ph) {
+    int32_t* sent_lds, uint32_t* tgt_lds, Phase& ph, int pos_lo = 0,
+    int pos_hi = 1 << 30) {
   // ---- subsample + wave compaction into LDS -----------------------------
   int L = 0;
   for (int p0 = 0; p0 < len; p0 += 64) {
@@ -285,6 +292,33 @@ __device__ __forceinline__ void walk_sentence_dev(
     if (!((lo < i) || (hi > i))) continue;
     const int n_ctx = (hi - lo + 1) - ((lo <= i && i <= hi) ? 1 : 0);
     const int total_slots = n_ctx * per_ctx;
+    if (i < pos_lo || i >= pos_hi) {
+      // outside this wave's position block: advance the pair numbering
+      // without touching rows.  Negative-collision discards still need the
+      // draws; do them 64 slots at a time like the main path.
+      for (int chunk = 0; chunk < total_slots; chunk += 64) {
+        const int slot = chunk + lane;
+        bool valid = false;
+        if (slot < total_slots) {
+          const int ctx_i = slot / per_ctx;
+          const int s_in = slot - ctx_i * per_ctx;
+          if (s_in == 0) {
+            valid = true;
+          } else {
+            int j = lo + ctx_i;
+            if (j >= i) ++j;
+            const int32_t t = sent_lds[j];
+            const uint64_t kbase = kNegBase +
+                (uint64_t)(i * (2 * window + 1) + (j - i + window)) *
+                    (uint64_t)n_neg;
+            const uint32_t un = draw_u32(base, kbase + (uint64_t)(s_in - 1));
+            valid = (table[un % table_size] != t);
+          }
+        }
+        pair_idx += __popcll(__ballot(valid));
+      }
+      continue;
+    }
     ph.begin_position(c);
     // Materialize the target list 64 slots at a time (all lanes draw in
     // parallel — negative RNG + table gathers vectorize across the wave),
@@ -455,12 +489,17 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void sgns_train_kernel(KernelA
   ph.lane = lane;
   ph.atomic_below = a.atomic_below;
 
+  // gridDim.y > 1 splits each sentence into position blocks of kPosBlock
+  // kept positions so long sentences fill the chip (hogwild across blocks,
+  // same class as across sentences)
+  const int pos_lo = (int)blockIdx.y * kPosBlock;
+  const int pos_hi = gridDim.y > 1 ? pos_lo + kPosBlock : (1 << 30);
   for (int64_t s = wave_gid; s < a.num_sentences; s += total_waves) {
     const uint64_t base = sent_base(a.seed, (uint64_t)(a.sent_id_base + s));
     walk_sentence_dev(a.tokens, a.offsets[s],
                       (int)(a.offsets[s + 1] - a.offsets[s]), base, a.keep_thr,
                       a.table, a.table_size, a.window, a.n_neg, a.ref_window,
-                      lane, sbuf[wave], tbuf[wave], ph);
+                      lane, sbuf[wave], tbuf[wave], ph, pos_lo, pos_hi);
   }
 
   // ---- stats (device-scope atomics, once per wave) -----------------------
@@ -834,13 +873,13 @@ __global__ __launch_bounds__(256) void norms_kernel(
 // ---------------------------------------------------------------------------
 template <typename T, int NC>
 static void launch_train_nc(const KernelArgs& a, bool atomic, int blocks,
-                            int threads, hipStream_t stream) {
+                            int pos_blocks, int threads, hipStream_t stream) {
   if (atomic)
-    hipLaunchKernelGGL((sgns_train_kernel<T, NC, true>), dim3(blocks),
-                       dim3(threads), 0, stream, a);
+    hipLaunchKernelGGL((sgns_train_kernel<T, NC, true>),
+                       dim3(blocks, pos_blocks), dim3(threads), 0, stream, a);
   else
-    hipLaunchKernelGGL((sgns_train_kernel<T, NC, false>), dim3(blocks),
-                       dim3(threads), 0, stream, a);
+    hipLaunchKernelGGL((sgns_train_kernel<T, NC, false>),
+                       dim3(blocks, pos_blocks), dim3(threads), 0, stream, a);
 }
 
 // Supported NC values (stride = 64*NC).  The wrapper rounds the row stride
@@ -850,11 +889,11 @@ static void launch_train_nc(const KernelArgs& a, bool atomic, int blocks,
 
 template <typename T>
 static void launch_train(const KernelArgs& a, int nc, bool atomic, int blocks,
-                         int threads, hipStream_t stream) {
+                         int pos_blocks, int threads, hipStream_t stream) {
   switch (nc) {
-#define CASE_NC(N)                                             \
-  case N:                                                      \
-    launch_train_nc<T, N>(a, atomic, blocks, threads, stream); \
+#define CASE_NC(N)                                                        \
+  case N:                                                                 \
+    launch_train_nc<T, N>(a, atomic, blocks, pos_blocks, threads, stream);\
     return;
     FOR_EACH_NC(CASE_NC)
 #undef CASE_NC
@@ -877,8 +916,8 @@ static void sgns_train(uintptr_t syn0, uintptr_t syn1, int is_bf16,
                        int window, int n_neg, uint64_t seed,
                        int64_t sent_id_base, int ref_window,
                        int64_t atomic_below,
-                       uintptr_t stats, int blocks, int threads,
-                       uintptr_t stream_ptr) {
+                       uintptr_t stats, int blocks, int pos_blocks,
+                       int threads, uintptr_t stream_ptr) {
   HIP_CLEAR_ERROR();
   if (threads != 64 && threads != 256)
     throw std::runtime_error("threads must be 64 (serial) or 256");
@@ -913,10 +952,13 @@ static void sgns_train(uintptr_t syn0, uintptr_t syn1, int is_bf16,
   }
   hipStream_t stream = (hipStream_t)stream_ptr;
   const bool use_atomic = atomic_below > 0;
+  if (pos_blocks < 1) pos_blocks = 1;
   if (is_bf16)
-    launch_train<uint16_t>(a, nc, use_atomic, blocks, threads, stream);
+    launch_train<uint16_t>(a, nc, use_atomic, blocks, pos_blocks, threads,
+                           stream);
   else
-    launch_train<float>(a, nc, use_atomic, blocks, threads, stream);
+    launch_train<float>(a, nc, use_atomic, blocks, pos_blocks, threads,
+                        stream);
   HIP_CHECK(hipGetLastError());
 }
 
@@ -1160,8 +1202,8 @@ PYBIND11_MODULE(_hip_native, m) {
         py::arg("table"), py::arg("table_size"), py::arg("alpha"),
         py::arg("window"), py::arg("n_neg"), py::arg("seed"),
         py::arg("sent_id_base"), py::arg("ref_window"), py::arg("atomic"),
-        py::arg("stats"), py::arg("blocks"), py::arg("threads"),
-        py::arg("stream"));
+        py::arg("stats"), py::arg("blocks"), py::arg("pos_blocks"),
+        py::arg("threads"), py::arg("stream"));
   m.def("count_pairs", &count_pairs);
   m.def("dots_slice", &dots_slice);
   m.def("update_slice", &update_slice);

@@ -125,14 +125,20 @@ class GpuSgns:
         if num_sent <= 0:
             return
         if serial:
-            nblocks, nthreads = 1, 64        # exactly one wave: oracle order
+            nblocks, nthreads, pos_blocks = 1, 64, 1   # one wave: oracle order
         else:
             nthreads = 256
             if blocks is None:
                 # 4 waves per block; oversubscribe the 256 CUs
-                nblocks = max(1, min((num_sent + 3) // 4, 8192))
+                nblocks = max(1, min((num_sent + 3) // 4, 2048))
             else:
                 nblocks = blocks
+            # long-sentence batches: split positions over grid.y so ~8k
+            # waves stay in flight regardless of sentence count
+            avg_len = max(1, int(tokens.numel()) // num_sent)
+            pos_blocks = min(max(1, 8192 // max(num_sent, 1)),
+                             (avg_len + 95) // 96)
+            pos_blocks = max(1, min(pos_blocks, 11))
         s = stream if stream is not None else torch.cuda.current_stream(self.device)
         self.native.sgns_train(
             self.syn0.data_ptr(), self.syn1.data_ptr(), int(self.is_bf16),
@@ -143,7 +149,8 @@ class GpuSgns:
             int(sent_id_base), int(window_mode == "reference"),
             0 if not atomic else (2 ** 31 - 1 if atomic_below is None
                                   else int(atomic_below)),
-            self._stats.data_ptr(), nblocks, nthreads, s.cuda_stream)
+            self._stats.data_ptr(), nblocks, pos_blocks, nthreads,
+            s.cuda_stream)
 
     def read_stats(self, reset: bool = True) -> GpuStats:
         h = self._stats.cpu()
