@@ -284,25 +284,25 @@ def test_model_gpu_serving_ops():
 
 
 def test_estimator_end_to_end_gpu():
+    # vocab ~200 so wave-level contention stays realistic (a 25-word corpus
+    # puts hundreds of concurrent waves on the same rows — pathological)
     rng = np.random.default_rng(5)
     sents = []
-    for _ in range(600):
+    for _ in range(3000):
         head = "aa" if rng.random() < 0.5 else "a2"
-        filler = [f"x{rng.integers(0, 20)}" for _ in range(3)]
+        filler = [f"x{rng.integers(0, 200)}" for _ in range(4)]
         sents.append([head, "bb"] + filler + [head, "bb"])
     from glint_word2vec_amd import GlintWord2Vec
     est = (GlintWord2Vec().setVectorSize(32).setMinCount(1).setSeed(4)
-           .setNumIterations(10).setWindowSize(2).setN(5)
+           .setNumIterations(6).setWindowSize(2).setN(5)
            .setUnigramTableSize(100000).setStepSize(0.05)
            .setSubsampleRatio(0.0))
     est.config.device = "cuda"
     m = est.fit(sents)
     assert np.isfinite(m.syn0).all()
-    # hogwild/atomic scheduling is nondeterministic on a 25-word corpus, so
-    # assert the value relation, not the exact rank: aa ~ a2 (identical
-    # context distribution) must beat aa ~ every filler on average
+    # aa ~ a2 (identical context distribution) must beat aa ~ fillers
     f = m.syn0 / np.linalg.norm(m.syn0, axis=1, keepdims=True)
     v = m.vocab
     sim_a2 = f[v["aa"]] @ f[v["a2"]]
-    sims_x = [f[v["aa"]] @ f[v[f"x{i}"]] for i in range(20) if f"x{i}" in v]
-    assert sim_a2 > np.mean(sims_x) + 0.15
+    sims_x = [f[v["aa"]] @ f[v[f"x{i}"]] for i in range(200) if f"x{i}" in v]
+    assert sim_a2 > np.mean(sims_x) + 0.2
