@@ -51,11 +51,16 @@ class Word2VecConfig:
     dtype: str = "float32"           # "float32" | "bfloat16" table storage
     device: str = "auto"             # "auto" | "cpu" | "cuda"
     words_per_step: int = 1 << 20    # tokens fed to the GPU per training step
-    # Row updates: atomics (no lost updates on hot rows — fp32 atomicAdd /
-    # gfx950 packed-bf16 atomic) vs plain hogwild read-modify-write (the
-    # reference's races-embraced semantics, slightly faster, loses updates
-    # under heavy contention).
-    atomic_updates: bool = True
+    # Row updates. False (default) = plain hogwild read-modify-write: the
+    # reference's races-embraced fire-and-forget semantics (mllib:425);
+    # concurrent updates to a hot row mostly collapse to one (self-limiting,
+    # like last-writer CPU hogwild).  True = atomics (fp32 atomicAdd /
+    # gfx950 packed-bf16): no lost updates, but ALL concurrent stale
+    # gradients land — on extremely hot rows (tiny vocabularies, no
+    # subsampling) the summed stale positives can run away (measured:
+    # diverges on a 200-word corpus; fine at vocab 1M + subsampling).
+    # ~3-5x slower.  Measure both for your corpus; see DESIGN.md.
+    atomic_updates: bool = False
     # --- multi-GPU engine --------------------------------------
     # "auto": fused single-GPU kernel at world 1; dim-sharded at world > 1.
     # "dim": dimension-sharded (CIKM scheme, RCCL allreduce of partial dots).
