@@ -59,8 +59,10 @@ class Word2VecConfig:
     # gradients land — on extremely hot rows (tiny vocabularies, no
     # subsampling) the summed stale positives can run away (measured:
     # diverges on a 200-word corpus; fine at vocab 1M + subsampling).
-    # ~3-5x slower.  Measure both for your corpus; see DESIGN.md.
-    atomic_updates: bool = False
+    # ~3-5x slower.  Default True: correctness-first for fit() on real
+    # corpora (the explosion regime is degenerate toy data); the benchmark
+    # measures hogwild explicitly.  See DESIGN.md.
+    atomic_updates: bool = True
     # --- multi-GPU engine --------------------------------------
     # "auto": fused single-GPU kernel at world 1; dim-sharded at world > 1.
     # "dim": dimension-sharded (CIKM scheme, RCCL allreduce of partial dots).
