@@ -34,6 +34,9 @@ class Word2VecConfig:
     # --- optimisation ------------------------------------------------------
     learning_rate: float = 0.01875   # stepSize (mllib:69)
     num_iterations: int = 1          # epochs (mllib:73)
+    # batchSize carried for API parity; the reference's batchSize*n*window
+    # <= 10000 constraint was an Akka payload limit (mllib:83-85) that has
+    # no xGMI analog — step granularity here is words_per_step/chunk_words.
     batch_size: int = 50             # positions per mini-batch (mllib:79)
     window: int = 5                  # max context window (mllib:77)
     n: int = 5                       # negatives per positive pair (mllib:81)

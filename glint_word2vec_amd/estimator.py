@@ -80,6 +80,21 @@ class GlintWord2Vec:
         # PS count maps to GPU shard count in this framework.
         return self._set(num_shards=v)
 
+    def setParameterServerHost(self, v):
+        # No separate PS cluster exists: the GPUs of this node are the
+        # servers.  Accepted for drop-in compatibility.
+        if v:
+            log.warning("setParameterServerHost(%r) ignored: parameter "
+                        "servers are the local GPUs; launch under torchrun "
+                        "for multi-GPU", v)
+        return self
+
+    def setParameterServerConfig(self, v):
+        if v:
+            log.warning("setParameterServerConfig ignored (no Akka/HOCON "
+                        "tier); use Word2VecConfig device/engine fields")
+        return self
+
     def setInputCol(self, v):
         self.input_col = v
         return self
