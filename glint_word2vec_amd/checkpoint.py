@@ -60,6 +60,49 @@ def save_model(path: str, config: Word2VecConfig, vocab: Vocabulary,
             syn1[rows].tofile(os.path.join(path, "shards", f"syn1-{s:05d}.bin"))
 
 
+def save_model_streaming(path: str, config: Word2VecConfig, vocab: Vocabulary,
+                         row_block_fn, num_shards: int = 8,
+                         block_rows: int = 1 << 20,
+                         has_syn1: bool = True) -> None:
+    """Checkpoint without materialising the full matrices on host (80M-vocab
+    models: 96 GB per table).  ``row_block_fn(which, r0, r1)`` returns the
+    f32 [r1-r0, dim] block of syn0 (which=0) / syn1 (which=1).  Shards are
+    contiguous row ranges ("row_range" layout; the reference's per-PS shard
+    files are the analog, mllib:493-498)."""
+    os.makedirs(os.path.join(path, "shards"), exist_ok=True)
+    V = vocab.num_words
+    bounds = [min(V, s * ((V + num_shards - 1) // num_shards))
+              for s in range(num_shards + 1)]
+    bounds[-1] = V
+    probe = row_block_fn(0, 0, min(1, V))
+    dim = probe.shape[1]
+    meta = {
+        "class": "glint_word2vec_amd.GlintWord2VecModel",
+        "timestamp": int(time.time() * 1000),
+        "numWords": V,
+        "vectorSize": dim,
+        "paramMap": config.to_dict(),
+    }
+    with open(os.path.join(path, "metadata"), "w") as f:
+        json.dump(meta, f, indent=2, sort_keys=True)
+    vocab.save_words(os.path.join(path, "words"))
+    np.save(os.path.join(path, "counts.npy"), vocab.counts)
+    index = {"num_shards": num_shards, "vocab": V, "dim": dim,
+             "dtype": "float32", "layout": "row_range", "bounds": bounds,
+             "has_syn1": has_syn1}
+    with open(os.path.join(path, "shards", "index.json"), "w") as f:
+        json.dump(index, f, indent=2)
+    for which, name in ((0, "syn0"), (1, "syn1"))[:2 if has_syn1 else 1]:
+        for s in range(num_shards):
+            with open(os.path.join(path, "shards", f"{name}-{s:05d}.bin"),
+                      "wb") as f:
+                for r0 in range(bounds[s], bounds[s + 1], block_rows):
+                    r1 = min(bounds[s + 1], r0 + block_rows)
+                    f.write(np.ascontiguousarray(
+                        row_block_fn(which, r0, r1), dtype=np.float32)
+                        .tobytes())
+
+
 def load_model(path: str) -> Tuple[Word2VecConfig, Vocabulary, np.ndarray, np.ndarray | None]:
     with open(os.path.join(path, "metadata")) as f:
         meta = json.load(f)
@@ -71,10 +114,15 @@ def load_model(path: str) -> Tuple[Word2VecConfig, Vocabulary, np.ndarray, np.nd
         index = json.load(f)
     V, D, k = index["vocab"], index["dim"], index["num_shards"]
     dt = np.dtype(index["dtype"])
+    layout = index.get("layout", "row_mod")
     syn0 = np.empty((V, D), dtype=dt)
     syn1 = np.empty((V, D), dtype=dt) if index.get("has_syn1") else None
     for s in range(k):
-        rows = _shard_rows(V, s, k)
+        if layout == "row_mod":
+            rows = _shard_rows(V, s, k)
+        else:  # row_range
+            b = index["bounds"]
+            rows = np.arange(b[s], b[s + 1], dtype=np.int64)
         buf = np.fromfile(os.path.join(path, "shards", f"syn0-{s:05d}.bin"), dtype=dt)
         syn0[rows] = buf.reshape(len(rows), D)
         if syn1 is not None:

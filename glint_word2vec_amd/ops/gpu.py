@@ -179,6 +179,19 @@ class GpuSgns:
                           2048, s.cuda_stream)
         return out
 
+    def save_checkpoint(self, path: str, config, vocab,
+                        num_shards: int = 8) -> None:
+        """Stream the tables from HBM to a checkpoint without materialising
+        them on host (80M-vocab models are ~100 GB on-device)."""
+        from ..checkpoint import save_model_streaming
+
+        def row_block(which, r0, r1):
+            t = self.syn0 if which == 0 else self.syn1
+            return t[r0:r1, :self.dim].float().cpu().numpy()
+
+        save_model_streaming(path, config, vocab, row_block,
+                             num_shards=num_shards)
+
     def multiply(self, vec: torch.Tensor) -> torch.Tensor:
         """Whole-table GEMV — a plain library GEMM-shaped op: rocBLAS via
         torch.matmul is the right tool (guide: hand-write only fused ops)."""

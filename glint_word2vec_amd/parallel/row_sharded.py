@@ -244,6 +244,65 @@ class RowShardedSgns:
                                    sum_fplus=0.0)
         return out
 
+    def save_checkpoint(self, path: str, config, vocab) -> None:
+        """Each rank writes its own shard file — the reference's per-PS
+        parallel save (matrix.save, mllib:493-498).  Shard layout is the
+        host checkpoint's row_mod interleave, so GlintWord2VecModel.load
+        reads it directly."""
+        import json
+        import os
+        if self.rank == 0:
+            os.makedirs(os.path.join(path, "shards"), exist_ok=True)
+            from ..checkpoint import save_model  # reuse metadata writer
+            import time as _t
+            meta = {
+                "class": "glint_word2vec_amd.GlintWord2VecModel",
+                "timestamp": int(_t.time() * 1000),
+                "numWords": vocab.num_words,
+                "vectorSize": self.dim,
+                "paramMap": config.to_dict(),
+            }
+            with open(os.path.join(path, "metadata"), "w") as f:
+                json.dump(meta, f, indent=2, sort_keys=True)
+            vocab.save_words(os.path.join(path, "words"))
+            np.save(os.path.join(path, "counts.npy"), vocab.counts)
+            with open(os.path.join(path, "shards", "index.json"), "w") as f:
+                json.dump({"num_shards": self.world, "vocab": self.vocab_size,
+                           "dim": self.dim, "dtype": "float32",
+                           "layout": "row_mod", "has_syn1": True}, f, indent=2)
+        comm.barrier()
+        block = 1 << 20
+        for t, name in ((self.syn0, "syn0"), (self.syn1, "syn1")):
+            with open(os.path.join(path, "shards",
+                                   f"{name}-{self.rank:05d}.bin"), "wb") as f:
+                for r0 in range(0, self.shard_size, block):
+                    r1 = min(self.shard_size, r0 + block)
+                    f.write(np.ascontiguousarray(
+                        t[r0:r1].float().cpu().numpy(), dtype=np.float32)
+                        .tobytes())
+        comm.barrier()
+
+    def load_checkpoint(self, path: str) -> None:
+        """Each rank reads its own shard (requires num_shards == world and
+        row_mod layout — i.e. a checkpoint written by this engine at the
+        same world size, or by save_model(num_shards=world))."""
+        import json
+        import os
+        with open(os.path.join(path, "shards", "index.json")) as f:
+            index = json.load(f)
+        if (index["num_shards"] != self.world or
+                index.get("layout", "row_mod") != "row_mod"):
+            raise ValueError(
+                f"checkpoint has {index['num_shards']} {index.get('layout')} "
+                f"shards; this engine needs {self.world} row_mod shards — "
+                "load via GlintWord2VecModel.load instead")
+        dt = np.dtype(index["dtype"])
+        for t, name in ((self.syn0, "syn0"), (self.syn1, "syn1")):
+            buf = np.fromfile(os.path.join(path, "shards",
+                                           f"{name}-{self.rank:05d}.bin"),
+                              dtype=dt).reshape(self.shard_size, self.dim)
+            t.copy_(torch.from_numpy(buf).to(t.dtype).to(self.device))
+
     def to_host(self) -> Tuple[np.ndarray, np.ndarray]:
         out = []
         for shard in (self.syn0, self.syn1):
