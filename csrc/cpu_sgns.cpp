@@ -58,6 +58,19 @@ static inline float sigmoid_clipped(float f) {
   return 1.0f / (1.0f + std::exp(-f));
 }
 
+// reference getSigmoid (mllib:292-302): floor-indexed 1000-entry LUT
+static inline float sigma_of(float f, const float* et, int64_t etn) {
+  if (f > kMaxExp) return 1.0f;
+  if (f < -kMaxExp) return 0.0f;
+  if (et) {
+    int64_t i = (int64_t)((f + kMaxExp) * (etn / (2.0f * kMaxExp)));
+    if (i >= etn) i = etn - 1;
+    if (i < 0) i = 0;
+    return et[i];
+  }
+  return 1.0f / (1.0f + std::exp(-f));
+}
+
 struct Stats {
   int64_t pairs = 0;
   int64_t positives = 0;
@@ -76,7 +89,8 @@ static void train_sentences(float* syn0, float* syn1, int64_t dim,
                             bool reference_window, Stats* stats,
                             std::vector<int32_t>& kept,
                             std::vector<float>& c_row,
-                            std::vector<float>& grad) {
+                            std::vector<float>& grad,
+                            const float* et = nullptr, int64_t etn = 0) {
   for (int64_t s = s_begin; s < s_end; ++s) {
     const int32_t* sent = tokens + offsets[s];
     int64_t len = offsets[s + 1] - offsets[s];
@@ -122,7 +136,7 @@ static void train_sentences(float* syn0, float* syn1, int64_t dim,
           float* t1 = syn1 + (int64_t)t * dim;
           float f = 0.0f;
           for (int64_t d = 0; d < dim; ++d) f += c_row[d] * t1[d];
-          float g = (1.0f - sigmoid_clipped(f)) * alpha;
+          float g = (1.0f - sigma_of(f, et, etn)) * alpha;
           for (int64_t d = 0; d < dim; ++d) {
             grad[d] += g * t1[d];
             t1[d] += g * c_row[d];
@@ -138,7 +152,7 @@ static void train_sentences(float* syn0, float* syn1, int64_t dim,
           float* t1 = syn1 + (int64_t)neg * dim;
           float f = 0.0f;
           for (int64_t d = 0; d < dim; ++d) f += c_row[d] * t1[d];
-          float g = (0.0f - sigmoid_clipped(f)) * alpha;
+          float g = (0.0f - sigma_of(f, et, etn)) * alpha;
           for (int64_t d = 0; d < dim; ++d) {
             grad[d] += g * t1[d];
             t1[d] += g * c_row[d];
@@ -464,7 +478,8 @@ static py::dict train_batch(
     py::array_t<int32_t, py::array::c_style> table,
     float alpha, int window, int n_neg,
     uint64_t seed, int64_t sent_id_base,
-    std::string window_mode, int num_threads) {
+    std::string window_mode, int num_threads,
+    py::object exp_table_obj = py::none()) {
   if (syn0.ndim() != 2 || syn1.ndim() != 2)
     throw std::runtime_error("syn0/syn1 must be 2-D float32");
   int64_t dim = syn0.shape(1);
@@ -478,6 +493,14 @@ static py::dict train_batch(
   }
   bool ref_window = (window_mode == "reference");
   if (window <= 0 || n_neg < 0) throw std::runtime_error("bad window/n");
+  const float* et = nullptr;
+  int64_t etn = 0;
+  py::array_t<float, py::array::c_style> et_arr;
+  if (!exp_table_obj.is_none()) {
+    et_arr = exp_table_obj.cast<py::array_t<float, py::array::c_style>>();
+    et = et_arr.data();
+    etn = et_arr.shape(0);
+  }
 
   float* s0 = syn0.mutable_data();
   float* s1 = syn1.mutable_data();
@@ -494,7 +517,7 @@ static py::dict train_batch(
       std::vector<int32_t> kept; std::vector<float> cr(dim), gr(dim);
       train_sentences(s0, s1, dim, tok, off, 0, num_sent, keep_prob, tab,
                       tab_size, alpha, window, n_neg, seed, sent_id_base,
-                      ref_window, &total, kept, cr, gr);
+                      ref_window, &total, kept, cr, gr, et, etn);
     } else {
       std::vector<std::thread> threads;
       std::vector<Stats> st(num_threads);
@@ -506,7 +529,7 @@ static py::dict train_batch(
           std::vector<int32_t> kept; std::vector<float> cr(dim), gr(dim);
           train_sentences(s0, s1, dim, tok, off, b, e, keep_prob, tab,
                           tab_size, alpha, window, n_neg, seed, sent_id_base,
-                          ref_window, &st[t], kept, cr, gr);
+                          ref_window, &st[t], kept, cr, gr, et, etn);
         });
       }
       for (auto& th : threads) th.join();
@@ -629,7 +652,8 @@ PYBIND11_MODULE(_cpu_native, m) {
         py::arg("offsets"), py::arg("keep_prob"), py::arg("table"),
         py::arg("alpha"), py::arg("window"), py::arg("n_neg"),
         py::arg("seed"), py::arg("sent_id_base") = 0,
-        py::arg("window_mode") = "canonical", py::arg("num_threads") = 1);
+        py::arg("window_mode") = "canonical", py::arg("num_threads") = 1,
+        py::arg("exp_table") = py::none());
   m.def("count_pairs", &count_pairs, py::arg("tokens"), py::arg("offsets"),
         py::arg("keep_prob"), py::arg("table"), py::arg("window"),
         py::arg("n_neg"), py::arg("seed"), py::arg("sent_id_base") = 0,

@@ -181,6 +181,18 @@ __device__ __forceinline__ float sigmoid_clipped(float f) {
   return 1.0f / (1.0f + expf(-f));
 }
 
+// reference getSigmoid parity mode (mllib:292-302): floor-indexed LUT
+__device__ __forceinline__ float sigma_of(float f, const float* et, int etn) {
+  if (et) {
+    if (f > 6.0f) return 1.0f;
+    if (f < -6.0f) return 0.0f;
+    int i = (int)((f + 6.0f) * (etn / 12.0f));
+    i = i >= etn ? etn - 1 : (i < 0 ? 0 : i);
+    return et[i];
+  }
+  return sigmoid_clipped(f);
+}
+
 // Fast full-wave (64-lane) float sum: 4x row_shr + row_bcast15/31 DPP adds
 // (bound_ctrl=0-fill), total in lane 63, broadcast via readlane.  Replaces
 // the 6-step __shfl_xor chain (ds_bpermute + mask setup, ~7x more VALU).
@@ -217,6 +229,8 @@ struct KernelArgs {
   int64_t stride;             // row stride in elements (= 64*NC)
   int ref_window;             // 0 canonical, 1 reference (B2) semantics
   int32_t atomic_below;       // rows < this use atomics (hot rows); rest plain
+  const float* exp_table;     // non-null: reference LUT sigmoid parity mode
+  int exp_table_size;
   // stats
   unsigned long long* d_pairs;
   unsigned long long* d_positives;
@@ -372,6 +386,8 @@ struct TrainPhase {
   float alpha;
   int lane;
   int32_t atomic_below;
+  const float* exp_table;
+  int exp_table_size;
   // per-position state
   T* c_ptr;
   int32_t c_idx;
@@ -400,7 +416,7 @@ struct TrainPhase {
 #pragma unroll
     for (int k = 0; k < NC; ++k) f += c_row[k] * t_row[k];
     f = wave_sum_f32(f);
-    const float g = (label - sigmoid_clipped(f)) * alpha;
+    const float g = (label - sigma_of(f, exp_table, exp_table_size)) * alpha;
     const bool use_atomic =
         ATOMIC && ((int32_t)(enc & 0x7FFFFFFFu) < atomic_below);
     if (use_atomic) {
@@ -488,6 +504,8 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void sgns_train_kernel(KernelA
   ph.alpha = a.alpha;
   ph.lane = lane;
   ph.atomic_below = a.atomic_below;
+  ph.exp_table = a.exp_table;
+  ph.exp_table_size = a.exp_table_size;
 
   // gridDim.y > 1 splits each sentence into position blocks of kPosBlock
   // kept positions so long sentences fill the chip (hogwild across blocks,
@@ -623,6 +641,8 @@ struct UpdateSlicePhase {
   float alpha;
   int lane;
   int32_t atomic_below;
+  const float* exp_table;
+  int exp_table_size;
   T* c_ptr;
   int32_t c_idx;
   float c_row[NC];
@@ -650,7 +670,7 @@ struct UpdateSlicePhase {
       fresh = wave_sum_f32(fresh);
       f += world_scale * (fresh - f_loc[idx]);
     }
-    const float g = (label - sigmoid_clipped(f)) * alpha;
+    const float g = (label - sigma_of(f, exp_table, exp_table_size)) * alpha;
     const bool use_atomic =
         ATOMIC && ((int32_t)(enc & 0x7FFFFFFFu) < atomic_below);
     if (use_atomic) {
@@ -736,6 +756,8 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void update_slice_kernel(
   ph.lane = lane;
   ph.world_scale = world_scale;
   ph.atomic_below = a.atomic_below;
+  ph.exp_table = a.exp_table;
+  ph.exp_table_size = a.exp_table_size;
   for (int64_t s = wave_gid; s < a.num_sentences; s += total_waves) {
     ph.f_base = f_in + pair_offsets[s];
     ph.f_loc = f_loc ? f_loc + pair_offsets[s] : nullptr;
@@ -917,7 +939,8 @@ static void sgns_train(uintptr_t syn0, uintptr_t syn1, int is_bf16,
                        int64_t sent_id_base, int ref_window,
                        int64_t atomic_below,
                        uintptr_t stats, int blocks, int pos_blocks,
-                       int threads, uintptr_t stream_ptr) {
+                       int threads, uintptr_t stream_ptr,
+                       uintptr_t exp_table, int exp_table_size) {
   HIP_CLEAR_ERROR();
   if (threads != 64 && threads != 256)
     throw std::runtime_error("threads must be 64 (serial) or 256");
@@ -943,6 +966,8 @@ static void sgns_train(uintptr_t syn0, uintptr_t syn1, int is_bf16,
   a.stride = stride;
   a.ref_window = ref_window;
   a.atomic_below = (int32_t)std::min<int64_t>(atomic_below, 0x7FFFFFFFLL);
+  a.exp_table = (const float*)exp_table;
+  a.exp_table_size = exp_table_size;
   unsigned long long* st = (unsigned long long*)stats;
   if (st) {
     a.d_pairs = st + 0;
@@ -1056,13 +1081,16 @@ static void update_slice(uintptr_t syn0, uintptr_t syn1, int is_bf16,
                          uintptr_t f_loc, double world_scale,
                          int64_t atomic_below,
                          uintptr_t stats, int blocks, int threads,
-                         uintptr_t stream_ptr) {
+                         uintptr_t stream_ptr,
+                         uintptr_t exp_table = 0, int exp_table_size = 0) {
   HIP_CLEAR_ERROR();
   KernelArgs a = make_walk_args(syn0, syn1, stride, tokens, offsets,
                                 num_sentences, keep_thr, table, table_size,
                                 alpha, window, n_neg, seed, sent_id_base,
                                 ref_window, stats);
   a.atomic_below = (int32_t)std::min<int64_t>(atomic_below, 0x7FFFFFFFLL);
+  a.exp_table = (const float*)exp_table;
+  a.exp_table_size = exp_table_size;
   const int atomic = atomic_below > 0;
   const int nc = (int)(stride / 64);
   hipStream_t stream = (hipStream_t)stream_ptr;
@@ -1203,7 +1231,8 @@ PYBIND11_MODULE(_hip_native, m) {
         py::arg("window"), py::arg("n_neg"), py::arg("seed"),
         py::arg("sent_id_base"), py::arg("ref_window"), py::arg("atomic"),
         py::arg("stats"), py::arg("blocks"), py::arg("pos_blocks"),
-        py::arg("threads"), py::arg("stream"));
+        py::arg("threads"), py::arg("stream"),
+        py::arg("exp_table") = 0, py::arg("exp_table_size") = 0);
   m.def("count_pairs", &count_pairs);
   m.def("dots_slice", &dots_slice);
   m.def("update_slice", &update_slice);

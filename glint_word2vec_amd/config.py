@@ -85,6 +85,9 @@ class Word2VecConfig:
     # context (mllib:385-387).  Default is the canonical symmetric shrunk
     # window; "reference" reproduces B2.
     window_mode: str = "canonical"   # "canonical" | "reference"
+    # sigmoid: "exact" clipped sigmoid, or "lut" = the reference's
+    # EXP_TABLE_SIZE-entry lookup (createExpTable/getSigmoid, mllib:281-302)
+    sigmoid_mode: str = "exact"      # "exact" | "lut"
 
     def __post_init__(self) -> None:
         self.validate()
@@ -112,6 +115,8 @@ class Word2VecConfig:
             raise ValueError(f"unsupported window_mode {self.window_mode!r}")
         if self.engine not in ("auto", "fused", "dim", "row", "dp"):
             raise ValueError(f"unsupported engine {self.engine!r}")
+        if self.sigmoid_mode not in ("exact", "lut"):
+            raise ValueError(f"unsupported sigmoid_mode {self.sigmoid_mode!r}")
 
     # -- (de)serialisation used by the checkpoint metadata ------------------
     def to_dict(self) -> dict:

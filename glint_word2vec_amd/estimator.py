@@ -171,6 +171,8 @@ class GlintWord2Vec:
                 kp = None
         table = build_unigram_table(vocab.counts, cfg.unigram_table_size,
                                     cfg.unigram_power)
+        exp_table = (sgns.create_exp_table() if cfg.sigmoid_mode == "lut"
+                     else None)
         try:
             from . import _cpu_native
         except ImportError:
@@ -191,7 +193,7 @@ class GlintWord2Vec:
                     st = _cpu_native.train_batch(
                         syn0, syn1, batch.tokens, batch.offsets, kp, table,
                         alpha, cfg.window, cfg.n, seed, sent_base,
-                        cfg.window_mode, cfg.num_partitions)
+                        cfg.window_mode, cfg.num_partitions, exp_table)
                     npos, sum_fp = st["positives"], st["sum_fplus"]
                 else:
                     plan = sgns.make_plan(batch.tokens, batch.offsets, kp,

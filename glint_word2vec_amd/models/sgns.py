@@ -23,6 +23,13 @@ import numpy as np
 from ..config import MAX_EXP, Word2VecConfig
 
 
+def create_exp_table(size: int = 1000, max_exp: float = MAX_EXP) -> np.ndarray:
+    """The reference's 1000-entry sigmoid lookup table (createExpTable,
+    mllib:281-290): entry i = sigma((i/size*2-1)*max_exp)."""
+    x = np.exp((np.arange(size, dtype=np.float64) / size * 2.0 - 1.0) * max_exp)
+    return (x / (x + 1.0)).astype(np.float32)
+
+
 def init_tables(vocab_size: int, dim: int, seed: int) -> Tuple[np.ndarray, np.ndarray]:
     """Canonical word2vec init: syn0 ~ U(-0.5/dim, 0.5/dim), syn1 = 0."""
     rng = np.random.default_rng(seed)

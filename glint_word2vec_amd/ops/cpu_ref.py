@@ -36,6 +36,19 @@ def sigmoid_clipped(f: float) -> float:
     return 1.0 / (1.0 + math.exp(-f))
 
 
+def sigmoid_lut(f: float, table: "np.ndarray") -> float:
+    """Reference getSigmoid semantics (mllib:292-302): clip at +-MAX_EXP,
+    else floor-indexed table lookup."""
+    if f > MAX_EXP:
+        return 1.0
+    if f < -MAX_EXP:
+        return 0.0
+    ind = int((f + MAX_EXP) * (len(table) / (2.0 * MAX_EXP)))
+    if ind >= len(table):
+        ind = len(table) - 1
+    return float(table[ind])
+
+
 @dataclass
 class TrainStats:
     pairs: int = 0
@@ -50,13 +63,16 @@ def train_batch_oracle(syn0: np.ndarray, syn1: np.ndarray,
                        table: np.ndarray,
                        alpha: float, window: int, n_neg: int,
                        seed: int, sent_id_base: int = 0,
-                       window_mode: str = "canonical") -> TrainStats:
+                       window_mode: str = "canonical",
+                       exp_table: np.ndarray | None = None) -> TrainStats:
     """In-place SGNS update over one batch.  syn0/syn1 float32 [vocab, dim].
 
     ``keep_prob`` None => subsampling off (no RNG draws for it).
     ``sent_id_base`` + local sentence index seeds each sentence's RNG stream.
     """
     stats = TrainStats()
+    sig = (sigmoid_clipped if exp_table is None
+           else (lambda f: sigmoid_lut(f, exp_table)))
     table_size = len(table)
     num_sentences = len(offsets) - 1
     do_subsample = keep_prob is not None
@@ -94,8 +110,7 @@ def train_batch_oracle(syn0: np.ndarray, syn1: np.ndarray,
                 t = kept[j]
                 # positive
                 f = float(np.dot(c_row, syn1[t]))
-                sig = sigmoid_clipped(f)
-                g = (1.0 - sig) * alpha
+                g = (1.0 - sig(f)) * alpha
                 grad += g * syn1[t]
                 syn1[t] += g * c_row
                 stats.pairs += 1
@@ -109,8 +124,7 @@ def train_batch_oracle(syn0: np.ndarray, syn1: np.ndarray,
                     if neg == t:
                         continue
                     fn = float(np.dot(c_row, syn1[neg]))
-                    sign = sigmoid_clipped(fn)
-                    gn = (0.0 - sign) * alpha
+                    gn = (0.0 - sig(fn)) * alpha
                     grad += gn * syn1[neg]
                     syn1[neg] += gn * c_row
                     stats.pairs += 1

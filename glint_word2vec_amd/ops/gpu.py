@@ -82,6 +82,7 @@ class GpuSgns:
         self._stats = torch.zeros(4, dtype=torch.int64, device=self.device)
         self.keep_thr: Optional[torch.Tensor] = None
         self.table: Optional[torch.Tensor] = None
+        self.exp_table: Optional[torch.Tensor] = None
 
     # --- state management --------------------------------------------------
     def load_host(self, syn0: np.ndarray, syn1: Optional[np.ndarray]) -> None:
@@ -105,6 +106,12 @@ class GpuSgns:
         kp = keep_probabilities(counts, train_words, ratio).astype(np.float64)
         thr = np.minimum(kp * 4294967296.0, 4294967295.0).astype(np.uint32)
         self.keep_thr = torch.from_numpy(thr.view(np.int32)).to(self.device)
+
+    def set_sigmoid_lut(self, table: Optional[np.ndarray]) -> None:
+        """Reference getSigmoid parity mode (EXP_TABLE_SIZE-entry LUT)."""
+        self.exp_table = (None if table is None else
+                          torch.from_numpy(np.ascontiguousarray(
+                              table, dtype=np.float32)).to(self.device))
 
     def set_table(self, table: np.ndarray) -> None:
         self.table = torch.from_numpy(np.ascontiguousarray(table, dtype=np.int32)) \
@@ -150,7 +157,9 @@ class GpuSgns:
             0 if not atomic else (2 ** 31 - 1 if atomic_below is None
                                   else int(atomic_below)),
             self._stats.data_ptr(), nblocks, pos_blocks, nthreads,
-            s.cuda_stream)
+            s.cuda_stream,
+            0 if self.exp_table is None else self.exp_table.data_ptr(),
+            0 if self.exp_table is None else int(self.exp_table.numel()))
 
     def read_stats(self, reset: bool = True) -> GpuStats:
         h = self._stats.cpu()

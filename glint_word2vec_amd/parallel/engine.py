@@ -49,6 +49,9 @@ def train_gpu(cfg: Word2VecConfig, vocab: Vocabulary, batches_fn: Callable,
                          cfg.subsample_ratio)
     gs.set_table(build_unigram_table(vocab.counts, cfg.unigram_table_size,
                                      cfg.unigram_power))
+    if cfg.sigmoid_mode == "lut":
+        from ..models.sgns import create_exp_table
+        gs.set_sigmoid_lut(create_exp_table())
     copy_stream = torch.cuda.Stream(device)
     compute_stream = torch.cuda.current_stream(device)
 

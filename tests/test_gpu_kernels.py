@@ -57,6 +57,25 @@ def test_serial_kernel_matches_oracle(window_mode):
     np.testing.assert_allclose(g1, a1, rtol=2e-4, atol=2e-6)
 
 
+def test_serial_kernel_lut_sigmoid_matches_oracle():
+    """Reference LUT-sigmoid parity mode on the GPU."""
+    from glint_word2vec_amd.models.sgns import create_exp_table
+    tokens, offsets, counts, table, syn0, syn1 = _problem()
+    et = create_exp_table()
+    a0, a1 = syn0.copy(), syn1.copy()
+    st_py = cpu_ref.train_batch_oracle(a0, a1, tokens, offsets, None, table,
+                                       0.03, 3, 4, seed=77, exp_table=et)
+    gs = _gpu_setup(syn0, syn1, table)
+    gs.set_sigmoid_lut(et)
+    gs.train_batch(_to_dev(tokens), _to_dev(offsets), 0.03, 3, 4, 77,
+                   serial=True, atomic=False)
+    torch.cuda.synchronize()
+    st = gs.read_stats()
+    assert st.pairs == st_py.pairs
+    g0, g1 = gs.to_host()
+    np.testing.assert_allclose(g0, a0, rtol=2e-4, atol=2e-6)
+
+
 def test_serial_kernel_with_subsampling():
     tokens, offsets, counts, table, syn0, syn1 = _problem()
     total = int(counts.sum())

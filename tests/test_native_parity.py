@@ -40,6 +40,29 @@ def test_cpp_matches_python_oracle(window_mode, subsample):
     np.testing.assert_allclose(a1, b1, rtol=1e-5, atol=1e-7)
 
 
+def test_lut_sigmoid_parity_and_semantics():
+    """LUT mode (reference getSigmoid): C++ == Python oracle, and LUT
+    differs measurably from exact sigmoid (it is a real mode)."""
+    from glint_word2vec_amd.models.sgns import create_exp_table
+    tokens, offsets, counts, table, syn0, syn1 = _problem()
+    et = create_exp_table()
+    assert et.shape == (1000,)
+    assert abs(float(et[500]) - 0.5) < 0.01
+    a0, a1 = syn0.copy(), syn1.copy()
+    st_py = cpu_ref.train_batch_oracle(a0, a1, tokens, offsets, None, table,
+                                       0.03, 3, 4, seed=99, exp_table=et)
+    b0, b1 = syn0.copy(), syn1.copy()
+    st_c = native.train_batch(b0, b1, tokens, offsets, None, table,
+                              0.03, 3, 4, 99, 0, "canonical", 1, et)
+    assert st_c["pairs"] == st_py.pairs
+    np.testing.assert_allclose(a0, b0, rtol=1e-5, atol=1e-7)
+    # differs from exact mode
+    c0, c1 = syn0.copy(), syn1.copy()
+    native.train_batch(c0, c1, tokens, offsets, None, table,
+                       0.03, 3, 4, 99, 0, "canonical", 1)
+    assert not np.allclose(a0, c0, rtol=1e-6, atol=1e-9)
+
+
 def test_cpp_multithread_stats_close():
     """Hogwild threads race on rows but must process the same pair count."""
     tokens, offsets, counts, table, syn0, syn1 = _problem(n_tokens=200)
