@@ -38,14 +38,18 @@ def sigmoid_clipped(f: float) -> float:
 
 def sigmoid_lut(f: float, table: "np.ndarray") -> float:
     """Reference getSigmoid semantics (mllib:292-302): clip at +-MAX_EXP,
-    else floor-indexed table lookup."""
+    else floor-indexed table lookup.  Index math in float32 to match the
+    C++/HIP implementations bit-for-bit."""
     if f > MAX_EXP:
         return 1.0
     if f < -MAX_EXP:
         return 0.0
-    ind = int((f + MAX_EXP) * (len(table) / (2.0 * MAX_EXP)))
+    scale = np.float32(len(table)) / np.float32(2.0 * MAX_EXP)
+    ind = int(np.float32(np.float32(f) + np.float32(MAX_EXP)) * scale)
     if ind >= len(table):
         ind = len(table) - 1
+    if ind < 0:
+        ind = 0
     return float(table[ind])
 
 
