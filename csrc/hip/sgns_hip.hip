@@ -1082,7 +1082,7 @@ static void update_slice(uintptr_t syn0, uintptr_t syn1, int is_bf16,
                          int64_t atomic_below,
                          uintptr_t stats, int blocks, int threads,
                          uintptr_t stream_ptr,
-                         uintptr_t exp_table = 0, int exp_table_size = 0) {
+                         uintptr_t exp_table, int exp_table_size) {
   HIP_CLEAR_ERROR();
   KernelArgs a = make_walk_args(syn0, syn1, stride, tokens, offsets,
                                 num_sentences, keep_thr, table, table_size,
