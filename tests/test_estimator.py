@@ -118,5 +118,29 @@ def test_fit_from_path(tmp_path):
     assert set(m.vocab.words) == {"a", "b"}
 
 
+def test_transform_words_batched(tiny_model):
+    # mllib:529-543: iterator-of-words batched lookup
+    out = tiny_model.transform_words(["aa", "bb", "aa"])
+    assert out.shape == (3, 24)
+    np.testing.assert_array_equal(out[0], out[2])
+
+
+def test_metadata_roundtrips_all_knobs(tiny_model, tmp_path):
+    """Every config knob (incl. semantics switches) survives save/load —
+    the reference persists all params in metadata (ml:187-194, 504-560)."""
+    tiny_model.config.window_mode = "reference"
+    tiny_model.config.sigmoid_mode = "lut"
+    p = str(tmp_path / "m")
+    tiny_model.save(p)
+    from glint_word2vec_amd import GlintWord2VecModel
+    m2 = GlintWord2VecModel.load(p)
+    assert m2.config.window_mode == "reference"
+    assert m2.config.sigmoid_mode == "lut"
+    assert m2.config.vector_size == tiny_model.config.vector_size
+    assert m2.config.unigram_table_size == tiny_model.config.unigram_table_size
+    tiny_model.config.window_mode = "canonical"
+    tiny_model.config.sigmoid_mode = "exact"
+
+
 def test_stop_noop(tiny_model):
     tiny_model.stop()   # must not raise without dist initialised
