@@ -170,6 +170,103 @@ struct RowIO<uint16_t, NC> {
 };
 
 // ---------------------------------------------------------------------------
+// 32-lane row I/O: each HALF-wave owns one row (two pairs per wave — 2x the
+// rows in flight per wave at the same instruction count).  Element of slot
+// m (m < NCH, NCH = stride/32): pairs of slots 2q/2q+1 = elements
+// 64*q + 2*(lane&31) + {0,1}; odd-NCH tail slot = 32*(NCH-1) + (lane&31).
+// ---------------------------------------------------------------------------
+template <typename T, int NCH>
+struct RowIO32;
+
+template <int NCH>
+struct RowIO32<float, NCH> {
+  static __device__ __forceinline__ void load(const float* row, float v[NCH], int l32) {
+#pragma unroll
+    for (int q = 0; q < NCH / 2; ++q) {
+      const float2 p = *reinterpret_cast<const float2*>(row + 64 * q + 2 * l32);
+      v[2 * q] = p.x;
+      v[2 * q + 1] = p.y;
+    }
+    if (NCH & 1) v[NCH - 1] = row[32 * (NCH - 1) + l32];
+  }
+  static __device__ __forceinline__ void store(float* row, const float v[NCH], int l32) {
+#pragma unroll
+    for (int q = 0; q < NCH / 2; ++q)
+      *reinterpret_cast<float2*>(row + 64 * q + 2 * l32) =
+          make_float2(v[2 * q], v[2 * q + 1]);
+    if (NCH & 1) row[32 * (NCH - 1) + l32] = v[NCH - 1];
+  }
+  static __device__ __forceinline__ void atomic_add(float* row, const float v[NCH], int l32) {
+#pragma unroll
+    for (int q = 0; q < NCH / 2; ++q) {
+      atomicAdd(row + 64 * q + 2 * l32, v[2 * q]);
+      atomicAdd(row + 64 * q + 2 * l32 + 1, v[2 * q + 1]);
+    }
+    if (NCH & 1) atomicAdd(row + 32 * (NCH - 1) + l32, v[NCH - 1]);
+  }
+};
+
+template <int NCH>
+struct RowIO32<uint16_t, NCH> {
+  static __device__ __forceinline__ void load(const uint16_t* row, float v[NCH], int l32) {
+#pragma unroll
+    for (int q = 0; q < NCH / 2; ++q) {
+      const uint32_t p = *reinterpret_cast<const uint32_t*>(row + 64 * q + 2 * l32);
+      v[2 * q] = bf16_to_f32((uint16_t)(p & 0xFFFF));
+      v[2 * q + 1] = bf16_to_f32((uint16_t)(p >> 16));
+    }
+    if (NCH & 1) v[NCH - 1] = bf16_to_f32(row[32 * (NCH - 1) + l32]);
+  }
+  static __device__ __forceinline__ void store(uint16_t* row, const float v[NCH], int l32) {
+#pragma unroll
+    for (int q = 0; q < NCH / 2; ++q) {
+      v2bf16 d;
+      d[0] = (__bf16)v[2 * q];
+      d[1] = (__bf16)v[2 * q + 1];
+      *reinterpret_cast<v2bf16*>(row + 64 * q + 2 * l32) = d;
+    }
+    if (NCH & 1)
+      *reinterpret_cast<__bf16*>(row + 32 * (NCH - 1) + l32) = (__bf16)v[NCH - 1];
+  }
+  static __device__ __forceinline__ void atomic_add(uint16_t* row, const float v[NCH], int l32) {
+#pragma unroll
+    for (int q = 0; q < NCH / 2; ++q) {
+      v2bf16 d;
+      d[0] = (__bf16)v[2 * q];
+      d[1] = (__bf16)v[2 * q + 1];
+      __builtin_amdgcn_global_atomic_fadd_v2bf16(
+          (v2bf16*)(row + 64 * q + 2 * l32), d);
+    }
+    if (NCH & 1) {
+      uint16_t* p = row + 32 * (NCH - 1) + l32;
+      const bool odd = ((uintptr_t)p >> 1) & 1;
+      v2bf16 d;
+      d[0] = odd ? (__bf16)0.0f : (__bf16)v[NCH - 1];
+      d[1] = odd ? (__bf16)v[NCH - 1] : (__bf16)0.0f;
+      __builtin_amdgcn_global_atomic_fadd_v2bf16(
+          (v2bf16*)((uintptr_t)p & ~(uintptr_t)3), d);
+    }
+  }
+};
+
+// Per-half (32-lane) sum: 4x row_shr + row_bcast15 leaves each half's total
+// in its lane 31/63; one width-32 shuffle broadcasts it within the half.
+__device__ __forceinline__ float half_sum_f32(float v) {
+  typedef int i32;
+  i32 x = __float_as_int(v);
+#define DPPH_ADD(ctrl)                                                       \
+  x = __float_as_int(__int_as_float(x) +                                     \
+      __int_as_float(__builtin_amdgcn_update_dpp(0, x, ctrl, 0xF, 0xF, true)))
+  DPPH_ADD(0x111);
+  DPPH_ADD(0x112);
+  DPPH_ADD(0x114);
+  DPPH_ADD(0x118);
+  DPPH_ADD(0x142);   // row_bcast:15 -> lane31 (half0) / lane63 (half1) total
+#undef DPPH_ADD
+  return __shfl(__int_as_float(x), 31, 32);
+}
+
+// ---------------------------------------------------------------------------
 // sigmoid with the reference's +-MAX_EXP clip (mllib:281-302 semantics,
 // exact sigmoid instead of the 1000-entry LUT approximation)
 // ---------------------------------------------------------------------------
@@ -526,6 +623,140 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void sgns_train_kernel(KernelA
     atomicAdd(a.d_positives, ph.w_pos);
     atomicAdd(a.d_words, ph.w_words);
     atomicAdd(a.d_sum_fplus, (double)ph.w_fplus);
+  }
+}
+
+// ---- Phase: fused train, TWO pairs per wave (32 lanes each) --------------
+// Each half-wave owns one pair's target row: 2x rows in flight per wave at
+// the same instruction count (halves process their own data in the same
+// VALU instructions).  Pair 2k+1 reads rows before pair 2k's update lands —
+// one extra pair of hogwild staleness; the serial/parity path keeps the
+// 64-lane kernel.  Center grads accumulate per half and are combined with
+// one cross-half shuffle at position end.
+template <typename T, int NCH, bool ATOMIC>
+struct TrainPhase2 {
+  T* syn0;
+  T* syn1;
+  int64_t stride;
+  float alpha;
+  int lane;     // 0..63
+  int l32;      // lane & 31
+  int half;     // lane >> 5
+  int32_t atomic_below;
+  const float* exp_table;
+  int exp_table_size;
+  T* c_ptr;
+  int32_t c_idx;
+  float c_row[NCH];
+  float grad[NCH];
+  uint32_t w_pairs = 0, w_pos = 0, w_words = 0;
+  float w_fplus = 0.0f;
+
+  __device__ __forceinline__ void begin_position(int32_t c) {
+    c_idx = c;
+    c_ptr = syn0 + (int64_t)c * stride;
+    RowIO32<T, NCH>::load(c_ptr, c_row, l32);
+#pragma unroll
+    for (int k = 0; k < NCH; ++k) grad[k] = 0.0f;
+  }
+
+  __device__ __forceinline__ void process_pairs(const uint32_t* tl, int count,
+                                                int64_t) {
+    for (int k = 0; k < count; k += 2) {
+      const int my = k + half;
+      const bool active = my < count;
+      const uint32_t enc = tl[active ? my : k];
+      T* t_ptr = syn1 + (int64_t)(enc & 0x7FFFFFFFu) * stride;
+      float t_row[NCH];
+      RowIO32<T, NCH>::load(t_ptr, t_row, l32);
+      float f = 0.0f;
+#pragma unroll
+      for (int m = 0; m < NCH; ++m) f += c_row[m] * t_row[m];
+      f = half_sum_f32(f);
+      const float label = (enc & 0x80000000u) ? 1.0f : 0.0f;
+      const float g0 = (label - sigma_of(f, exp_table, exp_table_size)) * alpha;
+      const float g = active ? g0 : 0.0f;   // idle half: zero contribution
+#pragma unroll
+      for (int m = 0; m < NCH; ++m) grad[m] += g * t_row[m];
+      if (active) {
+        const bool use_atomic =
+            ATOMIC && ((int32_t)(enc & 0x7FFFFFFFu) < atomic_below);
+        if (use_atomic) {
+          float delta[NCH];
+#pragma unroll
+          for (int m = 0; m < NCH; ++m) delta[m] = g * c_row[m];
+          RowIO32<T, NCH>::atomic_add(t_ptr, delta, l32);
+        } else {
+#pragma unroll
+          for (int m = 0; m < NCH; ++m) t_row[m] += g * c_row[m];
+          RowIO32<T, NCH>::store(t_ptr, t_row, l32);
+        }
+        ++w_pairs;
+        if (label > 0.5f) {
+          ++w_pos;
+          w_fplus += f;
+        }
+      }
+    }
+  }
+
+  __device__ __forceinline__ void end_position(int32_t) {
+    // combine the halves' grads (same elements live at lane l and l+32)
+#pragma unroll
+    for (int m = 0; m < NCH; ++m) grad[m] += __shfl_xor(grad[m], 32, 64);
+    if (ATOMIC && (c_idx < atomic_below)) {
+      if (half == 0) RowIO32<T, NCH>::atomic_add(c_ptr, grad, l32);
+    } else {
+      float cur[NCH];
+      RowIO32<T, NCH>::load(c_ptr, cur, l32);
+#pragma unroll
+      for (int m = 0; m < NCH; ++m) cur[m] += grad[m];
+      RowIO32<T, NCH>::store(c_ptr, cur, l32);   // both halves store same values
+    }
+    ++w_words;
+  }
+};
+
+template <typename T, int NCH, bool ATOMIC>
+__global__ __launch_bounds__(64 * kWavesPerBlock) void sgns_train2_kernel(KernelArgs a) {
+  __shared__ int32_t sbuf[kWavesPerBlock][kMaxSent];
+  __shared__ uint32_t tbuf[kWavesPerBlock][64];
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int waves_in_block = blockDim.x >> 6;
+  const int64_t wave_gid = (int64_t)blockIdx.x * waves_in_block + wave;
+  const int64_t total_waves = (int64_t)gridDim.x * waves_in_block;
+
+  TrainPhase2<T, NCH, ATOMIC> ph{};
+  ph.syn0 = (T*)a.syn0;
+  ph.syn1 = (T*)a.syn1;
+  ph.stride = a.stride;
+  ph.alpha = a.alpha;
+  ph.lane = lane;
+  ph.l32 = lane & 31;
+  ph.half = lane >> 5;
+  ph.atomic_below = a.atomic_below;
+  ph.exp_table = a.exp_table;
+  ph.exp_table_size = a.exp_table_size;
+
+  const int pos_lo = (int)blockIdx.y * kPosBlock;
+  const int pos_hi = gridDim.y > 1 ? pos_lo + kPosBlock : (1 << 30);
+  for (int64_t s = wave_gid; s < a.num_sentences; s += total_waves) {
+    const uint64_t base = sent_base(a.seed, (uint64_t)(a.sent_id_base + s));
+    walk_sentence_dev(a.tokens, a.offsets[s],
+                      (int)(a.offsets[s + 1] - a.offsets[s]), base, a.keep_thr,
+                      a.table, a.table_size, a.window, a.n_neg, a.ref_window,
+                      lane, sbuf[wave], tbuf[wave], ph, pos_lo, pos_hi);
+  }
+  // combine half stats then one atomic from lane 0
+  unsigned int p2 = ph.w_pairs + __shfl_xor(ph.w_pairs, 32, 64);
+  unsigned int o2 = ph.w_pos + __shfl_xor(ph.w_pos, 32, 64);
+  float f2 = ph.w_fplus + __shfl_xor(ph.w_fplus, 32, 64);
+  if (lane == 0 && a.d_pairs) {
+    atomicAdd(a.d_pairs, (unsigned long long)p2);
+    atomicAdd(a.d_positives, (unsigned long long)o2);
+    atomicAdd(a.d_words, (unsigned long long)ph.w_words);
+    atomicAdd(a.d_sum_fplus, (double)f2);
   }
 }
 
@@ -909,6 +1140,34 @@ static void launch_train_nc(const KernelArgs& a, bool atomic, int blocks,
 #define FOR_EACH_NC(X) \
   X(1) X(2) X(3) X(4) X(5) X(6) X(8) X(10) X(12) X(16) X(20) X(24) X(32)
 
+template <typename T, int NCH>
+static void launch_train2_nch(const KernelArgs& a, bool atomic, int blocks,
+                              int pos_blocks, int threads,
+                              hipStream_t stream) {
+  if (atomic)
+    hipLaunchKernelGGL((sgns_train2_kernel<T, NCH, true>),
+                       dim3(blocks, pos_blocks), dim3(threads), 0, stream, a);
+  else
+    hipLaunchKernelGGL((sgns_train2_kernel<T, NCH, false>),
+                       dim3(blocks, pos_blocks), dim3(threads), 0, stream, a);
+}
+
+template <typename T>
+static void launch_train2(const KernelArgs& a, int nc, bool atomic, int blocks,
+                          int pos_blocks, int threads, hipStream_t stream) {
+  switch (nc) {
+#define CASE_NC2(N)                                                          \
+  case N:                                                                    \
+    launch_train2_nch<T, 2 * N>(a, atomic, blocks, pos_blocks, threads,      \
+                                stream);                                     \
+    return;
+    FOR_EACH_NC(CASE_NC2)
+#undef CASE_NC2
+    default:
+      throw std::runtime_error("unsupported NC=" + std::to_string(nc));
+  }
+}
+
 template <typename T>
 static void launch_train(const KernelArgs& a, int nc, bool atomic, int blocks,
                          int pos_blocks, int threads, hipStream_t stream) {
@@ -940,7 +1199,8 @@ static void sgns_train(uintptr_t syn0, uintptr_t syn1, int is_bf16,
                        int64_t atomic_below,
                        uintptr_t stats, int blocks, int pos_blocks,
                        int threads, uintptr_t stream_ptr,
-                       uintptr_t exp_table, int exp_table_size) {
+                       uintptr_t exp_table, int exp_table_size,
+                       int pair2) {
   HIP_CLEAR_ERROR();
   if (threads != 64 && threads != 256)
     throw std::runtime_error("threads must be 64 (serial) or 256");
@@ -978,12 +1238,20 @@ static void sgns_train(uintptr_t syn0, uintptr_t syn1, int is_bf16,
   hipStream_t stream = (hipStream_t)stream_ptr;
   const bool use_atomic = atomic_below > 0;
   if (pos_blocks < 1) pos_blocks = 1;
-  if (is_bf16)
+  if (pair2 && threads == 256) {
+    if (is_bf16)
+      launch_train2<uint16_t>(a, nc, use_atomic, blocks, pos_blocks, threads,
+                              stream);
+    else
+      launch_train2<float>(a, nc, use_atomic, blocks, pos_blocks, threads,
+                           stream);
+  } else if (is_bf16) {
     launch_train<uint16_t>(a, nc, use_atomic, blocks, pos_blocks, threads,
                            stream);
-  else
+  } else {
     launch_train<float>(a, nc, use_atomic, blocks, pos_blocks, threads,
                         stream);
+  }
   HIP_CHECK(hipGetLastError());
 }
 
@@ -1232,7 +1500,8 @@ PYBIND11_MODULE(_hip_native, m) {
         py::arg("sent_id_base"), py::arg("ref_window"), py::arg("atomic"),
         py::arg("stats"), py::arg("blocks"), py::arg("pos_blocks"),
         py::arg("threads"), py::arg("stream"),
-        py::arg("exp_table") = 0, py::arg("exp_table_size") = 0);
+        py::arg("exp_table") = 0, py::arg("exp_table_size") = 0,
+        py::arg("pair2") = 0);
   m.def("count_pairs", &count_pairs);
   m.def("dots_slice", &dots_slice);
   m.def("update_slice", &update_slice);

@@ -123,7 +123,7 @@ class GpuSgns:
                     sent_id_base: int = 0, window_mode: str = "canonical",
                     atomic: bool = True, atomic_below: Optional[int] = None,
                     blocks: Optional[int] = None,
-                    serial: bool = False,
+                    serial: bool = False, pair2: bool = True,
                     stream: Optional[torch.cuda.Stream] = None) -> None:
         """Launch the fused train kernel (async on the given/current stream).
         Stats accumulate on-device; read with read_stats()."""
@@ -159,7 +159,8 @@ class GpuSgns:
             self._stats.data_ptr(), nblocks, pos_blocks, nthreads,
             s.cuda_stream,
             0 if self.exp_table is None else self.exp_table.data_ptr(),
-            0 if self.exp_table is None else int(self.exp_table.numel()))
+            0 if self.exp_table is None else int(self.exp_table.numel()),
+            int(pair2 and not serial))
 
     def read_stats(self, reset: bool = True) -> GpuStats:
         h = self._stats.cpu()
