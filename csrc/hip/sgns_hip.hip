@@ -249,6 +249,99 @@ struct RowIO32<uint16_t, NCH> {
   }
 };
 
+// 16-lane row I/O (four pairs per wave).  Element of slot m (m < NCQ,
+// NCQ = stride/16): pairs of slots 2q/2q+1 = elements 32*q + 2*(lane&15)
+// + {0,1}; odd tail slot = 16*(NCQ-1) + (lane&15).
+template <typename T, int NCQ>
+struct RowIO16;
+
+template <int NCQ>
+struct RowIO16<float, NCQ> {
+  static __device__ __forceinline__ void load(const float* row, float v[NCQ], int l16) {
+#pragma unroll
+    for (int q = 0; q < NCQ / 2; ++q) {
+      const float2 p = *reinterpret_cast<const float2*>(row + 32 * q + 2 * l16);
+      v[2 * q] = p.x;
+      v[2 * q + 1] = p.y;
+    }
+    if (NCQ & 1) v[NCQ - 1] = row[16 * (NCQ - 1) + l16];
+  }
+  static __device__ __forceinline__ void store(float* row, const float v[NCQ], int l16) {
+#pragma unroll
+    for (int q = 0; q < NCQ / 2; ++q)
+      *reinterpret_cast<float2*>(row + 32 * q + 2 * l16) =
+          make_float2(v[2 * q], v[2 * q + 1]);
+    if (NCQ & 1) row[16 * (NCQ - 1) + l16] = v[NCQ - 1];
+  }
+  static __device__ __forceinline__ void atomic_add(float* row, const float v[NCQ], int l16) {
+#pragma unroll
+    for (int q = 0; q < NCQ / 2; ++q) {
+      atomicAdd(row + 32 * q + 2 * l16, v[2 * q]);
+      atomicAdd(row + 32 * q + 2 * l16 + 1, v[2 * q + 1]);
+    }
+    if (NCQ & 1) atomicAdd(row + 16 * (NCQ - 1) + l16, v[NCQ - 1]);
+  }
+};
+
+template <int NCQ>
+struct RowIO16<uint16_t, NCQ> {
+  static __device__ __forceinline__ void load(const uint16_t* row, float v[NCQ], int l16) {
+#pragma unroll
+    for (int q = 0; q < NCQ / 2; ++q) {
+      const uint32_t p = *reinterpret_cast<const uint32_t*>(row + 32 * q + 2 * l16);
+      v[2 * q] = bf16_to_f32((uint16_t)(p & 0xFFFF));
+      v[2 * q + 1] = bf16_to_f32((uint16_t)(p >> 16));
+    }
+    if (NCQ & 1) v[NCQ - 1] = bf16_to_f32(row[16 * (NCQ - 1) + l16]);
+  }
+  static __device__ __forceinline__ void store(uint16_t* row, const float v[NCQ], int l16) {
+#pragma unroll
+    for (int q = 0; q < NCQ / 2; ++q) {
+      v2bf16 d;
+      d[0] = (__bf16)v[2 * q];
+      d[1] = (__bf16)v[2 * q + 1];
+      *reinterpret_cast<v2bf16*>(row + 32 * q + 2 * l16) = d;
+    }
+    if (NCQ & 1)
+      *reinterpret_cast<__bf16*>(row + 16 * (NCQ - 1) + l16) = (__bf16)v[NCQ - 1];
+  }
+  static __device__ __forceinline__ void atomic_add(uint16_t* row, const float v[NCQ], int l16) {
+#pragma unroll
+    for (int q = 0; q < NCQ / 2; ++q) {
+      v2bf16 d;
+      d[0] = (__bf16)v[2 * q];
+      d[1] = (__bf16)v[2 * q + 1];
+      __builtin_amdgcn_global_atomic_fadd_v2bf16(
+          (v2bf16*)(row + 32 * q + 2 * l16), d);
+    }
+    if (NCQ & 1) {
+      uint16_t* p = row + 16 * (NCQ - 1) + l16;
+      const bool odd = ((uintptr_t)p >> 1) & 1;
+      v2bf16 d;
+      d[0] = odd ? (__bf16)0.0f : (__bf16)v[NCQ - 1];
+      d[1] = odd ? (__bf16)v[NCQ - 1] : (__bf16)0.0f;
+      __builtin_amdgcn_global_atomic_fadd_v2bf16(
+          (v2bf16*)((uintptr_t)p & ~(uintptr_t)3), d);
+    }
+  }
+};
+
+// Per-quarter (16-lane) sum: row_shr prefix leaves each 16-row's total in
+// its lane 15; width-16 shuffle broadcasts it within the quarter.
+__device__ __forceinline__ float quarter_sum_f32(float v) {
+  typedef int i32;
+  i32 x = __float_as_int(v);
+#define DPPQ_ADD(ctrl)                                                       \
+  x = __float_as_int(__int_as_float(x) +                                     \
+      __int_as_float(__builtin_amdgcn_update_dpp(0, x, ctrl, 0xF, 0xF, true)))
+  DPPQ_ADD(0x111);
+  DPPQ_ADD(0x112);
+  DPPQ_ADD(0x114);
+  DPPQ_ADD(0x118);
+#undef DPPQ_ADD
+  return __shfl(__int_as_float(x), 15, 16);
+}
+
 // Per-half (32-lane) sum: 4x row_shr + row_bcast15 leaves each half's total
 // in its lane 31/63; one width-32 shuffle broadcasts it within the half.
 __device__ __forceinline__ float half_sum_f32(float v) {
@@ -760,6 +853,135 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void sgns_train2_kernel(Kernel
   }
 }
 
+// ---- Phase: fused train, FOUR pairs per wave (16-lane quarters) ----------
+template <typename T, int NCQ, bool ATOMIC>
+struct TrainPhase4 {
+  T* syn0;
+  T* syn1;
+  int64_t stride;
+  float alpha;
+  int l16;      // lane & 15
+  int quarter;  // lane >> 4
+  int32_t atomic_below;
+  const float* exp_table;
+  int exp_table_size;
+  T* c_ptr;
+  int32_t c_idx;
+  float c_row[NCQ];
+  float grad[NCQ];
+  uint32_t w_pairs = 0, w_pos = 0, w_words = 0;
+  float w_fplus = 0.0f;
+
+  __device__ __forceinline__ void begin_position(int32_t c) {
+    c_idx = c;
+    c_ptr = syn0 + (int64_t)c * stride;
+    RowIO16<T, NCQ>::load(c_ptr, c_row, l16);
+#pragma unroll
+    for (int k = 0; k < NCQ; ++k) grad[k] = 0.0f;
+  }
+
+  __device__ __forceinline__ void process_pairs(const uint32_t* tl, int count,
+                                                int64_t) {
+    for (int k = 0; k < count; k += 4) {
+      const int my = k + quarter;
+      const bool active = my < count;
+      const uint32_t enc = tl[active ? my : k];
+      T* t_ptr = syn1 + (int64_t)(enc & 0x7FFFFFFFu) * stride;
+      float t_row[NCQ];
+      RowIO16<T, NCQ>::load(t_ptr, t_row, l16);
+      float f = 0.0f;
+#pragma unroll
+      for (int m = 0; m < NCQ; ++m) f += c_row[m] * t_row[m];
+      f = quarter_sum_f32(f);
+      const float label = (enc & 0x80000000u) ? 1.0f : 0.0f;
+      const float g0 = (label - sigma_of(f, exp_table, exp_table_size)) * alpha;
+      const float g = active ? g0 : 0.0f;
+#pragma unroll
+      for (int m = 0; m < NCQ; ++m) grad[m] += g * t_row[m];
+      if (active) {
+        const bool use_atomic =
+            ATOMIC && ((int32_t)(enc & 0x7FFFFFFFu) < atomic_below);
+        if (use_atomic) {
+          float delta[NCQ];
+#pragma unroll
+          for (int m = 0; m < NCQ; ++m) delta[m] = g * c_row[m];
+          RowIO16<T, NCQ>::atomic_add(t_ptr, delta, l16);
+        } else {
+#pragma unroll
+          for (int m = 0; m < NCQ; ++m) t_row[m] += g * c_row[m];
+          RowIO16<T, NCQ>::store(t_ptr, t_row, l16);
+        }
+        ++w_pairs;
+        if (label > 0.5f) {
+          ++w_pos;
+          w_fplus += f;
+        }
+      }
+    }
+  }
+
+  __device__ __forceinline__ void end_position(int32_t) {
+#pragma unroll
+    for (int m = 0; m < NCQ; ++m) {
+      grad[m] += __shfl_xor(grad[m], 16, 64);
+      grad[m] += __shfl_xor(grad[m], 32, 64);
+    }
+    if (ATOMIC && (c_idx < atomic_below)) {
+      if (quarter == 0) RowIO16<T, NCQ>::atomic_add(c_ptr, grad, l16);
+    } else {
+      float cur[NCQ];
+      RowIO16<T, NCQ>::load(c_ptr, cur, l16);
+#pragma unroll
+      for (int m = 0; m < NCQ; ++m) cur[m] += grad[m];
+      RowIO16<T, NCQ>::store(c_ptr, cur, l16);
+    }
+    ++w_words;
+  }
+};
+
+template <typename T, int NCQ, bool ATOMIC>
+__global__ __launch_bounds__(64 * kWavesPerBlock) void sgns_train4_kernel(KernelArgs a) {
+  __shared__ int32_t sbuf[kWavesPerBlock][kMaxSent];
+  __shared__ uint32_t tbuf[kWavesPerBlock][64];
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int waves_in_block = blockDim.x >> 6;
+  const int64_t wave_gid = (int64_t)blockIdx.x * waves_in_block + wave;
+  const int64_t total_waves = (int64_t)gridDim.x * waves_in_block;
+
+  TrainPhase4<T, NCQ, ATOMIC> ph{};
+  ph.syn0 = (T*)a.syn0;
+  ph.syn1 = (T*)a.syn1;
+  ph.stride = a.stride;
+  ph.alpha = a.alpha;
+  ph.l16 = lane & 15;
+  ph.quarter = lane >> 4;
+  ph.atomic_below = a.atomic_below;
+  ph.exp_table = a.exp_table;
+  ph.exp_table_size = a.exp_table_size;
+
+  const int pos_lo = (int)blockIdx.y * kPosBlock;
+  const int pos_hi = gridDim.y > 1 ? pos_lo + kPosBlock : (1 << 30);
+  for (int64_t s = wave_gid; s < a.num_sentences; s += total_waves) {
+    const uint64_t base = sent_base(a.seed, (uint64_t)(a.sent_id_base + s));
+    walk_sentence_dev(a.tokens, a.offsets[s],
+                      (int)(a.offsets[s + 1] - a.offsets[s]), base, a.keep_thr,
+                      a.table, a.table_size, a.window, a.n_neg, a.ref_window,
+                      lane, sbuf[wave], tbuf[wave], ph, pos_lo, pos_hi);
+  }
+  unsigned int p2 = ph.w_pairs, o2 = ph.w_pos;
+  float f2 = ph.w_fplus;
+  p2 += __shfl_xor(p2, 16, 64); p2 += __shfl_xor(p2, 32, 64);
+  o2 += __shfl_xor(o2, 16, 64); o2 += __shfl_xor(o2, 32, 64);
+  f2 += __shfl_xor(f2, 16, 64); f2 += __shfl_xor(f2, 32, 64);
+  if (lane == 0 && a.d_pairs) {
+    atomicAdd(a.d_pairs, (unsigned long long)p2);
+    atomicAdd(a.d_positives, (unsigned long long)o2);
+    atomicAdd(a.d_words, (unsigned long long)ph.w_words);
+    atomicAdd(a.d_sum_fplus, (double)f2);
+  }
+}
+
 // ---- Phase: count pairs (dim-sharded phase 0) ----------------------------
 struct CountPhase {
   int64_t pairs = 0;
@@ -1168,6 +1390,34 @@ static void launch_train2(const KernelArgs& a, int nc, bool atomic, int blocks,
   }
 }
 
+template <typename T, int NCQ>
+static void launch_train4_ncq(const KernelArgs& a, bool atomic, int blocks,
+                              int pos_blocks, int threads,
+                              hipStream_t stream) {
+  if (atomic)
+    hipLaunchKernelGGL((sgns_train4_kernel<T, NCQ, true>),
+                       dim3(blocks, pos_blocks), dim3(threads), 0, stream, a);
+  else
+    hipLaunchKernelGGL((sgns_train4_kernel<T, NCQ, false>),
+                       dim3(blocks, pos_blocks), dim3(threads), 0, stream, a);
+}
+
+template <typename T>
+static void launch_train4(const KernelArgs& a, int nc, bool atomic, int blocks,
+                          int pos_blocks, int threads, hipStream_t stream) {
+  switch (nc) {
+#define CASE_NC4(N)                                                          \
+  case N:                                                                    \
+    launch_train4_ncq<T, 4 * N>(a, atomic, blocks, pos_blocks, threads,      \
+                                stream);                                     \
+    return;
+    FOR_EACH_NC(CASE_NC4)
+#undef CASE_NC4
+    default:
+      throw std::runtime_error("unsupported NC=" + std::to_string(nc));
+  }
+}
+
 template <typename T>
 static void launch_train(const KernelArgs& a, int nc, bool atomic, int blocks,
                          int pos_blocks, int threads, hipStream_t stream) {
@@ -1238,7 +1488,14 @@ static void sgns_train(uintptr_t syn0, uintptr_t syn1, int is_bf16,
   hipStream_t stream = (hipStream_t)stream_ptr;
   const bool use_atomic = atomic_below > 0;
   if (pos_blocks < 1) pos_blocks = 1;
-  if (pair2 && threads == 256) {
+  if (pair2 == 2 && threads == 256) {
+    if (is_bf16)
+      launch_train4<uint16_t>(a, nc, use_atomic, blocks, pos_blocks, threads,
+                              stream);
+    else
+      launch_train4<float>(a, nc, use_atomic, blocks, pos_blocks, threads,
+                           stream);
+  } else if (pair2 && threads == 256) {
     if (is_bf16)
       launch_train2<uint16_t>(a, nc, use_atomic, blocks, pos_blocks, threads,
                               stream);

@@ -123,7 +123,7 @@ class GpuSgns:
                     sent_id_base: int = 0, window_mode: str = "canonical",
                     atomic: bool = True, atomic_below: Optional[int] = None,
                     blocks: Optional[int] = None,
-                    serial: bool = False, pair2: bool = True,
+                    serial: bool = False, pair_mode: int = 1,
                     stream: Optional[torch.cuda.Stream] = None) -> None:
         """Launch the fused train kernel (async on the given/current stream).
         Stats accumulate on-device; read with read_stats()."""
@@ -160,7 +160,7 @@ class GpuSgns:
             s.cuda_stream,
             0 if self.exp_table is None else self.exp_table.data_ptr(),
             0 if self.exp_table is None else int(self.exp_table.numel()),
-            int(pair2 and not serial))
+            0 if serial else int(pair_mode))
 
     def read_stats(self, reset: bool = True) -> GpuStats:
         h = self._stats.cpu()
