@@ -74,6 +74,12 @@ def train_gpu(cfg: Word2VecConfig, vocab: Vocabulary, batches_fn: Callable,
             prev_done = (tok, off)
             sent_base += batch.num_sentences
             processed += batch.num_tokens
+            if log.isEnabledFor(logging.INFO):
+                st = gs.read_stats(reset=False)
+                wps = processed / max(time.time() - t0, 1e-9)
+                log.info("iter %d: %d/%d words, alpha=%.5f, %.0f words/s, "
+                         "mean_fplus=%.4f", it, processed, total_words, alpha,
+                         wps, st.sum_fplus / max(st.positives, 1))
     torch.cuda.synchronize(device)
     st = gs.read_stats()
     dt = time.time() - t0
