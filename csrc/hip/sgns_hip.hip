@@ -1228,6 +1228,194 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void update_slice_kernel(
   }
 }
 
+// ---- Phase: dim-sharded dots / update, TWO pairs per wave ---------------
+template <typename T, int NCH>
+struct DotPhase2 {
+  const T* syn0;
+  const T* syn1;
+  int64_t stride;
+  float* f_base;
+  int l32;
+  int half;
+  float c_row[NCH];
+  __device__ __forceinline__ void begin_position(int32_t c) {
+    RowIO32<T, NCH>::load(syn0 + (int64_t)c * stride, c_row, l32);
+  }
+  __device__ __forceinline__ void process_pairs(const uint32_t* tl, int count,
+                                                int64_t idx_base) {
+    for (int k = 0; k < count; k += 2) {
+      const int my = k + half;
+      const bool active = my < count;
+      const uint32_t enc = tl[active ? my : k];
+      const T* t_ptr = syn1 + (int64_t)(enc & 0x7FFFFFFFu) * stride;
+      float t_row[NCH];
+      RowIO32<T, NCH>::load(t_ptr, t_row, l32);
+      float f = 0.0f;
+#pragma unroll
+      for (int m = 0; m < NCH; ++m) f += c_row[m] * t_row[m];
+      f = half_sum_f32(f);
+      if (active && l32 == 0) f_base[idx_base + my] = f;
+    }
+  }
+  __device__ __forceinline__ void end_position(int32_t) {}
+};
+
+template <typename T, int NCH>
+__global__ __launch_bounds__(64 * kWavesPerBlock) void dots_slice2_kernel(
+    KernelArgs a, const int64_t* __restrict__ pair_offsets,
+    float* __restrict__ f_out) {
+  __shared__ int32_t sbuf[kWavesPerBlock][kMaxSent];
+  __shared__ uint32_t tbuf[kWavesPerBlock][64];
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int waves_in_block = blockDim.x >> 6;
+  const int64_t wave_gid = (int64_t)blockIdx.x * waves_in_block + wave;
+  const int64_t total_waves = (int64_t)gridDim.x * waves_in_block;
+  for (int64_t s = wave_gid; s < a.num_sentences; s += total_waves) {
+    DotPhase2<T, NCH> ph{};
+    ph.syn0 = (const T*)a.syn0;
+    ph.syn1 = (const T*)a.syn1;
+    ph.stride = a.stride;
+    ph.f_base = f_out + pair_offsets[s];
+    ph.l32 = lane & 31;
+    ph.half = lane >> 5;
+    const uint64_t base = sent_base(a.seed, (uint64_t)(a.sent_id_base + s));
+    walk_sentence_dev(a.tokens, a.offsets[s],
+                      (int)(a.offsets[s + 1] - a.offsets[s]), base, a.keep_thr,
+                      a.table, a.table_size, a.window, a.n_neg, a.ref_window,
+                      lane, sbuf[wave], tbuf[wave], ph);
+  }
+}
+
+template <typename T, int NCH, bool ATOMIC>
+struct UpdateSlicePhase2 {
+  T* syn0;
+  T* syn1;
+  int64_t stride;
+  const float* f_base;
+  const float* f_loc;
+  float world_scale;
+  float alpha;
+  int l32;
+  int half;
+  int32_t atomic_below;
+  const float* exp_table;
+  int exp_table_size;
+  T* c_ptr;
+  int32_t c_idx;
+  float c_row[NCH];
+  float grad[NCH];
+  uint32_t w_pairs = 0, w_pos = 0, w_words = 0;
+  float w_fplus = 0.0f;
+  __device__ __forceinline__ void begin_position(int32_t c) {
+    c_idx = c;
+    c_ptr = syn0 + (int64_t)c * stride;
+    RowIO32<T, NCH>::load(c_ptr, c_row, l32);
+#pragma unroll
+    for (int m = 0; m < NCH; ++m) grad[m] = 0.0f;
+  }
+  __device__ __forceinline__ void process_pairs(const uint32_t* tl, int count,
+                                                int64_t idx_base) {
+    for (int k = 0; k < count; k += 2) {
+      const int my = k + half;
+      const bool active = my < count;
+      const uint32_t enc = tl[active ? my : k];
+      T* t_ptr = syn1 + (int64_t)(enc & 0x7FFFFFFFu) * stride;
+      float t_row[NCH];
+      RowIO32<T, NCH>::load(t_ptr, t_row, l32);
+      float f = f_base[idx_base + (active ? my : k)];
+      if (f_loc) {
+        float fresh = 0.0f;
+#pragma unroll
+        for (int m = 0; m < NCH; ++m) fresh += c_row[m] * t_row[m];
+        fresh = half_sum_f32(fresh);
+        f += world_scale * (fresh - f_loc[idx_base + (active ? my : k)]);
+      }
+      const float label = (enc & 0x80000000u) ? 1.0f : 0.0f;
+      const float g0 = (label - sigma_of(f, exp_table, exp_table_size)) * alpha;
+      const float g = active ? g0 : 0.0f;
+#pragma unroll
+      for (int m = 0; m < NCH; ++m) grad[m] += g * t_row[m];
+      if (active) {
+        const bool use_atomic =
+            ATOMIC && ((int32_t)(enc & 0x7FFFFFFFu) < atomic_below);
+        if (use_atomic) {
+          float delta[NCH];
+#pragma unroll
+          for (int m = 0; m < NCH; ++m) delta[m] = g * c_row[m];
+          RowIO32<T, NCH>::atomic_add(t_ptr, delta, l32);
+        } else {
+#pragma unroll
+          for (int m = 0; m < NCH; ++m) t_row[m] += g * c_row[m];
+          RowIO32<T, NCH>::store(t_ptr, t_row, l32);
+        }
+        ++w_pairs;
+        if (label > 0.5f) {
+          ++w_pos;
+          w_fplus += f;
+        }
+      }
+    }
+  }
+  __device__ __forceinline__ void end_position(int32_t) {
+#pragma unroll
+    for (int m = 0; m < NCH; ++m) grad[m] += __shfl_xor(grad[m], 32, 64);
+    if (ATOMIC && (c_idx < atomic_below)) {
+      if (half == 0) RowIO32<T, NCH>::atomic_add(c_ptr, grad, l32);
+    } else {
+      float cur[NCH];
+      RowIO32<T, NCH>::load(c_ptr, cur, l32);
+#pragma unroll
+      for (int m = 0; m < NCH; ++m) cur[m] += grad[m];
+      RowIO32<T, NCH>::store(c_ptr, cur, l32);
+    }
+    ++w_words;
+  }
+};
+
+template <typename T, int NCH, bool ATOMIC>
+__global__ __launch_bounds__(64 * kWavesPerBlock) void update_slice2_kernel(
+    KernelArgs a, const int64_t* __restrict__ pair_offsets,
+    const float* __restrict__ f_in, const float* __restrict__ f_loc,
+    float world_scale) {
+  __shared__ int32_t sbuf[kWavesPerBlock][kMaxSent];
+  __shared__ uint32_t tbuf[kWavesPerBlock][64];
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int waves_in_block = blockDim.x >> 6;
+  const int64_t wave_gid = (int64_t)blockIdx.x * waves_in_block + wave;
+  const int64_t total_waves = (int64_t)gridDim.x * waves_in_block;
+  UpdateSlicePhase2<T, NCH, ATOMIC> ph{};
+  ph.syn0 = (T*)a.syn0;
+  ph.syn1 = (T*)a.syn1;
+  ph.stride = a.stride;
+  ph.alpha = a.alpha;
+  ph.l32 = lane & 31;
+  ph.half = lane >> 5;
+  ph.world_scale = world_scale;
+  ph.atomic_below = a.atomic_below;
+  ph.exp_table = a.exp_table;
+  ph.exp_table_size = a.exp_table_size;
+  for (int64_t s = wave_gid; s < a.num_sentences; s += total_waves) {
+    ph.f_base = f_in + pair_offsets[s];
+    ph.f_loc = f_loc ? f_loc + pair_offsets[s] : nullptr;
+    const uint64_t base = sent_base(a.seed, (uint64_t)(a.sent_id_base + s));
+    walk_sentence_dev(a.tokens, a.offsets[s],
+                      (int)(a.offsets[s + 1] - a.offsets[s]), base, a.keep_thr,
+                      a.table, a.table_size, a.window, a.n_neg, a.ref_window,
+                      lane, sbuf[wave], tbuf[wave], ph);
+  }
+  unsigned int p2 = ph.w_pairs + __shfl_xor(ph.w_pairs, 32, 64);
+  unsigned int o2 = ph.w_pos + __shfl_xor(ph.w_pos, 32, 64);
+  float f2 = ph.w_fplus + __shfl_xor(ph.w_fplus, 32, 64);
+  if (lane == 0 && a.d_pairs) {
+    atomicAdd(a.d_pairs, (unsigned long long)p2);
+    atomicAdd(a.d_positives, (unsigned long long)o2);
+    atomicAdd(a.d_words, (unsigned long long)ph.w_words);
+    atomicAdd(a.d_sum_fplus, (double)f2);
+  }
+}
+
 // ---- pairs trainer (row-sharded engine): grouped explicit plan against
 // local f32 caches; hogwild across groups with fp32 atomics on the shared
 // cache rows.  Group = one center position; targets contiguous. ------------
@@ -1570,7 +1758,8 @@ static void dots_slice(uintptr_t syn0, uintptr_t syn1, int is_bf16,
                        uintptr_t table, int64_t table_size, int window,
                        int n_neg, uint64_t seed, int64_t sent_id_base,
                        int ref_window, uintptr_t pair_offsets, uintptr_t f_out,
-                       int blocks, int threads, uintptr_t stream_ptr) {
+                       int blocks, int threads, uintptr_t stream_ptr,
+                       int pair_mode) {
   HIP_CLEAR_ERROR();
   KernelArgs a = make_walk_args(syn0, syn1, stride, tokens, offsets,
                                 num_sentences, keep_thr, table, table_size,
@@ -1579,8 +1768,16 @@ static void dots_slice(uintptr_t syn0, uintptr_t syn1, int is_bf16,
   const int nc = (int)(stride / 64);
   hipStream_t stream = (hipStream_t)stream_ptr;
 #define DOTS_CASE(T, N)                                                       \
-  hipLaunchKernelGGL((dots_slice_kernel<T, N>), dim3(blocks), dim3(threads),  \
-                     0, stream, a, (const int64_t*)pair_offsets, (float*)f_out)
+  do {                                                                        \
+    if (pair_mode && threads == 256)                                          \
+      hipLaunchKernelGGL((dots_slice2_kernel<T, 2 * N>), dim3(blocks),        \
+                         dim3(threads), 0, stream, a,                         \
+                         (const int64_t*)pair_offsets, (float*)f_out);        \
+    else                                                                      \
+      hipLaunchKernelGGL((dots_slice_kernel<T, N>), dim3(blocks),             \
+                         dim3(threads), 0, stream, a,                         \
+                         (const int64_t*)pair_offsets, (float*)f_out);        \
+  } while (0)
   switch (nc) {
 #define CASE_NC(N)                                   \
   case N:                                            \
@@ -1607,7 +1804,8 @@ static void update_slice(uintptr_t syn0, uintptr_t syn1, int is_bf16,
                          int64_t atomic_below,
                          uintptr_t stats, int blocks, int threads,
                          uintptr_t stream_ptr,
-                         uintptr_t exp_table, int exp_table_size) {
+                         uintptr_t exp_table, int exp_table_size,
+                         int pair_mode) {
   HIP_CLEAR_ERROR();
   KernelArgs a = make_walk_args(syn0, syn1, stride, tokens, offsets,
                                 num_sentences, keep_thr, table, table_size,
@@ -1621,16 +1819,28 @@ static void update_slice(uintptr_t syn0, uintptr_t syn1, int is_bf16,
   hipStream_t stream = (hipStream_t)stream_ptr;
 #define UPD_CASE(T, N)                                                        \
   do {                                                                        \
-    if (atomic)                                                               \
+    if (pair_mode && threads == 256) {                                        \
+      if (atomic)                                                             \
+        hipLaunchKernelGGL((update_slice2_kernel<T, 2 * N, true>),            \
+                           dim3(blocks), dim3(threads), 0, stream, a,         \
+                           (const int64_t*)pair_offsets, (const float*)f_in,  \
+                           (const float*)f_loc, (float)world_scale);          \
+      else                                                                    \
+        hipLaunchKernelGGL((update_slice2_kernel<T, 2 * N, false>),           \
+                           dim3(blocks), dim3(threads), 0, stream, a,         \
+                           (const int64_t*)pair_offsets, (const float*)f_in,  \
+                           (const float*)f_loc, (float)world_scale);          \
+    } else if (atomic) {                                                      \
       hipLaunchKernelGGL((update_slice_kernel<T, N, true>), dim3(blocks),     \
                          dim3(threads), 0, stream, a,                         \
                          (const int64_t*)pair_offsets, (const float*)f_in,    \
                          (const float*)f_loc, (float)world_scale);            \
-    else                                                                      \
+    } else {                                                                  \
       hipLaunchKernelGGL((update_slice_kernel<T, N, false>), dim3(blocks),    \
                          dim3(threads), 0, stream, a,                         \
                          (const int64_t*)pair_offsets, (const float*)f_in,    \
                          (const float*)f_loc, (float)world_scale);            \
+    }                                                                         \
   } while (0)
   switch (nc) {
 #define CASE_NC(N)                                  \

@@ -182,7 +182,8 @@ class DimShardedSgns:
                 poff[c[0]:c[1] + 1].data_ptr(), f.data_ptr(),
                 0 if f_loc is None else f_loc.data_ptr(), float(self.world),
                 (2 ** 31 - 1) if self.atomic else 0, self._stats.data_ptr(),
-                nb, nt, comp.cuda_stream, 0, 0)
+                nb, nt, comp.cuda_stream, 0, 0,
+                0 if self.serial else 1)
 
         for (a, b) in chunks:
             n = b - a
