@@ -607,8 +607,11 @@ struct TrainPhase {
     for (int k = 0; k < NC; ++k) f += c_row[k] * t_row[k];
     f = wave_sum_f32(f);
     const float g = (label - sigma_of(f, exp_table, exp_table_size)) * alpha;
+    // atomic_below < 0: atomics for POSITIVE pairs only (they drive the
+    // learning signal; negatives self-limit) — ~1/6 of the update traffic
     const bool use_atomic =
-        ATOMIC && ((int32_t)(enc & 0x7FFFFFFFu) < atomic_below);
+        ATOMIC && (atomic_below < 0 ? (label > 0.5f)
+                                    : ((int32_t)(enc & 0x7FFFFFFFu) < atomic_below));
     if (use_atomic) {
       float delta[NC];
 #pragma unroll
@@ -664,7 +667,7 @@ struct TrainPhase {
 
   __device__ __forceinline__ void end_position(int32_t) {
     // center row update (hogwild: re-read current value, add, store)
-    if (ATOMIC && (c_idx < atomic_below)) {
+    if (ATOMIC && (atomic_below < 0 || c_idx < atomic_below)) {
       RowIO<T, NC>::atomic_add(c_ptr, grad, lane);
     } else {
       float cur[NC];
@@ -773,7 +776,8 @@ struct TrainPhase2 {
       for (int m = 0; m < NCH; ++m) grad[m] += g * t_row[m];
       if (active) {
         const bool use_atomic =
-            ATOMIC && ((int32_t)(enc & 0x7FFFFFFFu) < atomic_below);
+            ATOMIC && (atomic_below < 0 ? (label > 0.5f)
+                                        : ((int32_t)(enc & 0x7FFFFFFFu) < atomic_below));
         if (use_atomic) {
           float delta[NCH];
 #pragma unroll
@@ -797,7 +801,7 @@ struct TrainPhase2 {
     // combine the halves' grads (same elements live at lane l and l+32)
 #pragma unroll
     for (int m = 0; m < NCH; ++m) grad[m] += __shfl_xor(grad[m], 32, 64);
-    if (ATOMIC && (c_idx < atomic_below)) {
+    if (ATOMIC && (atomic_below < 0 || c_idx < atomic_below)) {
       if (half == 0) RowIO32<T, NCH>::atomic_add(c_ptr, grad, l32);
     } else {
       float cur[NCH];
@@ -1124,8 +1128,11 @@ struct UpdateSlicePhase {
       f += world_scale * (fresh - f_loc[idx]);
     }
     const float g = (label - sigma_of(f, exp_table, exp_table_size)) * alpha;
+    // atomic_below < 0: atomics for POSITIVE pairs only (they drive the
+    // learning signal; negatives self-limit) — ~1/6 of the update traffic
     const bool use_atomic =
-        ATOMIC && ((int32_t)(enc & 0x7FFFFFFFu) < atomic_below);
+        ATOMIC && (atomic_below < 0 ? (label > 0.5f)
+                                    : ((int32_t)(enc & 0x7FFFFFFFu) < atomic_below));
     if (use_atomic) {
       float delta[NC];
 #pragma unroll
@@ -1674,7 +1681,7 @@ static void sgns_train(uintptr_t syn0, uintptr_t syn1, int is_bf16,
     a.d_sum_fplus = (double*)(st + 3);
   }
   hipStream_t stream = (hipStream_t)stream_ptr;
-  const bool use_atomic = atomic_below > 0;
+  const bool use_atomic = atomic_below != 0;
   if (pos_blocks < 1) pos_blocks = 1;
   if (pair2 == 2 && threads == 256) {
     if (is_bf16)

@@ -126,7 +126,10 @@ class GpuSgns:
                     serial: bool = False, pair_mode: int = 1,
                     stream: Optional[torch.cuda.Stream] = None) -> None:
         """Launch the fused train kernel (async on the given/current stream).
-        Stats accumulate on-device; read with read_stats()."""
+        Stats accumulate on-device; read with read_stats().
+        atomic_below: None = all rows atomic; K > 0 = rows < K only;
+        -1 = positive pairs (+ centers) only — the quality-critical 1/6 of
+        update traffic (see benchmarks/quality_probe.py results)."""
         assert self.table is not None, "call set_table first"
         num_sent = int(offsets.numel() - 1)
         if num_sent <= 0:
