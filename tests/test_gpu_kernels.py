@@ -119,6 +119,19 @@ def test_parallel_kernel_exact_counts_stable_values():
         np.linalg.norm(a0 - syn0), rel=0.3)
 
 
+def test_atomic_positives_only_mode():
+    tokens, offsets, counts, table, syn0, syn1 = _problem()
+    gs = _gpu_setup(syn0, syn1, table)
+    gs.train_batch(_to_dev(tokens), _to_dev(offsets), 0.03, 3, 4, 77,
+                   atomic=True, atomic_below=-1)
+    torch.cuda.synchronize()
+    st = gs.read_stats()
+    assert st.pairs > 0
+    g0, g1 = gs.to_host()
+    assert np.isfinite(g0).all() and np.isfinite(g1).all()
+    assert not np.allclose(g1, syn1)
+
+
 def test_atomic_variant_fp32():
     tokens, offsets, counts, table, syn0, syn1 = _problem()
     gs = _gpu_setup(syn0, syn1, table)
