@@ -10,12 +10,17 @@
 //   reduction) -> clipped sigmoid + gradient -> SGD scatter-add on both
 //   tables.
 //
-// Execution model: one 64-lane wave per sentence (4 waves per 256-thread
-// workgroup), sentences assigned round-robin over the launched waves.  The
-// per-GPU asynchrony of the reference's numPartitions workers (mllib:120-127)
-// becomes hogwild waves: row updates are plain read-modify-write by default
-// (races embraced exactly as the reference embraces async PS updates), with
-// an atomicAdd variant for fp32 when determinism-vs-throughput is preferred.
+// Execution model: one wave per sentence (4 waves per 256-thread
+// workgroup), sentences round-robin over the launched waves; long
+// sentences additionally split into position blocks over gridDim.y.  The
+// shipped variant runs TWO pairs per wave (32-lane halves — sgns_train2),
+// with 64-lane (exact oracle order, used for serial parity) and 16-lane
+// four-pair variants selectable via pair_mode.  The per-GPU asynchrony of
+// the reference's numPartitions workers (mllib:120-127) becomes hogwild
+// waves: row updates are plain read-modify-write (the reference's
+// fire-and-forget adjust), with atomic modes (all rows / hot rows /
+// positive pairs) for the measured quality/speed frontier
+// (benchmarks/results.md).
 //
 // Rows are laid out [vocab][stride] with stride a multiple of 64 elements;
 // each lane owns elements in chunk-pairs of 128 (2 adjacent elements per

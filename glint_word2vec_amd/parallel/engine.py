@@ -1,11 +1,11 @@
 """Training engines.
 
-``train_gpu``: full-table single-GPU training (BASELINE.json config 2) —
-both embedding matrices resident in HBM, one fused-kernel launch per step,
-host->device token upload overlapped with compute via a separate copy
-stream + double-buffered pinned staging.
-
-The multi-GPU row-sharded engine lives in sharded.py.
+``train_gpu`` dispatches on world size and ``config.engine``:
+  * fused   — full-table single-GPU (BASELINE config 2): both matrices in
+    HBM, one fused-kernel launch per step, H2D upload on a copy stream;
+  * dp      — replicated tables + periodic delta-allreduce (replicated.py);
+  * dim     — dimension-sharded CIKM scheme (dim_sharded.py);
+  * row     — row-sharded alltoallv pull/push (row_sharded.py).
 """
 from __future__ import annotations
 
