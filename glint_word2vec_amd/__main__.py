@@ -35,9 +35,11 @@ def _add_train(sub):
     p.add_argument("--seed", type=int, default=None)
     p.add_argument("--unigram-table-size", type=int, default=100_000_000)
     p.add_argument("--max-sentence-length", type=int, default=1000)
-    p.add_argument("--dtype", choices=["float32", "bfloat16"], default="float32")
+    p.add_argument("--dtype", choices=["auto", "float32", "bfloat16"],
+                   default="auto")
     p.add_argument("--device", choices=["auto", "cpu", "cuda"], default="auto")
-    p.add_argument("--engine", choices=["auto", "fused", "dim", "row"],
+    p.add_argument("--engine",
+                   choices=["auto", "fused", "dim", "row", "dp"],
                    default="auto")
     p.add_argument("--window-mode", choices=["canonical", "reference"],
                    default="canonical")

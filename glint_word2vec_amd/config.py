@@ -51,7 +51,10 @@ class Word2VecConfig:
     num_partitions: int = 1          # concurrent async workers per GPU
     num_shards: Optional[int] = None  # row shards; None = world size
     # --- device ------------------------------------------------------------
-    dtype: str = "float32"           # "float32" | "bfloat16" table storage
+    # table storage dtype. "auto" = bfloat16 on GPU, float32 on CPU: the
+    # planted-synonym probe measured bf16 as BOTH faster and higher-quality
+    # than fp32 on GPU (implicit regularization — benchmarks/results.md).
+    dtype: str = "auto"              # "auto" | "float32" | "bfloat16"
     device: str = "auto"             # "auto" | "cpu" | "cuda"
     words_per_step: int = 1 << 20    # tokens fed to the GPU per training step
     # Row updates. False (default) = plain hogwild read-modify-write: the
@@ -109,7 +112,7 @@ class Word2VecConfig:
             raise ValueError("max_sentence_length must be > 0")
         if self.unigram_table_size <= 0:
             raise ValueError("unigram_table_size must be > 0")
-        if self.dtype not in ("float32", "bfloat16"):
+        if self.dtype not in ("auto", "float32", "bfloat16"):
             raise ValueError(f"unsupported dtype {self.dtype!r}")
         if self.window_mode not in ("canonical", "reference"):
             raise ValueError(f"unsupported window_mode {self.window_mode!r}")

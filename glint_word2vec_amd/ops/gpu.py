@@ -57,6 +57,8 @@ class GpuSgns:
                  init: str = "word2vec",
                  syn0_host: Optional[np.ndarray] = None,
                  syn1_host: Optional[np.ndarray] = None):
+        if dtype not in ("float32", "bfloat16"):
+            raise ValueError(f"GpuSgns needs a concrete dtype, got {dtype!r}")
         self.native = _load_native()
         self.device = torch.device(device)
         self.vocab_size = vocab_size

@@ -42,7 +42,8 @@ def train_gpu(cfg: Word2VecConfig, vocab: Vocabulary, batches_fn: Callable,
     if engine in ("dim", "row", "dp") and world >= 1:
         return _train_sharded(cfg, vocab, batches_fn, seed, engine, rank, world)
     device = torch.device("cuda", torch.cuda.current_device())
-    gs = GpuSgns(vocab.num_words, cfg.vector_size, cfg.dtype,
+    dtype = "bfloat16" if cfg.dtype == "auto" else cfg.dtype
+    gs = GpuSgns(vocab.num_words, cfg.vector_size, dtype,
                  device=str(device), seed=seed)
     if not cfg.legacy_subsample and cfg.subsample_ratio > 0:
         gs.set_subsample(vocab.counts, vocab.train_words_count,
@@ -99,7 +100,9 @@ def _train_sharded(cfg: Word2VecConfig, vocab: Vocabulary,
     device = ("cuda:" + str(torch.cuda.current_device())
               if torch.cuda.is_available() else "cpu")
     subsample = 0.0 if cfg.legacy_subsample else cfg.subsample_ratio
-    common = dict(dtype=cfg.dtype if device != "cpu" else "float32",
+    dtype = ("float32" if device == "cpu"
+             else ("bfloat16" if cfg.dtype == "auto" else cfg.dtype))
+    common = dict(dtype=dtype,
                   device=device, seed=seed, counts=vocab.counts,
                   table_size=cfg.unigram_table_size, subsample=subsample,
                   window_mode=cfg.window_mode)
