@@ -104,7 +104,13 @@ class GlintWord2Vec:
         return self
 
     # ----------------------------------------------------------------------
-    def fit(self, corpus: CorpusLike) -> "GlintWord2VecModel":
+    def fit(self, corpus: CorpusLike, save_path: Optional[str] = None,
+            materialize: bool = True) -> "GlintWord2VecModel":
+        """Train.  ``save_path``: checkpoint straight from the engine
+        (streamed from HBM / per-rank shards — required path for 80M-vocab
+        models whose matrices must not materialise on host).
+        ``materialize=False``: skip host assembly and return None (load
+        later with GlintWord2VecModel.load)."""
         cfg = self.config
         seed = cfg.seed if cfg.seed is not None else np.random.SeedSequence().entropy % (2 ** 63)
         seed = int(seed)
@@ -153,9 +159,16 @@ class GlintWord2Vec:
             # single- or multi-GPU engines; dim/row/dp also run on CPU under
             # gloo (multi-process tests, torchrun without GPUs)
             from .parallel.engine import train_gpu
-            syn0, syn1 = train_gpu(cfg, vocab, batches, seed)
+            syn0, syn1 = train_gpu(cfg, vocab, batches, seed,
+                                   save_path=save_path,
+                                   materialize=materialize)
         else:
             syn0, syn1 = self._fit_cpu(cfg, vocab, batches, seed)
+            if save_path is not None:
+                from .checkpoint import save_model
+                save_model(save_path, cfg, vocab, syn0, syn1)
+        if not materialize:
+            return None
         return GlintWord2VecModel(config=cfg, vocab=vocab, syn0=syn0, syn1=syn1,
                                   output_col=self.output_col, input_col=self.input_col)
 

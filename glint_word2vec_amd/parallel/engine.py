@@ -25,7 +25,8 @@ log = logging.getLogger("glint_word2vec_amd")
 
 
 def train_gpu(cfg: Word2VecConfig, vocab: Vocabulary, batches_fn: Callable,
-              seed: int) -> Tuple[np.ndarray, np.ndarray]:
+              seed: int, save_path=None, materialize: bool = True
+              ) -> Tuple[np.ndarray, np.ndarray]:
     """Dispatch to the right engine: fused single-GPU kernel at world 1,
     dp/dim/row-sharded at world > 1 (one rank per GPU, launched via
     torchrun; every rank calls fit() with the same corpus).
@@ -40,7 +41,8 @@ def train_gpu(cfg: Word2VecConfig, vocab: Vocabulary, batches_fn: Callable,
     if engine == "auto":
         engine = "fused" if world == 1 else "dim"
     if engine in ("dim", "row", "dp") and world >= 1:
-        return _train_sharded(cfg, vocab, batches_fn, seed, engine, rank, world)
+        return _train_sharded(cfg, vocab, batches_fn, seed, engine, rank,
+                              world, save_path, materialize)
     device = torch.device("cuda", torch.cuda.current_device())
     dtype = "bfloat16" if cfg.dtype == "auto" else cfg.dtype
     gs = GpuSgns(vocab.num_words, cfg.vector_size, dtype,
@@ -87,12 +89,18 @@ def train_gpu(cfg: Word2VecConfig, vocab: Vocabulary, batches_fn: Callable,
     log.info("GPU training done: %d words in %.2fs (%.0f words/s), "
              "%d pairs, mean_fplus=%.4f", processed, dt, processed / max(dt, 1e-9),
              st.pairs, st.sum_fplus / max(st.positives, 1))
+    if save_path:
+        # streamed from HBM — no full-matrix host materialisation
+        gs.save_checkpoint(save_path, cfg, vocab)
+    if not materialize:
+        return None, None
     return gs.to_host()
 
 
 def _train_sharded(cfg: Word2VecConfig, vocab: Vocabulary,
                    batches_fn: Callable, seed: int, engine: str, rank: int,
-                   world: int) -> Tuple[np.ndarray, np.ndarray]:
+                   world: int, save_path=None, materialize: bool = True
+                   ) -> Tuple[np.ndarray, np.ndarray]:
     """Multi-GPU engines; also usable on CPU (gloo) for tests.  Dim-sharded:
     every rank walks the same data (compute split by dimension).  Row-
     sharded: corpus partitioned by rank (data parallel), rows sharded."""
@@ -183,4 +191,19 @@ def _train_sharded(cfg: Word2VecConfig, vocab: Vocabulary,
     log.info("%s-sharded training (rank %d/%d): %d words in %.2fs, %d pairs, "
              "mean_fplus=%.4f", engine, rank, world, processed, dt, st.pairs,
              st.sum_fplus / max(st.positives, 1))
+    if save_path:
+        if hasattr(eng, "save_checkpoint"):
+            eng.save_checkpoint(save_path, cfg, vocab)
+        else:
+            # dim/dp engines: assemble on host (round-2: per-slice streaming)
+            s0, s1 = eng.to_host()
+            if rank == 0:
+                from ..checkpoint import save_model
+                save_model(save_path, cfg, vocab, s0, s1,
+                           num_shards=max(world, 1))
+            if not materialize:
+                return None, None
+            return s0, s1
+    if not materialize:
+        return None, None
     return eng.to_host()

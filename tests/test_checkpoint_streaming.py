@@ -64,6 +64,24 @@ def _worker(rank, world, rdv, out_dir):
         dist.destroy_process_group()
 
 
+def test_fit_save_path_no_materialize(tmp_path):
+    """fit(save_path=..., materialize=False): checkpoint written engine-side,
+    nothing assembled on host (the 80M-vocab production path)."""
+    from glint_word2vec_amd import GlintWord2Vec, GlintWord2VecModel
+    sents = [["a", "b", "c", "a", "b"]] * 100
+    est = (GlintWord2Vec().setVectorSize(8).setMinCount(1).setSeed(2)
+           .setUnigramTableSize(1000).setNumIterations(2)
+           .setSubsampleRatio(0.0))
+    est.config.device = "cpu"
+    est.config.engine = "row"     # has engine-side save_checkpoint
+    p = str(tmp_path / "big_model")
+    out = est.fit(sents, save_path=p, materialize=False)
+    assert out is None
+    m = GlintWord2VecModel.load(p)
+    assert set(m.vocab.words) == {"a", "b", "c"}
+    assert np.isfinite(m.syn0).all()
+
+
 def test_row_sharded_distributed_checkpoint(tmp_path):
     rdv = str(tmp_path / "rdv")
     mp.spawn(_worker, args=(2, rdv, str(tmp_path)), nprocs=2, join=True)
