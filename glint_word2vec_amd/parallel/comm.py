@@ -77,6 +77,26 @@ def all_reduce_sum(t: torch.Tensor) -> torch.Tensor:
     return t
 
 
+def bf16_collectives_ok() -> bool:
+    """bf16 tensors over collectives: RCCL yes, gloo no."""
+    return (dist.is_initialized() and dist.get_backend() == "nccl")
+
+
+def all_reduce_sum_compressed(t: torch.Tensor) -> torch.Tensor:
+    """Sum-allreduce with bf16 transport when the backend supports it
+    (halves xGMI bytes; ~0.4% relative rounding on the summand — fine for
+    SGNS dot partials and table deltas)."""
+    if not (dist.is_initialized() and dist.get_world_size() > 1):
+        return t
+    if t.dtype == torch.float32 and bf16_collectives_ok():
+        c = t.bfloat16()
+        dist.all_reduce(c, op=dist.ReduceOp.SUM)
+        t.copy_(c.float())
+    else:
+        dist.all_reduce(t, op=dist.ReduceOp.SUM)
+    return t
+
+
 def broadcast_(t: torch.Tensor, src: int = 0) -> torch.Tensor:
     if dist.is_initialized() and dist.get_world_size() > 1:
         dist.broadcast(t, src=src)

@@ -204,7 +204,7 @@ class DimShardedSgns:
             ev.record(comp)
             with torch.cuda.stream(cs):
                 cs.wait_event(ev)
-                comm.all_reduce_sum(f[lo:hi])
+                comm.all_reduce_sum_compressed(f[lo:hi])
                 ar_ev = torch.cuda.Event()
                 ar_ev.record(cs)
             if prev is not None:

@@ -107,7 +107,7 @@ class ReplicatedSgns:
         for syn, master in ((self.syn0, self.master0),
                             (self.syn1, self.master1)):
             delta = syn.float() - master
-            comm.all_reduce_sum(delta)
+            comm.all_reduce_sum_compressed(delta)
             master += delta
             syn.copy_(master.to(syn.dtype))
 
