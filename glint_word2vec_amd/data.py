@@ -8,9 +8,8 @@ array (CSR layout): tokens int32 [total], offsets int32 [num_sentences+1].
 """
 from __future__ import annotations
 
-import itertools
 from dataclasses import dataclass
-from typing import Iterable, Iterator, List, Sequence
+from typing import Iterable, Iterator, List
 
 import numpy as np
 

@@ -9,7 +9,6 @@ fluent ``setX`` setters are provided as aliases.
 from __future__ import annotations
 
 import logging
-import math
 import time
 from typing import Iterable, List, Optional, Sequence, Union
 

@@ -16,11 +16,11 @@ the oracle's job, not this path's.
 from __future__ import annotations
 
 from dataclasses import dataclass
-from typing import Iterator, Tuple
+from typing import Tuple
 
 import numpy as np
 
-from ..config import MAX_EXP, Word2VecConfig
+from ..config import MAX_EXP
 
 
 def create_exp_table(size: int = 1000, max_exp: float = MAX_EXP) -> np.ndarray:

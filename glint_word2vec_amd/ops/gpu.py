@@ -16,7 +16,6 @@ from typing import Optional
 import numpy as np
 import torch
 
-from ..config import Word2VecConfig
 from ..vocab import keep_probabilities
 
 

@@ -17,9 +17,8 @@ import numpy as np
 import torch
 
 from ..config import Word2VecConfig
-from ..data import batch_sentences
 from ..ops.gpu import GpuSgns
-from ..vocab import Vocabulary, build_unigram_table, encode_sentences
+from ..vocab import Vocabulary, build_unigram_table
 
 log = logging.getLogger("glint_word2vec_amd")
 
