@@ -365,6 +365,13 @@ class GlintWord2VecModel:
                 break
         return out
 
+    def find_synonyms_df(self, word_or_vec, num: int):
+        """DataFrame variant of findSynonyms (ml:390-420 returns a
+        (word, similarity) DataFrame)."""
+        import pandas as pd
+        rows = self.find_synonyms(word_or_vec, num)
+        return pd.DataFrame(rows, columns=["word", "similarity"])
+
     def analogy(self, pos: List[str], neg: List[str], num: int = 10) -> List[tuple]:
         """wien - oesterreich + deutschland -> berlin style queries
         (IT spec :327-382)."""

@@ -118,6 +118,14 @@ def test_fit_from_path(tmp_path):
     assert set(m.vocab.words) == {"a", "b"}
 
 
+def test_find_synonyms_df(tiny_model):
+    pd = pytest.importorskip("pandas")
+    df = tiny_model.find_synonyms_df("aa", 4)
+    assert list(df.columns) == ["word", "similarity"]   # ml:390-420 shape
+    assert len(df) == 4
+    assert df["similarity"].is_monotonic_decreasing
+
+
 def test_transform_words_batched(tiny_model):
     # mllib:529-543: iterator-of-words batched lookup
     out = tiny_model.transform_words(["aa", "bb", "aa"])

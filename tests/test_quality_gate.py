@@ -57,6 +57,23 @@ def test_get_vectors_count(de_model):
     assert len(de_model.get_vectors()) == de_model.num_words
 
 
+def test_gate_with_lut_sigmoid(de_corpus_path):
+    """Maximum-parity stack: reference window semantics (B2) + the
+    reference's 1000-entry sigmoid LUT.  The synonym gate must still hold."""
+    est = (GlintWord2Vec()
+           .setSeed(1).setStepSize(0.025).setUnigramTableSize(1_000_000)
+           .setSubsampleRatio(0.0).setNumIterations(2))
+    est.config.device = "cpu"
+    est.config.num_partitions = 1
+    est.config.window_mode = "reference"
+    est.config.sigmoid_mode = "lut"
+    m = est.fit(de_corpus_path)
+    syns = m.find_synonyms("österreich", 10)
+    assert "wien" in [w for w, _ in syns]
+    res = m.analogy(["wien", "deutschland"], ["österreich"], 10)
+    assert "berlin" in [w for w, _ in res]
+
+
 def test_save_load_preserves_gate(de_model, tmp_path):
     # Spec:137-155 (load) + synonyms on the loaded model
     p = str(tmp_path / "de_model")
