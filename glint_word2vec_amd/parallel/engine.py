@@ -165,26 +165,46 @@ def _train_sharded(cfg: Word2VecConfig, vocab: Vocabulary,
                 sent_base += max(len(offsets) - 1, 0)
                 processed += len(tokens)
     else:  # row
+        from concurrent.futures import ThreadPoolExecutor
         from .row_sharded import RowShardedSgns
         eng = RowShardedSgns(vocab.num_words, cfg.vector_size, **common)
         rng = np.random.default_rng(seed + 17 * rank)
         empty = (np.zeros(0, dtype=np.int32), np.zeros(1, dtype=np.int32))
-        for it in range(cfg.num_iterations):
-            batches = [b for i, b in enumerate(batches_fn())
-                       if i % world == rank]
-            # every rank must make the same number of collective calls
-            n_steps = len(batches)
-            if world > 1:
-                t = torch.tensor([n_steps])
-                torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
-                n_steps = int(t.item())
-            for k in range(n_steps):
-                tokens, offsets = ((batches[k].tokens, batches[k].offsets)
-                                   if k < len(batches) else empty)
-                alpha = cfg.learning_rate * max(
-                    1e-4, 1.0 - processed / (total_words // world + 1))
-                eng.train_step(tokens, offsets, alpha, cfg.window, cfg.n, rng)
-                processed += len(tokens)
+        pool = ThreadPoolExecutor(max_workers=1)
+        try:
+            for it in range(cfg.num_iterations):
+                batches = [b for i, b in enumerate(batches_fn())
+                           if i % world == rank]
+                # every rank must make the same number of collective calls
+                n_steps = len(batches)
+                if world > 1:
+                    t = torch.tensor([n_steps])
+                    torch.distributed.all_reduce(
+                        t, op=torch.distributed.ReduceOp.MAX)
+                    n_steps = int(t.item())
+
+                def batch_at(k):
+                    return ((batches[k].tokens, batches[k].offsets)
+                            if k < len(batches) else empty)
+
+                # prefetch next step's host plan while this step trains
+                next_plan = (pool.submit(eng.make_plan, *batch_at(0),
+                                         cfg.window, cfg.n, rng)
+                             if n_steps else None)
+                for k in range(n_steps):
+                    tokens, offsets = batch_at(k)
+                    plan = next_plan.result()
+                    if k + 1 < n_steps:
+                        next_plan = pool.submit(eng.make_plan,
+                                                *batch_at(k + 1),
+                                                cfg.window, cfg.n, rng)
+                    alpha = cfg.learning_rate * max(
+                        1e-4, 1.0 - processed / (total_words // world + 1))
+                    eng.train_step(tokens, offsets, alpha, cfg.window, cfg.n,
+                                   rng, plan=plan)
+                    processed += len(tokens)
+        finally:
+            pool.shutdown(wait=True)
     st = eng.read_stats()
     dt = time.time() - t0
     log.info("%s-sharded training (rank %d/%d): %d words in %.2fs, %d pairs, "

@@ -180,14 +180,23 @@ class RowShardedSgns:
             shard.index_copy_(0, idx_recv[s], cur.to(shard.dtype))
 
     # ------------------------------------------------------------------
-    def train_step(self, tokens: np.ndarray, offsets: np.ndarray,
-                   alpha: float, window: int, n_neg: int,
-                   rng: np.random.Generator) -> None:
-        """One data-parallel step over this rank's batch (host arrays).
-        Ranks with no data still participate in the collectives."""
-        plan = sgns.make_grouped_plan(tokens, offsets, self.keep_prob,
+    def make_plan(self, tokens: np.ndarray, offsets: np.ndarray,
+                  window: int, n_neg: int, rng: np.random.Generator):
+        """Host-side pair planning for one batch — independent of table
+        state, so it can be prefetched on a worker thread while the GPU
+        trains the previous step (the reference's async mini-batch workers
+        overlapping compute, SURVEY §2.2 pipeline note)."""
+        return sgns.make_grouped_plan(tokens, offsets, self.keep_prob,
                                       self.table, window, n_neg, rng,
                                       self.window_mode)
+
+    def train_step(self, tokens: np.ndarray, offsets: np.ndarray,
+                   alpha: float, window: int, n_neg: int,
+                   rng: np.random.Generator, plan=None) -> None:
+        """One data-parallel step over this rank's batch (host arrays).
+        Ranks with no data still participate in the collectives."""
+        if plan is None:
+            plan = self.make_plan(tokens, offsets, window, n_neg, rng)
         uc, inv_c = np.unique(plan.group_center, return_inverse=True)
         ut, inv_t = np.unique(plan.pair_target, return_inverse=True)
         cache0 = self.pull(uc.astype(np.int64), 0)
