@@ -167,6 +167,49 @@ def test_bf16_kernel_counts_and_direction():
     assert den > 0 and num / den > 0.98
 
 
+def test_serial_parity_fuzz():
+    """Randomized config sweep: serial GPU kernel vs oracle across dims,
+    windows, negatives, modes, subsampling — exact counts + close values."""
+    rng = np.random.default_rng(123)
+    for trial in range(10):
+        vocab = int(rng.integers(5, 400))
+        dim = int(rng.choice([7, 17, 50, 64, 129, 300]))
+        window = int(rng.integers(1, 9))
+        n_neg = int(rng.choice([0, 1, 3, 7]))
+        wm = str(rng.choice(["canonical", "reference"]))
+        sub = bool(rng.integers(0, 2))
+        n_tokens = int(rng.integers(30, 400))
+        n_sent = int(rng.integers(1, 8))
+        tokens = rng.integers(0, vocab, n_tokens).astype(np.int32)
+        offsets = np.sort(rng.choice(np.arange(1, n_tokens),
+                                     size=min(n_sent - 1, n_tokens - 1),
+                                     replace=False)).astype(np.int32)             if n_sent > 1 and n_tokens > 1 else np.zeros(0, np.int32)
+        offsets = np.concatenate([[0], offsets, [n_tokens]]).astype(np.int32)
+        counts = np.bincount(tokens, minlength=vocab).astype(np.int64) + 1
+        table = build_unigram_table(counts, int(rng.integers(50, 2000)))
+        syn0, syn1 = sgns.init_tables(vocab, dim, trial)
+        kp = (keep_probabilities(counts, int(counts.sum()), 0.02)
+              if sub else None)
+        a0, a1 = syn0.copy(), syn1.copy()
+        st_py = cpu_ref.train_batch_oracle(
+            a0, a1, tokens, offsets, kp, table, 0.04, window, n_neg,
+            seed=trial * 7, sent_id_base=trial, window_mode=wm)
+        gs = _gpu_setup(syn0, syn1, table)
+        if sub:
+            gs.set_subsample(counts, int(counts.sum()), 0.02)
+        gs.train_batch(_to_dev(tokens), _to_dev(offsets), 0.04, window,
+                       n_neg, trial * 7, sent_id_base=trial, window_mode=wm,
+                       serial=True, atomic=False)
+        torch.cuda.synchronize()
+        st = gs.read_stats()
+        ctx = f"trial={trial} vocab={vocab} dim={dim} w={window} n={n_neg} "               f"wm={wm} sub={sub}"
+        assert st.pairs == st_py.pairs, ctx
+        assert st.words_trained == st_py.words_trained, ctx
+        g0, g1 = gs.to_host()
+        np.testing.assert_allclose(g0, a0, rtol=5e-4, atol=5e-6, err_msg=ctx)
+        np.testing.assert_allclose(g1, a1, rtol=5e-4, atol=5e-6, err_msg=ctx)
+
+
 def test_pull_average_matches_numpy():
     tokens, offsets, counts, table, syn0, syn1 = _problem()
     gs = _gpu_setup(syn0, syn1, table)
