@@ -336,6 +336,41 @@ def test_estimator_german_corpus_gpu():
     assert "berlin" in [w for w, _ in res]
 
 
+def test_gpu_streaming_checkpoint_matches_to_host(tmp_path_factory):
+    """GpuSgns.save_checkpoint streams from HBM; must equal the host-side
+    matrices (bf16 table -> f32 checkpoint)."""
+    from glint_word2vec_amd.checkpoint import load_model
+    from glint_word2vec_amd.config import Word2VecConfig
+    from glint_word2vec_amd.vocab import build_vocab
+    tokens, offsets, counts, table, syn0, syn1 = _problem(vocab=37, dim=24)
+    gs = _gpu_setup(syn0, syn1, table, dtype="bfloat16")
+    gs.train_batch(_to_dev(tokens), _to_dev(offsets), 0.03, 3, 4, 7)
+    torch.cuda.synchronize()
+    vocab = build_vocab([[f"w{i}"] * (40 - i) for i in range(37)], min_count=1)
+    path = str(tmp_path_factory.mktemp("ckpt") / "m")
+    gs.save_checkpoint(path, Word2VecConfig(vector_size=24), vocab,
+                       num_shards=3)
+    _, _, s0, s1 = load_model(path)
+    h0, h1 = gs.to_host()
+    np.testing.assert_array_equal(s0, h0)
+    np.testing.assert_array_equal(s1, h1)
+
+
+def test_pull_average_and_norms_bf16():
+    tokens, offsets, counts, table, syn0, syn1 = _problem(vocab=80, dim=40)
+    gs = _gpu_setup(syn0, syn1, table, dtype="bfloat16")
+    out = gs.pull_average(_to_dev(tokens), _to_dev(offsets))
+    torch.cuda.synchronize()
+    sentences = [tokens[offsets[i]:offsets[i + 1]]
+                 for i in range(len(offsets) - 1)]
+    # reference computed on the bf16-quantized table
+    q = torch.from_numpy(syn0).bfloat16().float().numpy()
+    ref = cpu_ref.pull_average(q, sentences)
+    np.testing.assert_allclose(out.cpu().numpy(), ref, rtol=1e-5, atol=1e-6)
+    nr = gs.norms().cpu().numpy()
+    np.testing.assert_allclose(nr, cpu_ref.norms(q), rtol=1e-5, atol=1e-6)
+
+
 def test_model_gpu_serving_ops():
     """to_gpu(): device-side findSynonyms + sentence-average transform must
     agree with the host implementations."""
