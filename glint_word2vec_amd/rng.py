@@ -19,8 +19,10 @@ Draw-index layout per sentence (normative — every implementation must match):
     a draw whose table entry equals the positive target is discarded (that
     negative slot is skipped, as in canonical word2vec.c).
 
-Sentence ids must stay < 2^63; draw indices stay < 2^22 for the supported
-max_sentence_length of 1024 (window <= 2^5-ish), far below any collision.
+Sentence ids must stay < 2^63.  The three draw-index streams never
+collide for any window/negative count: subsample uses k < 1024, window
+draws use [WIN_BASE, WIN_BASE + 1024), negatives use [NEG_BASE, inf) and
+NEG_BASE > WIN_BASE + 1024.
 """
 from __future__ import annotations
 

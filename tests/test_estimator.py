@@ -150,5 +150,21 @@ def test_metadata_roundtrips_all_knobs(tiny_model, tmp_path):
     tiny_model.config.sigmoid_mode = "exact"
 
 
+def test_fit_reproducible_same_seed():
+    """Same seed, single worker -> bitwise-identical model (counter-based
+    RNG; no hidden global state)."""
+    sents = [["p", "q", "r", "p", "q"]] * 80
+    def go():
+        est = (GlintWord2Vec().setVectorSize(12).setMinCount(1).setSeed(9)
+               .setUnigramTableSize(500).setNumIterations(2)
+               .setSubsampleRatio(0.0))
+        est.config.device = "cpu"
+        est.config.num_partitions = 1
+        return est.fit(sents)
+    m1, m2 = go(), go()
+    np.testing.assert_array_equal(m1.syn0, m2.syn0)
+    np.testing.assert_array_equal(m1.syn1, m2.syn1)
+
+
 def test_stop_noop(tiny_model):
     tiny_model.stop()   # must not raise without dist initialised
