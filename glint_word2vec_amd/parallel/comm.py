@@ -77,9 +77,14 @@ def all_reduce_sum(t: torch.Tensor) -> torch.Tensor:
     return t
 
 
+_BF16_OK = True
+
+
 def bf16_collectives_ok() -> bool:
-    """bf16 tensors over collectives: RCCL yes, gloo no."""
-    return (dist.is_initialized() and dist.get_backend() == "nccl")
+    """bf16 tensors over collectives: RCCL yes, gloo no; sticky-disabled if
+    a bf16 allreduce ever fails at runtime."""
+    return (_BF16_OK and dist.is_initialized()
+            and dist.get_backend() == "nccl")
 
 
 def all_reduce_sum_compressed(t: torch.Tensor) -> torch.Tensor:
@@ -89,11 +94,17 @@ def all_reduce_sum_compressed(t: torch.Tensor) -> torch.Tensor:
     if not (dist.is_initialized() and dist.get_world_size() > 1):
         return t
     if t.dtype == torch.float32 and bf16_collectives_ok():
-        c = t.bfloat16()
-        dist.all_reduce(c, op=dist.ReduceOp.SUM)
-        t.copy_(c.float())
-    else:
-        dist.all_reduce(t, op=dist.ReduceOp.SUM)
+        try:
+            c = t.bfloat16()
+            dist.all_reduce(c, op=dist.ReduceOp.SUM)
+            t.copy_(c.float())
+            return t
+        except RuntimeError:
+            # dtype unsupported by this RCCL build: raises uniformly on all
+            # ranks before any transfer — safe to retry uncompressed
+            global _BF16_OK
+            _BF16_OK = False
+    dist.all_reduce(t, op=dist.ReduceOp.SUM)
     return t
 
 
