@@ -808,12 +808,13 @@ struct TrainPhase2 {
     for (int m = 0; m < NCH; ++m) grad[m] += __shfl_xor(grad[m], 32, 64);
     if (ATOMIC && (atomic_below < 0 || c_idx < atomic_below)) {
       if (half == 0) RowIO32<T, NCH>::atomic_add(c_ptr, grad, l32);
-    } else {
+    } else if (half == 0) {
+      // one half does the RMW — the other would duplicate identical bytes
       float cur[NCH];
       RowIO32<T, NCH>::load(c_ptr, cur, l32);
 #pragma unroll
       for (int m = 0; m < NCH; ++m) cur[m] += grad[m];
-      RowIO32<T, NCH>::store(c_ptr, cur, l32);   // both halves store same values
+      RowIO32<T, NCH>::store(c_ptr, cur, l32);
     }
     ++w_words;
   }
@@ -937,7 +938,7 @@ struct TrainPhase4 {
     }
     if (ATOMIC && (c_idx < atomic_below)) {
       if (quarter == 0) RowIO16<T, NCQ>::atomic_add(c_ptr, grad, l16);
-    } else {
+    } else if (quarter == 0) {
       float cur[NCQ];
       RowIO16<T, NCQ>::load(c_ptr, cur, l16);
 #pragma unroll
@@ -1374,7 +1375,7 @@ struct UpdateSlicePhase2 {
     for (int m = 0; m < NCH; ++m) grad[m] += __shfl_xor(grad[m], 32, 64);
     if (ATOMIC && (c_idx < atomic_below)) {
       if (half == 0) RowIO32<T, NCH>::atomic_add(c_ptr, grad, l32);
-    } else {
+    } else if (half == 0) {
       float cur[NCH];
       RowIO32<T, NCH>::load(c_ptr, cur, l32);
 #pragma unroll
