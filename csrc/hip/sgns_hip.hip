@@ -347,6 +347,102 @@ __device__ __forceinline__ float quarter_sum_f32(float v) {
   return __shfl(__int_as_float(x), 15, 16);
 }
 
+// Masked 32-lane row I/O for dim-slice tables whose width is not a
+// multiple of 64 (dim-sharded slices, e.g. dim 300 over 8 GPUs = 38 wide):
+// rows are stored at stride = round_up(width, 8) elements and lanes beyond
+// the width contribute zeros.  Only the dim-sharded phases pay the guard
+// cost; the fused train kernels keep the unmasked RowIO32.
+template <typename T, int NCH>
+struct RowIO32M;
+
+template <int NCH>
+struct RowIO32M<float, NCH> {
+  static __device__ __forceinline__ void load(const float* row, float v[NCH],
+                                              int l32, int width) {
+#pragma unroll
+    for (int q = 0; q < NCH / 2; ++q) {
+      const int idx = 64 * q + 2 * l32;
+      if (idx + 1 < width) {
+        const float2 p = *reinterpret_cast<const float2*>(row + idx);
+        v[2 * q] = p.x;
+        v[2 * q + 1] = p.y;
+      } else if (idx < width) {
+        v[2 * q] = row[idx];
+        v[2 * q + 1] = 0.0f;
+      } else {
+        v[2 * q] = 0.0f;
+        v[2 * q + 1] = 0.0f;
+      }
+    }
+    if (NCH & 1) {
+      const int idx = 32 * (NCH - 1) + l32;
+      v[NCH - 1] = idx < width ? row[idx] : 0.0f;
+    }
+  }
+  static __device__ __forceinline__ void store(float* row, const float v[NCH],
+                                               int l32, int width) {
+#pragma unroll
+    for (int q = 0; q < NCH / 2; ++q) {
+      const int idx = 64 * q + 2 * l32;
+      if (idx + 1 < width) {
+        *reinterpret_cast<float2*>(row + idx) =
+            make_float2(v[2 * q], v[2 * q + 1]);
+      } else if (idx < width) {
+        row[idx] = v[2 * q];
+      }
+    }
+    if (NCH & 1) {
+      const int idx = 32 * (NCH - 1) + l32;
+      if (idx < width) row[idx] = v[NCH - 1];
+    }
+  }
+};
+
+template <int NCH>
+struct RowIO32M<uint16_t, NCH> {
+  static __device__ __forceinline__ void load(const uint16_t* row, float v[NCH],
+                                              int l32, int width) {
+#pragma unroll
+    for (int q = 0; q < NCH / 2; ++q) {
+      const int idx = 64 * q + 2 * l32;
+      if (idx + 1 < width) {
+        const uint32_t p = *reinterpret_cast<const uint32_t*>(row + idx);
+        v[2 * q] = bf16_to_f32((uint16_t)(p & 0xFFFF));
+        v[2 * q + 1] = bf16_to_f32((uint16_t)(p >> 16));
+      } else if (idx < width) {
+        v[2 * q] = bf16_to_f32(row[idx]);
+        v[2 * q + 1] = 0.0f;
+      } else {
+        v[2 * q] = 0.0f;
+        v[2 * q + 1] = 0.0f;
+      }
+    }
+    if (NCH & 1) {
+      const int idx = 32 * (NCH - 1) + l32;
+      v[NCH - 1] = idx < width ? bf16_to_f32(row[idx]) : 0.0f;
+    }
+  }
+  static __device__ __forceinline__ void store(uint16_t* row, const float v[NCH],
+                                               int l32, int width) {
+#pragma unroll
+    for (int q = 0; q < NCH / 2; ++q) {
+      const int idx = 64 * q + 2 * l32;
+      if (idx + 1 < width) {
+        v2bf16 d;
+        d[0] = (__bf16)v[2 * q];
+        d[1] = (__bf16)v[2 * q + 1];
+        *reinterpret_cast<v2bf16*>(row + idx) = d;
+      } else if (idx < width) {
+        *reinterpret_cast<__bf16*>(row + idx) = (__bf16)v[2 * q];
+      }
+    }
+    if (NCH & 1) {
+      const int idx = 32 * (NCH - 1) + l32;
+      if (idx < width) *reinterpret_cast<__bf16*>(row + idx) = (__bf16)v[NCH - 1];
+    }
+  }
+};
+
 // Per-half (32-lane) sum: 4x row_shr + row_bcast15 leaves each half's total
 // in its lane 31/63; one width-32 shuffle broadcasts it within the half.
 __device__ __forceinline__ float half_sum_f32(float v) {
@@ -426,6 +522,7 @@ struct KernelArgs {
   int32_t atomic_below;       // rows < this use atomics (hot rows); rest plain
   const float* exp_table;     // non-null: reference LUT sigmoid parity mode
   int exp_table_size;
+  int width;                  // valid elements per row (masked dim phases)
   // stats
   unsigned long long* d_pairs;
   unsigned long long* d_positives;
@@ -1250,9 +1347,10 @@ struct DotPhase2 {
   float* f_base;
   int l32;
   int half;
+  int width;
   float c_row[NCH];
   __device__ __forceinline__ void begin_position(int32_t c) {
-    RowIO32<T, NCH>::load(syn0 + (int64_t)c * stride, c_row, l32);
+    RowIO32M<T, NCH>::load(syn0 + (int64_t)c * stride, c_row, l32, width);
   }
   __device__ __forceinline__ void process_pairs(const uint32_t* tl, int count,
                                                 int64_t idx_base) {
@@ -1262,7 +1360,7 @@ struct DotPhase2 {
       const uint32_t enc = tl[active ? my : k];
       const T* t_ptr = syn1 + (int64_t)(enc & 0x7FFFFFFFu) * stride;
       float t_row[NCH];
-      RowIO32<T, NCH>::load(t_ptr, t_row, l32);
+      RowIO32M<T, NCH>::load(t_ptr, t_row, l32, width);
       float f = 0.0f;
 #pragma unroll
       for (int m = 0; m < NCH; ++m) f += c_row[m] * t_row[m];
@@ -1292,6 +1390,7 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void dots_slice2_kernel(
     ph.f_base = f_out + pair_offsets[s];
     ph.l32 = lane & 31;
     ph.half = lane >> 5;
+    ph.width = a.width;
     const uint64_t base = sent_base(a.seed, (uint64_t)(a.sent_id_base + s));
     walk_sentence_dev(a.tokens, a.offsets[s],
                       (int)(a.offsets[s + 1] - a.offsets[s]), base, a.keep_thr,
@@ -1311,6 +1410,7 @@ struct UpdateSlicePhase2 {
   float alpha;
   int l32;
   int half;
+  int width;
   int32_t atomic_below;
   const float* exp_table;
   int exp_table_size;
@@ -1323,7 +1423,7 @@ struct UpdateSlicePhase2 {
   __device__ __forceinline__ void begin_position(int32_t c) {
     c_idx = c;
     c_ptr = syn0 + (int64_t)c * stride;
-    RowIO32<T, NCH>::load(c_ptr, c_row, l32);
+    RowIO32M<T, NCH>::load(c_ptr, c_row, l32, width);
 #pragma unroll
     for (int m = 0; m < NCH; ++m) grad[m] = 0.0f;
   }
@@ -1335,7 +1435,7 @@ struct UpdateSlicePhase2 {
       const uint32_t enc = tl[active ? my : k];
       T* t_ptr = syn1 + (int64_t)(enc & 0x7FFFFFFFu) * stride;
       float t_row[NCH];
-      RowIO32<T, NCH>::load(t_ptr, t_row, l32);
+      RowIO32M<T, NCH>::load(t_ptr, t_row, l32, width);
       float f = f_base[idx_base + (active ? my : k)];
       if (f_loc) {
         float fresh = 0.0f;
@@ -1405,6 +1505,7 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void update_slice2_kernel(
   ph.alpha = a.alpha;
   ph.l32 = lane & 31;
   ph.half = lane >> 5;
+  ph.width = a.width;
   ph.world_scale = world_scale;
   ph.atomic_below = a.atomic_below;
   ph.exp_table = a.exp_table;
@@ -1721,7 +1822,8 @@ static KernelArgs make_walk_args(uintptr_t syn0, uintptr_t syn1,
                                  int n_neg, uint64_t seed,
                                  int64_t sent_id_base, int ref_window,
                                  uintptr_t stats) {
-  if (stride % 64 != 0) throw std::runtime_error("stride must be a multiple of 64");
+  if (stride % 8 != 0)
+    throw std::runtime_error("stride must be a multiple of 8");
   if (table_size <= 0 || table_size > 0xFFFFFFFFLL)
     throw std::runtime_error("table_size out of range");
   KernelArgs a{};
@@ -1772,13 +1874,17 @@ static void dots_slice(uintptr_t syn0, uintptr_t syn1, int is_bf16,
                        int n_neg, uint64_t seed, int64_t sent_id_base,
                        int ref_window, uintptr_t pair_offsets, uintptr_t f_out,
                        int blocks, int threads, uintptr_t stream_ptr,
-                       int pair_mode) {
+                       int pair_mode, int width) {
   HIP_CLEAR_ERROR();
+  if (width <= 0) width = (int)stride;
   KernelArgs a = make_walk_args(syn0, syn1, stride, tokens, offsets,
                                 num_sentences, keep_thr, table, table_size,
                                 0.0, window, n_neg, seed, sent_id_base,
                                 ref_window, 0);
-  const int nc = (int)(stride / 64);
+  a.width = width;
+  const int nc = supported_nc((width + 63) / 64);
+  if (width != (int)stride && !(pair_mode && threads == 256))
+    throw std::runtime_error("masked width requires pair_mode kernels");
   hipStream_t stream = (hipStream_t)stream_ptr;
 #define DOTS_CASE(T, N)                                                       \
   do {                                                                        \
@@ -1818,17 +1924,21 @@ static void update_slice(uintptr_t syn0, uintptr_t syn1, int is_bf16,
                          uintptr_t stats, int blocks, int threads,
                          uintptr_t stream_ptr,
                          uintptr_t exp_table, int exp_table_size,
-                         int pair_mode) {
+                         int pair_mode, int width) {
   HIP_CLEAR_ERROR();
+  if (width <= 0) width = (int)stride;
   KernelArgs a = make_walk_args(syn0, syn1, stride, tokens, offsets,
                                 num_sentences, keep_thr, table, table_size,
                                 alpha, window, n_neg, seed, sent_id_base,
                                 ref_window, stats);
+  a.width = width;
   a.atomic_below = (int32_t)std::min<int64_t>(atomic_below, 0x7FFFFFFFLL);
   a.exp_table = (const float*)exp_table;
   a.exp_table_size = exp_table_size;
   const int atomic = atomic_below > 0;
-  const int nc = (int)(stride / 64);
+  const int nc = supported_nc((width + 63) / 64);
+  if (width != (int)stride && !(pair_mode && threads == 256))
+    throw std::runtime_error("masked width requires pair_mode kernels");
   hipStream_t stream = (hipStream_t)stream_ptr;
 #define UPD_CASE(T, N)                                                        \
   do {                                                                        \

@@ -44,6 +44,7 @@ class DimShardedSgns:
                  table_size: int = 1_000_000, subsample: float = 0.0,
                  window_mode: str = "canonical", chunk_words: int = 1 << 19,
                  f_correction: bool = True, atomic: bool = True,
+                 narrow: "bool | None" = None,
                  init_full_limit: int = 1 << 28):
         self.rank, self.world = (comm.init_from_env() if torch.distributed.is_available()
                                  else (0, 1))
@@ -62,10 +63,19 @@ class DimShardedSgns:
         if self.is_cuda:
             from .. import _hip_native
             self.native = _hip_native
-            self.stride = self.native.round_stride(max(self.width, 1))
+            # narrow storage: stride = round_up(width, 8) removes the
+            # 64-element padding waste for thin dim slices (NOTES backlog
+            # #2, measured in benchmarks/results.md); requires the pair2
+            # masked kernels, so serial-parity tests pass narrow=False
+            if narrow is None:
+                narrow = self.width % 64 != 0
+            self.narrow = bool(narrow)
+            self.stride = ((self.width + 7) // 8 * 8 if self.narrow
+                           else self.native.round_stride(max(self.width, 1)))
         else:
             from .. import _cpu_native
             self.native = _cpu_native
+            self.narrow = False
             self.stride = self.width
         tdtype = torch.bfloat16 if self.is_bf16 else torch.float32
         if self.is_bf16 and not self.is_cuda:
@@ -129,6 +139,8 @@ class DimShardedSgns:
                 chunks.append((s0, s + 1))
                 s0, acc = s + 1, 0
         if self.is_cuda:
+            if self.narrow and self.serial:
+                raise ValueError("serial parity mode requires narrow=False")
             self._train_step_gpu(tokens, offsets, chunks, alpha, window,
                                  n_neg, seed, sent_id_base)
         else:
@@ -183,7 +195,7 @@ class DimShardedSgns:
                 0 if f_loc is None else f_loc.data_ptr(), float(self.world),
                 (2 ** 31 - 1) if self.atomic else 0, self._stats.data_ptr(),
                 nb, nt, comp.cuda_stream, 0, 0,
-                0 if self.serial else 1)
+                0 if self.serial else 1, self.width)
 
         for (a, b) in chunks:
             n = b - a

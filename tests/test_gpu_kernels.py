@@ -253,7 +253,7 @@ def test_dim_sharded_gpu_world1_matches_sequential():
                                 table, 0.03, 3, 4, 42, 0, "canonical", 1)
     eng = DimShardedSgns(60, 24, device="cuda", seed=3, counts=counts,
                          table_size=1009, chunk_words=10 ** 9,
-                         f_correction=True, atomic=False)
+                         f_correction=True, atomic=False, narrow=False)
     eng.serial = True
     tok = torch.from_numpy(batch.tokens).cuda()
     off = torch.from_numpy(batch.offsets).cuda()
@@ -316,6 +316,46 @@ def test_row_sharded_gpu_world1_trains():
     p0, p1, pst = run("cuda", serial=False)
     assert pst.pairs == cst.pairs
     assert np.isfinite(p0).all() and np.isfinite(p1).all()
+
+
+def test_dim_sharded_narrow_slices_gpu():
+    """Masked narrow-slice storage (stride = round_up(width, 8)) must match
+    the padded-to-64 storage and the CPU engine.  Single-sentence batch ->
+    one active wave -> deterministic order."""
+    from glint_word2vec_amd.parallel.dim_sharded import DimShardedSgns
+    rng = np.random.default_rng(4)
+    dim = 38                       # the dim=300-over-8-GPUs slice width
+    tokens = rng.integers(0, 40, 120).astype(np.int32)
+    offsets = np.array([0, 120], dtype=np.int32)
+    counts = np.bincount(tokens, minlength=40).astype(np.int64) + 1
+
+    def run(device, narrow=None):
+        eng = DimShardedSgns(40, dim, device=device, seed=3, counts=counts,
+                             table_size=1009, chunk_words=10 ** 9,
+                             f_correction=True, atomic=False, narrow=narrow)
+        if device == "cuda":
+            assert eng.narrow == (narrow if narrow is not None else True)
+            tok = torch.from_numpy(tokens).cuda()
+            off = torch.from_numpy(offsets).cuda()
+        else:
+            tok = torch.from_numpy(tokens)
+            off = torch.from_numpy(offsets)
+        eng.train_step(tok, off, 0.04, 3, 4, seed=11, offsets_host=offsets)
+        if device == "cuda":
+            torch.cuda.synchronize()
+        st = eng.read_stats()
+        s0, s1 = eng.to_host()
+        return s0, s1, st
+
+    n0, n1, nst = run("cuda", narrow=True)
+    p0, p1, pst = run("cuda", narrow=False)
+    c0, c1, cst = run("cpu")
+    assert nst.pairs == pst.pairs == cst.pairs
+    # narrow vs padded GPU: identical order, only storage differs
+    np.testing.assert_allclose(n0, p0, rtol=1e-5, atol=1e-7)
+    np.testing.assert_allclose(n1, p1, rtol=1e-5, atol=1e-7)
+    # vs CPU sequential: pair2 halves introduce adjacent-pair concurrency
+    np.testing.assert_allclose(n0, c0, rtol=2e-2, atol=1e-4)
 
 
 def test_estimator_german_corpus_gpu():
