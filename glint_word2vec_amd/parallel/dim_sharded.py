@@ -209,7 +209,7 @@ class DimShardedSgns:
                 n, kthr, self.table.data_ptr(), int(self.table.numel()),
                 window, n_neg, seed, sent_id_base + a, ref,
                 poff[a:b + 1].data_ptr(), f.data_ptr(), nb, nt,
-                comp.cuda_stream, 0 if self.serial else 1)
+                comp.cuda_stream, 0 if self.serial else 1, self.width)
             if f_loc is not None:
                 f_loc[lo:hi] = f[lo:hi]
             ev = torch.cuda.Event()
