@@ -396,6 +396,20 @@ struct RowIO32M<float, NCH> {
       if (idx < width) row[idx] = v[NCH - 1];
     }
   }
+  static __device__ __forceinline__ void atomic_add(float* row,
+                                                    const float v[NCH],
+                                                    int l32, int width) {
+#pragma unroll
+    for (int q = 0; q < NCH / 2; ++q) {
+      const int idx = 64 * q + 2 * l32;
+      if (idx < width) atomicAdd(row + idx, v[2 * q]);
+      if (idx + 1 < width) atomicAdd(row + idx + 1, v[2 * q + 1]);
+    }
+    if (NCH & 1) {
+      const int idx = 32 * (NCH - 1) + l32;
+      if (idx < width) atomicAdd(row + idx, v[NCH - 1]);
+    }
+  }
 };
 
 template <int NCH>
@@ -439,6 +453,38 @@ struct RowIO32M<uint16_t, NCH> {
     if (NCH & 1) {
       const int idx = 32 * (NCH - 1) + l32;
       if (idx < width) *reinterpret_cast<__bf16*>(row + idx) = (__bf16)v[NCH - 1];
+    }
+  }
+  static __device__ __forceinline__ void atomic_add(uint16_t* row,
+                                                    const float v[NCH],
+                                                    int l32, int width) {
+#pragma unroll
+    for (int q = 0; q < NCH / 2; ++q) {
+      const int idx = 64 * q + 2 * l32;
+      if (idx + 1 < width) {
+        v2bf16 d;
+        d[0] = (__bf16)v[2 * q];
+        d[1] = (__bf16)v[2 * q + 1];
+        __builtin_amdgcn_global_atomic_fadd_v2bf16((v2bf16*)(row + idx), d);
+      } else if (idx < width) {
+        // lone element at an even offset: pack with a zero partner
+        v2bf16 d;
+        d[0] = (__bf16)v[2 * q];
+        d[1] = (__bf16)0.0f;
+        __builtin_amdgcn_global_atomic_fadd_v2bf16((v2bf16*)(row + idx), d);
+      }
+    }
+    if (NCH & 1) {
+      const int idx = 32 * (NCH - 1) + l32;
+      if (idx < width) {
+        uint16_t* p = row + idx;
+        const bool odd = ((uintptr_t)p >> 1) & 1;
+        v2bf16 d;
+        d[0] = odd ? (__bf16)0.0f : (__bf16)v[NCH - 1];
+        d[1] = odd ? (__bf16)v[NCH - 1] : (__bf16)0.0f;
+        __builtin_amdgcn_global_atomic_fadd_v2bf16(
+            (v2bf16*)((uintptr_t)p & ~(uintptr_t)3), d);
+      }
     }
   }
 };
@@ -1456,11 +1502,11 @@ struct UpdateSlicePhase2 {
           float delta[NCH];
 #pragma unroll
           for (int m = 0; m < NCH; ++m) delta[m] = g * c_row[m];
-          RowIO32<T, NCH>::atomic_add(t_ptr, delta, l32);
+          RowIO32M<T, NCH>::atomic_add(t_ptr, delta, l32, width);
         } else {
 #pragma unroll
           for (int m = 0; m < NCH; ++m) t_row[m] += g * c_row[m];
-          RowIO32<T, NCH>::store(t_ptr, t_row, l32);
+          RowIO32M<T, NCH>::store(t_ptr, t_row, l32, width);
         }
         ++w_pairs;
         if (label > 0.5f) {
@@ -1474,13 +1520,13 @@ struct UpdateSlicePhase2 {
 #pragma unroll
     for (int m = 0; m < NCH; ++m) grad[m] += __shfl_xor(grad[m], 32, 64);
     if (ATOMIC && (c_idx < atomic_below)) {
-      if (half == 0) RowIO32<T, NCH>::atomic_add(c_ptr, grad, l32);
+      if (half == 0) RowIO32M<T, NCH>::atomic_add(c_ptr, grad, l32, width);
     } else if (half == 0) {
       float cur[NCH];
-      RowIO32<T, NCH>::load(c_ptr, cur, l32);
+      RowIO32M<T, NCH>::load(c_ptr, cur, l32, width);
 #pragma unroll
       for (int m = 0; m < NCH; ++m) cur[m] += grad[m];
-      RowIO32<T, NCH>::store(c_ptr, cur, l32);
+      RowIO32M<T, NCH>::store(c_ptr, cur, l32, width);
     }
     ++w_words;
   }

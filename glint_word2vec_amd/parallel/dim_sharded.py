@@ -195,7 +195,8 @@ class DimShardedSgns:
                 0 if f_loc is None else f_loc.data_ptr(), float(self.world),
                 (2 ** 31 - 1) if self.atomic else 0, self._stats.data_ptr(),
                 nb, nt, comp.cuda_stream, 0, 0,
-                0 if self.serial else 1, self.width)
+                0 if self.serial else 1,
+                0 if self.serial else self.width)
 
         for (a, b) in chunks:
             n = b - a
@@ -209,7 +210,8 @@ class DimShardedSgns:
                 n, kthr, self.table.data_ptr(), int(self.table.numel()),
                 window, n_neg, seed, sent_id_base + a, ref,
                 poff[a:b + 1].data_ptr(), f.data_ptr(), nb, nt,
-                comp.cuda_stream, 0 if self.serial else 1, self.width)
+                comp.cuda_stream, 0 if self.serial else 1,
+                0 if self.serial else self.width)
             if f_loc is not None:
                 f_loc[lo:hi] = f[lo:hi]
             ev = torch.cuda.Event()
