@@ -1,0 +1,80 @@
+"""Perf probe: dim-sharded kernels at the 8-GPU slice shape (dim 300 /
+8 ranks -> width 38) comparing narrow storage (stride 40) vs padded
+(stride 64) on ONE GPU.  Dist is not initialised; the slice geometry is
+overridden after construction, so the numbers measure exactly the
+per-rank kernel work of an 8-GPU run (comm excluded).
+
+Run: python benchmarks/narrow_probe.py [--vocab N] [--steps K]
+"""
+from __future__ import annotations
+
+import argparse
+import os
+import sys
+import time
+
+import numpy as np
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+from glint_word2vec_amd.data import synthetic_corpus
+from glint_word2vec_amd.parallel.dim_sharded import DimShardedSgns
+
+
+def make_engine(vocab, width, stride, counts, dtype):
+    eng = DimShardedSgns(vocab, 300, dtype=dtype, device="cuda", seed=3,
+                         counts=counts, table_size=10_000_000,
+                         subsample=1e-4, chunk_words=1 << 20,
+                         f_correction=True, atomic=False, narrow=False)
+    # override slice geometry to the 8-GPU shape (world stays 1: the
+    # allreduce is a no-op; kernel work per rank is what we measure)
+    eng.lo, eng.hi, eng.width = 0, width, width
+    eng.narrow = stride < 64
+    eng.stride = stride
+    tdtype = torch.bfloat16 if dtype == "bfloat16" else torch.float32
+    eng.syn0 = torch.zeros((vocab, stride), dtype=tdtype, device="cuda")
+    eng.syn1 = torch.zeros((vocab, stride), dtype=tdtype, device="cuda")
+    eng._init_slices(3, 1 << 28)
+    return eng
+
+
+def run(eng, tokens, offsets, offs_host, steps, warmup):
+    for s in range(warmup):
+        eng.train_step(tokens, offsets, 0.025, 5, 5, seed=100 + s,
+                       offsets_host=offs_host)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for s in range(steps):
+        eng.train_step(tokens, offsets, 0.025, 5, 5, seed=200 + s,
+                       offsets_host=offs_host)
+    torch.cuda.synchronize()
+    dt = time.perf_counter() - t0
+    return int(offs_host[-1]) * steps / dt
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--vocab", type=int, default=1_000_000)
+    ap.add_argument("--words", type=int, default=2_000_000)
+    ap.add_argument("--steps", type=int, default=8)
+    ap.add_argument("--warmup", type=int, default=2)
+    ap.add_argument("--dtype", default="bfloat16")
+    args = ap.parse_args()
+
+    batch = synthetic_corpus(args.vocab, args.words, sentence_len=100)
+    counts = np.bincount(batch.tokens, minlength=args.vocab).astype(np.int64) + 1
+    tokens = torch.from_numpy(batch.tokens).cuda()
+    offsets = torch.from_numpy(batch.offsets).cuda()
+
+    for label, stride in (("narrow stride=40", 40), ("padded stride=64", 64)):
+        eng = make_engine(args.vocab, 38, stride, counts, args.dtype)
+        wps = run(eng, tokens, offsets, batch.offsets, args.steps, args.warmup)
+        print(f"{label}: {wps/1e6:.1f}M words/s (per-rank kernel rate, "
+              f"width=38 dtype={args.dtype})")
+        del eng
+        torch.cuda.empty_cache()
+
+
+if __name__ == "__main__":
+    main()

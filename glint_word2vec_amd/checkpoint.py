@@ -63,44 +63,56 @@ def save_model(path: str, config: Word2VecConfig, vocab: Vocabulary,
 def save_model_streaming(path: str, config: Word2VecConfig, vocab: Vocabulary,
                          row_block_fn, num_shards: int = 8,
                          block_rows: int = 1 << 20,
-                         has_syn1: bool = True) -> None:
+                         has_syn1: bool = True, write: bool = True) -> None:
     """Checkpoint without materialising the full matrices on host (80M-vocab
     models: 96 GB per table).  ``row_block_fn(which, r0, r1)`` returns the
     f32 [r1-r0, dim] block of syn0 (which=0) / syn1 (which=1).  Shards are
     contiguous row ranges ("row_range" layout; the reference's per-PS shard
-    files are the analog, mllib:493-498)."""
-    os.makedirs(os.path.join(path, "shards"), exist_ok=True)
+    files are the analog, mllib:493-498).
+
+    ``write=False`` performs the exact same sequence of ``row_block_fn``
+    calls but no file IO: non-zero ranks of an engine whose row_block_fn is
+    collective (dim-sharded allgather) call with write=False so every rank
+    walks the identical block schedule."""
+    if write:
+        os.makedirs(os.path.join(path, "shards"), exist_ok=True)
     V = vocab.num_words
     bounds = [min(V, s * ((V + num_shards - 1) // num_shards))
               for s in range(num_shards + 1)]
     bounds[-1] = V
     probe = row_block_fn(0, 0, min(1, V))
     dim = probe.shape[1]
-    meta = {
-        "class": "glint_word2vec_amd.GlintWord2VecModel",
-        "timestamp": int(time.time() * 1000),
-        "numWords": V,
-        "vectorSize": dim,
-        "paramMap": config.to_dict(),
-    }
-    with open(os.path.join(path, "metadata"), "w") as f:
-        json.dump(meta, f, indent=2, sort_keys=True)
-    vocab.save_words(os.path.join(path, "words"))
-    np.save(os.path.join(path, "counts.npy"), vocab.counts)
-    index = {"num_shards": num_shards, "vocab": V, "dim": dim,
-             "dtype": "float32", "layout": "row_range", "bounds": bounds,
-             "has_syn1": has_syn1}
-    with open(os.path.join(path, "shards", "index.json"), "w") as f:
-        json.dump(index, f, indent=2)
+    if write:
+        meta = {
+            "class": "glint_word2vec_amd.GlintWord2VecModel",
+            "timestamp": int(time.time() * 1000),
+            "numWords": V,
+            "vectorSize": dim,
+            "paramMap": config.to_dict(),
+        }
+        with open(os.path.join(path, "metadata"), "w") as f:
+            json.dump(meta, f, indent=2, sort_keys=True)
+        vocab.save_words(os.path.join(path, "words"))
+        np.save(os.path.join(path, "counts.npy"), vocab.counts)
+        index = {"num_shards": num_shards, "vocab": V, "dim": dim,
+                 "dtype": "float32", "layout": "row_range", "bounds": bounds,
+                 "has_syn1": has_syn1}
+        with open(os.path.join(path, "shards", "index.json"), "w") as f:
+            json.dump(index, f, indent=2)
     for which, name in ((0, "syn0"), (1, "syn1"))[:2 if has_syn1 else 1]:
         for s in range(num_shards):
-            with open(os.path.join(path, "shards", f"{name}-{s:05d}.bin"),
-                      "wb") as f:
+            f = (open(os.path.join(path, "shards", f"{name}-{s:05d}.bin"),
+                      "wb") if write else None)
+            try:
                 for r0 in range(bounds[s], bounds[s + 1], block_rows):
                     r1 = min(bounds[s + 1], r0 + block_rows)
-                    f.write(np.ascontiguousarray(
+                    block = np.ascontiguousarray(
                         row_block_fn(which, r0, r1), dtype=np.float32)
-                        .tobytes())
+                    if f is not None:
+                        f.write(block.tobytes())
+            finally:
+                if f is not None:
+                    f.close()
 
 
 def load_model(path: str) -> Tuple[Word2VecConfig, Vocabulary, np.ndarray, np.ndarray | None]:

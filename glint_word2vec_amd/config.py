@@ -79,6 +79,11 @@ class Word2VecConfig:
     chunk_words: int = 1 << 19       # dim-sharded feedback chunk
     f_correction: bool = True        # dim-sharded local-drift freshening
     sync_every: int = 4              # dp engine: steps between delta merges
+    # mid-training checkpoints every N steps (0 = off; the reference has
+    # none — saves only at the end, mllib:493-498).  Written next to the
+    # final save_path as "<save_path>-step<N>"; each is a complete
+    # loadable model directory.
+    checkpoint_every: int = 0
     # --- semantics switches ---------------------
     # The reference's subsampling is a de-facto no-op (integer-division bug,
     # mllib:375-377).  We implement the intended math; set
@@ -120,6 +125,8 @@ class Word2VecConfig:
             raise ValueError(f"unsupported engine {self.engine!r}")
         if self.sigmoid_mode not in ("exact", "lut"):
             raise ValueError(f"unsupported sigmoid_mode {self.sigmoid_mode!r}")
+        if self.checkpoint_every < 0:
+            raise ValueError("checkpoint_every must be >= 0")
 
     # -- (de)serialisation used by the checkpoint metadata ------------------
     def to_dict(self) -> dict:

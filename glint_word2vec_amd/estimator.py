@@ -75,6 +75,7 @@ class GlintWord2Vec:
     def setN(self, v):                 return self._set(n=v)
     def setSubsampleRatio(self, v):    return self._set(subsample_ratio=v)
     def setUnigramTableSize(self, v):  return self._set(unigram_table_size=v)
+    def setCheckpointEvery(self, v):   return self._set(checkpoint_every=v)
     def setNumParameterServers(self, v):
         # PS count maps to GPU shard count in this framework.
         return self._set(num_shards=v)

@@ -283,6 +283,39 @@ class DimShardedSgns:
                                    sum_fplus=0.0)
         return out
 
+    def save_checkpoint(self, path: str, config, vocab,
+                        num_shards: int = 8,
+                        block_rows: int = 1 << 20) -> None:
+        """Streamed checkpoint: row blocks are assembled from the per-rank
+        dim slices (allgather over the world) and written by rank 0 —
+        host memory stays O(block_rows * dim) instead of O(vocab * dim)
+        (NOTES backlog #5).  Collective: every rank must call this."""
+        from ..checkpoint import save_model_streaming
+        maxw = max(slice_bounds(self.dim, r, self.world)[1] -
+                   slice_bounds(self.dim, r, self.world)[0]
+                   for r in range(self.world))
+
+        def row_block(which, r0, r1):
+            t = self.syn0 if which == 0 else self.syn1
+            my = t[r0:r1, :self.width].float()
+            if self.world == 1:
+                return my.cpu().numpy()
+            padded = torch.zeros((r1 - r0, maxw), dtype=torch.float32,
+                                 device=self.device)
+            padded[:, :self.width] = my
+            gathered = [torch.empty_like(padded) for _ in range(self.world)]
+            torch.distributed.all_gather(gathered, padded)
+            full = np.empty((r1 - r0, self.dim), dtype=np.float32)
+            for r in range(self.world):
+                lo, hi = slice_bounds(self.dim, r, self.world)
+                full[:, lo:hi] = gathered[r][:, :hi - lo].cpu().numpy()
+            return full
+
+        save_model_streaming(path, config, vocab, row_block,
+                             num_shards=num_shards, block_rows=block_rows,
+                             write=self.rank == 0)
+        comm.barrier()
+
     def to_host(self) -> tuple[np.ndarray, np.ndarray]:
         """Assemble the full [vocab, dim] matrices on every rank (allgather
         of slices; small-model path for save/model ops)."""

@@ -23,6 +23,19 @@ from ..vocab import Vocabulary, build_unigram_table
 log = logging.getLogger("glint_word2vec_amd")
 
 
+def _maybe_mid_checkpoint(cfg: Word2VecConfig, save_path, step: int,
+                          save_fn) -> None:
+    """Mid-training checkpoint every cfg.checkpoint_every steps (0 = off).
+    Complete loadable model dirs at "<save_path>-step<N>".  ``save_fn`` must
+    be collective for engines whose save is (dim/dp/row: every rank calls
+    this at the same step because the step loops are lockstep)."""
+    if cfg.checkpoint_every <= 0 or not save_path or step == 0:
+        return
+    if step % cfg.checkpoint_every == 0:
+        save_fn(f"{save_path}-step{step}")
+        log.info("mid-training checkpoint at step %d", step)
+
+
 def train_gpu(cfg: Word2VecConfig, vocab: Vocabulary, batches_fn: Callable,
               seed: int, save_path=None, materialize: bool = True
               ) -> Tuple[np.ndarray, np.ndarray]:
@@ -60,6 +73,7 @@ def train_gpu(cfg: Word2VecConfig, vocab: Vocabulary, batches_fn: Callable,
     total_words = vocab.train_words_count * cfg.num_iterations
     processed = 0
     sent_base = 0
+    step = 0
     t0 = time.time()
     prev_done = None
     for it in range(cfg.num_iterations):
@@ -76,6 +90,9 @@ def train_gpu(cfg: Word2VecConfig, vocab: Vocabulary, batches_fn: Callable,
             prev_done = (tok, off)
             sent_base += batch.num_sentences
             processed += batch.num_tokens
+            step += 1
+            _maybe_mid_checkpoint(cfg, save_path, step,
+                                  lambda p: gs.save_checkpoint(p, cfg, vocab))
             if log.isEnabledFor(logging.INFO):
                 st = gs.read_stats(reset=False)
                 wps = processed / max(time.time() - t0, 1e-9)
@@ -116,6 +133,7 @@ def _train_sharded(cfg: Word2VecConfig, vocab: Vocabulary,
     total_words = vocab.train_words_count * cfg.num_iterations
     processed = 0
     sent_base = 0
+    step = 0
     t0 = time.time()
     if engine == "dim":
         from .dim_sharded import DimShardedSgns
@@ -134,6 +152,10 @@ def _train_sharded(cfg: Word2VecConfig, vocab: Vocabulary,
                                offsets_host=batch.offsets)
                 sent_base += batch.num_sentences
                 processed += batch.num_tokens
+                step += 1
+                _maybe_mid_checkpoint(
+                    cfg, save_path, step,
+                    lambda p: eng.save_checkpoint(p, cfg, vocab))
     elif engine == "dp":
         from .replicated import ReplicatedSgns
         eng = ReplicatedSgns(vocab.num_words, cfg.vector_size,
@@ -164,6 +186,10 @@ def _train_sharded(cfg: Word2VecConfig, vocab: Vocabulary,
                                sent_id_base=sent_base)
                 sent_base += max(len(offsets) - 1, 0)
                 processed += len(tokens)
+                step += 1
+                _maybe_mid_checkpoint(
+                    cfg, save_path, step,
+                    lambda p: eng.save_checkpoint(p, cfg, vocab))
     else:  # row
         from concurrent.futures import ThreadPoolExecutor
         from .row_sharded import RowShardedSgns
@@ -203,6 +229,10 @@ def _train_sharded(cfg: Word2VecConfig, vocab: Vocabulary,
                     eng.train_step(tokens, offsets, alpha, cfg.window, cfg.n,
                                    rng, plan=plan)
                     processed += len(tokens)
+                    step += 1
+                    _maybe_mid_checkpoint(
+                        cfg, save_path, step,
+                        lambda p: eng.save_checkpoint(p, cfg, vocab))
         finally:
             pool.shutdown(wait=True)
     st = eng.read_stats()
@@ -214,7 +244,8 @@ def _train_sharded(cfg: Word2VecConfig, vocab: Vocabulary,
         if hasattr(eng, "save_checkpoint"):
             eng.save_checkpoint(save_path, cfg, vocab)
         else:
-            # dim/dp engines: assemble on host (round-2: per-slice streaming)
+            # defensive fallback (all current engines stream their own
+            # checkpoints via save_checkpoint)
             s0, s1 = eng.to_host()
             if rank == 0:
                 from ..checkpoint import save_model

@@ -123,6 +123,27 @@ class ReplicatedSgns:
                                    sum_fplus=0.0)
         return out
 
+    def save_checkpoint(self, path: str, config, vocab,
+                        num_shards: int = 8,
+                        block_rows: int = 1 << 20) -> None:
+        """Streamed checkpoint from the fp32 masters (replicated — rank 0
+        writes, host memory O(block_rows * dim)).  Collective: sync() flushes
+        pending deltas on every rank first."""
+        self.sync()
+        if self.rank == 0:
+            from ..checkpoint import save_model_streaming
+
+            def row_block(which, r0, r1):
+                m = self.master0 if which == 0 else self.master1
+                blk = m[r0:r1, :self.dim]
+                return (blk.cpu().numpy() if self.is_cuda
+                        else blk.numpy())
+
+            save_model_streaming(path, config, vocab, row_block,
+                                 num_shards=num_shards,
+                                 block_rows=block_rows)
+        comm.barrier()
+
     def to_host(self):
         self.sync()
         d = self.dim

@@ -51,3 +51,80 @@ def test_word2vec_text_format(tmp_path):
     lines = open(p).read().strip().split("\n")
     assert lines[0] == "2 2"
     assert lines[1].startswith("a 1 2")
+
+
+def test_dim_engine_streaming_save(tmp_path):
+    """DimShardedSgns.save_checkpoint streams row blocks (allgathered dim
+    slices) without full host materialisation; the written checkpoint must
+    equal to_host()."""
+    import torch
+    from glint_word2vec_amd.parallel.dim_sharded import DimShardedSgns
+
+    rng = np.random.default_rng(7)
+    v, _, _ = _make(vocab_size=23, dim=10)
+    eng = DimShardedSgns(23, 10, device="cpu", seed=5,
+                         counts=v.counts, table_size=997)
+    tokens = torch.from_numpy(rng.integers(0, 23, 200).astype(np.int32))
+    offsets = torch.from_numpy(np.array([0, 100, 200], dtype=np.int32))
+    eng.train_step(tokens, offsets, 0.03, 3, 4, seed=2,
+                   offsets_host=offsets.numpy())
+    cfg = Word2VecConfig(vector_size=10)
+    eng.save_checkpoint(str(tmp_path / "m"), cfg, v, num_shards=3,
+                        block_rows=7)   # forces multiple blocks per shard
+    _, v2, s0, s1 = load_model(str(tmp_path / "m"))
+    h0, h1 = eng.to_host()
+    assert v2.words == v.words
+    np.testing.assert_array_equal(s0, h0)
+    np.testing.assert_array_equal(s1, h1)
+
+
+def test_dp_engine_streaming_save(tmp_path):
+    """ReplicatedSgns.save_checkpoint streams from the fp32 master."""
+    import torch
+    from glint_word2vec_amd.parallel.replicated import ReplicatedSgns
+
+    rng = np.random.default_rng(8)
+    v, _, _ = _make(vocab_size=17, dim=8)
+    eng = ReplicatedSgns(17, 8, device="cpu", seed=5, counts=v.counts,
+                         table_size=997)
+    tokens = rng.integers(0, 17, 150).astype(np.int32)
+    offsets = np.array([0, 150], dtype=np.int32)
+    eng.train_step(tokens, offsets, 0.03, 3, 4, seed=2)
+    cfg = Word2VecConfig(vector_size=8)
+    eng.save_checkpoint(str(tmp_path / "m"), cfg, v, num_shards=2,
+                        block_rows=5)
+    _, _, s0, s1 = load_model(str(tmp_path / "m"))
+    h0, h1 = eng.to_host()
+    np.testing.assert_array_equal(s0, h0)
+    np.testing.assert_array_equal(s1, h1)
+
+
+def test_mid_training_checkpoints(tmp_path):
+    """checkpoint_every=N writes complete loadable models at
+    <save_path>-step<k*N> during training (engine loops, engine.py)."""
+    from glint_word2vec_amd import GlintWord2Vec, GlintWord2VecModel
+
+    rng = np.random.default_rng(9)
+    sents = [[f"w{rng.integers(0, 15)}" for _ in range(20)]
+             for _ in range(30)]
+    est = (GlintWord2Vec().setVectorSize(8).setMinCount(1).setSeed(3)
+           .setNumIterations(1).setWindowSize(2).setN(3)
+           .setUnigramTableSize(10000).setSubsampleRatio(0.0)
+           .setCheckpointEvery(2))
+    est.config.device = "cpu"
+    est.config.engine = "dim"          # CPU-capable engine loop
+    est.config.words_per_step = 150    # ~600 tokens -> 4 steps
+    final = str(tmp_path / "model")
+    model = est.fit(sents, save_path=final)
+    mid = tmp_path / "model-step2"
+    assert mid.is_dir(), "mid-training checkpoint missing"
+    m2 = GlintWord2VecModel.load(str(mid))
+    assert m2.num_words == model.num_words
+    v2 = m2.to_local().vectors
+    assert np.isfinite(v2).all()
+    mfinal = GlintWord2VecModel.load(final)
+    np.testing.assert_allclose(mfinal.to_local().vectors,
+                               model.to_local().vectors,
+                               rtol=1e-6, atol=1e-7)
+    # the mid checkpoint differs from the final state (training continued)
+    assert not np.array_equal(v2, mfinal.to_local().vectors)
