@@ -214,6 +214,20 @@ def train_plan_minibatched(syn0: np.ndarray, syn1: np.ndarray, plan: PairPlan,
     mllib:392-433).  Returns (num_positive_pairs, sum_f_plus) for the
     divergence canary (mllib:411-412)."""
     import torch
+    # intra-op thread-pool sync costs ~ms per op under contention — a 500x
+    # slowdown for tiny tables.  Single-thread below a work threshold.
+    prev_nt = torch.get_num_threads()
+    if syn0.size < (1 << 18):
+        torch.set_num_threads(1)
+    try:
+        return _train_plan_minibatched_impl(syn0, syn1, plan, alpha,
+                                            minibatch)
+    finally:
+        torch.set_num_threads(prev_nt)
+
+
+def _train_plan_minibatched_impl(syn0, syn1, plan, alpha, minibatch):
+    import torch
     s0 = torch.from_numpy(syn0)
     s1 = torch.from_numpy(syn1)
     c_all = torch.from_numpy(plan.center.astype(np.int64))
