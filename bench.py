@@ -169,8 +169,14 @@ def main():
             trainer.train_step(tok, off, alpha, args.window, args.neg,
                                seed=99 + rank, sent_id_base=i * nsent)
         elif dist_mode == "row":
-            trainer.train_step(batch.tokens, batch.offsets, alpha,
-                               args.window, args.neg, row_rng)
+            if trainer.is_cuda:
+                plan = trainer.make_plan_device(tok, off, args.window,
+                                                args.neg, 99 + rank + 7919 * i)
+                trainer.train_step(batch.tokens, batch.offsets, alpha,
+                                   args.window, args.neg, row_rng, plan=plan)
+            else:
+                trainer.train_step(batch.tokens, batch.offsets, alpha,
+                                   args.window, args.neg, row_rng)
         elif dist_mode:
             trainer.train_step(tok, off, alpha, args.window, args.neg,
                                seed=99, sent_id_base=i * nsent,

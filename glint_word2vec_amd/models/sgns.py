@@ -252,3 +252,122 @@ def _train_plan_minibatched_impl(syn0, syn1, plan, alpha, minibatch):
         n_pos += int(pos_mask.sum())
         sum_fplus += float(f[pos_mask].sum())
     return n_pos, sum_fplus
+
+
+# ---------------------------------------------------------------------------
+# Device-side planning (torch ops — GPU-resident in production, CPU in tests)
+# ---------------------------------------------------------------------------
+
+@dataclass
+class GroupedPlanT:
+    """GroupedPlan twin with torch tensors, produced entirely with tensor
+    ops on the training device.  The numpy planner costs ~45 s of host time
+    per 2M-word batch (benchmarks/results.md); this one runs in
+    milliseconds on the GPU and feeds the row-sharded engine without any
+    host round-trip."""
+    group_center: "object"      # int32 [G] center WORD ids (torch.Tensor)
+    group_offsets: "object"     # int64 [G+1] into pair arrays
+    pair_target: "object"       # int32 [P] target WORD ids
+    pair_label: "object"        # float32 [P]
+
+    @property
+    def num_groups(self) -> int:
+        return int(self.group_center.numel())
+
+    @property
+    def num_pairs(self) -> int:
+        return int(self.pair_target.numel())
+
+
+def make_grouped_plan_torch(tokens, offsets, keep_prob, table,
+                            window: int, n_neg: int, gen,
+                            window_mode: str = "canonical") -> GroupedPlanT:
+    """Torch twin of make_grouped_plan: identical pair-enumeration
+    semantics (subsample -> compacted sentences -> shrunk window ->
+    negatives with positive-collision drop; targets of one center
+    contiguous, ascending).  Randomness comes from ``gen``
+    (torch.Generator on the same device), so draws differ from the numpy
+    planner's — the *structure* is equivalent (tested with randomness
+    pinned) but streams are not bit-identical.
+
+    tokens int32 [T0], offsets int32/int64 [S+1], keep_prob float32
+    [vocab] or None, table int32 [table_size] — all on the same device.
+    """
+    import torch
+    dev = tokens.device
+    S = offsets.numel() - 1
+    lengths = (offsets[1:] - offsets[:-1]).long()
+    sent_id = torch.repeat_interleave(torch.arange(S, device=dev), lengths)
+    if keep_prob is not None and tokens.numel():
+        u = torch.rand(tokens.numel(), generator=gen, device=dev)
+        keep = u < keep_prob[tokens.long()]
+        toks = tokens[keep]
+        sid = sent_id[keep]
+    else:
+        toks, sid = tokens, sent_id
+    T = toks.numel()
+    ei32 = torch.zeros(0, dtype=torch.int32, device=dev)
+    empty = GroupedPlanT(ei32, torch.zeros(1, dtype=torch.int64, device=dev),
+                         ei32.clone(), torch.zeros(0, device=dev))
+    if T == 0:
+        return empty
+    cnt = torch.bincount(sid, minlength=S)
+    offs = torch.zeros(S + 1, dtype=torch.int64, device=dev)
+    torch.cumsum(cnt, 0, out=offs[1:])
+    sent_lo = offs[:-1][sid]
+    sent_hi = offs[1:][sid]
+    pos = torch.arange(T, dtype=torch.int64, device=dev)
+    if window_mode == "canonical":
+        b = torch.randint(1, window + 1, (T,), generator=gen, device=dev)
+        left, right = b, b
+    else:   # reference mode (B2): left b, right b-1, b in [0, window-1]
+        b = torch.randint(0, window, (T,), generator=gen, device=dev)
+        left = b
+        right = torch.clamp(b - 1, min=0) * (b > 0)
+    # candidate grid [T, 2W] laid out center-major with ascending target
+    # offsets — flattening row-major yields the numpy planner's
+    # (center, target)-sorted order without a sort
+    d = torch.cat([torch.arange(-window, 0, device=dev),
+                   torch.arange(1, window + 1, device=dev)])
+    tpos_grid = pos[:, None] + d[None, :]
+    is_left = d < 0
+    reach = torch.where(is_left[None, :], left[:, None], right[:, None])
+    valid = (d.abs()[None, :] <= reach) & \
+        torch.where(is_left[None, :], tpos_grid >= sent_lo[:, None],
+                    tpos_grid < sent_hi[:, None])
+    flat = valid.reshape(-1)
+    cpos = pos[:, None].expand(-1, 2 * window).reshape(-1)[flat]
+    tpos = tpos_grid.reshape(-1)[flat]
+    pt = toks[tpos].to(torch.int32)
+    P = int(pt.numel())
+    if P == 0:
+        return empty
+    if n_neg > 0:
+        nidx = torch.randint(0, table.numel(), (P, n_neg), generator=gen,
+                             device=dev)
+        negs = table[nidx].to(torch.int32)
+        nvalid = negs != pt[:, None]
+        all_target = torch.cat([pt[:, None], negs], dim=1)
+        all_label = torch.cat(
+            [torch.ones(P, 1, device=dev),
+             torch.zeros(P, n_neg, device=dev)], dim=1)
+        all_valid = torch.cat(
+            [torch.ones(P, 1, dtype=torch.bool, device=dev), nvalid], dim=1)
+        all_cpos = cpos[:, None].expand(-1, 1 + n_neg)
+        fmask = all_valid.reshape(-1)
+        pair_target = all_target.reshape(-1)[fmask]
+        pair_label = all_label.reshape(-1)[fmask]
+        pair_cpos = all_cpos.reshape(-1)[fmask]
+    else:
+        pair_target, pair_label, pair_cpos = pt, \
+            torch.ones(P, device=dev), cpos
+    # group boundaries: runs of equal center position
+    Np = pair_cpos.numel()
+    change = torch.nonzero(pair_cpos[1:] != pair_cpos[:-1]).reshape(-1) + 1
+    starts = torch.cat([torch.zeros(1, dtype=torch.int64, device=dev),
+                        change])
+    group_offsets = torch.cat(
+        [starts, torch.tensor([Np], dtype=torch.int64, device=dev)])
+    group_center = toks[pair_cpos[starts]].to(torch.int32)
+    return GroupedPlanT(group_center, group_offsets, pair_target.contiguous(),
+                        pair_label.contiguous())

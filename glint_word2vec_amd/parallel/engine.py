@@ -213,7 +213,27 @@ def _train_sharded(cfg: Word2VecConfig, vocab: Vocabulary,
                     return ((batches[k].tokens, batches[k].offsets)
                             if k < len(batches) else empty)
 
-                # prefetch next step's host plan while this step trains
+                if eng.is_cuda:
+                    # device-side planning (torch ops on the GPU):
+                    # the numpy planner costs ~45 s of host time per
+                    # 2M-word batch — on-GPU it is milliseconds
+                    for k in range(n_steps):
+                        tokens, offsets = batch_at(k)
+                        plan = eng.make_plan_device(
+                            tokens, offsets, cfg.window, cfg.n,
+                            seed + 1_000_003 * (step + 31 * rank))
+                        alpha = cfg.learning_rate * max(
+                            1e-4,
+                            1.0 - processed / (total_words // world + 1))
+                        eng.train_step(tokens, offsets, alpha, cfg.window,
+                                       cfg.n, rng, plan=plan)
+                        processed += len(tokens)
+                        step += 1
+                        _maybe_mid_checkpoint(
+                            cfg, save_path, step,
+                            lambda p: eng.save_checkpoint(p, cfg, vocab))
+                    continue
+                # CPU: prefetch next step's host plan while this step trains
                 next_plan = (pool.submit(eng.make_plan, *batch_at(0),
                                          cfg.window, cfg.n, rng)
                              if n_steps else None)

@@ -96,13 +96,20 @@ class RowShardedSgns:
     # ------------------------------------------------------------------
     # pull / push: the alltoallv exchanges (Glint pull / adjust push)
     # ------------------------------------------------------------------
-    def _route(self, ids: np.ndarray):
-        """Sort global row ids by owner.  Returns (perm, send_counts,
-        sorted_local_ids)."""
+    def _to_ids(self, ids) -> torch.Tensor:
+        """Accept numpy or torch row ids; return int64 on the engine device."""
+        if isinstance(ids, np.ndarray):
+            return torch.from_numpy(np.ascontiguousarray(ids)).long() \
+                .to(self.device)
+        return ids.long().to(self.device)
+
+    def _route(self, ids: torch.Tensor):
+        """Sort global row ids by owner.  Returns (perm, send_counts_host,
+        sorted_local_ids) — all device tensors except the host counts."""
         owner = ids % self.world
-        perm = np.argsort(owner, kind="stable")
-        send_counts = np.bincount(owner, minlength=self.world)
-        local = (ids[perm] // self.world).astype(np.int64)
+        perm = torch.argsort(owner, stable=True)
+        send_counts = torch.bincount(owner, minlength=self.world)
+        local = ids[perm] // self.world
         return perm, send_counts, local
 
     def _exchange(self, send_chunks, recv_shapes, dtype):
@@ -112,23 +119,25 @@ class RowShardedSgns:
         comm.all_to_all_v(recv, send_chunks)
         return recv
 
-    def pull(self, ids: np.ndarray, which: int) -> torch.Tensor:
-        """Gather rows `ids` (global) of syn0 (which=0) / syn1 (which=1)
-        from their owners.  Returns f32 [len(ids), cache_stride]."""
+    def pull(self, ids, which: int) -> torch.Tensor:
+        """Gather rows `ids` (global, numpy or torch) of syn0 (which=0) /
+        syn1 (which=1) from their owners.  Returns f32
+        [len(ids), cache_stride] on the engine device."""
         shard = self.syn0 if which == 0 else self.syn1
+        idx = self._to_ids(ids)
         if self.world == 1:
-            out = torch.zeros((len(ids), self.cache_stride),
+            out = torch.zeros((idx.numel(), self.cache_stride),
                               dtype=torch.float32, device=self.device)
-            idx = torch.from_numpy(ids // 1).to(self.device)
             out[:, :self.dim] = shard.index_select(0, idx).float()
             return out
-        perm, send_counts, local_sorted = self._route(ids)
-        # 1) exchange request sizes + index lists
-        sc = torch.from_numpy(send_counts.astype(np.int64))
-        rc = comm.exchange_counts(sc).numpy()
-        bounds = np.concatenate([[0], np.cumsum(send_counts)])
-        idx_send = [torch.from_numpy(local_sorted[bounds[d]:bounds[d + 1]])
-                    .to(self.device) for d in range(self.world)]
+        perm, send_counts, local_sorted = self._route(idx)
+        # 1) exchange request sizes + index lists (counts stay on the
+        # engine device: RCCL collectives need device tensors, gloo CPU)
+        rc = comm.exchange_counts(send_counts).cpu().numpy()
+        bounds = np.concatenate([[0],
+                                 np.cumsum(send_counts.cpu().numpy())])
+        idx_send = [local_sorted[bounds[d]:bounds[d + 1]].contiguous()
+                    for d in range(self.world)]
         idx_recv = self._exchange(idx_send, [int(rc[s]) for s in range(self.world)],
                                   torch.int64)
         # 2) owners gather rows, reply
@@ -140,31 +149,30 @@ class RowShardedSgns:
                 rows[:, :self.dim] = shard.index_select(0, idx_recv[s]).float()
             row_send.append(rows)
         row_recv = self._exchange(
-            row_send, [(int(send_counts[d]), self.cache_stride)
+            row_send, [(int(bounds[d + 1] - bounds[d]), self.cache_stride)
                        for d in range(self.world)], torch.float32)
         sorted_rows = torch.cat(row_recv, dim=0)
         out = torch.empty_like(sorted_rows)
-        out[torch.from_numpy(perm).to(self.device)] = sorted_rows
+        out[perm] = sorted_rows
         return out
 
-    def push_add(self, ids: np.ndarray, deltas: torch.Tensor, which: int) -> None:
+    def push_add(self, ids, deltas: torch.Tensor, which: int) -> None:
         """Scatter-add row deltas back to their owners (the adjust push)."""
         shard = self.syn0 if which == 0 else self.syn1
+        idx = self._to_ids(ids)
         if self.world == 1:
-            idx = torch.from_numpy(ids // 1).to(self.device)
             upd = shard.index_select(0, idx).float() + deltas[:, :self.dim]
             shard.index_copy_(0, idx, upd.to(shard.dtype))
             return
-        perm, send_counts, local_sorted = self._route(ids)
-        sc = torch.from_numpy(send_counts.astype(np.int64))
-        rc = comm.exchange_counts(sc).numpy()
-        bounds = np.concatenate([[0], np.cumsum(send_counts)])
-        idx_send = [torch.from_numpy(local_sorted[bounds[d]:bounds[d + 1]])
-                    .to(self.device) for d in range(self.world)]
+        perm, send_counts, local_sorted = self._route(idx)
+        rc = comm.exchange_counts(send_counts).cpu().numpy()
+        bounds = np.concatenate([[0],
+                                 np.cumsum(send_counts.cpu().numpy())])
+        idx_send = [local_sorted[bounds[d]:bounds[d + 1]].contiguous()
+                    for d in range(self.world)]
         idx_recv = self._exchange(idx_send, [int(rc[s]) for s in range(self.world)],
                                   torch.int64)
-        perm_t = torch.from_numpy(perm).to(self.device)
-        deltas_sorted = deltas.index_select(0, perm_t)
+        deltas_sorted = deltas.index_select(0, perm)
         del_send = [deltas_sorted[bounds[d]:bounds[d + 1]].contiguous()
                     for d in range(self.world)]
         del_recv = self._exchange(
@@ -190,13 +198,51 @@ class RowShardedSgns:
                                       self.table, window, n_neg, rng,
                                       self.window_mode)
 
+    def make_plan_device(self, tokens, offsets, window: int, n_neg: int,
+                         seed: int):
+        """Device-side planning (torch ops, models/sgns.py
+        make_grouped_plan_torch): ~45 s of numpy host time per 2M-word
+        batch becomes milliseconds on the GPU.  ``tokens``/``offsets``
+        numpy or torch; deterministic per (engine device, seed)."""
+        if not hasattr(self, "_table_t"):
+            self._table_t = torch.from_numpy(self.table).to(self.device)
+            self._keep_prob_t = (None if self.keep_prob is None else
+                                 torch.from_numpy(self.keep_prob)
+                                 .to(self.device))
+            self._gen = torch.Generator(device=self.device)
+        tok = (torch.from_numpy(tokens) if isinstance(tokens, np.ndarray)
+               else tokens).to(self.device)
+        off = (torch.from_numpy(offsets) if isinstance(offsets, np.ndarray)
+               else offsets).to(self.device)
+        self._gen.manual_seed(int(seed) & 0x7FFFFFFFFFFFFFFF)
+        return sgns.make_grouped_plan_torch(tok, off, self._keep_prob_t,
+                                            self._table_t, window, n_neg,
+                                            self._gen, self.window_mode)
+
     def train_step(self, tokens: np.ndarray, offsets: np.ndarray,
                    alpha: float, window: int, n_neg: int,
                    rng: np.random.Generator, plan=None) -> None:
-        """One data-parallel step over this rank's batch (host arrays).
+        """One data-parallel step over this rank's batch.  ``plan`` may be
+        a host GroupedPlan or a device GroupedPlanT (make_plan_device).
         Ranks with no data still participate in the collectives."""
         if plan is None:
             plan = self.make_plan(tokens, offsets, window, n_neg, rng)
+        if isinstance(plan, sgns.GroupedPlanT):
+            uc, inv_c = torch.unique(plan.group_center.long(),
+                                     return_inverse=True)
+            ut, inv_t = torch.unique(plan.pair_target.long(),
+                                     return_inverse=True)
+            cache0 = self.pull(uc, 0)
+            cache1 = self.pull(ut, 1)
+            orig0 = cache0.clone()
+            orig1 = cache1.clone()
+            if plan.num_pairs > 0:
+                self._train_pairs_t(cache0, cache1, inv_c.int(),
+                                    plan.group_offsets, inv_t.int(),
+                                    plan.pair_label, alpha)
+            self.push_add(uc, cache0 - orig0, 0)
+            self.push_add(ut, cache1 - orig1, 1)
+            return
         uc, inv_c = np.unique(plan.group_center, return_inverse=True)
         ut, inv_t = np.unique(plan.pair_target, return_inverse=True)
         cache0 = self.pull(uc.astype(np.int64), 0)
@@ -209,6 +255,35 @@ class RowShardedSgns:
                               inv_t.astype(np.int32), plan.pair_label, alpha)
         self.push_add(uc.astype(np.int64), cache0 - orig0, 0)
         self.push_add(ut.astype(np.int64), cache1 - orig1, 1)
+
+    def _train_pairs_t(self, cache0, cache1, group_center, group_offsets,
+                       pair_target, pair_label, alpha):
+        """Device-tensor twin of _train_pairs: no host conversion, no
+        stream sync (everything stays queued on the compute stream)."""
+        if self.is_cuda:
+            G = int(group_center.numel())
+            nb = 1 if self.serial else max(1, min((G + 3) // 4, 2048))
+            nt = 64 if self.serial else 256
+            stream = torch.cuda.current_stream(self.device)
+            gc = group_center.contiguous()
+            go = group_offsets.contiguous()
+            pt = pair_target.contiguous()
+            pl = pair_label.contiguous()
+            self.native.train_pairs(
+                cache0.data_ptr(), cache1.data_ptr(), self.cache_stride,
+                gc.data_ptr(), go.data_ptr(), G, pt.data_ptr(), pl.data_ptr(),
+                float(alpha), self._stats.data_ptr(), nb, nt,
+                stream.cuda_stream)
+            # keep args alive until the kernel drains
+            self._inflight = (gc, go, pt, pl, cache0, cache1)
+        else:
+            st = self.native.train_pairs(
+                cache0.numpy(), cache1.numpy(), group_center.numpy(),
+                group_offsets.numpy(), pair_target.numpy(),
+                pair_label.numpy(), float(alpha))
+            for k in ("pairs", "positives", "words_trained"):
+                self._cpu_stats[k] += st[k]
+            self._cpu_stats["sum_fplus"] += st["sum_fplus"]
 
     def _train_pairs(self, cache0, cache1, group_center, group_offsets,
                      pair_target, pair_label, alpha):

@@ -119,3 +119,49 @@ def test_pull_roundtrip_identity():
     for i, r in enumerate(ids):
         expect[r] = s0[r] + 1.0
     np.testing.assert_allclose(s0b[ids], expect[ids], rtol=1e-6)
+
+
+def _worker_tplan(rank, world, rdv_file, out_dir):
+    """World-2, device (torch) plans: rank-0-only data — comm paths with
+    torch-routed ids must match the world-1 torch-plan run exactly."""
+    import torch.distributed as dist
+    dist.init_process_group("gloo", rank=rank, world_size=world,
+                            init_method=f"file://{rdv_file}")
+    try:
+        from glint_word2vec_amd.parallel.row_sharded import RowShardedSgns
+        batch = _make_batch()
+        counts = np.bincount(batch.tokens, minlength=VOCAB).astype(np.int64) + 1
+        eng = RowShardedSgns(VOCAB, DIM, device="cpu", seed=3, counts=counts,
+                             table_size=1009)
+        if rank == 0:
+            tokens, offsets = batch.tokens, batch.offsets
+        else:
+            tokens = np.zeros(0, dtype=np.int32)
+            offsets = np.zeros(1, dtype=np.int32)
+        plan = eng.make_plan_device(tokens, offsets, 3, 4, seed=21)
+        eng.train_step(tokens, offsets, 0.03, 3, 4,
+                       np.random.default_rng(17), plan=plan)
+        s0, s1 = eng.to_host()          # collective: every rank calls
+        if rank == 0:
+            np.save(os.path.join(out_dir, "t0.npy"), s0)
+            np.save(os.path.join(out_dir, "t1.npy"), s1)
+    finally:
+        dist.destroy_process_group()
+
+
+def test_world2_torch_plan_matches_world1(tmp_path):
+    from glint_word2vec_amd.parallel.row_sharded import RowShardedSgns
+    batch = _make_batch()
+    counts = np.bincount(batch.tokens, minlength=VOCAB).astype(np.int64) + 1
+    eng = RowShardedSgns(VOCAB, DIM, device="cpu", seed=3, counts=counts,
+                         table_size=1009)
+    plan = eng.make_plan_device(batch.tokens, batch.offsets, 3, 4, seed=21)
+    eng.train_step(batch.tokens, batch.offsets, 0.03, 3, 4,
+                   np.random.default_rng(17), plan=plan)
+    r0, r1 = eng.to_host()
+    rdv = str(tmp_path / "rdv3")
+    mp.spawn(_worker_tplan, args=(2, rdv, str(tmp_path)), nprocs=2, join=True)
+    s0 = np.load(tmp_path / "t0.npy")
+    s1 = np.load(tmp_path / "t1.npy")
+    np.testing.assert_allclose(s0, r0, rtol=1e-5, atol=1e-7)
+    np.testing.assert_allclose(s1, r1, rtol=1e-5, atol=1e-7)
