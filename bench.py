@@ -145,7 +145,8 @@ def main():
         trainer = RowShardedSgns(args.vocab, args.dim, dtype=dtype,
                                  device=str(device), seed=1, counts=counts,
                                  table_size=args.table_size,
-                                 subsample=args.subsample)
+                                 subsample=args.subsample,
+                                 atomic=bool(args.atomic or args.atomic_below))
         dist_mode = "row"
     else:
         from glint_word2vec_amd.ops.gpu import GpuSgns

@@ -1636,6 +1636,95 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void train_pairs_kernel(
   }
 }
 
+// Two pairs per wave (32-lane halves, same structure as TrainPhase2) for
+// the pairs trainer: 2x rows in flight per wave at the same instruction
+// count.  ATOMIC=false is the hogwild variant (plain RMW on the cache,
+// same update class as the fused kernel's default); serial parity keeps
+// the one-pair kernel above.
+template <int NCH, bool ATOMIC>
+__global__ __launch_bounds__(64 * kWavesPerBlock) void train_pairs2_kernel(
+    float* __restrict__ cache0, float* __restrict__ cache1, int64_t stride,
+    const int32_t* __restrict__ group_center,
+    const int64_t* __restrict__ group_offsets, int64_t num_groups,
+    const int32_t* __restrict__ pair_target,
+    const float* __restrict__ pair_label, float alpha,
+    unsigned long long* d_pairs, unsigned long long* d_positives,
+    unsigned long long* d_words, double* d_sum_fplus) {
+  const int lane = threadIdx.x & 63;
+  const int l32 = lane & 31;
+  const int half = lane >> 5;
+  const int wave = threadIdx.x >> 6;
+  const int waves_in_block = blockDim.x >> 6;
+  const int64_t wave_gid = (int64_t)blockIdx.x * waves_in_block + wave;
+  const int64_t total_waves = (int64_t)gridDim.x * waves_in_block;
+  unsigned int w_pairs = 0, w_pos = 0;
+  unsigned long long w_words = 0;
+  float w_fplus = 0.0f;
+  for (int64_t g = wave_gid; g < num_groups; g += total_waves) {
+    float* c_ptr = cache0 + (int64_t)group_center[g] * stride;
+    float c_row[NCH], grad[NCH];
+    RowIO32<float, NCH>::load(c_ptr, c_row, l32);
+#pragma unroll
+    for (int k = 0; k < NCH; ++k) grad[k] = 0.0f;
+    const int64_t pe = group_offsets[g + 1];
+    for (int64_t p = group_offsets[g]; p < pe; p += 2) {
+      const int64_t my = p + half;
+      const bool active = my < pe;
+      const int64_t pid = active ? my : p;
+      float* t_ptr = cache1 + (int64_t)pair_target[pid] * stride;
+      float t_row[NCH];
+      RowIO32<float, NCH>::load(t_ptr, t_row, l32);
+      float f = 0.0f;
+#pragma unroll
+      for (int k = 0; k < NCH; ++k) f += c_row[k] * t_row[k];
+      f = half_sum_f32(f);
+      const float label = pair_label[pid];
+      const float g0 = (label - sigmoid_clipped(f)) * alpha;
+      const float gg = active ? g0 : 0.0f;   // idle half: zero contribution
+#pragma unroll
+      for (int k = 0; k < NCH; ++k) grad[k] += gg * t_row[k];
+      if (active) {
+        if (ATOMIC) {
+          float delta[NCH];
+#pragma unroll
+          for (int k = 0; k < NCH; ++k) delta[k] = gg * c_row[k];
+          RowIO32<float, NCH>::atomic_add(t_ptr, delta, l32);
+        } else {
+#pragma unroll
+          for (int k = 0; k < NCH; ++k) t_row[k] += gg * c_row[k];
+          RowIO32<float, NCH>::store(t_ptr, t_row, l32);
+        }
+        ++w_pairs;
+        if (label > 0.5f) {
+          ++w_pos;
+          w_fplus += f;
+        }
+      }
+    }
+#pragma unroll
+    for (int k = 0; k < NCH; ++k) grad[k] += __shfl_xor(grad[k], 32, 64);
+    if (ATOMIC) {
+      if (half == 0) RowIO32<float, NCH>::atomic_add(c_ptr, grad, l32);
+    } else if (half == 0) {
+      float cur[NCH];
+      RowIO32<float, NCH>::load(c_ptr, cur, l32);
+#pragma unroll
+      for (int k = 0; k < NCH; ++k) cur[k] += grad[k];
+      RowIO32<float, NCH>::store(c_ptr, cur, l32);
+    }
+    ++w_words;
+  }
+  const unsigned int p2 = w_pairs + __shfl_xor(w_pairs, 32, 64);
+  const unsigned int o2 = w_pos + __shfl_xor(w_pos, 32, 64);
+  const float f2 = w_fplus + __shfl_xor(w_fplus, 32, 64);
+  if (lane == 0 && d_pairs) {
+    atomicAdd(d_pairs, (unsigned long long)p2);
+    atomicAdd(d_positives, (unsigned long long)o2);
+    atomicAdd(d_words, w_words);
+    atomicAdd(d_sum_fplus, (double)f2);
+  }
+}
+
 // ---------------------------------------------------------------------------
 // pullAverage: per-sentence mean of syn0 rows (Glint pullAverage, ml:453).
 // One wave per sentence; output f32 [num_sentences][dim_out<=stride].
@@ -2030,7 +2119,8 @@ static void train_pairs(uintptr_t cache0, uintptr_t cache1, int64_t stride,
                         uintptr_t group_center, uintptr_t group_offsets,
                         int64_t num_groups, uintptr_t pair_target,
                         uintptr_t pair_label, double alpha, uintptr_t stats,
-                        int blocks, int threads, uintptr_t stream_ptr) {
+                        int blocks, int threads, uintptr_t stream_ptr,
+                        int pair_mode, int atomic) {
   HIP_CLEAR_ERROR();
   if (stride % 64 != 0) throw std::runtime_error("stride must be a multiple of 64");
   const int nc = (int)(stride / 64);
@@ -2039,14 +2129,40 @@ static void train_pairs(uintptr_t cache0, uintptr_t cache1, int64_t stride,
   switch (nc) {
 #define CASE_NC(N)                                                            \
   case N:                                                                     \
-    hipLaunchKernelGGL((train_pairs_kernel<N>), dim3(blocks), dim3(threads),  \
-                       0, stream, (float*)cache0, (float*)cache1, stride,     \
-                       (const int32_t*)group_center,                          \
-                       (const int64_t*)group_offsets, num_groups,             \
-                       (const int32_t*)pair_target, (const float*)pair_label, \
-                       (float)alpha, st ? st + 0 : nullptr,                   \
-                       st ? st + 1 : nullptr, st ? st + 2 : nullptr,          \
-                       st ? (double*)(st + 3) : nullptr);                     \
+    if (pair_mode == 0) {                                                     \
+      hipLaunchKernelGGL((train_pairs_kernel<N>), dim3(blocks),               \
+                         dim3(threads), 0, stream, (float*)cache0,            \
+                         (float*)cache1, stride,                              \
+                         (const int32_t*)group_center,                        \
+                         (const int64_t*)group_offsets, num_groups,           \
+                         (const int32_t*)pair_target,                         \
+                         (const float*)pair_label, (float)alpha,              \
+                         st ? st + 0 : nullptr, st ? st + 1 : nullptr,        \
+                         st ? st + 2 : nullptr,                               \
+                         st ? (double*)(st + 3) : nullptr);                   \
+    } else if (atomic) {                                                      \
+      hipLaunchKernelGGL((train_pairs2_kernel<2 * N, true>), dim3(blocks),    \
+                         dim3(threads), 0, stream, (float*)cache0,            \
+                         (float*)cache1, stride,                              \
+                         (const int32_t*)group_center,                        \
+                         (const int64_t*)group_offsets, num_groups,           \
+                         (const int32_t*)pair_target,                         \
+                         (const float*)pair_label, (float)alpha,              \
+                         st ? st + 0 : nullptr, st ? st + 1 : nullptr,        \
+                         st ? st + 2 : nullptr,                               \
+                         st ? (double*)(st + 3) : nullptr);                   \
+    } else {                                                                  \
+      hipLaunchKernelGGL((train_pairs2_kernel<2 * N, false>), dim3(blocks),   \
+                         dim3(threads), 0, stream, (float*)cache0,            \
+                         (float*)cache1, stride,                              \
+                         (const int32_t*)group_center,                        \
+                         (const int64_t*)group_offsets, num_groups,           \
+                         (const int32_t*)pair_target,                         \
+                         (const float*)pair_label, (float)alpha,              \
+                         st ? st + 0 : nullptr, st ? st + 1 : nullptr,        \
+                         st ? st + 2 : nullptr,                               \
+                         st ? (double*)(st + 3) : nullptr);                   \
+    }                                                                         \
     break;
     FOR_EACH_NC(CASE_NC)
 #undef CASE_NC

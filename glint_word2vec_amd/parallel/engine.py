@@ -193,7 +193,8 @@ def _train_sharded(cfg: Word2VecConfig, vocab: Vocabulary,
     else:  # row
         from concurrent.futures import ThreadPoolExecutor
         from .row_sharded import RowShardedSgns
-        eng = RowShardedSgns(vocab.num_words, cfg.vector_size, **common)
+        eng = RowShardedSgns(vocab.num_words, cfg.vector_size,
+                             atomic=cfg.atomic_updates, **common)
         rng = np.random.default_rng(seed + 17 * rank)
         empty = (np.zeros(0, dtype=np.int32), np.zeros(1, dtype=np.int32))
         pool = ThreadPoolExecutor(max_workers=1)

@@ -40,7 +40,7 @@ class RowShardedSgns:
                  device: str = "cuda", seed: int = 1,
                  counts: Optional[np.ndarray] = None,
                  table_size: int = 1_000_000, subsample: float = 0.0,
-                 window_mode: str = "canonical",
+                 window_mode: str = "canonical", atomic: bool = True,
                  init_full_limit: int = 1 << 28):
         self.rank, self.world = comm.init_from_env()
         self.vocab_size = vocab_size
@@ -49,6 +49,10 @@ class RowShardedSgns:
         self.is_cuda = self.device.type == "cuda"
         self.is_bf16 = dtype == "bfloat16"
         self.window_mode = window_mode
+        # cache-update mode for the GPU pairs kernel: True = fp32 atomics
+        # (no lost updates; the PS adjust semantics), False = hogwild RMW
+        # (the fused kernel's default class; ~2.6x faster)
+        self.atomic = atomic
         if self.is_cuda:
             from .. import _hip_native
             self.native = _hip_native
@@ -273,7 +277,9 @@ class RowShardedSgns:
                 cache0.data_ptr(), cache1.data_ptr(), self.cache_stride,
                 gc.data_ptr(), go.data_ptr(), G, pt.data_ptr(), pl.data_ptr(),
                 float(alpha), self._stats.data_ptr(), nb, nt,
-                stream.cuda_stream)
+                stream.cuda_stream,
+                0 if (self.serial or self.atomic) else 1,
+                int(self.atomic))
             # keep args alive until the kernel drains
             self._inflight = (gc, go, pt, pl, cache0, cache1)
         else:
@@ -300,7 +306,9 @@ class RowShardedSgns:
                 cache0.data_ptr(), cache1.data_ptr(), self.cache_stride,
                 gc.data_ptr(), go.data_ptr(), G, pt.data_ptr(), pl.data_ptr(),
                 float(alpha), self._stats.data_ptr(), nb, nt,
-                stream.cuda_stream)
+                stream.cuda_stream,
+                0 if (self.serial or self.atomic) else 1,
+                int(self.atomic))
             torch.cuda.current_stream(self.device).synchronize()
         else:
             st = self.native.train_pairs(cache0.numpy(), cache1.numpy(),
