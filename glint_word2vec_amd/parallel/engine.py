@@ -51,7 +51,15 @@ def train_gpu(cfg: Word2VecConfig, vocab: Vocabulary, batches_fn: Callable,
     rank, world = comm.init_from_env()
     engine = cfg.engine
     if engine == "auto":
-        engine = "fused" if world == 1 else "dim"
+        if world == 1:
+            engine = "fused"
+        else:
+            # same policy as bench.py: replicate while both tables sit
+            # comfortably in 288 GB HBM3E (delta-allreduce dp is the
+            # fastest engine), shard by dimension beyond that
+            dtype_bytes = 4 if cfg.dtype == "float32" else 2
+            table_bytes = 2 * vocab.num_words * cfg.vector_size * dtype_bytes
+            engine = "dp" if table_bytes <= (8 << 30) else "dim"
     if engine in ("dim", "row", "dp") and world >= 1:
         return _train_sharded(cfg, vocab, batches_fn, seed, engine, rank,
                               world, save_path, materialize)
