@@ -132,7 +132,10 @@ class RowShardedSgns:
         if self.world == 1:
             out = torch.zeros((idx.numel(), self.cache_stride),
                               dtype=torch.float32, device=self.device)
-            out[:, :self.dim] = shard.index_select(0, idx).float()
+            B = 1 << 22   # blockwise: bounds the f32 temp at ~7 GB
+            for i in range(0, idx.numel(), B):
+                out[i:i + B, :self.dim] = \
+                    shard.index_select(0, idx[i:i + B]).float()
             return out
         perm, send_counts, local_sorted = self._route(idx)
         # 1) exchange request sizes + index lists (counts stay on the
@@ -238,27 +241,47 @@ class RowShardedSgns:
                                      return_inverse=True)
             cache0 = self.pull(uc, 0)
             cache1 = self.pull(ut, 1)
-            orig0 = cache0.clone()
-            orig1 = cache1.clone()
+            # world 1: the trained cache IS the new row value — write it
+            # back directly and skip the orig clones + delta temps
+            # (3 extra cache-sized buffers; matters at 80M-vocab scale)
+            local = self.world == 1
+            orig0 = None if local else cache0.clone()
+            orig1 = None if local else cache1.clone()
             if plan.num_pairs > 0:
                 self._train_pairs_t(cache0, cache1, inv_c.int(),
                                     plan.group_offsets, inv_t.int(),
                                     plan.pair_label, alpha)
-            self.push_add(uc, cache0 - orig0, 0)
-            self.push_add(ut, cache1 - orig1, 1)
+            if local:
+                self._write_back(uc, cache0, 0)
+                self._write_back(ut, cache1, 1)
+            else:
+                self.push_add(uc, cache0 - orig0, 0)
+                self.push_add(ut, cache1 - orig1, 1)
             return
         uc, inv_c = np.unique(plan.group_center, return_inverse=True)
         ut, inv_t = np.unique(plan.pair_target, return_inverse=True)
         cache0 = self.pull(uc.astype(np.int64), 0)
         cache1 = self.pull(ut.astype(np.int64), 1)
-        orig0 = cache0.clone()
-        orig1 = cache1.clone()
+        local = self.world == 1
+        orig0 = None if local else cache0.clone()
+        orig1 = None if local else cache1.clone()
         if plan.num_pairs > 0:
             self._train_pairs(cache0, cache1, inv_c.astype(np.int32),
                               plan.group_offsets,
                               inv_t.astype(np.int32), plan.pair_label, alpha)
-        self.push_add(uc.astype(np.int64), cache0 - orig0, 0)
-        self.push_add(ut.astype(np.int64), cache1 - orig1, 1)
+        if local:
+            self._write_back(self._to_ids(uc.astype(np.int64)), cache0, 0)
+            self._write_back(self._to_ids(ut.astype(np.int64)), cache1, 1)
+        else:
+            self.push_add(uc.astype(np.int64), cache0 - orig0, 0)
+            self.push_add(ut.astype(np.int64), cache1 - orig1, 1)
+
+    def _write_back(self, ids: torch.Tensor, rows: torch.Tensor,
+                    which: int) -> None:
+        """World-1 fast path: replace owned rows with the trained cache
+        (ids unique, so copy == add-delta)."""
+        shard = self.syn0 if which == 0 else self.syn1
+        shard.index_copy_(0, ids, rows[:, :self.dim].to(shard.dtype))
 
     def _train_pairs_t(self, cache0, cache1, group_center, group_offsets,
                        pair_target, pair_label, alpha):
