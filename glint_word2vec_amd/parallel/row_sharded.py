@@ -235,9 +235,10 @@ class RowShardedSgns:
         if plan is None:
             plan = self.make_plan(tokens, offsets, window, n_neg, rng)
         if isinstance(plan, sgns.GroupedPlanT):
-            uc, inv_c = torch.unique(plan.group_center.long(),
+            # unique on int32 ids: half the radix-sort bytes of .long()
+            uc, inv_c = torch.unique(plan.group_center,
                                      return_inverse=True)
-            ut, inv_t = torch.unique(plan.pair_target.long(),
+            ut, inv_t = torch.unique(plan.pair_target,
                                      return_inverse=True)
             cache0 = self.pull(uc, 0)
             cache1 = self.pull(ut, 1)
@@ -281,7 +282,8 @@ class RowShardedSgns:
         """World-1 fast path: replace owned rows with the trained cache
         (ids unique, so copy == add-delta)."""
         shard = self.syn0 if which == 0 else self.syn1
-        shard.index_copy_(0, ids, rows[:, :self.dim].to(shard.dtype))
+        shard.index_copy_(0, self._to_ids(ids),
+                          rows[:, :self.dim].to(shard.dtype))
 
     def _train_pairs_t(self, cache0, cache1, group_center, group_offsets,
                        pair_target, pair_label, alpha):
