@@ -1640,10 +1640,12 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void train_pairs_kernel(
 // the pairs trainer: 2x rows in flight per wave at the same instruction
 // count.  ATOMIC=false is the hogwild variant (plain RMW on the cache,
 // same update class as the fused kernel's default); serial parity keeps
-// the one-pair kernel above.
-template <int NCH, bool ATOMIC>
+// the one-pair kernel above.  T = float (world>1 delta caches) or
+// uint16_t/bf16 (world-1 native-dtype cache: half the bytes, the fused
+// kernel's precision class).
+template <typename T, int NCH, bool ATOMIC>
 __global__ __launch_bounds__(64 * kWavesPerBlock) void train_pairs2_kernel(
-    float* __restrict__ cache0, float* __restrict__ cache1, int64_t stride,
+    T* __restrict__ cache0, T* __restrict__ cache1, int64_t stride,
     const int32_t* __restrict__ group_center,
     const int64_t* __restrict__ group_offsets, int64_t num_groups,
     const int32_t* __restrict__ pair_target,
@@ -1661,9 +1663,9 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void train_pairs2_kernel(
   unsigned long long w_words = 0;
   float w_fplus = 0.0f;
   for (int64_t g = wave_gid; g < num_groups; g += total_waves) {
-    float* c_ptr = cache0 + (int64_t)group_center[g] * stride;
+    T* c_ptr = cache0 + (int64_t)group_center[g] * stride;
     float c_row[NCH], grad[NCH];
-    RowIO32<float, NCH>::load(c_ptr, c_row, l32);
+    RowIO32<T, NCH>::load(c_ptr, c_row, l32);
 #pragma unroll
     for (int k = 0; k < NCH; ++k) grad[k] = 0.0f;
     const int64_t pe = group_offsets[g + 1];
@@ -1671,9 +1673,9 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void train_pairs2_kernel(
       const int64_t my = p + half;
       const bool active = my < pe;
       const int64_t pid = active ? my : p;
-      float* t_ptr = cache1 + (int64_t)pair_target[pid] * stride;
+      T* t_ptr = cache1 + (int64_t)pair_target[pid] * stride;
       float t_row[NCH];
-      RowIO32<float, NCH>::load(t_ptr, t_row, l32);
+      RowIO32<T, NCH>::load(t_ptr, t_row, l32);
       float f = 0.0f;
 #pragma unroll
       for (int k = 0; k < NCH; ++k) f += c_row[k] * t_row[k];
@@ -1688,11 +1690,11 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void train_pairs2_kernel(
           float delta[NCH];
 #pragma unroll
           for (int k = 0; k < NCH; ++k) delta[k] = gg * c_row[k];
-          RowIO32<float, NCH>::atomic_add(t_ptr, delta, l32);
+          RowIO32<T, NCH>::atomic_add(t_ptr, delta, l32);
         } else {
 #pragma unroll
           for (int k = 0; k < NCH; ++k) t_row[k] += gg * c_row[k];
-          RowIO32<float, NCH>::store(t_ptr, t_row, l32);
+          RowIO32<T, NCH>::store(t_ptr, t_row, l32);
         }
         ++w_pairs;
         if (label > 0.5f) {
@@ -1704,13 +1706,13 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void train_pairs2_kernel(
 #pragma unroll
     for (int k = 0; k < NCH; ++k) grad[k] += __shfl_xor(grad[k], 32, 64);
     if (ATOMIC) {
-      if (half == 0) RowIO32<float, NCH>::atomic_add(c_ptr, grad, l32);
+      if (half == 0) RowIO32<T, NCH>::atomic_add(c_ptr, grad, l32);
     } else if (half == 0) {
       float cur[NCH];
-      RowIO32<float, NCH>::load(c_ptr, cur, l32);
+      RowIO32<T, NCH>::load(c_ptr, cur, l32);
 #pragma unroll
       for (int k = 0; k < NCH; ++k) cur[k] += grad[k];
-      RowIO32<float, NCH>::store(c_ptr, cur, l32);
+      RowIO32<T, NCH>::store(c_ptr, cur, l32);
     }
     ++w_words;
   }
@@ -2120,9 +2122,11 @@ static void train_pairs(uintptr_t cache0, uintptr_t cache1, int64_t stride,
                         int64_t num_groups, uintptr_t pair_target,
                         uintptr_t pair_label, double alpha, uintptr_t stats,
                         int blocks, int threads, uintptr_t stream_ptr,
-                        int pair_mode, int atomic) {
+                        int pair_mode, int atomic, int is_bf16) {
   HIP_CLEAR_ERROR();
   if (stride % 64 != 0) throw std::runtime_error("stride must be a multiple of 64");
+  if (is_bf16 && pair_mode == 0)
+    throw std::runtime_error("bf16 pairs cache requires pair_mode=1");
   const int nc = (int)(stride / 64);
   hipStream_t stream = (hipStream_t)stream_ptr;
   unsigned long long* st = (unsigned long long*)stats;
@@ -2140,9 +2144,33 @@ static void train_pairs(uintptr_t cache0, uintptr_t cache1, int64_t stride,
                          st ? st + 0 : nullptr, st ? st + 1 : nullptr,        \
                          st ? st + 2 : nullptr,                               \
                          st ? (double*)(st + 3) : nullptr);                   \
+    } else if (is_bf16) {                                                     \
+      if (atomic)                                                             \
+        hipLaunchKernelGGL((train_pairs2_kernel<uint16_t, 2 * N, true>),      \
+                           dim3(blocks), dim3(threads), 0, stream,            \
+                           (uint16_t*)cache0, (uint16_t*)cache1, stride,      \
+                           (const int32_t*)group_center,                      \
+                           (const int64_t*)group_offsets, num_groups,         \
+                           (const int32_t*)pair_target,                       \
+                           (const float*)pair_label, (float)alpha,            \
+                           st ? st + 0 : nullptr, st ? st + 1 : nullptr,      \
+                           st ? st + 2 : nullptr,                             \
+                           st ? (double*)(st + 3) : nullptr);                 \
+      else                                                                    \
+        hipLaunchKernelGGL((train_pairs2_kernel<uint16_t, 2 * N, false>),     \
+                           dim3(blocks), dim3(threads), 0, stream,            \
+                           (uint16_t*)cache0, (uint16_t*)cache1, stride,      \
+                           (const int32_t*)group_center,                      \
+                           (const int64_t*)group_offsets, num_groups,         \
+                           (const int32_t*)pair_target,                       \
+                           (const float*)pair_label, (float)alpha,            \
+                           st ? st + 0 : nullptr, st ? st + 1 : nullptr,      \
+                           st ? st + 2 : nullptr,                             \
+                           st ? (double*)(st + 3) : nullptr);                 \
     } else if (atomic) {                                                      \
-      hipLaunchKernelGGL((train_pairs2_kernel<2 * N, true>), dim3(blocks),    \
-                         dim3(threads), 0, stream, (float*)cache0,            \
+      hipLaunchKernelGGL((train_pairs2_kernel<float, 2 * N, true>),           \
+                         dim3(blocks), dim3(threads), 0, stream,              \
+                         (float*)cache0,                                      \
                          (float*)cache1, stride,                              \
                          (const int32_t*)group_center,                        \
                          (const int64_t*)group_offsets, num_groups,           \
@@ -2152,8 +2180,9 @@ static void train_pairs(uintptr_t cache0, uintptr_t cache1, int64_t stride,
                          st ? st + 2 : nullptr,                               \
                          st ? (double*)(st + 3) : nullptr);                   \
     } else {                                                                  \
-      hipLaunchKernelGGL((train_pairs2_kernel<2 * N, false>), dim3(blocks),   \
-                         dim3(threads), 0, stream, (float*)cache0,            \
+      hipLaunchKernelGGL((train_pairs2_kernel<float, 2 * N, false>),          \
+                         dim3(blocks), dim3(threads), 0, stream,              \
+                         (float*)cache0,                                      \
                          (float*)cache1, stride,                              \
                          (const int32_t*)group_center,                        \
                          (const int64_t*)group_offsets, num_groups,           \

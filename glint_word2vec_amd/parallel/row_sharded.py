@@ -163,6 +163,22 @@ class RowShardedSgns:
         out[perm] = sorted_rows
         return out
 
+    def _pull_cache(self, ids, which: int) -> torch.Tensor:
+        """Training-cache pull.  World-1 bf16 hogwild keeps the cache in
+        the shard's native dtype (half the kernel bytes — the fused
+        kernel's precision class); every other case uses the f32 pull."""
+        if (self.world == 1 and self.is_cuda and self.is_bf16
+                and not self.atomic and not self.serial):
+            shard = self.syn0 if which == 0 else self.syn1
+            idx = self._to_ids(ids)
+            out = torch.zeros((idx.numel(), self.cache_stride),
+                              dtype=shard.dtype, device=self.device)
+            B = 1 << 22
+            for i in range(0, idx.numel(), B):
+                out[i:i + B, :self.dim] = shard.index_select(0, idx[i:i + B])
+            return out
+        return self.pull(ids, which)
+
     def push_add(self, ids, deltas: torch.Tensor, which: int) -> None:
         """Scatter-add row deltas back to their owners (the adjust push)."""
         shard = self.syn0 if which == 0 else self.syn1
@@ -240,8 +256,8 @@ class RowShardedSgns:
                                      return_inverse=True)
             ut, inv_t = torch.unique(plan.pair_target,
                                      return_inverse=True)
-            cache0 = self.pull(uc, 0)
-            cache1 = self.pull(ut, 1)
+            cache0 = self._pull_cache(uc, 0)
+            cache1 = self._pull_cache(ut, 1)
             # world 1: the trained cache IS the new row value — write it
             # back directly and skip the orig clones + delta temps
             # (3 extra cache-sized buffers; matters at 80M-vocab scale)
@@ -304,7 +320,7 @@ class RowShardedSgns:
                 float(alpha), self._stats.data_ptr(), nb, nt,
                 stream.cuda_stream,
                 0 if (self.serial or self.atomic) else 1,
-                int(self.atomic))
+                int(self.atomic), int(cache0.dtype == torch.bfloat16))
             # keep args alive until the kernel drains
             self._inflight = (gc, go, pt, pl, cache0, cache1)
         else:
@@ -333,7 +349,7 @@ class RowShardedSgns:
                 float(alpha), self._stats.data_ptr(), nb, nt,
                 stream.cuda_stream,
                 0 if (self.serial or self.atomic) else 1,
-                int(self.atomic))
+                int(self.atomic), 0)
             torch.cuda.current_stream(self.device).synchronize()
         else:
             st = self.native.train_pairs(cache0.numpy(), cache1.numpy(),
