@@ -105,32 +105,57 @@ class GlintWord2Vec:
 
     # ----------------------------------------------------------------------
     def fit(self, corpus: CorpusLike, save_path: Optional[str] = None,
-            materialize: bool = True) -> "GlintWord2VecModel":
+            materialize: bool = True,
+            init_from: Optional[str] = None) -> "GlintWord2VecModel":
         """Train.  ``save_path``: checkpoint straight from the engine
         (streamed from HBM / per-rank shards — required path for 80M-vocab
         models whose matrices must not materialise on host).
         ``materialize=False``: skip host assembly and return None (load
-        later with GlintWord2VecModel.load)."""
+        later with GlintWord2VecModel.load).
+        ``init_from``: resume training from a saved checkpoint — its vocab
+        and matrices seed the run; the corpus is encoded against the
+        checkpoint's vocabulary (continue training on new data)."""
         cfg = self.config
         seed = cfg.seed if cfg.seed is not None else np.random.SeedSequence().entropy % (2 ** 63)
         seed = int(seed)
-        native = None
-        if isinstance(corpus, str):
-            try:
-                from . import _cpu_native as native
-            except ImportError:
-                native = None
-        if native is not None:
-            # native corpus pipeline (csrc/cpu_sgns.cpp): count/sort/encode
-            # in C++ — same semantics as the Python path (tested)
-            words, counts, total = native.build_vocab_file(corpus, cfg.min_count)
-            vocab = Vocabulary(words=list(words), counts=counts,
-                               index={w: i for i, w in enumerate(words)},
-                               train_words_count=int(total))
+        init_tables = None
+        if init_from is not None:
+            from .checkpoint import load_model
+            ck_cfg, vocab, ck0, ck1 = load_model(init_from)
+            if ck1 is None:
+                raise ValueError("checkpoint has no syn1 — cannot resume "
+                                 "training (saved with has_syn1=False)")
+            if ck0.shape[1] != cfg.vector_size:
+                raise ValueError(
+                    f"checkpoint dim {ck0.shape[1]} != vector_size "
+                    f"{cfg.vector_size}")
+            init_tables = (ck0, ck1)
             reader = _corpus_reader(corpus, self.input_col)
+            native = None
+            if isinstance(corpus, str):
+                try:
+                    from . import _cpu_native as native
+                except ImportError:
+                    native = None
         else:
-            reader = _corpus_reader(corpus, self.input_col)
-            vocab = build_vocab(reader(), min_count=cfg.min_count)
+            native = None
+            if isinstance(corpus, str):
+                try:
+                    from . import _cpu_native as native
+                except ImportError:
+                    native = None
+            if native is not None:
+                # native corpus pipeline (csrc/cpu_sgns.cpp): count/sort/
+                # encode in C++ — same semantics as the Python path (tested)
+                words, counts, total = native.build_vocab_file(corpus,
+                                                               cfg.min_count)
+                vocab = Vocabulary(words=list(words), counts=counts,
+                                   index={w: i for i, w in enumerate(words)},
+                                   train_words_count=int(total))
+                reader = _corpus_reader(corpus, self.input_col)
+            else:
+                reader = _corpus_reader(corpus, self.input_col)
+                vocab = build_vocab(reader(), min_count=cfg.min_count)
         if vocab.num_words == 0:
             raise ValueError("empty vocabulary — corpus has no word above min_count")
         log.info("vocab: %d words, %d train words", vocab.num_words, vocab.train_words_count)
@@ -161,9 +186,11 @@ class GlintWord2Vec:
             from .parallel.engine import train_gpu
             syn0, syn1 = train_gpu(cfg, vocab, batches, seed,
                                    save_path=save_path,
-                                   materialize=materialize)
+                                   materialize=materialize,
+                                   init_tables=init_tables)
         else:
-            syn0, syn1 = self._fit_cpu(cfg, vocab, batches, seed)
+            syn0, syn1 = self._fit_cpu(cfg, vocab, batches, seed,
+                                       init_tables=init_tables)
             if save_path is not None:
                 from .checkpoint import save_model
                 save_model(save_path, cfg, vocab, syn0, syn1)
@@ -174,8 +201,13 @@ class GlintWord2Vec:
 
     # --- single-process CPU trainer (BASELINE config 1) --------------------
     def _fit_cpu(self, cfg: Word2VecConfig, vocab: Vocabulary, batches_fn,
-                 seed: int):
-        syn0, syn1 = sgns.init_tables(vocab.num_words, cfg.vector_size, seed)
+                 seed: int, init_tables=None):
+        if init_tables is not None:
+            syn0 = np.ascontiguousarray(init_tables[0], dtype=np.float32)
+            syn1 = np.ascontiguousarray(init_tables[1], dtype=np.float32)
+        else:
+            syn0, syn1 = sgns.init_tables(vocab.num_words, cfg.vector_size,
+                                          seed)
         kp = None
         if not cfg.legacy_subsample and cfg.subsample_ratio > 0:
             kp = keep_probabilities(vocab.counts, vocab.train_words_count,

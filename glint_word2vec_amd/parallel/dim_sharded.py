@@ -116,6 +116,15 @@ class DimShardedSgns:
             sl = torch.from_numpy(np.ascontiguousarray(full[:, self.lo:self.hi]))
             self.syn0[r0:r1, :self.width] = sl.to(tdtype).to(self.device)
 
+    def load_host(self, syn0, syn1) -> None:
+        """Initialise this rank's dim slice from full host f32 matrices
+        (training resume)."""
+        tdtype = self.syn0.dtype
+        for host, dev in ((syn0, self.syn0), (syn1, self.syn1)):
+            sl = torch.from_numpy(np.ascontiguousarray(
+                host[:, self.lo:self.hi], dtype=np.float32))
+            dev[:, :self.width] = sl.to(tdtype).to(self.device)
+
     # ------------------------------------------------------------------
     def train_step(self, tokens: torch.Tensor, offsets: torch.Tensor,
                    alpha: float, window: int, n_neg: int, seed: int,

@@ -37,8 +37,8 @@ def _maybe_mid_checkpoint(cfg: Word2VecConfig, save_path, step: int,
 
 
 def train_gpu(cfg: Word2VecConfig, vocab: Vocabulary, batches_fn: Callable,
-              seed: int, save_path=None, materialize: bool = True
-              ) -> Tuple[np.ndarray, np.ndarray]:
+              seed: int, save_path=None, materialize: bool = True,
+              init_tables=None) -> Tuple[np.ndarray, np.ndarray]:
     """Dispatch to the right engine: fused single-GPU kernel at world 1,
     dp/dim/row-sharded at world > 1 (one rank per GPU, launched via
     torchrun; every rank calls fit() with the same corpus).
@@ -62,11 +62,13 @@ def train_gpu(cfg: Word2VecConfig, vocab: Vocabulary, batches_fn: Callable,
             engine = "dp" if table_bytes <= (8 << 30) else "dim"
     if engine in ("dim", "row", "dp") and world >= 1:
         return _train_sharded(cfg, vocab, batches_fn, seed, engine, rank,
-                              world, save_path, materialize)
+                              world, save_path, materialize, init_tables)
     device = torch.device("cuda", torch.cuda.current_device())
     dtype = "bfloat16" if cfg.dtype == "auto" else cfg.dtype
     gs = GpuSgns(vocab.num_words, cfg.vector_size, dtype,
-                 device=str(device), seed=seed)
+                 device=str(device), seed=seed,
+                 syn0_host=None if init_tables is None else init_tables[0],
+                 syn1_host=None if init_tables is None else init_tables[1])
     if not cfg.legacy_subsample and cfg.subsample_ratio > 0:
         gs.set_subsample(vocab.counts, vocab.train_words_count,
                          cfg.subsample_ratio)
@@ -123,8 +125,8 @@ def train_gpu(cfg: Word2VecConfig, vocab: Vocabulary, batches_fn: Callable,
 
 def _train_sharded(cfg: Word2VecConfig, vocab: Vocabulary,
                    batches_fn: Callable, seed: int, engine: str, rank: int,
-                   world: int, save_path=None, materialize: bool = True
-                   ) -> Tuple[np.ndarray, np.ndarray]:
+                   world: int, save_path=None, materialize: bool = True,
+                   init_tables=None) -> Tuple[np.ndarray, np.ndarray]:
     """Multi-GPU engines; also usable on CPU (gloo) for tests.  Dim-sharded:
     every rank walks the same data (compute split by dimension).  Row-
     sharded: corpus partitioned by rank (data parallel), rows sharded."""
@@ -149,6 +151,8 @@ def _train_sharded(cfg: Word2VecConfig, vocab: Vocabulary,
                              chunk_words=cfg.chunk_words,
                              f_correction=cfg.f_correction,
                              atomic=cfg.atomic_updates, **common)
+        if init_tables is not None:
+            eng.load_host(*init_tables)
         for it in range(cfg.num_iterations):
             for batch in batches_fn():
                 alpha = cfg.learning_rate * max(
@@ -169,6 +173,8 @@ def _train_sharded(cfg: Word2VecConfig, vocab: Vocabulary,
         eng = ReplicatedSgns(vocab.num_words, cfg.vector_size,
                              sync_every=cfg.sync_every,
                              atomic=cfg.atomic_updates, **common)
+        if init_tables is not None:
+            eng.load_host(*init_tables)
         empty = (np.zeros(0, dtype=np.int32), np.zeros(1, dtype=np.int32))
         sent_base = 10_000_000 * rank   # disjoint RNG streams per rank
         for it in range(cfg.num_iterations):
@@ -203,6 +209,8 @@ def _train_sharded(cfg: Word2VecConfig, vocab: Vocabulary,
         from .row_sharded import RowShardedSgns
         eng = RowShardedSgns(vocab.num_words, cfg.vector_size,
                              atomic=cfg.atomic_updates, **common)
+        if init_tables is not None:
+            eng.load_host(*init_tables)
         rng = np.random.default_rng(seed + 17 * rank)
         empty = (np.zeros(0, dtype=np.int32), np.zeros(1, dtype=np.int32))
         pool = ThreadPoolExecutor(max_workers=1)

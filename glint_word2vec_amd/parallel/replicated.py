@@ -151,6 +151,18 @@ class ReplicatedSgns:
         if prev is not None:
             apply(prev)
 
+    def load_host(self, syn0, syn1) -> None:
+        """Initialise from full host f32 matrices (training resume)."""
+        import torch as _t
+        s0 = _t.from_numpy(np.ascontiguousarray(syn0, dtype=np.float32))
+        s1 = _t.from_numpy(np.ascontiguousarray(syn1, dtype=np.float32))
+        d = self.dim
+        self.master0[:, :d] = s0.to(self.master0.device)
+        self.master1[:, :d] = s1.to(self.master1.device)
+        self.syn0.copy_(self.master0.to(self.syn0.dtype))
+        self.syn1.copy_(self.master1.to(self.syn1.dtype))
+        self._steps_since_sync = 0
+
     # ------------------------------------------------------------------
     def read_stats(self, reset: bool = True) -> GpuStats:
         if self.is_cuda:

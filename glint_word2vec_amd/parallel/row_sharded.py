@@ -97,6 +97,14 @@ class RowShardedSgns:
             self.syn0[lo:lo + len(mine)] = \
                 torch.from_numpy(np.ascontiguousarray(mine)).to(tdtype).to(self.device)
 
+    def load_host(self, syn0, syn1) -> None:
+        """Initialise this rank's row shard from full host f32 matrices
+        (training resume)."""
+        tdtype = self.syn0.dtype
+        for host, dev in ((syn0, self.syn0), (syn1, self.syn1)):
+            mine = np.ascontiguousarray(host[self.my_rows], dtype=np.float32)
+            dev.copy_(torch.from_numpy(mine).to(tdtype).to(self.device))
+
     # ------------------------------------------------------------------
     # pull / push: the alltoallv exchanges (Glint pull / adjust push)
     # ------------------------------------------------------------------
