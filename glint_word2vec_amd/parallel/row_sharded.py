@@ -65,10 +65,14 @@ class RowShardedSgns:
                                  dtype=np.int64)
         self.shard_size = len(self.my_rows)
         tdtype = torch.bfloat16 if self.is_bf16 else torch.float32
-        self.syn0 = torch.zeros((self.shard_size, dim), dtype=tdtype,
-                                device=self.device)
-        self.syn1 = torch.zeros((self.shard_size, dim), dtype=tdtype,
-                                device=self.device)
+        # shards stored padded to the kernel stride on GPU so the pairs
+        # kernel can train DIRECTLY on them at world 1 (no pull/unique/
+        # write-back); padding columns provably stay zero
+        self.store_stride = self.cache_stride if self.is_cuda else dim
+        self.syn0 = torch.zeros((self.shard_size, self.store_stride),
+                                dtype=tdtype, device=self.device)
+        self.syn1 = torch.zeros((self.shard_size, self.store_stride),
+                                dtype=tdtype, device=self.device)
         self._init_shard(seed, init_full_limit)
         counts = (np.ones(vocab_size, dtype=np.int64) if counts is None
                   else counts)
@@ -94,7 +98,7 @@ class RowShardedSgns:
             full = (rng.random((r1 - r0, self.dim), dtype=np.float32) - 0.5) / self.dim
             mine = full[(np.arange(r0, r1) % self.world) == self.rank]
             lo = (r0 + self.world - 1 - self.rank) // self.world
-            self.syn0[lo:lo + len(mine)] = \
+            self.syn0[lo:lo + len(mine), :self.dim] = \
                 torch.from_numpy(np.ascontiguousarray(mine)).to(tdtype).to(self.device)
 
     def load_host(self, syn0, syn1) -> None:
@@ -103,7 +107,8 @@ class RowShardedSgns:
         tdtype = self.syn0.dtype
         for host, dev in ((syn0, self.syn0), (syn1, self.syn1)):
             mine = np.ascontiguousarray(host[self.my_rows], dtype=np.float32)
-            dev.copy_(torch.from_numpy(mine).to(tdtype).to(self.device))
+            dev[:, :self.dim] = \
+                torch.from_numpy(mine).to(tdtype).to(self.device)
 
     # ------------------------------------------------------------------
     # pull / push: the alltoallv exchanges (Glint pull / adjust push)
@@ -143,7 +148,7 @@ class RowShardedSgns:
             B = 1 << 22   # blockwise: bounds the f32 temp at ~7 GB
             for i in range(0, idx.numel(), B):
                 out[i:i + B, :self.dim] = \
-                    shard.index_select(0, idx[i:i + B]).float()
+                    shard.index_select(0, idx[i:i + B])[:, :self.dim].float()
             return out
         perm, send_counts, local_sorted = self._route(idx)
         # 1) exchange request sizes + index lists (counts stay on the
@@ -161,7 +166,8 @@ class RowShardedSgns:
             rows = torch.zeros((int(rc[s]), self.cache_stride),
                                dtype=torch.float32, device=self.device)
             if int(rc[s]):
-                rows[:, :self.dim] = shard.index_select(0, idx_recv[s]).float()
+                rows[:, :self.dim] = \
+                    shard.index_select(0, idx_recv[s])[:, :self.dim].float()
             row_send.append(rows)
         row_recv = self._exchange(
             row_send, [(int(bounds[d + 1] - bounds[d]), self.cache_stride)
@@ -179,11 +185,11 @@ class RowShardedSgns:
                 and not self.atomic and not self.serial):
             shard = self.syn0 if which == 0 else self.syn1
             idx = self._to_ids(ids)
-            out = torch.zeros((idx.numel(), self.cache_stride),
+            out = torch.empty((idx.numel(), self.cache_stride),
                               dtype=shard.dtype, device=self.device)
             B = 1 << 22
             for i in range(0, idx.numel(), B):
-                out[i:i + B, :self.dim] = shard.index_select(0, idx[i:i + B])
+                out[i:i + B] = shard.index_select(0, idx[i:i + B])
             return out
         return self.pull(ids, which)
 
@@ -192,8 +198,10 @@ class RowShardedSgns:
         shard = self.syn0 if which == 0 else self.syn1
         idx = self._to_ids(ids)
         if self.world == 1:
-            upd = shard.index_select(0, idx).float() + deltas[:, :self.dim]
-            shard.index_copy_(0, idx, upd.to(shard.dtype))
+            sel = shard.index_select(0, idx)
+            sel[:, :self.dim] = (sel[:, :self.dim].float()
+                                 + deltas[:, :self.dim]).to(shard.dtype)
+            shard.index_copy_(0, idx, sel)
             return
         perm, send_counts, local_sorted = self._route(idx)
         rc = comm.exchange_counts(send_counts).cpu().numpy()
@@ -213,7 +221,7 @@ class RowShardedSgns:
             if int(rc[s]) == 0:
                 continue
             cur = shard.index_select(0, idx_recv[s]).float()
-            cur += del_recv[s][:, :self.dim]
+            cur[:, :self.dim] += del_recv[s][:, :self.dim]
             # duplicate local indices across sources are rare (each source
             # deduplicates); across sources index_copy applies sequentially
             shard.index_copy_(0, idx_recv[s], cur.to(shard.dtype))
@@ -259,6 +267,12 @@ class RowShardedSgns:
         if plan is None:
             plan = self.make_plan(tokens, offsets, window, n_neg, rng)
         if isinstance(plan, sgns.GroupedPlanT):
+            if self.world == 1 and self.is_cuda and not self.serial:
+                # world 1: row r = word r and the shard is stored at the
+                # kernel stride — train DIRECTLY on the tables (no
+                # unique/pull/write-back at all)
+                self._train_pairs_direct(plan, alpha)
+                return
             # unique on int32 ids: half the radix-sort bytes of .long()
             uc, inv_c = torch.unique(plan.group_center,
                                      return_inverse=True)
@@ -306,8 +320,34 @@ class RowShardedSgns:
         """World-1 fast path: replace owned rows with the trained cache
         (ids unique, so copy == add-delta)."""
         shard = self.syn0 if which == 0 else self.syn1
-        shard.index_copy_(0, self._to_ids(ids),
-                          rows[:, :self.dim].to(shard.dtype))
+        if rows.shape[1] == self.store_stride:
+            shard.index_copy_(0, self._to_ids(ids), rows.to(shard.dtype))
+        else:
+            shard[:, self.dim:] = 0   # unreachable today; keep pad invariant
+            shard.index_copy_(0, self._to_ids(ids),
+                              torch.nn.functional.pad(
+                                  rows[:, :self.dim],
+                                  (0, self.store_stride - self.dim))
+                              .to(shard.dtype))
+
+    def _train_pairs_direct(self, plan, alpha: float) -> None:
+        if plan.num_pairs == 0:
+            return
+        G = plan.num_groups
+        nb = max(1, min((G + 3) // 4, 2048))
+        stream = torch.cuda.current_stream(self.device)
+        gc = plan.group_center.contiguous()   # global word id == shard row
+        go = plan.group_offsets.contiguous()
+        pt = plan.pair_target.contiguous()
+        pl = plan.pair_label.contiguous()
+        pair_mode = 0 if (self.atomic and not self.is_bf16) else 1
+        self.native.train_pairs(
+            self.syn0.data_ptr(), self.syn1.data_ptr(), self.store_stride,
+            gc.data_ptr(), go.data_ptr(), G, pt.data_ptr(), pl.data_ptr(),
+            float(alpha), self._stats.data_ptr(), nb, 256,
+            stream.cuda_stream, pair_mode, int(self.atomic),
+            int(self.is_bf16))
+        self._inflight = (gc, go, pt, pl)
 
     def _train_pairs_t(self, cache0, cache1, group_center, group_offsets,
                        pair_target, pair_label, alpha):
@@ -419,8 +459,8 @@ class RowShardedSgns:
                 for r0 in range(0, self.shard_size, block):
                     r1 = min(self.shard_size, r0 + block)
                     f.write(np.ascontiguousarray(
-                        t[r0:r1].float().cpu().numpy(), dtype=np.float32)
-                        .tobytes())
+                        t[r0:r1, :self.dim].float().cpu().numpy(),
+                        dtype=np.float32).tobytes())
         comm.barrier()
 
     def load_checkpoint(self, path: str) -> None:
@@ -442,12 +482,13 @@ class RowShardedSgns:
             buf = np.fromfile(os.path.join(path, "shards",
                                            f"{name}-{self.rank:05d}.bin"),
                               dtype=dt).reshape(self.shard_size, self.dim)
-            t.copy_(torch.from_numpy(buf).to(t.dtype).to(self.device))
+            t[:, :self.dim] = \
+                torch.from_numpy(buf).to(t.dtype).to(self.device)
 
     def to_host(self) -> Tuple[np.ndarray, np.ndarray]:
         out = []
         for shard in (self.syn0, self.syn1):
-            my = shard.float()
+            my = shard[:, :self.dim].float()
             if self.world == 1:
                 out.append(my.cpu().numpy().copy())
                 continue
