@@ -130,3 +130,48 @@ def test_row_engine_device_plan_end_to_end():
     assert not np.array_equal(before, after)
     st = eng.read_stats()
     assert st.pairs > 0 and st.positives > 0
+
+
+@pytest.mark.parametrize("case", ["empty", "single_word", "one_sent_1tok",
+                                  "window_bigger_than_sent"])
+def test_planner_edge_cases(case):
+    table = torch.from_numpy(
+        build_unigram_table(np.ones(20, np.int64), 101))
+    gen = torch.Generator().manual_seed(1)
+    if case == "empty":
+        tokens = torch.zeros(0, dtype=torch.int32)
+        offsets = torch.zeros(1, dtype=torch.int32)
+    elif case == "single_word":
+        tokens = torch.tensor([7], dtype=torch.int32)
+        offsets = torch.tensor([0, 1], dtype=torch.int32)
+    elif case == "one_sent_1tok":
+        tokens = torch.tensor([3, 7, 3], dtype=torch.int32)
+        offsets = torch.tensor([0, 1, 2, 3], dtype=torch.int32)
+    else:  # window 10 >> sentence length 3
+        tokens = torch.tensor([1, 2, 3], dtype=torch.int32)
+        offsets = torch.tensor([0, 3], dtype=torch.int32)
+    p = sgns.make_grouped_plan_torch(tokens, offsets, None, table, 10, 3,
+                                     gen)
+    go = p.group_offsets.numpy()
+    assert go[0] == 0 and go[-1] == p.num_pairs
+    if p.num_pairs:
+        pl = p.pair_label.numpy()
+        assert (pl[go[:-1]] == 1.0).all()
+        # targets only come from within the same sentence
+        pt = p.pair_target.numpy()
+        assert set(pt[pl == 1].tolist()) <= set(tokens.numpy().tolist())
+    if case in ("empty", "single_word"):
+        assert p.num_pairs == 0
+
+
+def test_planner_full_subsample_drop():
+    """keep_prob 0 drops everything: empty plan, no crash."""
+    table = torch.from_numpy(
+        build_unigram_table(np.ones(20, np.int64), 101))
+    gen = torch.Generator().manual_seed(2)
+    tokens = torch.from_numpy(
+        np.random.default_rng(0).integers(0, 20, 100).astype(np.int32))
+    offsets = torch.tensor([0, 100], dtype=torch.int32)
+    kp = torch.zeros(20)
+    p = sgns.make_grouped_plan_torch(tokens, offsets, kp, table, 3, 2, gen)
+    assert p.num_pairs == 0 and p.num_groups == 0
