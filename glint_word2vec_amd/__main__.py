@@ -47,6 +47,11 @@ def _add_train(sub):
                    help="CPU hogwild threads (numPartitions analog)")
     p.add_argument("--shards", type=int, default=1,
                    help="checkpoint shard files (numParameterServers analog)")
+    p.add_argument("--resume-from", default=None, metavar="CKPT",
+                   help="continue training from a saved model directory")
+    p.add_argument("--checkpoint-every", type=int, default=0, metavar="N",
+                   help="write a loadable checkpoint every N steps "
+                        "(<model>-step<N> directories; 0 = off)")
 
 
 def main(argv=None):
@@ -77,8 +82,12 @@ def main(argv=None):
             unigram_table_size=args.unigram_table_size,
             max_sentence_length=args.max_sentence_length,
             dtype=args.dtype, device=args.device, engine=args.engine,
-            window_mode=args.window_mode, num_partitions=args.workers)
-        model = est.fit(args.corpus)
+            window_mode=args.window_mode, num_partitions=args.workers,
+            checkpoint_every=args.checkpoint_every)
+        model = est.fit(args.corpus,
+                        save_path=(args.model if args.checkpoint_every
+                                   else None),
+                        init_from=args.resume_from)
         import os
         rank = int(os.environ.get("RANK", "0"))
         if rank == 0:

@@ -46,3 +46,26 @@ def test_cli_roundtrip(tmp_path):
     r = _run(["info", model_dir])
     assert r.returncode == 0
     assert "vectorSize" in r.stdout
+
+
+def test_cli_resume_and_mid_checkpoints(tmp_path):
+    corpus = tmp_path / "c.txt"
+    import numpy as np
+    rng = np.random.default_rng(5)
+    corpus.write_text("\n".join(
+        " ".join(f"w{rng.integers(0, 10)}" for _ in range(15))
+        for _ in range(40)))
+    m1 = str(tmp_path / "m1")
+    r = _run(["train", str(corpus), m1, "--vector-size", "8",
+              "--min-count", "1", "--iterations", "1", "--window", "2",
+              "--subsample", "0", "--seed", "3", "--device", "cpu",
+              "--unigram-table-size", "10000"])
+    assert r.returncode == 0, r.stderr
+    m2 = str(tmp_path / "m2")
+    r = _run(["train", str(corpus), m2, "--vector-size", "8",
+              "--min-count", "1", "--iterations", "1", "--window", "2",
+              "--subsample", "0", "--seed", "3", "--device", "cpu",
+              "--unigram-table-size", "10000", "--resume-from", m1])
+    assert r.returncode == 0, r.stderr
+    r = _run(["info", m2])
+    assert r.returncode == 0
