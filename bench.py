@@ -54,7 +54,7 @@ def parse_args():
                         "delta-allreduce) when tables fit, else dim-sharded")
     p.add_argument("--sync-every", type=int, default=4,
                    help="dp engine: steps between delta merges")
-    p.add_argument("--chunk-words", type=int, default=1 << 19,
+    p.add_argument("--chunk-words", type=int, default=1 << 20,
                    help="dim-sharded feedback chunk size")
     return p.parse_args()
 

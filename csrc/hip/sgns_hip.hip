@@ -489,6 +489,29 @@ struct RowIO32M<uint16_t, NCH> {
   }
 };
 
+// Compile-time masked/unmasked row-I/O selector: padded storage keeps the
+// original unmasked vector loads (the masked variant's per-chunk guards
+// cost real throughput even when always-true); narrow storage masks.
+template <typename T, int NCH, bool MASKED>
+struct RIO32 {
+  static __device__ __forceinline__ void load(const T* row, float v[NCH],
+                                              int l32, int width) {
+    if (MASKED) RowIO32M<T, NCH>::load(row, v, l32, width);
+    else RowIO32<T, NCH>::load(row, v, l32);
+  }
+  static __device__ __forceinline__ void store(T* row, const float v[NCH],
+                                               int l32, int width) {
+    if (MASKED) RowIO32M<T, NCH>::store(row, v, l32, width);
+    else RowIO32<T, NCH>::store(row, v, l32);
+  }
+  static __device__ __forceinline__ void atomic_add(T* row,
+                                                    const float v[NCH],
+                                                    int l32, int width) {
+    if (MASKED) RowIO32M<T, NCH>::atomic_add(row, v, l32, width);
+    else RowIO32<T, NCH>::atomic_add(row, v, l32);
+  }
+};
+
 // Per-half (32-lane) sum: 4x row_shr + row_bcast15 leaves each half's total
 // in its lane 31/63; one width-32 shuffle broadcasts it within the half.
 __device__ __forceinline__ float half_sum_f32(float v) {
@@ -1385,7 +1408,7 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void update_slice_kernel(
 }
 
 // ---- Phase: dim-sharded dots / update, TWO pairs per wave ---------------
-template <typename T, int NCH>
+template <typename T, int NCH, bool MASKED>
 struct DotPhase2 {
   const T* syn0;
   const T* syn1;
@@ -1396,7 +1419,8 @@ struct DotPhase2 {
   int width;
   float c_row[NCH];
   __device__ __forceinline__ void begin_position(int32_t c) {
-    RowIO32M<T, NCH>::load(syn0 + (int64_t)c * stride, c_row, l32, width);
+    RIO32<T, NCH, MASKED>::load(syn0 + (int64_t)c * stride, c_row, l32,
+                                width);
   }
   __device__ __forceinline__ void process_pairs(const uint32_t* tl, int count,
                                                 int64_t idx_base) {
@@ -1406,7 +1430,7 @@ struct DotPhase2 {
       const uint32_t enc = tl[active ? my : k];
       const T* t_ptr = syn1 + (int64_t)(enc & 0x7FFFFFFFu) * stride;
       float t_row[NCH];
-      RowIO32M<T, NCH>::load(t_ptr, t_row, l32, width);
+      RIO32<T, NCH, MASKED>::load(t_ptr, t_row, l32, width);
       float f = 0.0f;
 #pragma unroll
       for (int m = 0; m < NCH; ++m) f += c_row[m] * t_row[m];
@@ -1417,7 +1441,7 @@ struct DotPhase2 {
   __device__ __forceinline__ void end_position(int32_t) {}
 };
 
-template <typename T, int NCH>
+template <typename T, int NCH, bool MASKED>
 __global__ __launch_bounds__(64 * kWavesPerBlock) void dots_slice2_kernel(
     KernelArgs a, const int64_t* __restrict__ pair_offsets,
     float* __restrict__ f_out) {
@@ -1429,7 +1453,7 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void dots_slice2_kernel(
   const int64_t wave_gid = (int64_t)blockIdx.x * waves_in_block + wave;
   const int64_t total_waves = (int64_t)gridDim.x * waves_in_block;
   for (int64_t s = wave_gid; s < a.num_sentences; s += total_waves) {
-    DotPhase2<T, NCH> ph{};
+    DotPhase2<T, NCH, MASKED> ph{};
     ph.syn0 = (const T*)a.syn0;
     ph.syn1 = (const T*)a.syn1;
     ph.stride = a.stride;
@@ -1445,7 +1469,7 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void dots_slice2_kernel(
   }
 }
 
-template <typename T, int NCH, bool ATOMIC>
+template <typename T, int NCH, bool ATOMIC, bool MASKED>
 struct UpdateSlicePhase2 {
   T* syn0;
   T* syn1;
@@ -1469,7 +1493,7 @@ struct UpdateSlicePhase2 {
   __device__ __forceinline__ void begin_position(int32_t c) {
     c_idx = c;
     c_ptr = syn0 + (int64_t)c * stride;
-    RowIO32M<T, NCH>::load(c_ptr, c_row, l32, width);
+    RIO32<T, NCH, MASKED>::load(c_ptr, c_row, l32, width);
 #pragma unroll
     for (int m = 0; m < NCH; ++m) grad[m] = 0.0f;
   }
@@ -1481,7 +1505,7 @@ struct UpdateSlicePhase2 {
       const uint32_t enc = tl[active ? my : k];
       T* t_ptr = syn1 + (int64_t)(enc & 0x7FFFFFFFu) * stride;
       float t_row[NCH];
-      RowIO32M<T, NCH>::load(t_ptr, t_row, l32, width);
+      RIO32<T, NCH, MASKED>::load(t_ptr, t_row, l32, width);
       float f = f_base[idx_base + (active ? my : k)];
       if (f_loc) {
         float fresh = 0.0f;
@@ -1502,11 +1526,11 @@ struct UpdateSlicePhase2 {
           float delta[NCH];
 #pragma unroll
           for (int m = 0; m < NCH; ++m) delta[m] = g * c_row[m];
-          RowIO32M<T, NCH>::atomic_add(t_ptr, delta, l32, width);
+          RIO32<T, NCH, MASKED>::atomic_add(t_ptr, delta, l32, width);
         } else {
 #pragma unroll
           for (int m = 0; m < NCH; ++m) t_row[m] += g * c_row[m];
-          RowIO32M<T, NCH>::store(t_ptr, t_row, l32, width);
+          RIO32<T, NCH, MASKED>::store(t_ptr, t_row, l32, width);
         }
         ++w_pairs;
         if (label > 0.5f) {
@@ -1520,19 +1544,20 @@ struct UpdateSlicePhase2 {
 #pragma unroll
     for (int m = 0; m < NCH; ++m) grad[m] += __shfl_xor(grad[m], 32, 64);
     if (ATOMIC && (c_idx < atomic_below)) {
-      if (half == 0) RowIO32M<T, NCH>::atomic_add(c_ptr, grad, l32, width);
+      if (half == 0)
+        RIO32<T, NCH, MASKED>::atomic_add(c_ptr, grad, l32, width);
     } else if (half == 0) {
       float cur[NCH];
-      RowIO32M<T, NCH>::load(c_ptr, cur, l32, width);
+      RIO32<T, NCH, MASKED>::load(c_ptr, cur, l32, width);
 #pragma unroll
       for (int m = 0; m < NCH; ++m) cur[m] += grad[m];
-      RowIO32M<T, NCH>::store(c_ptr, cur, l32, width);
+      RIO32<T, NCH, MASKED>::store(c_ptr, cur, l32, width);
     }
     ++w_words;
   }
 };
 
-template <typename T, int NCH, bool ATOMIC>
+template <typename T, int NCH, bool ATOMIC, bool MASKED>
 __global__ __launch_bounds__(64 * kWavesPerBlock) void update_slice2_kernel(
     KernelArgs a, const int64_t* __restrict__ pair_offsets,
     const float* __restrict__ f_in, const float* __restrict__ f_loc,
@@ -1544,7 +1569,7 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void update_slice2_kernel(
   const int waves_in_block = blockDim.x >> 6;
   const int64_t wave_gid = (int64_t)blockIdx.x * waves_in_block + wave;
   const int64_t total_waves = (int64_t)gridDim.x * waves_in_block;
-  UpdateSlicePhase2<T, NCH, ATOMIC> ph{};
+  UpdateSlicePhase2<T, NCH, ATOMIC, MASKED> ph{};
   ph.syn0 = (T*)a.syn0;
   ph.syn1 = (T*)a.syn1;
   ph.stride = a.stride;
@@ -2023,10 +2048,15 @@ static void dots_slice(uintptr_t syn0, uintptr_t syn1, int is_bf16,
   if (width != (int)stride && !(pair_mode && threads == 256))
     throw std::runtime_error("masked width requires pair_mode kernels");
   hipStream_t stream = (hipStream_t)stream_ptr;
+  const bool masked = width != (int)stride;
 #define DOTS_CASE(T, N)                                                       \
   do {                                                                        \
-    if (pair_mode && threads == 256)                                          \
-      hipLaunchKernelGGL((dots_slice2_kernel<T, 2 * N>), dim3(blocks),        \
+    if (pair_mode && threads == 256 && masked)                                \
+      hipLaunchKernelGGL((dots_slice2_kernel<T, 2 * N, true>), dim3(blocks),  \
+                         dim3(threads), 0, stream, a,                         \
+                         (const int64_t*)pair_offsets, (float*)f_out);        \
+    else if (pair_mode && threads == 256)                                     \
+      hipLaunchKernelGGL((dots_slice2_kernel<T, 2 * N, false>), dim3(blocks), \
                          dim3(threads), 0, stream, a,                         \
                          (const int64_t*)pair_offsets, (float*)f_out);        \
     else                                                                      \
@@ -2077,16 +2107,27 @@ static void update_slice(uintptr_t syn0, uintptr_t syn1, int is_bf16,
   if (width != (int)stride && !(pair_mode && threads == 256))
     throw std::runtime_error("masked width requires pair_mode kernels");
   hipStream_t stream = (hipStream_t)stream_ptr;
+  const bool masked = width != (int)stride;
 #define UPD_CASE(T, N)                                                        \
   do {                                                                        \
     if (pair_mode && threads == 256) {                                        \
-      if (atomic)                                                             \
-        hipLaunchKernelGGL((update_slice2_kernel<T, 2 * N, true>),            \
+      if (atomic && masked)                                                   \
+        hipLaunchKernelGGL((update_slice2_kernel<T, 2 * N, true, true>),      \
+                           dim3(blocks), dim3(threads), 0, stream, a,         \
+                           (const int64_t*)pair_offsets, (const float*)f_in,  \
+                           (const float*)f_loc, (float)world_scale);          \
+      else if (atomic)                                                        \
+        hipLaunchKernelGGL((update_slice2_kernel<T, 2 * N, true, false>),     \
+                           dim3(blocks), dim3(threads), 0, stream, a,         \
+                           (const int64_t*)pair_offsets, (const float*)f_in,  \
+                           (const float*)f_loc, (float)world_scale);          \
+      else if (masked)                                                        \
+        hipLaunchKernelGGL((update_slice2_kernel<T, 2 * N, false, true>),     \
                            dim3(blocks), dim3(threads), 0, stream, a,         \
                            (const int64_t*)pair_offsets, (const float*)f_in,  \
                            (const float*)f_loc, (float)world_scale);          \
       else                                                                    \
-        hipLaunchKernelGGL((update_slice2_kernel<T, 2 * N, false>),           \
+        hipLaunchKernelGGL((update_slice2_kernel<T, 2 * N, false, false>),    \
                            dim3(blocks), dim3(threads), 0, stream, a,         \
                            (const int64_t*)pair_offsets, (const float*)f_in,  \
                            (const float*)f_loc, (float)world_scale);          \

@@ -76,7 +76,7 @@ class Word2VecConfig:
     # "dp": replicated tables + periodic delta-allreduce (fused kernel per
     # GPU; for vocabularies whose tables fit comfortably in HBM).
     engine: str = "auto"             # "auto" | "fused" | "dim" | "row" | "dp"
-    chunk_words: int = 1 << 19       # dim-sharded feedback chunk
+    chunk_words: int = 1 << 20       # dim-sharded feedback chunk
     f_correction: bool = True        # dim-sharded local-drift freshening
     sync_every: int = 4              # dp engine: steps between delta merges
     # mid-training checkpoints every N steps (0 = off; the reference has

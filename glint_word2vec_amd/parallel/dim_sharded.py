@@ -66,9 +66,16 @@ class DimShardedSgns:
             # narrow storage: stride = round_up(width, 8) removes the
             # 64-element padding waste for thin dim slices (NOTES backlog
             # #2, measured in benchmarks/results.md); requires the pair2
-            # masked kernels, so serial-parity tests pass narrow=False
+            # masked kernels, so serial-parity tests pass narrow=False.
+            # Auto only when padding waste is >= 25%: a non-64-multiple
+            # stride misaligns row starts, which costs far more than a
+            # small byte saving on wide bandwidth-bound slices (measured:
+            # dim 300 at stride 304 runs 2.4x SLOWER than padded 320;
+            # width 38 at stride 40 runs 2.6% faster than padded 64)
             if narrow is None:
-                narrow = self.width % 64 != 0
+                padded = self.native.round_stride(max(self.width, 1))
+                narrow = (self.width % 64 != 0
+                          and (self.width + 7) // 8 * 8 * 4 <= padded * 3)
             self.narrow = bool(narrow)
             self.stride = ((self.width + 7) // 8 * 8 if self.narrow
                            else self.native.round_stride(max(self.width, 1)))
@@ -205,7 +212,7 @@ class DimShardedSgns:
                 (2 ** 31 - 1) if self.atomic else 0, self._stats.data_ptr(),
                 nb, nt, comp.cuda_stream, 0, 0,
                 0 if self.serial else 1,
-                0 if self.serial else self.width)
+                self.width if self.narrow else 0)
 
         for (a, b) in chunks:
             n = b - a
@@ -220,7 +227,7 @@ class DimShardedSgns:
                 window, n_neg, seed, sent_id_base + a, ref,
                 poff[a:b + 1].data_ptr(), f.data_ptr(), nb, nt,
                 comp.cuda_stream, 0 if self.serial else 1,
-                0 if self.serial else self.width)
+                self.width if self.narrow else 0)
             if f_loc is not None:
                 f_loc[lo:hi] = f[lo:hi]
             ev = torch.cuda.Event()
