@@ -479,3 +479,42 @@ def test_dp_pipelined_sync_mechanics():
     assert torch.equal(eng.syn0, syn_before)
     assert torch.equal(eng.master0, eng.syn0.float())
     assert torch.equal(eng.master1, eng.syn1.float())
+
+
+def test_row_engine_direct_mode():
+    """World-1 direct mode (pairs kernel straight on padded shard tables)
+    must train, keep padding zero, and leave usable embeddings."""
+    from glint_word2vec_amd.parallel.row_sharded import RowShardedSgns
+    rng = np.random.default_rng(11)
+    vocab, dim = 5000, 48
+    tokens = rng.integers(0, vocab, 20000).astype(np.int32)
+    offsets = np.arange(0, 20001, 100, dtype=np.int32)
+    counts = np.bincount(tokens, minlength=vocab).astype(np.int64) + 1
+    for dtype in ("bfloat16", "float32"):
+        eng = RowShardedSgns(vocab, dim, dtype=dtype, device="cuda", seed=3,
+                             counts=counts, table_size=100003, atomic=False)
+        before = eng.to_host()[0].copy()
+        for s in range(3):
+            plan = eng.make_plan_device(tokens, offsets, 4, 5, seed=50 + s)
+            eng.train_step(tokens, offsets, 0.05, 4, 5,
+                           np.random.default_rng(1), plan=plan)
+        torch.cuda.synchronize()
+        st = eng.read_stats()
+        assert st.pairs > 0 and st.positives > 0
+        # padding columns beyond dim stay exactly zero
+        assert eng.store_stride > dim
+        assert torch.all(eng.syn0[:, dim:] == 0)
+        assert torch.all(eng.syn1[:, dim:] == 0)
+        s0, s1 = eng.to_host()
+        assert np.isfinite(s0).all() and np.isfinite(s1).all()
+        assert not np.array_equal(before, s0)
+        assert np.abs(s0).max() < 10.0
+        # atomic direct variant as well (bf16 packed / f32 atomics)
+        eng2 = RowShardedSgns(vocab, dim, dtype=dtype, device="cuda", seed=3,
+                              counts=counts, table_size=100003, atomic=True)
+        plan = eng2.make_plan_device(tokens, offsets, 4, 5, seed=99)
+        eng2.train_step(tokens, offsets, 0.05, 4, 5,
+                        np.random.default_rng(1), plan=plan)
+        torch.cuda.synchronize()
+        assert eng2.read_stats().pairs > 0
+        assert np.isfinite(eng2.to_host()[0]).all()
