@@ -334,7 +334,7 @@ class RowShardedSgns:
         if plan.num_pairs == 0:
             return
         G = plan.num_groups
-        nb = max(1, min((G + 3) // 4, 2048))
+        nb = max(1, min((G + 3) // 4, 8192))
         stream = torch.cuda.current_stream(self.device)
         gc = plan.group_center.contiguous()   # global word id == shard row
         go = plan.group_offsets.contiguous()
@@ -355,7 +355,7 @@ class RowShardedSgns:
         stream sync (everything stays queued on the compute stream)."""
         if self.is_cuda:
             G = int(group_center.numel())
-            nb = 1 if self.serial else max(1, min((G + 3) // 4, 2048))
+            nb = 1 if self.serial else max(1, min((G + 3) // 4, 8192))
             nt = 64 if self.serial else 256
             stream = torch.cuda.current_stream(self.device)
             gc = group_center.contiguous()
@@ -389,7 +389,7 @@ class RowShardedSgns:
             pl = torch.from_numpy(pair_label).to(self.device)
             stream = torch.cuda.current_stream(self.device)
             G = len(group_center)
-            nb = 1 if self.serial else max(1, min((G + 3) // 4, 2048))
+            nb = 1 if self.serial else max(1, min((G + 3) // 4, 8192))
             nt = 64 if self.serial else 256
             self.native.train_pairs(
                 cache0.data_ptr(), cache1.data_ptr(), self.cache_stride,
