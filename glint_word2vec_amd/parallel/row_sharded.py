@@ -5,11 +5,16 @@ Row r of both tables lives on rank r % world (round-robin interleave keeps
 Zipf-hot rows balanced).  Each rank is simultaneously a *worker* over its
 own corpus partition and the *server* for its row shard:
 
-  plan pairs (vectorized host plan)  ->  unique touched rows  ->
-  alltoallv index request to owners  ->  owners gather rows   ->
-  alltoallv rows back (pull)         ->  train on the local f32 cache
-  (fused pairs kernel, live hogwild feedback within the step)  ->
-  alltoallv row deltas to owners     ->  owners scatter-add (push/adjust)
+  plan pairs (torch ops on the device) ->  unique touched rows  ->
+  alltoallv index request to owners    ->  owners gather rows   ->
+  alltoallv rows back (pull)           ->  train on the local f32 cache
+  (pairs kernel, live hogwild feedback within the step)  ->
+  alltoallv row deltas to owners       ->  owners scatter-add (push/adjust)
+
+At world 1 the cycle collapses: shard row r IS table row r and shards are
+stored at the kernel stride, so the pairs kernel trains directly on the
+tables — no unique, no pull, no write-back (143M words/s at the headline
+config, benchmarks/results.md).
 
 This is the reference's dotprod/adjust split with the network role of Akka
 messages taken by two alltoallv exchanges; traffic scales with unique rows
@@ -17,7 +22,7 @@ x dim, so it is the engine of choice when the touched-row set is sparse
 (very large vocabularies) — the dim-sharded engine (dim_sharded.py) covers
 the dense case with dimension-independent traffic.
 
-Works on CUDA (HIP train_pairs kernel) and CPU (C++ twin; gloo tests).
+Works on CUDA (HIP train_pairs kernels) and CPU (C++ twin; gloo tests).
 """
 from __future__ import annotations
 
@@ -320,15 +325,8 @@ class RowShardedSgns:
         """World-1 fast path: replace owned rows with the trained cache
         (ids unique, so copy == add-delta)."""
         shard = self.syn0 if which == 0 else self.syn1
-        if rows.shape[1] == self.store_stride:
-            shard.index_copy_(0, self._to_ids(ids), rows.to(shard.dtype))
-        else:
-            shard[:, self.dim:] = 0   # unreachable today; keep pad invariant
-            shard.index_copy_(0, self._to_ids(ids),
-                              torch.nn.functional.pad(
-                                  rows[:, :self.dim],
-                                  (0, self.store_stride - self.dim))
-                              .to(shard.dtype))
+        assert rows.shape[1] == self.store_stride
+        shard.index_copy_(0, self._to_ids(ids), rows.to(shard.dtype))
 
     def _train_pairs_direct(self, plan, alpha: float) -> None:
         if plan.num_pairs == 0:
