@@ -344,8 +344,8 @@ def make_grouped_plan_torch(tokens, offsets, keep_prob, table,
         return empty
     if n_neg > 0:
         nidx = torch.randint(0, table.numel(), (P, n_neg), generator=gen,
-                             device=dev)
-        negs = table[nidx].to(torch.int32)
+                             device=dev, dtype=torch.int32)
+        negs = table[nidx.long()].to(torch.int32)
         nvalid = negs != pt[:, None]
         all_target = torch.cat([pt[:, None], negs], dim=1)
         all_label = torch.cat(
@@ -353,7 +353,7 @@ def make_grouped_plan_torch(tokens, offsets, keep_prob, table,
              torch.zeros(P, n_neg, device=dev)], dim=1)
         all_valid = torch.cat(
             [torch.ones(P, 1, dtype=torch.bool, device=dev), nvalid], dim=1)
-        all_cpos = cpos[:, None].expand(-1, 1 + n_neg)
+        all_cpos = cpos.to(torch.int32)[:, None].expand(-1, 1 + n_neg)
         fmask = all_valid.reshape(-1)
         pair_target = all_target.reshape(-1)[fmask]
         pair_label = all_label.reshape(-1)[fmask]
@@ -368,6 +368,6 @@ def make_grouped_plan_torch(tokens, offsets, keep_prob, table,
                         change])
     group_offsets = torch.cat(
         [starts, torch.tensor([Np], dtype=torch.int64, device=dev)])
-    group_center = toks[pair_cpos[starts]].to(torch.int32)
+    group_center = toks[pair_cpos[starts].long()].to(torch.int32)
     return GroupedPlanT(group_center, group_offsets, pair_target.contiguous(),
                         pair_label.contiguous())
