@@ -171,8 +171,9 @@ def main():
                                seed=99 + rank, sent_id_base=i * nsent)
         elif dist_mode == "row":
             if trainer.is_cuda:
-                plan = trainer.make_plan_device(tok, off, args.window,
-                                                args.neg, 99 + rank + 7919 * i)
+                plan = trainer.make_plan_counter(
+                    tok, off, args.window, args.neg, 99 + rank,
+                    sent_id_base=i * nsent)
                 trainer.train_step(batch.tokens, batch.offsets, alpha,
                                    args.window, args.neg, row_rng, plan=plan)
             else:

@@ -1169,6 +1169,67 @@ struct CountPhase {
   __device__ __forceinline__ void end_position(int32_t) {}
 };
 
+// ---- plan emission (row-sharded engine): materialise the walker's pair
+// enumeration — counter-based RNG, identical to the fused kernel's — into
+// explicit plan arrays.  Each sentence is written by exactly one wave at
+// offsets fixed by pair_offsets, so the output is deterministic at any
+// grid size.  out_start marks the first pair of each center position
+// (group boundaries). -------------------------------------------------------
+struct PlanEmitPhase {
+  int32_t* out_target;
+  float* out_label;
+  uint8_t* out_start;
+  int32_t* out_center;
+  int64_t base;          // pair_offsets[s]
+  int lane;
+  int32_t c_word;
+  bool at_pos_start;
+  __device__ __forceinline__ void begin_position(int32_t c) {
+    c_word = c;
+    at_pos_start = true;
+  }
+  __device__ __forceinline__ void process_pairs(const uint32_t* tl, int count,
+                                                int64_t idx_base) {
+    if (lane < count) {
+      const uint32_t enc = tl[lane];
+      const int64_t idx = base + idx_base + lane;
+      out_target[idx] = (int32_t)(enc & 0x7FFFFFFFu);
+      out_label[idx] = (enc & 0x80000000u) ? 1.0f : 0.0f;
+      out_center[idx] = c_word;
+      out_start[idx] = (at_pos_start && lane == 0) ? 1 : 0;
+    }
+    at_pos_start = false;
+  }
+  __device__ __forceinline__ void end_position(int32_t) {}
+};
+
+__global__ __launch_bounds__(64 * kWavesPerBlock) void plan_emit_kernel(
+    KernelArgs a, const int64_t* __restrict__ pair_offsets,
+    int32_t* __restrict__ out_target, float* __restrict__ out_label,
+    uint8_t* __restrict__ out_start, int32_t* __restrict__ out_center) {
+  __shared__ int32_t sbuf[kWavesPerBlock][kMaxSent];
+  __shared__ uint32_t tbuf[kWavesPerBlock][64];
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int waves_in_block = blockDim.x >> 6;
+  const int64_t wave_gid = (int64_t)blockIdx.x * waves_in_block + wave;
+  const int64_t total_waves = (int64_t)gridDim.x * waves_in_block;
+  for (int64_t s = wave_gid; s < a.num_sentences; s += total_waves) {
+    PlanEmitPhase ph{};
+    ph.out_target = out_target;
+    ph.out_label = out_label;
+    ph.out_start = out_start;
+    ph.out_center = out_center;
+    ph.base = pair_offsets[s];
+    ph.lane = lane;
+    const uint64_t base = sent_base(a.seed, (uint64_t)(a.sent_id_base + s));
+    walk_sentence_dev(a.tokens, a.offsets[s],
+                      (int)(a.offsets[s + 1] - a.offsets[s]), base, a.keep_thr,
+                      a.table, a.table_size, a.window, a.n_neg, a.ref_window,
+                      lane, sbuf[wave], tbuf[wave], ph);
+  }
+}
+
 __global__ __launch_bounds__(64 * kWavesPerBlock) void count_pairs_kernel(
     KernelArgs a, int64_t* __restrict__ counts) {
   __shared__ int32_t sbuf[kWavesPerBlock][kMaxSent];
@@ -2029,6 +2090,26 @@ static void count_pairs(uintptr_t tokens, uintptr_t offsets,
   HIP_CHECK(hipGetLastError());
 }
 
+static void plan_emit(uintptr_t tokens, uintptr_t offsets,
+                      int64_t num_sentences, uintptr_t keep_thr,
+                      uintptr_t table, int64_t table_size, int window,
+                      int n_neg, uint64_t seed, int64_t sent_id_base,
+                      int ref_window, uintptr_t pair_offsets,
+                      uintptr_t out_target, uintptr_t out_label,
+                      uintptr_t out_start, uintptr_t out_center, int blocks,
+                      int threads, uintptr_t stream_ptr) {
+  HIP_CLEAR_ERROR();
+  KernelArgs a = make_walk_args(0, 0, 64, tokens, offsets, num_sentences,
+                                keep_thr, table, table_size, 0.0, window,
+                                n_neg, seed, sent_id_base, ref_window, 0);
+  hipLaunchKernelGGL(plan_emit_kernel, dim3(blocks), dim3(threads), 0,
+                     (hipStream_t)stream_ptr, a,
+                     (const int64_t*)pair_offsets, (int32_t*)out_target,
+                     (float*)out_label, (uint8_t*)out_start,
+                     (int32_t*)out_center);
+  HIP_CHECK(hipGetLastError());
+}
+
 static void dots_slice(uintptr_t syn0, uintptr_t syn1, int is_bf16,
                        int64_t stride, uintptr_t tokens, uintptr_t offsets,
                        int64_t num_sentences, uintptr_t keep_thr,
@@ -2328,6 +2409,7 @@ PYBIND11_MODULE(_hip_native, m) {
   m.def("dots_slice", &dots_slice);
   m.def("update_slice", &update_slice);
   m.def("train_pairs", &train_pairs);
+  m.def("plan_emit", &plan_emit);
   m.def("pull_average", &pull_average);
   m.def("norms", &norms);
   m.def("round_stride", &round_stride);

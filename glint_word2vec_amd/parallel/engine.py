@@ -231,14 +231,17 @@ def _train_sharded(cfg: Word2VecConfig, vocab: Vocabulary,
                             if k < len(batches) else empty)
 
                 if eng.is_cuda:
-                    # device-side planning (torch ops on the GPU):
-                    # the numpy planner costs ~45 s of host time per
-                    # 2M-word batch — on-GPU it is milliseconds
+                    # GPU planning via the fused kernel's walker (counter
+                    # RNG, milliseconds; the numpy planner costs ~45 s of
+                    # host time per 2M-word batch).  Disjoint sentence-id
+                    # streams per rank (data parallel).
+                    sbase = 10_000_000 * rank + sent_base
                     for k in range(n_steps):
                         tokens, offsets = batch_at(k)
-                        plan = eng.make_plan_device(
-                            tokens, offsets, cfg.window, cfg.n,
-                            seed + 1_000_003 * (step + 31 * rank))
+                        plan = eng.make_plan_counter(
+                            tokens, offsets, cfg.window, cfg.n, seed,
+                            sent_id_base=sbase)
+                        sbase += max(len(offsets) - 1, 0)
                         alpha = cfg.learning_rate * max(
                             1e-4,
                             1.0 - processed / (total_words // world + 1))
@@ -249,6 +252,7 @@ def _train_sharded(cfg: Word2VecConfig, vocab: Vocabulary,
                         _maybe_mid_checkpoint(
                             cfg, save_path, step,
                             lambda p: eng.save_checkpoint(p, cfg, vocab))
+                    sent_base = sbase - 10_000_000 * rank
                     continue
                 # CPU: prefetch next step's host plan while this step trains
                 next_plan = (pool.submit(eng.make_plan, *batch_at(0),

@@ -263,6 +263,74 @@ class RowShardedSgns:
                                             self._table_t, window, n_neg,
                                             self._gen, self.window_mode)
 
+    def make_plan_counter(self, tokens, offsets, window: int, n_neg: int,
+                          seed: int, sent_id_base: int = 0,
+                          window_mode_ref: "bool | None" = None):
+        """GPU plan via the fused kernel's walker (counter-based RNG,
+        rng.py contract): count_pairs -> cumsum -> plan_emit.  Pair
+        enumeration is bit-identical to the fused kernel / CPU oracle for
+        the same (seed, sent_id_base) — unlike make_plan_device, whose
+        draws come from torch's RNG.  CUDA only."""
+        assert self.is_cuda, "counter planner runs the HIP walker"
+        if not hasattr(self, "_table_t"):
+            self._table_t = torch.from_numpy(self.table).to(self.device)
+            self._keep_prob_t = (None if self.keep_prob is None else
+                                 torch.from_numpy(self.keep_prob)
+                                 .to(self.device))
+            self._gen = torch.Generator(device=self.device)
+        if not hasattr(self, "_keep_thr_t"):
+            if self.keep_prob is None:
+                self._keep_thr_t = None
+            else:
+                thr = np.minimum(self.keep_prob.astype(np.float64)
+                                 * 4294967296.0,
+                                 4294967295.0).astype(np.uint32)
+                self._keep_thr_t = torch.from_numpy(
+                    thr.view(np.int32)).to(self.device)
+        tok = (torch.from_numpy(tokens) if isinstance(tokens, np.ndarray)
+               else tokens).to(self.device)
+        off = (torch.from_numpy(offsets) if isinstance(offsets, np.ndarray)
+               else offsets).to(self.device)
+        num_sent = off.numel() - 1
+        ref = int(self.window_mode == "reference"
+                  if window_mode_ref is None else window_mode_ref)
+        kthr = (0 if self._keep_thr_t is None
+                else self._keep_thr_t.data_ptr())
+        stream = torch.cuda.current_stream(self.device)
+        nb = 1 if self.serial else max(1, min((num_sent + 3) // 4, 8192))
+        nt = 64 if self.serial else 256
+        counts = torch.empty(num_sent, dtype=torch.int64, device=self.device)
+        self.native.count_pairs(
+            tok.data_ptr(), off.data_ptr(), num_sent, kthr,
+            self._table_t.data_ptr(), int(self._table_t.numel()), window,
+            n_neg, seed & 0xFFFFFFFFFFFFFFFF, sent_id_base, ref,
+            counts.data_ptr(), nb, nt, stream.cuda_stream)
+        poff = torch.zeros(num_sent + 1, dtype=torch.int64,
+                           device=self.device)
+        torch.cumsum(counts, 0, out=poff[1:])
+        total = int(poff[-1].item())
+        ei32 = torch.zeros(0, dtype=torch.int32, device=self.device)
+        if total == 0:
+            return sgns.GroupedPlanT(
+                ei32, torch.zeros(1, dtype=torch.int64, device=self.device),
+                ei32.clone(), torch.zeros(0, device=self.device))
+        target = torch.empty(total, dtype=torch.int32, device=self.device)
+        label = torch.empty(total, dtype=torch.float32, device=self.device)
+        start = torch.empty(total, dtype=torch.uint8, device=self.device)
+        center = torch.empty(total, dtype=torch.int32, device=self.device)
+        self.native.plan_emit(
+            tok.data_ptr(), off.data_ptr(), num_sent, kthr,
+            self._table_t.data_ptr(), int(self._table_t.numel()), window,
+            n_neg, seed & 0xFFFFFFFFFFFFFFFF, sent_id_base, ref,
+            poff.data_ptr(), target.data_ptr(), label.data_ptr(),
+            start.data_ptr(), center.data_ptr(), nb, nt, stream.cuda_stream)
+        starts = start.nonzero().reshape(-1)
+        group_offsets = torch.cat(
+            [starts, torch.tensor([total], dtype=torch.int64,
+                                  device=self.device)])
+        group_center = center[starts]
+        return sgns.GroupedPlanT(group_center, group_offsets, target, label)
+
     def train_step(self, tokens: np.ndarray, offsets: np.ndarray,
                    alpha: float, window: int, n_neg: int,
                    rng: np.random.Generator, plan=None) -> None:

@@ -555,3 +555,51 @@ def test_dim_sharded_narrow_atomic_gpu():
         np.testing.assert_allclose(n0, p0, **tol)
         np.testing.assert_allclose(n1, p1, **tol)
         assert np.isfinite(n0).all()
+
+
+def test_row_counter_plan_matches_oracle():
+    """The row engine's counter-RNG GPU planner (plan_emit kernel) must
+    enumerate exactly the fused kernel's / oracle's pairs, and serial
+    training on that plan must reproduce the oracle numerically."""
+    from glint_word2vec_amd.parallel.row_sharded import RowShardedSgns
+    tokens, offsets, counts, table, syn0, syn1 = _problem()
+    a0, a1 = syn0.copy(), syn1.copy()
+    st_py = cpu_ref.train_batch_oracle(a0, a1, tokens, offsets, None, table,
+                                       0.03, 3, 4, seed=77, sent_id_base=5)
+    eng = RowShardedSgns(50, 20, device="cuda", seed=1, counts=counts,
+                         table_size=1009, atomic=False)
+    eng.load_host(syn0, syn1)
+    eng.serial = True
+    plan = eng.make_plan_counter(tokens, offsets, 3, 4, seed=77,
+                                 sent_id_base=5)
+    # exact enumeration parity with the oracle
+    assert plan.num_pairs == st_py.pairs
+    pl = plan.pair_label.cpu().numpy()
+    assert int(pl.sum()) == st_py.positives
+    assert plan.num_groups == st_py.words_trained
+    eng.train_step(tokens, offsets, 0.03, 3, 4, np.random.default_rng(0),
+                   plan=plan)
+    torch.cuda.synchronize()
+    st = eng.read_stats()
+    assert st.pairs == st_py.pairs and st.positives == st_py.positives
+    g0, g1 = eng.to_host()
+    np.testing.assert_allclose(g0, a0, rtol=2e-4, atol=2e-6)
+    np.testing.assert_allclose(g1, a1, rtol=2e-4, atol=2e-6)
+
+
+def test_row_counter_plan_with_subsampling():
+    """Counter planner with in-kernel subsampling draws: pair totals match
+    count_pairs (by construction) and the oracle's enumeration."""
+    from glint_word2vec_amd.parallel.row_sharded import RowShardedSgns
+    tokens, offsets, counts, table, syn0, syn1 = _problem(n_tokens=400)
+    kp = keep_probabilities(counts, int(counts.sum()), 0.05)
+    st_py = cpu_ref.train_batch_oracle(syn0.copy(), syn1.copy(), tokens,
+                                       offsets, kp, table, 0.03, 3, 4,
+                                       seed=13, sent_id_base=2)
+    eng = RowShardedSgns(50, 20, device="cuda", seed=1, counts=counts,
+                         table_size=1009, subsample=0.05, atomic=False)
+    # engine keep_prob comes from the same counts/ratio -> same thresholds
+    plan = eng.make_plan_counter(tokens, offsets, 3, 4, seed=13,
+                                 sent_id_base=2)
+    assert plan.num_pairs == st_py.pairs
+    assert int(plan.pair_label.cpu().numpy().sum()) == st_py.positives
