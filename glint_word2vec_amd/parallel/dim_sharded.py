@@ -18,7 +18,6 @@ multi-process tests).
 from __future__ import annotations
 
 import logging
-from dataclasses import dataclass
 from typing import Optional
 
 import numpy as np

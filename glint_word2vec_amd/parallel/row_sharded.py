@@ -500,7 +500,6 @@ class RowShardedSgns:
         import os
         if self.rank == 0:
             os.makedirs(os.path.join(path, "shards"), exist_ok=True)
-            from ..checkpoint import save_model  # reuse metadata writer
             import time as _t
             meta = {
                 "class": "glint_word2vec_amd.GlintWord2VecModel",
