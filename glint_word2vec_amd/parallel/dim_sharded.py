@@ -207,7 +207,8 @@ class DimShardedSgns:
                 self.table.data_ptr(), int(self.table.numel()), float(alpha),
                 window, n_neg, seed, sent_id_base + c[0], ref,
                 poff[c[0]:c[1] + 1].data_ptr(), f.data_ptr(),
-                0 if f_loc is None else f_loc.data_ptr(), float(self.world),
+                0 if f_loc is None else f_loc.data_ptr(),
+                float(self.dim) / max(self.width, 1),
                 (2 ** 31 - 1) if self.atomic else 0, self._stats.data_ptr(),
                 nb, nt, comp.cuda_stream, 0, 0,
                 0 if self.serial else 1,
@@ -276,7 +277,8 @@ class DimShardedSgns:
                                           window, n_neg,
                                           seed & 0xFFFFFFFFFFFFFFFF, base, wm,
                                           poff, f.numpy(), f_loc,
-                                          float(self.world))
+                                          float(self.dim) /
+                                          max(self.width, 1))
             for k in ("pairs", "positives", "words_trained"):
                 self._cpu_stats[k] += st[k]
             self._cpu_stats["sum_fplus"] += st["sum_fplus"]
