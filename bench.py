@@ -38,12 +38,24 @@ def parse_args():
     p.add_argument("--subsample", type=float, default=1e-4,
                    help="subsample ratio (drawn in-kernel; 0 disables)")
     p.add_argument("--blocks", type=int, default=0, help="grid blocks override")
+    p.add_argument("--updates", choices=["hogwild", "atomic", "hybrid"],
+                   default="hybrid",
+                   help="row-update mode (same default as fit()): hybrid = "
+                        "atomics on the hot Zipf head (rows < hot-rows), "
+                        "plain hogwild RMW on the cold tail; hogwild = the "
+                        "reference's fire-and-forget adjust semantics "
+                        "(mllib:425); atomic = atomics everywhere")
+    p.add_argument("--hot-rows", type=int, default=None,
+                   help="hybrid mode: rows < K use atomics (default: "
+                        "Word2VecConfig.hybrid_hot_rows)")
     p.add_argument("--atomic", action="store_true",
-                   help="atomic row updates (no lost updates; ~3-5x slower). "
-                        "Default is hogwild plain RMW — the reference's "
-                        "fire-and-forget adjust semantics (mllib:425)")
+                   help="alias for --updates atomic")
     p.add_argument("--atomic-below", type=int, default=None,
-                   help="atomics only for rows < K (hot rows)")
+                   help="alias for --updates hybrid --hot-rows K")
+    p.add_argument("--pair-mode", type=int, default=None, choices=[0, 1, 2],
+                   help="fused-kernel variant: 0=one pair/wave (64-lane), "
+                        "1=two pairs (32-lane halves, default), "
+                        "2=four pairs (16-lane quarters)")
     p.add_argument("--profile-steps", type=int, default=0,
                    help="run only this many steps, no warmup JSON (rocprof)")
     p.add_argument("--device", choices=["cuda", "cpu"], default="cuda",
@@ -62,6 +74,19 @@ def parse_args():
 def main():
     args = parse_args()
     import torch
+    from glint_word2vec_amd.config import Word2VecConfig, choose_engine
+
+    # resolve the update mode (aliases kept for round-1 scripts)
+    if args.atomic:
+        args.updates = "atomic"
+    elif args.atomic_below is not None:
+        args.updates = "hybrid"
+        args.hot_rows = args.atomic_below
+    if args.hot_rows is None:
+        args.hot_rows = Word2VecConfig.hybrid_hot_rows
+    upd_atomic = args.updates != "hogwild"       # kernel atomic flag
+    upd_below = (0 if args.updates == "hogwild" else
+                 (2 ** 31 - 1 if args.updates == "atomic" else args.hot_rows))
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
@@ -93,8 +118,8 @@ def main():
     if engine == "auto":
         if distributed or use_cpu:
             esize = 2 if args.dtype == "bf16" else 4
-            table_bytes = 2 * args.vocab * (args.dim + 64) * esize
-            engine = "dp" if table_bytes <= (8 << 30) else "dim"
+            engine = choose_engine(args.vocab, args.dim, esize,
+                                   world if distributed else 2)
         else:
             engine = "fused"
 
@@ -129,7 +154,7 @@ def main():
                                  table_size=args.table_size,
                                  subsample=args.subsample,
                                  chunk_words=args.chunk_words,
-                                 atomic=bool(args.atomic or args.atomic_below))
+                                 atomic=upd_atomic, atomic_below=upd_below)
         dist_mode = True
     elif engine == "dp":
         from glint_word2vec_amd.parallel.replicated import ReplicatedSgns
@@ -138,7 +163,7 @@ def main():
                                  table_size=args.table_size,
                                  subsample=args.subsample,
                                  sync_every=args.sync_every,
-                                 atomic=bool(args.atomic or args.atomic_below))
+                                 atomic=upd_atomic, atomic_below=upd_below)
         dist_mode = "dp"
     elif engine == "row":
         from glint_word2vec_amd.parallel.row_sharded import RowShardedSgns
@@ -146,7 +171,8 @@ def main():
                                  device=str(device), seed=1, counts=counts,
                                  table_size=args.table_size,
                                  subsample=args.subsample,
-                                 atomic=bool(args.atomic or args.atomic_below))
+                                 atomic=args.updates == "atomic",
+                                 atomic_below=upd_below)
         dist_mode = "row"
     else:
         from glint_word2vec_amd.ops.gpu import GpuSgns
@@ -186,9 +212,12 @@ def main():
         else:
             trainer.train_batch(tok, off, alpha, args.window, args.neg, 99,
                                 sent_id_base=i * nsent,
-                                atomic=bool(args.atomic or args.atomic_below),
-                                atomic_below=args.atomic_below,
-                                blocks=args.blocks or None)
+                                atomic=upd_atomic,
+                                atomic_below=(None if args.updates == "atomic"
+                                              else upd_below),
+                                blocks=args.blocks or None,
+                                **({} if args.pair_mode is None
+                                   else {"pair_mode": args.pair_mode}))
 
     def barrier_sync():
         if not use_cpu:
@@ -249,8 +278,8 @@ def main():
                 "seq_len": args.sentence_len,
                 "parallelism": (f"{engine}shard-rccl-x{world}" if dist_mode
                                 else "hogwild-1gpu"),
-                "updates": ("atomic" if (args.atomic or args.atomic_below)
-                            else "hogwild"),
+                "updates": (args.updates if args.updates != "hybrid" else
+                            f"hybrid(atomic rows<{args.hot_rows})"),
             },
             "pairs_per_step": st.pairs / max(args.steps, 1),
             "mean_fplus": st.sum_fplus / max(st.positives, 1),

@@ -17,6 +17,41 @@ from typing import Optional
 MAX_EXP = 6.0          # sigmoid clip range [-6, 6] (mllib:248, 281-302)
 EXP_TABLE_SIZE = 1000  # only used by the optional LUT-parity sigmoid mode
 
+# Row strides the HIP kernels support (stride = 64 * NC; csrc/hip
+# FOR_EACH_NC).  round_stride_py mirrors _hip_native.round_stride so CPU-only
+# code (bench policy, memory estimates) agrees with the kernels exactly.
+KERNEL_NCS = (1, 2, 3, 4, 5, 6, 8, 10, 12, 16, 20, 24, 32)
+
+
+def round_stride_py(dim: int) -> int:
+    nc_min = (dim + 63) // 64
+    for nc in KERNEL_NCS:
+        if nc >= nc_min:
+            return 64 * nc
+    raise ValueError(f"dim {dim} too large (max {64 * KERNEL_NCS[-1]})")
+
+
+# dp-vs-dim auto-engine threshold: the dp engine allreduces full-table
+# deltas every sync_every steps; at bf16 wire dtype a sync moves
+# ~2*vocab*stride*2 bytes over ring xGMI (~153 GB/s per link), so tables
+# beyond a few GB make the merge cost exceed the step time and the
+# dimension-sharded engine (traffic ~4 bytes/pair, dimension-independent)
+# wins even though its per-GPU kernel rate is lower.  4 GiB of padded
+# bf16 tables ~= vocab 3M at dim 300: sync ~18 ms vs ~12 ms step at
+# sync_every=4 — the crossover region.  (288 GB HBM is NOT the binding
+# constraint for dp; xGMI merge bandwidth is.)
+DP_MAX_TABLE_BYTES = 4 << 30
+
+
+def choose_engine(vocab_size: int, dim: int, dtype_bytes: int,
+                  world: int) -> str:
+    """Auto engine policy, shared by engine.py and bench.py (both use the
+    padded kernel stride so the estimate matches the real allocation)."""
+    if world == 1:
+        return "fused"
+    table_bytes = 2 * vocab_size * round_stride_py(dim) * dtype_bytes
+    return "dp" if table_bytes <= DP_MAX_TABLE_BYTES else "dim"
+
 
 @dataclass
 class Word2VecConfig:
@@ -48,7 +83,12 @@ class Word2VecConfig:
     unigram_table_size: int = 100_000_000  # negative-sampling table (mllib:85)
     unigram_power: float = 0.75      # classic word2vec table exponent
     # --- parallelism (replaces numPartitions / numParameterServers) --------
-    num_partitions: int = 1          # concurrent async workers per GPU
+    # numPartitions (mllib:120-127) = concurrent async workers.  On the CPU
+    # trainer this is the hogwild thread count.  On GPU it is accepted but
+    # INERT by design: the reference used it to create concurrency, and the
+    # fused kernel already runs ~8k concurrent hogwild waves per launch —
+    # the GPU analog of numPartitions is the wave count, not a user knob.
+    num_partitions: int = 1
     num_shards: Optional[int] = None  # row shards; None = world size
     # --- device ------------------------------------------------------------
     # table storage dtype. "auto" = bfloat16 on GPU, float32 on CPU: the
@@ -57,18 +97,29 @@ class Word2VecConfig:
     dtype: str = "auto"              # "auto" | "float32" | "bfloat16"
     device: str = "auto"             # "auto" | "cpu" | "cuda"
     words_per_step: int = 1 << 20    # tokens fed to the GPU per training step
-    # Row updates. False (default) = plain hogwild read-modify-write: the
-    # reference's races-embraced fire-and-forget semantics (mllib:425);
-    # concurrent updates to a hot row mostly collapse to one (self-limiting,
-    # like last-writer CPU hogwild).  True = atomics (fp32 atomicAdd /
-    # gfx950 packed-bf16): no lost updates, but ALL concurrent stale
-    # gradients land — on extremely hot rows (tiny vocabularies, no
-    # subsampling) the summed stale positives can run away (measured:
-    # diverges on a 200-word corpus; fine at vocab 1M + subsampling).
-    # ~3-5x slower.  Default True: correctness-first for fit() on real
-    # corpora (the explosion regime is degenerate toy data); the benchmark
-    # measures hogwild explicitly.  See DESIGN.md.
-    atomic_updates: bool = True
+    # Row-update mode for the massively concurrent GPU kernels:
+    #   "hogwild" — plain read-modify-write: the reference's races-embraced
+    #     fire-and-forget semantics (mllib:425); fastest, but lost updates
+    #     cap hot-word quality (planted-NN acc plateaus ~0.61,
+    #     benchmarks/results.md).
+    #   "atomic"  — atomics on every row (fp32 atomicAdd / gfx950 packed
+    #     bf16): no lost updates, best quality (acc ~0.81), ~2.5x slower.
+    #     On degenerate corpora (hundreds of words, no subsampling) the
+    #     summed stale gradients on ultra-hot rows can run away; fine on
+    #     real vocabularies.
+    #   "hybrid" (default) — atomics only on rows < hybrid_hot_rows (the
+    #     vocabulary is sorted by count, so these ARE the contended Zipf
+    #     head where hogwild loses updates); the cold tail — rarely
+    #     contended, hogwild == atomic there in practice — takes the
+    #     cheap path.  Measured on the planted-synonym probe: quality of
+    #     full atomics at most of hogwild's speed (benchmarks/results.md).
+    # The benchmark and fit() share this default (same semantics measured
+    # as shipped).  See DESIGN.md.
+    update_mode: str = "hybrid"      # "hogwild" | "atomic" | "hybrid"
+    hybrid_hot_rows: int = 1 << 18   # atomic rows for update_mode="hybrid"
+    # Deprecated alias (round-1 API): True -> "atomic", False -> "hogwild".
+    # None (default) leaves update_mode in charge.
+    atomic_updates: Optional[bool] = None
     # --- multi-GPU engine --------------------------------------
     # "auto": fused single-GPU kernel at world 1; dim-sharded at world > 1.
     # "dim": dimension-sharded (CIKM scheme, RCCL allreduce of partial dots).
@@ -98,7 +149,26 @@ class Word2VecConfig:
     sigmoid_mode: str = "exact"      # "exact" | "lut"
 
     def __post_init__(self) -> None:
+        if self.atomic_updates is not None:
+            self.update_mode = "atomic" if self.atomic_updates else "hogwild"
         self.validate()
+
+    def resolved_update_mode(self) -> str:
+        """update_mode, with the deprecated atomic_updates alias (possibly
+        assigned after construction) taking precedence when set."""
+        if self.atomic_updates is not None:
+            return "atomic" if self.atomic_updates else "hogwild"
+        return self.update_mode
+
+    def effective_atomic_below(self) -> int:
+        """Row-id threshold below which updates use atomics (the engines'
+        kernel argument): 0 = hogwild everywhere, 2^31-1 = everywhere."""
+        mode = self.resolved_update_mode()
+        if mode == "hogwild":
+            return 0
+        if mode == "atomic":
+            return 2 ** 31 - 1
+        return int(self.hybrid_hot_rows)
 
     def validate(self) -> None:
         if self.vector_size <= 0:
@@ -119,6 +189,10 @@ class Word2VecConfig:
             raise ValueError("unigram_table_size must be > 0")
         if self.dtype not in ("auto", "float32", "bfloat16"):
             raise ValueError(f"unsupported dtype {self.dtype!r}")
+        if self.update_mode not in ("hogwild", "atomic", "hybrid"):
+            raise ValueError(f"unsupported update_mode {self.update_mode!r}")
+        if self.hybrid_hot_rows < 0:
+            raise ValueError("hybrid_hot_rows must be >= 0")
         if self.window_mode not in ("canonical", "reference"):
             raise ValueError(f"unsupported window_mode {self.window_mode!r}")
         if self.engine not in ("auto", "fused", "dim", "row", "dp"):

@@ -161,6 +161,16 @@ class GlintWord2Vec:
         log.info("vocab: %d words, %d train words", vocab.num_words, vocab.train_words_count)
 
         max_sent = min(cfg.max_sentence_length, 1024)
+        if cfg.max_sentence_length > 1024:
+            # the GPU walker compacts each sentence into a 1024-entry LDS
+            # buffer (csrc/hip kMaxSent; 4 waves x 4 KB LDS per block), so
+            # longer sentences are chunked at 1024 — the same semantics the
+            # reference applies at its default maxSentenceLength=1000
+            # (mllib:88-97), just at a different boundary.
+            log.warning(
+                "max_sentence_length=%d exceeds the kernel sentence buffer; "
+                "sentences are chunked at 1024 tokens instead",
+                cfg.max_sentence_length)
         if native is not None:
             enc_tokens, enc_offsets = native.encode_corpus(
                 corpus, list(vocab.words), max_sent)

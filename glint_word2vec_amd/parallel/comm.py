@@ -16,11 +16,16 @@ import torch.distributed as dist
 
 def init_from_env(backend: Optional[str] = None) -> tuple[int, int]:
     """Initialise the default process group from torchrun env vars.
-    Returns (rank, world)."""
+    Returns (rank, world).
+
+    A world-1 run under torchrun (MASTER_ADDR set) still initialises the
+    group: RCCL communicator creation and every world-1 collective then
+    exercise the exact code path an 8-GPU launch uses — the production
+    path is the tested path, not a bypass."""
     if dist.is_initialized():
         return dist.get_rank(), dist.get_world_size()
     world = int(os.environ.get("WORLD_SIZE", "1"))
-    if world == 1:
+    if world == 1 and "MASTER_ADDR" not in os.environ:
         return 0, 1
     if backend is None:
         backend = "nccl" if torch.cuda.is_available() else "gloo"
@@ -57,6 +62,42 @@ def all_to_all_v(output_chunks: List[torch.Tensor],
             r.wait()
     else:
         dist.all_to_all(output_chunks, list(input_chunks))
+
+
+def all_to_all_single_v(output: torch.Tensor, input: torch.Tensor,
+                        out_splits, in_splits) -> None:
+    """Flat-buffer alltoallv: ``input`` rows [sum(in_splits), ...] where the
+    d-th contiguous segment goes to rank d; ``output`` receives rank s's
+    segment at the s-th position.  One RCCL op instead of world tensors —
+    the preferred form for the row-sharded pull/push exchanges."""
+    if not dist.is_initialized() or dist.get_world_size() == 1:
+        output.copy_(input)
+        return
+    if backend_is_gloo():
+        rank = dist.get_rank()
+        world = dist.get_world_size()
+        ob = [0]
+        ib = [0]
+        for s in range(world):
+            ob.append(ob[-1] + int(out_splits[s]))
+            ib.append(ib[-1] + int(in_splits[s]))
+        reqs = []
+        for peer in range(world):
+            if peer == rank:
+                output[ob[rank]:ob[rank + 1]].copy_(
+                    input[ib[rank]:ib[rank + 1]])
+                continue
+            if int(in_splits[peer]) > 0:
+                reqs.append(dist.isend(
+                    input[ib[peer]:ib[peer + 1]].contiguous(), peer))
+            if int(out_splits[peer]) > 0:
+                reqs.append(dist.irecv(output[ob[peer]:ob[peer + 1]], peer))
+        for r in reqs:
+            r.wait()
+    else:
+        dist.all_to_all_single(output, input,
+                               output_split_sizes=[int(x) for x in out_splits],
+                               input_split_sizes=[int(x) for x in in_splits])
 
 
 def exchange_counts(counts: torch.Tensor) -> torch.Tensor:

@@ -43,6 +43,7 @@ class DimShardedSgns:
                  table_size: int = 1_000_000, subsample: float = 0.0,
                  window_mode: str = "canonical", chunk_words: int = 1 << 19,
                  f_correction: bool = True, atomic: bool = True,
+                 atomic_below: "int | None" = None,
                  narrow: "bool | None" = None,
                  init_full_limit: int = 1 << 28):
         self.rank, self.world = (comm.init_from_env() if torch.distributed.is_available()
@@ -56,6 +57,10 @@ class DimShardedSgns:
         self.chunk_words = chunk_words
         self.f_correction = f_correction
         self.atomic = atomic
+        # row-id threshold for atomics (update_mode="hybrid"); None derives
+        # all-or-nothing from the atomic flag
+        self.atomic_below = ((2 ** 31 - 1 if atomic else 0)
+                             if atomic_below is None else int(atomic_below))
         self.lo, self.hi = slice_bounds(dim, self.rank, self.world)
         self.width = self.hi - self.lo
 
@@ -209,7 +214,7 @@ class DimShardedSgns:
                 poff[c[0]:c[1] + 1].data_ptr(), f.data_ptr(),
                 0 if f_loc is None else f_loc.data_ptr(),
                 float(self.dim) / max(self.width, 1),
-                (2 ** 31 - 1) if self.atomic else 0, self._stats.data_ptr(),
+                self.atomic_below, self._stats.data_ptr(),
                 nb, nt, comp.cuda_stream, 0, 0,
                 0 if self.serial else 1,
                 self.width if self.narrow else 0)

@@ -51,15 +51,14 @@ def train_gpu(cfg: Word2VecConfig, vocab: Vocabulary, batches_fn: Callable,
     rank, world = comm.init_from_env()
     engine = cfg.engine
     if engine == "auto":
-        if world == 1:
-            engine = "fused"
-        else:
-            # same policy as bench.py: replicate while both tables sit
-            # comfortably in 288 GB HBM3E (delta-allreduce dp is the
-            # fastest engine), shard by dimension beyond that
-            dtype_bytes = 4 if cfg.dtype == "float32" else 2
-            table_bytes = 2 * vocab.num_words * cfg.vector_size * dtype_bytes
-            engine = "dp" if table_bytes <= (8 << 30) else "dim"
+        # shared policy (config.choose_engine, same as bench.py): dp while
+        # the delta-allreduce merge stays cheap relative to the step,
+        # dim-sharded beyond (the crossover is xGMI merge bandwidth, not
+        # HBM capacity — see config.DP_MAX_TABLE_BYTES)
+        from ..config import choose_engine
+        dtype_bytes = 4 if cfg.dtype == "float32" else 2
+        engine = choose_engine(vocab.num_words, cfg.vector_size,
+                               dtype_bytes, world)
     if engine in ("dim", "row", "dp") and world >= 1:
         return _train_sharded(cfg, vocab, batches_fn, seed, engine, rank,
                               world, save_path, materialize, init_tables)
@@ -85,7 +84,6 @@ def train_gpu(cfg: Word2VecConfig, vocab: Vocabulary, batches_fn: Callable,
     sent_base = 0
     step = 0
     t0 = time.time()
-    prev_done = None
     for it in range(cfg.num_iterations):
         for batch in batches_fn():
             alpha = cfg.learning_rate * max(1e-4, 1.0 - processed / (total_words + 1))
@@ -93,11 +91,17 @@ def train_gpu(cfg: Word2VecConfig, vocab: Vocabulary, batches_fn: Callable,
                 tok = torch.from_numpy(batch.tokens).to(device, non_blocking=True)
                 off = torch.from_numpy(batch.offsets).to(device, non_blocking=True)
             compute_stream.wait_stream(copy_stream)
+            # tok/off were allocated on copy_stream; tell the caching
+            # allocator the compute-stream kernel reads them, so the
+            # buffers cannot be recycled for a later H2D copy while a
+            # queued kernel still reads them (the host can run many steps
+            # ahead when INFO logging is off)
+            tok.record_stream(compute_stream)
+            off.record_stream(compute_stream)
             gs.train_batch(tok, off, alpha, cfg.window, cfg.n, seed,
                            sent_id_base=sent_base, window_mode=cfg.window_mode,
-                           atomic=cfg.atomic_updates)
-            # keep tensors alive until the kernel is done
-            prev_done = (tok, off)
+                           atomic=cfg.resolved_update_mode() != "hogwild",
+                           atomic_below=cfg.effective_atomic_below())
             sent_base += batch.num_sentences
             processed += batch.num_tokens
             step += 1
@@ -150,7 +154,9 @@ def _train_sharded(cfg: Word2VecConfig, vocab: Vocabulary,
         eng = DimShardedSgns(vocab.num_words, cfg.vector_size,
                              chunk_words=cfg.chunk_words,
                              f_correction=cfg.f_correction,
-                             atomic=cfg.atomic_updates, **common)
+                             atomic=cfg.resolved_update_mode() != "hogwild",
+                             atomic_below=cfg.effective_atomic_below(),
+                             **common)
         if init_tables is not None:
             eng.load_host(*init_tables)
         for it in range(cfg.num_iterations):
@@ -172,11 +178,17 @@ def _train_sharded(cfg: Word2VecConfig, vocab: Vocabulary,
         from .replicated import ReplicatedSgns
         eng = ReplicatedSgns(vocab.num_words, cfg.vector_size,
                              sync_every=cfg.sync_every,
-                             atomic=cfg.atomic_updates, **common)
+                             atomic=cfg.resolved_update_mode() != "hogwild",
+                             atomic_below=cfg.effective_atomic_below(),
+                             **common)
         if init_tables is not None:
             eng.load_host(*init_tables)
         empty = (np.zeros(0, dtype=np.int32), np.zeros(1, dtype=np.int32))
-        sent_base = 10_000_000 * rank   # disjoint RNG streams per rank
+        # disjoint per-rank counter-RNG streams: rank in the high bits of
+        # the sentence-id space (2^48 sentences per rank before any
+        # overlap — unreachable), instead of a fixed decimal stride that a
+        # long-running rank could walk past
+        sent_base = rank << 48
         for it in range(cfg.num_iterations):
             # data parallel: round-robin whole batches across ranks
             batches = [b for i, b in enumerate(batches_fn())
@@ -208,7 +220,9 @@ def _train_sharded(cfg: Word2VecConfig, vocab: Vocabulary,
         from concurrent.futures import ThreadPoolExecutor
         from .row_sharded import RowShardedSgns
         eng = RowShardedSgns(vocab.num_words, cfg.vector_size,
-                             atomic=cfg.atomic_updates, **common)
+                             atomic=cfg.resolved_update_mode() == "atomic",
+                             atomic_below=cfg.effective_atomic_below(),
+                             **common)
         if init_tables is not None:
             eng.load_host(*init_tables)
         rng = np.random.default_rng(seed + 17 * rank)
@@ -235,7 +249,7 @@ def _train_sharded(cfg: Word2VecConfig, vocab: Vocabulary,
                     # RNG, milliseconds; the numpy planner costs ~45 s of
                     # host time per 2M-word batch).  Disjoint sentence-id
                     # streams per rank (data parallel).
-                    sbase = 10_000_000 * rank + sent_base
+                    sbase = (rank << 48) + sent_base
                     for k in range(n_steps):
                         tokens, offsets = batch_at(k)
                         plan = eng.make_plan_counter(
@@ -252,7 +266,7 @@ def _train_sharded(cfg: Word2VecConfig, vocab: Vocabulary,
                         _maybe_mid_checkpoint(
                             cfg, save_path, step,
                             lambda p: eng.save_checkpoint(p, cfg, vocab))
-                    sent_base = sbase - 10_000_000 * rank
+                    sent_base = sbase - (rank << 48)
                     continue
                 # CPU: prefetch next step's host plan while this step trains
                 next_plan = (pool.submit(eng.make_plan, *batch_at(0),

@@ -34,7 +34,7 @@ class ReplicatedSgns:
                  counts: Optional[np.ndarray] = None,
                  table_size: int = 1_000_000, subsample: float = 0.0,
                  window_mode: str = "canonical", sync_every: int = 4,
-                 atomic: bool = False):
+                 atomic: bool = False, atomic_below: "int | None" = None):
         self.rank, self.world = comm.init_from_env()
         self.vocab_size = vocab_size
         self.dim = dim
@@ -43,6 +43,8 @@ class ReplicatedSgns:
         self.window_mode = window_mode
         self.sync_every = max(1, sync_every)
         self.atomic = atomic
+        self.atomic_below = ((2 ** 31 - 1 if atomic else 0)
+                             if atomic_below is None else int(atomic_below))
         self._steps_since_sync = 0
         counts = (np.ones(vocab_size, dtype=np.int64) if counts is None
                   else counts)
@@ -81,7 +83,10 @@ class ReplicatedSgns:
             self.gs.train_batch(tokens, offsets, alpha, window, n_neg, seed,
                                 sent_id_base=sent_id_base,
                                 window_mode=self.window_mode,
-                                atomic=self.atomic)
+                                atomic=self.atomic,
+                                atomic_below=(None if self.atomic_below
+                                              >= 2 ** 31 - 1
+                                              else self.atomic_below))
         else:
             st = self.native.train_batch(
                 self.syn0.numpy(), self.syn1.numpy(),

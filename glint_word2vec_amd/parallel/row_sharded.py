@@ -46,6 +46,7 @@ class RowShardedSgns:
                  counts: Optional[np.ndarray] = None,
                  table_size: int = 1_000_000, subsample: float = 0.0,
                  window_mode: str = "canonical", atomic: bool = True,
+                 atomic_below: "int | None" = None,
                  init_full_limit: int = 1 << 28):
         self.rank, self.world = comm.init_from_env()
         self.vocab_size = vocab_size
@@ -56,8 +57,14 @@ class RowShardedSgns:
         self.window_mode = window_mode
         # cache-update mode for the GPU pairs kernel: True = fp32 atomics
         # (no lost updates; the PS adjust semantics), False = hogwild RMW
-        # (the fused kernel's default class; ~2.6x faster)
+        # (the fused kernel's default class; ~2.6x faster).
+        # atomic_below: hybrid threshold used by the world-1 DIRECT mode,
+        # where plan ids are global rows (sorted by count, so rows < K are
+        # the contended Zipf head); the cache modes use the plain bool
+        # (cache-local ids carry no frequency meaning).
         self.atomic = atomic
+        self.atomic_below = ((2 ** 31 - 1 if atomic else 0)
+                             if atomic_below is None else int(atomic_below))
         if self.is_cuda:
             from .. import _hip_native
             self.native = _hip_native
