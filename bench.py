@@ -190,6 +190,7 @@ def main():
     nsent = batch.num_sentences
 
     row_rng = np.random.default_rng(99 + rank)
+    row_pending = []   # pipelined row engine: one pull in flight
 
     def step(i):
         if dist_mode == "dp":
@@ -200,11 +201,20 @@ def main():
                 plan = trainer.make_plan_counter(
                     tok, off, args.window, args.neg, 99 + rank,
                     sent_id_base=i * nsent)
+            else:
+                plan = trainer._to_plan_t(trainer.make_plan(
+                    batch.tokens, batch.offsets, args.window, args.neg,
+                    row_rng))
+            if world == 1 and trainer.is_cuda:
                 trainer.train_step(batch.tokens, batch.offsets, alpha,
                                    args.window, args.neg, row_rng, plan=plan)
-            else:
-                trainer.train_step(batch.tokens, batch.offsets, alpha,
-                                   args.window, args.neg, row_rng)
+                return
+            # pipelined: issue this step's pull, then train+push the
+            # previous step (pull k+1 overlaps train k — row_sharded.py)
+            st = trainer.pull_begin(plan)
+            if row_pending:
+                trainer.train_push(row_pending.pop(), alpha)
+            row_pending.append(st)
         elif dist_mode:
             trainer.train_step(tok, off, alpha, args.window, args.neg,
                                seed=99, sent_id_base=i * nsent,
@@ -220,6 +230,8 @@ def main():
                                    else {"pair_mode": args.pair_mode}))
 
     def barrier_sync():
+        if row_pending:
+            trainer.train_push(row_pending.pop(), alpha)
         if not use_cpu:
             torch.cuda.synchronize(device)
         if distributed:

@@ -1729,13 +1729,17 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void train_pairs_kernel(
 // the one-pair kernel above.  T = float (world>1 delta caches) or
 // uint16_t/bf16 (world-1 native-dtype cache: half the bytes, the fused
 // kernel's precision class).
+// atomic_below (ATOMIC only): row ids < atomic_below take the atomic path,
+// the rest plain RMW — the hybrid update mode.  Meaningful when ids are
+// global words (direct mode: vocab sorted by count, ids < K = Zipf head);
+// cache modes pass INT32_MAX (all atomic).
 template <typename T, int NCH, bool ATOMIC>
 __global__ __launch_bounds__(64 * kWavesPerBlock) void train_pairs2_kernel(
     T* __restrict__ cache0, T* __restrict__ cache1, int64_t stride,
     const int32_t* __restrict__ group_center,
     const int64_t* __restrict__ group_offsets, int64_t num_groups,
     const int32_t* __restrict__ pair_target,
-    const float* __restrict__ pair_label, float alpha,
+    const float* __restrict__ pair_label, float alpha, int32_t atomic_below,
     unsigned long long* d_pairs, unsigned long long* d_positives,
     unsigned long long* d_words, double* d_sum_fplus) {
   const int lane = threadIdx.x & 63;
@@ -1759,7 +1763,8 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void train_pairs2_kernel(
       const int64_t my = p + half;
       const bool active = my < pe;
       const int64_t pid = active ? my : p;
-      T* t_ptr = cache1 + (int64_t)pair_target[pid] * stride;
+      const int32_t tid = pair_target[pid];
+      T* t_ptr = cache1 + (int64_t)tid * stride;
       float t_row[NCH];
       RowIO32<T, NCH>::load(t_ptr, t_row, l32);
       float f = 0.0f;
@@ -1772,7 +1777,7 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void train_pairs2_kernel(
 #pragma unroll
       for (int k = 0; k < NCH; ++k) grad[k] += gg * t_row[k];
       if (active) {
-        if (ATOMIC) {
+        if (ATOMIC && tid < atomic_below) {
           float delta[NCH];
 #pragma unroll
           for (int k = 0; k < NCH; ++k) delta[k] = gg * c_row[k];
@@ -1791,7 +1796,7 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void train_pairs2_kernel(
     }
 #pragma unroll
     for (int k = 0; k < NCH; ++k) grad[k] += __shfl_xor(grad[k], 32, 64);
-    if (ATOMIC) {
+    if (ATOMIC && group_center[g] < atomic_below) {
       if (half == 0) RowIO32<T, NCH>::atomic_add(c_ptr, grad, l32);
     } else if (half == 0) {
       float cur[NCH];
@@ -1843,6 +1848,82 @@ __global__ __launch_bounds__(256) void pull_average_kernel(
 #pragma unroll
     for (int k = 0; k < NC; ++k) acc[k] *= inv;
     RowIO<float, NC>::store(out + s * stride, acc, lane);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Row gather / scatter-add / subtract — the row-sharded engine's pull/push
+// hot ops (Glint pull / adjust), fused so the alltoallv wire
+// buffers are built in ONE pass at the shard's native dtype: no f32
+// inflation, no zero-fill pass, no torch temporaries.  Two rows per wave
+// (32-lane halves), dwordx2-coalesced via RowIO32.
+// ---------------------------------------------------------------------------
+template <typename T, int NCH>
+__global__ __launch_bounds__(256) void gather_rows_kernel(
+    const T* __restrict__ src, int64_t stride,
+    const int32_t* __restrict__ ids, int64_t n, T* __restrict__ dst) {
+  const int lane = threadIdx.x & 63;
+  const int l32 = lane & 31;
+  const int half = lane >> 5;
+  const int64_t wave_gid =
+      (((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6);
+  const int64_t total_hw = (((int64_t)gridDim.x * blockDim.x) >> 6) * 2;
+  for (int64_t i = 2 * wave_gid + half; i < n; i += total_hw) {
+    float v[NCH];
+    RowIO32<T, NCH>::load(src + (int64_t)ids[i] * stride, v, l32);
+    RowIO32<T, NCH>::store(dst + i * stride, v, l32);
+  }
+}
+
+// dst[ids[i]] += src[i]; atomics (fp32 atomicAdd / gfx950 packed-bf16) so
+// duplicate ids — the same shard row pulled by several ranks — sum exactly.
+template <typename T, int NCH>
+__global__ __launch_bounds__(256) void scatter_add_rows_kernel(
+    T* __restrict__ dst, int64_t stride, const int32_t* __restrict__ ids,
+    int64_t n, const T* __restrict__ src) {
+  const int lane = threadIdx.x & 63;
+  const int l32 = lane & 31;
+  const int half = lane >> 5;
+  const int64_t wave_gid =
+      (((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6);
+  const int64_t total_hw = (((int64_t)gridDim.x * blockDim.x) >> 6) * 2;
+  for (int64_t i = 2 * wave_gid + half; i < n; i += total_hw) {
+    float v[NCH];
+    RowIO32<T, NCH>::load(src + i * stride, v, l32);
+    RowIO32<T, NCH>::atomic_add(dst + (int64_t)ids[i] * stride, v, l32);
+  }
+}
+
+// out = a - b, flat elementwise over n2 element-PAIRS (strides are
+// multiples of 64 so totals are even); f32 math, one pass (the delta
+// computation for the push — torch would spill two f32 temporaries).
+__global__ __launch_bounds__(256) void sub_bf16_kernel(
+    const uint16_t* __restrict__ a, const uint16_t* __restrict__ b,
+    uint16_t* __restrict__ out, int64_t n2) {
+  const int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t tot = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = tid; i < n2; i += tot) {
+    const uint32_t pa = *reinterpret_cast<const uint32_t*>(a + 2 * i);
+    const uint32_t pb = *reinterpret_cast<const uint32_t*>(b + 2 * i);
+    v2bf16 d;
+    d[0] = (__bf16)(bf16_to_f32((uint16_t)(pa & 0xFFFF)) -
+                    bf16_to_f32((uint16_t)(pb & 0xFFFF)));
+    d[1] = (__bf16)(bf16_to_f32((uint16_t)(pa >> 16)) -
+                    bf16_to_f32((uint16_t)(pb >> 16)));
+    *reinterpret_cast<v2bf16*>(out + 2 * i) = d;
+  }
+}
+
+__global__ __launch_bounds__(256) void sub_f32_kernel(
+    const float* __restrict__ a, const float* __restrict__ b,
+    float* __restrict__ out, int64_t n2) {
+  const int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t tot = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = tid; i < n2; i += tot) {
+    const float2 pa = *reinterpret_cast<const float2*>(a + 2 * i);
+    const float2 pb = *reinterpret_cast<const float2*>(b + 2 * i);
+    *reinterpret_cast<float2*>(out + 2 * i) =
+        make_float2(pa.x - pb.x, pa.y - pb.y);
   }
 }
 
@@ -2244,12 +2325,15 @@ static void train_pairs(uintptr_t cache0, uintptr_t cache1, int64_t stride,
                         int64_t num_groups, uintptr_t pair_target,
                         uintptr_t pair_label, double alpha, uintptr_t stats,
                         int blocks, int threads, uintptr_t stream_ptr,
-                        int pair_mode, int atomic, int is_bf16) {
+                        int pair_mode, int atomic, int is_bf16,
+                        int64_t atomic_below) {
   HIP_CLEAR_ERROR();
   if (stride % 64 != 0) throw std::runtime_error("stride must be a multiple of 64");
   if (is_bf16 && pair_mode == 0)
     throw std::runtime_error("bf16 pairs cache requires pair_mode=1");
   const int nc = (int)(stride / 64);
+  const int32_t abelow =
+      (int32_t)std::min<int64_t>(atomic_below, 0x7FFFFFFFLL);
   hipStream_t stream = (hipStream_t)stream_ptr;
   unsigned long long* st = (unsigned long long*)stats;
   switch (nc) {
@@ -2274,7 +2358,7 @@ static void train_pairs(uintptr_t cache0, uintptr_t cache1, int64_t stride,
                            (const int32_t*)group_center,                      \
                            (const int64_t*)group_offsets, num_groups,         \
                            (const int32_t*)pair_target,                       \
-                           (const float*)pair_label, (float)alpha,            \
+                           (const float*)pair_label, (float)alpha, abelow,    \
                            st ? st + 0 : nullptr, st ? st + 1 : nullptr,      \
                            st ? st + 2 : nullptr,                             \
                            st ? (double*)(st + 3) : nullptr);                 \
@@ -2285,7 +2369,7 @@ static void train_pairs(uintptr_t cache0, uintptr_t cache1, int64_t stride,
                            (const int32_t*)group_center,                      \
                            (const int64_t*)group_offsets, num_groups,         \
                            (const int32_t*)pair_target,                       \
-                           (const float*)pair_label, (float)alpha,            \
+                           (const float*)pair_label, (float)alpha, abelow,    \
                            st ? st + 0 : nullptr, st ? st + 1 : nullptr,      \
                            st ? st + 2 : nullptr,                             \
                            st ? (double*)(st + 3) : nullptr);                 \
@@ -2297,7 +2381,7 @@ static void train_pairs(uintptr_t cache0, uintptr_t cache1, int64_t stride,
                          (const int32_t*)group_center,                        \
                          (const int64_t*)group_offsets, num_groups,           \
                          (const int32_t*)pair_target,                         \
-                         (const float*)pair_label, (float)alpha,              \
+                         (const float*)pair_label, (float)alpha, abelow,      \
                          st ? st + 0 : nullptr, st ? st + 1 : nullptr,        \
                          st ? st + 2 : nullptr,                               \
                          st ? (double*)(st + 3) : nullptr);                   \
@@ -2309,7 +2393,7 @@ static void train_pairs(uintptr_t cache0, uintptr_t cache1, int64_t stride,
                          (const int32_t*)group_center,                        \
                          (const int64_t*)group_offsets, num_groups,           \
                          (const int32_t*)pair_target,                         \
-                         (const float*)pair_label, (float)alpha,              \
+                         (const float*)pair_label, (float)alpha, abelow,      \
                          st ? st + 0 : nullptr, st ? st + 1 : nullptr,        \
                          st ? st + 2 : nullptr,                               \
                          st ? (double*)(st + 3) : nullptr);                   \
@@ -2387,6 +2471,91 @@ static void norms(uintptr_t syn0, int is_bf16, int64_t vocab, int64_t stride,
   HIP_CHECK(hipGetLastError());
 }
 
+template <typename T>
+static void launch_gather_scatter(const char* which, uintptr_t table,
+                                  int64_t stride, uintptr_t ids, int64_t n,
+                                  uintptr_t buf, int blocks,
+                                  hipStream_t stream) {
+  const int nc = (int)(stride / 64);
+  switch (nc) {
+#define CASE_NC(N)                                                           \
+  case N:                                                                    \
+    if (which[0] == 'g')                                                     \
+      hipLaunchKernelGGL((gather_rows_kernel<T, 2 * N>), dim3(blocks),       \
+                         dim3(256), 0, stream, (const T*)table, stride,      \
+                         (const int32_t*)ids, n, (T*)buf);                   \
+    else                                                                     \
+      hipLaunchKernelGGL((scatter_add_rows_kernel<T, 2 * N>), dim3(blocks),  \
+                         dim3(256), 0, stream, (T*)table, stride,            \
+                         (const int32_t*)ids, n, (const T*)buf);             \
+    return;
+    FOR_EACH_NC(CASE_NC)
+#undef CASE_NC
+    default:
+      throw std::runtime_error("unsupported NC");
+  }
+}
+
+static int rows_blocks(int64_t n) {
+  // 2 rows per wave, 4 waves per block -> n/8 blocks fills; cap 8192
+  return (int)std::max<int64_t>(1, std::min<int64_t>((n + 7) / 8, 8192));
+}
+
+static void gather_rows(uintptr_t src, int is_bf16, int64_t stride,
+                        uintptr_t ids, int64_t n, uintptr_t dst,
+                        uintptr_t stream_ptr) {
+  HIP_CLEAR_ERROR();
+  if (stride % 64 != 0)
+    throw std::runtime_error("stride must be a multiple of 64");
+  if (n == 0) return;
+  hipStream_t stream = (hipStream_t)stream_ptr;
+  if (is_bf16)
+    launch_gather_scatter<uint16_t>("g", src, stride, ids, n, dst,
+                                    rows_blocks(n), stream);
+  else
+    launch_gather_scatter<float>("g", src, stride, ids, n, dst,
+                                 rows_blocks(n), stream);
+  HIP_CHECK(hipGetLastError());
+}
+
+static void scatter_add_rows(uintptr_t dst, int is_bf16, int64_t stride,
+                             uintptr_t ids, int64_t n, uintptr_t src,
+                             uintptr_t stream_ptr) {
+  HIP_CLEAR_ERROR();
+  if (stride % 64 != 0)
+    throw std::runtime_error("stride must be a multiple of 64");
+  if (n == 0) return;
+  hipStream_t stream = (hipStream_t)stream_ptr;
+  if (is_bf16)
+    launch_gather_scatter<uint16_t>("s", dst, stride, ids, n, src,
+                                    rows_blocks(n), stream);
+  else
+    launch_gather_scatter<float>("s", dst, stride, ids, n, src,
+                                 rows_blocks(n), stream);
+  HIP_CHECK(hipGetLastError());
+}
+
+static void sub_rows(uintptr_t a, uintptr_t b, int is_bf16,
+                     int64_t total_elems, uintptr_t out,
+                     uintptr_t stream_ptr) {
+  HIP_CLEAR_ERROR();
+  if (total_elems % 2 != 0)
+    throw std::runtime_error("total_elems must be even");
+  if (total_elems == 0) return;
+  const int64_t n2 = total_elems / 2;
+  const int blocks =
+      (int)std::max<int64_t>(1, std::min<int64_t>((n2 + 255) / 256, 8192));
+  hipStream_t stream = (hipStream_t)stream_ptr;
+  if (is_bf16)
+    hipLaunchKernelGGL(sub_bf16_kernel, dim3(blocks), dim3(256), 0, stream,
+                       (const uint16_t*)a, (const uint16_t*)b, (uint16_t*)out,
+                       n2);
+  else
+    hipLaunchKernelGGL(sub_f32_kernel, dim3(blocks), dim3(256), 0, stream,
+                       (const float*)a, (const float*)b, (float*)out, n2);
+  HIP_CHECK(hipGetLastError());
+}
+
 static int round_stride(int dim) {
   int nc = supported_nc((dim + 63) / 64);
   if (nc < 0) throw std::runtime_error("dim too large (max 2048)");
@@ -2412,6 +2581,9 @@ PYBIND11_MODULE(_hip_native, m) {
   m.def("plan_emit", &plan_emit);
   m.def("pull_average", &pull_average);
   m.def("norms", &norms);
+  m.def("gather_rows", &gather_rows);
+  m.def("scatter_add_rows", &scatter_add_rows);
+  m.def("sub_rows", &sub_rows);
   m.def("round_stride", &round_stride);
   m.def("max_sentence_length", []() { return (int)kMaxSent; });
 }

@@ -217,7 +217,6 @@ def _train_sharded(cfg: Word2VecConfig, vocab: Vocabulary,
                     cfg, save_path, step,
                     lambda p: eng.save_checkpoint(p, cfg, vocab))
     else:  # row
-        from concurrent.futures import ThreadPoolExecutor
         from .row_sharded import RowShardedSgns
         eng = RowShardedSgns(vocab.num_words, cfg.vector_size,
                              atomic=cfg.resolved_update_mode() == "atomic",
@@ -227,69 +226,79 @@ def _train_sharded(cfg: Word2VecConfig, vocab: Vocabulary,
             eng.load_host(*init_tables)
         rng = np.random.default_rng(seed + 17 * rank)
         empty = (np.zeros(0, dtype=np.int32), np.zeros(1, dtype=np.int32))
-        pool = ThreadPoolExecutor(max_workers=1)
-        try:
-            for it in range(cfg.num_iterations):
-                batches = [b for i, b in enumerate(batches_fn())
-                           if i % world == rank]
-                # every rank must make the same number of collective calls
-                n_steps = len(batches)
-                if world > 1:
-                    t = torch.tensor([n_steps])
-                    torch.distributed.all_reduce(
-                        t, op=torch.distributed.ReduceOp.MAX)
-                    n_steps = int(t.item())
 
-                def batch_at(k):
-                    return ((batches[k].tokens, batches[k].offsets)
-                            if k < len(batches) else empty)
+        def alpha_now():
+            return cfg.learning_rate * max(
+                1e-4, 1.0 - processed / (total_words // world + 1))
 
-                if eng.is_cuda:
-                    # GPU planning via the fused kernel's walker (counter
-                    # RNG, milliseconds; the numpy planner costs ~45 s of
-                    # host time per 2M-word batch).  Disjoint sentence-id
-                    # streams per rank (data parallel).
-                    sbase = (rank << 48) + sent_base
-                    for k in range(n_steps):
-                        tokens, offsets = batch_at(k)
-                        plan = eng.make_plan_counter(
-                            tokens, offsets, cfg.window, cfg.n, seed,
-                            sent_id_base=sbase)
-                        sbase += max(len(offsets) - 1, 0)
-                        alpha = cfg.learning_rate * max(
-                            1e-4,
-                            1.0 - processed / (total_words // world + 1))
-                        eng.train_step(tokens, offsets, alpha, cfg.window,
-                                       cfg.n, rng, plan=plan)
-                        processed += len(tokens)
-                        step += 1
-                        _maybe_mid_checkpoint(
-                            cfg, save_path, step,
-                            lambda p: eng.save_checkpoint(p, cfg, vocab))
-                    sent_base = sbase - (rank << 48)
-                    continue
-                # CPU: prefetch next step's host plan while this step trains
-                next_plan = (pool.submit(eng.make_plan, *batch_at(0),
-                                         cfg.window, cfg.n, rng)
-                             if n_steps else None)
+        for it in range(cfg.num_iterations):
+            batches = [b for i, b in enumerate(batches_fn())
+                       if i % world == rank]
+            # every rank must make the same number of collective calls
+            n_steps = len(batches)
+            if world > 1:
+                t = torch.tensor([n_steps])
+                torch.distributed.all_reduce(
+                    t, op=torch.distributed.ReduceOp.MAX)
+                n_steps = int(t.item())
+
+            def batch_at(k):
+                return ((batches[k].tokens, batches[k].offsets)
+                        if k < len(batches) else empty)
+
+            # Disjoint per-rank sentence-id streams (data parallel).
+            sbase = (rank << 48) + sent_base
+            if eng.is_cuda and world == 1:
+                # direct mode: the pairs kernel trains straight on the
+                # padded shard tables — no pull cycle at all
                 for k in range(n_steps):
                     tokens, offsets = batch_at(k)
-                    plan = next_plan.result()
-                    if k + 1 < n_steps:
-                        next_plan = pool.submit(eng.make_plan,
-                                                *batch_at(k + 1),
-                                                cfg.window, cfg.n, rng)
-                    alpha = cfg.learning_rate * max(
-                        1e-4, 1.0 - processed / (total_words // world + 1))
-                    eng.train_step(tokens, offsets, alpha, cfg.window, cfg.n,
-                                   rng, plan=plan)
+                    plan = eng.make_plan_counter(
+                        tokens, offsets, cfg.window, cfg.n, seed,
+                        sent_id_base=sbase)
+                    sbase += max(len(offsets) - 1, 0)
+                    eng.train_step(tokens, offsets, alpha_now(), cfg.window,
+                                   cfg.n, rng, plan=plan)
                     processed += len(tokens)
                     step += 1
                     _maybe_mid_checkpoint(
                         cfg, save_path, step,
                         lambda p: eng.save_checkpoint(p, cfg, vocab))
-        finally:
-            pool.shutdown(wait=True)
+                sent_base = sbase - (rank << 48)
+                continue
+            # world > 1 (or CPU): pipelined pull/train/push — step k+1's
+            # pull is issued before step k's train+push, so the alltoallv
+            # row fetch overlaps the train kernel (the reference's
+            # dotprod/adjust pipeline, mllib:419-429); identical collective
+            # order on every rank.
+            pending = None   # (state, num_tokens)
+            for k in range(n_steps):
+                tokens, offsets = batch_at(k)
+                if eng.is_cuda:
+                    plan = eng.make_plan_counter(
+                        tokens, offsets, cfg.window, cfg.n, seed,
+                        sent_id_base=sbase)
+                else:
+                    plan = eng._to_plan_t(eng.make_plan(
+                        tokens, offsets, cfg.window, cfg.n, rng))
+                sbase += max(len(offsets) - 1, 0)
+                state = eng.pull_begin(plan)
+                if pending is not None:
+                    eng.train_push(pending[0], alpha_now())
+                    processed += pending[1]
+                    step += 1
+                    _maybe_mid_checkpoint(
+                        cfg, save_path, step,
+                        lambda p: eng.save_checkpoint(p, cfg, vocab))
+                pending = (state, len(tokens))
+            if pending is not None:
+                eng.train_push(pending[0], alpha_now())
+                processed += pending[1]
+                step += 1
+                _maybe_mid_checkpoint(
+                    cfg, save_path, step,
+                    lambda p: eng.save_checkpoint(p, cfg, vocab))
+            sent_base = sbase - (rank << 48)
     st = eng.read_stats()
     dt = time.time() - t0
     log.info("%s-sharded training (rank %d/%d): %d words in %.2fs, %d pairs, "

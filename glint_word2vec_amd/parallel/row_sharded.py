@@ -189,22 +189,6 @@ class RowShardedSgns:
         out[perm] = sorted_rows
         return out
 
-    def _pull_cache(self, ids, which: int) -> torch.Tensor:
-        """Training-cache pull.  World-1 bf16 hogwild keeps the cache in
-        the shard's native dtype (half the kernel bytes — the fused
-        kernel's precision class); every other case uses the f32 pull."""
-        if (self.world == 1 and self.is_cuda and self.is_bf16
-                and not self.atomic and not self.serial):
-            shard = self.syn0 if which == 0 else self.syn1
-            idx = self._to_ids(ids)
-            out = torch.empty((idx.numel(), self.cache_stride),
-                              dtype=shard.dtype, device=self.device)
-            B = 1 << 22
-            for i in range(0, idx.numel(), B):
-                out[i:i + B] = shard.index_select(0, idx[i:i + B])
-            return out
-        return self.pull(ids, which)
-
     def push_add(self, ids, deltas: torch.Tensor, which: int) -> None:
         """Scatter-add row deltas back to their owners (the adjust push)."""
         shard = self.syn0 if which == 0 else self.syn1
@@ -338,62 +322,277 @@ class RowShardedSgns:
         group_center = center[starts]
         return sgns.GroupedPlanT(group_center, group_offsets, target, label)
 
+    # ------------------------------------------------------------------
+    # Pipelined pull/train/push cycle (the Glint dotprod/adjust message
+    # pipeline, mllib:419-429: the next batch's dotprod overlaps the
+    # previous batch's adjust).  pull_begin(k+1) may be issued before
+    # train_push(k): its row reads then predate step k's push — one step
+    # of hogwild staleness, exactly the reference's fire-and-forget
+    # semantics.  All collectives are enqueued in a deterministic order
+    # (pull k+1 before push k on every rank), on a dedicated stream so
+    # xGMI transfers overlap the compute-stream train kernel.
+    # ------------------------------------------------------------------
+    def _route32(self, ids: torch.Tensor):
+        """Sort unique int32 ids by owner (id % world).  Returns
+        (pos, send_counts, sorted_local_ids_i32): pos maps an id's position
+        in `ids` to its position in the owner-sorted order."""
+        owner = torch.remainder(ids, self.world)
+        perm = torch.argsort(owner, stable=True)
+        counts = torch.bincount(owner, minlength=self.world)
+        local = torch.div(ids.index_select(0, perm), self.world,
+                          rounding_mode="floor").int().contiguous()
+        pos = torch.empty(ids.numel(), dtype=torch.int32, device=ids.device)
+        pos.index_copy_(0, perm, torch.arange(ids.numel(), dtype=torch.int32,
+                                              device=ids.device))
+        return pos, counts, local
+
+    def _exchange_counts2(self, c0: torch.Tensor, c1: torch.Tensor):
+        """One allgather for both tables' request counts.  Returns
+        (recv0, recv1) as host numpy arrays."""
+        both = torch.stack([c0, c1]).to(self.device)
+        if self.world == 1 or not torch.distributed.is_initialized():
+            h = both.cpu().numpy()
+            return h[0], h[1]
+        gathered = [torch.empty_like(both) for _ in range(self.world)]
+        torch.distributed.all_gather(gathered, both)
+        h = [g.cpu().numpy() for g in gathered]
+        r0 = np.array([h[s][0][self.rank] for s in range(self.world)])
+        r1 = np.array([h[s][1][self.rank] for s in range(self.world)])
+        return r0, r1
+
+    def _gather_native(self, shard: torch.Tensor, idx_i32: torch.Tensor,
+                       out: torch.Tensor) -> None:
+        """out[i] = shard[idx[i]] at native dtype (fused HIP gather on GPU:
+        one pass, no f32 inflation, padding columns travel as-is)."""
+        n = int(idx_i32.numel())
+        if n == 0:
+            return
+        if self.is_cuda:
+            s = torch.cuda.current_stream(self.device)
+            self.native.gather_rows(shard.data_ptr(), int(self.is_bf16),
+                                    self.store_stride, idx_i32.data_ptr(), n,
+                                    out.data_ptr(), s.cuda_stream)
+        else:
+            out.copy_(shard.index_select(0, idx_i32.long()))
+
+    def _scatter_add_native(self, shard: torch.Tensor, idx_i32: torch.Tensor,
+                            deltas: torch.Tensor) -> None:
+        """shard[idx[i]] += deltas[i] (atomic: duplicate ids across source
+        ranks sum exactly — the adjust semantics)."""
+        n = int(idx_i32.numel())
+        if n == 0:
+            return
+        if self.is_cuda:
+            s = torch.cuda.current_stream(self.device)
+            self.native.scatter_add_rows(shard.data_ptr(), int(self.is_bf16),
+                                         self.store_stride,
+                                         idx_i32.data_ptr(), n,
+                                         deltas.data_ptr(), s.cuda_stream)
+        else:
+            shard.index_add_(0, idx_i32.long(), deltas)
+
+    def _sub_native(self, a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+        """a - b at native dtype, one fused pass (the push delta)."""
+        out = torch.empty_like(a)
+        if a.numel() == 0:
+            return out
+        if self.is_cuda:
+            s = torch.cuda.current_stream(self.device)
+            self.native.sub_rows(a.data_ptr(), b.data_ptr(),
+                                 int(self.is_bf16), a.numel(),
+                                 out.data_ptr(), s.cuda_stream)
+        else:
+            torch.sub(a, b, out=out)
+        return out
+
+    def _comm_ctx(self):
+        """Stream the pull/push collectives run on (CUDA: dedicated stream
+        overlapping the compute stream's train kernel)."""
+        if not self.is_cuda:
+            import contextlib
+            return contextlib.nullcontext()
+        if not hasattr(self, "_pull_stream"):
+            self._pull_stream = torch.cuda.Stream(self.device)
+        return torch.cuda.stream(self._pull_stream)
+
+    def pull_begin(self, plan) -> dict:
+        """Stage 1 of a step: route the plan's unique rows to their owners,
+        exchange indices, gather + exchange the rows (native dtype on the
+        wire — bf16 halves xGMI bytes).  Returns the state consumed by
+        train_push().  May be called for step k+1 before train_push(k)."""
+        dev = self.device
+        tdtype = self.syn0.dtype
+        uc, inv_c = torch.unique(plan.group_center, return_inverse=True)
+        ut, inv_t = torch.unique(plan.pair_target, return_inverse=True)
+        st = {"plan": plan}
+        if self.world == 1:
+            # no routing: gather straight from the shard (serving-path
+            # probes; production world-1 training uses the direct mode)
+            cache0 = torch.empty((uc.numel(), self.store_stride),
+                                 dtype=tdtype, device=dev)
+            cache1 = torch.empty((ut.numel(), self.store_stride),
+                                 dtype=tdtype, device=dev)
+            self._gather_native(self.syn0, uc.int().contiguous(), cache0)
+            self._gather_native(self.syn1, ut.int().contiguous(), cache1)
+            st["gc"] = inv_c.int().contiguous()
+            st["gt"] = inv_t.int().contiguous()
+            st["cache0"], st["cache1"] = cache0, cache1
+            st["orig0"] = st["orig1"] = None
+            st["uc_sorted"] = uc
+            st["ut_sorted"] = ut
+            return st
+        pos0, cnt0, loc0 = self._route32(uc)
+        pos1, cnt1, loc1 = self._route32(ut)
+        rc0, rc1 = self._exchange_counts2(cnt0, cnt1)
+        cnt0_h = cnt0.cpu().numpy()
+        cnt1_h = cnt1.cpu().numpy()
+        if self.is_cuda:
+            ev = torch.cuda.Event()
+            ev.record(torch.cuda.current_stream(dev))
+        with self._comm_ctx():
+            if self.is_cuda:
+                torch.cuda.current_stream(dev).wait_event(ev)
+            # index request exchange (int32 local rows)
+            idx_recv0 = torch.empty(int(rc0.sum()), dtype=torch.int32,
+                                    device=dev)
+            idx_recv1 = torch.empty(int(rc1.sum()), dtype=torch.int32,
+                                    device=dev)
+            comm.all_to_all_single_v(idx_recv0, loc0, rc0, cnt0_h)
+            comm.all_to_all_single_v(idx_recv1, loc1, rc1, cnt1_h)
+            # owners gather requested rows; rows return in owner-major order
+            rows_send0 = torch.empty((idx_recv0.numel(), self.store_stride),
+                                     dtype=tdtype, device=dev)
+            rows_send1 = torch.empty((idx_recv1.numel(), self.store_stride),
+                                     dtype=tdtype, device=dev)
+            self._gather_native(self.syn0, idx_recv0, rows_send0)
+            self._gather_native(self.syn1, idx_recv1, rows_send1)
+            cache0 = torch.empty((uc.numel(), self.store_stride),
+                                 dtype=tdtype, device=dev)
+            cache1 = torch.empty((ut.numel(), self.store_stride),
+                                 dtype=tdtype, device=dev)
+            comm.all_to_all_single_v(cache0, rows_send0, cnt0_h, rc0)
+            comm.all_to_all_single_v(cache1, rows_send1, cnt1_h, rc1)
+            if self.is_cuda:
+                done = torch.cuda.Event()
+                done.record(torch.cuda.current_stream(dev))
+                st["pull_done"] = done
+        # cache rows are owner-sorted; remap the plan's inverse indices
+        st["gc"] = pos0.index_select(0, inv_c).contiguous()
+        st["gt"] = pos1.index_select(0, inv_t).contiguous()
+        st["cache0"], st["cache1"] = cache0, cache1
+        st["orig0"] = cache0.clone()
+        st["orig1"] = cache1.clone()
+        st["idx_recv"] = (idx_recv0, idx_recv1)
+        st["splits"] = (cnt0_h, rc0, cnt1_h, rc1)
+        return st
+
+    def train_push(self, st: dict, alpha: float) -> None:
+        """Stage 2: train the pairs kernel on the pulled cache (compute
+        stream), compute native-dtype deltas, exchange them back and
+        scatter-add into the owners' shards (adjust).  Push collectives are
+        enqueued after any pull_begin already issued — same order on every
+        rank."""
+        plan = st["plan"]
+        if self.is_cuda and "pull_done" in st:
+            torch.cuda.current_stream(self.device).wait_event(st["pull_done"])
+        if plan.num_pairs > 0:
+            self._train_pairs_cache(st["cache0"], st["cache1"], st["gc"],
+                                    plan.group_offsets, st["gt"],
+                                    plan.pair_label, alpha)
+        if self.world == 1:
+            if st["uc_sorted"].numel():
+                self._write_back(st["uc_sorted"], st["cache0"], 0)
+            if st["ut_sorted"].numel():
+                self._write_back(st["ut_sorted"], st["cache1"], 1)
+            return
+        cnt0_h, rc0, cnt1_h, rc1 = st["splits"]
+        idx_recv0, idx_recv1 = st["idx_recv"]
+        delta0 = self._sub_native(st["cache0"], st["orig0"])
+        delta1 = self._sub_native(st["cache1"], st["orig1"])
+        if self.is_cuda:
+            ev = torch.cuda.Event()
+            ev.record(torch.cuda.current_stream(self.device))
+        with self._comm_ctx():
+            if self.is_cuda:
+                torch.cuda.current_stream(self.device).wait_event(ev)
+            del_recv0 = torch.empty((idx_recv0.numel(), self.store_stride),
+                                    dtype=delta0.dtype, device=self.device)
+            del_recv1 = torch.empty((idx_recv1.numel(), self.store_stride),
+                                    dtype=delta1.dtype, device=self.device)
+            comm.all_to_all_single_v(del_recv0, delta0, rc0, cnt0_h)
+            comm.all_to_all_single_v(del_recv1, delta1, rc1, cnt1_h)
+            self._scatter_add_native(self.syn0, idx_recv0, del_recv0)
+            self._scatter_add_native(self.syn1, idx_recv1, del_recv1)
+            if self.is_cuda:
+                # later compute-stream work (next train) must not overtake
+                push_done = torch.cuda.Event()
+                push_done.record(torch.cuda.current_stream(self.device))
+                self._last_push = push_done
+
+    def _train_pairs_cache(self, cache0, cache1, gc, go, gt, pl, alpha):
+        """Pairs kernel on the native-dtype cache (GPU) / f32 cache (CPU).
+        Cache ids are local, so the hybrid threshold does not apply here —
+        atomic is all-or-nothing (self.atomic)."""
+        if self.is_cuda:
+            G = int(gc.numel())
+            nb = 1 if self.serial else max(1, min((G + 3) // 4, 8192))
+            nt = 64 if self.serial else 256
+            if self.serial and self.is_bf16:
+                raise ValueError("serial parity mode requires fp32 caches")
+            stream = torch.cuda.current_stream(self.device)
+            go = go.contiguous()
+            pl = pl.contiguous()
+            self.native.train_pairs(
+                cache0.data_ptr(), cache1.data_ptr(), self.store_stride,
+                gc.data_ptr(), go.data_ptr(), G, gt.data_ptr(), pl.data_ptr(),
+                float(alpha), self._stats.data_ptr(), nb, nt,
+                stream.cuda_stream, 0 if self.serial else 1,
+                int(self.atomic), int(self.is_bf16), 2 ** 31 - 1)
+            self._inflight = (gc, go, gt, pl, cache0, cache1)
+        else:
+            st = self.native.train_pairs(
+                cache0.numpy(), cache1.numpy(), gc.numpy(),
+                go.numpy() if hasattr(go, "numpy") else go,
+                gt.numpy(), pl.numpy() if hasattr(pl, "numpy") else pl,
+                float(alpha))
+            for k in ("pairs", "positives", "words_trained"):
+                self._cpu_stats[k] += st[k]
+            self._cpu_stats["sum_fplus"] += st["sum_fplus"]
+
+    def _to_plan_t(self, plan) -> "sgns.GroupedPlanT":
+        if isinstance(plan, sgns.GroupedPlanT):
+            return plan
+        dev = self.device
+        return sgns.GroupedPlanT(
+            torch.from_numpy(np.ascontiguousarray(
+                plan.group_center, dtype=np.int32)).to(dev),
+            torch.from_numpy(np.ascontiguousarray(
+                plan.group_offsets, dtype=np.int64)).to(dev),
+            torch.from_numpy(np.ascontiguousarray(
+                plan.pair_target, dtype=np.int32)).to(dev),
+            torch.from_numpy(np.ascontiguousarray(
+                plan.pair_label, dtype=np.float32)).to(dev))
+
     def train_step(self, tokens: np.ndarray, offsets: np.ndarray,
                    alpha: float, window: int, n_neg: int,
                    rng: np.random.Generator, plan=None) -> None:
-        """One data-parallel step over this rank's batch.  ``plan`` may be
-        a host GroupedPlan or a device GroupedPlanT (make_plan_device).
-        Ranks with no data still participate in the collectives."""
+        """One (unpipelined) data-parallel step over this rank's batch.
+        ``plan`` may be a host GroupedPlan or a device GroupedPlanT.
+        Ranks with no data still participate in the collectives.  The
+        pipelined form is pull_begin()/train_push() (engine.py uses it)."""
         if plan is None:
             plan = self.make_plan(tokens, offsets, window, n_neg, rng)
-        if isinstance(plan, sgns.GroupedPlanT):
-            if self.world == 1 and self.is_cuda and not self.serial:
-                # world 1: row r = word r and the shard is stored at the
-                # kernel stride — train DIRECTLY on the tables (no
-                # unique/pull/write-back at all)
-                self._train_pairs_direct(plan, alpha)
-                return
-            # unique on int32 ids: half the radix-sort bytes of .long()
-            uc, inv_c = torch.unique(plan.group_center,
-                                     return_inverse=True)
-            ut, inv_t = torch.unique(plan.pair_target,
-                                     return_inverse=True)
-            cache0 = self._pull_cache(uc, 0)
-            cache1 = self._pull_cache(ut, 1)
-            # world 1: the trained cache IS the new row value — write it
-            # back directly and skip the orig clones + delta temps
-            # (3 extra cache-sized buffers; matters at 80M-vocab scale)
-            local = self.world == 1
-            orig0 = None if local else cache0.clone()
-            orig1 = None if local else cache1.clone()
-            if plan.num_pairs > 0:
-                self._train_pairs_t(cache0, cache1, inv_c.int(),
-                                    plan.group_offsets, inv_t.int(),
-                                    plan.pair_label, alpha)
-            if local:
-                self._write_back(uc, cache0, 0)
-                self._write_back(ut, cache1, 1)
-            else:
-                self.push_add(uc, cache0 - orig0, 0)
-                self.push_add(ut, cache1 - orig1, 1)
+        plan = self._to_plan_t(plan)
+        if (self.world == 1 and self.is_cuda and not self.serial
+                and getattr(self, "use_direct", True)
+                and isinstance(plan, sgns.GroupedPlanT)):
+            # world 1: row r = word r and the shard is stored at the
+            # kernel stride — train DIRECTLY on the tables (no
+            # unique/pull/write-back at all)
+            self._train_pairs_direct(plan, alpha)
             return
-        uc, inv_c = np.unique(plan.group_center, return_inverse=True)
-        ut, inv_t = np.unique(plan.pair_target, return_inverse=True)
-        cache0 = self.pull(uc.astype(np.int64), 0)
-        cache1 = self.pull(ut.astype(np.int64), 1)
-        local = self.world == 1
-        orig0 = None if local else cache0.clone()
-        orig1 = None if local else cache1.clone()
-        if plan.num_pairs > 0:
-            self._train_pairs(cache0, cache1, inv_c.astype(np.int32),
-                              plan.group_offsets,
-                              inv_t.astype(np.int32), plan.pair_label, alpha)
-        if local:
-            self._write_back(self._to_ids(uc.astype(np.int64)), cache0, 0)
-            self._write_back(self._to_ids(ut.astype(np.int64)), cache1, 1)
-        else:
-            self.push_add(uc.astype(np.int64), cache0 - orig0, 0)
-            self.push_add(ut.astype(np.int64), cache1 - orig1, 1)
+        st = self.pull_begin(plan)
+        self.train_push(st, alpha)
 
     def _write_back(self, ids: torch.Tensor, rows: torch.Tensor,
                     which: int) -> None:
@@ -413,73 +612,16 @@ class RowShardedSgns:
         go = plan.group_offsets.contiguous()
         pt = plan.pair_target.contiguous()
         pl = plan.pair_label.contiguous()
-        pair_mode = 0 if (self.atomic and not self.is_bf16) else 1
+        # direct mode: plan ids are global words, so the hybrid
+        # atomic_below threshold applies (atomics on the Zipf head only)
+        atomic_flag = self.atomic_below > 0
         self.native.train_pairs(
             self.syn0.data_ptr(), self.syn1.data_ptr(), self.store_stride,
             gc.data_ptr(), go.data_ptr(), G, pt.data_ptr(), pl.data_ptr(),
             float(alpha), self._stats.data_ptr(), nb, 256,
-            stream.cuda_stream, pair_mode, int(self.atomic),
-            int(self.is_bf16))
+            stream.cuda_stream, 1, int(atomic_flag),
+            int(self.is_bf16), self.atomic_below)
         self._inflight = (gc, go, pt, pl)
-
-    def _train_pairs_t(self, cache0, cache1, group_center, group_offsets,
-                       pair_target, pair_label, alpha):
-        """Device-tensor twin of _train_pairs: no host conversion, no
-        stream sync (everything stays queued on the compute stream)."""
-        if self.is_cuda:
-            G = int(group_center.numel())
-            nb = 1 if self.serial else max(1, min((G + 3) // 4, 8192))
-            nt = 64 if self.serial else 256
-            stream = torch.cuda.current_stream(self.device)
-            gc = group_center.contiguous()
-            go = group_offsets.contiguous()
-            pt = pair_target.contiguous()
-            pl = pair_label.contiguous()
-            self.native.train_pairs(
-                cache0.data_ptr(), cache1.data_ptr(), self.cache_stride,
-                gc.data_ptr(), go.data_ptr(), G, pt.data_ptr(), pl.data_ptr(),
-                float(alpha), self._stats.data_ptr(), nb, nt,
-                stream.cuda_stream,
-                0 if (self.serial or self.atomic) else 1,
-                int(self.atomic), int(cache0.dtype == torch.bfloat16))
-            # keep args alive until the kernel drains
-            self._inflight = (gc, go, pt, pl, cache0, cache1)
-        else:
-            st = self.native.train_pairs(
-                cache0.numpy(), cache1.numpy(), group_center.numpy(),
-                group_offsets.numpy(), pair_target.numpy(),
-                pair_label.numpy(), float(alpha))
-            for k in ("pairs", "positives", "words_trained"):
-                self._cpu_stats[k] += st[k]
-            self._cpu_stats["sum_fplus"] += st["sum_fplus"]
-
-    def _train_pairs(self, cache0, cache1, group_center, group_offsets,
-                     pair_target, pair_label, alpha):
-        if self.is_cuda:
-            gc = torch.from_numpy(group_center).to(self.device)
-            go = torch.from_numpy(group_offsets.astype(np.int64)).to(self.device)
-            pt = torch.from_numpy(pair_target).to(self.device)
-            pl = torch.from_numpy(pair_label).to(self.device)
-            stream = torch.cuda.current_stream(self.device)
-            G = len(group_center)
-            nb = 1 if self.serial else max(1, min((G + 3) // 4, 8192))
-            nt = 64 if self.serial else 256
-            self.native.train_pairs(
-                cache0.data_ptr(), cache1.data_ptr(), self.cache_stride,
-                gc.data_ptr(), go.data_ptr(), G, pt.data_ptr(), pl.data_ptr(),
-                float(alpha), self._stats.data_ptr(), nb, nt,
-                stream.cuda_stream,
-                0 if (self.serial or self.atomic) else 1,
-                int(self.atomic), 0)
-            torch.cuda.current_stream(self.device).synchronize()
-        else:
-            st = self.native.train_pairs(cache0.numpy(), cache1.numpy(),
-                                         group_center,
-                                         group_offsets.astype(np.int64),
-                                         pair_target, pair_label, float(alpha))
-            for k in ("pairs", "positives", "words_trained"):
-                self._cpu_stats[k] += st[k]
-            self._cpu_stats["sum_fplus"] += st["sum_fplus"]
 
     # ------------------------------------------------------------------
     def read_stats(self, reset: bool = True) -> GpuStats:
