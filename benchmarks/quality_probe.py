@@ -66,7 +66,13 @@ def main():
     ap.add_argument("--dim", type=int, default=128)
     ap.add_argument("--words", type=int, default=50_000_000)
     ap.add_argument("--sentence-len", type=int, default=100)
-    ap.add_argument("--mode", choices=["hogwild", "atomic"], default="hogwild")
+    ap.add_argument("--mode",
+                    choices=["hogwild", "atomic", "hybrid", "positives"],
+                    default="hogwild")
+    ap.add_argument("--hot-rows", type=int, default=1 << 18,
+                    help="hybrid: atomics for rows < K (word id ~ 2*Zipf "
+                         "rank in this corpus, so K covers the K/2 hottest "
+                         "concepts)")
     ap.add_argument("--device", choices=["cuda", "cpu"], default="cuda")
     ap.add_argument("--lr", type=float, default=0.025)
     ap.add_argument("--neg", type=int, default=5)
@@ -93,8 +99,10 @@ def main():
             gs.set_subsample(counts, int(counts.sum()), args.subsample)
         tok = torch.from_numpy(tokens).cuda()
         off = torch.from_numpy(offsets).cuda()
+        abelow = {"hogwild": None, "atomic": None, "positives": -1,
+                  "hybrid": args.hot_rows}[args.mode]
         gs.train_batch(tok, off, args.lr, args.window, args.neg, 42,
-                       atomic=(args.mode == "atomic"))
+                       atomic=(args.mode != "hogwild"), atomic_below=abelow)
         torch.cuda.synchronize()
         st = gs.read_stats()
         syn0, _ = gs.to_host()
@@ -110,7 +118,9 @@ def main():
     dt = time.time() - t0
     # eval on mid-frequency concepts (Zipf rank = concept id)
     acc = nn_accuracy(syn0, range(args.eval_lo, args.eval_hi))
-    print(f"mode={args.mode} device={args.device} vocab={args.vocab} "
+    mode = (f"hybrid[K={args.hot_rows}]" if args.mode == "hybrid"
+            else args.mode)
+    print(f"mode={mode} device={args.device} vocab={args.vocab} "
           f"words={args.words}: planted-NN acc={acc:.3f} "
           f"({args.words / dt / 1e6:.1f}M words/s incl. setup)")
 

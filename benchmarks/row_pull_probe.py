@@ -10,6 +10,9 @@ copies — and times each stage with CUDA events:
 Run: python benchmarks/row_pull_probe.py --vocab 80000000 [--steps 5]
 """
 import argparse
+import os
+import sys
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import json
 import time
 
