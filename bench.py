@@ -48,6 +48,10 @@ def parse_args():
     p.add_argument("--hot-rows", type=int, default=None,
                    help="hybrid mode: rows < K use atomics (default: "
                         "Word2VecConfig.hybrid_hot_rows)")
+    p.add_argument("--hot-floor", type=int, default=None,
+                   help="hybrid mode: rows < F stay hogwild (ultra-head "
+                        "contention escape; default "
+                        "Word2VecConfig.hybrid_skip_rows)")
     p.add_argument("--atomic", action="store_true",
                    help="alias for --updates atomic")
     p.add_argument("--atomic-below", type=int, default=None,
@@ -84,6 +88,9 @@ def main():
         args.hot_rows = args.atomic_below
     if args.hot_rows is None:
         args.hot_rows = Word2VecConfig.hybrid_hot_rows
+    if args.hot_floor is None:
+        args.hot_floor = (Word2VecConfig.hybrid_skip_rows
+                          if args.updates == "hybrid" else 0)
     upd_atomic = args.updates != "hogwild"       # kernel atomic flag
     upd_below = (0 if args.updates == "hogwild" else
                  (2 ** 31 - 1 if args.updates == "atomic" else args.hot_rows))
@@ -154,7 +161,8 @@ def main():
                                  table_size=args.table_size,
                                  subsample=args.subsample,
                                  chunk_words=args.chunk_words,
-                                 atomic=upd_atomic, atomic_below=upd_below)
+                                 atomic=upd_atomic, atomic_below=upd_below,
+                                 atomic_floor=args.hot_floor)
         dist_mode = True
     elif engine == "dp":
         from glint_word2vec_amd.parallel.replicated import ReplicatedSgns
@@ -163,7 +171,8 @@ def main():
                                  table_size=args.table_size,
                                  subsample=args.subsample,
                                  sync_every=args.sync_every,
-                                 atomic=upd_atomic, atomic_below=upd_below)
+                                 atomic=upd_atomic, atomic_below=upd_below,
+                                 atomic_floor=args.hot_floor)
         dist_mode = "dp"
     elif engine == "row":
         from glint_word2vec_amd.parallel.row_sharded import RowShardedSgns
@@ -172,7 +181,8 @@ def main():
                                  table_size=args.table_size,
                                  subsample=args.subsample,
                                  atomic=args.updates == "atomic",
-                                 atomic_below=upd_below)
+                                 atomic_below=upd_below,
+                                 atomic_floor=args.hot_floor)
         dist_mode = "row"
     else:
         from glint_word2vec_amd.ops.gpu import GpuSgns
@@ -225,6 +235,7 @@ def main():
                                 atomic=upd_atomic,
                                 atomic_below=(None if args.updates == "atomic"
                                               else upd_below),
+                                atomic_floor=args.hot_floor,
                                 blocks=args.blocks or None,
                                 **({} if args.pair_mode is None
                                    else {"pair_mode": args.pair_mode}))
@@ -291,7 +302,8 @@ def main():
                 "parallelism": (f"{engine}shard-rccl-x{world}" if dist_mode
                                 else "hogwild-1gpu"),
                 "updates": (args.updates if args.updates != "hybrid" else
-                            f"hybrid(atomic rows<{args.hot_rows})"),
+                            f"hybrid(atomic rows "
+                            f"{args.hot_floor}..{args.hot_rows})"),
             },
             "pairs_per_step": st.pairs / max(args.steps, 1),
             "mean_fplus": st.sum_fplus / max(st.positives, 1),

@@ -69,7 +69,9 @@ def main():
     ap.add_argument("--mode",
                     choices=["hogwild", "atomic", "hybrid", "positives"],
                     default="hogwild")
-    ap.add_argument("--hot-rows", type=int, default=1 << 18,
+    ap.add_argument("--hot-floor", type=int, default=0,
+                    help="hybrid: rows < F stay hogwild (contention escape)")
+    ap.add_argument("--hot-rows", type=int, default=8192,
                     help="hybrid: atomics for rows < K (word id ~ 2*Zipf "
                          "rank in this corpus, so K covers the K/2 hottest "
                          "concepts)")
@@ -102,7 +104,9 @@ def main():
         abelow = {"hogwild": None, "atomic": None, "positives": -1,
                   "hybrid": args.hot_rows}[args.mode]
         gs.train_batch(tok, off, args.lr, args.window, args.neg, 42,
-                       atomic=(args.mode != "hogwild"), atomic_below=abelow)
+                       atomic=(args.mode != "hogwild"), atomic_below=abelow,
+                       atomic_floor=(args.hot_floor
+                                     if args.mode == "hybrid" else 0))
         torch.cuda.synchronize()
         st = gs.read_stats()
         syn0, _ = gs.to_host()
@@ -118,8 +122,8 @@ def main():
     dt = time.time() - t0
     # eval on mid-frequency concepts (Zipf rank = concept id)
     acc = nn_accuracy(syn0, range(args.eval_lo, args.eval_hi))
-    mode = (f"hybrid[K={args.hot_rows}]" if args.mode == "hybrid"
-            else args.mode)
+    mode = (f"hybrid[{args.hot_floor}..{args.hot_rows}]"
+            if args.mode == "hybrid" else args.mode)
     print(f"mode={mode} device={args.device} vocab={args.vocab} "
           f"words={args.words}: planted-NN acc={acc:.3f} "
           f"({args.words / dt / 1e6:.1f}M words/s incl. setup)")

@@ -589,6 +589,13 @@ struct KernelArgs {
   int64_t stride;             // row stride in elements (= 64*NC)
   int ref_window;             // 0 canonical, 1 reference (B2) semantics
   int32_t atomic_below;       // rows < this use atomics (hot rows); rest plain
+  // rows < atomic_floor take the plain path even in atomic modes: the
+  // ultra-hot head (top ~100 Zipf rows) is hit by a double-digit share of
+  // all negative draws, so atomics there serialize on a few cachelines —
+  // the hybrid cliff.  Those rows are subsample-suppressed as
+  // centers/contexts anyway; hogwild on them costs no measurable quality
+  // (benchmarks/results.md round-2 sweep).
+  int32_t atomic_floor;
   const float* exp_table;     // non-null: reference LUT sigmoid parity mode
   int exp_table_size;
   int width;                  // valid elements per row (masked dim phases)
@@ -747,6 +754,7 @@ struct TrainPhase {
   float alpha;
   int lane;
   int32_t atomic_below;
+  int32_t atomic_floor;
   const float* exp_table;
   int exp_table_size;
   // per-position state
@@ -780,17 +788,19 @@ struct TrainPhase {
     const float g = (label - sigma_of(f, exp_table, exp_table_size)) * alpha;
     // atomic_below < 0: atomics for POSITIVE pairs only (they drive the
     // learning signal; negatives self-limit) — ~1/6 of the update traffic
+    const int32_t rid = (int32_t)(enc & 0x7FFFFFFFu);
     const bool use_atomic =
         ATOMIC && (atomic_below < 0 ? (label > 0.5f)
-                                    : ((int32_t)(enc & 0x7FFFFFFFu) < atomic_below));
+                                    : (rid < atomic_below && rid >= atomic_floor));
     if (use_atomic) {
-      float delta[NC];
+      // t_row doubles as the delta buffer (it is dead after this pair) —
+      // no extra NC registers for the atomic path
 #pragma unroll
       for (int k = 0; k < NC; ++k) {
         grad[k] += g * t_row[k];
-        delta[k] = g * c_row[k];
+        t_row[k] = g * c_row[k];
       }
-      RowIO<T, NC>::atomic_add(t_ptr, delta, lane);
+      RowIO<T, NC>::atomic_add(t_ptr, t_row, lane);
     } else {
 #pragma unroll
       for (int k = 0; k < NC; ++k) {
@@ -838,7 +848,8 @@ struct TrainPhase {
 
   __device__ __forceinline__ void end_position(int32_t) {
     // center row update (hogwild: re-read current value, add, store)
-    if (ATOMIC && (atomic_below < 0 || c_idx < atomic_below)) {
+    if (ATOMIC && (atomic_below < 0 ||
+                   (c_idx < atomic_below && c_idx >= atomic_floor))) {
       RowIO<T, NC>::atomic_add(c_ptr, grad, lane);
     } else {
       float cur[NC];
@@ -868,6 +879,7 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void sgns_train_kernel(KernelA
   ph.alpha = a.alpha;
   ph.lane = lane;
   ph.atomic_below = a.atomic_below;
+  ph.atomic_floor = a.atomic_floor;
   ph.exp_table = a.exp_table;
   ph.exp_table_size = a.exp_table_size;
 
@@ -910,6 +922,7 @@ struct TrainPhase2 {
   int l32;      // lane & 31
   int half;     // lane >> 5
   int32_t atomic_below;
+  int32_t atomic_floor;
   const float* exp_table;
   int exp_table_size;
   T* c_ptr;
@@ -946,14 +959,14 @@ struct TrainPhase2 {
 #pragma unroll
       for (int m = 0; m < NCH; ++m) grad[m] += g * t_row[m];
       if (active) {
+        const int32_t rid = (int32_t)(enc & 0x7FFFFFFFu);
         const bool use_atomic =
             ATOMIC && (atomic_below < 0 ? (label > 0.5f)
-                                        : ((int32_t)(enc & 0x7FFFFFFFu) < atomic_below));
+                                        : (rid < atomic_below && rid >= atomic_floor));
         if (use_atomic) {
-          float delta[NCH];
 #pragma unroll
-          for (int m = 0; m < NCH; ++m) delta[m] = g * c_row[m];
-          RowIO32<T, NCH>::atomic_add(t_ptr, delta, l32);
+          for (int m = 0; m < NCH; ++m) t_row[m] = g * c_row[m];
+          RowIO32<T, NCH>::atomic_add(t_ptr, t_row, l32);
         } else {
 #pragma unroll
           for (int m = 0; m < NCH; ++m) t_row[m] += g * c_row[m];
@@ -972,7 +985,8 @@ struct TrainPhase2 {
     // combine the halves' grads (same elements live at lane l and l+32)
 #pragma unroll
     for (int m = 0; m < NCH; ++m) grad[m] += __shfl_xor(grad[m], 32, 64);
-    if (ATOMIC && (atomic_below < 0 || c_idx < atomic_below)) {
+    if (ATOMIC && (atomic_below < 0 ||
+                   (c_idx < atomic_below && c_idx >= atomic_floor))) {
       if (half == 0) RowIO32<T, NCH>::atomic_add(c_ptr, grad, l32);
     } else if (half == 0) {
       // one half does the RMW — the other would duplicate identical bytes
@@ -1005,6 +1019,7 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void sgns_train2_kernel(Kernel
   ph.l32 = lane & 31;
   ph.half = lane >> 5;
   ph.atomic_below = a.atomic_below;
+  ph.atomic_floor = a.atomic_floor;
   ph.exp_table = a.exp_table;
   ph.exp_table_size = a.exp_table_size;
 
@@ -1039,6 +1054,7 @@ struct TrainPhase4 {
   int l16;      // lane & 15
   int quarter;  // lane >> 4
   int32_t atomic_below;
+  int32_t atomic_floor;
   const float* exp_table;
   int exp_table_size;
   T* c_ptr;
@@ -1075,13 +1091,13 @@ struct TrainPhase4 {
 #pragma unroll
       for (int m = 0; m < NCQ; ++m) grad[m] += g * t_row[m];
       if (active) {
+        const int32_t rid = (int32_t)(enc & 0x7FFFFFFFu);
         const bool use_atomic =
-            ATOMIC && ((int32_t)(enc & 0x7FFFFFFFu) < atomic_below);
+            ATOMIC && rid < atomic_below && rid >= atomic_floor;
         if (use_atomic) {
-          float delta[NCQ];
 #pragma unroll
-          for (int m = 0; m < NCQ; ++m) delta[m] = g * c_row[m];
-          RowIO16<T, NCQ>::atomic_add(t_ptr, delta, l16);
+          for (int m = 0; m < NCQ; ++m) t_row[m] = g * c_row[m];
+          RowIO16<T, NCQ>::atomic_add(t_ptr, t_row, l16);
         } else {
 #pragma unroll
           for (int m = 0; m < NCQ; ++m) t_row[m] += g * c_row[m];
@@ -1102,7 +1118,7 @@ struct TrainPhase4 {
       grad[m] += __shfl_xor(grad[m], 16, 64);
       grad[m] += __shfl_xor(grad[m], 32, 64);
     }
-    if (ATOMIC && (c_idx < atomic_below)) {
+    if (ATOMIC && c_idx < atomic_below && c_idx >= atomic_floor) {
       if (quarter == 0) RowIO16<T, NCQ>::atomic_add(c_ptr, grad, l16);
     } else if (quarter == 0) {
       float cur[NCQ];
@@ -1133,6 +1149,7 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void sgns_train4_kernel(Kernel
   ph.l16 = lane & 15;
   ph.quarter = lane >> 4;
   ph.atomic_below = a.atomic_below;
+  ph.atomic_floor = a.atomic_floor;
   ph.exp_table = a.exp_table;
   ph.exp_table_size = a.exp_table_size;
 
@@ -1331,6 +1348,7 @@ struct UpdateSlicePhase {
   float alpha;
   int lane;
   int32_t atomic_below;
+  int32_t atomic_floor;
   const float* exp_table;
   int exp_table_size;
   T* c_ptr;
@@ -1363,17 +1381,19 @@ struct UpdateSlicePhase {
     const float g = (label - sigma_of(f, exp_table, exp_table_size)) * alpha;
     // atomic_below < 0: atomics for POSITIVE pairs only (they drive the
     // learning signal; negatives self-limit) — ~1/6 of the update traffic
+    const int32_t rid = (int32_t)(enc & 0x7FFFFFFFu);
     const bool use_atomic =
         ATOMIC && (atomic_below < 0 ? (label > 0.5f)
-                                    : ((int32_t)(enc & 0x7FFFFFFFu) < atomic_below));
+                                    : (rid < atomic_below && rid >= atomic_floor));
     if (use_atomic) {
-      float delta[NC];
+      // t_row doubles as the delta buffer (it is dead after this pair) —
+      // no extra NC registers for the atomic path
 #pragma unroll
       for (int k = 0; k < NC; ++k) {
         grad[k] += g * t_row[k];
-        delta[k] = g * c_row[k];
+        t_row[k] = g * c_row[k];
       }
-      RowIO<T, NC>::atomic_add(t_ptr, delta, lane);
+      RowIO<T, NC>::atomic_add(t_ptr, t_row, lane);
     } else {
 #pragma unroll
       for (int k = 0; k < NC; ++k) {
@@ -1416,7 +1436,7 @@ struct UpdateSlicePhase {
     }
   }
   __device__ __forceinline__ void end_position(int32_t) {
-    if (ATOMIC && (c_idx < atomic_below)) {
+    if (ATOMIC && c_idx < atomic_below && c_idx >= atomic_floor) {
       RowIO<T, NC>::atomic_add(c_ptr, grad, lane);
     } else {
       float cur[NC];
@@ -1449,6 +1469,7 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void update_slice_kernel(
   ph.lane = lane;
   ph.world_scale = world_scale;
   ph.atomic_below = a.atomic_below;
+  ph.atomic_floor = a.atomic_floor;
   ph.exp_table = a.exp_table;
   ph.exp_table_size = a.exp_table_size;
   for (int64_t s = wave_gid; s < a.num_sentences; s += total_waves) {
@@ -1543,6 +1564,7 @@ struct UpdateSlicePhase2 {
   int half;
   int width;
   int32_t atomic_below;
+  int32_t atomic_floor;
   const float* exp_table;
   int exp_table_size;
   T* c_ptr;
@@ -1581,13 +1603,13 @@ struct UpdateSlicePhase2 {
 #pragma unroll
       for (int m = 0; m < NCH; ++m) grad[m] += g * t_row[m];
       if (active) {
+        const int32_t rid = (int32_t)(enc & 0x7FFFFFFFu);
         const bool use_atomic =
-            ATOMIC && ((int32_t)(enc & 0x7FFFFFFFu) < atomic_below);
+            ATOMIC && rid < atomic_below && rid >= atomic_floor;
         if (use_atomic) {
-          float delta[NCH];
 #pragma unroll
-          for (int m = 0; m < NCH; ++m) delta[m] = g * c_row[m];
-          RIO32<T, NCH, MASKED>::atomic_add(t_ptr, delta, l32, width);
+          for (int m = 0; m < NCH; ++m) t_row[m] = g * c_row[m];
+          RIO32<T, NCH, MASKED>::atomic_add(t_ptr, t_row, l32, width);
         } else {
 #pragma unroll
           for (int m = 0; m < NCH; ++m) t_row[m] += g * c_row[m];
@@ -1604,7 +1626,7 @@ struct UpdateSlicePhase2 {
   __device__ __forceinline__ void end_position(int32_t) {
 #pragma unroll
     for (int m = 0; m < NCH; ++m) grad[m] += __shfl_xor(grad[m], 32, 64);
-    if (ATOMIC && (c_idx < atomic_below)) {
+    if (ATOMIC && c_idx < atomic_below && c_idx >= atomic_floor) {
       if (half == 0)
         RIO32<T, NCH, MASKED>::atomic_add(c_ptr, grad, l32, width);
     } else if (half == 0) {
@@ -1640,6 +1662,7 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void update_slice2_kernel(
   ph.width = a.width;
   ph.world_scale = world_scale;
   ph.atomic_below = a.atomic_below;
+  ph.atomic_floor = a.atomic_floor;
   ph.exp_table = a.exp_table;
   ph.exp_table_size = a.exp_table_size;
   for (int64_t s = wave_gid; s < a.num_sentences; s += total_waves) {
@@ -1740,6 +1763,7 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void train_pairs2_kernel(
     const int64_t* __restrict__ group_offsets, int64_t num_groups,
     const int32_t* __restrict__ pair_target,
     const float* __restrict__ pair_label, float alpha, int32_t atomic_below,
+    int32_t atomic_floor,
     unsigned long long* d_pairs, unsigned long long* d_positives,
     unsigned long long* d_words, double* d_sum_fplus) {
   const int lane = threadIdx.x & 63;
@@ -1777,11 +1801,10 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void train_pairs2_kernel(
 #pragma unroll
       for (int k = 0; k < NCH; ++k) grad[k] += gg * t_row[k];
       if (active) {
-        if (ATOMIC && tid < atomic_below) {
-          float delta[NCH];
+        if (ATOMIC && tid < atomic_below && tid >= atomic_floor) {
 #pragma unroll
-          for (int k = 0; k < NCH; ++k) delta[k] = gg * c_row[k];
-          RowIO32<T, NCH>::atomic_add(t_ptr, delta, l32);
+          for (int k = 0; k < NCH; ++k) t_row[k] = gg * c_row[k];
+          RowIO32<T, NCH>::atomic_add(t_ptr, t_row, l32);
         } else {
 #pragma unroll
           for (int k = 0; k < NCH; ++k) t_row[k] += gg * c_row[k];
@@ -1796,7 +1819,8 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void train_pairs2_kernel(
     }
 #pragma unroll
     for (int k = 0; k < NCH; ++k) grad[k] += __shfl_xor(grad[k], 32, 64);
-    if (ATOMIC && group_center[g] < atomic_below) {
+    if (ATOMIC && group_center[g] < atomic_below &&
+        group_center[g] >= atomic_floor) {
       if (half == 0) RowIO32<T, NCH>::atomic_add(c_ptr, grad, l32);
     } else if (half == 0) {
       float cur[NCH];
@@ -2056,7 +2080,7 @@ static void sgns_train(uintptr_t syn0, uintptr_t syn1, int is_bf16,
                        uintptr_t stats, int blocks, int pos_blocks,
                        int threads, uintptr_t stream_ptr,
                        uintptr_t exp_table, int exp_table_size,
-                       int pair2) {
+                       int pair2, int64_t atomic_floor) {
   HIP_CLEAR_ERROR();
   if (threads != 64 && threads != 256)
     throw std::runtime_error("threads must be 64 (serial) or 256");
@@ -2082,6 +2106,7 @@ static void sgns_train(uintptr_t syn0, uintptr_t syn1, int is_bf16,
   a.stride = stride;
   a.ref_window = ref_window;
   a.atomic_below = (int32_t)std::min<int64_t>(atomic_below, 0x7FFFFFFFLL);
+  a.atomic_floor = (int32_t)std::min<int64_t>(atomic_floor, 0x7FFFFFFFLL);
   a.exp_table = (const float*)exp_table;
   a.exp_table_size = exp_table_size;
   unsigned long long* st = (unsigned long long*)stats;
@@ -2253,7 +2278,7 @@ static void update_slice(uintptr_t syn0, uintptr_t syn1, int is_bf16,
                          uintptr_t stats, int blocks, int threads,
                          uintptr_t stream_ptr,
                          uintptr_t exp_table, int exp_table_size,
-                         int pair_mode, int width) {
+                         int pair_mode, int width, int64_t atomic_floor) {
   HIP_CLEAR_ERROR();
   if (width <= 0) width = (int)stride;
   KernelArgs a = make_walk_args(syn0, syn1, stride, tokens, offsets,
@@ -2262,6 +2287,7 @@ static void update_slice(uintptr_t syn0, uintptr_t syn1, int is_bf16,
                                 ref_window, stats);
   a.width = width;
   a.atomic_below = (int32_t)std::min<int64_t>(atomic_below, 0x7FFFFFFFLL);
+  a.atomic_floor = (int32_t)std::min<int64_t>(atomic_floor, 0x7FFFFFFFLL);
   a.exp_table = (const float*)exp_table;
   a.exp_table_size = exp_table_size;
   const int atomic = atomic_below > 0;
@@ -2326,7 +2352,7 @@ static void train_pairs(uintptr_t cache0, uintptr_t cache1, int64_t stride,
                         uintptr_t pair_label, double alpha, uintptr_t stats,
                         int blocks, int threads, uintptr_t stream_ptr,
                         int pair_mode, int atomic, int is_bf16,
-                        int64_t atomic_below) {
+                        int64_t atomic_below, int64_t atomic_floor) {
   HIP_CLEAR_ERROR();
   if (stride % 64 != 0) throw std::runtime_error("stride must be a multiple of 64");
   if (is_bf16 && pair_mode == 0)
@@ -2334,6 +2360,8 @@ static void train_pairs(uintptr_t cache0, uintptr_t cache1, int64_t stride,
   const int nc = (int)(stride / 64);
   const int32_t abelow =
       (int32_t)std::min<int64_t>(atomic_below, 0x7FFFFFFFLL);
+  const int32_t afloor =
+      (int32_t)std::min<int64_t>(atomic_floor, 0x7FFFFFFFLL);
   hipStream_t stream = (hipStream_t)stream_ptr;
   unsigned long long* st = (unsigned long long*)stats;
   switch (nc) {
@@ -2358,7 +2386,7 @@ static void train_pairs(uintptr_t cache0, uintptr_t cache1, int64_t stride,
                            (const int32_t*)group_center,                      \
                            (const int64_t*)group_offsets, num_groups,         \
                            (const int32_t*)pair_target,                       \
-                           (const float*)pair_label, (float)alpha, abelow,    \
+                           (const float*)pair_label, (float)alpha, abelow, afloor,    \
                            st ? st + 0 : nullptr, st ? st + 1 : nullptr,      \
                            st ? st + 2 : nullptr,                             \
                            st ? (double*)(st + 3) : nullptr);                 \
@@ -2369,7 +2397,7 @@ static void train_pairs(uintptr_t cache0, uintptr_t cache1, int64_t stride,
                            (const int32_t*)group_center,                      \
                            (const int64_t*)group_offsets, num_groups,         \
                            (const int32_t*)pair_target,                       \
-                           (const float*)pair_label, (float)alpha, abelow,    \
+                           (const float*)pair_label, (float)alpha, abelow, afloor,    \
                            st ? st + 0 : nullptr, st ? st + 1 : nullptr,      \
                            st ? st + 2 : nullptr,                             \
                            st ? (double*)(st + 3) : nullptr);                 \
@@ -2381,7 +2409,7 @@ static void train_pairs(uintptr_t cache0, uintptr_t cache1, int64_t stride,
                          (const int32_t*)group_center,                        \
                          (const int64_t*)group_offsets, num_groups,           \
                          (const int32_t*)pair_target,                         \
-                         (const float*)pair_label, (float)alpha, abelow,      \
+                         (const float*)pair_label, (float)alpha, abelow, afloor,      \
                          st ? st + 0 : nullptr, st ? st + 1 : nullptr,        \
                          st ? st + 2 : nullptr,                               \
                          st ? (double*)(st + 3) : nullptr);                   \
@@ -2393,7 +2421,7 @@ static void train_pairs(uintptr_t cache0, uintptr_t cache1, int64_t stride,
                          (const int32_t*)group_center,                        \
                          (const int64_t*)group_offsets, num_groups,           \
                          (const int32_t*)pair_target,                         \
-                         (const float*)pair_label, (float)alpha, abelow,      \
+                         (const float*)pair_label, (float)alpha, abelow, afloor,      \
                          st ? st + 0 : nullptr, st ? st + 1 : nullptr,        \
                          st ? st + 2 : nullptr,                               \
                          st ? (double*)(st + 3) : nullptr);                   \
@@ -2573,7 +2601,7 @@ PYBIND11_MODULE(_hip_native, m) {
         py::arg("stats"), py::arg("blocks"), py::arg("pos_blocks"),
         py::arg("threads"), py::arg("stream"),
         py::arg("exp_table") = 0, py::arg("exp_table_size") = 0,
-        py::arg("pair2") = 0);
+        py::arg("pair2") = 0, py::arg("atomic_floor") = 0);
   m.def("count_pairs", &count_pairs);
   m.def("dots_slice", &dots_slice);
   m.def("update_slice", &update_slice);

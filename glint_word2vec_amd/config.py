@@ -116,7 +116,13 @@ class Word2VecConfig:
     # The benchmark and fit() share this default (same semantics measured
     # as shipped).  See DESIGN.md.
     update_mode: str = "hybrid"      # "hogwild" | "atomic" | "hybrid"
-    hybrid_hot_rows: int = 1 << 18   # atomic rows for update_mode="hybrid"
+    hybrid_hot_rows: int = 8192      # hybrid: atomics for rows < this
+    # hybrid: rows < this stay hogwild even inside the atomic head — the
+    # ultra-hot top rows take a double-digit share of all negative-table
+    # draws, and atomics there serialize on a handful of cachelines (the
+    # measured hybrid cliff); they are subsample-suppressed as centers
+    # anyway, so the quality cost is nil (benchmarks/results.md).
+    hybrid_skip_rows: int = 128
     # Deprecated alias (round-1 API): True -> "atomic", False -> "hogwild".
     # None (default) leaves update_mode in charge.
     atomic_updates: Optional[bool] = None
@@ -170,6 +176,13 @@ class Word2VecConfig:
             return 2 ** 31 - 1
         return int(self.hybrid_hot_rows)
 
+    def effective_atomic_floor(self) -> int:
+        # Rows below this stay hogwild even in atomic ranges (hybrid
+        # ultra-head contention escape; 0 elsewhere).
+        if self.resolved_update_mode() == "hybrid":
+            return int(self.hybrid_skip_rows)
+        return 0
+
     def validate(self) -> None:
         if self.vector_size <= 0:
             raise ValueError("vector_size must be > 0")
@@ -193,6 +206,8 @@ class Word2VecConfig:
             raise ValueError(f"unsupported update_mode {self.update_mode!r}")
         if self.hybrid_hot_rows < 0:
             raise ValueError("hybrid_hot_rows must be >= 0")
+        if self.hybrid_skip_rows < 0:
+            raise ValueError("hybrid_skip_rows must be >= 0")
         if self.window_mode not in ("canonical", "reference"):
             raise ValueError(f"unsupported window_mode {self.window_mode!r}")
         if self.engine not in ("auto", "fused", "dim", "row", "dp"):

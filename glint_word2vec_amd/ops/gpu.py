@@ -123,6 +123,7 @@ class GpuSgns:
                     alpha: float, window: int, n_neg: int, seed: int,
                     sent_id_base: int = 0, window_mode: str = "canonical",
                     atomic: bool = True, atomic_below: Optional[int] = None,
+                    atomic_floor: int = 0,
                     blocks: Optional[int] = None,
                     serial: bool = False, pair_mode: int = 1,
                     stream: Optional[torch.cuda.Stream] = None) -> None:
@@ -164,7 +165,7 @@ class GpuSgns:
             s.cuda_stream,
             0 if self.exp_table is None else self.exp_table.data_ptr(),
             0 if self.exp_table is None else int(self.exp_table.numel()),
-            0 if serial else int(pair_mode))
+            0 if serial else int(pair_mode), int(atomic_floor))
 
     def read_stats(self, reset: bool = True) -> GpuStats:
         h = self._stats.cpu()

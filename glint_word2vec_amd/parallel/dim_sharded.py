@@ -43,7 +43,7 @@ class DimShardedSgns:
                  table_size: int = 1_000_000, subsample: float = 0.0,
                  window_mode: str = "canonical", chunk_words: int = 1 << 19,
                  f_correction: bool = True, atomic: bool = True,
-                 atomic_below: "int | None" = None,
+                 atomic_below: "int | None" = None, atomic_floor: int = 0,
                  narrow: "bool | None" = None,
                  init_full_limit: int = 1 << 28):
         self.rank, self.world = (comm.init_from_env() if torch.distributed.is_available()
@@ -61,6 +61,7 @@ class DimShardedSgns:
         # all-or-nothing from the atomic flag
         self.atomic_below = ((2 ** 31 - 1 if atomic else 0)
                              if atomic_below is None else int(atomic_below))
+        self.atomic_floor = int(atomic_floor)
         self.lo, self.hi = slice_bounds(dim, self.rank, self.world)
         self.width = self.hi - self.lo
 
@@ -217,7 +218,7 @@ class DimShardedSgns:
                 self.atomic_below, self._stats.data_ptr(),
                 nb, nt, comp.cuda_stream, 0, 0,
                 0 if self.serial else 1,
-                self.width if self.narrow else 0)
+                self.width if self.narrow else 0, self.atomic_floor)
 
         for (a, b) in chunks:
             n = b - a

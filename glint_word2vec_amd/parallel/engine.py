@@ -101,7 +101,8 @@ def train_gpu(cfg: Word2VecConfig, vocab: Vocabulary, batches_fn: Callable,
             gs.train_batch(tok, off, alpha, cfg.window, cfg.n, seed,
                            sent_id_base=sent_base, window_mode=cfg.window_mode,
                            atomic=cfg.resolved_update_mode() != "hogwild",
-                           atomic_below=cfg.effective_atomic_below())
+                           atomic_below=cfg.effective_atomic_below(),
+                           atomic_floor=cfg.effective_atomic_floor())
             sent_base += batch.num_sentences
             processed += batch.num_tokens
             step += 1
@@ -156,6 +157,7 @@ def _train_sharded(cfg: Word2VecConfig, vocab: Vocabulary,
                              f_correction=cfg.f_correction,
                              atomic=cfg.resolved_update_mode() != "hogwild",
                              atomic_below=cfg.effective_atomic_below(),
+                             atomic_floor=cfg.effective_atomic_floor(),
                              **common)
         if init_tables is not None:
             eng.load_host(*init_tables)
@@ -180,6 +182,7 @@ def _train_sharded(cfg: Word2VecConfig, vocab: Vocabulary,
                              sync_every=cfg.sync_every,
                              atomic=cfg.resolved_update_mode() != "hogwild",
                              atomic_below=cfg.effective_atomic_below(),
+                             atomic_floor=cfg.effective_atomic_floor(),
                              **common)
         if init_tables is not None:
             eng.load_host(*init_tables)
@@ -221,6 +224,7 @@ def _train_sharded(cfg: Word2VecConfig, vocab: Vocabulary,
         eng = RowShardedSgns(vocab.num_words, cfg.vector_size,
                              atomic=cfg.resolved_update_mode() == "atomic",
                              atomic_below=cfg.effective_atomic_below(),
+                             atomic_floor=cfg.effective_atomic_floor(),
                              **common)
         if init_tables is not None:
             eng.load_host(*init_tables)

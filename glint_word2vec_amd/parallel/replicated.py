@@ -34,7 +34,8 @@ class ReplicatedSgns:
                  counts: Optional[np.ndarray] = None,
                  table_size: int = 1_000_000, subsample: float = 0.0,
                  window_mode: str = "canonical", sync_every: int = 4,
-                 atomic: bool = False, atomic_below: "int | None" = None):
+                 atomic: bool = False, atomic_below: "int | None" = None,
+                 atomic_floor: int = 0):
         self.rank, self.world = comm.init_from_env()
         self.vocab_size = vocab_size
         self.dim = dim
@@ -45,6 +46,7 @@ class ReplicatedSgns:
         self.atomic = atomic
         self.atomic_below = ((2 ** 31 - 1 if atomic else 0)
                              if atomic_below is None else int(atomic_below))
+        self.atomic_floor = int(atomic_floor)
         self._steps_since_sync = 0
         counts = (np.ones(vocab_size, dtype=np.int64) if counts is None
                   else counts)
@@ -86,7 +88,8 @@ class ReplicatedSgns:
                                 atomic=self.atomic,
                                 atomic_below=(None if self.atomic_below
                                               >= 2 ** 31 - 1
-                                              else self.atomic_below))
+                                              else self.atomic_below),
+                                atomic_floor=self.atomic_floor)
         else:
             st = self.native.train_batch(
                 self.syn0.numpy(), self.syn1.numpy(),
