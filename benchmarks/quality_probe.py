@@ -71,7 +71,7 @@ def main():
                     default="hogwild")
     ap.add_argument("--hot-floor", type=int, default=0,
                     help="hybrid: rows < F stay hogwild (contention escape)")
-    ap.add_argument("--hot-rows", type=int, default=8192,
+    ap.add_argument("--hot-rows", type=int, default=32768,
                     help="hybrid: atomics for rows < K (word id ~ 2*Zipf "
                          "rank in this corpus, so K covers the K/2 hottest "
                          "concepts)")

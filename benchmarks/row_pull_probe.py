@@ -56,25 +56,35 @@ def main():
     acc = {s: 0.0 for s in stages}
 
     def run_step(i, record):
-        evs = [torch.cuda.Event(enable_timing=True) for _ in range(6)]
+        # replicate the world-1 pull_begin internals so each sub-stage gets
+        # its own event pair (keep in sync with row_sharded.pull_begin)
+        evs = [torch.cuda.Event(enable_timing=True) for _ in range(8)]
         evs[0].record()
         plan = eng.make_plan_counter(tok, off, args.window, args.neg, 99,
                                      sent_id_base=i * nsent)
         evs[1].record()
-        # pull_begin internals split for timing: do unique+route inline
-        st = eng.pull_begin(plan)
+        uc, inv_c = torch.unique(plan.group_center, return_inverse=True)
+        ut, inv_t = torch.unique(plan.pair_target, return_inverse=True)
         evs[2].record()
+        cache0 = eng._ws("c0", uc.numel(), eng.store_stride, eng.syn0.dtype)
+        cache1 = eng._ws("c1", ut.numel(), eng.store_stride, eng.syn0.dtype)
+        eng._gather_native(eng.syn0, uc.int().contiguous(), cache0)
+        eng._gather_native(eng.syn1, ut.int().contiguous(), cache1)
         evs[3].record()
-        eng.train_push(st, 0.01875)
+        st = {"plan": plan, "gc": inv_c.int().contiguous(),
+              "gt": inv_t.int().contiguous(), "cache0": cache0,
+              "cache1": cache1, "orig0": None, "orig1": None,
+              "uc_sorted": uc, "ut_sorted": ut}
         evs[4].record()
+        eng.train_push(st, 0.01875)
         evs[5].record()
         torch.cuda.synchronize(dev)
         if record:
             acc["plan"] += evs[0].elapsed_time(evs[1])
-            acc["unique_route"] += 0.0
-            acc["pull"] += evs[1].elapsed_time(evs[2])
-            acc["train"] += evs[3].elapsed_time(evs[4])
-            acc["push"] += 0.0
+            acc["unique_route"] += evs[1].elapsed_time(evs[2])
+            acc["pull"] += evs[2].elapsed_time(evs[3])
+            acc["train"] += evs[4].elapsed_time(evs[5])
+            acc["push"] += evs[3].elapsed_time(evs[4])
         return plan
 
     plan = run_step(0, False)  # warmup

@@ -641,7 +641,123 @@ static py::tuple encode_corpus(const std::string& path, py::list words,
   return py::make_tuple(tok, off);
 }
 
+// ---------------------------------------------------------------------------
+// WordFileIndex: word <-> row-index lookups straight off the checkpoint's
+// `words` file (one word per line, line number == row index — the
+// reference's layout, mllib:714-715) WITHOUT materialising 10s of millions
+// of Python strings.  The reference's client holds the whole word->index
+// map in an 8 GB Spark broadcast and documents a <80M-word ceiling
+// (README.md:71-73); here the file is mmap'd and indexed with an
+// open-addressing FNV-1a table: ~24 bytes/word of host memory, so an
+// 80M-word vocabulary costs ~2 GB and loads in seconds.  Used by the
+// sharded serving model (glint_word2vec_amd/serving.py).
+// ---------------------------------------------------------------------------
+#include <fcntl.h>
+#include <sys/mman.h>
+#include <sys/stat.h>
+#include <unistd.h>
+
+class WordFileIndex {
+ public:
+  explicit WordFileIndex(const std::string& path) {
+    fd_ = ::open(path.c_str(), O_RDONLY);
+    if (fd_ < 0) throw std::runtime_error("cannot open words file: " + path);
+    struct stat st;
+    if (fstat(fd_, &st) != 0) {
+      ::close(fd_);
+      throw std::runtime_error("cannot stat words file: " + path);
+    }
+    size_ = st.st_size;
+    data_ = size_ ? (const char*)mmap(nullptr, size_, PROT_READ, MAP_PRIVATE,
+                                      fd_, 0)
+                  : nullptr;
+    if (size_ && data_ == MAP_FAILED) {
+      ::close(fd_);
+      throw std::runtime_error("mmap failed: " + path);
+    }
+    py::gil_scoped_release nogil;
+    offsets_.push_back(0);
+    const char* p = data_;
+    const char* end = data_ + size_;
+    while (p < end) {
+      const char* nl = (const char*)memchr(p, '\n', end - p);
+      if (!nl) break;
+      offsets_.push_back((int64_t)(nl + 1 - data_));
+      p = nl + 1;
+    }
+    if (offsets_.back() < size_) offsets_.push_back(size_);  // no final \n
+    n_ = (int64_t)offsets_.size() - 1;
+    int64_t cap = 16;
+    while (cap < 2 * n_) cap <<= 1;
+    mask_ = (uint64_t)cap - 1;
+    table_.assign((size_t)cap, 0);
+    for (int64_t i = 0; i < n_; ++i) {
+      uint64_t h = hash(ptr(i), len(i)) & mask_;
+      while (table_[h]) h = (h + 1) & mask_;
+      table_[h] = (uint64_t)(i + 1);
+    }
+  }
+  ~WordFileIndex() {
+    if (data_ && data_ != MAP_FAILED) munmap((void*)data_, size_);
+    if (fd_ >= 0) ::close(fd_);
+  }
+  WordFileIndex(const WordFileIndex&) = delete;
+
+  int64_t lookup(const std::string& w) const {
+    uint64_t h = hash(w.data(), (int64_t)w.size()) & mask_;
+    while (table_[h]) {
+      const int64_t i = (int64_t)table_[h] - 1;
+      if (len(i) == (int64_t)w.size() &&
+          memcmp(ptr(i), w.data(), w.size()) == 0)
+        return i;
+      h = (h + 1) & mask_;
+    }
+    return -1;
+  }
+  std::string word(int64_t i) const {
+    if (i < 0 || i >= n_) throw std::out_of_range("word index");
+    return std::string(ptr(i), (size_t)len(i));
+  }
+  py::array_t<int64_t> lookup_many(py::list words) const {
+    auto out = py::array_t<int64_t>((py::ssize_t)words.size());
+    int64_t* o = out.mutable_data();
+    for (size_t i = 0; i < words.size(); ++i)
+      o[i] = lookup(py::cast<std::string>(words[i]));
+    return out;
+  }
+  int64_t size() const { return n_; }
+
+ private:
+  const char* ptr(int64_t i) const { return data_ + offsets_[i]; }
+  int64_t len(int64_t i) const {
+    int64_t l = offsets_[i + 1] - offsets_[i];
+    if (l > 0 && data_[offsets_[i + 1] - 1] == '\n') --l;
+    return l;
+  }
+  static uint64_t hash(const char* s, int64_t n) {
+    uint64_t h = 1469598103934665603ULL;  // FNV-1a 64
+    for (int64_t i = 0; i < n; ++i) {
+      h ^= (unsigned char)s[i];
+      h *= 1099511628211ULL;
+    }
+    return h;
+  }
+  int fd_ = -1;
+  int64_t size_ = 0;
+  const char* data_ = nullptr;
+  std::vector<int64_t> offsets_;
+  int64_t n_ = 0;
+  uint64_t mask_ = 0;
+  std::vector<uint64_t> table_;
+};
+
 PYBIND11_MODULE(_cpu_native, m) {
+  py::class_<WordFileIndex>(m, "WordFileIndex")
+      .def(py::init<const std::string&>(), py::arg("path"))
+      .def("lookup", &WordFileIndex::lookup)
+      .def("word", &WordFileIndex::word)
+      .def("lookup_many", &WordFileIndex::lookup_many)
+      .def("__len__", &WordFileIndex::size);
   m.def("build_vocab_file", &build_vocab_file, py::arg("path"),
         py::arg("min_count") = 5);
   m.def("encode_corpus", &encode_corpus, py::arg("path"), py::arg("words"),

@@ -116,7 +116,7 @@ class Word2VecConfig:
     # The benchmark and fit() share this default (same semantics measured
     # as shipped).  See DESIGN.md.
     update_mode: str = "hybrid"      # "hogwild" | "atomic" | "hybrid"
-    hybrid_hot_rows: int = 8192      # hybrid: atomics for rows < this
+    hybrid_hot_rows: int = 32768     # hybrid: atomics for rows < this
     # hybrid: rows < this stay hogwild even inside the atomic head — the
     # ultra-hot top rows take a double-digit share of all negative-table
     # draws, and atomics there serialize on a handful of cachelines (the

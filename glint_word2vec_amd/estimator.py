@@ -444,6 +444,18 @@ class GlintWord2VecModel:
         config, vocab, syn0, syn1 = load_model(path)
         return cls(config=config, vocab=vocab, syn0=syn0, syn1=syn1)
 
+    @classmethod
+    def load_sharded(cls, path: str, device: str = "auto",
+                     dtype: str = "auto"):
+        """Load WITHOUT host materialisation: rows stream from the shard
+        files straight to each rank's device and serving ops (findSynonyms
+        / transform / getVector) run sharded — the reference's PS-side
+        load+serve (mllib:683-726, :598).  Works at world 1 (one GPU) or
+        under torchrun (one rank per GPU); returns a
+        serving.ShardedWord2VecModel."""
+        from .serving import ShardedWord2VecModel
+        return ShardedWord2VecModel.load(path, device=device, dtype=dtype)
+
     def stop(self, terminate_other_clients: bool = False) -> None:
         """Release training resources (reference: stops PS cluster,
         mllib:664-667).  Tears down the torch.distributed group if this

@@ -708,27 +708,42 @@ class RowShardedSgns:
                         dtype=np.float32).tobytes())
         comm.barrier()
 
-    def load_checkpoint(self, path: str) -> None:
-        """Each rank reads its own shard (requires num_shards == world and
-        row_mod layout — i.e. a checkpoint written by this engine at the
-        same world size, or by save_model(num_shards=world))."""
+    def load_checkpoint(self, path: str, block_rows: int = 1 << 20) -> None:
+        """Streaming resume from ANY checkpoint layout/shard count: each
+        rank scans the shard files block-wise and keeps the rows it owns
+        (r % world == rank).  Host memory stays O(block_rows*dim); a
+        same-world row_mod checkpoint reduces to a straight shard read."""
         import json
         import os
+        from ..serving import _shard_row_iter
         with open(os.path.join(path, "shards", "index.json")) as f:
             index = json.load(f)
-        if (index["num_shards"] != self.world or
-                index.get("layout", "row_mod") != "row_mod"):
+        if index["vocab"] != self.vocab_size or index["dim"] != self.dim:
             raise ValueError(
-                f"checkpoint has {index['num_shards']} {index.get('layout')} "
-                f"shards; this engine needs {self.world} row_mod shards — "
-                "load via GlintWord2VecModel.load instead")
+                f"checkpoint is vocab={index['vocab']} dim={index['dim']}; "
+                f"engine is vocab={self.vocab_size} dim={self.dim}")
+        if not index.get("has_syn1", False):
+            raise ValueError("checkpoint has no syn1 — cannot resume")
         dt = np.dtype(index["dtype"])
+        D = self.dim
         for t, name in ((self.syn0, "syn0"), (self.syn1, "syn1")):
-            buf = np.fromfile(os.path.join(path, "shards",
-                                           f"{name}-{self.rank:05d}.bin"),
-                              dtype=dt).reshape(self.shard_size, self.dim)
-            t[:, :self.dim] = \
-                torch.from_numpy(buf).to(t.dtype).to(self.device)
+            for s in range(index["num_shards"]):
+                first, count, stride = _shard_row_iter(index, s)
+                fp = os.path.join(path, "shards", f"{name}-{s:05d}.bin")
+                with open(fp, "rb") as f:
+                    for j0 in range(0, count, block_rows):
+                        j1 = min(count, j0 + block_rows)
+                        block = np.fromfile(f, dtype=dt, count=(j1 - j0) * D) \
+                            .reshape(j1 - j0, D)
+                        gids = first + np.arange(j0, j1,
+                                                 dtype=np.int64) * stride
+                        mine = (gids % self.world) == self.rank
+                        if not mine.any():
+                            continue
+                        rows = torch.from_numpy(np.ascontiguousarray(
+                            block[mine], dtype=np.float32))
+                        t[torch.from_numpy(gids[mine] // self.world),
+                          :D] = rows.to(t.dtype).to(self.device)
 
     def to_host(self) -> Tuple[np.ndarray, np.ndarray]:
         out = []

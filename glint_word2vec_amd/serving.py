@@ -1,0 +1,318 @@
+"""Sharded model serving — the reference's distributed-model load+serve
+path (mllib:683-726: ``load`` restarts the PS cluster and loads shards
+SERVER-side, keeping the model sharded; findSynonyms/transform then run
+against the sharded state, mllib:598/486, ml:453).
+
+Here the GPUs are the servers: ``ShardedWord2VecModel.load`` streams each
+rank's rows (row r lives on rank r % world) straight from the checkpoint
+shard files into HBM — host memory stays O(block), never O(vocab*dim) —
+and the model ops run as per-shard kernels + small collectives:
+
+  findSynonyms  = per-shard GEMM scores (rocBLAS) + local top-k
+                  -> allgather of k candidates -> merge     (mllib:598)
+  norms         = per-shard row-norm kernel, cached         (mllib:486)
+  transform     = per-shard partial sentence sums -> allreduce (ml:453)
+
+Word <-> index lookups come from the mmap'd C++ ``WordFileIndex`` (one
+word per line, line number == row index — the reference's words-file
+layout, mllib:714-715), ~24 bytes/word instead of the reference's 8 GB
+driver-side broadcast map (README.md:71-73).
+
+Runs at any world size (world 1 = single-GPU serving) and on CPU under
+gloo for tests.  Every rank calls every op collectively and every rank
+returns the full result (the natural shape for a replicated-frontend
+serving tier).
+"""
+from __future__ import annotations
+
+import json
+import logging
+import os
+from typing import Iterable, List, Optional, Sequence, Union
+
+import numpy as np
+import torch
+
+from . import _cpu_native
+from .config import Word2VecConfig
+from .parallel import comm
+
+log = logging.getLogger("glint_word2vec_amd")
+
+
+class LazyVocab:
+    """Vocabulary view over the checkpoint's words file (mmap'd C++ index):
+    word->row and row->word without materialising Python strings."""
+
+    def __init__(self, words_path: str):
+        self._ix = _cpu_native.WordFileIndex(words_path)
+
+    @property
+    def num_words(self) -> int:
+        return len(self._ix)
+
+    def __contains__(self, word: str) -> bool:
+        return self._ix.lookup(word) >= 0
+
+    def __getitem__(self, word: str) -> int:
+        i = self._ix.lookup(word)
+        if i < 0:
+            raise KeyError(word)
+        return i
+
+    def get(self, word: str, default: int = -1) -> int:
+        i = self._ix.lookup(word)
+        return i if i >= 0 else default
+
+    def word(self, i: int) -> str:
+        return self._ix.word(i)
+
+    def lookup_many(self, words: Sequence[str]) -> np.ndarray:
+        return self._ix.lookup_many(list(words))
+
+
+def _shard_row_iter(index: dict, shard: int):
+    """Yield (global_row_of_first, rows_in_shard) addressing for a shard
+    file: row j of shard s is global ``s + j*k`` (row_mod) or
+    ``bounds[s] + j`` (row_range)."""
+    k = index["num_shards"]
+    if index.get("layout", "row_mod") == "row_mod":
+        first = shard
+        count = (index["vocab"] - shard + k - 1) // k
+        stride = k
+    else:
+        b = index["bounds"]
+        first = b[shard]
+        count = b[shard + 1] - b[shard]
+        stride = 1
+    return first, count, stride
+
+
+class ShardedWord2VecModel:
+    """Distributed (or single-GPU) serving model over a checkpoint.
+
+    ``load`` is collective: every rank streams its own rows of syn0 from
+    whatever shard layout/count the checkpoint was written with.  syn1 is
+    not loaded (serving needs only the input embeddings, like the
+    reference's model ops)."""
+
+    def __init__(self, config: Word2VecConfig, vocab, shard: torch.Tensor,
+                 rank: int, world: int, dim: int):
+        self.config = config
+        self.vocab = vocab
+        self.shard = shard            # [shard_size, dim] on the device
+        self.rank = rank
+        self.world = world
+        self.dim = dim
+        self.device = shard.device
+        self.is_cuda = shard.device.type == "cuda"
+        self._norms: Optional[torch.Tensor] = None   # local shard norms
+
+    # ------------------------------------------------------------------
+    @classmethod
+    def load(cls, path: str, device: str = "auto", dtype: str = "auto",
+             block_rows: int = 1 << 20) -> "ShardedWord2VecModel":
+        """Streaming sharded load (the PS-side load, mllib:717-722).
+        Host RSS stays O(block_rows*dim) + the word index (~24 B/word)."""
+        rank, world = comm.init_from_env()
+        if device == "auto":
+            device = "cuda" if torch.cuda.is_available() else "cpu"
+        if device.startswith("cuda") and torch.cuda.is_available():
+            device = f"cuda:{torch.cuda.current_device()}"
+        with open(os.path.join(path, "metadata")) as f:
+            meta = json.load(f)
+        config = Word2VecConfig.from_dict(meta["paramMap"])
+        with open(os.path.join(path, "shards", "index.json")) as f:
+            index = json.load(f)
+        V, D = index["vocab"], index["dim"]
+        tdtype = (torch.bfloat16 if (dtype == "bfloat16" or (
+            dtype == "auto" and device.startswith("cuda")))
+            else torch.float32)
+        file_dt = np.dtype(index["dtype"])
+        shard_size = (V - rank + world - 1) // world
+        shard = torch.empty((shard_size, D), dtype=tdtype,
+                            device=torch.device(device))
+        for s in range(index["num_shards"]):
+            first, count, stride = _shard_row_iter(index, s)
+            fp = os.path.join(path, "shards", f"syn0-{s:05d}.bin")
+            with open(fp, "rb") as f:
+                for j0 in range(0, count, block_rows):
+                    j1 = min(count, j0 + block_rows)
+                    buf = np.fromfile(f, dtype=file_dt,
+                                      count=(j1 - j0) * D)
+                    block = buf.reshape(j1 - j0, D)
+                    # global ids of this block, then the ones this rank owns
+                    gids = first + np.arange(j0, j1, dtype=np.int64) * stride
+                    mine = (gids % world) == rank
+                    if not mine.any():
+                        continue
+                    rows = torch.from_numpy(
+                        np.ascontiguousarray(block[mine], dtype=np.float32))
+                    shard[torch.from_numpy(gids[mine] // world)] = \
+                        rows.to(tdtype).to(shard.device)
+        vocab = LazyVocab(os.path.join(path, "words"))
+        if vocab.num_words != V:
+            raise ValueError(f"words file has {vocab.num_words} entries, "
+                             f"index says {V}")
+        m = cls(config, vocab, shard, rank, world, D)
+        log.info("sharded load: rank %d/%d holds %d of %d rows (%s, %s)",
+                 rank, world, shard_size, V, device, tdtype)
+        return m
+
+    # ------------------------------------------------------------------
+    @property
+    def num_words(self) -> int:
+        return self.vocab.num_words
+
+    @property
+    def vector_size(self) -> int:
+        return self.dim
+
+    def _local_norms(self) -> torch.Tensor:
+        """Lazily cached per-shard row norms (mllib:486)."""
+        if self._norms is None:
+            self._norms = self.shard.float().norm(dim=1)
+        return self._norms
+
+    def _global_ids(self, local: torch.Tensor) -> torch.Tensor:
+        return local * self.world + self.rank
+
+    def get_vector(self, word_or_id: Union[str, int]) -> np.ndarray:
+        """Row pull from its owner + broadcast (the Glint single-row pull,
+        mllib:514)."""
+        idx = (self.vocab[word_or_id] if isinstance(word_or_id, str)
+               else int(word_or_id))
+        owner = idx % self.world
+        v = torch.zeros(self.dim, dtype=torch.float32, device=self.device)
+        if self.rank == owner:
+            v.copy_(self.shard[idx // self.world].float())
+        comm.broadcast_(v, src=owner)
+        return v.cpu().numpy()
+
+    transform_word = get_vector
+
+    def transform_words(self, words: Iterable[str]) -> np.ndarray:
+        """Batched word lookup (mllib:529-543): owner-partial scatter +
+        allreduce."""
+        ids = [self.vocab[w] for w in words]
+        out = torch.zeros((len(ids), self.dim), dtype=torch.float32,
+                          device=self.device)
+        idx = torch.tensor(ids, dtype=torch.long, device=self.device)
+        mine = (idx % self.world) == self.rank
+        if bool(mine.any()):
+            out[mine] = self.shard[idx[mine] // self.world].float()
+        comm.all_reduce_sum(out)
+        return out.cpu().numpy()
+
+    def transform_sentences(self, sentences: Sequence[Sequence[str]]
+                            ) -> np.ndarray:
+        """Sentence-average transform (pullAverage, ml:453): each rank sums
+        the rows it owns per sentence, one allreduce merges."""
+        sums = torch.zeros((len(sentences), self.dim), dtype=torch.float32,
+                           device=self.device)
+        counts = np.zeros(len(sentences), dtype=np.float32)
+        sent_idx: List[int] = []
+        tok_ids: List[int] = []
+        for si, s in enumerate(sentences):
+            for w in s:
+                i = self.vocab.get(w)
+                if i >= 0:
+                    counts[si] += 1
+                    if i % self.world == self.rank:
+                        sent_idx.append(si)
+                        tok_ids.append(i // self.world)
+        if tok_ids:
+            rows = self.shard[torch.tensor(tok_ids, dtype=torch.long,
+                                           device=self.device)].float()
+            sums.index_add_(0, torch.tensor(sent_idx, dtype=torch.long,
+                                            device=self.device), rows)
+        comm.all_reduce_sum(sums)
+        out = sums.cpu().numpy()
+        nz = counts > 0
+        out[nz] /= counts[nz, None]
+        return out
+
+    # alias matching GlintWord2VecModel's GPU serving name
+    transform_sentences_gpu = transform_sentences
+
+    def _query_vec(self, word_or_vec) -> tuple:
+        if isinstance(word_or_vec, str):
+            vec = self.get_vector(word_or_vec)
+            return word_or_vec, vec
+        return None, np.asarray(word_or_vec, dtype=np.float32)
+
+    def find_synonyms(self, word_or_vec, num: int) -> List[tuple]:
+        """Top-``num`` cosine neighbours: sharded GEMV + local top-k +
+        allgather merge (mllib:554-630; the multiply op at :598)."""
+        return self.find_synonyms_batch([word_or_vec], num)[0]
+
+    def find_synonyms_batch(self, queries: Sequence, num: int
+                            ) -> List[List[tuple]]:
+        """Batched multi-query findSynonyms: ONE sharded GEMM over all
+        queries + one topk per shard + one allgather (serving throughput
+        path; benchmarks/serving_probe.py)."""
+        Q = len(queries)
+        skip_words = []
+        vecs = np.empty((Q, self.dim), dtype=np.float32)
+        for qi, q in enumerate(queries):
+            w, v = self._query_vec(q)
+            skip_words.append(w)
+            n = np.linalg.norm(v)
+            vecs[qi] = v / n if n > 0 else v
+        qs = torch.from_numpy(vecs).to(self.shard.dtype).to(self.device)
+        scores = (self.shard @ qs.T).float()            # [shard, Q]
+        norms = self._local_norms()
+        cos = torch.where(norms[:, None] > 0, scores / norms[:, None],
+                          torch.zeros((), device=self.device))
+        k = min(num + 1, cos.shape[0])
+        val, loc = torch.topk(cos, k, dim=0)            # [k, Q]
+        gid = self._global_ids(loc.long())
+        # pad to a fixed k across ranks (tiny shards), then allgather
+        kmax = min(num + 1, self.num_words)
+        if k < kmax:
+            pad = torch.full((kmax - k, Q), -2.0, device=self.device)
+            val = torch.cat([val, pad])
+            gid = torch.cat([gid, torch.zeros_like(pad, dtype=torch.long)])
+        if self.world > 1:
+            gval = [torch.empty_like(val) for _ in range(self.world)]
+            ggid = [torch.empty_like(gid) for _ in range(self.world)]
+            torch.distributed.all_gather(gval, val.contiguous())
+            torch.distributed.all_gather(ggid, gid.contiguous())
+            val = torch.cat(gval)                        # [world*kmax, Q]
+            gid = torch.cat(ggid)
+        val = val.cpu().numpy()
+        gid = gid.cpu().numpy()
+        results: List[List[tuple]] = []
+        for qi in range(Q):
+            order = np.argsort(-val[:, qi])
+            out = []
+            for j in order:
+                if val[j, qi] <= -2.0:
+                    continue
+                w = self.vocab.word(int(gid[j, qi]))
+                if w == skip_words[qi]:
+                    continue
+                out.append((w, float(val[j, qi])))
+                if len(out) == num:
+                    break
+            results.append(out)
+        return results
+
+    def analogy(self, pos: List[str], neg: List[str],
+                num: int = 10) -> List[tuple]:
+        v = np.zeros(self.dim, dtype=np.float32)
+        for w in pos:
+            v += self.get_vector(w)
+        for w in neg:
+            v -= self.get_vector(w)
+        skip = set(pos) | set(neg)
+        res = self.find_synonyms(v, num + len(skip))
+        return [(w, c) for w, c in res if w not in skip][:num]
+
+    def stop(self, terminate_other_clients: bool = False) -> None:
+        try:
+            import torch.distributed as dist
+            if dist.is_initialized():
+                dist.destroy_process_group()
+        except Exception:
+            pass
