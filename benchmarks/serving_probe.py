@@ -63,5 +63,26 @@ def main():
           f"{reps / dt:.0f} queries/s ({dt / reps * 1e3:.3f} ms/query)")
 
 
+
+    # --- batched findSynonyms: one GEMM over Q queries + one topk
+    norms_c = norms.clamp_min(1e-12)[:, None]
+    for Q in (256, 1024, 4096):
+        qb = torch.randn(Q, dim)
+        for _ in range(3):
+            cos = gs.multiply_batch(qb) / norms_c
+            torch.topk(cos, 10, dim=0)
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        reps = 20
+        for _ in range(reps):
+            cos = gs.multiply_batch(qb) / norms_c
+            torch.topk(cos, 10, dim=0)
+        torch.cuda.synchronize()
+        dt = time.perf_counter() - t0
+        print(f"findSynonyms batched Q={Q}: "
+              f"{reps * Q / dt / 1e3:.1f}k queries/s "
+              f"({dt / reps * 1e3:.2f} ms per batch)")
+
+
 if __name__ == "__main__":
     main()

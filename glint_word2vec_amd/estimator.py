@@ -408,6 +408,59 @@ class GlintWord2VecModel:
                 break
         return out
 
+    def find_synonyms_batch(self, queries, num: int):
+        """Batched multi-query findSynonyms: one GEMM over all queries +
+        one top-k (GPU when to_gpu() was called, else BLAS on host).
+        Returns a list of (word, cosine) lists, query word excluded."""
+        Q = len(queries)
+        vecs = np.empty((Q, self.vector_size), dtype=np.float32)
+        skip = []
+        for i, q in enumerate(queries):
+            if isinstance(q, str):
+                skip.append(q)
+                vecs[i] = self._f32()[self.vocab[q]]
+            else:
+                skip.append(None)
+                vecs[i] = np.asarray(q, dtype=np.float32)
+        qn = np.linalg.norm(vecs, axis=1, keepdims=True)
+        np.divide(vecs, qn, out=vecs, where=qn > 0)
+        if self._gpu is not None:
+            import torch
+            scores = self._gpu.multiply_batch(torch.from_numpy(vecs))
+            norms = (self._gpu.norms() if self._norms is None
+                     else torch.from_numpy(self._norms)
+                     .to(self._gpu.device))
+            cos = torch.where(norms[:, None] > 0, scores / norms[:, None],
+                              torch.zeros((), device=scores.device))
+            k = min(num + 1, cos.shape[0])
+            val, idx = torch.topk(cos, k, dim=0)
+            val = val.cpu().numpy().T
+            idx = idx.cpu().numpy().T
+        else:
+            scores = self._f32() @ vecs.T
+            norms = self.norms()
+            with np.errstate(divide="ignore", invalid="ignore"):
+                cos = np.where(norms[:, None] > 0, scores / norms[:, None],
+                               0.0)
+            k = min(num + 1, cos.shape[0])
+            part = np.argpartition(-cos, k - 1, axis=0)[:k]
+            idx = np.take_along_axis(
+                part, np.argsort(-np.take_along_axis(cos, part, axis=0),
+                                 axis=0), axis=0).T
+            val = np.take_along_axis(cos.T, idx, axis=1)
+        out = []
+        for qi in range(Q):
+            res = []
+            for j in range(idx.shape[1]):
+                w = self.vocab.words[int(idx[qi, j])]
+                if w == skip[qi]:
+                    continue
+                res.append((w, float(val[qi, j])))
+                if len(res) == num:
+                    break
+            out.append(res)
+        return out
+
     def find_synonyms_df(self, word_or_vec, num: int):
         """DataFrame variant of findSynonyms (ml:390-420 returns a
         (word, similarity) DataFrame)."""

@@ -213,3 +213,13 @@ class GpuSgns:
         v = torch.zeros(self.stride, dtype=self.syn0.dtype, device=self.device)
         v[:self.dim] = vec.to(self.syn0.dtype)
         return (self.syn0 @ v).float()
+
+    def multiply_batch(self, vecs: torch.Tensor) -> torch.Tensor:
+        """Multi-query scores: one rocBLAS GEMM [vocab, stride] x
+        [stride, Q] -> [vocab, Q] — the batched-serving findSynonyms path
+        (one GEMV per query is launch-latency-bound at ~2k q/s; the GEMM
+        amortises it to >100k q/s, benchmarks/serving_probe.py)."""
+        q = torch.zeros((self.stride, vecs.shape[0]), dtype=self.syn0.dtype,
+                        device=self.device)
+        q[:self.dim] = vecs.T.to(self.syn0.dtype).to(self.device)
+        return (self.syn0 @ q).float()

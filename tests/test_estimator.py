@@ -168,3 +168,24 @@ def test_fit_reproducible_same_seed():
 
 def test_stop_noop(tiny_model):
     tiny_model.stop()   # must not raise without dist initialised
+
+
+def test_find_synonyms_batch_matches_single():
+    import numpy as np
+    from glint_word2vec_amd.config import Word2VecConfig
+    from glint_word2vec_amd.estimator import GlintWord2VecModel
+    from glint_word2vec_amd.vocab import Vocabulary
+    rng = np.random.default_rng(2)
+    vocab, dim = 40, 12
+    words = [f"w{i}" for i in range(vocab)]
+    voc = Vocabulary(words=words, counts=np.ones(vocab, dtype=np.int64),
+                     index={w: i for i, w in enumerate(words)},
+                     train_words_count=vocab)
+    syn0 = rng.standard_normal((vocab, dim)).astype(np.float32)
+    m = GlintWord2VecModel(Word2VecConfig(vector_size=dim), voc, syn0)
+    batch = m.find_synonyms_batch(["w3", syn0[7] * 2.0], 5)
+    for got, q in zip(batch, ["w3", syn0[7] * 2.0]):
+        ref = m.find_synonyms(q, 5)
+        assert [w for w, _ in got] == [w for w, _ in ref]
+        np.testing.assert_allclose([c for _, c in got],
+                                   [c for _, c in ref], rtol=1e-5)

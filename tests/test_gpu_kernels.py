@@ -679,3 +679,53 @@ def test_row_engine_pull_path_world1():
         s0, s1 = eng.to_host()
         assert np.isfinite(s0).all() and np.isfinite(s1).all()
         assert not np.array_equal(before, s0)
+
+
+def test_multiply_batch_matches_gemv():
+    _, _, _, table, syn0, syn1 = _problem(vocab=300, dim=64)
+    gs = _gpu_setup(syn0, syn1, table, dtype="float32")
+    q = torch.randn(8, 64)
+    batch = gs.multiply_batch(q)
+    for i in range(8):
+        single = gs.multiply(q[i].cuda())
+        torch.testing.assert_close(batch[:, i], single, rtol=1e-4,
+                                   atol=1e-5)
+
+
+def test_sharded_serving_gpu_world1(tmp_path):
+    """load_sharded onto the GPU (bf16 shard): findSynonyms ranking and
+    transform parity vs the dense f32 model on a cluster-separated
+    embedding (robust to bf16 rounding)."""
+    import numpy as np
+    from glint_word2vec_amd.checkpoint import save_model
+    from glint_word2vec_amd.config import Word2VecConfig
+    from glint_word2vec_amd.estimator import GlintWord2VecModel
+    from glint_word2vec_amd.vocab import Vocabulary
+    rng = np.random.default_rng(5)
+    vocab, dim, ncl = 600, 48, 20
+    centers = rng.standard_normal((ncl, dim)).astype(np.float32) * 5
+    syn0 = (centers[np.arange(vocab) % ncl]
+            + rng.standard_normal((vocab, dim)).astype(np.float32) * 0.05)
+    words = [f"w{i:03d}" for i in range(vocab)]
+    voc = Vocabulary(words=words, counts=np.ones(vocab, dtype=np.int64),
+                     index={w: i for i, w in enumerate(words)},
+                     train_words_count=vocab)
+    path = str(tmp_path / "m")
+    save_model(path, Word2VecConfig(vector_size=dim), voc, syn0,
+               num_shards=3)
+    dense = GlintWord2VecModel.load(path)
+    sharded = GlintWord2VecModel.load_sharded(path, device="cuda")
+    assert sharded.shard.device.type == "cuda"
+    for q in ("w005", "w013"):
+        d = [w for w, _ in dense.find_synonyms(q, 10)]
+        s = [w for w, _ in sharded.find_synonyms(q, 10)]
+        assert set(d) == set(s), (d, s)
+    np.testing.assert_allclose(sharded.get_vector("w077"), syn0[77],
+                               rtol=2e-2, atol=1e-2)
+    got = sharded.transform_sentences([["w001", "w002"]])
+    np.testing.assert_allclose(got[0], syn0[[1, 2]].mean(0), rtol=2e-2,
+                               atol=2e-2)
+    # batched multi-query path on the GPU shard
+    batch = sharded.find_synonyms_batch(["w005", "w013"], 10)
+    assert set(w for w, _ in batch[0]) == \
+        set(w for w, _ in sharded.find_synonyms("w005", 10))
