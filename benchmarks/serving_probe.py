@@ -65,18 +65,18 @@ def main():
 
 
     # --- batched findSynonyms: one GEMM over Q queries + one topk
-    norms_c = norms.clamp_min(1e-12)[:, None]
+    norms_r = norms.clamp_min(1e-12)[None, :]
     for Q in (256, 1024, 4096):
         qb = torch.randn(Q, dim)
         for _ in range(3):
-            cos = gs.multiply_batch(qb) / norms_c
-            torch.topk(cos, 10, dim=0)
+            cos = gs.multiply_batch(qb) / norms_r
+            torch.topk(cos, 10, dim=1)
         torch.cuda.synchronize()
         t0 = time.perf_counter()
         reps = 20
         for _ in range(reps):
-            cos = gs.multiply_batch(qb) / norms_c
-            torch.topk(cos, 10, dim=0)
+            cos = gs.multiply_batch(qb) / norms_r
+            torch.topk(cos, 10, dim=1)
         torch.cuda.synchronize()
         dt = time.perf_counter() - t0
         print(f"findSynonyms batched Q={Q}: "

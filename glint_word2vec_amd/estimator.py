@@ -430,12 +430,12 @@ class GlintWord2VecModel:
             norms = (self._gpu.norms() if self._norms is None
                      else torch.from_numpy(self._norms)
                      .to(self._gpu.device))
-            cos = torch.where(norms[:, None] > 0, scores / norms[:, None],
+            cos = torch.where(norms[None, :] > 0, scores / norms[None, :],
                               torch.zeros((), device=scores.device))
-            k = min(num + 1, cos.shape[0])
-            val, idx = torch.topk(cos, k, dim=0)
-            val = val.cpu().numpy().T
-            idx = idx.cpu().numpy().T
+            k = min(num + 1, cos.shape[1])
+            val, idx = torch.topk(cos, k, dim=1)
+            val = val.cpu().numpy()
+            idx = idx.cpu().numpy()
         else:
             scores = self._f32() @ vecs.T
             norms = self.norms()

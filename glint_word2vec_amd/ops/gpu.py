@@ -125,7 +125,8 @@ class GpuSgns:
                     atomic: bool = True, atomic_below: Optional[int] = None,
                     atomic_floor: int = 0,
                     blocks: Optional[int] = None,
-                    serial: bool = False, pair_mode: int = 1,
+                    serial: bool = False,
+                    pair_mode: Optional[int] = None,
                     stream: Optional[torch.cuda.Stream] = None) -> None:
         """Launch the fused train kernel (async on the given/current stream).
         Stats accumulate on-device; read with read_stats().
@@ -133,6 +134,13 @@ class GpuSgns:
         -1 = positive pairs (+ centers) only — the quality-critical 1/6 of
         update traffic (see benchmarks/quality_probe.py results)."""
         assert self.table is not None, "call set_table first"
+        if pair_mode is None:
+            # two pairs per wave (32-lane halves) wins up to dim ~512; at
+            # NC > 8 the doubled per-wave registers (c_row/grad/t_row of
+            # 2*NC floats each) collapse occupancy — one 64-lane pair per
+            # wave runs 2.5x faster at dim 1024 (benchmarks/results.md
+            # round-2 A/B: 15.0M vs 5.9M words/s at dim=1024 neg=25)
+            pair_mode = 1 if self.stride <= 512 else 0
         num_sent = int(offsets.numel() - 1)
         if num_sent <= 0:
             return
@@ -215,11 +223,13 @@ class GpuSgns:
         return (self.syn0 @ v).float()
 
     def multiply_batch(self, vecs: torch.Tensor) -> torch.Tensor:
-        """Multi-query scores: one rocBLAS GEMM [vocab, stride] x
-        [stride, Q] -> [vocab, Q] — the batched-serving findSynonyms path
-        (one GEMV per query is launch-latency-bound at ~2k q/s; the GEMM
-        amortises it to >100k q/s, benchmarks/serving_probe.py)."""
-        q = torch.zeros((self.stride, vecs.shape[0]), dtype=self.syn0.dtype,
+        """Multi-query scores: one rocBLAS GEMM [Q, stride] x
+        [stride, vocab] -> [Q, vocab] — the batched-serving findSynonyms
+        path.  The [Q, vocab] layout matters: the follow-up top-k then
+        reduces along the CONTIGUOUS last dim (topk over dim 0 of a
+        [vocab, Q] tensor ran 30x slower at Q=4096,
+        benchmarks/serving_probe.py)."""
+        q = torch.zeros((vecs.shape[0], self.stride), dtype=self.syn0.dtype,
                         device=self.device)
-        q[:self.dim] = vecs.T.to(self.syn0.dtype).to(self.device)
-        return (self.syn0 @ q).float()
+        q[:, :self.dim] = vecs.to(self.syn0.dtype).to(self.device)
+        return (q @ self.syn0.T).float()

@@ -260,13 +260,14 @@ class ShardedWord2VecModel:
             n = np.linalg.norm(v)
             vecs[qi] = v / n if n > 0 else v
         qs = torch.from_numpy(vecs).to(self.shard.dtype).to(self.device)
-        scores = (self.shard @ qs.T).float()            # [shard, Q]
+        scores = (qs @ self.shard.T).float()            # [Q, shard]
         norms = self._local_norms()
-        cos = torch.where(norms[:, None] > 0, scores / norms[:, None],
+        cos = torch.where(norms[None, :] > 0, scores / norms[None, :],
                           torch.zeros((), device=self.device))
-        k = min(num + 1, cos.shape[0])
-        val, loc = torch.topk(cos, k, dim=0)            # [k, Q]
-        gid = self._global_ids(loc.long())
+        k = min(num + 1, cos.shape[1])
+        val, loc = torch.topk(cos, k, dim=1)            # [Q, k]
+        val = val.T.contiguous()                        # [k, Q] for gather
+        gid = self._global_ids(loc.long()).T.contiguous()
         # pad to a fixed k across ranks (tiny shards), then allgather
         kmax = min(num + 1, self.num_words)
         if k < kmax:
