@@ -818,6 +818,20 @@ struct TrainPhase {
 
   __device__ __forceinline__ void process_pairs(const uint32_t* tl, int count,
                                                 int64_t) {
+    if (NC >= 16) {
+      // wide rows (dim >= ~1024): the 2-deep prefetch costs an extra NC
+      // registers (b1) that collapse occupancy below what HBM latency
+      // hiding needs — a plain load-use loop keeps ~2 more waves/SIMD in
+      // flight and wins (A/B in benchmarks/results.md round 2)
+      for (int k = 0; k < count; ++k) {
+        const uint32_t e = tl[k];
+        T* p = rowptr(e);
+        float b[NC];
+        RowIO<T, NC>::load(p, b, lane);
+        do_pair(e, b, p);
+      }
+      return;
+    }
     float b0[NC], b1[NC];
     uint32_t e0 = tl[0], e1 = 0;
     T* p0 = rowptr(e0);
