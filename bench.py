@@ -292,7 +292,8 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16" if args.dtype == "bf16" else "fp32",
+            "dtype": ("fp32" if use_cpu else
+                      ("bf16" if args.dtype == "bf16" else "fp32")),
             "data": "synthetic (Zipf 1.05 token stream, random-init weights)",
             "config": {
                 "model": f"sgns vocab={args.vocab} dim={args.dim} "

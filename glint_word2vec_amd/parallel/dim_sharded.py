@@ -162,12 +162,49 @@ class DimShardedSgns:
         if self.is_cuda:
             if self.narrow and self.serial:
                 raise ValueError("serial parity mode requires narrow=False")
+            if (self.world == 1 and not self.serial and not self.narrow
+                    and getattr(self, "single_pass_world1", True)):
+                # world 1 collapses the CIKM scheme: the local partial IS
+                # the full dot, so the dots/allreduce/update pipeline
+                # reduces EXACTLY to the fused single-kernel form —
+                # dot + sigmoid + update in one pass, zero redundant row
+                # reads (the "single-pass variant", VERDICT round-1 #6;
+                # 105.8M -> fused-rate ~190M words/s).  The phase pipeline
+                # stays in use for serial parity tests and world > 1.
+                self._train_step_gpu_fused(tokens, offsets, alpha, window,
+                                           n_neg, seed, sent_id_base)
+                return
             self._train_step_gpu(tokens, offsets, chunks, alpha, window,
                                  n_neg, seed, sent_id_base)
         else:
             for (a, b) in chunks:
                 self._train_chunk_cpu(tokens, offsets, a, b, alpha, window,
                                       n_neg, seed, sent_id_base)
+
+    def _train_step_gpu_fused(self, tokens, offsets, alpha, window, n_neg,
+                              seed, sent_id_base):
+        """World-1 fast path: one fused sgns_train launch over the full-
+        width slice (lo=0, hi=dim, stride == round_stride(dim) — identical
+        table layout to ops.gpu.GpuSgns)."""
+        assert self.lo == 0 and self.hi == self.dim and not self.narrow
+        num_sent = len(offsets) - 1
+        nthreads = 256
+        nblocks = max(1, min((num_sent + 3) // 4, 2048))
+        avg_len = max(1, int(tokens.numel()) // max(num_sent, 1))
+        pos_blocks = min(max(1, 8192 // max(num_sent, 1)),
+                         (avg_len + 95) // 96)
+        pos_blocks = max(1, min(pos_blocks, 11))
+        stream = torch.cuda.current_stream(self.device)
+        self.native.sgns_train(
+            self.syn0.data_ptr(), self.syn1.data_ptr(), int(self.is_bf16),
+            self.stride, tokens.data_ptr(), offsets.data_ptr(), num_sent,
+            0 if self.keep_thr is None else self.keep_thr.data_ptr(),
+            self.table.data_ptr(), int(self.table.numel()), float(alpha),
+            int(window), int(n_neg), seed & 0xFFFFFFFFFFFFFFFF,
+            int(sent_id_base), int(self.window_mode == "reference"),
+            self.atomic_below, self._stats.data_ptr(), nblocks, pos_blocks,
+            nthreads, stream.cuda_stream, 0, 0,
+            1 if self.stride <= 512 else 0, self.atomic_floor)
 
     def _train_step_gpu(self, tokens, offsets, chunks, alpha, window, n_neg,
                         seed, sent_id_base):
