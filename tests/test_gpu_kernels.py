@@ -736,3 +736,38 @@ def test_sharded_serving_gpu_world1(tmp_path):
         assert len(batch[qi]) == 10
         for w, c in batch[qi]:
             assert int(w[1:]) % ncl == q % ncl and c > 0.99
+
+
+def test_dim_engine_world1_single_pass_gpu():
+    """World-1 dim engine auto-collapses to the fused one-kernel form;
+    pair/positive counts must match the phase pipeline exactly (same
+    counter-RNG walker), tables finite with padding intact."""
+    from glint_word2vec_amd.parallel.dim_sharded import DimShardedSgns
+    rng = np.random.default_rng(8)
+    vocab, dim = 4000, 48
+    tokens = torch.from_numpy(
+        rng.integers(0, vocab, 30000).astype(np.int32)).cuda()
+    offsets = torch.from_numpy(
+        np.arange(0, 30001, 100, dtype=np.int32)).cuda()
+    counts = np.bincount(tokens.cpu().numpy(),
+                         minlength=vocab).astype(np.int64) + 1
+
+    def run(single_pass):
+        eng = DimShardedSgns(vocab, dim, dtype="bfloat16", device="cuda",
+                             seed=3, counts=counts, table_size=10007,
+                             atomic=False, narrow=False)
+        eng.single_pass_world1 = single_pass
+        eng.train_step(tokens, offsets, 0.03, 4, 5, seed=11)
+        torch.cuda.synchronize()
+        st = eng.read_stats()
+        s0, s1 = eng.to_host()
+        assert torch.all(eng.syn0[:, dim:] == 0)
+        return st, s0, s1
+
+    fused_st, f0, f1 = run(True)
+    phase_st, p0, p1 = run(False)
+    assert fused_st.pairs == phase_st.pairs > 0
+    assert fused_st.positives == phase_st.positives
+    assert np.isfinite(f0).all() and np.isfinite(f1).all()
+    # same update class: aggregate movement within hogwild-race tolerance
+    assert abs(np.abs(f0).sum() - np.abs(p0).sum()) / np.abs(p0).sum() < 0.2
