@@ -176,12 +176,21 @@ class Word2VecConfig:
             return 2 ** 31 - 1
         return int(self.hybrid_hot_rows)
 
-    def effective_atomic_floor(self) -> int:
-        # Rows below this stay hogwild even in atomic ranges (hybrid
-        # ultra-head contention escape; 0 elsewhere).
-        if self.resolved_update_mode() == "hybrid":
-            return int(self.hybrid_skip_rows)
-        return 0
+    def effective_atomic_floor(self, vocab_size: "int | None" = None
+                                ) -> int:
+        # Rows below this stay hogwild even in atomic ranges (the hybrid
+        # ultra-head contention escape; 0 elsewhere).  The escape exists
+        # for million-row Zipf heads where one row absorbs ~1% of all
+        # negative draws and serializes its cacheline; on small
+        # vocabularies those same top rows ARE the content words (a 3.6k
+        # vocabulary's "berlin" sits in the top 128), so the floor scales
+        # down with vocabulary size.
+        if self.resolved_update_mode() != "hybrid":
+            return 0
+        floor = int(self.hybrid_skip_rows)
+        if vocab_size is not None:
+            floor = min(floor, int(vocab_size) // 1024)
+        return floor
 
     def validate(self) -> None:
         if self.vector_size <= 0:

@@ -102,7 +102,7 @@ def train_gpu(cfg: Word2VecConfig, vocab: Vocabulary, batches_fn: Callable,
                            sent_id_base=sent_base, window_mode=cfg.window_mode,
                            atomic=cfg.resolved_update_mode() != "hogwild",
                            atomic_below=cfg.effective_atomic_below(),
-                           atomic_floor=cfg.effective_atomic_floor())
+                           atomic_floor=cfg.effective_atomic_floor(vocab.num_words))
             sent_base += batch.num_sentences
             processed += batch.num_tokens
             step += 1
@@ -157,7 +157,7 @@ def _train_sharded(cfg: Word2VecConfig, vocab: Vocabulary,
                              f_correction=cfg.f_correction,
                              atomic=cfg.resolved_update_mode() != "hogwild",
                              atomic_below=cfg.effective_atomic_below(),
-                             atomic_floor=cfg.effective_atomic_floor(),
+                             atomic_floor=cfg.effective_atomic_floor(vocab.num_words),
                              **common)
         if init_tables is not None:
             eng.load_host(*init_tables)
@@ -182,7 +182,7 @@ def _train_sharded(cfg: Word2VecConfig, vocab: Vocabulary,
                              sync_every=cfg.sync_every,
                              atomic=cfg.resolved_update_mode() != "hogwild",
                              atomic_below=cfg.effective_atomic_below(),
-                             atomic_floor=cfg.effective_atomic_floor(),
+                             atomic_floor=cfg.effective_atomic_floor(vocab.num_words),
                              **common)
         if init_tables is not None:
             eng.load_host(*init_tables)
@@ -224,7 +224,7 @@ def _train_sharded(cfg: Word2VecConfig, vocab: Vocabulary,
         eng = RowShardedSgns(vocab.num_words, cfg.vector_size,
                              atomic=cfg.resolved_update_mode() == "atomic",
                              atomic_below=cfg.effective_atomic_below(),
-                             atomic_floor=cfg.effective_atomic_floor(),
+                             atomic_floor=cfg.effective_atomic_floor(vocab.num_words),
                              **common)
         if init_tables is not None:
             eng.load_host(*init_tables)

@@ -333,6 +333,9 @@ def test_dim_sharded_narrow_slices_gpu():
         eng = DimShardedSgns(40, dim, device=device, seed=3, counts=counts,
                              table_size=1009, chunk_words=10 ** 9,
                              f_correction=True, atomic=False, narrow=narrow)
+        # comparing narrow vs padded STORAGE: both runs must use the same
+        # (phase) pipeline, not the world-1 fused shortcut
+        eng.single_pass_world1 = False
         if device == "cuda":
             assert eng.narrow == (narrow if narrow is not None else True)
             tok = torch.from_numpy(tokens).cuda()
@@ -536,6 +539,7 @@ def test_dim_sharded_narrow_atomic_gpu():
                              counts=counts, table_size=1009,
                              chunk_words=10 ** 9, f_correction=True,
                              atomic=True, narrow=narrow)
+        eng.single_pass_world1 = False   # storage A/B needs one pipeline
         tok = torch.from_numpy(tokens).cuda()
         off = torch.from_numpy(offsets).cuda()
         eng.train_step(tok, off, 0.04, 3, 4, seed=11, offsets_host=offsets)
@@ -685,10 +689,10 @@ def test_multiply_batch_matches_gemv():
     _, _, _, table, syn0, syn1 = _problem(vocab=300, dim=64)
     gs = _gpu_setup(syn0, syn1, table, dtype="float32")
     q = torch.randn(8, 64)
-    batch = gs.multiply_batch(q)
+    batch = gs.multiply_batch(q)          # [Q, vocab]
     for i in range(8):
         single = gs.multiply(q[i].cuda())
-        torch.testing.assert_close(batch[:, i], single, rtol=1e-4,
+        torch.testing.assert_close(batch[i], single, rtol=1e-4,
                                    atol=1e-5)
 
 
