@@ -64,6 +64,19 @@ def main():
 
 
 
+    # --- graph-replayed single query (hipGraph: one launch per query)
+    qv = torch.randn(dim).cuda()
+    gs.synonyms_query(qv, 11)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    reps = 2000
+    for _ in range(reps):
+        gs.synonyms_query(qv, 11)
+    torch.cuda.synchronize()
+    dt = time.perf_counter() - t0
+    print(f"findSynonyms graphed single query: {reps / dt:.0f} queries/s "
+          f"({dt / reps * 1e3:.3f} ms/query)")
+
     # --- batched findSynonyms: one GEMM over Q queries + one topk
     norms_r = norms.clamp_min(1e-12)[None, :]
     for Q in (256, 1024, 4096):

@@ -52,6 +52,11 @@ def _add_train(sub):
     p.add_argument("--checkpoint-every", type=int, default=0, metavar="N",
                    help="write a loadable checkpoint every N steps "
                         "(<model>-step<N> directories; 0 = off)")
+    p.add_argument("--updates", choices=["hogwild", "atomic", "hybrid"],
+                   default="hybrid",
+                   help="row-update mode (hybrid: atomics on the hot "
+                        "Zipf band, hogwild elsewhere — the measured "
+                        "quality/speed default)")
 
 
 def main(argv=None):
@@ -63,10 +68,14 @@ def main(argv=None):
     p = sub.add_parser("similar", help="top-N cosine-similar words")
     p.add_argument("model"); p.add_argument("word")
     p.add_argument("-n", type=int, default=10)
+    p.add_argument("--sharded", action="store_true",
+                   help="stream the model to the GPU(s) shard-wise "
+                        "(no host materialisation; torchrun for multi-GPU)")
     p = sub.add_parser("analogy", help="pos... - neg... vector arithmetic")
     p.add_argument("model"); p.add_argument("pos", nargs="+")
     p.add_argument("-m", "--minus", nargs="+", default=[])
     p.add_argument("-n", type=int, default=10)
+    p.add_argument("--sharded", action="store_true")
     p = sub.add_parser("export", help="write word2vec text format (toLocal)")
     p.add_argument("model"); p.add_argument("out")
     p = sub.add_parser("info", help="print model metadata")
@@ -83,7 +92,8 @@ def main(argv=None):
             max_sentence_length=args.max_sentence_length,
             dtype=args.dtype, device=args.device, engine=args.engine,
             window_mode=args.window_mode, num_partitions=args.workers,
-            checkpoint_every=args.checkpoint_every)
+            checkpoint_every=args.checkpoint_every,
+            update_mode=args.updates)
         model = est.fit(args.corpus,
                         save_path=(args.model if args.checkpoint_every
                                    else None),
@@ -96,11 +106,13 @@ def main(argv=None):
                   f"({model.num_words} words, dim {model.vector_size})")
         model.stop()
     elif args.cmd == "similar":
-        model = GlintWord2VecModel.load(args.model)
+        model = (GlintWord2VecModel.load_sharded(args.model) if args.sharded
+                 else GlintWord2VecModel.load(args.model))
         for w, c in model.find_synonyms(args.word, args.n):
             print(f"{c:.4f}\t{w}")
     elif args.cmd == "analogy":
-        model = GlintWord2VecModel.load(args.model)
+        model = (GlintWord2VecModel.load_sharded(args.model) if args.sharded
+                 else GlintWord2VecModel.load(args.model))
         for w, c in model.analogy(args.pos, args.minus, args.n):
             print(f"{c:.4f}\t{w}")
     elif args.cmd == "export":

@@ -388,8 +388,25 @@ class GlintWord2VecModel:
             vec = vec / qn
         if self._gpu is not None:
             import torch
-            scores = self._gpu.multiply(
-                torch.from_numpy(vec).to(self._gpu.device)).cpu().numpy()
+            try:
+                # hipGraph-replayed GEMV+topk (launch-latency bound op)
+                k = min(num + 1, self.num_words)
+                val, idx = self._gpu.synonyms_query(torch.from_numpy(vec), k)
+                val = val.cpu().numpy()
+                idx = idx.cpu().numpy()
+                out = []
+                for j in range(len(idx)):
+                    w = self.vocab.words[int(idx[j])]
+                    if w == query_word:
+                        continue
+                    out.append((w, float(val[j])))
+                    if len(out) == num:
+                        break
+                return out
+            except RuntimeError:
+                # graph capture unavailable: plain kernel-by-kernel path
+                scores = self._gpu.multiply(
+                    torch.from_numpy(vec).to(self._gpu.device)).cpu().numpy()
         else:
             scores = self._f32() @ vec          # `multiply` (mllib:598)
         norms = self.norms()

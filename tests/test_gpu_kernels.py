@@ -775,3 +775,18 @@ def test_dim_engine_world1_single_pass_gpu():
     assert np.isfinite(f0).all() and np.isfinite(f1).all()
     # same update class: aggregate movement within hogwild-race tolerance
     assert abs(np.abs(f0).sum() - np.abs(p0).sum()) / np.abs(p0).sum() < 0.2
+
+
+def test_synonyms_query_graphed_matches_plain():
+    """hipGraph-replayed GEMV+topk must equal the kernel-by-kernel path,
+    including after a second query (replay reuses captured buffers)."""
+    _, _, _, table, syn0, syn1 = _problem(vocab=500, dim=64)
+    gs = _gpu_setup(syn0, syn1, table, dtype="float32")
+    norms = gs.norms().clamp_min(1e-12)
+    for seed in (0, 1):
+        q = torch.randn(64, generator=torch.Generator().manual_seed(seed))
+        val, idx = gs.synonyms_query(q, 7)
+        ref = (gs.multiply(q.cuda()) / norms)
+        rv, ri = torch.topk(ref, 7)
+        torch.testing.assert_close(val, rv, rtol=1e-4, atol=1e-5)
+        assert torch.equal(idx, ri)
