@@ -104,60 +104,82 @@ class ReplicatedSgns:
         if self._steps_since_sync >= self.sync_every:
             self.sync()
 
-    def sync(self) -> None:
-        """delta = replica - master; allreduce(sum); master += delta;
-        replica <- master.  Applies every rank's updates exactly once.
+    def sync(self, flush: bool = False) -> None:
+        """Merge table deltas across ranks, applied ONE sync period late so
+        the allreduce overlaps training (round-2 overlap design):
 
-        On CUDA the tables are processed in row chunks with the allreduce
-        on a separate stream, so chunk k's xGMI transfer overlaps chunk
-        k+1's delta compute — same result, about half the sync wall time.
-        """
+          round r:  capture delta_r = replica - master; start
+                    allreduce(delta_r) on the comm stream; training of the
+                    next sync_every steps proceeds concurrently.
+          round r+1: wait allreduce(delta_{r}); master += sum(delta_r);
+                    replica += sum(delta_r) - own(delta_r)  (the replica
+                    already contains its own delta plus newer local
+                    updates — every update still lands exactly once).
+
+        This is the reference's fire-and-forget adjust at the engine
+        level: cross-rank state is bounded-stale by <= 2*sync_every steps
+        instead of sync blocking the step loop (~15 ms per sync at world 8
+        for vocab-1M tables).  ``flush=True`` drains the pipeline AND runs
+        one synchronous round — callers that need the agreed state
+        (save/to_host) get it."""
         self._steps_since_sync = 0
         if self.world == 1:
             self.master0.copy_(self.syn0.float())
             self.master1.copy_(self.syn1.float())
             return
-        if self.is_cuda:
-            self._sync_cuda_pipelined()
-            return
-        for syn, master in ((self.syn0, self.master0),
-                            (self.syn1, self.master1)):
-            delta = syn.float() - master
-            comm.all_reduce_sum_compressed(delta)
-            master += delta
-            syn.copy_(master.to(syn.dtype))
+        self._apply_pending()
+        self._begin_round()
+        if flush:
+            self._apply_pending()
 
-    def _sync_cuda_pipelined(self, chunk_rows: int = 1 << 19) -> None:
+    # -- async round machinery ------------------------------------------
+    def _comm_ctx(self):
+        if not self.is_cuda:
+            import contextlib
+            return contextlib.nullcontext()
         if not hasattr(self, "_comm_stream"):
             self._comm_stream = torch.cuda.Stream(self.device)
-        cs = self._comm_stream
-        comp = torch.cuda.current_stream(self.device)
-        prev = None   # (syn_c, master_c, delta, ar_event)
+        return torch.cuda.stream(self._comm_stream)
 
-        def apply(entry):
-            syn_c, master_c, delta, ev = entry
-            comp.wait_event(ev)
-            master_c += delta
-            syn_c.copy_(master_c.to(syn_c.dtype))
-
+    def _begin_round(self) -> None:
+        """Capture deltas and launch their allreduce on the comm stream."""
+        pend = []
         for syn, master in ((self.syn0, self.master0),
                             (self.syn1, self.master1)):
-            for r0 in range(0, syn.shape[0], chunk_rows):
-                syn_c = syn[r0:r0 + chunk_rows]
-                master_c = master[r0:r0 + chunk_rows]
-                delta = syn_c.float() - master_c
-                ev = torch.cuda.Event()
-                ev.record(comp)
-                with torch.cuda.stream(cs):
-                    cs.wait_event(ev)
-                    comm.all_reduce_sum_compressed(delta)
-                    ar_ev = torch.cuda.Event()
-                    ar_ev.record(cs)
-                if prev is not None:
-                    apply(prev)
-                prev = (syn_c, master_c, delta, ar_ev)
-        if prev is not None:
-            apply(prev)
+            own = syn.float() - master          # on compute stream
+            total = own.clone()
+            pend.append((own, total))
+        if self.is_cuda:
+            ev = torch.cuda.Event()
+            ev.record(torch.cuda.current_stream(self.device))
+        with self._comm_ctx():
+            if self.is_cuda:
+                torch.cuda.current_stream(self.device).wait_event(ev)
+            for own, total in pend:
+                comm.all_reduce_sum_compressed(total)
+            if self.is_cuda:
+                done = torch.cuda.Event()
+                done.record(torch.cuda.current_stream(self.device))
+                self._round_done = done
+        self._pending = pend
+
+    def _apply_pending(self) -> None:
+        """Fold a completed round into master + replica (compute stream)."""
+        pend = getattr(self, "_pending", None)
+        if pend is None:
+            return
+        self._pending = None
+        if self.is_cuda:
+            torch.cuda.current_stream(self.device).wait_event(
+                self._round_done)
+        for (own, total), (syn, master) in zip(
+                pend, ((self.syn0, self.master0),
+                       (self.syn1, self.master1))):
+            master += total
+            # replica already holds `own` (and newer local updates):
+            # add only the other ranks' contribution
+            total -= own
+            syn += total.to(syn.dtype)
 
     def load_host(self, syn0, syn1) -> None:
         """Initialise from full host f32 matrices (training resume)."""
@@ -189,7 +211,7 @@ class ReplicatedSgns:
         """Streamed checkpoint from the fp32 masters (replicated — rank 0
         writes, host memory O(block_rows * dim)).  Collective: sync() flushes
         pending deltas on every rank first."""
-        self.sync()
+        self.sync(flush=True)
         if self.rank == 0:
             from ..checkpoint import save_model_streaming
 
@@ -205,7 +227,7 @@ class ReplicatedSgns:
         comm.barrier()
 
     def to_host(self):
-        self.sync()
+        self.sync(flush=True)
         d = self.dim
         if self.is_cuda:
             return (self.master0[:, :d].cpu().numpy().copy(),

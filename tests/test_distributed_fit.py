@@ -142,3 +142,46 @@ def test_distributed_streaming_save(tmp_path, engine):
     live1 = np.load(tmp_path / "syn1_live.npy")
     np.testing.assert_allclose(s0, live0, rtol=1e-6, atol=1e-7)
     np.testing.assert_allclose(s1, live1, rtol=1e-6, atol=1e-7)
+
+
+def _dp_conservation_worker(rank, world, rdv, out_dir):
+    """Controlled dp-sync check: each rank injects known constant deltas
+    between syncs; after flush every update must have landed exactly once
+    on every rank (master identical everywhere = init + sum of all
+    injections)."""
+    import torch
+    import torch.distributed as dist
+    dist.init_process_group("gloo", rank=rank, world_size=world,
+                            init_method=f"file://{rdv}")
+    try:
+        from glint_word2vec_amd.parallel.replicated import ReplicatedSgns
+        eng = ReplicatedSgns(50, 8, device="cpu", seed=3, table_size=101,
+                             sync_every=1)
+        init0 = eng.master0.clone()
+        total_injected = torch.zeros_like(eng.syn0)
+        for r in range(5):
+            inj = float((rank + 1) * (r + 1))   # distinct per rank/round
+            eng.syn0 += inj
+            total_injected += inj
+            eng.sync()
+        eng.sync(flush=True)
+        # sum over ranks of each round's injection
+        world_inj = torch.zeros_like(total_injected)
+        for rk in range(world):
+            for r in range(5):
+                world_inj += float((rk + 1) * (r + 1))
+        torch.testing.assert_close(eng.master0, init0 + world_inj)
+        torch.testing.assert_close(eng.syn0, eng.master0)
+        np.save(os.path.join(out_dir, f"dpcons_{rank}.npy"),
+                eng.master0.numpy())
+    finally:
+        dist.destroy_process_group()
+
+
+def test_dp_async_sync_applies_exactly_once(tmp_path):
+    rdv = str(tmp_path / "rdv_dpc")
+    mp.spawn(_dp_conservation_worker, args=(2, rdv, str(tmp_path)), nprocs=2,
+             join=True)
+    a = np.load(tmp_path / "dpcons_0.npy")
+    b = np.load(tmp_path / "dpcons_1.npy")
+    np.testing.assert_allclose(a, b)

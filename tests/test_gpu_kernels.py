@@ -463,27 +463,6 @@ def test_estimator_end_to_end_gpu():
     assert len(m.find_synonyms("aa", 5)) == 5
 
 
-def test_dp_pipelined_sync_mechanics():
-    """The chunked comm-stream sync used at world > 1: at world 1 the
-    allreduce no-ops, so its stream/event pipeline must reproduce the
-    simple path exactly (master == syn.float(), replica unchanged)."""
-    from glint_word2vec_amd.parallel.replicated import ReplicatedSgns
-    rng = np.random.default_rng(3)
-    counts = rng.integers(1, 100, 3000).astype(np.int64)
-    eng = ReplicatedSgns(3000, 48, dtype="bfloat16", device="cuda", seed=5,
-                         counts=counts, table_size=10007, sync_every=10**9)
-    tokens = torch.from_numpy(rng.integers(0, 3000, 5000).astype(np.int32)).cuda()
-    offsets = torch.from_numpy(np.arange(0, 5001, 100, dtype=np.int32)).cuda()
-    eng.train_step(tokens, offsets, 0.03, 4, 5, seed=7)
-    torch.cuda.synchronize()
-    syn_before = eng.syn0.clone()
-    eng._sync_cuda_pipelined(chunk_rows=700)   # uneven chunks on purpose
-    torch.cuda.synchronize()
-    assert torch.equal(eng.syn0, syn_before)
-    assert torch.equal(eng.master0, eng.syn0.float())
-    assert torch.equal(eng.master1, eng.syn1.float())
-
-
 def test_row_engine_direct_mode():
     """World-1 direct mode (pairs kernel straight on padded shard tables)
     must train, keep padding zero, and leave usable embeddings."""
