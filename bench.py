@@ -207,6 +207,11 @@ def main():
             trainer.train_step(tok, off, alpha, args.window, args.neg,
                                seed=99 + rank, sent_id_base=i * nsent)
         elif dist_mode == "row":
+            if world == 1 and trainer.is_cuda:
+                trainer.train_batch_fused(tok, off, alpha, args.window,
+                                          args.neg, 99 + rank,
+                                          sent_id_base=i * nsent)
+                return
             if trainer.is_cuda:
                 plan = trainer.make_plan_counter(
                     tok, off, args.window, args.neg, 99 + rank,
@@ -215,10 +220,6 @@ def main():
                 plan = trainer._to_plan_t(trainer.make_plan(
                     batch.tokens, batch.offsets, args.window, args.neg,
                     row_rng))
-            if world == 1 and trainer.is_cuda:
-                trainer.train_step(batch.tokens, batch.offsets, alpha,
-                                   args.window, args.neg, row_rng, plan=plan)
-                return
             # pipelined: issue this step's pull, then train+push the
             # previous step (pull k+1 overlaps train k — row_sharded.py)
             st = trainer.pull_begin(plan)

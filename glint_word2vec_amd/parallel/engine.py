@@ -253,16 +253,15 @@ def _train_sharded(cfg: Word2VecConfig, vocab: Vocabulary,
             # Disjoint per-rank sentence-id streams (data parallel).
             sbase = (rank << 48) + sent_base
             if eng.is_cuda and world == 1:
-                # direct mode: the pairs kernel trains straight on the
-                # padded shard tables — no pull cycle at all
+                # fused fast path: the shard IS the full table at kernel
+                # stride — one fused launch per step, no plan cycle
                 for k in range(n_steps):
                     tokens, offsets = batch_at(k)
-                    plan = eng.make_plan_counter(
-                        tokens, offsets, cfg.window, cfg.n, seed,
-                        sent_id_base=sbase)
+                    tok = torch.from_numpy(tokens).to(eng.device)
+                    off = torch.from_numpy(offsets).to(eng.device)
+                    eng.train_batch_fused(tok, off, alpha_now(), cfg.window,
+                                          cfg.n, seed, sent_id_base=sbase)
                     sbase += max(len(offsets) - 1, 0)
-                    eng.train_step(tokens, offsets, alpha_now(), cfg.window,
-                                   cfg.n, rng, plan=plan)
                     processed += len(tokens)
                     step += 1
                     _maybe_mid_checkpoint(

@@ -769,3 +769,35 @@ def test_synonyms_query_graphed_matches_plain():
         rv, ri = torch.topk(ref, 7)
         torch.testing.assert_close(val, rv, rtol=1e-4, atol=1e-5)
         assert torch.equal(idx, ri)
+
+
+def test_row_engine_fused_fast_path_world1():
+    """train_batch_fused (world-1 row engine) must walk the same pairs as
+    the plan+direct path (identical counter RNG) and train the tables."""
+    from glint_word2vec_amd.parallel.row_sharded import RowShardedSgns
+    from glint_word2vec_amd.data import synthetic_corpus
+    batch = synthetic_corpus(vocab_size=2000, num_tokens=20000,
+                             sentence_len=100, seed=4, zipf_a=1.01)
+    counts = np.bincount(batch.tokens, minlength=2000).astype(np.int64) + 1
+    tok = torch.from_numpy(batch.tokens).cuda()
+    off = torch.from_numpy(batch.offsets).cuda()
+
+    eng = RowShardedSgns(2000, 64, dtype="bfloat16", device="cuda", seed=3,
+                         counts=counts, table_size=10007, atomic=False)
+    before = eng.to_host()[0].copy()
+    eng.train_batch_fused(tok, off, 0.03, 4, 5, seed=9)
+    torch.cuda.synchronize()
+    st_fused = eng.read_stats()
+
+    eng2 = RowShardedSgns(2000, 64, dtype="bfloat16", device="cuda", seed=3,
+                          counts=counts, table_size=10007, atomic=False)
+    plan = eng2.make_plan_counter(tok, off, 4, 5, seed=9)
+    eng2.train_step(batch.tokens, batch.offsets, 0.03, 4, 5,
+                    np.random.default_rng(1), plan=plan)
+    torch.cuda.synchronize()
+    st_plan = eng2.read_stats()
+
+    assert st_fused.pairs == st_plan.pairs > 0
+    assert st_fused.positives == st_plan.positives
+    s0 = eng.to_host()[0]
+    assert np.isfinite(s0).all() and not np.array_equal(before, s0)
