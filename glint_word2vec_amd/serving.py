@@ -310,6 +310,58 @@ class ShardedWord2VecModel:
         res = self.find_synonyms(v, num + len(skip))
         return [(w, c) for w, c in res if w not in skip][:num]
 
+    def to_local(self, max_bytes: int = 16 << 30):
+        """Assemble a host LocalWord2VecModel (the reference's toLocal,
+        mllib:651-659).  Deliberately guarded: materialising an 80M-vocab
+        matrix needs ~96 GB of host RAM — raise ``max_bytes`` explicitly
+        for giant models."""
+        need = self.num_words * self.dim * 4
+        if need > max_bytes:
+            raise ValueError(
+                f"toLocal would materialise {need / 2**30:.1f} GB on host; "
+                f"pass max_bytes >= {need} to allow it, or use "
+                "export_text() for a streaming export")
+        from .estimator import LocalWord2VecModel
+        full = np.empty((self.num_words, self.dim), dtype=np.float32)
+        block = 1 << 20
+        for r0 in range(0, self.num_words, block):
+            r1 = min(self.num_words, r0 + block)
+            full[r0:r1] = self._pull_range(r0, r1)
+        words = [self.vocab.word(i) for i in range(self.num_words)]
+        return LocalWord2VecModel(words, full)
+
+    def _pull_range(self, r0: int, r1: int) -> np.ndarray:
+        """Rows [r0, r1) assembled from their owners (partial + allreduce
+        merge, O(block) at a time)."""
+        ids = torch.arange(r0, r1, dtype=torch.long, device=self.device)
+        out = torch.zeros((r1 - r0, self.dim), dtype=torch.float32,
+                          device=self.device)
+        mine = (ids % self.world) == self.rank
+        if bool(mine.any()):
+            out[mine] = self.shard[ids[mine] // self.world].float()
+        comm.all_reduce_sum(out)
+        return out.cpu().numpy()
+
+    def export_text(self, path: str, block: int = 1 << 18) -> None:
+        """Streaming word2vec-text export (toLocal().save without the host
+        matrix): rank 0 writes; every rank participates in the pulls."""
+        f = open(path, "w", encoding="utf-8") if self.rank == 0 else None
+        try:
+            if f:
+                f.write(f"{self.num_words} {self.dim}\n")
+            for r0 in range(0, self.num_words, block):
+                r1 = min(self.num_words, r0 + block)
+                rows = self._pull_range(r0, r1)
+                if f:
+                    for i in range(r0, r1):
+                        v = rows[i - r0]
+                        f.write(self.vocab.word(i) + " "
+                                + " ".join(f"{x:.6g}" for x in v) + "\n")
+        finally:
+            if f:
+                f.close()
+        comm.barrier()
+
     def stop(self, terminate_other_clients: bool = False) -> None:
         try:
             import torch.distributed as dist

@@ -144,3 +144,21 @@ def test_row_engine_streaming_resume_any_shards(tmp_path):
     got0, got1 = eng.to_host()
     np.testing.assert_allclose(got0, syn0, rtol=1e-6)
     np.testing.assert_allclose(got1, syn1, rtol=1e-6)
+
+
+def test_sharded_to_local_and_export(tmp_path):
+    path = str(tmp_path / "model")
+    voc, syn0, _ = _make_checkpoint(path, num_shards=2)
+    m = GlintWord2VecModel.load_sharded(path, device="cpu")
+    local = m.to_local()
+    np.testing.assert_allclose(local.vectors, syn0, rtol=1e-6)
+    assert local.words[:3] == ["w000", "w001", "w002"]
+    with pytest.raises(ValueError):
+        m.to_local(max_bytes=10)
+    out = str(tmp_path / "vecs.txt")
+    m.export_text(out, block=17)
+    lines = open(out, encoding="utf-8").read().splitlines()
+    assert lines[0] == "60 24"
+    assert lines[1].split()[0] == "w000"
+    got = np.array([float(x) for x in lines[8].split()[1:]])
+    np.testing.assert_allclose(got, syn0[7], rtol=1e-4, atol=1e-5)
