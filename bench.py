@@ -56,6 +56,11 @@ def parse_args():
                    help="alias for --updates atomic")
     p.add_argument("--atomic-below", type=int, default=None,
                    help="alias for --updates hybrid --hot-rows K")
+    p.add_argument("--shared-negatives", action="store_true",
+                   help="HogBatch-style shared negatives: one draw set per "
+                        "position reused across its contexts (opt-in; cuts "
+                        "target-row traffic ~(1+n)/(1+n/2b)-fold at high "
+                        "n_neg — the dim-1024 neg-25 config)")
     p.add_argument("--pair-mode", type=int, default=None, choices=[0, 1, 2],
                    help="fused-kernel variant: 0=one pair/wave (64-lane), "
                         "1=two pairs (32-lane halves, default), "
@@ -162,7 +167,8 @@ def main():
                                  subsample=args.subsample,
                                  chunk_words=args.chunk_words,
                                  atomic=upd_atomic, atomic_below=upd_below,
-                                 atomic_floor=args.hot_floor)
+                                 atomic_floor=args.hot_floor,
+                                 shared_negatives=args.shared_negatives)
         dist_mode = True
     elif engine == "dp":
         from glint_word2vec_amd.parallel.replicated import ReplicatedSgns
@@ -172,7 +178,8 @@ def main():
                                  subsample=args.subsample,
                                  sync_every=args.sync_every,
                                  atomic=upd_atomic, atomic_below=upd_below,
-                                 atomic_floor=args.hot_floor)
+                                 atomic_floor=args.hot_floor,
+                                 shared_negatives=args.shared_negatives)
         dist_mode = "dp"
     elif engine == "row":
         from glint_word2vec_amd.parallel.row_sharded import RowShardedSgns
@@ -182,7 +189,8 @@ def main():
                                  subsample=args.subsample,
                                  atomic=args.updates == "atomic",
                                  atomic_below=upd_below,
-                                 atomic_floor=args.hot_floor)
+                                 atomic_floor=args.hot_floor,
+                                 shared_negatives=args.shared_negatives)
         dist_mode = "row"
     else:
         from glint_word2vec_amd.ops.gpu import GpuSgns
@@ -237,6 +245,7 @@ def main():
                                 atomic_below=(None if args.updates == "atomic"
                                               else upd_below),
                                 atomic_floor=args.hot_floor,
+                                shared_negatives=args.shared_negatives,
                                 blocks=args.blocks or None,
                                 **({} if args.pair_mode is None
                                    else {"pair_mode": args.pair_mode}))
@@ -306,6 +315,7 @@ def main():
                 "updates": (args.updates if args.updates != "hybrid" else
                             f"hybrid(atomic rows "
                             f"{args.hot_floor}..{args.hot_rows})"),
+                "shared_negatives": bool(args.shared_negatives),
             },
             "pairs_per_step": st.pairs / max(args.steps, 1),
             "mean_fplus": st.sum_fplus / max(st.positives, 1),

@@ -69,6 +69,7 @@ def main():
     ap.add_argument("--mode",
                     choices=["hogwild", "atomic", "hybrid", "positives"],
                     default="hogwild")
+    ap.add_argument("--shared-negatives", action="store_true")
     ap.add_argument("--hot-floor", type=int, default=0,
                     help="hybrid: rows < F stay hogwild (contention escape)")
     ap.add_argument("--hot-rows", type=int, default=32768,
@@ -106,7 +107,8 @@ def main():
         gs.train_batch(tok, off, args.lr, args.window, args.neg, 42,
                        atomic=(args.mode != "hogwild"), atomic_below=abelow,
                        atomic_floor=(args.hot_floor
-                                     if args.mode == "hybrid" else 0))
+                                     if args.mode == "hybrid" else 0),
+                       shared_negatives=args.shared_negatives)
         torch.cuda.synchronize()
         st = gs.read_stats()
         syn0, _ = gs.to_host()
@@ -124,6 +126,8 @@ def main():
     acc = nn_accuracy(syn0, range(args.eval_lo, args.eval_hi))
     mode = (f"hybrid[{args.hot_floor}..{args.hot_rows}]"
             if args.mode == "hybrid" else args.mode)
+    if args.shared_negatives:
+        mode += "+sharedneg"
     print(f"mode={mode} device={args.device} vocab={args.vocab} "
           f"words={args.words}: planted-NN acc={acc:.3f} "
           f"({args.words / dt / 1e6:.1f}M words/s incl. setup)")

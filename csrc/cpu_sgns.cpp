@@ -86,7 +86,8 @@ static void train_sentences(float* syn0, float* syn1, int64_t dim,
                             const int32_t* table, int64_t table_size,
                             float alpha, int window, int n_neg,
                             uint64_t seed, int64_t sent_id_base,
-                            bool reference_window, Stats* stats,
+                            bool reference_window, bool shared_neg,
+                            Stats* stats,
                             std::vector<int32_t>& kept,
                             std::vector<float>& c_row,
                             std::vector<float>& grad,
@@ -143,8 +144,11 @@ static void train_sentences(float* syn0, float* syn1, int64_t dim,
           }
           stats->pairs++; stats->positives++; stats->sum_fplus += f;
         }
-        uint64_t kbase = kNegBase +
-            (uint64_t)(i * (2 * window + 1) + (j - i + window)) * (uint64_t)n_neg;
+        uint64_t kbase = shared_neg
+            ? kNegBase + (uint64_t)i * (uint64_t)n_neg
+            : kNegBase +
+              (uint64_t)(i * (2 * window + 1) + (j - i + window)) *
+                  (uint64_t)n_neg;
         for (int k = 0; k < n_neg; ++k) {
           uint32_t un = draw_u32(base, kbase + (uint64_t)k);
           int32_t neg = table[un % (uint64_t)table_size];
@@ -180,8 +184,8 @@ template <typename Phase>
 static void walk_sentence(const int32_t* sent, int64_t len, uint64_t base,
                           const float* keep_prob, const int32_t* table,
                           int64_t table_size, int window, int n_neg,
-                          bool reference_window, std::vector<int32_t>& kept,
-                          Phase& ph) {
+                          bool reference_window, bool shared_neg,
+                          std::vector<int32_t>& kept, Phase& ph) {
   kept.clear();
   if (keep_prob) {
     for (int64_t p = 0; p < len; ++p) {
@@ -218,8 +222,11 @@ static void walk_sentence(const int32_t* sent, int64_t len, uint64_t base,
       if (j == i) continue;
       int32_t t = kept[j];
       ph.pair(c, t, 1.0f, pair_idx++);
-      uint64_t kbase = kNegBase +
-          (uint64_t)(i * (2 * window + 1) + (j - i + window)) * (uint64_t)n_neg;
+      uint64_t kbase = shared_neg
+          ? kNegBase + (uint64_t)i * (uint64_t)n_neg
+          : kNegBase +
+            (uint64_t)(i * (2 * window + 1) + (j - i + window)) *
+                (uint64_t)n_neg;
       for (int k = 0; k < n_neg; ++k) {
         uint32_t un = draw_u32(base, kbase + (uint64_t)k);
         int32_t neg = table[un % (uint64_t)table_size];
@@ -309,7 +316,7 @@ static py::array_t<int64_t> count_pairs(
     py::object keep_prob_obj,
     py::array_t<int32_t, py::array::c_style> table,
     int window, int n_neg, uint64_t seed, int64_t sent_id_base,
-    std::string window_mode) {
+    std::string window_mode, int shared_negatives) {
   int64_t num_sent = offsets.shape(0) - 1;
   const float* keep_prob = nullptr;
   py::array_t<float, py::array::c_style> kp_arr;
@@ -328,6 +335,7 @@ static py::array_t<int64_t> count_pairs(
     walk_sentence(tok + off[s], off[s + 1] - off[s],
                   sentence_base(seed, (uint64_t)(sent_id_base + s)), keep_prob,
                   table.data(), table.shape(0), window, n_neg, ref_window,
+                  shared_negatives != 0,
                   kept, ph);
     o[s] = ph.pairs;
   }
@@ -344,7 +352,7 @@ static void dots_slice(
     int window, int n_neg, uint64_t seed, int64_t sent_id_base,
     std::string window_mode,
     py::array_t<int64_t, py::array::c_style> pair_offsets,
-    py::array_t<float, py::array::c_style> f_out) {
+    py::array_t<float, py::array::c_style> f_out, int shared_negatives) {
   int64_t num_sent = offsets.shape(0) - 1;
   const float* keep_prob = nullptr;
   py::array_t<float, py::array::c_style> kp_arr;
@@ -363,6 +371,7 @@ static void dots_slice(
     walk_sentence(tok + off[s], off[s + 1] - off[s],
                   sentence_base(seed, (uint64_t)(sent_id_base + s)), keep_prob,
                   table.data(), table.shape(0), window, n_neg, ref_window,
+                  shared_negatives != 0,
                   kept, ph);
   }
 }
@@ -378,7 +387,7 @@ static py::dict update_slice(
     std::string window_mode,
     py::array_t<int64_t, py::array::c_style> pair_offsets,
     py::array_t<float, py::array::c_style> f_in,
-    py::object f_loc_obj, float world_scale) {
+    py::object f_loc_obj, float world_scale, int shared_negatives) {
   int64_t num_sent = offsets.shape(0) - 1;
   const float* keep_prob = nullptr;
   py::array_t<float, py::array::c_style> kp_arr;
@@ -408,6 +417,7 @@ static py::dict update_slice(
     walk_sentence(tok + off[s], off[s + 1] - off[s],
                   sentence_base(seed, (uint64_t)(sent_id_base + s)), keep_prob,
                   table.data(), table.shape(0), window, n_neg, ref_window,
+                  shared_negatives != 0,
                   kept, ph);
   }
   py::dict d;
@@ -479,7 +489,7 @@ static py::dict train_batch(
     float alpha, int window, int n_neg,
     uint64_t seed, int64_t sent_id_base,
     std::string window_mode, int num_threads,
-    py::object exp_table_obj = py::none()) {
+    py::object exp_table_obj = py::none(), int shared_negatives = 0) {
   if (syn0.ndim() != 2 || syn1.ndim() != 2)
     throw std::runtime_error("syn0/syn1 must be 2-D float32");
   int64_t dim = syn0.shape(1);
@@ -517,7 +527,8 @@ static py::dict train_batch(
       std::vector<int32_t> kept; std::vector<float> cr(dim), gr(dim);
       train_sentences(s0, s1, dim, tok, off, 0, num_sent, keep_prob, tab,
                       tab_size, alpha, window, n_neg, seed, sent_id_base,
-                      ref_window, &total, kept, cr, gr, et, etn);
+                      ref_window, shared_negatives != 0, &total, kept, cr,
+                      gr, et, etn);
     } else {
       std::vector<std::thread> threads;
       std::vector<Stats> st(num_threads);
@@ -528,8 +539,9 @@ static py::dict train_batch(
         threads.emplace_back([=, &st]() {
           std::vector<int32_t> kept; std::vector<float> cr(dim), gr(dim);
           train_sentences(s0, s1, dim, tok, off, b, e, keep_prob, tab,
-                          tab_size, alpha, window, n_neg, seed, sent_id_base,
-                          ref_window, &st[t], kept, cr, gr, et, etn);
+                          tab_size, alpha, window, n_neg, seed,
+                          sent_id_base, ref_window, shared_negatives != 0,
+                          &st[t], kept, cr, gr, et, etn);
         });
       }
       for (auto& th : threads) th.join();
@@ -769,22 +781,26 @@ PYBIND11_MODULE(_cpu_native, m) {
         py::arg("alpha"), py::arg("window"), py::arg("n_neg"),
         py::arg("seed"), py::arg("sent_id_base") = 0,
         py::arg("window_mode") = "canonical", py::arg("num_threads") = 1,
-        py::arg("exp_table") = py::none());
+        py::arg("exp_table") = py::none(),
+        py::arg("shared_negatives") = 0);
   m.def("count_pairs", &count_pairs, py::arg("tokens"), py::arg("offsets"),
         py::arg("keep_prob"), py::arg("table"), py::arg("window"),
         py::arg("n_neg"), py::arg("seed"), py::arg("sent_id_base") = 0,
-        py::arg("window_mode") = "canonical");
+        py::arg("window_mode") = "canonical",
+        py::arg("shared_negatives") = 0);
   m.def("dots_slice", &dots_slice, py::arg("syn0"), py::arg("syn1"),
         py::arg("tokens"), py::arg("offsets"), py::arg("keep_prob"),
         py::arg("table"), py::arg("window"), py::arg("n_neg"), py::arg("seed"),
         py::arg("sent_id_base"), py::arg("window_mode"),
-        py::arg("pair_offsets"), py::arg("f_out"));
+        py::arg("pair_offsets"), py::arg("f_out"),
+        py::arg("shared_negatives") = 0);
   m.def("update_slice", &update_slice, py::arg("syn0"), py::arg("syn1"),
         py::arg("tokens"), py::arg("offsets"), py::arg("keep_prob"),
         py::arg("table"), py::arg("alpha"), py::arg("window"), py::arg("n_neg"),
         py::arg("seed"), py::arg("sent_id_base"), py::arg("window_mode"),
         py::arg("pair_offsets"), py::arg("f_in"),
-        py::arg("f_loc") = py::none(), py::arg("world_scale") = 0.0f);
+        py::arg("f_loc") = py::none(), py::arg("world_scale") = 0.0f,
+        py::arg("shared_negatives") = 0);
   m.def("train_pairs", &train_pairs, py::arg("cache0"), py::arg("cache1"),
         py::arg("group_center"), py::arg("group_offsets"),
         py::arg("pair_target"), py::arg("pair_label"), py::arg("alpha"));

@@ -596,6 +596,11 @@ struct KernelArgs {
   // centers/contexts anyway; hogwild on them costs no measurable quality
   // (benchmarks/results.md round-2 sweep).
   int32_t atomic_floor;
+  // HogBatch-style shared negatives (rng.py layout: k = NEG_BASE + i*n + s,
+  // one draw set per position reused across its contexts) — opt-in,
+  // config.shared_negatives.  Cuts per-position target-row traffic from
+  // ~2b*(1+n) to ~2b+n rows (the contexts' negative rows stay LLC-hot).
+  int shared_neg;
   const float* exp_table;     // non-null: reference LUT sigmoid parity mode
   int exp_table_size;
   int width;                  // valid elements per row (masked dim phases)
@@ -628,9 +633,9 @@ template <typename Phase>
 __device__ __forceinline__ void walk_sentence_dev(
     const int32_t* __restrict__ tokens, int64_t off, int len, uint64_t base,
     const uint32_t* __restrict__ keep_thr, const int32_t* __restrict__ table,
-    uint32_t table_size, int window, int n_neg, int ref_window, int lane,
-    int32_t* sent_lds, uint32_t* tgt_lds, Phase& ph, int pos_lo = 0,
-    int pos_hi = 1 << 30) {
+    uint32_t table_size, int window, int n_neg, int ref_window,
+    int shared_neg, int lane, int32_t* sent_lds, uint32_t* tgt_lds,
+    Phase& ph, int pos_lo = 0, int pos_hi = 1 << 30) {
   // ---- subsample + wave compaction into LDS -----------------------------
   int L = 0;
   for (int p0 = 0; p0 < len; p0 += 64) {
@@ -690,9 +695,11 @@ __device__ __forceinline__ void walk_sentence_dev(
             int j = lo + ctx_i;
             if (j >= i) ++j;
             const int32_t t = sent_lds[j];
-            const uint64_t kbase = kNegBase +
-                (uint64_t)(i * (2 * window + 1) + (j - i + window)) *
-                    (uint64_t)n_neg;
+            const uint64_t kbase = shared_neg
+                ? kNegBase + (uint64_t)i * (uint64_t)n_neg
+                : kNegBase +
+                  (uint64_t)(i * (2 * window + 1) + (j - i + window)) *
+                      (uint64_t)n_neg;
             const uint32_t un = draw_u32(base, kbase + (uint64_t)(s_in - 1));
             valid = (table[un % table_size] != t);
           }
@@ -719,9 +726,11 @@ __device__ __forceinline__ void walk_sentence_dev(
           enc = (uint32_t)t | 0x80000000u;     // positive: bit 31 set
           valid = true;
         } else {
-          const uint64_t kbase = kNegBase +
-              (uint64_t)(i * (2 * window + 1) + (j - i + window)) *
-                  (uint64_t)n_neg;
+          const uint64_t kbase = shared_neg
+              ? kNegBase + (uint64_t)i * (uint64_t)n_neg
+              : kNegBase +
+                (uint64_t)(i * (2 * window + 1) + (j - i + window)) *
+                    (uint64_t)n_neg;
           const uint32_t un = draw_u32(base, kbase + (uint64_t)(s_in - 1));
           const int32_t neg = table[un % table_size];
           valid = (neg != t);                  // discard colliding negative
@@ -907,7 +916,8 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void sgns_train_kernel(KernelA
     walk_sentence_dev(a.tokens, a.offsets[s],
                       (int)(a.offsets[s + 1] - a.offsets[s]), base, a.keep_thr,
                       a.table, a.table_size, a.window, a.n_neg, a.ref_window,
-                      lane, sbuf[wave], tbuf[wave], ph, pos_lo, pos_hi);
+                      a.shared_neg, lane, sbuf[wave], tbuf[wave], ph, pos_lo,
+                      pos_hi);
   }
 
   // ---- stats (device-scope atomics, once per wave) -----------------------
@@ -1044,7 +1054,8 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void sgns_train2_kernel(Kernel
     walk_sentence_dev(a.tokens, a.offsets[s],
                       (int)(a.offsets[s + 1] - a.offsets[s]), base, a.keep_thr,
                       a.table, a.table_size, a.window, a.n_neg, a.ref_window,
-                      lane, sbuf[wave], tbuf[wave], ph, pos_lo, pos_hi);
+                      a.shared_neg, lane, sbuf[wave], tbuf[wave], ph, pos_lo,
+                      pos_hi);
   }
   // combine half stats then one atomic from lane 0
   unsigned int p2 = ph.w_pairs + __shfl_xor(ph.w_pairs, 32, 64);
@@ -1174,7 +1185,8 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void sgns_train4_kernel(Kernel
     walk_sentence_dev(a.tokens, a.offsets[s],
                       (int)(a.offsets[s + 1] - a.offsets[s]), base, a.keep_thr,
                       a.table, a.table_size, a.window, a.n_neg, a.ref_window,
-                      lane, sbuf[wave], tbuf[wave], ph, pos_lo, pos_hi);
+                      a.shared_neg, lane, sbuf[wave], tbuf[wave], ph, pos_lo,
+                      pos_hi);
   }
   unsigned int p2 = ph.w_pairs, o2 = ph.w_pos;
   float f2 = ph.w_fplus;
@@ -1257,7 +1269,7 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void plan_emit_kernel(
     walk_sentence_dev(a.tokens, a.offsets[s],
                       (int)(a.offsets[s + 1] - a.offsets[s]), base, a.keep_thr,
                       a.table, a.table_size, a.window, a.n_neg, a.ref_window,
-                      lane, sbuf[wave], tbuf[wave], ph);
+                      a.shared_neg, lane, sbuf[wave], tbuf[wave], ph);
   }
 }
 
@@ -1276,7 +1288,7 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void count_pairs_kernel(
     walk_sentence_dev(a.tokens, a.offsets[s],
                       (int)(a.offsets[s + 1] - a.offsets[s]), base, a.keep_thr,
                       a.table, a.table_size, a.window, a.n_neg, a.ref_window,
-                      lane, sbuf[wave], tbuf[wave], ph);
+                      a.shared_neg, lane, sbuf[wave], tbuf[wave], ph);
     if (lane == 0) counts[s] = ph.pairs;
   }
 }
@@ -1343,7 +1355,7 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void dots_slice_kernel(
     walk_sentence_dev(a.tokens, a.offsets[s],
                       (int)(a.offsets[s + 1] - a.offsets[s]), base, a.keep_thr,
                       a.table, a.table_size, a.window, a.n_neg, a.ref_window,
-                      lane, sbuf[wave], tbuf[wave], ph);
+                      a.shared_neg, lane, sbuf[wave], tbuf[wave], ph);
   }
 }
 
@@ -1493,7 +1505,7 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void update_slice_kernel(
     walk_sentence_dev(a.tokens, a.offsets[s],
                       (int)(a.offsets[s + 1] - a.offsets[s]), base, a.keep_thr,
                       a.table, a.table_size, a.window, a.n_neg, a.ref_window,
-                      lane, sbuf[wave], tbuf[wave], ph);
+                      a.shared_neg, lane, sbuf[wave], tbuf[wave], ph);
   }
   if (lane == 0 && a.d_pairs) {
     atomicAdd(a.d_pairs, ph.w_pairs);
@@ -1561,7 +1573,7 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void dots_slice2_kernel(
     walk_sentence_dev(a.tokens, a.offsets[s],
                       (int)(a.offsets[s + 1] - a.offsets[s]), base, a.keep_thr,
                       a.table, a.table_size, a.window, a.n_neg, a.ref_window,
-                      lane, sbuf[wave], tbuf[wave], ph);
+                      a.shared_neg, lane, sbuf[wave], tbuf[wave], ph);
   }
 }
 
@@ -1686,7 +1698,7 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void update_slice2_kernel(
     walk_sentence_dev(a.tokens, a.offsets[s],
                       (int)(a.offsets[s + 1] - a.offsets[s]), base, a.keep_thr,
                       a.table, a.table_size, a.window, a.n_neg, a.ref_window,
-                      lane, sbuf[wave], tbuf[wave], ph);
+                      a.shared_neg, lane, sbuf[wave], tbuf[wave], ph);
   }
   unsigned int p2 = ph.w_pairs + __shfl_xor(ph.w_pairs, 32, 64);
   unsigned int o2 = ph.w_pos + __shfl_xor(ph.w_pos, 32, 64);
@@ -2094,7 +2106,7 @@ static void sgns_train(uintptr_t syn0, uintptr_t syn1, int is_bf16,
                        uintptr_t stats, int blocks, int pos_blocks,
                        int threads, uintptr_t stream_ptr,
                        uintptr_t exp_table, int exp_table_size,
-                       int pair2, int64_t atomic_floor) {
+                       int pair2, int64_t atomic_floor, int shared_neg) {
   HIP_CLEAR_ERROR();
   if (threads != 64 && threads != 256)
     throw std::runtime_error("threads must be 64 (serial) or 256");
@@ -2121,6 +2133,7 @@ static void sgns_train(uintptr_t syn0, uintptr_t syn1, int is_bf16,
   a.ref_window = ref_window;
   a.atomic_below = (int32_t)std::min<int64_t>(atomic_below, 0x7FFFFFFFLL);
   a.atomic_floor = (int32_t)std::min<int64_t>(atomic_floor, 0x7FFFFFFFLL);
+  a.shared_neg = shared_neg;
   a.exp_table = (const float*)exp_table;
   a.exp_table_size = exp_table_size;
   unsigned long long* st = (unsigned long long*)stats;
@@ -2200,11 +2213,13 @@ static void count_pairs(uintptr_t tokens, uintptr_t offsets,
                         uintptr_t table, int64_t table_size, int window,
                         int n_neg, uint64_t seed, int64_t sent_id_base,
                         int ref_window, uintptr_t counts_out, int blocks,
-                        int threads, uintptr_t stream_ptr) {
+                        int threads, uintptr_t stream_ptr,
+                        int shared_neg) {
   HIP_CLEAR_ERROR();
   KernelArgs a = make_walk_args(0, 0, 64, tokens, offsets, num_sentences,
                                 keep_thr, table, table_size, 0.0, window,
                                 n_neg, seed, sent_id_base, ref_window, 0);
+  a.shared_neg = shared_neg;
   hipLaunchKernelGGL(count_pairs_kernel, dim3(blocks), dim3(threads), 0,
                      (hipStream_t)stream_ptr, a, (int64_t*)counts_out);
   HIP_CHECK(hipGetLastError());
@@ -2217,11 +2232,12 @@ static void plan_emit(uintptr_t tokens, uintptr_t offsets,
                       int ref_window, uintptr_t pair_offsets,
                       uintptr_t out_target, uintptr_t out_label,
                       uintptr_t out_start, uintptr_t out_center, int blocks,
-                      int threads, uintptr_t stream_ptr) {
+                      int threads, uintptr_t stream_ptr, int shared_neg) {
   HIP_CLEAR_ERROR();
   KernelArgs a = make_walk_args(0, 0, 64, tokens, offsets, num_sentences,
                                 keep_thr, table, table_size, 0.0, window,
                                 n_neg, seed, sent_id_base, ref_window, 0);
+  a.shared_neg = shared_neg;
   hipLaunchKernelGGL(plan_emit_kernel, dim3(blocks), dim3(threads), 0,
                      (hipStream_t)stream_ptr, a,
                      (const int64_t*)pair_offsets, (int32_t*)out_target,
@@ -2237,7 +2253,7 @@ static void dots_slice(uintptr_t syn0, uintptr_t syn1, int is_bf16,
                        int n_neg, uint64_t seed, int64_t sent_id_base,
                        int ref_window, uintptr_t pair_offsets, uintptr_t f_out,
                        int blocks, int threads, uintptr_t stream_ptr,
-                       int pair_mode, int width) {
+                       int pair_mode, int width, int shared_neg) {
   HIP_CLEAR_ERROR();
   if (width <= 0) width = (int)stride;
   KernelArgs a = make_walk_args(syn0, syn1, stride, tokens, offsets,
@@ -2245,6 +2261,7 @@ static void dots_slice(uintptr_t syn0, uintptr_t syn1, int is_bf16,
                                 0.0, window, n_neg, seed, sent_id_base,
                                 ref_window, 0);
   a.width = width;
+  a.shared_neg = shared_neg;
   const int nc = supported_nc((width + 63) / 64);
   if (width != (int)stride && !(pair_mode && threads == 256))
     throw std::runtime_error("masked width requires pair_mode kernels");
@@ -2292,7 +2309,8 @@ static void update_slice(uintptr_t syn0, uintptr_t syn1, int is_bf16,
                          uintptr_t stats, int blocks, int threads,
                          uintptr_t stream_ptr,
                          uintptr_t exp_table, int exp_table_size,
-                         int pair_mode, int width, int64_t atomic_floor) {
+                         int pair_mode, int width, int64_t atomic_floor,
+                         int shared_neg) {
   HIP_CLEAR_ERROR();
   if (width <= 0) width = (int)stride;
   KernelArgs a = make_walk_args(syn0, syn1, stride, tokens, offsets,
@@ -2300,6 +2318,7 @@ static void update_slice(uintptr_t syn0, uintptr_t syn1, int is_bf16,
                                 alpha, window, n_neg, seed, sent_id_base,
                                 ref_window, stats);
   a.width = width;
+  a.shared_neg = shared_neg;
   a.atomic_below = (int32_t)std::min<int64_t>(atomic_below, 0x7FFFFFFFLL);
   a.atomic_floor = (int32_t)std::min<int64_t>(atomic_floor, 0x7FFFFFFFLL);
   a.exp_table = (const float*)exp_table;
@@ -2615,7 +2634,8 @@ PYBIND11_MODULE(_hip_native, m) {
         py::arg("stats"), py::arg("blocks"), py::arg("pos_blocks"),
         py::arg("threads"), py::arg("stream"),
         py::arg("exp_table") = 0, py::arg("exp_table_size") = 0,
-        py::arg("pair2") = 0, py::arg("atomic_floor") = 0);
+        py::arg("pair2") = 0, py::arg("atomic_floor") = 0,
+        py::arg("shared_neg") = 0);
   m.def("count_pairs", &count_pairs);
   m.def("dots_slice", &dots_slice);
   m.def("update_slice", &update_slice);

@@ -150,6 +150,14 @@ class Word2VecConfig:
     # context (mllib:385-387).  Default is the canonical symmetric shrunk
     # window; "reference" reproduces B2.
     window_mode: str = "canonical"   # "canonical" | "reference"
+    # HogBatch-style shared negative sampling (OPT-IN; changes the draw
+    # layout, rng.py): one negative set per center position, reused
+    # across its contexts.  Cuts per-position target-row traffic from
+    # ~2b*(1+n) to ~2b+n rows — the only route past the measured HBM
+    # floor at dim=1024 neg=25 (profiles/round2_kernel_stats.md);
+    # quality measured on the planted-synonym probe
+    # (benchmarks/results.md).  Default off: reference draw semantics.
+    shared_negatives: bool = False
     # sigmoid: "exact" clipped sigmoid, or "lut" = the reference's
     # EXP_TABLE_SIZE-entry lookup (createExpTable/getSigmoid, mllib:281-302)
     sigmoid_mode: str = "exact"      # "exact" | "lut"

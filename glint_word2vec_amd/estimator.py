@@ -248,7 +248,8 @@ class GlintWord2Vec:
                     st = _cpu_native.train_batch(
                         syn0, syn1, batch.tokens, batch.offsets, kp, table,
                         alpha, cfg.window, cfg.n, seed, sent_base,
-                        cfg.window_mode, cfg.num_partitions, exp_table)
+                        cfg.window_mode, cfg.num_partitions, exp_table,
+                        int(cfg.shared_negatives))
                     npos, sum_fp = st["positives"], st["sum_fplus"]
                 else:
                     plan = sgns.make_plan(batch.tokens, batch.offsets, kp,

@@ -68,7 +68,8 @@ def train_batch_oracle(syn0: np.ndarray, syn1: np.ndarray,
                        alpha: float, window: int, n_neg: int,
                        seed: int, sent_id_base: int = 0,
                        window_mode: str = "canonical",
-                       exp_table: np.ndarray | None = None) -> TrainStats:
+                       exp_table: np.ndarray | None = None,
+                       shared_negatives: bool = False) -> TrainStats:
     """In-place SGNS update over one batch.  syn0/syn1 float32 [vocab, dim].
 
     ``keep_prob`` None => subsampling off (no RNG draws for it).
@@ -121,7 +122,9 @@ def train_batch_oracle(syn0: np.ndarray, syn1: np.ndarray,
                 stats.positives += 1
                 stats.sum_fplus += f
                 # negatives
-                kbase = NEG_BASE + (i * (2 * window + 1) + (j - i + window)) * n_neg
+                kbase = (NEG_BASE + i * n_neg if shared_negatives else
+                         NEG_BASE + (i * (2 * window + 1)
+                                     + (j - i + window)) * n_neg)
                 for k in range(n_neg):
                     u = draw_u32(base, kbase + k)
                     neg = int(table[u % table_size])

@@ -123,7 +123,7 @@ class GpuSgns:
                     alpha: float, window: int, n_neg: int, seed: int,
                     sent_id_base: int = 0, window_mode: str = "canonical",
                     atomic: bool = True, atomic_below: Optional[int] = None,
-                    atomic_floor: int = 0,
+                    atomic_floor: int = 0, shared_negatives: bool = False,
                     blocks: Optional[int] = None,
                     serial: bool = False,
                     pair_mode: Optional[int] = None,
@@ -173,7 +173,8 @@ class GpuSgns:
             s.cuda_stream,
             0 if self.exp_table is None else self.exp_table.data_ptr(),
             0 if self.exp_table is None else int(self.exp_table.numel()),
-            0 if serial else int(pair_mode), int(atomic_floor))
+            0 if serial else int(pair_mode), int(atomic_floor),
+            int(shared_negatives))
 
     def read_stats(self, reset: bool = True) -> GpuStats:
         h = self._stats.cpu()

@@ -44,6 +44,7 @@ class DimShardedSgns:
                  window_mode: str = "canonical", chunk_words: int = 1 << 19,
                  f_correction: bool = True, atomic: bool = True,
                  atomic_below: "int | None" = None, atomic_floor: int = 0,
+                 shared_negatives: bool = False,
                  narrow: "bool | None" = None,
                  init_full_limit: int = 1 << 28):
         self.rank, self.world = (comm.init_from_env() if torch.distributed.is_available()
@@ -62,6 +63,7 @@ class DimShardedSgns:
         self.atomic_below = ((2 ** 31 - 1 if atomic else 0)
                              if atomic_below is None else int(atomic_below))
         self.atomic_floor = int(atomic_floor)
+        self.shared_neg = int(shared_negatives)
         self.lo, self.hi = slice_bounds(dim, self.rank, self.world)
         self.width = self.hi - self.lo
 
@@ -204,7 +206,8 @@ class DimShardedSgns:
             int(sent_id_base), int(self.window_mode == "reference"),
             self.atomic_below, self._stats.data_ptr(), nblocks, pos_blocks,
             nthreads, stream.cuda_stream, 0, 0,
-            1 if self.stride <= 512 else 0, self.atomic_floor)
+            1 if self.stride <= 512 else 0, self.atomic_floor,
+            self.shared_neg)
 
     def _train_step_gpu(self, tokens, offsets, chunks, alpha, window, n_neg,
                         seed, sent_id_base):
@@ -229,7 +232,7 @@ class DimShardedSgns:
             tokens.data_ptr(), offsets.data_ptr(), num_sent, kthr,
             self.table.data_ptr(), int(self.table.numel()), window, n_neg,
             seed, sent_id_base, ref, counts.data_ptr(), nb_all, nt,
-            comp.cuda_stream)
+            comp.cuda_stream, self.shared_neg)
         poff = torch.zeros(num_sent + 1, dtype=torch.int64, device=device)
         torch.cumsum(counts, 0, out=poff[1:])
         poff_host = poff.cpu().numpy()          # one sync per step
@@ -255,7 +258,8 @@ class DimShardedSgns:
                 self.atomic_below, self._stats.data_ptr(),
                 nb, nt, comp.cuda_stream, 0, 0,
                 0 if self.serial else 1,
-                self.width if self.narrow else 0, self.atomic_floor)
+                self.width if self.narrow else 0, self.atomic_floor,
+                self.shared_neg)
 
         for (a, b) in chunks:
             n = b - a
@@ -270,7 +274,7 @@ class DimShardedSgns:
                 window, n_neg, seed, sent_id_base + a, ref,
                 poff[a:b + 1].data_ptr(), f.data_ptr(), nb, nt,
                 comp.cuda_stream, 0 if self.serial else 1,
-                self.width if self.narrow else 0)
+                self.width if self.narrow else 0, self.shared_neg)
             if f_loc is not None:
                 f_loc[lo:hi] = f[lo:hi]
             ev = torch.cuda.Event()
@@ -298,9 +302,10 @@ class DimShardedSgns:
             off_np = off_view.numpy()
             tab_np = self.table.numpy()
             wm = self.window_mode
-            cnts = self.native.count_pairs(tok_np, off_np, self.keep_prob,
-                                           tab_np, window, n_neg,
-                                           seed & 0xFFFFFFFFFFFFFFFF, base, wm)
+            cnts = self.native.count_pairs(
+                tok_np, off_np, self.keep_prob, tab_np, window, n_neg,
+                seed & 0xFFFFFFFFFFFFFFFF, base, wm,
+                shared_negatives=self.shared_neg)
             poff = np.zeros(n + 1, dtype=np.int64)
             np.cumsum(cnts, out=poff[1:])
             total = int(poff[-1])
@@ -312,7 +317,8 @@ class DimShardedSgns:
             self.native.dots_slice(s0_np, s1_np, tok_np, off_np,
                                    self.keep_prob, tab_np, window, n_neg,
                                    seed & 0xFFFFFFFFFFFFFFFF, base, wm, poff,
-                                   f.numpy())
+                                   f.numpy(),
+                                   shared_negatives=self.shared_neg)
             f_loc = f.numpy().copy() if self.f_correction else None
             comm.all_reduce_sum(f)
             st = self.native.update_slice(s0_np, s1_np, tok_np, off_np,
@@ -321,7 +327,8 @@ class DimShardedSgns:
                                           seed & 0xFFFFFFFFFFFFFFFF, base, wm,
                                           poff, f.numpy(), f_loc,
                                           float(self.dim) /
-                                          max(self.width, 1))
+                                          max(self.width, 1),
+                                          shared_negatives=self.shared_neg)
             for k in ("pairs", "positives", "words_trained"):
                 self._cpu_stats[k] += st[k]
             self._cpu_stats["sum_fplus"] += st["sum_fplus"]

@@ -102,7 +102,9 @@ def train_gpu(cfg: Word2VecConfig, vocab: Vocabulary, batches_fn: Callable,
                            sent_id_base=sent_base, window_mode=cfg.window_mode,
                            atomic=cfg.resolved_update_mode() != "hogwild",
                            atomic_below=cfg.effective_atomic_below(),
-                           atomic_floor=cfg.effective_atomic_floor(vocab.num_words))
+                           atomic_floor=cfg.effective_atomic_floor(
+                               vocab.num_words),
+                           shared_negatives=cfg.shared_negatives)
             sent_base += batch.num_sentences
             processed += batch.num_tokens
             step += 1
@@ -144,7 +146,8 @@ def _train_sharded(cfg: Word2VecConfig, vocab: Vocabulary,
     common = dict(dtype=dtype,
                   device=device, seed=seed, counts=vocab.counts,
                   table_size=cfg.unigram_table_size, subsample=subsample,
-                  window_mode=cfg.window_mode)
+                  window_mode=cfg.window_mode,
+                  shared_negatives=cfg.shared_negatives)
     total_words = vocab.train_words_count * cfg.num_iterations
     processed = 0
     sent_base = 0

@@ -35,7 +35,7 @@ class ReplicatedSgns:
                  table_size: int = 1_000_000, subsample: float = 0.0,
                  window_mode: str = "canonical", sync_every: int = 4,
                  atomic: bool = False, atomic_below: "int | None" = None,
-                 atomic_floor: int = 0):
+                 atomic_floor: int = 0, shared_negatives: bool = False):
         self.rank, self.world = comm.init_from_env()
         self.vocab_size = vocab_size
         self.dim = dim
@@ -47,6 +47,7 @@ class ReplicatedSgns:
         self.atomic_below = ((2 ** 31 - 1 if atomic else 0)
                              if atomic_below is None else int(atomic_below))
         self.atomic_floor = int(atomic_floor)
+        self.shared_neg = int(shared_negatives)
         self._steps_since_sync = 0
         counts = (np.ones(vocab_size, dtype=np.int64) if counts is None
                   else counts)
@@ -89,14 +90,16 @@ class ReplicatedSgns:
                                 atomic_below=(None if self.atomic_below
                                               >= 2 ** 31 - 1
                                               else self.atomic_below),
-                                atomic_floor=self.atomic_floor)
+                                atomic_floor=self.atomic_floor,
+                                shared_negatives=bool(self.shared_neg))
         else:
             st = self.native.train_batch(
                 self.syn0.numpy(), self.syn1.numpy(),
                 np.ascontiguousarray(tokens), np.ascontiguousarray(offsets),
                 self.keep_prob, self.table, float(alpha), int(window),
                 int(n_neg), seed & 0xFFFFFFFFFFFFFFFF, int(sent_id_base),
-                self.window_mode, 1)
+                self.window_mode, 1,
+                shared_negatives=self.shared_neg)
             for k in ("pairs", "positives", "words_trained"):
                 self._cpu_stats[k] += st[k]
             self._cpu_stats["sum_fplus"] += st["sum_fplus"]

@@ -47,6 +47,7 @@ class RowShardedSgns:
                  table_size: int = 1_000_000, subsample: float = 0.0,
                  window_mode: str = "canonical", atomic: bool = True,
                  atomic_below: "int | None" = None, atomic_floor: int = 0,
+                 shared_negatives: bool = False,
                  init_full_limit: int = 1 << 28):
         self.rank, self.world = comm.init_from_env()
         self.vocab_size = vocab_size
@@ -66,6 +67,7 @@ class RowShardedSgns:
         self.atomic_below = ((2 ** 31 - 1 if atomic else 0)
                              if atomic_below is None else int(atomic_below))
         self.atomic_floor = int(atomic_floor)
+        self.shared_neg = int(shared_negatives)
         if self.is_cuda:
             from .. import _hip_native
             self.native = _hip_native
@@ -230,6 +232,10 @@ class RowShardedSgns:
         state, so it can be prefetched on a worker thread while the GPU
         trains the previous step (the reference's async mini-batch workers
         overlapping compute, SURVEY §2.2 pipeline note)."""
+        if self.shared_neg:
+            raise NotImplementedError(
+                "shared_negatives needs the counter-RNG planner "
+                "(make_plan_counter, CUDA) or the fused world-1 path")
         return sgns.make_grouped_plan(tokens, offsets, self.keep_prob,
                                       self.table, window, n_neg, rng,
                                       self.window_mode)
@@ -296,7 +302,7 @@ class RowShardedSgns:
             tok.data_ptr(), off.data_ptr(), num_sent, kthr,
             self._table_t.data_ptr(), int(self._table_t.numel()), window,
             n_neg, seed & 0xFFFFFFFFFFFFFFFF, sent_id_base, ref,
-            counts.data_ptr(), nb, nt, stream.cuda_stream)
+            counts.data_ptr(), nb, nt, stream.cuda_stream, self.shared_neg)
         poff = torch.zeros(num_sent + 1, dtype=torch.int64,
                            device=self.device)
         torch.cumsum(counts, 0, out=poff[1:])
@@ -315,7 +321,8 @@ class RowShardedSgns:
             self._table_t.data_ptr(), int(self._table_t.numel()), window,
             n_neg, seed & 0xFFFFFFFFFFFFFFFF, sent_id_base, ref,
             poff.data_ptr(), target.data_ptr(), label.data_ptr(),
-            start.data_ptr(), center.data_ptr(), nb, nt, stream.cuda_stream)
+            start.data_ptr(), center.data_ptr(), nb, nt, stream.cuda_stream,
+            self.shared_neg)
         starts = start.nonzero().reshape(-1)
         group_offsets = torch.cat(
             [starts, torch.tensor([total], dtype=torch.int64,
@@ -643,7 +650,8 @@ class RowShardedSgns:
             int(self.window_mode == "reference"), self.atomic_below,
             self._stats.data_ptr(), nblocks, pos_blocks, 256,
             stream.cuda_stream, 0, 0,
-            1 if self.store_stride <= 512 else 0, self.atomic_floor)
+            1 if self.store_stride <= 512 else 0, self.atomic_floor,
+            self.shared_neg)
 
     def train_step(self, tokens: np.ndarray, offsets: np.ndarray,
                    alpha: float, window: int, n_neg: int,

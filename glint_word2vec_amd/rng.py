@@ -18,6 +18,14 @@ Draw-index layout per sentence (normative — every implementation must match):
         k = NEG_BASE + (i * (2*window+1) + (j - i + window)) * n + s
     a draw whose table entry equals the positive target is discarded (that
     negative slot is skipped, as in canonical word2vec.c).
+  * SHARED-NEGATIVES mode (config.shared_negatives=True, opt-in — the
+    HogBatch-style trick that reuses one negative set across all contexts
+    of a position, cutting target-row traffic per position from
+    ~2b*(1+n) to ~2b+n rows):
+        k = NEG_BASE + i * n + s          (independent of j)
+    The same collision-discard rule applies per context (the shared draw
+    is discarded only against that context's positive target), so pair
+    counts can differ between contexts.
 
 Sentence ids must stay < 2^63.  The three draw-index streams never
 collide for any window/negative count: subsample uses k < 1024, window

@@ -801,3 +801,44 @@ def test_row_engine_fused_fast_path_world1():
     assert st_fused.positives == st_plan.positives
     s0 = eng.to_host()[0]
     assert np.isfinite(s0).all() and not np.array_equal(before, s0)
+
+
+def test_shared_negatives_serial_matches_oracle():
+    """Shared-negative mode on the GPU walker: serial parity vs the Python
+    oracle (exact pair counts + close values)."""
+    tokens, offsets, counts, table, syn0, syn1 = _problem()
+    a0, a1 = syn0.copy(), syn1.copy()
+    st_py = cpu_ref.train_batch_oracle(a0, a1, tokens, offsets, None, table,
+                                       0.03, 3, 4, seed=77, sent_id_base=5,
+                                       shared_negatives=True)
+    gs = _gpu_setup(syn0, syn1, table)
+    gs.train_batch(_to_dev(tokens), _to_dev(offsets), 0.03, 3, 4, 77,
+                   sent_id_base=5, serial=True, atomic=False,
+                   shared_negatives=True)
+    torch.cuda.synchronize()
+    st = gs.read_stats()
+    assert st.pairs == st_py.pairs
+    assert st.positives == st_py.positives
+    g0, g1 = gs.to_host()
+    np.testing.assert_allclose(g0, a0, rtol=2e-4, atol=2e-6)
+    np.testing.assert_allclose(g1, a1, rtol=2e-4, atol=2e-6)
+
+
+def test_shared_negatives_row_planner_consistency():
+    """Counter planner with shared negatives: the plan's pair counts match
+    the fused kernel's own walk (same flag through count/emit)."""
+    from glint_word2vec_amd.parallel.row_sharded import RowShardedSgns
+    from glint_word2vec_amd.data import synthetic_corpus
+    batch = synthetic_corpus(vocab_size=500, num_tokens=5000,
+                             sentence_len=50, seed=5, zipf_a=1.01)
+    counts = np.bincount(batch.tokens, minlength=500).astype(np.int64) + 1
+    tok = torch.from_numpy(batch.tokens).cuda()
+    off = torch.from_numpy(batch.offsets).cuda()
+    eng = RowShardedSgns(500, 48, dtype="bfloat16", device="cuda", seed=3,
+                         counts=counts, table_size=1009, atomic=False,
+                         shared_negatives=True)
+    plan = eng.make_plan_counter(tok, off, 3, 4, seed=9)
+    eng.train_batch_fused(tok, off, 0.03, 3, 4, seed=9)
+    torch.cuda.synchronize()
+    st = eng.read_stats()
+    assert st.pairs == plan.num_pairs > 0

@@ -73,3 +73,38 @@ def test_cpp_multithread_stats_close():
                              table, 0.03, 3, 4, 5, 0, "canonical", 4)
     assert st1["pairs"] == st4["pairs"]
     assert st1["positives"] == st4["positives"]
+
+
+def test_shared_negatives_native_matches_oracle():
+    """Shared-negative draw layout (rng.py): C++ trainer vs Python oracle,
+    and distinct from the per-context layout."""
+    import numpy as np
+    from glint_word2vec_amd import _cpu_native
+    from glint_word2vec_amd.models import sgns
+    from glint_word2vec_amd.ops import cpu_ref
+    from glint_word2vec_amd.vocab import build_unigram_table
+    rng = np.random.default_rng(3)
+    vocab, dim = 80, 16
+    tokens = rng.integers(0, vocab, 300).astype(np.int32)
+    offsets = np.array([0, 100, 220, 300], dtype=np.int32)
+    counts = np.bincount(tokens, minlength=vocab).astype(np.int64) + 1
+    table = build_unigram_table(counts, 1009)
+    syn0, syn1 = sgns.init_tables(vocab, dim, 5)
+
+    a0, a1 = syn0.copy(), syn1.copy()
+    st_py = cpu_ref.train_batch_oracle(a0, a1, tokens, offsets, None, table,
+                                       0.03, 3, 4, seed=7,
+                                       shared_negatives=True)
+    b0, b1 = syn0.copy(), syn1.copy()
+    st_c = _cpu_native.train_batch(b0, b1, tokens, offsets, None, table,
+                                   0.03, 3, 4, 7, 0, "canonical", 1, None, 1)
+    assert st_c["pairs"] == st_py.pairs
+    assert st_c["positives"] == st_py.positives
+    np.testing.assert_allclose(b0, a0, rtol=1e-5, atol=1e-7)
+    np.testing.assert_allclose(b1, a1, rtol=1e-5, atol=1e-7)
+
+    # differs from the per-context layout (different negatives drawn)
+    c0, c1 = syn0.copy(), syn1.copy()
+    cpu_ref.train_batch_oracle(c0, c1, tokens, offsets, None, table,
+                               0.03, 3, 4, seed=7, shared_negatives=False)
+    assert not np.allclose(c1, a1)
