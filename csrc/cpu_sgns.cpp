@@ -144,15 +144,34 @@ static void train_sentences(float* syn0, float* syn1, int64_t dim,
           }
           stats->pairs++; stats->positives++; stats->sum_fplus += f;
         }
-        uint64_t kbase = shared_neg
-            ? kNegBase + (uint64_t)i * (uint64_t)n_neg
-            : kNegBase +
-              (uint64_t)(i * (2 * window + 1) + (j - i + window)) *
-                  (uint64_t)n_neg;
+        if (shared_neg) continue;   // negatives once per position, below
+        uint64_t kbase = kNegBase +
+            (uint64_t)(i * (2 * window + 1) + (j - i + window)) *
+                (uint64_t)n_neg;
         for (int k = 0; k < n_neg; ++k) {
           uint32_t un = draw_u32(base, kbase + (uint64_t)k);
           int32_t neg = table[un % (uint64_t)table_size];
           if (neg == t) continue;
+          float* t1 = syn1 + (int64_t)neg * dim;
+          float f = 0.0f;
+          for (int64_t d = 0; d < dim; ++d) f += c_row[d] * t1[d];
+          float g = (0.0f - sigma_of(f, et, etn)) * alpha;
+          for (int64_t d = 0; d < dim; ++d) {
+            grad[d] += g * t1[d];
+            t1[d] += g * c_row[d];
+          }
+          stats->pairs++;
+        }
+      }
+      if (shared_neg) {
+        // shared mode (rng.py): ONE negative set per position, applied
+        // once, discarded when the draw equals the center word
+        for (int k = 0; k < n_neg; ++k) {
+          uint32_t un = draw_u32(base,
+                                 kNegBase + (uint64_t)i * (uint64_t)n_neg +
+                                     (uint64_t)k);
+          int32_t neg = table[un % (uint64_t)table_size];
+          if (neg == c) continue;
           float* t1 = syn1 + (int64_t)neg * dim;
           float f = 0.0f;
           for (int64_t d = 0; d < dim; ++d) f += c_row[d] * t1[d];
@@ -222,15 +241,23 @@ static void walk_sentence(const int32_t* sent, int64_t len, uint64_t base,
       if (j == i) continue;
       int32_t t = kept[j];
       ph.pair(c, t, 1.0f, pair_idx++);
-      uint64_t kbase = shared_neg
-          ? kNegBase + (uint64_t)i * (uint64_t)n_neg
-          : kNegBase +
-            (uint64_t)(i * (2 * window + 1) + (j - i + window)) *
-                (uint64_t)n_neg;
+      if (shared_neg) continue;   // negatives once per position, below
+      uint64_t kbase = kNegBase +
+          (uint64_t)(i * (2 * window + 1) + (j - i + window)) *
+              (uint64_t)n_neg;
       for (int k = 0; k < n_neg; ++k) {
         uint32_t un = draw_u32(base, kbase + (uint64_t)k);
         int32_t neg = table[un % (uint64_t)table_size];
         if (neg == t) continue;
+        ph.pair(c, neg, 0.0f, pair_idx++);
+      }
+    }
+    if (shared_neg) {
+      for (int k = 0; k < n_neg; ++k) {
+        uint32_t un = draw_u32(
+            base, kNegBase + (uint64_t)i * (uint64_t)n_neg + (uint64_t)k);
+        int32_t neg = table[un % (uint64_t)table_size];
+        if (neg == c) continue;
         ph.pair(c, neg, 0.0f, pair_idx++);
       }
     }

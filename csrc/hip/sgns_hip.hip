@@ -678,7 +678,11 @@ __device__ __forceinline__ void walk_sentence_dev(
     }
     if (!((lo < i) || (hi > i))) continue;
     const int n_ctx = (hi - lo + 1) - ((lo <= i && i <= hi) ? 1 : 0);
-    const int total_slots = n_ctx * per_ctx;
+    // shared mode: slots = [n_ctx positives][n_neg shared negatives] —
+    // ONE negative set per position (rng.py shared layout), vs the
+    // per-context n_ctx*(1+n_neg) enumeration
+    const int total_slots =
+        shared_neg ? (n_ctx + n_neg) : n_ctx * per_ctx;
     if (i < pos_lo || i >= pos_hi) {
       // outside this wave's position block: advance the pair numbering
       // without touching rows.  Negative-collision discards still need the
@@ -687,21 +691,31 @@ __device__ __forceinline__ void walk_sentence_dev(
         const int slot = chunk + lane;
         bool valid = false;
         if (slot < total_slots) {
-          const int ctx_i = slot / per_ctx;
-          const int s_in = slot - ctx_i * per_ctx;
-          if (s_in == 0) {
-            valid = true;
+          if (shared_neg) {
+            if (slot < n_ctx) {
+              valid = true;
+            } else {
+              const uint32_t un = draw_u32(
+                  base, kNegBase + (uint64_t)i * (uint64_t)n_neg +
+                            (uint64_t)(slot - n_ctx));
+              valid = (table[un % table_size] != c);
+            }
           } else {
-            int j = lo + ctx_i;
-            if (j >= i) ++j;
-            const int32_t t = sent_lds[j];
-            const uint64_t kbase = shared_neg
-                ? kNegBase + (uint64_t)i * (uint64_t)n_neg
-                : kNegBase +
+            const int ctx_i = slot / per_ctx;
+            const int s_in = slot - ctx_i * per_ctx;
+            if (s_in == 0) {
+              valid = true;
+            } else {
+              int j = lo + ctx_i;
+              if (j >= i) ++j;
+              const int32_t t = sent_lds[j];
+              const uint64_t kbase = kNegBase +
                   (uint64_t)(i * (2 * window + 1) + (j - i + window)) *
                       (uint64_t)n_neg;
-            const uint32_t un = draw_u32(base, kbase + (uint64_t)(s_in - 1));
-            valid = (table[un % table_size] != t);
+              const uint32_t un =
+                  draw_u32(base, kbase + (uint64_t)(s_in - 1));
+              valid = (table[un % table_size] != t);
+            }
           }
         }
         pair_idx += __popcll(__ballot(valid));
@@ -717,24 +731,38 @@ __device__ __forceinline__ void walk_sentence_dev(
       bool valid = false;
       uint32_t enc = 0;
       if (slot < total_slots) {
-        const int ctx_i = slot / per_ctx;
-        const int s_in = slot - ctx_i * per_ctx;
-        int j = lo + ctx_i;
-        if (j >= i) ++j;                       // skip the center position
-        const int32_t t = sent_lds[j];
-        if (s_in == 0) {
-          enc = (uint32_t)t | 0x80000000u;     // positive: bit 31 set
-          valid = true;
+        if (shared_neg) {
+          if (slot < n_ctx) {
+            int j = lo + slot;
+            if (j >= i) ++j;                   // skip the center position
+            enc = (uint32_t)sent_lds[j] | 0x80000000u;
+            valid = true;
+          } else {
+            const uint32_t un = draw_u32(
+                base, kNegBase + (uint64_t)i * (uint64_t)n_neg +
+                          (uint64_t)(slot - n_ctx));
+            const int32_t neg = table[un % table_size];
+            valid = (neg != c);                // discard center collision
+            enc = (uint32_t)neg;
+          }
         } else {
-          const uint64_t kbase = shared_neg
-              ? kNegBase + (uint64_t)i * (uint64_t)n_neg
-              : kNegBase +
+          const int ctx_i = slot / per_ctx;
+          const int s_in = slot - ctx_i * per_ctx;
+          int j = lo + ctx_i;
+          if (j >= i) ++j;                     // skip the center position
+          const int32_t t = sent_lds[j];
+          if (s_in == 0) {
+            enc = (uint32_t)t | 0x80000000u;   // positive: bit 31 set
+            valid = true;
+          } else {
+            const uint64_t kbase = kNegBase +
                 (uint64_t)(i * (2 * window + 1) + (j - i + window)) *
                     (uint64_t)n_neg;
-          const uint32_t un = draw_u32(base, kbase + (uint64_t)(s_in - 1));
-          const int32_t neg = table[un % table_size];
-          valid = (neg != t);                  // discard colliding negative
-          enc = (uint32_t)neg;
+            const uint32_t un = draw_u32(base, kbase + (uint64_t)(s_in - 1));
+            const int32_t neg = table[un % table_size];
+            valid = (neg != t);                // discard colliding negative
+            enc = (uint32_t)neg;
+          }
         }
       }
       const uint64_t m = __ballot(valid);

@@ -111,6 +111,32 @@ def train_batch_oracle(syn0: np.ndarray, syn1: np.ndarray,
                 continue
             c_row = syn0[c].copy()
             grad = np.zeros_like(c_row)
+            if shared_negatives:
+                # one negative set per POSITION (rng.py shared layout):
+                # positives for every context first, then n_neg negatives
+                # applied once, discarded when the draw equals the center
+                for j in ctx:
+                    t = kept[j]
+                    f = float(np.dot(c_row, syn1[t]))
+                    g = (1.0 - sig(f)) * alpha
+                    grad += g * syn1[t]
+                    syn1[t] += g * c_row
+                    stats.pairs += 1
+                    stats.positives += 1
+                    stats.sum_fplus += f
+                for k in range(n_neg):
+                    u = draw_u32(base, NEG_BASE + i * n_neg + k)
+                    neg = int(table[u % table_size])
+                    if neg == c:
+                        continue
+                    f = float(np.dot(c_row, syn1[neg]))
+                    g = (0.0 - sig(f)) * alpha
+                    grad += g * syn1[neg]
+                    syn1[neg] += g * c_row
+                    stats.pairs += 1
+                syn0[c] += grad
+                stats.words_trained += 1
+                continue
             for j in ctx:
                 t = kept[j]
                 # positive
@@ -122,9 +148,8 @@ def train_batch_oracle(syn0: np.ndarray, syn1: np.ndarray,
                 stats.positives += 1
                 stats.sum_fplus += f
                 # negatives
-                kbase = (NEG_BASE + i * n_neg if shared_negatives else
-                         NEG_BASE + (i * (2 * window + 1)
-                                     + (j - i + window)) * n_neg)
+                kbase = NEG_BASE + (i * (2 * window + 1)
+                                    + (j - i + window)) * n_neg
                 for k in range(n_neg):
                     u = draw_u32(base, kbase + k)
                     neg = int(table[u % table_size])
