@@ -178,6 +178,7 @@ def test_serial_parity_fuzz():
         n_neg = int(rng.choice([0, 1, 3, 7]))
         wm = str(rng.choice(["canonical", "reference"]))
         sub = bool(rng.integers(0, 2))
+        shared = bool(rng.integers(0, 2))
         n_tokens = int(rng.integers(30, 400))
         n_sent = int(rng.integers(1, 8))
         tokens = rng.integers(0, vocab, n_tokens).astype(np.int32)
@@ -193,16 +194,17 @@ def test_serial_parity_fuzz():
         a0, a1 = syn0.copy(), syn1.copy()
         st_py = cpu_ref.train_batch_oracle(
             a0, a1, tokens, offsets, kp, table, 0.04, window, n_neg,
-            seed=trial * 7, sent_id_base=trial, window_mode=wm)
+            seed=trial * 7, sent_id_base=trial, window_mode=wm,
+            shared_negatives=shared)
         gs = _gpu_setup(syn0, syn1, table)
         if sub:
             gs.set_subsample(counts, int(counts.sum()), 0.02)
         gs.train_batch(_to_dev(tokens), _to_dev(offsets), 0.04, window,
                        n_neg, trial * 7, sent_id_base=trial, window_mode=wm,
-                       serial=True, atomic=False)
+                       serial=True, atomic=False, shared_negatives=shared)
         torch.cuda.synchronize()
         st = gs.read_stats()
-        ctx = f"trial={trial} vocab={vocab} dim={dim} w={window} n={n_neg} "               f"wm={wm} sub={sub}"
+        ctx = f"trial={trial} vocab={vocab} dim={dim} w={window} n={n_neg} "               f"wm={wm} sub={sub} shared={shared}"
         assert st.pairs == st_py.pairs, ctx
         assert st.words_trained == st_py.words_trained, ctx
         g0, g1 = gs.to_host()

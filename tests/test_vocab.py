@@ -62,3 +62,33 @@ def test_words_roundtrip(tmp_path):
     v2 = Vocabulary.load_words(p, v.counts)
     assert v2.words == v.words
     assert v2.index == v.index
+
+
+def test_update_mode_resolution():
+    from glint_word2vec_amd.config import Word2VecConfig
+    c = Word2VecConfig()
+    assert c.update_mode == "hybrid"
+    assert c.effective_atomic_below() == c.hybrid_hot_rows == 32768
+    assert c.effective_atomic_floor() == 128
+    assert c.effective_atomic_floor(1_000_000) == 128
+    assert c.effective_atomic_floor(3611) == 3      # small-vocab scaling
+    c2 = Word2VecConfig(update_mode="hogwild")
+    assert c2.effective_atomic_below() == 0
+    c3 = Word2VecConfig(update_mode="atomic")
+    assert c3.effective_atomic_below() == 2 ** 31 - 1
+    assert c3.effective_atomic_floor(100) == 0
+    # deprecated alias wins, even assigned post-construction
+    c4 = Word2VecConfig(atomic_updates=False)
+    assert c4.resolved_update_mode() == "hogwild"
+    c5 = Word2VecConfig()
+    c5.atomic_updates = True
+    assert c5.resolved_update_mode() == "atomic"
+
+
+def test_choose_engine_policy():
+    from glint_word2vec_amd.config import choose_engine, round_stride_py
+    assert choose_engine(1_000_000, 300, 2, 1) == "fused"
+    assert choose_engine(1_000_000, 300, 2, 8) == "dp"
+    assert choose_engine(10_000_000, 300, 2, 8) == "dim"
+    assert round_stride_py(300) == 320
+    assert round_stride_py(1024) == 1024
