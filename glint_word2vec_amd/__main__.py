@@ -78,6 +78,9 @@ def main(argv=None):
     p.add_argument("--sharded", action="store_true")
     p = sub.add_parser("export", help="write word2vec text format (toLocal)")
     p.add_argument("model"); p.add_argument("out")
+    p.add_argument("--sharded", action="store_true",
+                   help="stream the export from device shards (no host "
+                        "matrix; torchrun for multi-GPU)")
     p = sub.add_parser("info", help="print model metadata")
     p.add_argument("model")
     args = ap.parse_args(argv)
@@ -116,8 +119,12 @@ def main(argv=None):
         for w, c in model.analogy(args.pos, args.minus, args.n):
             print(f"{c:.4f}\t{w}")
     elif args.cmd == "export":
-        model = GlintWord2VecModel.load(args.model)
-        model.to_local().save(args.out)
+        if args.sharded:
+            model = GlintWord2VecModel.load_sharded(args.model)
+            model.export_text(args.out)
+        else:
+            model = GlintWord2VecModel.load(args.model)
+            model.to_local().save(args.out)
         print(f"wrote {args.out}")
     elif args.cmd == "info":
         import json

@@ -503,6 +503,15 @@ class GlintWord2VecModel:
         f = self._f32()
         return {w: f[i].copy() for i, w in enumerate(self.vocab.words)}
 
+    def get_vectors_df(self):
+        """DataFrame of (word, vector) — the ml-layer getVectors shape
+        (ml:342-364)."""
+        import pandas as pd
+        f = self._f32()
+        return pd.DataFrame({"word": list(self.vocab.words),
+                             "vector": [f[i].copy() for i in
+                                        range(self.num_words)]})
+
     def to_local(self) -> LocalWord2VecModel:
         return LocalWord2VecModel(list(self.vocab.words), self._f32().copy())
 

@@ -69,3 +69,24 @@ def test_cli_resume_and_mid_checkpoints(tmp_path):
     assert r.returncode == 0, r.stderr
     r = _run(["info", m2])
     assert r.returncode == 0
+
+
+def test_cli_sharded_similar_and_export(tmp_path):
+    import numpy as np
+    from glint_word2vec_amd.__main__ import main
+    from glint_word2vec_amd.checkpoint import save_model
+    from glint_word2vec_amd.config import Word2VecConfig
+    from glint_word2vec_amd.vocab import Vocabulary
+    rng = np.random.default_rng(1)
+    words = [f"w{i}" for i in range(30)]
+    voc = Vocabulary(words=words, counts=np.ones(30, dtype=np.int64),
+                     index={w: i for i, w in enumerate(words)},
+                     train_words_count=30)
+    syn0 = rng.standard_normal((30, 8)).astype(np.float32)
+    path = str(tmp_path / "m")
+    save_model(path, Word2VecConfig(vector_size=8), voc, syn0, num_shards=2)
+    assert main(["similar", path, "w3", "-n", "3", "--sharded"]) == 0
+    out = str(tmp_path / "v.txt")
+    assert main(["export", path, out, "--sharded"]) == 0
+    lines = open(out).read().splitlines()
+    assert lines[0] == "30 8" and len(lines) == 31

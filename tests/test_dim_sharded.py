@@ -171,3 +171,39 @@ def test_world3_uneven_widths_match_world1(tmp_path):
     assert np.abs(s0 - r0).max() < 0.02       # measured 0.008 worst-case
     assert np.abs(s0 - r0).mean() < 1e-3      # measured ~1e-4
     assert np.abs(s1 - r1).max() < 0.02
+
+
+def test_dim_cpu_shared_negatives_matches_oracle():
+    """Shared-negative flag through the dim engine's CPU phases: pair
+    counts and values match the oracle's shared enumeration."""
+    import numpy as np
+    import torch
+    from glint_word2vec_amd.ops import cpu_ref
+    from glint_word2vec_amd.models import sgns
+    from glint_word2vec_amd.parallel.dim_sharded import DimShardedSgns
+    from glint_word2vec_amd.vocab import build_unigram_table
+    rng = np.random.default_rng(4)
+    vocab, dim = 60, 12
+    tokens = rng.integers(0, vocab, 240).astype(np.int32)
+    offsets = np.array([0, 120, 240], dtype=np.int32)
+    counts = np.bincount(tokens, minlength=vocab).astype(np.int64) + 1
+    syn0, syn1 = sgns.init_tables(vocab, dim, 5)
+    a0, a1 = syn0.copy(), syn1.copy()
+    table = build_unigram_table(counts, 1009)
+    st_py = cpu_ref.train_batch_oracle(a0, a1, tokens, offsets, None, table,
+                                       0.04, 3, 4, seed=9,
+                                       shared_negatives=True)
+    eng = DimShardedSgns(vocab, dim, device="cpu", seed=1, counts=counts,
+                         table_size=1009, shared_negatives=True,
+                         chunk_words=10 ** 9)
+    eng.load_host(syn0, syn1)
+    # rebuild the table identically (constructor uses the same builder)
+    eng.table = torch.from_numpy(table)
+    eng.train_step(torch.from_numpy(tokens), torch.from_numpy(offsets),
+                   0.04, 3, 4, seed=9, offsets_host=offsets)
+    st = eng.read_stats()
+    assert st.pairs == st_py.pairs
+    assert st.positives == st_py.positives
+    s0, s1 = eng.to_host()
+    np.testing.assert_allclose(s0, a0, rtol=1e-4, atol=1e-6)
+    np.testing.assert_allclose(s1, a1, rtol=1e-4, atol=1e-6)
