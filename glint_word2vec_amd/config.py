@@ -120,9 +120,10 @@ class Word2VecConfig:
     # hybrid: rows < this stay hogwild even inside the atomic head — the
     # ultra-hot top rows take a double-digit share of all negative-table
     # draws, and atomics there serialize on a handful of cachelines (the
-    # measured hybrid cliff); they are subsample-suppressed as centers
-    # anyway, so the quality cost is nil (benchmarks/results.md).
-    hybrid_skip_rows: int = 128
+    # measured hybrid cliff).  Round-2 sweep at vocab 1M: floor 64 gives
+    # planted-NN 0.872 vs 0.893 full-atomic and 0.861 at floor 128, all
+    # at ~171M words/s (benchmarks/results.md).
+    hybrid_skip_rows: int = 64
     # Deprecated alias (round-1 API): True -> "atomic", False -> "hogwild".
     # None (default) leaves update_mode in charge.
     atomic_updates: Optional[bool] = None
