@@ -2006,6 +2006,34 @@ __global__ __launch_bounds__(256) void sub_f32_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// scores: out[r] = dot(syn0[r], q) [/ norms[r]] — the findSynonyms GEMV
+// (Glint multiply, mllib:598).  rocBLAS bf16 GEMV sustains only ~1.6 TB/s
+// on this shape; a wave-per-row kernel with the query held in registers
+// streams the table at HBM rate.  norms == nullptr skips the divide.
+// ---------------------------------------------------------------------------
+template <typename T, int NC>
+__global__ __launch_bounds__(256) void scores_kernel(
+    const T* __restrict__ syn0, int64_t vocab, int64_t stride,
+    const float* __restrict__ q, const float* __restrict__ norms,
+    float* __restrict__ out) {
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int64_t wave_gid = (int64_t)blockIdx.x * kWavesPerBlock + wave;
+  const int64_t total_waves = (int64_t)gridDim.x * kWavesPerBlock;
+  float qv[NC];
+  RowIO<float, NC>::load(q, qv, lane);   // query resident in registers
+  for (int64_t r = wave_gid; r < vocab; r += total_waves) {
+    float v[NC];
+    RowIO<T, NC>::load(syn0 + r * stride, v, lane);
+    float f = 0.0f;
+#pragma unroll
+    for (int k = 0; k < NC; ++k) f += qv[k] * v[k];
+    f = wave_sum_f32(f);
+    if (lane == 0) out[r] = norms ? f / norms[r] : f;
+  }
+}
+
+// ---------------------------------------------------------------------------
 // norms: Euclidean norm of every row (Glint norms, mllib:486).
 // ---------------------------------------------------------------------------
 template <typename T, int NC>
@@ -2533,6 +2561,39 @@ static void pull_average(uintptr_t syn0, int is_bf16, int64_t stride,
 }
 
 template <typename T>
+static void launch_scores_t(uintptr_t syn0, int64_t vocab, int64_t stride,
+                            uintptr_t q, uintptr_t norms, uintptr_t out,
+                            int blocks, hipStream_t stream) {
+  const int nc = (int)(stride / 64);
+  switch (nc) {
+#define CASE_NC(N)                                                         \
+  case N:                                                                  \
+    hipLaunchKernelGGL((scores_kernel<T, N>), dim3(blocks), dim3(256), 0,  \
+                       stream, (const T*)syn0, vocab, stride,              \
+                       (const float*)q, (const float*)norms, (float*)out); \
+    return;
+    FOR_EACH_NC(CASE_NC)
+#undef CASE_NC
+    default:
+      throw std::runtime_error("unsupported NC");
+  }
+}
+
+static void scores(uintptr_t syn0, int is_bf16, int64_t vocab,
+                   int64_t stride, uintptr_t q, uintptr_t norms,
+                   uintptr_t out, int blocks, uintptr_t stream_ptr) {
+  HIP_CLEAR_ERROR();
+  hipStream_t stream = (hipStream_t)stream_ptr;
+  if (is_bf16)
+    launch_scores_t<uint16_t>(syn0, vocab, stride, q, norms, out, blocks,
+                              stream);
+  else
+    launch_scores_t<float>(syn0, vocab, stride, q, norms, out, blocks,
+                           stream);
+  HIP_CHECK(hipGetLastError());
+}
+
+template <typename T>
 static void launch_norms_t(uintptr_t syn0, int64_t vocab, int64_t stride,
                            uintptr_t out, int blocks, hipStream_t stream) {
   const int nc = (int)(stride / 64);
@@ -2671,6 +2732,7 @@ PYBIND11_MODULE(_hip_native, m) {
   m.def("plan_emit", &plan_emit);
   m.def("pull_average", &pull_average);
   m.def("norms", &norms);
+  m.def("scores", &scores);
   m.def("gather_rows", &gather_rows);
   m.def("scatter_add_rows", &scatter_add_rows);
   m.def("sub_rows", &sub_rows);

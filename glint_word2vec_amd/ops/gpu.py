@@ -216,12 +216,24 @@ class GpuSgns:
         save_model_streaming(path, config, vocab, row_block,
                              num_shards=num_shards)
 
-    def multiply(self, vec: torch.Tensor) -> torch.Tensor:
-        """Whole-table GEMV — a plain library GEMM-shaped op: rocBLAS via
-        torch.matmul is the right tool (guide: hand-write only fused ops)."""
-        v = torch.zeros(self.stride, dtype=self.syn0.dtype, device=self.device)
-        v[:self.dim] = vec.to(self.syn0.dtype)
-        return (self.syn0 @ v).float()
+    def multiply(self, vec: torch.Tensor, norms: Optional[torch.Tensor]
+                 = None) -> torch.Tensor:
+        """Whole-table GEMV scores (Glint multiply, mllib:598) — a
+        hand-written wave-per-row kernel: rocBLAS bf16 GEMV sustains only
+        ~1.6 TB/s on this tall-skinny shape, the kernel streams the table
+        at HBM rate (~5x, benchmarks/serving_probe.py).  ``norms`` (>0,
+        pre-clamped) fuses the cosine divide."""
+        q = torch.zeros(self.stride, dtype=torch.float32, device=self.device)
+        q[:self.dim] = vec.float().to(self.device)
+        out = torch.empty(self.vocab_size, dtype=torch.float32,
+                          device=self.device)
+        st = torch.cuda.current_stream(self.device)
+        self.native.scores(self.syn0.data_ptr(), int(self.is_bf16),
+                           self.vocab_size, self.stride, q.data_ptr(),
+                           0 if norms is None else norms.data_ptr(),
+                           out.data_ptr(), 2048, st.cuda_stream)
+        self._scores_args = (q, out) if norms is None else (q, out, norms)
+        return out
 
     def synonyms_query(self, vec: torch.Tensor, k: int):
         """Single-query findSynonyms core as a replayed hipGraph: the
@@ -239,20 +251,18 @@ class GpuSgns:
             side.wait_stream(torch.cuda.current_stream(self.device))
             with torch.cuda.stream(side):
                 for _ in range(2):   # warmup allocations before capture
-                    cos = (self.multiply_batch(self._g_in)
-                           / self._syn_norms[None, :])
-                    torch.topk(cos, k, dim=1)
+                    cos = self.multiply(self._g_in[0], norms=self._syn_norms)
+                    torch.topk(cos, k)
             torch.cuda.current_stream(self.device).wait_stream(side)
             g = torch.cuda.CUDAGraph()
             with torch.cuda.graph(g):
-                cos = (self.multiply_batch(self._g_in)
-                       / self._syn_norms[None, :])
-                self._g_val, self._g_idx = torch.topk(cos, k, dim=1)
+                cos = self.multiply(self._g_in[0], norms=self._syn_norms)
+                self._g_val, self._g_idx = torch.topk(cos, k)
             self._syn_graph = g
             self._syn_graph_k = k
         self._g_in.copy_(vec.reshape(1, -1).to(torch.float32))
         self._syn_graph.replay()
-        return self._g_val[0], self._g_idx[0]
+        return self._g_val, self._g_idx
 
     def multiply_batch(self, vecs: torch.Tensor) -> torch.Tensor:
         """Multi-query scores: one rocBLAS GEMM [Q, stride] x
