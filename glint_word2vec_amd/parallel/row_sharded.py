@@ -503,6 +503,13 @@ class RowShardedSgns:
                               tdtype)
             comm.all_to_all_single_v(cache0, rows_send0, cnt0_h, rc0)
             comm.all_to_all_single_v(cache1, rows_send1, cnt1_h, rc1)
+            # orig snapshots MUST be ordered after the alltoall on the
+            # same (comm) stream — a compute-stream copy could read the
+            # cache before the exchange lands
+            st["orig0"] = self._ws("o0" + sl, *cache0.shape, tdtype)
+            st["orig1"] = self._ws("o1" + sl, *cache1.shape, tdtype)
+            st["orig0"].copy_(cache0)
+            st["orig1"].copy_(cache1)
             if self.is_cuda:
                 done = torch.cuda.Event()
                 done.record(torch.cuda.current_stream(dev))
@@ -511,10 +518,6 @@ class RowShardedSgns:
         st["gc"] = pos0.index_select(0, inv_c).contiguous()
         st["gt"] = pos1.index_select(0, inv_t).contiguous()
         st["cache0"], st["cache1"] = cache0, cache1
-        st["orig0"] = self._ws("o0" + sl, *cache0.shape, tdtype)
-        st["orig1"] = self._ws("o1" + sl, *cache1.shape, tdtype)
-        st["orig0"].copy_(cache0)
-        st["orig1"].copy_(cache1)
         st["slot"] = sl
         st["idx_recv"] = (idx_recv0, idx_recv1)
         st["splits"] = (cnt0_h, rc0, cnt1_h, rc1)
