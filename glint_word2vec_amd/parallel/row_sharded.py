@@ -490,6 +490,13 @@ class RowShardedSgns:
             idx_recv1 = self._ws("i1" + sl, int(rc1.sum()), 1, torch.int32)
             comm.all_to_all_single_v(idx_recv0, loc0, rc0, cnt0_h)
             comm.all_to_all_single_v(idx_recv1, loc1, rc1, cnt1_h)
+            if self.is_cuda:
+                # loc* were allocated on the compute stream and die with
+                # this function — tell the allocator the comm stream still
+                # reads them (else their blocks could be re-issued to a
+                # compute-stream tensor mid-send)
+                loc0.record_stream(torch.cuda.current_stream(dev))
+                loc1.record_stream(torch.cuda.current_stream(dev))
             # owners gather requested rows; rows return in owner-major order
             rows_send0 = self._ws("s0" + sl, idx_recv0.numel(),
                                   self.store_stride, tdtype)
