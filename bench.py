@@ -61,10 +61,11 @@ def parse_args():
                         "position reused across its contexts (opt-in; cuts "
                         "target-row traffic ~(1+n)/(1+n/2b)-fold at high "
                         "n_neg — the dim-1024 neg-25 config)")
-    p.add_argument("--pair-mode", type=int, default=None, choices=[0, 1, 2],
+    p.add_argument("--pair-mode", type=int, default=None, choices=[0, 1, 2, 3],
                    help="fused-kernel variant: 0=one pair/wave (64-lane), "
                         "1=two pairs (32-lane halves, default), "
-                        "2=four pairs (16-lane quarters)")
+                        "2=four pairs (16-lane quarters), "
+                        "3=two pairs software-pipelined 2-deep")
     p.add_argument("--profile-steps", type=int, default=0,
                    help="run only this many steps, no warmup JSON (rocprof)")
     p.add_argument("--device", choices=["cuda", "cpu"], default="cuda",

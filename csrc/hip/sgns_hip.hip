@@ -964,7 +964,12 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void sgns_train_kernel(KernelA
 // one extra pair of hogwild staleness; the serial/parity path keeps the
 // 64-lane kernel.  Center grads accumulate per half and are combined with
 // one cross-half shuffle at position end.
-template <typename T, int NCH, bool ATOMIC>
+// PIPE: software-pipeline the half-wave pair blocks 2-deep — the next
+// block's row loads issue BEFORE this block's update, so the update's
+// atomic/store traffic (which clogs vmcnt) no longer delays the next
+// load's waitcnt.  Costs NCH extra VGPRs; one extra block of hogwild
+// staleness (same race class as the half-split itself).
+template <typename T, int NCH, bool ATOMIC, bool PIPE = false>
 struct TrainPhase2 {
   T* syn0;
   T* syn1;
@@ -992,44 +997,75 @@ struct TrainPhase2 {
     for (int k = 0; k < NCH; ++k) grad[k] = 0.0f;
   }
 
+  __device__ __forceinline__ void do_block(uint32_t enc, bool active,
+                                           float (&t_row)[NCH], T* t_ptr) {
+    float f = 0.0f;
+#pragma unroll
+    for (int m = 0; m < NCH; ++m) f += c_row[m] * t_row[m];
+    f = half_sum_f32(f);
+    const float label = (enc & 0x80000000u) ? 1.0f : 0.0f;
+    const float g0 = (label - sigma_of(f, exp_table, exp_table_size)) * alpha;
+    const float g = active ? g0 : 0.0f;     // idle half: zero contribution
+#pragma unroll
+    for (int m = 0; m < NCH; ++m) grad[m] += g * t_row[m];
+    if (active) {
+      const int32_t rid = (int32_t)(enc & 0x7FFFFFFFu);
+      const bool use_atomic =
+          ATOMIC && (atomic_below < 0 ? (label > 0.5f)
+                                      : (rid < atomic_below && rid >= atomic_floor));
+      if (use_atomic) {
+#pragma unroll
+        for (int m = 0; m < NCH; ++m) t_row[m] = g * c_row[m];
+        RowIO32<T, NCH>::atomic_add(t_ptr, t_row, l32);
+      } else {
+#pragma unroll
+        for (int m = 0; m < NCH; ++m) t_row[m] += g * c_row[m];
+        RowIO32<T, NCH>::store(t_ptr, t_row, l32);
+      }
+      ++w_pairs;
+      if (label > 0.5f) {
+        ++w_pos;
+        w_fplus += f;
+      }
+    }
+  }
+
+  __device__ __forceinline__ void load_block(const uint32_t* tl, int count,
+                                             int k, uint32_t& enc, T*& ptr,
+                                             float (&buf)[NCH], bool& act) {
+    const int my = k + half;
+    act = my < count;
+    enc = tl[act ? my : k];
+    ptr = syn1 + (int64_t)(enc & 0x7FFFFFFFu) * stride;
+    RowIO32<T, NCH>::load(ptr, buf, l32);
+  }
+
   __device__ __forceinline__ void process_pairs(const uint32_t* tl, int count,
                                                 int64_t) {
-    for (int k = 0; k < count; k += 2) {
-      const int my = k + half;
-      const bool active = my < count;
-      const uint32_t enc = tl[active ? my : k];
-      T* t_ptr = syn1 + (int64_t)(enc & 0x7FFFFFFFu) * stride;
-      float t_row[NCH];
-      RowIO32<T, NCH>::load(t_ptr, t_row, l32);
-      float f = 0.0f;
-#pragma unroll
-      for (int m = 0; m < NCH; ++m) f += c_row[m] * t_row[m];
-      f = half_sum_f32(f);
-      const float label = (enc & 0x80000000u) ? 1.0f : 0.0f;
-      const float g0 = (label - sigma_of(f, exp_table, exp_table_size)) * alpha;
-      const float g = active ? g0 : 0.0f;   // idle half: zero contribution
-#pragma unroll
-      for (int m = 0; m < NCH; ++m) grad[m] += g * t_row[m];
-      if (active) {
-        const int32_t rid = (int32_t)(enc & 0x7FFFFFFFu);
-        const bool use_atomic =
-            ATOMIC && (atomic_below < 0 ? (label > 0.5f)
-                                        : (rid < atomic_below && rid >= atomic_floor));
-        if (use_atomic) {
-#pragma unroll
-          for (int m = 0; m < NCH; ++m) t_row[m] = g * c_row[m];
-          RowIO32<T, NCH>::atomic_add(t_ptr, t_row, l32);
-        } else {
-#pragma unroll
-          for (int m = 0; m < NCH; ++m) t_row[m] += g * c_row[m];
-          RowIO32<T, NCH>::store(t_ptr, t_row, l32);
-        }
-        ++w_pairs;
-        if (label > 0.5f) {
-          ++w_pos;
-          w_fplus += f;
-        }
+    if (!PIPE) {
+      for (int k = 0; k < count; k += 2) {
+        uint32_t enc;
+        T* t_ptr;
+        float t_row[NCH];
+        bool active;
+        load_block(tl, count, k, enc, t_ptr, t_row, active);
+        do_block(enc, active, t_row, t_ptr);
       }
+      return;
+    }
+    // 2-deep ping-pong (TrainPhase's pattern, per half-wave block)
+    uint32_t e0, e1;
+    T *p0, *p1;
+    float b0[NCH], b1[NCH];
+    bool a0, a1;
+    load_block(tl, count, 0, e0, p0, b0, a0);
+    for (int k = 0;; k += 4) {
+      if (k + 2 < count) load_block(tl, count, k + 2, e1, p1, b1, a1);
+      do_block(e0, a0, b0, p0);
+      if (k + 2 >= count) break;
+      if (k + 4 < count) load_block(tl, count, k + 4, e0, p0, b0, a0);
+      do_block(e1, a1, b1, p1);
+      if (k + 4 >= count) break;
     }
   }
 
@@ -1052,7 +1088,7 @@ struct TrainPhase2 {
   }
 };
 
-template <typename T, int NCH, bool ATOMIC>
+template <typename T, int NCH, bool ATOMIC, bool PIPE = false>
 __global__ __launch_bounds__(64 * kWavesPerBlock) void sgns_train2_kernel(KernelArgs a) {
   __shared__ int32_t sbuf[kWavesPerBlock][kMaxSent];
   __shared__ uint32_t tbuf[kWavesPerBlock][64];
@@ -1062,7 +1098,7 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void sgns_train2_kernel(Kernel
   const int64_t wave_gid = (int64_t)blockIdx.x * waves_in_block + wave;
   const int64_t total_waves = (int64_t)gridDim.x * waves_in_block;
 
-  TrainPhase2<T, NCH, ATOMIC> ph{};
+  TrainPhase2<T, NCH, ATOMIC, PIPE> ph{};
   ph.syn0 = (T*)a.syn0;
   ph.syn1 = (T*)a.syn1;
   ph.stride = a.stride;
@@ -2077,23 +2113,30 @@ static void launch_train_nc(const KernelArgs& a, bool atomic, int blocks,
 template <typename T, int NCH>
 static void launch_train2_nch(const KernelArgs& a, bool atomic, int blocks,
                               int pos_blocks, int threads,
-                              hipStream_t stream) {
-  if (atomic)
-    hipLaunchKernelGGL((sgns_train2_kernel<T, NCH, true>),
+                              hipStream_t stream, bool pipe) {
+  if (atomic && pipe)
+    hipLaunchKernelGGL((sgns_train2_kernel<T, NCH, true, true>),
+                       dim3(blocks, pos_blocks), dim3(threads), 0, stream, a);
+  else if (atomic)
+    hipLaunchKernelGGL((sgns_train2_kernel<T, NCH, true, false>),
+                       dim3(blocks, pos_blocks), dim3(threads), 0, stream, a);
+  else if (pipe)
+    hipLaunchKernelGGL((sgns_train2_kernel<T, NCH, false, true>),
                        dim3(blocks, pos_blocks), dim3(threads), 0, stream, a);
   else
-    hipLaunchKernelGGL((sgns_train2_kernel<T, NCH, false>),
+    hipLaunchKernelGGL((sgns_train2_kernel<T, NCH, false, false>),
                        dim3(blocks, pos_blocks), dim3(threads), 0, stream, a);
 }
 
 template <typename T>
 static void launch_train2(const KernelArgs& a, int nc, bool atomic, int blocks,
-                          int pos_blocks, int threads, hipStream_t stream) {
+                          int pos_blocks, int threads, hipStream_t stream,
+                          bool pipe = false) {
   switch (nc) {
 #define CASE_NC2(N)                                                          \
   case N:                                                                    \
     launch_train2_nch<T, 2 * N>(a, atomic, blocks, pos_blocks, threads,      \
-                                stream);                                     \
+                                stream, pipe);                               \
     return;
     FOR_EACH_NC(CASE_NC2)
 #undef CASE_NC2
@@ -2202,7 +2245,14 @@ static void sgns_train(uintptr_t syn0, uintptr_t syn1, int is_bf16,
   hipStream_t stream = (hipStream_t)stream_ptr;
   const bool use_atomic = atomic_below != 0;
   if (pos_blocks < 1) pos_blocks = 1;
-  if (pair2 == 2 && threads == 256) {
+  if (pair2 == 3 && threads == 256) {
+    if (is_bf16)
+      launch_train2<uint16_t>(a, nc, use_atomic, blocks, pos_blocks, threads,
+                              stream, true);
+    else
+      launch_train2<float>(a, nc, use_atomic, blocks, pos_blocks, threads,
+                           stream, true);
+  } else if (pair2 == 2 && threads == 256) {
     if (is_bf16)
       launch_train4<uint16_t>(a, nc, use_atomic, blocks, pos_blocks, threads,
                               stream);
