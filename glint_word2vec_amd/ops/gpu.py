@@ -140,7 +140,10 @@ class GpuSgns:
             # 2*NC floats each) collapse occupancy — one 64-lane pair per
             # wave runs 2.5x faster at dim 1024 (benchmarks/results.md
             # round-2 A/B: 15.0M vs 5.9M words/s at dim=1024 neg=25)
-            pair_mode = 1 if self.stride <= 512 else 0
+            # 3 = two pairs/wave with 2-deep block pipelining (round-2
+            # A/B: 169.7->173.4M hybrid, 138.8->153.5M at vocab 10M — the
+            # atomic/store traffic no longer clogs the next load's vmcnt)
+            pair_mode = 3 if self.stride <= 512 else 0
         num_sent = int(offsets.numel() - 1)
         if num_sent <= 0:
             return

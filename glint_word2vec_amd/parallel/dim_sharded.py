@@ -206,7 +206,7 @@ class DimShardedSgns:
             int(sent_id_base), int(self.window_mode == "reference"),
             self.atomic_below, self._stats.data_ptr(), nblocks, pos_blocks,
             nthreads, stream.cuda_stream, 0, 0,
-            1 if self.stride <= 512 else 0, self.atomic_floor,
+            3 if self.stride <= 512 else 0, self.atomic_floor,
             self.shared_neg)
 
     def _train_step_gpu(self, tokens, offsets, chunks, alpha, window, n_neg,

@@ -660,7 +660,7 @@ class RowShardedSgns:
             int(self.window_mode == "reference"), self.atomic_below,
             self._stats.data_ptr(), nblocks, pos_blocks, 256,
             stream.cuda_stream, 0, 0,
-            1 if self.store_stride <= 512 else 0, self.atomic_floor,
+            3 if self.store_stride <= 512 else 0, self.atomic_floor,
             self.shared_neg)
 
     def train_step(self, tokens: np.ndarray, offsets: np.ndarray,
