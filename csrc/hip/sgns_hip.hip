@@ -1846,7 +1846,7 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void train_pairs_kernel(
 // the rest plain RMW — the hybrid update mode.  Meaningful when ids are
 // global words (direct mode: vocab sorted by count, ids < K = Zipf head);
 // cache modes pass INT32_MAX (all atomic).
-template <typename T, int NCH, bool ATOMIC>
+template <typename T, int NCH, bool ATOMIC, bool PIPE = false>
 __global__ __launch_bounds__(64 * kWavesPerBlock) void train_pairs2_kernel(
     T* __restrict__ cache0, T* __restrict__ cache1, int64_t stride,
     const int32_t* __restrict__ group_center,
@@ -1872,25 +1872,29 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void train_pairs2_kernel(
     RowIO32<T, NCH>::load(c_ptr, c_row, l32);
 #pragma unroll
     for (int k = 0; k < NCH; ++k) grad[k] = 0.0f;
+    const int64_t ps = group_offsets[g];
     const int64_t pe = group_offsets[g + 1];
-    for (int64_t p = group_offsets[g]; p < pe; p += 2) {
+    auto load_blk = [&](int64_t p, int32_t& tid, T*& t_ptr,
+                        float (&t_row)[NCH], bool& act, float& label) {
       const int64_t my = p + half;
-      const bool active = my < pe;
-      const int64_t pid = active ? my : p;
-      const int32_t tid = pair_target[pid];
-      T* t_ptr = cache1 + (int64_t)tid * stride;
-      float t_row[NCH];
+      act = my < pe;
+      const int64_t pid = act ? my : p;
+      tid = pair_target[pid];
+      label = pair_label[pid];
+      t_ptr = cache1 + (int64_t)tid * stride;
       RowIO32<T, NCH>::load(t_ptr, t_row, l32);
+    };
+    auto do_blk = [&](int32_t tid, T* t_ptr, float (&t_row)[NCH], bool act,
+                      float label) {
       float f = 0.0f;
 #pragma unroll
       for (int k = 0; k < NCH; ++k) f += c_row[k] * t_row[k];
       f = half_sum_f32(f);
-      const float label = pair_label[pid];
       const float g0 = (label - sigmoid_clipped(f)) * alpha;
-      const float gg = active ? g0 : 0.0f;   // idle half: zero contribution
+      const float gg = act ? g0 : 0.0f;      // idle half: zero contribution
 #pragma unroll
       for (int k = 0; k < NCH; ++k) grad[k] += gg * t_row[k];
-      if (active) {
+      if (act) {
         if (ATOMIC && tid < atomic_below && tid >= atomic_floor) {
 #pragma unroll
           for (int k = 0; k < NCH; ++k) t_row[k] = gg * c_row[k];
@@ -1905,6 +1909,27 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void train_pairs2_kernel(
           ++w_pos;
           w_fplus += f;
         }
+      }
+    };
+    if (!PIPE) {
+      for (int64_t p = ps; p < pe; p += 2) {
+        int32_t tid; T* t_ptr; float t_row[NCH]; bool act; float lb;
+        load_blk(p, tid, t_ptr, t_row, act, lb);
+        do_blk(tid, t_ptr, t_row, act, lb);
+      }
+    } else if (ps < pe) {
+      // 2-deep block pipeline: next block's loads issue before this
+      // block's update traffic (same rationale as TrainPhase2 PIPE)
+      int32_t tid0, tid1; T *p0, *p1; float b0[NCH], b1[NCH];
+      bool a0, a1; float l0, l1;
+      load_blk(ps, tid0, p0, b0, a0, l0);
+      for (int64_t k = ps;; k += 4) {
+        if (k + 2 < pe) load_blk(k + 2, tid1, p1, b1, a1, l1);
+        do_blk(tid0, p0, b0, a0, l0);
+        if (k + 2 >= pe) break;
+        if (k + 4 < pe) load_blk(k + 4, tid0, p0, b0, a0, l0);
+        do_blk(tid1, p1, b1, a1, l1);
+        if (k + 4 >= pe) break;
       }
     }
 #pragma unroll
@@ -2501,6 +2526,7 @@ static void train_pairs(uintptr_t cache0, uintptr_t cache1, int64_t stride,
       (int32_t)std::min<int64_t>(atomic_below, 0x7FFFFFFFLL);
   const int32_t afloor =
       (int32_t)std::min<int64_t>(atomic_floor, 0x7FFFFFFFLL);
+  const bool pipe = pair_mode == 3;
   hipStream_t stream = (hipStream_t)stream_ptr;
   unsigned long long* st = (unsigned long long*)stats;
   switch (nc) {
@@ -2519,6 +2545,19 @@ static void train_pairs(uintptr_t cache0, uintptr_t cache1, int64_t stride,
                          st ? (double*)(st + 3) : nullptr);                   \
     } else if (is_bf16) {                                                     \
       if (atomic)                                                             \
+        if (pipe)                                                             \
+        hipLaunchKernelGGL((train_pairs2_kernel<uint16_t, 2 * N, true, true>), \
+                           dim3(blocks), dim3(threads), 0, stream,      \
+                           (uint16_t*)cache0, (uint16_t*)cache1, stride,  \
+                           (const int32_t*)group_center,                \
+                           (const int64_t*)group_offsets, num_groups,   \
+                           (const int32_t*)pair_target,                 \
+                           (const float*)pair_label, (float)alpha,      \
+                           abelow, afloor,                              \
+                           st ? st + 0 : nullptr, st ? st + 1 : nullptr,\
+                           st ? st + 2 : nullptr,                       \
+                           st ? (double*)(st + 3) : nullptr);           \
+      else                                                              \
         hipLaunchKernelGGL((train_pairs2_kernel<uint16_t, 2 * N, true>),      \
                            dim3(blocks), dim3(threads), 0, stream,            \
                            (uint16_t*)cache0, (uint16_t*)cache1, stride,      \
@@ -2530,6 +2569,19 @@ static void train_pairs(uintptr_t cache0, uintptr_t cache1, int64_t stride,
                            st ? st + 2 : nullptr,                             \
                            st ? (double*)(st + 3) : nullptr);                 \
       else                                                                    \
+        if (pipe)                                                             \
+        hipLaunchKernelGGL((train_pairs2_kernel<uint16_t, 2 * N, false, true>), \
+                           dim3(blocks), dim3(threads), 0, stream,      \
+                           (uint16_t*)cache0, (uint16_t*)cache1, stride,  \
+                           (const int32_t*)group_center,                \
+                           (const int64_t*)group_offsets, num_groups,   \
+                           (const int32_t*)pair_target,                 \
+                           (const float*)pair_label, (float)alpha,      \
+                           abelow, afloor,                              \
+                           st ? st + 0 : nullptr, st ? st + 1 : nullptr,\
+                           st ? st + 2 : nullptr,                       \
+                           st ? (double*)(st + 3) : nullptr);           \
+      else                                                              \
         hipLaunchKernelGGL((train_pairs2_kernel<uint16_t, 2 * N, false>),     \
                            dim3(blocks), dim3(threads), 0, stream,            \
                            (uint16_t*)cache0, (uint16_t*)cache1, stride,      \
@@ -2541,7 +2593,20 @@ static void train_pairs(uintptr_t cache0, uintptr_t cache1, int64_t stride,
                            st ? st + 2 : nullptr,                             \
                            st ? (double*)(st + 3) : nullptr);                 \
     } else if (atomic) {                                                      \
-      hipLaunchKernelGGL((train_pairs2_kernel<float, 2 * N, true>),           \
+      if (pipe)                                                             \
+        hipLaunchKernelGGL((train_pairs2_kernel<float, 2 * N, true, true>), \
+                           dim3(blocks), dim3(threads), 0, stream,      \
+                           (float*)cache0, (float*)cache1, stride,  \
+                           (const int32_t*)group_center,                \
+                           (const int64_t*)group_offsets, num_groups,   \
+                           (const int32_t*)pair_target,                 \
+                           (const float*)pair_label, (float)alpha,      \
+                           abelow, afloor,                              \
+                           st ? st + 0 : nullptr, st ? st + 1 : nullptr,\
+                           st ? st + 2 : nullptr,                       \
+                           st ? (double*)(st + 3) : nullptr);           \
+      else                                                              \
+        hipLaunchKernelGGL((train_pairs2_kernel<float, 2 * N, true>),           \
                          dim3(blocks), dim3(threads), 0, stream,              \
                          (float*)cache0,                                      \
                          (float*)cache1, stride,                              \
@@ -2553,7 +2618,20 @@ static void train_pairs(uintptr_t cache0, uintptr_t cache1, int64_t stride,
                          st ? st + 2 : nullptr,                               \
                          st ? (double*)(st + 3) : nullptr);                   \
     } else {                                                                  \
-      hipLaunchKernelGGL((train_pairs2_kernel<float, 2 * N, false>),          \
+      if (pipe)                                                             \
+        hipLaunchKernelGGL((train_pairs2_kernel<float, 2 * N, false, true>), \
+                           dim3(blocks), dim3(threads), 0, stream,      \
+                           (float*)cache0, (float*)cache1, stride,  \
+                           (const int32_t*)group_center,                \
+                           (const int64_t*)group_offsets, num_groups,   \
+                           (const int32_t*)pair_target,                 \
+                           (const float*)pair_label, (float)alpha,      \
+                           abelow, afloor,                              \
+                           st ? st + 0 : nullptr, st ? st + 1 : nullptr,\
+                           st ? st + 2 : nullptr,                       \
+                           st ? (double*)(st + 3) : nullptr);           \
+      else                                                              \
+        hipLaunchKernelGGL((train_pairs2_kernel<float, 2 * N, false>),          \
                          dim3(blocks), dim3(threads), 0, stream,              \
                          (float*)cache0,                                      \
                          (float*)cache1, stride,                              \
