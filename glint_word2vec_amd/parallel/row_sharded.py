@@ -594,7 +594,7 @@ class RowShardedSgns:
                 cache0.data_ptr(), cache1.data_ptr(), self.store_stride,
                 gc.data_ptr(), go.data_ptr(), G, gt.data_ptr(), pl.data_ptr(),
                 float(alpha), self._stats.data_ptr(), nb, nt,
-                stream.cuda_stream, 0 if self.serial else 3,
+                stream.cuda_stream, 0 if self.serial else 1,
                 int(self.atomic), int(self.is_bf16), 2 ** 31 - 1, 0)
             self._inflight = (gc, go, gt, pl, cache0, cache1)
         else:
@@ -709,7 +709,7 @@ class RowShardedSgns:
             self.syn0.data_ptr(), self.syn1.data_ptr(), self.store_stride,
             gc.data_ptr(), go.data_ptr(), G, pt.data_ptr(), pl.data_ptr(),
             float(alpha), self._stats.data_ptr(), nb, 256,
-            stream.cuda_stream, 3, int(atomic_flag),
+            stream.cuda_stream, 1, int(atomic_flag),
             int(self.is_bf16), self.atomic_below, self.atomic_floor)
         self._inflight = (gc, go, pt, pl)
 
