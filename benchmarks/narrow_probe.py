@@ -22,11 +22,13 @@ from glint_word2vec_amd.data import synthetic_corpus
 from glint_word2vec_amd.parallel.dim_sharded import DimShardedSgns
 
 
-def make_engine(vocab, width, stride, counts, dtype):
+def make_engine(vocab, width, stride, counts, dtype, pair_mode=1):
     eng = DimShardedSgns(vocab, 300, dtype=dtype, device="cuda", seed=3,
                          counts=counts, table_size=10_000_000,
                          subsample=1e-4, chunk_words=1 << 20,
                          f_correction=True, atomic=False, narrow=False)
+    eng.single_pass_world1 = False   # measure the PHASE path, not fused
+    eng.pair_mode = pair_mode
     # override slice geometry to the 8-GPU shape (world stays 1: the
     # allreduce is a no-op; kernel work per rank is what we measure)
     eng.lo, eng.hi, eng.width = 0, width, width
@@ -67,8 +69,11 @@ def main():
     tokens = torch.from_numpy(batch.tokens).cuda()
     offsets = torch.from_numpy(batch.offsets).cuda()
 
-    for label, stride in (("narrow stride=40", 40), ("padded stride=64", 64)):
-        eng = make_engine(args.vocab, 38, stride, counts, args.dtype)
+    for label, stride, pm in (("narrow stride=40", 40, 1),
+                              ("narrow stride=40 piped", 40, 3),
+                              ("padded stride=64", 64, 1),
+                              ("padded stride=64 piped", 64, 3)):
+        eng = make_engine(args.vocab, 38, stride, counts, args.dtype, pm)
         wps = run(eng, tokens, offsets, batch.offsets, args.steps, args.warmup)
         print(f"{label}: {wps/1e6:.1f}M words/s (per-rank kernel rate, "
               f"width=38 dtype={args.dtype})")

@@ -1641,7 +1641,7 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void dots_slice2_kernel(
   }
 }
 
-template <typename T, int NCH, bool ATOMIC, bool MASKED>
+template <typename T, int NCH, bool ATOMIC, bool MASKED, bool PIPE = false>
 struct UpdateSlicePhase2 {
   T* syn0;
   T* syn1;
@@ -1670,47 +1670,80 @@ struct UpdateSlicePhase2 {
 #pragma unroll
     for (int m = 0; m < NCH; ++m) grad[m] = 0.0f;
   }
+  __device__ __forceinline__ void load_blk(const uint32_t* tl, int count,
+                                           int k, uint32_t& enc, T*& ptr,
+                                           float (&buf)[NCH], bool& act) {
+    const int my = k + half;
+    act = my < count;
+    enc = tl[act ? my : k];
+    ptr = syn1 + (int64_t)(enc & 0x7FFFFFFFu) * stride;
+    RIO32<T, NCH, MASKED>::load(ptr, buf, l32, width);
+  }
+
+  __device__ __forceinline__ void do_blk(uint32_t enc, T* t_ptr,
+                                         float (&t_row)[NCH], bool active,
+                                         int64_t idx_base, int k) {
+    const int my = k + half;
+    float f = f_base[idx_base + (active ? my : k)];
+    if (f_loc) {
+      float fresh = 0.0f;
+#pragma unroll
+      for (int m = 0; m < NCH; ++m) fresh += c_row[m] * t_row[m];
+      fresh = half_sum_f32(fresh);
+      f += world_scale * (fresh - f_loc[idx_base + (active ? my : k)]);
+    }
+    const float label = (enc & 0x80000000u) ? 1.0f : 0.0f;
+    const float g0 = (label - sigma_of(f, exp_table, exp_table_size)) * alpha;
+    const float g = active ? g0 : 0.0f;
+#pragma unroll
+    for (int m = 0; m < NCH; ++m) grad[m] += g * t_row[m];
+    if (active) {
+      const int32_t rid = (int32_t)(enc & 0x7FFFFFFFu);
+      const bool use_atomic =
+          ATOMIC && rid < atomic_below && rid >= atomic_floor;
+      if (use_atomic) {
+#pragma unroll
+        for (int m = 0; m < NCH; ++m) t_row[m] = g * c_row[m];
+        RIO32<T, NCH, MASKED>::atomic_add(t_ptr, t_row, l32, width);
+      } else {
+#pragma unroll
+        for (int m = 0; m < NCH; ++m) t_row[m] += g * c_row[m];
+        RIO32<T, NCH, MASKED>::store(t_ptr, t_row, l32, width);
+      }
+      ++w_pairs;
+      if (label > 0.5f) {
+        ++w_pos;
+        w_fplus += f;
+      }
+    }
+  }
+
   __device__ __forceinline__ void process_pairs(const uint32_t* tl, int count,
                                                 int64_t idx_base) {
-    for (int k = 0; k < count; k += 2) {
-      const int my = k + half;
-      const bool active = my < count;
-      const uint32_t enc = tl[active ? my : k];
-      T* t_ptr = syn1 + (int64_t)(enc & 0x7FFFFFFFu) * stride;
-      float t_row[NCH];
-      RIO32<T, NCH, MASKED>::load(t_ptr, t_row, l32, width);
-      float f = f_base[idx_base + (active ? my : k)];
-      if (f_loc) {
-        float fresh = 0.0f;
-#pragma unroll
-        for (int m = 0; m < NCH; ++m) fresh += c_row[m] * t_row[m];
-        fresh = half_sum_f32(fresh);
-        f += world_scale * (fresh - f_loc[idx_base + (active ? my : k)]);
+    if (!PIPE) {
+      for (int k = 0; k < count; k += 2) {
+        uint32_t enc;
+        T* t_ptr;
+        float t_row[NCH];
+        bool active;
+        load_blk(tl, count, k, enc, t_ptr, t_row, active);
+        do_blk(enc, t_ptr, t_row, active, idx_base, k);
       }
-      const float label = (enc & 0x80000000u) ? 1.0f : 0.0f;
-      const float g0 = (label - sigma_of(f, exp_table, exp_table_size)) * alpha;
-      const float g = active ? g0 : 0.0f;
-#pragma unroll
-      for (int m = 0; m < NCH; ++m) grad[m] += g * t_row[m];
-      if (active) {
-        const int32_t rid = (int32_t)(enc & 0x7FFFFFFFu);
-        const bool use_atomic =
-            ATOMIC && rid < atomic_below && rid >= atomic_floor;
-        if (use_atomic) {
-#pragma unroll
-          for (int m = 0; m < NCH; ++m) t_row[m] = g * c_row[m];
-          RIO32<T, NCH, MASKED>::atomic_add(t_ptr, t_row, l32, width);
-        } else {
-#pragma unroll
-          for (int m = 0; m < NCH; ++m) t_row[m] += g * c_row[m];
-          RIO32<T, NCH, MASKED>::store(t_ptr, t_row, l32, width);
-        }
-        ++w_pairs;
-        if (label > 0.5f) {
-          ++w_pos;
-          w_fplus += f;
-        }
-      }
+      return;
+    }
+    if (count <= 0) return;
+    uint32_t e0, e1;
+    T *p0, *p1;
+    float b0[NCH], b1[NCH];
+    bool a0, a1;
+    load_blk(tl, count, 0, e0, p0, b0, a0);
+    for (int k = 0;; k += 4) {
+      if (k + 2 < count) load_blk(tl, count, k + 2, e1, p1, b1, a1);
+      do_blk(e0, p0, b0, a0, idx_base, k);
+      if (k + 2 >= count) break;
+      if (k + 4 < count) load_blk(tl, count, k + 4, e0, p0, b0, a0);
+      do_blk(e1, p1, b1, a1, idx_base, k + 2);
+      if (k + 4 >= count) break;
     }
   }
   __device__ __forceinline__ void end_position(int32_t) {
@@ -1730,7 +1763,7 @@ struct UpdateSlicePhase2 {
   }
 };
 
-template <typename T, int NCH, bool ATOMIC, bool MASKED>
+template <typename T, int NCH, bool ATOMIC, bool MASKED, bool PIPE = false>
 __global__ __launch_bounds__(64 * kWavesPerBlock) void update_slice2_kernel(
     KernelArgs a, const int64_t* __restrict__ pair_offsets,
     const float* __restrict__ f_in, const float* __restrict__ f_loc,
@@ -1742,7 +1775,7 @@ __global__ __launch_bounds__(64 * kWavesPerBlock) void update_slice2_kernel(
   const int waves_in_block = blockDim.x >> 6;
   const int64_t wave_gid = (int64_t)blockIdx.x * waves_in_block + wave;
   const int64_t total_waves = (int64_t)gridDim.x * waves_in_block;
-  UpdateSlicePhase2<T, NCH, ATOMIC, MASKED> ph{};
+  UpdateSlicePhase2<T, NCH, ATOMIC, MASKED, PIPE> ph{};
   ph.syn0 = (T*)a.syn0;
   ph.syn1 = (T*)a.syn1;
   ph.stride = a.stride;
@@ -2428,6 +2461,31 @@ static void dots_slice(uintptr_t syn0, uintptr_t syn1, int is_bf16,
   HIP_CHECK(hipGetLastError());
 }
 
+template <typename T, int NCH>
+static void launch_update2_k(const KernelArgs& a, bool atomic, bool masked,
+                             bool pipe, int blocks, int threads,
+                             hipStream_t stream, const int64_t* poff,
+                             const float* fin, const float* floc, float ws) {
+#define UPD2(AT, MS, PP)                                                   \
+  hipLaunchKernelGGL((update_slice2_kernel<T, NCH, AT, MS, PP>),           \
+                     dim3(blocks), dim3(threads), 0, stream, a, poff, fin, \
+                     floc, ws)
+  if (atomic) {
+    if (masked) {
+      if (pipe) UPD2(true, true, true); else UPD2(true, true, false);
+    } else {
+      if (pipe) UPD2(true, false, true); else UPD2(true, false, false);
+    }
+  } else {
+    if (masked) {
+      if (pipe) UPD2(false, true, true); else UPD2(false, true, false);
+    } else {
+      if (pipe) UPD2(false, false, true); else UPD2(false, false, false);
+    }
+  }
+#undef UPD2
+}
+
 static void update_slice(uintptr_t syn0, uintptr_t syn1, int is_bf16,
                          int64_t stride, uintptr_t tokens, uintptr_t offsets,
                          int64_t num_sentences, uintptr_t keep_thr,
@@ -2463,26 +2521,11 @@ static void update_slice(uintptr_t syn0, uintptr_t syn1, int is_bf16,
 #define UPD_CASE(T, N)                                                        \
   do {                                                                        \
     if (pair_mode && threads == 256) {                                        \
-      if (atomic && masked)                                                   \
-        hipLaunchKernelGGL((update_slice2_kernel<T, 2 * N, true, true>),      \
-                           dim3(blocks), dim3(threads), 0, stream, a,         \
-                           (const int64_t*)pair_offsets, (const float*)f_in,  \
-                           (const float*)f_loc, (float)world_scale);          \
-      else if (atomic)                                                        \
-        hipLaunchKernelGGL((update_slice2_kernel<T, 2 * N, true, false>),     \
-                           dim3(blocks), dim3(threads), 0, stream, a,         \
-                           (const int64_t*)pair_offsets, (const float*)f_in,  \
-                           (const float*)f_loc, (float)world_scale);          \
-      else if (masked)                                                        \
-        hipLaunchKernelGGL((update_slice2_kernel<T, 2 * N, false, true>),     \
-                           dim3(blocks), dim3(threads), 0, stream, a,         \
-                           (const int64_t*)pair_offsets, (const float*)f_in,  \
-                           (const float*)f_loc, (float)world_scale);          \
-      else                                                                    \
-        hipLaunchKernelGGL((update_slice2_kernel<T, 2 * N, false, false>),    \
-                           dim3(blocks), dim3(threads), 0, stream, a,         \
-                           (const int64_t*)pair_offsets, (const float*)f_in,  \
-                           (const float*)f_loc, (float)world_scale);          \
+      launch_update2_k<T, 2 * N>(a, atomic, masked, pair_mode == 3, blocks,   \
+                                 threads, stream,                             \
+                                 (const int64_t*)pair_offsets,                \
+                                 (const float*)f_in, (const float*)f_loc,     \
+                                 (float)world_scale);                         \
     } else if (atomic) {                                                      \
       hipLaunchKernelGGL((update_slice_kernel<T, N, true>), dim3(blocks),     \
                          dim3(threads), 0, stream, a,                         \

@@ -116,6 +116,9 @@ class DimShardedSgns:
         self._cpu_stats = dict(pairs=0, positives=0, words_trained=0,
                                sum_fplus=0.0)
         self.serial = False   # tests: single-wave launches (oracle order)
+        # update-phase kernel variant: 1 = two pairs/wave, 3 = + 2-deep
+        # block pipelining (A/B via benchmarks/narrow_probe.py)
+        self.pair_mode = 1
 
     def _init_slices(self, seed: int, full_limit: int) -> None:
         """Same init as the single-GPU path: full-matrix U(-.5/dim,.5/dim)
@@ -257,7 +260,7 @@ class DimShardedSgns:
                 float(self.dim) / max(self.width, 1),
                 self.atomic_below, self._stats.data_ptr(),
                 nb, nt, comp.cuda_stream, 0, 0,
-                0 if self.serial else 1,
+                0 if self.serial else self.pair_mode,
                 self.width if self.narrow else 0, self.atomic_floor,
                 self.shared_neg)
 
