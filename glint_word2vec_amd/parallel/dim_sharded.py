@@ -79,10 +79,15 @@ class DimShardedSgns:
             # small byte saving on wide bandwidth-bound slices (measured:
             # dim 300 at stride 304 runs 2.4x SLOWER than padded 320;
             # width 38 at stride 40 runs 2.6% faster than padded 64)
+            # round-2 re-measure at the 8-GPU slice shape (width 38):
+            # padded stride 64 = 230.4M vs narrow stride 40 = 216.0M
+            # words/s per rank (the kernels' predicates changed since the
+            # round-1 +2.6% narrow result), and the capacity saving is
+            # irrelevant against 288 GB HBM (config 4 at world 8: 12.8 vs
+            # 20.5 GB).  Auto-narrow is therefore OFF; explicit
+            # narrow=True keeps the masked storage available.
             if narrow is None:
-                padded = self.native.round_stride(max(self.width, 1))
-                narrow = (self.width % 64 != 0
-                          and (self.width + 7) // 8 * 8 * 4 <= padded * 3)
+                narrow = False
             self.narrow = bool(narrow)
             self.stride = ((self.width + 7) // 8 * 8 if self.narrow
                            else self.native.round_stride(max(self.width, 1)))
