@@ -339,7 +339,7 @@ def test_dim_sharded_narrow_slices_gpu():
         # (phase) pipeline, not the world-1 fused shortcut
         eng.single_pass_world1 = False
         if device == "cuda":
-            assert eng.narrow == (narrow if narrow is not None else True)
+            assert eng.narrow == bool(narrow)   # auto-narrow is off
             tok = torch.from_numpy(tokens).cuda()
             off = torch.from_numpy(offsets).cuda()
         else:
