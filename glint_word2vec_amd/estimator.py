@@ -203,7 +203,8 @@ class GlintWord2Vec:
                                        init_tables=init_tables)
             if save_path is not None:
                 from .checkpoint import save_model
-                save_model(save_path, cfg, vocab, syn0, syn1)
+                save_model(save_path, cfg, vocab, syn0, syn1,
+                           num_shards=cfg.num_shards or 1)
         if not materialize:
             return None
         return GlintWord2VecModel(config=cfg, vocab=vocab, syn0=syn0, syn1=syn1,

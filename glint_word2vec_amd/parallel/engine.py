@@ -124,7 +124,8 @@ def train_gpu(cfg: Word2VecConfig, vocab: Vocabulary, batches_fn: Callable,
              st.pairs, st.sum_fplus / max(st.positives, 1))
     if save_path:
         # streamed from HBM — no full-matrix host materialisation
-        gs.save_checkpoint(save_path, cfg, vocab)
+        gs.save_checkpoint(save_path, cfg, vocab,
+                           num_shards=cfg.num_shards or 8)
     if not materialize:
         return None, None
     return gs.to_host()
@@ -312,7 +313,13 @@ def _train_sharded(cfg: Word2VecConfig, vocab: Vocabulary,
              st.sum_fplus / max(st.positives, 1))
     if save_path:
         if hasattr(eng, "save_checkpoint"):
-            eng.save_checkpoint(save_path, cfg, vocab)
+            # dim/dp honour numParameterServers -> num_shards; the row
+            # engine writes one shard per rank by construction
+            try:
+                eng.save_checkpoint(save_path, cfg, vocab,
+                                    num_shards=cfg.num_shards or 8)
+            except TypeError:
+                eng.save_checkpoint(save_path, cfg, vocab)
         else:
             # defensive fallback (all current engines stream their own
             # checkpoints via save_checkpoint)

@@ -92,3 +92,22 @@ def test_row_sharded_distributed_checkpoint(tmp_path):
     from glint_word2vec_amd import GlintWord2VecModel
     m = GlintWord2VecModel.load(str(tmp_path / "ckpt"))
     np.testing.assert_allclose(m.syn0, orig, rtol=1e-6)
+
+
+def test_num_shards_knob_controls_shard_count(tmp_path):
+    """setNumParameterServers -> num_shards controls how many shard files
+    the fit-time checkpoint writes (the reference's per-PS shard files)."""
+    import os
+    import numpy as np
+    from glint_word2vec_amd import GlintWord2Vec
+    rng = np.random.default_rng(0)
+    sents = [[f"w{rng.integers(0, 30)}" for _ in range(8)]
+             for _ in range(200)]
+    est = (GlintWord2Vec().setVectorSize(8).setMinCount(1).setSeed(1)
+           .setNumIterations(1).setUnigramTableSize(1000)
+           .setNumParameterServers(3))
+    est.config.device = "cpu"
+    out = str(tmp_path / "m")
+    est.fit(sents, save_path=out, materialize=False)
+    files = sorted(os.listdir(os.path.join(out, "shards")))
+    assert "syn0-00002.bin" in files and "syn0-00003.bin" not in files
