@@ -41,7 +41,8 @@ def save_model(path: str, config: Word2VecConfig, vocab: Vocabulary,
     with open(os.path.join(path, "metadata"), "w") as f:
         json.dump(meta, f, indent=2, sort_keys=True)
     vocab.save_words(os.path.join(path, "words"))
-    np.save(os.path.join(path, "counts.npy"), vocab.counts)
+    if vocab.counts is not None:
+        np.save(os.path.join(path, "counts.npy"), vocab.counts)
     dt = syn0.dtype
     index = {
         "num_shards": num_shards,
@@ -93,7 +94,8 @@ def save_model_streaming(path: str, config: Word2VecConfig, vocab: Vocabulary,
         with open(os.path.join(path, "metadata"), "w") as f:
             json.dump(meta, f, indent=2, sort_keys=True)
         vocab.save_words(os.path.join(path, "words"))
-        np.save(os.path.join(path, "counts.npy"), vocab.counts)
+        if getattr(vocab, "counts", None) is not None:
+            np.save(os.path.join(path, "counts.npy"), vocab.counts)
         index = {"num_shards": num_shards, "vocab": V, "dim": dim,
                  "dtype": "float32", "layout": "row_range", "bounds": bounds,
                  "has_syn1": has_syn1}

@@ -362,6 +362,31 @@ class ShardedWord2VecModel:
                 f.close()
         comm.barrier()
 
+    def save(self, path: str, num_shards: int = 8,
+             block: int = 1 << 20) -> None:
+        """Re-save the (serving, syn0-only) model as a checkpoint —
+        streaming, rank 0 writes (the reference's matrix.save,
+        mllib:493-498).  Collective."""
+        from .checkpoint import save_model_streaming
+
+        class _V:
+            num_words = self.num_words
+            counts = None
+
+            @staticmethod
+            def save_words(p):
+                with open(p, "w", encoding="utf-8") as f:
+                    for i in range(self.num_words):
+                        f.write(self.vocab.word(i) + "\n")
+
+        def row_block(which, r0, r1):
+            return self._pull_range(r0, r1)
+
+        save_model_streaming(path, self.config, _V, row_block,
+                             num_shards=num_shards, block_rows=block,
+                             has_syn1=False, write=self.rank == 0)
+        comm.barrier()
+
     def stop(self, terminate_other_clients: bool = False) -> None:
         try:
             import torch.distributed as dist

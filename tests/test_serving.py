@@ -162,3 +162,14 @@ def test_sharded_to_local_and_export(tmp_path):
     assert lines[1].split()[0] == "w000"
     got = np.array([float(x) for x in lines[8].split()[1:]])
     np.testing.assert_allclose(got, syn0[7], rtol=1e-4, atol=1e-5)
+
+
+def test_sharded_model_resave_roundtrip(tmp_path):
+    path = str(tmp_path / "model")
+    _, syn0, _ = _make_checkpoint(path, num_shards=3)
+    m = GlintWord2VecModel.load_sharded(path, device="cpu")
+    out = str(tmp_path / "resaved")
+    m.save(out, num_shards=2)
+    m2 = GlintWord2VecModel.load_sharded(out, device="cpu")
+    np.testing.assert_allclose(m2.get_vector("w007"), syn0[7], rtol=1e-6)
+    assert m2.num_words == 60
