@@ -210,22 +210,27 @@ class ShardedWord2VecModel:
         the rows it owns per sentence, one allreduce merges."""
         sums = torch.zeros((len(sentences), self.dim), dtype=torch.float32,
                            device=self.device)
-        counts = np.zeros(len(sentences), dtype=np.float32)
-        sent_idx: List[int] = []
-        tok_ids: List[int] = []
-        for si, s in enumerate(sentences):
-            for w in s:
-                i = self.vocab.get(w)
-                if i >= 0:
-                    counts[si] += 1
-                    if i % self.world == self.rank:
-                        sent_idx.append(si)
-                        tok_ids.append(i // self.world)
-        if tok_ids:
-            rows = self.shard[torch.tensor(tok_ids, dtype=torch.long,
-                                           device=self.device)].float()
-            sums.index_add_(0, torch.tensor(sent_idx, dtype=torch.long,
-                                            device=self.device), rows)
+        # one batched word->id lookup over the whole request (C++ mmap
+        # index), then owner-partial sums
+        flat: List[str] = []
+        sent_of = np.empty(sum(len(x) for x in sentences), dtype=np.int64)
+        pos = 0
+        for si, sent in enumerate(sentences):
+            flat.extend(sent)
+            sent_of[pos:pos + len(sent)] = si
+            pos += len(sent)
+        ids = (self.vocab.lookup_many(flat)
+               if hasattr(self.vocab, "lookup_many") else
+               np.array([self.vocab.get(w) for w in flat], dtype=np.int64))
+        known = ids >= 0
+        counts = np.bincount(sent_of[known],
+                             minlength=len(sentences)).astype(np.float32)
+        mine = known & (ids % self.world == self.rank)
+        if mine.any():
+            rows = self.shard[torch.from_numpy(ids[mine] // self.world)
+                              .to(self.device)].float()
+            sums.index_add_(0, torch.from_numpy(sent_of[mine])
+                            .to(self.device), rows)
         comm.all_reduce_sum(sums)
         out = sums.cpu().numpy()
         nz = counts > 0
