@@ -39,13 +39,15 @@ def write_checkpoint(path, vocab, dim, shards=8, block=1 << 20):
                    "dtype": "float32", "layout": "row_range",
                    "bounds": bounds, "has_syn1": False}, f)
     rng = np.random.default_rng(1)
+    # one random block, reused — generation must be disk-bound, not
+    # RNG-bound, at 96 GB (content is irrelevant to the RSS/latency proof)
+    proto = rng.standard_normal(block * dim).astype(np.float32)
     for s in range(shards):
         with open(os.path.join(path, "shards", f"syn0-{s:05d}.bin"),
                   "wb") as f:
             for r0 in range(bounds[s], bounds[s + 1], block):
                 r1 = min(bounds[s + 1], r0 + block)
-                f.write(rng.standard_normal((r1 - r0) * dim)
-                        .astype(np.float32).tobytes())
+                f.write(proto[:(r1 - r0) * dim].tobytes())
 
 
 def rss_gb():
