@@ -20,7 +20,8 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import numpy as np
 
 
-def write_checkpoint(path, vocab, dim, shards=8, block=1 << 20):
+def write_checkpoint(path, vocab, dim, shards=8, block=1 << 20,
+                     disk_dtype="float32"):
     from glint_word2vec_amd.config import Word2VecConfig
     os.makedirs(os.path.join(path, "shards"), exist_ok=True)
     meta = {"class": "glint_word2vec_amd.GlintWord2VecModel",
@@ -36,12 +37,15 @@ def write_checkpoint(path, vocab, dim, shards=8, block=1 << 20):
     bounds[-1] = vocab
     with open(os.path.join(path, "shards", "index.json"), "w") as f:
         json.dump({"num_shards": shards, "vocab": vocab, "dim": dim,
-                   "dtype": "float32", "layout": "row_range",
+                   "dtype": disk_dtype, "layout": "row_range",
                    "bounds": bounds, "has_syn1": False}, f)
     rng = np.random.default_rng(1)
     # one random block, reused — generation must be disk-bound, not
     # RNG-bound, at 96 GB (content is irrelevant to the RSS/latency proof)
     proto = rng.standard_normal(block * dim).astype(np.float32)
+    if disk_dtype == "bfloat16":
+        import torch
+        proto = torch.from_numpy(proto).bfloat16().view(torch.uint16).numpy()
     for s in range(shards):
         with open(os.path.join(path, "shards", f"syn0-{s:05d}.bin"),
                   "wb") as f:
@@ -59,10 +63,13 @@ def main():
     ap.add_argument("--vocab", type=int, default=10_000_000)
     ap.add_argument("--dim", type=int, default=300)
     ap.add_argument("--path", default="/tmp/sharded_load_probe")
+    ap.add_argument("--disk-dtype", default="float32",
+                    choices=["float32", "bfloat16"])
     args = ap.parse_args()
     t0 = time.time()
     if not os.path.exists(os.path.join(args.path, "metadata")):
-        write_checkpoint(args.path, args.vocab, args.dim)
+        write_checkpoint(args.path, args.vocab, args.dim,
+                         disk_dtype=args.disk_dtype)
     gen_s = time.time() - t0
     rss_before = rss_gb()
     from glint_word2vec_amd.estimator import GlintWord2VecModel
@@ -80,6 +87,7 @@ def main():
     t2 = time.time()
     print(json.dumps({
         "vocab": args.vocab, "dim": args.dim,
+        "disk_dtype": args.disk_dtype,
         "dense_load_would_need_gb": round(
             args.vocab * args.dim * 4 / 2**30, 1),
         "gen_s": round(gen_s, 1), "load_s": round(load_s, 1),

@@ -128,7 +128,11 @@ class ShardedWord2VecModel:
         tdtype = (torch.bfloat16 if (dtype == "bfloat16" or (
             dtype == "auto" and device.startswith("cuda")))
             else torch.float32)
-        file_dt = np.dtype(index["dtype"])
+        # "bfloat16" checkpoints are stored as raw u16 (checkpoint.py
+        # _DTYPES): half the disk/IO of f32 — the natural format for
+        # 100 GB-class models
+        is_bf16_file = index["dtype"] == "bfloat16"
+        file_dt = np.uint16 if is_bf16_file else np.dtype(index["dtype"])
         shard_size = (V - rank + world - 1) // world
         shard = torch.empty((shard_size, D), dtype=tdtype,
                             device=torch.device(device))
@@ -146,8 +150,11 @@ class ShardedWord2VecModel:
                     mine = (gids % world) == rank
                     if not mine.any():
                         continue
-                    rows = torch.from_numpy(
-                        np.ascontiguousarray(block[mine], dtype=np.float32))
+                    picked = np.ascontiguousarray(block[mine])
+                    rows = (torch.from_numpy(picked).view(torch.bfloat16)
+                            .float() if is_bf16_file else
+                            torch.from_numpy(picked.astype(np.float32,
+                                                           copy=False)))
                     shard[torch.from_numpy(gids[mine] // world)] = \
                         rows.to(tdtype).to(shard.device)
         vocab = LazyVocab(os.path.join(path, "words"))

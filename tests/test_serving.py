@@ -173,3 +173,29 @@ def test_sharded_model_resave_roundtrip(tmp_path):
     m2 = GlintWord2VecModel.load_sharded(out, device="cpu")
     np.testing.assert_allclose(m2.get_vector("w007"), syn0[7], rtol=1e-6)
     assert m2.num_words == 60
+
+
+def test_sharded_load_bf16_checkpoint(tmp_path):
+    """bf16-on-disk checkpoints (raw u16) load through the sharded path."""
+    import torch
+    from glint_word2vec_amd.serving import ShardedWord2VecModel
+    rng = np.random.default_rng(7)
+    vocab, dim = 40, 16
+    syn0 = rng.standard_normal((vocab, dim)).astype(np.float32)
+    syn0_bf = torch.from_numpy(syn0).bfloat16()
+    path = tmp_path / "mbf"
+    os.makedirs(path / "shards")
+    with open(path / "metadata", "w") as f:
+        json.dump({"numWords": vocab, "vectorSize": dim,
+                   "paramMap": Word2VecConfig(vector_size=dim).to_dict()}, f)
+    with open(path / "words", "w") as f:
+        f.writelines(f"w{i:03d}\n" for i in range(vocab))
+    with open(path / "shards" / "index.json", "w") as f:
+        json.dump({"num_shards": 1, "vocab": vocab, "dim": dim,
+                   "dtype": "bfloat16", "layout": "row_mod",
+                   "has_syn1": False}, f)
+    syn0_bf.view(torch.uint16).numpy().tofile(path / "shards"
+                                              / "syn0-00000.bin")
+    m = ShardedWord2VecModel.load(str(path), device="cpu")
+    np.testing.assert_allclose(m.get_vector("w007"),
+                               syn0_bf[7].float().numpy(), rtol=1e-6)
