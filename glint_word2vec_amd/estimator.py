@@ -427,10 +427,20 @@ class GlintWord2VecModel:
                 break
         return out
 
-    def find_synonyms_batch(self, queries, num: int):
+    def find_synonyms_batch(self, queries, num: int,
+                            max_score_bytes: int = 4 << 30):
         """Batched multi-query findSynonyms: one GEMM over all queries +
         one top-k (GPU when to_gpu() was called, else BLAS on host).
-        Returns a list of (word, cosine) lists, query word excluded."""
+        Returns a list of (word, cosine) lists, query word excluded.
+        Chunks the query batch so the [q, vocab] score tensor stays
+        under ``max_score_bytes``."""
+        q_chunk = max(1, int(max_score_bytes // (max(self.num_words, 1) * 4)))
+        if len(queries) > q_chunk:
+            out = []
+            for i in range(0, len(queries), q_chunk):
+                out.extend(self.find_synonyms_batch(
+                    queries[i:i + q_chunk], num, max_score_bytes))
+            return out
         Q = len(queries)
         vecs = np.empty((Q, self.vector_size), dtype=np.float32)
         skip = []

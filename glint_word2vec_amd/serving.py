@@ -258,11 +258,23 @@ class ShardedWord2VecModel:
         allgather merge (mllib:554-630; the multiply op at :598)."""
         return self.find_synonyms_batch([word_or_vec], num)[0]
 
-    def find_synonyms_batch(self, queries: Sequence, num: int
+    def find_synonyms_batch(self, queries: Sequence, num: int,
+                            max_score_bytes: int = 4 << 30
                             ) -> List[List[tuple]]:
-        """Batched multi-query findSynonyms: ONE sharded GEMM over all
-        queries + one topk per shard + one allgather (serving throughput
-        path; benchmarks/serving_probe.py)."""
+        """Batched multi-query findSynonyms: sharded GEMM + topk per
+        shard + one allgather (serving throughput path;
+        benchmarks/serving_probe.py).  Internally chunks the query batch
+        so the [q, shard] f32 score tensor stays under
+        ``max_score_bytes`` (an 80M-row shard at Q=4096 would otherwise
+        materialise 1.3 TB)."""
+        shard_rows = max(int(self.shard.shape[0]), 1)
+        q_chunk = max(1, int(max_score_bytes // (shard_rows * 4)))
+        if len(queries) > q_chunk:
+            out: List[List[tuple]] = []
+            for i in range(0, len(queries), q_chunk):
+                out.extend(self.find_synonyms_batch(
+                    queries[i:i + q_chunk], num, max_score_bytes))
+            return out
         Q = len(queries)
         skip_words = []
         vecs = np.empty((Q, self.dim), dtype=np.float32)
