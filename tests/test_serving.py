@@ -199,3 +199,19 @@ def test_sharded_load_bf16_checkpoint(tmp_path):
     m = ShardedWord2VecModel.load(str(path), device="cpu")
     np.testing.assert_allclose(m.get_vector("w007"),
                                syn0_bf[7].float().numpy(), rtol=1e-6)
+
+
+def test_find_synonyms_batch_chunking_equivalence(tmp_path):
+    """Tiny max_score_bytes forces internal query chunking; results must
+    equal the unchunked path (both sharded and dense models)."""
+    path = str(tmp_path / "model")
+    _, syn0, _ = _make_checkpoint(path, num_shards=2)
+    qs = [syn0[3] * 1.5, syn0[11], syn0[20] * 0.2, syn0[33]]
+    sharded = GlintWord2VecModel.load_sharded(path, device="cpu")
+    a = sharded.find_synonyms_batch(qs, 5)
+    b = sharded.find_synonyms_batch(qs, 5, max_score_bytes=1)
+    assert a == b
+    dense = GlintWord2VecModel.load(path)
+    c = dense.find_synonyms_batch(qs, 5)
+    d = dense.find_synonyms_batch(qs, 5, max_score_bytes=1)
+    assert [[w for w, _ in r] for r in c] == [[w for w, _ in r] for r in d]
