@@ -57,6 +57,9 @@ def _add_train(sub):
                    help="row-update mode (hybrid: atomics on the hot "
                         "Zipf band, hogwild elsewhere — the measured "
                         "quality/speed default)")
+    p.add_argument("--shared-negatives", action="store_true",
+                   help="one negative set per position (HogBatch-style; "
+                        "4.6x at high negative counts, see results.md)")
 
 
 def main(argv=None):
@@ -96,7 +99,8 @@ def main(argv=None):
             dtype=args.dtype, device=args.device, engine=args.engine,
             window_mode=args.window_mode, num_partitions=args.workers,
             checkpoint_every=args.checkpoint_every,
-            update_mode=args.updates)
+            update_mode=args.updates,
+            shared_negatives=args.shared_negatives)
         model = est.fit(args.corpus,
                         save_path=(args.model if args.checkpoint_every
                                    else None),
