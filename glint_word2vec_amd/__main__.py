@@ -86,6 +86,14 @@ def main(argv=None):
                         "matrix; torchrun for multi-GPU)")
     p = sub.add_parser("info", help="print model metadata")
     p.add_argument("model")
+    p = sub.add_parser("serve", help="HTTP serving endpoint (FastAPI)")
+    p.add_argument("model")
+    p.add_argument("--host", default="127.0.0.1")
+    p.add_argument("--port", type=int, default=8000)
+    p.add_argument("--sharded", action="store_true",
+                   help="stream the model to device shards (torchrun for "
+                        "multi-GPU; every rank runs the app)")
+    p.add_argument("--device", default=None)
     args = ap.parse_args(argv)
 
     if args.cmd == "train":
@@ -130,6 +138,10 @@ def main(argv=None):
             model = GlintWord2VecModel.load(args.model)
             model.to_local().save(args.out)
         print(f"wrote {args.out}")
+    elif args.cmd == "serve":
+        from .server import serve
+        serve(args.model, host=args.host, port=args.port,
+              sharded=args.sharded, device=args.device)
     elif args.cmd == "info":
         import json
         import os
