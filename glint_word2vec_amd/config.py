@@ -46,11 +46,21 @@ DP_MAX_TABLE_BYTES = 4 << 30
 def choose_engine(vocab_size: int, dim: int, dtype_bytes: int,
                   world: int) -> str:
     """Auto engine policy, shared by engine.py and bench.py (both use the
-    padded kernel stride so the estimate matches the real allocation)."""
+    padded kernel stride so the estimate matches the real allocation).
+
+    world 1 -> fused.  Small tables -> dp (replicated + overlapped
+    delta-allreduce; per-GPU fused rate).  Beyond the dp merge budget ->
+    ROW-sharded (the north-star PS shape): data-parallel, per-rank pull
+    cycle measured at 63M words/s at vocab 10M / 33M at 80M (world-1
+    proxy, benchmarks/results.md) -> ~8x that per node, vs the
+    dim-sharded engine whose ranks all walk the SAME batch (~230M/node
+    ceiling at the 8-GPU slice shape).  dim remains selectable: its
+    traffic is dimension-independent (~4 B/pair), the right tool if
+    xGMI, not HBM, turns out to bind."""
     if world == 1:
         return "fused"
     table_bytes = 2 * vocab_size * round_stride_py(dim) * dtype_bytes
-    return "dp" if table_bytes <= DP_MAX_TABLE_BYTES else "dim"
+    return "dp" if table_bytes <= DP_MAX_TABLE_BYTES else "row"
 
 
 @dataclass

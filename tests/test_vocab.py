@@ -89,6 +89,6 @@ def test_choose_engine_policy():
     from glint_word2vec_amd.config import choose_engine, round_stride_py
     assert choose_engine(1_000_000, 300, 2, 1) == "fused"
     assert choose_engine(1_000_000, 300, 2, 8) == "dp"
-    assert choose_engine(10_000_000, 300, 2, 8) == "dim"
+    assert choose_engine(10_000_000, 300, 2, 8) == "row"
     assert round_stride_py(300) == 320
     assert round_stride_py(1024) == 1024
