@@ -138,7 +138,8 @@ class Word2VecConfig:
     # None (default) leaves update_mode in charge.
     atomic_updates: Optional[bool] = None
     # --- multi-GPU engine --------------------------------------
-    # "auto": fused single-GPU kernel at world 1; dim-sharded at world > 1.
+    # "auto": fused at world 1; dp while tables are small; row-sharded
+    # beyond (choose_engine above).
     # "dim": dimension-sharded (CIKM scheme, RCCL allreduce of partial dots).
     # "row": row-sharded parameter-server shape (RCCL alltoallv pull/push).
     # "dp": replicated tables + periodic delta-allreduce (fused kernel per
