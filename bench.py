@@ -186,6 +186,14 @@ def main():
         dist_mode = "dp"
     elif engine == "row":
         from glint_word2vec_amd.parallel.row_sharded import RowShardedSgns
+        est_unique = min(args.vocab,
+                         args.words_per_step * (args.neg + 1) * 6)
+        cache_gb = 8 * est_unique * (args.dim + 64) * \
+            (2 if args.dtype == "bf16" else 4) / 2 ** 30
+        if cache_gb > 200 and rank == 0:
+            print(f"# warning: row-engine pull caches may need ~"
+                  f"{cache_gb:.0f} GB/GPU at this vocab/step size; "
+                  f"reduce --words-per-step", file=sys.stderr)
         trainer = RowShardedSgns(args.vocab, args.dim, dtype=dtype,
                                  device=str(device), seed=1, counts=counts,
                                  table_size=args.table_size,

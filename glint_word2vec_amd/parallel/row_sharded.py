@@ -23,6 +23,12 @@ x dim, so it is the engine of choice when the touched-row set is sparse
 the dense case with dimension-independent traffic.
 
 Works on CUDA (HIP train_pairs kernels) and CPU (C++ twin; gloo tests).
+
+Memory: the pull cycle stages ~4 cache-sized buffers per in-flight step
+(cache+orig per table) of unique_rows x stride elements, and the
+pipelined loop keeps two steps in flight — at very large vocabularies
+size words_per_step so 8 x unique_rows x stride x dtype fits HBM
+(e.g. 80M vocab / 1M-word steps ~= 90 GB; bench warns beyond 200 GB).
 """
 from __future__ import annotations
 
