@@ -91,3 +91,34 @@ def test_estimator_empty_corpus():
     est.config.device = "cpu"
     with pytest.raises(ValueError, match="empty vocabulary"):
         est.fit([["rare", "words", "only"]])
+
+
+def test_row_cpu_shared_negatives_raises(tmp_path):
+    """shared_negatives needs the counter-RNG planner; the CPU row engine
+    must refuse loudly rather than silently train different semantics."""
+    import numpy as np
+    import pytest
+    from glint_word2vec_amd.parallel.row_sharded import RowShardedSgns
+    eng = RowShardedSgns(50, 8, device="cpu", seed=1, table_size=101,
+                         shared_negatives=True)
+    with pytest.raises(NotImplementedError):
+        eng.make_plan(np.zeros(10, dtype=np.int32),
+                      np.array([0, 10], dtype=np.int32), 2, 3,
+                      np.random.default_rng(0))
+
+
+def test_sharded_load_missing_files(tmp_path):
+    import pytest
+    from glint_word2vec_amd.serving import ShardedWord2VecModel
+    with pytest.raises(FileNotFoundError):
+        ShardedWord2VecModel.load(str(tmp_path / "nope"), device="cpu")
+
+
+def test_config_rejects_bad_values():
+    import pytest
+    from glint_word2vec_amd.config import Word2VecConfig
+    for kw in ({"update_mode": "x"}, {"hybrid_hot_rows": -1},
+               {"hybrid_skip_rows": -2}, {"engine": "zz"},
+               {"vector_size": 0}):
+        with pytest.raises(ValueError):
+            Word2VecConfig(**kw)
