@@ -70,7 +70,7 @@ def main():
                     choices=["hogwild", "atomic", "hybrid", "positives"],
                     default="hogwild")
     ap.add_argument("--shared-negatives", action="store_true")
-    ap.add_argument("--hot-floor", type=int, default=64,
+    ap.add_argument("--hot-floor", type=int, default=16,
                     help="hybrid: rows < F stay hogwild (contention escape)")
     ap.add_argument("--hot-rows", type=int, default=32768,
                     help="hybrid: atomics for rows < K (word id ~ 2*Zipf "

@@ -130,10 +130,13 @@ class Word2VecConfig:
     # hybrid: rows < this stay hogwild even inside the atomic head — the
     # ultra-hot top rows take a double-digit share of all negative-table
     # draws, and atomics there serialize on a handful of cachelines (the
-    # measured hybrid cliff).  Round-2 sweep at vocab 1M: floor 64 gives
-    # planted-NN 0.872 vs 0.893 full-atomic and 0.861 at floor 128, all
-    # at ~171M words/s (benchmarks/results.md).
-    hybrid_skip_rows: int = 64
+    # measured hybrid cliff: floor 0 runs 65M words/s).  Round-2 final
+    # sweep at vocab 1M / 400M words (2 runs each): floor 16 -> 0.899
+    # planted-NN at 157M words/s (within 0.03 of full atomic's 0.927 at
+    # 2.1x its speed); 32 -> 0.882 @ 165M; 64 -> 0.861 @ 171M
+    # (benchmarks/results.md).  Default 16: the quality-defensible point;
+    # raise it for more speed.
+    hybrid_skip_rows: int = 16
     # Deprecated alias (round-1 API): True -> "atomic", False -> "hogwild".
     # None (default) leaves update_mode in charge.
     atomic_updates: Optional[bool] = None

@@ -69,8 +69,8 @@ def test_update_mode_resolution():
     c = Word2VecConfig()
     assert c.update_mode == "hybrid"
     assert c.effective_atomic_below() == c.hybrid_hot_rows == 32768
-    assert c.effective_atomic_floor() == 64
-    assert c.effective_atomic_floor(1_000_000) == 64
+    assert c.effective_atomic_floor() == 16
+    assert c.effective_atomic_floor(1_000_000) == 16
     assert c.effective_atomic_floor(3611) == 3      # small-vocab scaling
     c2 = Word2VecConfig(update_mode="hogwild")
     assert c2.effective_atomic_below() == 0
