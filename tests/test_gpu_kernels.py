@@ -379,6 +379,11 @@ def test_estimator_german_corpus_gpu():
     assert 3500 < m.num_words < 3700
     res = m.analogy(["wien", "deutschland"], ["österreich"], 10)
     assert "berlin" in [w for w, _ in res]
+    # the reference IT gate asserts wien in top-10 of österreich with
+    # cosine > 0.9 (Spec:301); measured ~0.99 here
+    syn = m.find_synonyms("österreich", 10)
+    assert "wien" in [w for w, _ in syn]
+    assert all(c > 0.9 for _, c in syn)
 
 
 def test_gpu_streaming_checkpoint_matches_to_host(tmp_path_factory):
