@@ -743,6 +743,11 @@ class RowShardedSgns:
         reads it directly."""
         import json
         import os
+        if self.is_cuda:
+            # quiesce in-flight comm-stream pushes/pulls so the D2H
+            # snapshot reads a settled table (mid-training checkpoints
+            # run inside the pipelined loop)
+            torch.cuda.synchronize(self.device)
         if self.rank == 0:
             os.makedirs(os.path.join(path, "shards"), exist_ok=True)
             import time as _t
