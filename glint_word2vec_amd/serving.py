@@ -323,6 +323,12 @@ class ShardedWord2VecModel:
             results.append(out)
         return results
 
+    def find_synonyms_df(self, word_or_vec, num: int):
+        """DataFrame variant (ml:390-420 parity, like the dense model)."""
+        import pandas as pd
+        return pd.DataFrame(self.find_synonyms(word_or_vec, num),
+                            columns=["word", "similarity"])
+
     def analogy(self, pos: List[str], neg: List[str],
                 num: int = 10) -> List[tuple]:
         v = np.zeros(self.dim, dtype=np.float32)

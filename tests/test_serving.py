@@ -66,6 +66,8 @@ def test_sharded_load_world1_matches_dense(tmp_path, num_shards):
     np.testing.assert_allclose(got[2], 0.0)
     # analogy runs
     assert len(sharded.analogy(["w001", "w002"], ["w003"], 3)) == 3
+    df = sharded.find_synonyms_df("w005", 4)
+    assert list(df.columns) == ["word", "similarity"] and len(df) == 4
 
 
 def test_sharded_load_row_range_layout(tmp_path):
