@@ -11,8 +11,14 @@ from .config import Word2VecConfig
 from .estimator import GlintWord2Vec, GlintWord2VecModel, LocalWord2VecModel
 from .vocab import Vocabulary, build_vocab
 
+
+def load_sharded(path, device="auto", dtype="auto"):
+    """Convenience: GlintWord2VecModel.load_sharded (streaming,
+    device-resident serving model)."""
+    return GlintWord2VecModel.load_sharded(path, device=device, dtype=dtype)
+
 __version__ = "0.1.0"
 __all__ = [
     "Word2VecConfig", "GlintWord2Vec", "GlintWord2VecModel",
-    "LocalWord2VecModel", "Vocabulary", "build_vocab",
+    "LocalWord2VecModel", "Vocabulary", "build_vocab", "load_sharded",
 ]

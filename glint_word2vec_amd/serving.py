@@ -247,6 +247,20 @@ class ShardedWord2VecModel:
     # alias matching GlintWord2VecModel's GPU serving name
     transform_sentences_gpu = transform_sentences
 
+    def transform(self, x, input_col: str = "sentence",
+                  output_col: str = "vector"):
+        """Drop-in transform (the dense model's polymorphic surface,
+        ml:432-460): word -> vector; token sequence -> average vector;
+        DataFrame -> copy with an output column of averaged vectors."""
+        if isinstance(x, str):
+            return self.get_vector(x)
+        if hasattr(x, "columns"):   # DataFrame
+            col = input_col if input_col in x.columns else x.columns[0]
+            out = x.copy()
+            out[output_col] = list(self.transform_sentences(list(x[col])))
+            return out
+        return self.transform_sentences([list(x)])[0]
+
     def _query_vec(self, word_or_vec) -> tuple:
         if isinstance(word_or_vec, str):
             vec = self.get_vector(word_or_vec)

@@ -220,3 +220,19 @@ def test_find_synonyms_batch_chunking_equivalence(tmp_path):
     c = dense.find_synonyms_batch(qs, 5)
     d = dense.find_synonyms_batch(qs, 5, max_score_bytes=1)
     assert [[w for w, _ in r] for r in c] == [[w for w, _ in r] for r in d]
+
+
+def test_sharded_transform_polymorphic(tmp_path):
+    import pandas as pd
+    path = str(tmp_path / "model")
+    _, syn0, _ = _make_checkpoint(path, num_shards=2)
+    import glint_word2vec_amd as g
+    m = g.load_sharded(path, device="cpu")
+    np.testing.assert_allclose(m.transform("w004"), syn0[4], rtol=1e-6)
+    np.testing.assert_allclose(m.transform(["w001", "w002"]),
+                               syn0[[1, 2]].mean(0), rtol=1e-5, atol=1e-6)
+    df = pd.DataFrame({"sentence": [["w001", "w002"], ["w010"]],
+                       "other": [1, 2]})
+    out = m.transform(df)
+    assert list(out.columns) == ["sentence", "other", "vector"]
+    np.testing.assert_allclose(out["vector"][1], syn0[10], rtol=1e-6)
