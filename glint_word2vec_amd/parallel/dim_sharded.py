@@ -179,7 +179,8 @@ class DimShardedSgns:
                 # reduces EXACTLY to the fused single-kernel form —
                 # dot + sigmoid + update in one pass, zero redundant row
                 # reads (the "single-pass variant", VERDICT round-1 #6;
-                # 105.8M -> fused-rate ~190M words/s).  The phase pipeline
+                # 105.8M -> fused-rate, ~169M at the final hybrid default).
+                # The phase pipeline
                 # stays in use for serial parity tests and world > 1.
                 self._train_step_gpu_fused(tokens, offsets, alpha, window,
                                            n_neg, seed, sent_id_base)
