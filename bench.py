@@ -64,10 +64,11 @@ def parse_args():
                         "target-row traffic ~(1+n)/(1+n/2b)-fold at high "
                         "n_neg — the dim-1024 neg-25 config)")
     p.add_argument("--pair-mode", type=int, default=None, choices=[0, 1, 2, 3],
-                   help="fused-kernel variant: 0=one pair/wave (64-lane), "
-                        "1=two pairs (32-lane halves, default), "
-                        "2=four pairs (16-lane quarters), "
-                        "3=two pairs software-pipelined 2-deep")
+                   help="fused-kernel variant: 0=one pair/wave (64-lane; "
+                        "auto default for stride>512), 1=two pairs "
+                        "(32-lane halves), 2=four pairs (16-lane), "
+                        "3=two pairs software-pipelined 2-deep (auto "
+                        "default for stride<=512)")
     p.add_argument("--profile-steps", type=int, default=0,
                    help="run only this many steps, no warmup JSON (rocprof)")
     p.add_argument("--device", choices=["cuda", "cpu"], default="cuda",
@@ -75,7 +76,8 @@ def parse_args():
     p.add_argument("--engine", choices=["auto", "fused", "dim", "row", "dp"],
                    default="auto",
                    help="auto: fused at world 1; beyond: dp (replicated + "
-                        "delta-allreduce) when tables fit, else dim-sharded")
+                        "overlapped delta-allreduce) while tables are "
+                        "small, else row-sharded (alltoallv pull/push)")
     p.add_argument("--sync-every", type=int, default=4,
                    help="dp engine: steps between delta merges")
     p.add_argument("--chunk-words", type=int, default=1 << 20,
