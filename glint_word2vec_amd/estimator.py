@@ -330,6 +330,7 @@ class GlintWord2VecModel:
         self._gpu = GpuSgns(self.num_words, self.vector_size, dtype=dtype,
                             device=device, syn0_host=self._f32())
         self._norms = None
+        self._norms_t = None
         return self
 
     def transform_sentences_gpu(self, sentences) -> np.ndarray:
@@ -394,8 +395,10 @@ class GlintWord2VecModel:
             # hipGraph replay once multiply() became a hand-written
             # kernel (2.9k vs 2.4k q/s — the graph re-runs the query
             # staging copies; synonyms_query remains available)
-            norms_t = self._gpu.norms().clamp_min(1e-12)
-            cos_t = self._gpu.multiply(torch.from_numpy(vec), norms=norms_t)
+            if getattr(self, "_norms_t", None) is None:
+                self._norms_t = self._gpu.norms().clamp_min(1e-12)
+            cos_t = self._gpu.multiply(torch.from_numpy(vec),
+                                       norms=self._norms_t)
             k = min(num + 1, self.num_words)
             val, idx = torch.topk(cos_t, k)
             val = val.cpu().numpy()
