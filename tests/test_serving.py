@@ -236,3 +236,37 @@ def test_sharded_transform_polymorphic(tmp_path):
     out = m.transform(df)
     assert list(out.columns) == ["sentence", "other", "vector"]
     np.testing.assert_allclose(out["vector"][1], syn0[10], rtol=1e-6)
+
+
+def _serve_worker_small(rank, world, rdv, path, out_dir):
+    import torch.distributed as dist
+    dist.init_process_group("gloo", rank=rank, world_size=world,
+                            init_method=f"file://{rdv}")
+    try:
+        m = GlintWord2VecModel.load_sharded(path, device="cpu")
+        out = {"syn": m.find_synonyms("w005", 5),
+               "vec": m.get_vector("w010").tolist(),
+               "avg": m.transform_sentences([["w001", "w002"]]).tolist()}
+        with open(os.path.join(out_dir, f"serve3_{rank}.json"), "w") as f:
+            json.dump(out, f)
+    finally:
+        dist.destroy_process_group()
+
+
+def test_sharded_serving_world3_uneven(tmp_path):
+    """World 3 over a 2-shard checkpoint: uneven per-rank row counts, all
+    ranks agree and match world-1."""
+    path = str(tmp_path / "model")
+    _, syn0, _ = _make_checkpoint(path, num_shards=2, vocab=31)  # 31 % 3 != 0
+    rdv = str(tmp_path / "rdv3")
+    mp.spawn(_serve_worker_small, args=(3, rdv, path, str(tmp_path)),
+             nprocs=3, join=True)
+    outs = []
+    for r in range(3):
+        with open(tmp_path / f"serve3_{r}.json") as f:
+            outs.append(json.load(f))
+    assert outs[0] == outs[1] == outs[2]
+    np.testing.assert_allclose(outs[0]["vec"], syn0[10], rtol=1e-5)
+    one = GlintWord2VecModel.load_sharded(path, device="cpu")
+    ref = one.find_synonyms("w005", 5)
+    assert [w for w, _ in ref] == [w for w, _ in outs[0]["syn"]]
