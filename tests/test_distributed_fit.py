@@ -185,3 +185,16 @@ def test_dp_async_sync_applies_exactly_once(tmp_path):
     a = np.load(tmp_path / "dpcons_0.npy")
     b = np.load(tmp_path / "dpcons_1.npy")
     np.testing.assert_allclose(a, b)
+
+
+def test_row_fit_world3(tmp_path):
+    """Pipelined row engine at world 3 (odd world, uneven batch counts):
+    all ranks converge to identical assembled models."""
+    rdv = str(tmp_path / "rdv_row3")
+    mp.spawn(_worker, args=(3, rdv, str(tmp_path), "row"), nprocs=3,
+             join=True)
+    s0 = np.load(tmp_path / "syn0_0.npy")
+    for r in (1, 2):
+        np.testing.assert_allclose(s0, np.load(tmp_path / f"syn0_{r}.npy"),
+                                   rtol=1e-5, atol=1e-7)
+    assert np.isfinite(s0).all()
