@@ -127,10 +127,22 @@ def load_model(path: str) -> Tuple[Word2VecConfig, Vocabulary, np.ndarray, np.nd
     with open(os.path.join(path, "shards", "index.json")) as f:
         index = json.load(f)
     V, D, k = index["vocab"], index["dim"], index["num_shards"]
-    dt = np.dtype(index["dtype"])
+    # "bfloat16" shards are raw u16 (half the disk of f32); decode to f32
+    is_bf16 = index["dtype"] == "bfloat16"
+    dt = np.uint16 if is_bf16 else np.dtype(index["dtype"])
+    out_dt = np.float32 if is_bf16 else dt
     layout = index.get("layout", "row_mod")
-    syn0 = np.empty((V, D), dtype=dt)
-    syn1 = np.empty((V, D), dtype=dt) if index.get("has_syn1") else None
+    syn0 = np.empty((V, D), dtype=out_dt)
+    syn1 = np.empty((V, D), dtype=out_dt) if index.get("has_syn1") else None
+
+    def _decode(buf, rows):
+        block = buf.reshape(len(rows), D)
+        if is_bf16:
+            import torch
+            block = torch.from_numpy(block).view(torch.bfloat16) \
+                .float().numpy()
+        return block
+
     for s in range(k):
         if layout == "row_mod":
             rows = _shard_rows(V, s, k)
@@ -138,10 +150,10 @@ def load_model(path: str) -> Tuple[Word2VecConfig, Vocabulary, np.ndarray, np.nd
             b = index["bounds"]
             rows = np.arange(b[s], b[s + 1], dtype=np.int64)
         buf = np.fromfile(os.path.join(path, "shards", f"syn0-{s:05d}.bin"), dtype=dt)
-        syn0[rows] = buf.reshape(len(rows), D)
+        syn0[rows] = _decode(buf, rows)
         if syn1 is not None:
             buf = np.fromfile(os.path.join(path, "shards", f"syn1-{s:05d}.bin"), dtype=dt)
-            syn1[rows] = buf.reshape(len(rows), D)
+            syn1[rows] = _decode(buf, rows)
     return config, vocab, syn0, syn1
 
 

@@ -189,3 +189,33 @@ def test_fit_resume_dim_engine(tmp_path):
     v2 = m2.to_local().vectors
     assert np.isfinite(v2).all()
     assert not np.array_equal(m1.to_local().vectors, v2)
+
+
+def test_dense_load_bf16_checkpoint(tmp_path):
+    """load_model decodes bfloat16-on-disk shards (raw u16) to f32."""
+    import json
+    import os
+    import numpy as np
+    import torch
+    from glint_word2vec_amd.checkpoint import load_model
+    from glint_word2vec_amd.config import Word2VecConfig
+    vocab, dim = 20, 8
+    rng = np.random.default_rng(2)
+    syn0_bf = torch.from_numpy(
+        rng.standard_normal((vocab, dim)).astype(np.float32)).bfloat16()
+    path = tmp_path / "bf"
+    os.makedirs(path / "shards")
+    with open(path / "metadata", "w") as f:
+        json.dump({"numWords": vocab, "vectorSize": dim,
+                   "paramMap": Word2VecConfig(vector_size=dim).to_dict()}, f)
+    with open(path / "words", "w") as f:
+        f.writelines(f"w{i}\n" for i in range(vocab))
+    with open(path / "shards" / "index.json", "w") as f:
+        json.dump({"num_shards": 1, "vocab": vocab, "dim": dim,
+                   "dtype": "bfloat16", "layout": "row_mod",
+                   "has_syn1": False}, f)
+    syn0_bf.view(torch.uint16).numpy().tofile(
+        path / "shards" / "syn0-00000.bin")
+    _, _, s0, s1 = load_model(str(path))
+    assert s1 is None and s0.dtype == np.float32
+    np.testing.assert_allclose(s0, syn0_bf.float().numpy(), rtol=1e-6)
