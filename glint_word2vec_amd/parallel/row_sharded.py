@@ -794,7 +794,8 @@ class RowShardedSgns:
                 f"engine is vocab={self.vocab_size} dim={self.dim}")
         if not index.get("has_syn1", False):
             raise ValueError("checkpoint has no syn1 — cannot resume")
-        dt = np.dtype(index["dtype"])
+        is_bf16_file = index["dtype"] == "bfloat16"
+        dt = np.uint16 if is_bf16_file else np.dtype(index["dtype"])
         D = self.dim
         for t, name in ((self.syn0, "syn0"), (self.syn1, "syn1")):
             for s in range(index["num_shards"]):
@@ -810,8 +811,12 @@ class RowShardedSgns:
                         mine = (gids % self.world) == self.rank
                         if not mine.any():
                             continue
-                        rows = torch.from_numpy(np.ascontiguousarray(
-                            block[mine], dtype=np.float32))
+                        picked = np.ascontiguousarray(block[mine])
+                        rows = (torch.from_numpy(picked)
+                                .view(torch.bfloat16).float()
+                                if is_bf16_file else
+                                torch.from_numpy(picked.astype(
+                                    np.float32, copy=False)))
                         t[torch.from_numpy(gids[mine] // self.world),
                           :D] = rows.to(t.dtype).to(self.device)
 
